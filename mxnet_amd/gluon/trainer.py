@@ -1,0 +1,185 @@
+"""Gluon Trainer (reference python/mxnet/gluon/trainer.py).
+
+step(batch_size) = allreduce_grads (KVStore pushpull, per-parameter,
+priority = -index so late layers' gradients — produced first by backward —
+communicate first, reference trainer.py:385-409) + update (fused optimizer
+step per device).  In the one-process-per-GPU RCCL layout the pushpull is
+an async all-reduce launched in backward-completion order, overlapping the
+remaining backward exactly like the reference's kGPUPrioritized engine ops.
+"""
+import torch
+
+from ..parallel import kvstore as kvs_mod
+from .. import optimizer as opt_mod
+from ..ndarray.ndarray import NDArray
+from .parameter import ParameterDict
+
+
+class Trainer:
+    def __init__(self, params, optimizer, optimizer_params=None, kvstore='device',
+                 compression_params=None, update_on_kvstore=None):
+        if isinstance(params, (dict, ParameterDict)):
+            params = list(params.values())
+        self._params = [p for p in params if p.grad_req != 'null']
+        self._scale = 1.0
+        param_dict = {i: p for i, p in enumerate(self._params)}
+        if isinstance(optimizer, opt_mod.Optimizer):
+            self._optimizer = optimizer
+            self._optimizer.param_dict = param_dict
+        else:
+            self._optimizer = opt_mod.create(
+                optimizer, param_dict=param_dict, **(optimizer_params or {}))
+        self._states = [None] * len(self._params)
+        self._states_init = [False] * len(self._params)
+        self._kvstore = None
+        self._kv_initialized = False
+        self._update_on_kvstore = update_on_kvstore
+        self._kvstore_kind = kvstore
+        self._distributed = False
+
+    # ------------------------------------------------------------------
+    @property
+    def optimizer(self):
+        return self._optimizer
+
+    @property
+    def learning_rate(self):
+        return self._optimizer.learning_rate
+
+    def set_learning_rate(self, lr):
+        self._optimizer.set_learning_rate(lr)
+
+    def _init_kvstore(self):
+        if self._kv_initialized:
+            return
+        kind = self._kvstore_kind
+        if kind is None:
+            self._kvstore = None
+        elif isinstance(kind, kvs_mod.KVStoreBase):
+            self._kvstore = kind
+        else:
+            self._kvstore = kvs_mod.create(kind)
+        self._distributed = isinstance(self._kvstore, kvs_mod.DistKVStore)
+        if self._distributed:
+            # sync initial parameters across ranks (reference: kv.init
+            # broadcasts rank-0 values)
+            for i, p in enumerate(self._params):
+                for d in p.list_data():
+                    self._kvstore.broadcast(i, d, d)
+        self._kv_initialized = True
+
+    def _check_states(self, i, p):
+        if not self._states_init[i]:
+            w = p.list_data()[0]
+            self._states[i] = self._optimizer.create_state_multi_precision(i, w)
+            self._states_init[i] = True
+
+    # ------------------------------------------------------------------
+    def step(self, batch_size, ignore_stale_grad=False):
+        self._init_kvstore()
+        self._optimizer.rescale_grad = self._scale / batch_size
+        self._allreduce_grads()
+        self._update(ignore_stale_grad)
+
+    def allreduce_grads(self):
+        self._init_kvstore()
+        self._allreduce_grads()
+
+    def _allreduce_grads(self):
+        if self._kvstore is None:
+            return
+        if self._distributed:
+            # late layers first (their grads were computed first)
+            handles = []
+            for i in reversed(range(len(self._params))):
+                grads = self._params[i].list_grad()
+                h = self._kvstore.pushpull(i, grads[0], priority=-i,
+                                           async_op=True)
+                handles.append(h)
+            for h in handles:
+                if h is not None:
+                    h.wait()
+            # mean over workers
+            n = self._kvstore.num_workers
+            if n > 1:
+                with torch.no_grad():
+                    for p in self._params:
+                        for g in p.list_grad():
+                            g._t.div_(n)
+            return
+        # single-process multi-device: reduce over devices then broadcast
+        for i, p in enumerate(self._params):
+            grads = p.list_grad()
+            if len(grads) > 1 or isinstance(self._kvstore, kvs_mod.KVStore):
+                self._kvstore.pushpull(i, grads, out=grads, priority=-i)
+
+    def update(self, batch_size, ignore_stale_grad=False):
+        self._init_kvstore()
+        self._optimizer.rescale_grad = self._scale / batch_size
+        self._update(ignore_stale_grad)
+
+    def _update(self, ignore_stale_grad=False):
+        for i, p in enumerate(self._params):
+            self._check_states(i, p)
+            if len(p.list_data()) == 1:
+                self._optimizer.update_multi_precision(
+                    i, p.list_data()[0], p.list_grad()[0], self._states[i])
+            else:
+                # replicated params: update each device copy with the
+                # already-reduced gradient (identical results)
+                for w, g in zip(p.list_data(), p.list_grad()):
+                    self._optimizer.update_multi_precision(i, w, g, self._states[i])
+            # grad_req='write' semantics: grads are consumed by the step
+            # (torch accumulates, the reference overwrites — clearing here
+            # restores reference behavior; 'add' keeps accumulating)
+            if p.grad_req == 'write':
+                p.zero_grad()
+
+    # -- AMP hook (loss scaler rescales via _scale) ----------------------
+    @property
+    def _amp_loss_scale(self):
+        return self._scale
+
+    def _set_scale(self, scale):
+        self._scale = scale
+
+    # ------------------------------------------------------------------
+    def save_states(self, fname):
+        import pickle
+        cpu_states = []
+        for s in self._states:
+            cpu_states.append(_state_to_cpu(s))
+        with open(fname, 'wb') as f:
+            pickle.dump({'states': cpu_states,
+                         'num_update': self._optimizer.num_update}, f)
+
+    def load_states(self, fname):
+        import pickle
+        with open(fname, 'rb') as f:
+            blob = pickle.load(f)
+        self._init_kvstore()
+        for i, p in enumerate(self._params):
+            self._check_states(i, p)
+        dev_states = []
+        for s, p in zip(blob['states'], self._params):
+            dev = p.list_data()[0]._t.device
+            dev_states.append(_state_to_device(s, dev))
+        self._states = dev_states
+        self._states_init = [True] * len(self._params)
+        self._optimizer.num_update = blob['num_update']
+
+
+def _state_to_cpu(s):
+    if isinstance(s, torch.Tensor):
+        return s.cpu()
+    if isinstance(s, tuple):
+        return tuple(_state_to_cpu(x) for x in s)
+    return s
+
+
+def _state_to_device(s, dev):
+    if isinstance(s, torch.Tensor):
+        return s.to(dev)
+    if isinstance(s, tuple):
+        return tuple(_state_to_device(x, dev) for x in s)
+    return s

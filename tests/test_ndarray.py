@@ -1,0 +1,115 @@
+import numpy as np
+import pytest
+
+import mxnet_amd as mx
+from mxnet_amd import nd
+
+
+def test_create_and_convert():
+    a = nd.array([[1, 2], [3, 4]])
+    assert a.shape == (2, 2)
+    assert a.dtype == np.float32
+    assert np.allclose(a.asnumpy(), [[1, 2], [3, 4]])
+    b = nd.array(np.arange(6, dtype='int32').reshape(2, 3))
+    assert b.dtype == np.int32
+
+
+def test_creation_ops():
+    assert nd.zeros((2, 3)).asnumpy().sum() == 0
+    assert nd.ones((2, 3)).asnumpy().sum() == 6
+    assert nd.full((2, 2), 7).asnumpy().sum() == 28
+    assert np.allclose(nd.arange(5).asnumpy(), np.arange(5))
+
+
+def test_arithmetic():
+    a = nd.array([1.0, 2.0, 3.0])
+    b = nd.array([4.0, 5.0, 6.0])
+    assert np.allclose((a + b).asnumpy(), [5, 7, 9])
+    assert np.allclose((a - b).asnumpy(), [-3, -3, -3])
+    assert np.allclose((a * b).asnumpy(), [4, 10, 18])
+    assert np.allclose((b / a).asnumpy(), [4, 2.5, 2])
+    assert np.allclose((a ** 2).asnumpy(), [1, 4, 9])
+    assert np.allclose((2 + a).asnumpy(), [3, 4, 5])
+    a += 1
+    assert np.allclose(a.asnumpy(), [2, 3, 4])
+
+
+def test_reductions():
+    a = nd.array(np.arange(12, dtype='float32').reshape(3, 4))
+    assert a.sum().asscalar() == 66
+    assert np.allclose(a.sum(axis=0).asnumpy(), [12, 15, 18, 21])
+    assert np.allclose(a.mean(axis=1).asnumpy(), [1.5, 5.5, 9.5])
+    assert a.max().asscalar() == 11
+    assert a.argmax(axis=1).asnumpy().tolist() == [3, 3, 3]
+
+
+def test_shape_ops():
+    a = nd.array(np.arange(24, dtype='float32').reshape(2, 3, 4))
+    assert a.reshape((6, 4)).shape == (6, 4)
+    assert a.reshape((-1, 4)).shape == (6, 4)
+    assert a.reshape((0, -1)).shape == (2, 12)
+    assert a.transpose().shape == (4, 3, 2)
+    assert a.transpose((1, 0, 2)).shape == (3, 2, 4)
+    assert a.expand_dims(0).shape == (1, 2, 3, 4)
+    assert a.flatten().shape == (2, 12)
+    assert nd.concat(a, a, dim=1).shape == (2, 6, 4)
+    assert nd.stack(a, a, axis=0).shape == (2, 2, 3, 4)
+
+
+def test_indexing():
+    a = nd.array(np.arange(20, dtype='float32').reshape(4, 5))
+    assert a[1, 2].asscalar() == 7
+    assert a[1:3].shape == (2, 5)
+    a[0, :] = 0
+    assert a[0].asnumpy().sum() == 0
+    idx = nd.array([0, 2], dtype='int64')
+    assert nd.take(a, idx).shape == (2, 5)
+
+
+def test_nn_ops_cpu():
+    x = nd.random_uniform(shape=(2, 8))
+    assert np.allclose(nd.softmax(x).asnumpy().sum(axis=1), 1, atol=1e-5)
+    r = nd.relu(nd.array([-1.0, 1.0]))
+    assert np.allclose(r.asnumpy(), [0, 1])
+    s = nd.sigmoid(nd.zeros((3,)))
+    assert np.allclose(s.asnumpy(), 0.5)
+
+
+def test_dtype_cast():
+    a = nd.ones((2, 2))
+    h = a.astype('float16')
+    assert h.dtype == np.float16
+    back = h.astype('float32')
+    assert np.allclose(back.asnumpy(), 1)
+
+
+def test_context():
+    a = nd.ones((2,), ctx=mx.cpu())
+    assert a.context == mx.cpu()
+    b = a.as_in_context(mx.cpu())
+    assert b.context.device_type == 'cpu'
+
+
+def test_broadcast_binary():
+    a = nd.ones((2, 1, 3))
+    b = nd.ones((1, 4, 3))
+    assert (a + b).shape == (2, 4, 3)
+    assert np.allclose(nd.broadcast_maximum(a * 3, b).asnumpy(), 3)
+
+
+def test_where_clip():
+    a = nd.array([-2.0, 0.5, 3.0])
+    assert np.allclose(a.clip(-1, 1).asnumpy(), [-1, 0.5, 1])
+    cond = nd.array([1.0, 0.0, 1.0])
+    w = nd.where(cond, a, -a)
+    assert np.allclose(w.asnumpy(), [-2, -0.5, 3])
+
+
+def test_topk_sort():
+    a = nd.array([3.0, 1.0, 2.0])
+    assert nd.topk(a, k=2).asnumpy().tolist() == [0, 2]
+    assert nd.sort(a).asnumpy().tolist() == [1, 2, 3]
+
+
+def test_waitall():
+    nd.waitall()  # no-op on CPU, must not raise
