@@ -1,0 +1,646 @@
+// NHWC convolution for gfx950: implicit-GEMM MFMA kernels (fwd, bwd-data,
+// bwd-weight) with an im2col+GEMM fallback for shapes the implicit path
+// cannot tile (C % 8 != 0, e.g. the RGB stem; fp32).
+//
+// Reference parity: src/operator/nn/convolution.cu:37-213 (cuDNN autotune /
+// im2col there).  MI355X design: the conv IS a GEMM on this hardware -
+//   fwd:   y[N*P*Q, K]  = sum_(r,s,c) x[n, p*s+r-ph, q*s+s-pw, c] * w[k,r,s,c]
+//   data:  dx[N*H*W, C] = sum_(r,s,k) dy[gather] * w~[r,s,c,k]
+//   weight:dw[K, r,s,c] = sum_(n,p,q) dy[m,k] * x[gather]
+// NHWC makes every reduction segment channel-contiguous, so the MFMA
+// staging is the same global_load_lds pattern as gemm.hip with a gather
+// on the pixel address; out-of-window taps redirect to the zero page
+// (branch-free padding).  Weights keep the mxnet [K,R,S,C] layout -- the
+// forward B-operand reads it contiguously as-is; bwd-data uses a one-off
+// [R,S,C,K] permuted copy (cheap: weights are KB-MB).
+#include "torch_common.h"
+
+DEV_INLINE void gload_lds16c(const void* g, void* lds) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)(uintptr_t)lds, 16, 0,
+      0);
+}
+
+// ---------------------------------------------------------------------------
+// im2col / col2im (generic fallback; also builds the stem's GEMM operand)
+// col[M, R*S*C] with M = N*P*Q
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void im2col_nhwc_kernel(const T* __restrict__ x, T* __restrict__ col,
+                                   long total, int H, int W, int C, int P,
+                                   int Q, int R, int S, int sh, int sw, int ph,
+                                   int pw, int dh, int dw) {
+  long RSC = (long)R * S * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int s = t % S;
+    t /= S;
+    int r = t % R;
+    long m = t / R;
+    int q = m % Q;
+    long t2 = m / Q;
+    int p = t2 % P;
+    int n = t2 / P;
+    int h = p * sh - ph + r * dh, w = q * sw - pw + s * dw;
+    T v = (T)0;
+    if (h >= 0 && h < H && w >= 0 && w < W)
+      v = x[(((long)n * H + h) * W + w) * C + c];
+    col[m * RSC + ((long)r * S + s) * C + c] = v;
+  }
+}
+
+// gather form (no atomics): dx[n,h,w,c] = sum over (r,s) hitting (h,w)
+template <typename T>
+__global__ void col2im_nhwc_kernel(const T* __restrict__ col,
+                                   T* __restrict__ dx, long total, int H,
+                                   int W, int C, int P, int Q, int R, int S,
+                                   int sh, int sw, int ph, int pw, int dh,
+                                   int dw) {
+  long RSC = (long)R * S * C;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int w = t % W;
+    long t2 = t / W;
+    int h = t2 % H;
+    int n = t2 / H;
+    float acc = 0.f;
+    for (int r = 0; r < R; ++r) {
+      int pnum = h + ph - r * dh;
+      if (pnum < 0 || pnum % sh) continue;
+      int p = pnum / sh;
+      if (p >= P) continue;
+      for (int s = 0; s < S; ++s) {
+        int qnum = w + pw - s * dw;
+        if (qnum < 0 || qnum % sw) continue;
+        int q = qnum / sw;
+        if (q >= Q) continue;
+        long m = ((long)n * P + p) * Q + q;
+        acc += (float)col[m * RSC + ((long)r * S + s) * C + c];
+      }
+    }
+    dx[i] = (T)acc;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// implicit-GEMM forward: 128(pixels) x 128(out-channels) tile, BK=64
+// same wave/fragment geometry as gemm_nt_mfma_kernel (gemm.hip)
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
+    const T* __restrict__ x, const T* __restrict__ w,
+    const float* __restrict__ bias, T* __restrict__ y, int NB, int H, int W,
+    int C, int Kout, int P, int Q, int R, int S, int sh, int sw, int ph,
+    int pw, int dh, int dw, const T* __restrict__ zpage, bool relu) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BM = 128, BN = 128, BK = 64;
+  __shared__ T As[2][BM * BK];
+  __shared__ T Bs[2][BN * BK];
+
+  const long M = (long)NB * P * Q;
+  const long RSC = (long)R * S * C;
+  const int nTn = (Kout + BN - 1) / BN;
+  const int nwg = (int)(((M + BM - 1) / BM) * nTn);
+  const int bid = xcd_swizzle(blockIdx.x, nwg);
+  const long m0 = (long)(bid / nTn) * BM;
+  const long n0 = (long)(bid % nTn) * BN;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int s_row = t >> 3;
+  const int s_col = (t & 7) * 8;
+
+  // per-thread pixel bases for the 4 staged rows (fixed for the tile)
+  int pb_n[4], pb_h[4], pb_w[4];
+  bool pb_ok[4];
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    long pix = m0 + rnd * 32 + s_row;
+    pb_ok[rnd] = pix < M;
+    long pp = pb_ok[rnd] ? pix : 0;
+    int q = (int)(pp % Q);
+    long t2 = pp / Q;
+    int p = (int)(t2 % P);
+    pb_n[rnd] = (int)(t2 / P);
+    pb_h[rnd] = p * sh - ph;
+    pb_w[rnd] = q * sw - pw;
+  }
+
+  const int cpl = (C + BK - 1) / BK;  // c-chunks per (r,s) plane
+  const int nk = R * S * cpl;
+
+  float4_t acc[4][4] = {};
+
+  auto stage = [&](int buf, int step) {
+    const int rs = step / cpl;
+    const int c0 = (step % cpl) * BK;
+    const int r = rs / S, s = rs % S;
+    const int cseg = c0 + s_col;
+    const bool c_ok = cseg + 8 <= C;
+    const long wbase = (long)rs * C + cseg;
+#pragma unroll
+    for (int rnd = 0; rnd < 4; ++rnd) {
+      const int ih = pb_h[rnd] + r * dh, iw = pb_w[rnd] + s * dw;
+      const bool ok = pb_ok[rnd] && c_ok && ih >= 0 && ih < H && iw >= 0 &&
+                      iw < W;
+      const T* ga = ok
+          ? x + (((long)pb_n[rnd] * H + ih) * W + iw) * C + cseg
+          : zpage;
+      gload_lds16c(ga, &As[buf][(rnd * 256 + t) * 8]);
+      const long kb = n0 + rnd * 32 + s_row;
+      const T* gb = (kb < Kout && c_ok) ? w + kb * RSC + wbase : zpage;
+      gload_lds16c(gb, &Bs[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nk) stage(buf ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[4], bf[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        af[m] = *(const Frag*)&As[buf][(wr * 64 + m * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        bf[n] = *(const Frag*)&Bs[buf][(wc * 64 + n * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const long col = n0 + wc * 64 + n * 16 + d_col;
+    if (col >= Kout) continue;
+    const float b = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const long row_base = m0 + wr * 64 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long row = row_base + j;
+        if (row < M) {
+          float v = acc[m][n][j] + b;
+          if (relu) v = fmaxf(v, 0.f);
+          y[row * Kout + col] = (T)v;
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// implicit-GEMM backward-data: 128(input pixels) x 128(C) tile; reduction
+// over (r,s) x Kout chunks; wt is the [R,S,C,K] permuted weight
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
+    const T* __restrict__ dy, const T* __restrict__ wt, T* __restrict__ dx,
+    int NB, int H, int W, int C, int Kout, int P, int Q, int R, int S,
+    int sh, int sw, int ph, int pw, int dh, int dw,
+    const T* __restrict__ zpage) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BM = 128, BN = 128, BK = 64;
+  __shared__ T As[2][BM * BK];
+  __shared__ T Bs[2][BN * BK];
+
+  const long M = (long)NB * H * W;
+  const int nTn = (C + BN - 1) / BN;
+  const int nwg = (int)(((M + BM - 1) / BM) * nTn);
+  const int bid = xcd_swizzle(blockIdx.x, nwg);
+  const long m0 = (long)(bid / nTn) * BM;
+  const long n0 = (long)(bid % nTn) * BN;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int s_row = t >> 3;
+  const int s_col = (t & 7) * 8;
+
+  int pb_n[4], pb_h[4], pb_w[4];
+  bool pb_ok[4];
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    long pix = m0 + rnd * 32 + s_row;
+    pb_ok[rnd] = pix < M;
+    long pp = pb_ok[rnd] ? pix : 0;
+    pb_w[rnd] = (int)(pp % W);
+    long t2 = pp / W;
+    pb_h[rnd] = (int)(t2 % H);
+    pb_n[rnd] = (int)(t2 / H);
+  }
+
+  const int kpl = (Kout + BK - 1) / BK;
+  const int nk = R * S * kpl;
+
+  float4_t acc[4][4] = {};
+
+  auto stage = [&](int buf, int step) {
+    const int rs = step / kpl;
+    const int k0 = (step % kpl) * BK;
+    const int r = rs / S, s = rs % S;
+    const int kseg = k0 + s_col;
+    const bool k_ok = kseg + 8 <= Kout;
+#pragma unroll
+    for (int rnd = 0; rnd < 4; ++rnd) {
+      // which output pixel (p,q) feeds input (h,w) through tap (r,s)?
+      const int pnum = pb_h[rnd] + ph - r * dh;
+      const int qnum = pb_w[rnd] + pw - s * dw;
+      const int p = pnum / sh, q = qnum / sw;
+      const bool ok = pb_ok[rnd] && k_ok && pnum >= 0 && qnum >= 0 &&
+                      pnum % sh == 0 && qnum % sw == 0 && p < P && q < Q;
+      const T* ga = ok
+          ? dy + (((long)pb_n[rnd] * P + p) * Q + q) * Kout + kseg
+          : zpage;
+      gload_lds16c(ga, &As[buf][(rnd * 256 + t) * 8]);
+      const long cb = n0 + rnd * 32 + s_row;
+      const T* gb = (cb < C && k_ok)
+          ? wt + ((long)rs * C + cb) * Kout + kseg
+          : zpage;
+      gload_lds16c(gb, &Bs[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nk) stage(buf ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[4], bf[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        af[m] = *(const Frag*)&As[buf][(wr * 64 + m * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        bf[n] = *(const Frag*)&Bs[buf][(wc * 64 + n * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const long col = n0 + wc * 64 + n * 16 + d_col;
+    if (col >= C) continue;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const long row_base = m0 + wr * 64 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long row = row_base + j;
+        if (row < M) dx[row * C + col] = (T)acc[m][n][j];
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// implicit backward-weight: dw[k, (r,s,c)] = sum_m dy[m,k] * x_tap[m,(r,s,c)]
+// TN-shaped: reduction m is the stored row dim of both operands, so tiles
+// are transposed while staging (reg -> 8x ds_write_b16).  64x64 output
+// tile per block (4 waves, 2x2, 32x32 each), fp32 atomics over the
+// grid.y m-split into the dw workspace.
+// pixtab[m] = (n, h_base, w_base, valid) precomputed once per call.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
+    int H, int W, int C, int Kout, int R, int S, int dh, int dw,
+    long m_per_slice) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BI = 64, BJ = 64, BKM = 64;
+  __shared__ T DyT[2][BI * BKM];  // [i(out-ch)][m]
+  __shared__ T XT[2][BJ * BKM];   // [j(c within chunk)][m]
+
+  const int cpl = (C + BJ - 1) / BJ;
+  const int nTj = R * S * cpl;
+  const int bid = blockIdx.x;
+  const int i0 = (bid / nTj) * BI;
+  const int jt = bid % nTj;
+  const int rs = jt / cpl;
+  const int c0 = (jt % cpl) * BJ;
+  const int r = rs / S, s = rs % S;
+  const int roff = r * dh, soff = s * dw;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  // staging map: task idx = rnd*256+t -> m_local = idx/8 in 0..31 (x2 rnds),
+  // seg = idx%8 -> 8 channels
+  const int sm_half = t >> 3;   // m_local within half (0..31)
+  const int seg = t & 7;
+
+  float4_t acc[2][2] = {};
+
+  auto stage = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      // dy side: read 8 out-channels [i0+seg*8 .. +8) at row m_g
+      {
+        const int iseg = i0 + seg * 8;
+        half8_t zero8 = {};
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok && iseg + 8 <= Kout)
+          v = *(const V8*)(dy + m_g * Kout + iseg);
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+        (void)zero8;
+      }
+      // x side: gathered tap (r,s) at channels [c0+seg*8 .. +8)
+      {
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok) {
+          int4_t pt = pixtab[m_g];
+          const int ih = pt[1] + roff, iw = pt[2] + soff;
+          const int cseg = c0 + seg * 8;
+          if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+              cseg + 8 <= C)
+            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C + cseg);
+        }
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          XT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+      }
+    }
+  };
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  int buf = 0;
+  stage(0, ms0);
+  __syncthreads();
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[2], bf[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+        af[m] = *(const Frag*)&DyT[buf][(wr * 32 + m * 16 + a_row) * BKM +
+                                        kk * 32 + k_off];
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        bf[n] = *(const Frag*)&XT[buf][(wc * 32 + n * 16 + a_row) * BKM +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    if (mc + BKM < ms1) {
+      stage(buf ^ 1, mc + BKM);
+      buf ^= 1;
+      __syncthreads();
+    }
+  }
+
+  const long RSC = (long)R * S * C;
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    const long c = c0 + wc * 32 + n * 16 + d_col;
+    if (c >= C) continue;
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long i = i_base + j;
+        if (i < Kout) {
+          float* dst = dw32 + i * RSC + (long)rs * C + c;
+          if (gridDim.y == 1) *dst += acc[m][n][j];
+          else atomicAdd(dst, acc[m][n][j]);
+        }
+      }
+    }
+  }
+}
+
+// pixtab builder: m -> (n, h_base, w_base, valid)
+__global__ void build_pixtab_kernel(int4_t* __restrict__ tab, long M, int P,
+                                    int Q, int sh, int sw, int ph, int pw) {
+  for (long m = (long)blockIdx.x * blockDim.x + threadIdx.x; m < M;
+       m += (long)gridDim.x * blockDim.x) {
+    int q = (int)(m % Q);
+    long t2 = m / Q;
+    int p = (int)(t2 % P);
+    int n = (int)(t2 / P);
+    int4_t v;
+    v[0] = n;
+    v[1] = p * sh - ph;
+    v[2] = q * sw - pw;
+    v[3] = 1;
+    tab[m] = v;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// host wrappers
+// ---------------------------------------------------------------------------
+
+static at::Tensor im2col_nhwc(const at::Tensor& x, int P, int Q, int R, int S,
+                              int sh, int sw, int ph, int pw, int dh, int dw) {
+  int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  long M = (long)NB * P * Q;
+  auto col = at::empty({M, (long)R * S * C}, x.options());
+  long total = col.numel();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "im2col", [&] {
+    im2col_nhwc_kernel<scalar_t><<<ew_grid(total), 256, 0, cur_stream()>>>(
+        (const scalar_t*)x.data_ptr(), (scalar_t*)col.data_ptr(), total, H,
+        W, C, P, Q, R, S, sh, sw, ph, pw, dh, dw);
+  });
+  HIP_CHECK_LAST();
+  return col;
+}
+
+at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
+                           c10::optional<at::Tensor> bias, int64_t sh,
+                           int64_t sw, int64_t ph, int64_t pw, int64_t dh,
+                           int64_t dw, int64_t groups) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
+  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
+  int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int Kout = w.size(0), R = w.size(1), S = w.size(2);
+  int P = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
+  int Q = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  long M = (long)NB * P * Q;
+  auto y = at::empty({NB, P, Q, Kout}, x.options());
+  if (y.numel() == 0) return y;
+  bool mfma_ok = (x.scalar_type() == at::kHalf ||
+                  x.scalar_type() == at::kBFloat16) && C % 8 == 0;
+  at::Tensor b32;
+  const float* bias_ptr = nullptr;
+  if (bias && bias->defined() && bias->numel() > 0) {
+    b32 = bias->to(at::kFloat).contiguous();
+    bias_ptr = b32.data_ptr<float>();
+  }
+  if (mfma_ok && R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
+      pw == 0) {
+    // 1x1 stride-1: a plain NT GEMM on the flattened pixels
+    auto y2 = gemm_nt_core(x.view({M, C}), w.view({Kout, C}),
+                           bias ? c10::optional<at::Tensor>(b32)
+                                : c10::nullopt,
+                           false);
+    return y2.view({NB, P, Q, Kout});
+  }
+  if (mfma_ok) {
+    int nwg = (int)(((M + 127) / 128) * ((Kout + 127) / 128));
+    DISPATCH_HALF_TYPES(x.scalar_type(), "conv_fwd", [&] {
+      conv_fwd_igemm_kernel<scalar_t><<<nwg, 256, 0, cur_stream()>>>(
+          (const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
+          bias_ptr, (scalar_t*)y.data_ptr(), NB, H, W, C, Kout, P, Q, R, S,
+          sh, sw, ph, pw, dh, dw, (const scalar_t*)zero_page(x), false);
+    });
+    HIP_CHECK_LAST();
+    return y;
+  }
+  // generic: im2col + NT GEMM (stem 7x7 C=3, fp32, odd C)
+  auto col = im2col_nhwc(x, P, Q, R, S, sh, sw, ph, pw, dh, dw);
+  auto y2 = gemm_nt_core(col, w.view({Kout, (long)R * S * C}),
+                         bias_ptr ? c10::optional<at::Tensor>(b32)
+                                  : c10::nullopt,
+                         false);
+  return y2.view({NB, P, Q, Kout});
+}
+
+at::Tensor conv2d_nhwc_bwd_data(const at::Tensor& dy, const at::Tensor& w,
+                                int64_t H, int64_t W, int64_t sh, int64_t sw,
+                                int64_t ph, int64_t pw, int64_t dh,
+                                int64_t dw, int64_t groups) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(w);
+  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
+  int NB = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  int R = w.size(1), S = w.size(2), C = w.size(3);
+  auto dx = at::empty({NB, (long)H, (long)W, (long)C}, dy.options());
+  long M2 = (long)NB * H * W;
+  bool mfma_ok = (dy.scalar_type() == at::kHalf ||
+                  dy.scalar_type() == at::kBFloat16) && Kout % 8 == 0;
+  if (mfma_ok && R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
+      pw == 0) {
+    auto wt = transpose2d(w.view({Kout, C}));  // [C, Kout]
+    auto d2 = gemm_nt_core(dy.view({(long)NB * P * Q, Kout}), wt,
+                           c10::nullopt, false);
+    return d2.view({NB, (long)H, (long)W, (long)C});
+  }
+  if (mfma_ok) {
+    auto wt = w.permute({1, 2, 3, 0}).contiguous();  // [R,S,C,K]
+    int nwg = (int)(((M2 + 127) / 128) * ((C + 127) / 128));
+    DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_data", [&] {
+      conv_bwd_data_igemm_kernel<scalar_t><<<nwg, 256, 0, cur_stream()>>>(
+          (const scalar_t*)dy.data_ptr(), (const scalar_t*)wt.data_ptr(),
+          (scalar_t*)dx.data_ptr(), NB, H, W, C, Kout, P, Q, R, S, sh, sw,
+          ph, pw, dh, dw, (const scalar_t*)zero_page(dy));
+    });
+    HIP_CHECK_LAST();
+    return dx;
+  }
+  // generic: dcol = dy @ w_flat, then col2im
+  auto wt = transpose2d(w.view({Kout, (long)R * S * C}));  // [RSC, K]
+  auto dcol = gemm_nt_core(dy.view({(long)NB * P * Q, Kout}), wt,
+                           c10::nullopt, false);  // [M, RSC]
+  long total = dx.numel();
+  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "col2im", [&] {
+    col2im_nhwc_kernel<scalar_t><<<ew_grid(total), 256, 0, cur_stream()>>>(
+        (const scalar_t*)dcol.data_ptr(), (scalar_t*)dx.data_ptr(), total, H,
+        W, C, P, Q, R, S, sh, sw, ph, pw, dh, dw);
+  });
+  HIP_CHECK_LAST();
+  return dx;
+}
+
+at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
+                                  int64_t R, int64_t S, int64_t sh,
+                                  int64_t sw, int64_t ph, int64_t pw,
+                                  int64_t dh, int64_t dw, int64_t groups) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
+  int NB = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
+  int H = x.size(1), W = x.size(2), C = x.size(3);
+  long M = (long)NB * P * Q;
+  bool mfma_ok = (dy.scalar_type() == at::kHalf ||
+                  dy.scalar_type() == at::kBFloat16) && C % 8 == 0 &&
+                 Kout % 8 == 0;
+  if (mfma_ok) {
+    auto dw32 = at::zeros({(long)Kout, (long)R * S * C},
+                          dy.options().dtype(at::kFloat));
+    auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
+    build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
+        (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
+    int cpl = (C + 63) / 64;
+    int nwg = (int)(((Kout + 63) / 64) * (long)R * S * cpl);
+    // split the reduction so the grid fills the chip
+    long want_blocks = 2048;
+    long yb = std::max<long>(
+        1, std::min<long>((M + 63) / 64, want_blocks / std::max(nwg, 1)));
+    long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
+    yb = (M + m_per_slice - 1) / m_per_slice;
+    dim3 grid((unsigned)nwg, (unsigned)yb);
+    DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
+      conv_bwd_w_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
+          (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+          (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W, C,
+          Kout, R, S, dh, dw, m_per_slice);
+    });
+    HIP_CHECK_LAST();
+    return dw32.view({(long)Kout, R, S, (long)C}).to(dy.scalar_type());
+  }
+  // generic: dw = dy^T @ col (both transposed into NT form)
+  auto col = im2col_nhwc(x, P, Q, R, S, sh, sw, ph, pw, dh, dw);
+  auto dyT = transpose2d(dy.view({M, (long)Kout}).contiguous());
+  auto colT = transpose2d(col);
+  auto dwf = gemm_nt_core(dyT, colT, c10::nullopt, false);  // [K, RSC]
+  return dwf.view({(long)Kout, R, S, (long)C});
+}

@@ -1,0 +1,577 @@
+#include "hip/hip_runtime.h"
+// BatchNorm (NHWC, fused residual-add + ReLU) and LayerNorm.
+//
+// Reference parity: src/operator/nn/batch_norm.cu:238-660 (fwd/bwd incl.
+// the fused BNReLU/BNAddReLU variants), src/operator/nn/layer_norm.cu.
+//
+// MI355X design: NHWC puts channels contiguous, so per-channel statistics
+// are column reductions: blocks own a 64-column chunk x row range,
+// accumulate in fp32 registers, cross-wave reduce in LDS, one atomicAdd
+// per column into an fp32 workspace.  The apply pass fuses
+// normalize + residual-add + ReLU (the reference needs a graph fusion
+// pass for this; here it is a single kernel by construction).
+#include "torch_common_hip.h"
+
+// ---------------------------------------------------------------------------
+// column-chunk reduction: each block covers cols [c0,c0+64) and a row range,
+// threads = 64 cols x 4 row-lanes.
+// Computes sum and sum-of-squares (fwd) or the two backward sums.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void bn_reduce_kernel(const T* __restrict__ x, long M, long C,
+                                 long rows_per_block,
+                                 float* __restrict__ sum,
+                                 float* __restrict__ sumsq) {
+  __shared__ float s0[4][64], s1[4][64];
+  int cc = threadIdx.x & 63;        // column within chunk
+  int rl = threadIdx.x >> 6;        // row lane 0..3
+  long c = (long)blockIdx.x * 64 + cc;
+  long r0 = (long)blockIdx.y * rows_per_block;
+  long r1 = min(M, r0 + rows_per_block);
+  float a0 = 0.f, a1 = 0.f;
+  if (c < C) {
+    for (long r = r0 + rl; r < r1; r += 4) {
+      float v = (float)x[r * C + c];
+      a0 += v;
+      a1 += v * v;
+    }
+  }
+  s0[rl][cc] = a0;
+  s1[rl][cc] = a1;
+  __syncthreads();
+  if (rl == 0 && c < C) {
+    a0 = s0[0][cc] + s0[1][cc] + s0[2][cc] + s0[3][cc];
+    a1 = s1[0][cc] + s1[1][cc] + s1[2][cc] + s1[3][cc];
+    if (gridDim.y == 1) {
+      sum[c] = a0;
+      sumsq[c] = a1;
+    } else {
+      atomicAdd(sum + c, a0);
+      atomicAdd(sumsq + c, a1);
+    }
+  }
+}
+
+// finalize training stats: mean/istd + running-stat update
+// (running = momentum*running + (1-momentum)*batch; running var unbiased,
+// matching the torch CPU oracle)
+__global__ void bn_finalize_kernel(const float* __restrict__ sum,
+                                   const float* __restrict__ sumsq, long M,
+                                   long C, float momentum, float eps,
+                                   float* __restrict__ save_mean,
+                                   float* __restrict__ save_istd,
+                                   float* __restrict__ rmean,
+                                   float* __restrict__ rvar) {
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += (long)gridDim.x * blockDim.x) {
+    float mean = sum[c] / M;
+    float var = fmaxf(sumsq[c] / M - mean * mean, 0.f);
+    save_mean[c] = mean;
+    save_istd[c] = rsqrtf(var + eps);
+    if (rmean) {
+      float unbias = M > 1 ? var * M / (M - 1) : var;
+      rmean[c] = momentum * rmean[c] + (1.f - momentum) * mean;
+      rvar[c] = momentum * rvar[c] + (1.f - momentum) * unbias;
+    }
+  }
+}
+
+// inference: scale/shift from running stats
+__global__ void bn_scale_shift_kernel(const float* __restrict__ gamma,
+                                      const float* __restrict__ beta,
+                                      const float* __restrict__ mean,
+                                      const float* __restrict__ var_or_istd,
+                                      bool is_istd, float eps, long C,
+                                      float* __restrict__ scale,
+                                      float* __restrict__ shift) {
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < C;
+       c += (long)gridDim.x * blockDim.x) {
+    float istd = is_istd ? var_or_istd[c] : rsqrtf(var_or_istd[c] + eps);
+    float sc = gamma[c] * istd;
+    scale[c] = sc;
+    shift[c] = beta[c] - mean[c] * sc;
+  }
+}
+
+// apply: y = x*scale[c] + shift[c] (+ residual) (relu)
+// scale/shift staged in LDS when C fits (<=4096).
+template <typename T, bool RELU, bool RES>
+__global__ void bn_apply_kernel(const T* __restrict__ x,
+                                const T* __restrict__ res,
+                                T* __restrict__ y, long total, long C,
+                                const float* __restrict__ scale,
+                                const float* __restrict__ shift) {
+  extern __shared__ float lds[];
+  float* s_scale = lds;
+  float* s_shift = lds + C;
+  const bool use_lds = C <= 4096;
+  if (use_lds) {
+    for (long c = threadIdx.x; c < C; c += blockDim.x) {
+      s_scale[c] = scale[c];
+      s_shift[c] = shift[c];
+    }
+    __syncthreads();
+  }
+  const float* sc = use_lds ? s_scale : scale;
+  const float* sh = use_lds ? s_shift : shift;
+  using VecT = T __attribute__((ext_vector_type(8)));
+  if (C % 8 == 0) {
+    long nv = total / 8;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+      long c0 = (i * 8) % C;
+      VecT v = reinterpret_cast<const VecT*>(x)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float z = (float)v[j] * sc[c0 + j] + sh[c0 + j];
+        if (RES) z += (float)res[i * 8 + j];
+        o[j] = (T)(RELU ? fmaxf(z, 0.f) : z);
+      }
+      reinterpret_cast<VecT*>(y)[i] = o;
+    }
+  } else {
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+         i += (long)gridDim.x * blockDim.x) {
+      long c = i % C;
+      float z = (float)x[i] * sc[c] + sh[c];
+      if (RES) z += (float)res[i];
+      y[i] = (T)(RELU ? fmaxf(z, 0.f) : z);
+    }
+  }
+}
+
+// backward column reduction: s1 = sum(dy_eff), s2 = sum(dy_eff * xhat)
+// dy_eff = relu-masked dy (mask from the saved post-activation y)
+template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const T* __restrict__ y, long M, long C,
+                                     long rows_per_block,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ istd,
+                                     float* __restrict__ s1,
+                                     float* __restrict__ s2) {
+  __shared__ float l1[4][64], l2[4][64];
+  int cc = threadIdx.x & 63;
+  int rl = threadIdx.x >> 6;
+  long c = (long)blockIdx.x * 64 + cc;
+  long r0 = (long)blockIdx.y * rows_per_block;
+  long r1 = min(M, r0 + rows_per_block);
+  float a1 = 0.f, a2 = 0.f;
+  if (c < C) {
+    float mu = mean[c], is = istd[c];
+    for (long r = r0 + rl; r < r1; r += 4) {
+      long idx = r * C + c;
+      float g = (float)dy[idx];
+      if (RELU && (float)y[idx] <= 0.f) g = 0.f;
+      a1 += g;
+      a2 += g * ((float)x[idx] - mu) * is;
+    }
+  }
+  l1[rl][cc] = a1;
+  l2[rl][cc] = a2;
+  __syncthreads();
+  if (rl == 0 && c < C) {
+    a1 = l1[0][cc] + l1[1][cc] + l1[2][cc] + l1[3][cc];
+    a2 = l2[0][cc] + l2[1][cc] + l2[2][cc] + l2[3][cc];
+    if (gridDim.y == 1) {
+      s1[c] = a1;
+      s2[c] = a2;
+    } else {
+      atomicAdd(s1 + c, a1);
+      atomicAdd(s2 + c, a2);
+    }
+  }
+}
+
+// backward apply:
+// dx = gamma*istd * (dy_eff - (s1 + xhat*s2)/M); dres = dy_eff
+template <typename T, bool RELU, bool RES>
+__global__ void bn_bwd_apply_kernel(const T* __restrict__ dy,
+                                    const T* __restrict__ x,
+                                    const T* __restrict__ y,
+                                    T* __restrict__ dx, T* __restrict__ dres,
+                                    long total, long C, float invM,
+                                    const float* __restrict__ gamma,
+                                    const float* __restrict__ mean,
+                                    const float* __restrict__ istd,
+                                    const float* __restrict__ s1,
+                                    const float* __restrict__ s2) {
+  extern __shared__ float lds[];  // [C] x5: ga*is, mean, istd, s1/M, s2/M
+  float* c_gis = lds;
+  float* c_mu = lds + C;
+  float* c_is = lds + 2 * C;
+  float* c_s1 = lds + 3 * C;
+  float* c_s2 = lds + 4 * C;
+  const bool use_lds = C <= 4096;
+  if (use_lds) {
+    for (long c = threadIdx.x; c < C; c += blockDim.x) {
+      c_gis[c] = gamma[c] * istd[c];
+      c_mu[c] = mean[c];
+      c_is[c] = istd[c];
+      c_s1[c] = s1[c] * invM;
+      c_s2[c] = s2[c] * invM;
+    }
+    __syncthreads();
+  }
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    long c = i % C;
+    float g = (float)dy[i];
+    if (RELU && (float)y[i] <= 0.f) g = 0.f;
+    float gis, mu, m1, m2, is;
+    if (use_lds) {
+      gis = c_gis[c]; mu = c_mu[c]; is = c_is[c]; m1 = c_s1[c]; m2 = c_s2[c];
+    } else {
+      gis = gamma[c] * istd[c]; mu = mean[c]; is = istd[c];
+      m1 = s1[c] * invM; m2 = s2[c] * invM;
+    }
+    float xhat = ((float)x[i] - mu) * is;
+    dx[i] = (T)(gis * (g - m1 - xhat * m2));
+    if (RES) dres[i] = (T)g;
+  }
+}
+
+// -- host wrappers ----------------------------------------------------------
+
+static dim3 bn_reduce_grid(long M, long C, long* rows_per_block) {
+  long xb = (C + 63) / 64;
+  long yb = std::max<long>(1, std::min<long>((M + 255) / 256,
+                                             2048 / std::max<long>(xb, 1)));
+  *rows_per_block = (M + yb - 1) / yb;
+  return dim3((unsigned)xb, (unsigned)yb);
+}
+
+std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
+                                          const at::Tensor& gamma,
+                                          const at::Tensor& beta,
+                                          at::Tensor rmean, at::Tensor rvar,
+                                          double momentum, double eps,
+                                          bool fuse_relu,
+                                          const at::Tensor& residual) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  long C = x.size(-1), M = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto sum = at::zeros({C}, opts), sumsq = at::zeros({C}, opts);
+  auto save_mean = at::empty({C}, opts), save_istd = at::empty({C}, opts);
+  auto scale = at::empty({C}, opts), shift = at::empty({C}, opts);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto rm32 = rmean.to(at::kFloat).contiguous();
+  auto rv32 = rvar.to(at::kFloat).contiguous();
+  bool has_res = residual.numel() > 0;
+  long rpb;
+  dim3 grid = bn_reduce_grid(M, C, &rpb);
+  auto stream = cur_stream();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_fwd", [&] {
+   hipLaunchKernelGGL(( bn_reduce_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
+        (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
+        sumsq.data_ptr<float>());
+   hipLaunchKernelGGL(( bn_finalize_kernel), dim3((int)((C + 255) / 256)), dim3(256), 0, stream, 
+        sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C,
+        (float)momentum, (float)eps, save_mean.data_ptr<float>(),
+        save_istd.data_ptr<float>(), rm32.data_ptr<float>(),
+        rv32.data_ptr<float>());
+   hipLaunchKernelGGL(( bn_scale_shift_kernel), dim3((int)((C + 255) / 256)), dim3(256), 0, stream, 
+        g32.data_ptr<float>(), b32.data_ptr<float>(),
+        save_mean.data_ptr<float>(), save_istd.data_ptr<float>(), true, 0.f,
+        C, scale.data_ptr<float>(), shift.data_ptr<float>());
+    long total = x.numel();
+    size_t lds = C <= 4096 ? 2 * C * sizeof(float) : 0;
+    auto launch_apply = [&](auto relu_c, auto res_c) {
+     hipLaunchKernelGGL(( bn_apply_kernel<scalar_t, decltype(relu_c)::value, decltype(res_c)::value>)
+          , dim3(ew_grid(total / 8 + 1)), dim3(256), lds, stream, 
+              (const scalar_t*)x.data_ptr(),
+              has_res ? (const scalar_t*)residual.data_ptr() : nullptr,
+              (scalar_t*)y.data_ptr(), total, C, scale.data_ptr<float>(),
+              shift.data_ptr<float>());
+    };
+    if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
+    else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});
+    else if (has_res) launch_apply(std::false_type{}, std::true_type{});
+    else launch_apply(std::false_type{}, std::false_type{});
+  });
+  HIP_CHECK_LAST();
+  // write back running stats in their original dtype
+  rmean.copy_(rm32.to(rmean.scalar_type()));
+  rvar.copy_(rv32.to(rvar.scalar_type()));
+  return {y, save_mean, save_istd};
+}
+
+at::Tensor bn_nhwc_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
+                             const at::Tensor& beta, const at::Tensor& rmean,
+                             const at::Tensor& rvar, double eps,
+                             bool fuse_relu, const at::Tensor& residual) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  long C = x.size(-1);
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto scale = at::empty({C}, opts), shift = at::empty({C}, opts);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  auto rm32 = rmean.to(at::kFloat).contiguous();
+  auto rv32 = rvar.to(at::kFloat).contiguous();
+  bool has_res = residual.numel() > 0;
+  auto stream = cur_stream();
+ hipLaunchKernelGGL(( bn_scale_shift_kernel), dim3((int)((C + 255) / 256)), dim3(256), 0, stream, 
+      g32.data_ptr<float>(), b32.data_ptr<float>(), rm32.data_ptr<float>(),
+      rv32.data_ptr<float>(), false, (float)eps, C, scale.data_ptr<float>(),
+      shift.data_ptr<float>());
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_infer", [&] {
+    long total = x.numel();
+    size_t lds = C <= 4096 ? 2 * C * sizeof(float) : 0;
+    auto launch_apply = [&](auto relu_c, auto res_c) {
+     hipLaunchKernelGGL(( bn_apply_kernel<scalar_t, decltype(relu_c)::value, decltype(res_c)::value>)
+          , dim3(ew_grid(total / 8 + 1)), dim3(256), lds, stream, 
+              (const scalar_t*)x.data_ptr(),
+              has_res ? (const scalar_t*)residual.data_ptr() : nullptr,
+              (scalar_t*)y.data_ptr(), total, C, scale.data_ptr<float>(),
+              shift.data_ptr<float>());
+    };
+    if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
+    else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});
+    else if (has_res) launch_apply(std::false_type{}, std::true_type{});
+    else launch_apply(std::false_type{}, std::false_type{});
+  });
+  HIP_CHECK_LAST();
+  return y;
+}
+
+std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                    const at::Tensor& gamma,
+                                    const at::Tensor& save_mean,
+                                    const at::Tensor& save_istd,
+                                    bool fuse_relu, const at::Tensor& y,
+                                    bool has_res) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  long C = x.size(-1), M = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto dres = has_res ? at::empty_like(x) : at::empty({0}, x.options());
+  auto opts = x.options().dtype(at::kFloat);
+  auto s1 = at::zeros({C}, opts), s2 = at::zeros({C}, opts);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  long rpb;
+  dim3 grid = bn_reduce_grid(M, C, &rpb);
+  auto stream = cur_stream();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_bwd", [&] {
+    auto launch_red = [&](auto relu_c) {
+     hipLaunchKernelGGL(( bn_bwd_reduce_kernel<scalar_t, decltype(relu_c)::value>)
+          , dim3(grid), dim3(256), 0, stream, 
+              (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+              (const scalar_t*)y.data_ptr(), M, C, rpb,
+              save_mean.data_ptr<float>(), save_istd.data_ptr<float>(),
+              s1.data_ptr<float>(), s2.data_ptr<float>());
+    };
+    if (fuse_relu) launch_red(std::true_type{});
+    else launch_red(std::false_type{});
+    long total = x.numel();
+    size_t lds = C <= 4096 ? 5 * C * sizeof(float) : 0;
+    auto launch_apply = [&](auto relu_c, auto res_c) {
+     hipLaunchKernelGGL(( bn_bwd_apply_kernel<scalar_t, decltype(relu_c)::value,
+                          decltype(res_c)::value>)
+          , dim3(ew_grid(total)), dim3(256), lds, stream, 
+              (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+              (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(),
+              has_res ? (scalar_t*)dres.data_ptr() : nullptr, total, C,
+              1.f / M, g32.data_ptr<float>(), save_mean.data_ptr<float>(),
+              save_istd.data_ptr<float>(), s1.data_ptr<float>(),
+              s2.data_ptr<float>());
+    };
+    if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
+    else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});
+    else if (has_res) launch_apply(std::false_type{}, std::true_type{});
+    else launch_apply(std::false_type{}, std::false_type{});
+  });
+  HIP_CHECK_LAST();
+  auto dgamma = s2.to(gamma.scalar_type());
+  auto dbeta = s1.to(gamma.scalar_type());
+  return {dx, dgamma, dbeta, dres};
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm over the last axis (reference layer_norm.cu:172-560)
+// block per row; fp32 stats; saved mean/istd for backward
+// ---------------------------------------------------------------------------
+template <typename T, int VEC>
+__global__ void ln_fwd_kernel(const T* __restrict__ x,
+                              const float* __restrict__ gamma,
+                              const float* __restrict__ beta,
+                              T* __restrict__ y, float* __restrict__ omean,
+                              float* __restrict__ oistd, long rows, long C,
+                              float eps) {
+  __shared__ float sred[16];
+  using VecT = T __attribute__((ext_vector_type(VEC)));
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* xr = x + r * C;
+    T* yr = y + r * C;
+    long cv = C / VEC;
+    float s = 0.f, sq = 0.f;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float f = (float)v[j];
+        s += f;
+        sq += f * f;
+      }
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      float f = (float)xr[i];
+      s += f;
+      sq += f * f;
+    }
+    s = block_reduce(s, sred, SumOp(), 0.f);
+    __syncthreads();
+    sq = block_reduce(sq, sred, SumOp(), 0.f);
+    float mean = s / C;
+    float istd = rsqrtf(fmaxf(sq / C - mean * mean, 0.f) + eps);
+    if (threadIdx.x == 0) {
+      omean[r] = mean;
+      oistd[r] = istd;
+    }
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        long c = i * VEC + j;
+        o[j] = (T)(((float)v[j] - mean) * istd * gamma[c] + beta[c]);
+      }
+      reinterpret_cast<VecT*>(yr)[i] = o;
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      yr[i] = (T)(((float)xr[i] - mean) * istd * gamma[i] + beta[i]);
+    __syncthreads();
+  }
+}
+
+// dx = istd * (dy*g - mean_c(dy*g) - xhat * mean_c(dy*g*xhat))
+template <typename T>
+__global__ void ln_bwd_dx_kernel(const T* __restrict__ dy,
+                                 const T* __restrict__ x,
+                                 const float* __restrict__ gamma,
+                                 const float* __restrict__ mean,
+                                 const float* __restrict__ istd,
+                                 T* __restrict__ dx, long rows, long C) {
+  __shared__ float sred[16];
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* gr = dy + r * C;
+    const T* xr = x + r * C;
+    T* dr = dx + r * C;
+    float mu = mean[r], is = istd[r];
+    float a = 0.f, b = 0.f;
+    for (long i = threadIdx.x; i < C; i += blockDim.x) {
+      float gg = (float)gr[i] * gamma[i];
+      float xh = ((float)xr[i] - mu) * is;
+      a += gg * xh;
+      b += gg;
+    }
+    a = block_reduce(a, sred, SumOp(), 0.f);
+    __syncthreads();
+    b = block_reduce(b, sred, SumOp(), 0.f);
+    a /= C;
+    b /= C;
+    for (long i = threadIdx.x; i < C; i += blockDim.x) {
+      float gg = (float)gr[i] * gamma[i];
+      float xh = ((float)xr[i] - mu) * is;
+      dr[i] = (T)(is * (gg - b - xh * a));
+    }
+    __syncthreads();
+  }
+}
+
+// dgamma[c] = sum_r dy*xhat ; dbeta[c] = sum_r dy  (column reduction)
+template <typename T>
+__global__ void ln_bwd_dgamma_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ istd,
+                                     float* __restrict__ dgamma,
+                                     float* __restrict__ dbeta, long rows,
+                                     long C, long rows_per_block) {
+  __shared__ float l1[4][64], l2[4][64];
+  int cc = threadIdx.x & 63;
+  int rl = threadIdx.x >> 6;
+  long c = (long)blockIdx.x * 64 + cc;
+  long r0 = (long)blockIdx.y * rows_per_block;
+  long r1 = min(rows, r0 + rows_per_block);
+  float a1 = 0.f, a2 = 0.f;
+  if (c < C) {
+    for (long r = r0 + rl; r < r1; r += 4) {
+      float g = (float)dy[r * C + c];
+      float xh = ((float)x[r * C + c] - mean[r]) * istd[r];
+      a1 += g * xh;
+      a2 += g;
+    }
+  }
+  l1[rl][cc] = a1;
+  l2[rl][cc] = a2;
+  __syncthreads();
+  if (rl == 0 && c < C) {
+    a1 = l1[0][cc] + l1[1][cc] + l1[2][cc] + l1[3][cc];
+    a2 = l2[0][cc] + l2[1][cc] + l2[2][cc] + l2[3][cc];
+    if (gridDim.y == 1) {
+      dgamma[c] = a1;
+      dbeta[c] = a2;
+    } else {
+      atomicAdd(dgamma + c, a1);
+      atomicAdd(dbeta + c, a2);
+    }
+  }
+}
+
+std::vector<at::Tensor> layernorm_fwd(const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& beta, double eps) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  long C = x.size(-1), rows = x.numel() / C;
+  auto y = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({rows}, opts), istd = at::empty({rows}, opts);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  auto b32 = beta.to(at::kFloat).contiguous();
+  int grid = (int)std::min<long>(rows, 4096);
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "ln_fwd", [&] {
+    constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
+   hipLaunchKernelGGL(( ln_fwd_kernel<scalar_t, VEC>), dim3(grid), dim3(256), 0, cur_stream(), 
+        (const scalar_t*)x.data_ptr(), g32.data_ptr<float>(),
+        b32.data_ptr<float>(), (scalar_t*)y.data_ptr(),
+        mean.data_ptr<float>(), istd.data_ptr<float>(), rows, C, (float)eps);
+  });
+  HIP_CHECK_LAST();
+  return {y, mean, istd};
+}
+
+std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
+                                      const at::Tensor& x,
+                                      const at::Tensor& gamma,
+                                      const at::Tensor& mean,
+                                      const at::Tensor& istd) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  long C = x.size(-1), rows = x.numel() / C;
+  auto dx = at::empty_like(x);
+  auto opts = x.options().dtype(at::kFloat);
+  auto dg = at::zeros({C}, opts), db = at::zeros({C}, opts);
+  auto g32 = gamma.to(at::kFloat).contiguous();
+  int grid = (int)std::min<long>(rows, 4096);
+  long xb = (C + 63) / 64;
+  long yb = std::max<long>(1, std::min<long>((rows + 255) / 256,
+                                             2048 / std::max<long>(xb, 1)));
+  long rpb = (rows + yb - 1) / yb;
+  dim3 cgrid((unsigned)xb, (unsigned)yb);
+  auto stream = cur_stream();
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "ln_bwd", [&] {
+   hipLaunchKernelGGL(( ln_bwd_dx_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+        g32.data_ptr<float>(), mean.data_ptr<float>(),
+        istd.data_ptr<float>(), (scalar_t*)dx.data_ptr(), rows, C);
+   hipLaunchKernelGGL(( ln_bwd_dgamma_kernel<scalar_t>), dim3(cgrid), dim3(256), 0, stream, 
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+        mean.data_ptr<float>(), istd.data_ptr<float>(),
+        dg.data_ptr<float>(), db.data_ptr<float>(), rows, C, rpb);
+  });
+  HIP_CHECK_LAST();
+  return {dx, dg.to(gamma.scalar_type()), db.to(gamma.scalar_type())};
+}

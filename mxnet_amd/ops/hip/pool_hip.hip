@@ -1,0 +1,138 @@
+#include "hip/hip_runtime.h"
+// NHWC pooling (max / avg) fwd + bwd.
+//
+// Reference parity: src/operator/nn/pool.cuh (pool_max_2d / pool_sum_2d and
+// unpool kernels).  MI355X design: channels-contiguous NHWC so one thread
+// per output element reads coalesced C-segments; max-pool saves the argmax
+// plane index, backward scatters with fp32 atomics (windows overlap).
+#include "torch_common_hip.h"
+
+template <typename T, bool IS_MAX>
+__global__ void pool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                int* __restrict__ arg, long total, int N,
+                                int H, int W, int C, int P, int Q, int kh,
+                                int kw, int sh, int sw, int ph, int pw,
+                                bool cip) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int q = t % Q;
+    t /= Q;
+    int p = t % P;
+    int n = t / P;
+    int h0 = p * sh - ph, w0 = q * sw - pw;
+    int h1 = min(h0 + kh, H), w1 = min(w0 + kw, W);
+    int hs = max(h0, 0), ws = max(w0, 0);
+    const T* xn = x + (long)n * H * W * C;
+    if (IS_MAX) {
+      float best = -INFINITY;
+      int best_idx = hs * W + ws;
+      for (int h = hs; h < h1; ++h)
+        for (int w = ws; w < w1; ++w) {
+          float v = (float)xn[((long)h * W + w) * C + c];
+          if (v > best) {
+            best = v;
+            best_idx = h * W + w;
+          }
+        }
+      y[i] = (T)best;
+      arg[i] = best_idx;
+    } else {
+      float s = 0.f;
+      for (int h = hs; h < h1; ++h)
+        for (int w = ws; w < w1; ++w) s += (float)xn[((long)h * W + w) * C + c];
+      int cnt = cip ? kh * kw : (h1 - hs) * (w1 - ws);
+      y[i] = (T)(s / cnt);
+    }
+  }
+}
+
+template <typename T, bool IS_MAX>
+__global__ void pool_bwd_kernel(const T* __restrict__ dy,
+                                const int* __restrict__ arg,
+                                float* __restrict__ dx32, long total, int N,
+                                int H, int W, int C, int P, int Q, int kh,
+                                int kw, int sh, int sw, int ph, int pw,
+                                bool cip) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int q = t % Q;
+    t /= Q;
+    int p = t % P;
+    int n = t / P;
+    float g = (float)dy[i];
+    float* dxn = dx32 + (long)n * H * W * C;
+    if (IS_MAX) {
+      atomicAdd(dxn + (long)arg[i] * C + c, g);
+    } else {
+      int h0 = p * sh - ph, w0 = q * sw - pw;
+      int h1 = min(h0 + kh, H), w1 = min(w0 + kw, W);
+      int hs = max(h0, 0), ws = max(w0, 0);
+      int cnt = cip ? kh * kw : (h1 - hs) * (w1 - ws);
+      float share = g / cnt;
+      for (int h = hs; h < h1; ++h)
+        for (int w = ws; w < w1; ++w)
+          atomicAdd(dxn + ((long)h * W + w) * C + c, share);
+    }
+  }
+}
+
+std::vector<at::Tensor> pool_nhwc_fwd(const at::Tensor& x,
+                                      const std::string& kind, int64_t kh,
+                                      int64_t kw, int64_t sh, int64_t sw,
+                                      int64_t ph, int64_t pw, bool cip) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  TORCH_CHECK(x.dim() == 4, "pool expects NHWC 4-D input");
+  int N = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
+  int P = (H + 2 * ph - kh) / sh + 1;
+  int Q = (W + 2 * pw - kw) / sw + 1;
+  auto y = at::empty({N, P, Q, C}, x.options());
+  bool is_max = kind == "max";
+  auto arg = is_max ? at::empty({N, P, Q, C}, x.options().dtype(at::kInt))
+                    : at::empty({0}, x.options().dtype(at::kInt));
+  long total = y.numel();
+  if (total == 0) return {y, arg};
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "pool_fwd", [&] {
+    if (is_max)
+     hipLaunchKernelGGL(( pool_fwd_kernel<scalar_t, true>), dim3(ew_grid(total)), dim3(256), 0, cur_stream(), 
+          (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(),
+          arg.data_ptr<int>(), total, N, H, W, C, P, Q, kh, kw, sh, sw, ph,
+          pw, cip);
+    else
+     hipLaunchKernelGGL(( pool_fwd_kernel<scalar_t, false>), dim3(ew_grid(total)), dim3(256), 0,
+                                         cur_stream(), 
+          (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), nullptr,
+          total, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw, cip);
+  });
+  HIP_CHECK_LAST();
+  return {y, arg};
+}
+
+at::Tensor pool_nhwc_bwd(const at::Tensor& dy, const at::Tensor& arg,
+                         const std::string& kind, int64_t H, int64_t W,
+                         int64_t kh, int64_t kw, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw, bool cip) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy);
+  int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
+  auto dx32 = at::zeros({N, H, W, C}, dy.options().dtype(at::kFloat));
+  long total = dy.numel();
+  bool is_max = kind == "max";
+  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "pool_bwd", [&] {
+    if (is_max)
+     hipLaunchKernelGGL(( pool_bwd_kernel<scalar_t, true>), dim3(ew_grid(total)), dim3(256), 0,
+                                        cur_stream(), 
+          (const scalar_t*)dy.data_ptr(), arg.data_ptr<int>(),
+          dx32.data_ptr<float>(), total, N, H, W, C, P, Q, kh, kw, sh, sw,
+          ph, pw, cip);
+    else
+     hipLaunchKernelGGL(( pool_bwd_kernel<scalar_t, false>), dim3(ew_grid(total)), dim3(256), 0,
+                                         cur_stream(), 
+          (const scalar_t*)dy.data_ptr(), nullptr, dx32.data_ptr<float>(),
+          total, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw, cip);
+  });
+  HIP_CHECK_LAST();
+  return dx32.to(dy.scalar_type());
+}

@@ -1,0 +1,179 @@
+// Row softmax / log-softmax fwd+bwd, and column-sum (bias gradient).
+//
+// Reference parity: src/operator/nn/softmax-inl.h:351-820 (softmax_compute,
+// softmax_gradient), broadcast_reduce sum for the bias grad.
+//
+// MI355X design: one 256-thread block per row (4 waves), fp32 accumulation,
+// wave64 __shfl_xor + LDS block reduction, 8-wide vector loads for 16-bit
+// dtypes (Guideline 13).  Rows are the contiguous last axis.
+#include "torch_common.h"
+
+template <typename T, int VEC>
+__global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   long rows, long C, bool log_mode,
+                                   float invT) {
+  __shared__ float sred[16];
+  using VecT = T __attribute__((ext_vector_type(VEC)));
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* xr = x + r * C;
+    T* yr = y + r * C;
+    long cv = C / VEC;
+    // pass 1: max
+    float m = -INFINITY;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) m = fmaxf(m, (float)v[j]);
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      m = fmaxf(m, (float)xr[i]);
+    m = block_reduce(m, sred, MaxOp(), -INFINITY);
+    // pass 2: sum of exp
+    float s = 0.f;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) s += __expf(((float)v[j] - m) * invT);
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      s += __expf(((float)xr[i] - m) * invT);
+    __syncthreads();  // reuse of sred
+    s = block_reduce(s, sred, SumOp(), 0.f);
+    float inv_s = 1.f / s, log_s = __logf(s);
+    // pass 3: write
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float z = ((float)v[j] - m) * invT;
+        o[j] = (T)(log_mode ? z - log_s : __expf(z) * inv_s);
+      }
+      reinterpret_cast<VecT*>(yr)[i] = o;
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      float z = ((float)xr[i] - m) * invT;
+      yr[i] = (T)(log_mode ? z - log_s : __expf(z) * inv_s);
+    }
+    __syncthreads();
+  }
+}
+
+// softmax:     dx = (dy - sum(dy*y)) * y * invT
+// log_softmax: dx = (dy - exp(y) * sum(dy)) * invT
+template <typename T, int VEC>
+__global__ void softmax_bwd_kernel(const T* __restrict__ dy,
+                                   const T* __restrict__ y,
+                                   T* __restrict__ dx, long rows, long C,
+                                   bool log_mode, float invT) {
+  __shared__ float sred[16];
+  using VecT = T __attribute__((ext_vector_type(VEC)));
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* gr = dy + r * C;
+    const T* yr = y + r * C;
+    T* dr = dx + r * C;
+    long cv = C / VEC;
+    float s = 0.f;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT g = reinterpret_cast<const VecT*>(gr)[i];
+      VecT v = reinterpret_cast<const VecT*>(yr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        s += log_mode ? (float)g[j] : (float)g[j] * (float)v[j];
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      s += log_mode ? (float)gr[i] : (float)gr[i] * (float)yr[i];
+    s = block_reduce(s, sred, SumOp(), 0.f);
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT g = reinterpret_cast<const VecT*>(gr)[i];
+      VecT v = reinterpret_cast<const VecT*>(yr)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float d = log_mode ? (float)g[j] - __expf((float)v[j]) * s
+                           : ((float)g[j] - s) * (float)v[j];
+        o[j] = (T)(d * invT);
+      }
+      reinterpret_cast<VecT*>(dr)[i] = o;
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      float d = log_mode ? (float)gr[i] - __expf((float)yr[i]) * s
+                         : ((float)gr[i] - s) * (float)yr[i];
+      dr[i] = (T)(d * invT);
+    }
+    __syncthreads();
+  }
+}
+
+at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode,
+                       double temperature) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  auto y = at::empty_like(x);
+  long C = x.size(-1), rows = x.numel() / (C > 0 ? C : 1);
+  if (x.numel() == 0) return y;
+  int grid = (int)std::min<long>(rows, 4096);
+  DISPATCH_FLOAT_TYPES(x.scalar_type(), "softmax_fwd", [&] {
+    constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
+    softmax_fwd_kernel<scalar_t, VEC><<<grid, 256, 0, cur_stream()>>>(
+        (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), rows, C,
+        log_mode, (float)(1.0 / temperature));
+  });
+  HIP_CHECK_LAST();
+  return y;
+}
+
+at::Tensor softmax_bwd(const at::Tensor& dy, const at::Tensor& y,
+                       bool log_mode, double temperature) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(y);
+  auto dx = at::empty_like(dy);
+  long C = dy.size(-1), rows = dy.numel() / (C > 0 ? C : 1);
+  if (dy.numel() == 0) return dx;
+  int grid = (int)std::min<long>(rows, 4096);
+  DISPATCH_FLOAT_TYPES(dy.scalar_type(), "softmax_bwd", [&] {
+    constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
+    softmax_bwd_kernel<scalar_t, VEC><<<grid, 256, 0, cur_stream()>>>(
+        (const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+        (scalar_t*)dx.data_ptr(), rows, C, log_mode,
+        (float)(1.0 / temperature));
+  });
+  HIP_CHECK_LAST();
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+// column sum: out[c] = sum_m in[m, c]   (bias gradient)
+// threads stride columns (coalesced row segments); fp32 atomics over
+// row-chunks, cast at the end.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ in, float* __restrict__ out,
+                              long M, long N, long rows_per_block) {
+  long r0 = (long)blockIdx.y * rows_per_block;
+  long r1 = min(M, r0 + rows_per_block);
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < N;
+       c += (long)gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (long r = r0; r < r1; ++r) acc += (float)in[r * N + c];
+    if (gridDim.y == 1) out[c] = acc;
+    else atomicAdd(out + c, acc);
+  }
+}
+
+at::Tensor colsum(const at::Tensor& in) {
+  CHECK_GPU(in); CHECK_CONTIG(in);
+  long N = in.size(-1), M = in.numel() / (N > 0 ? N : 1);
+  auto out = at::zeros({N}, in.options().dtype(at::kFloat));
+  // enough row-chunks to fill the chip when N is small
+  long target_blocks = 2048;
+  long xblocks = (N + 255) / 256;
+  long yblocks = std::max<long>(1, std::min<long>(M, target_blocks / std::max<long>(xblocks, 1)));
+  long rows_per_block = (M + yblocks - 1) / yblocks;
+  dim3 grid((unsigned)std::min<long>(xblocks, 65535), (unsigned)yblocks);
+  DISPATCH_FLOAT_TYPES(in.scalar_type(), "colsum", [&] {
+    colsum_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
+        (const scalar_t*)in.data_ptr(), out.data_ptr<float>(), M, N,
+        rows_per_block);
+  });
+  HIP_CHECK_LAST();
+  return out.to(in.scalar_type());
+}
