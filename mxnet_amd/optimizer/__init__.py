@@ -132,6 +132,30 @@ class SGD(Optimizer):
                 upd = g
             w.sub_((lr * upd).to(w.dtype))
 
+    def update_multi_precision(self, index, weight, grad, state):
+        """Single fused HIP kernel on GPU (reference mp_sgd_mom_update):
+        rescale+clip+wd+momentum+master-weight update+fp16 cast in one pass."""
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        if w.is_cuda and g.dtype == w.dtype:
+            from ..ops.dispatch import hipops, use_hip
+            ext = hipops() if use_hip(w) else None
+            if ext is not None:
+                self._update_count(index)
+                lr, wd = self._get_lr(index), self._get_wd(index)
+                if self.multi_precision and isinstance(state, tuple) and \
+                        isinstance(state[0], torch.Tensor) and \
+                        state[0].dtype is torch.float32 and \
+                        w.dtype in (torch.float16, torch.bfloat16):
+                    master, mom = state
+                else:
+                    master, mom = None, state
+                ext.sgd_update(w, master, g.contiguous(), mom, lr,
+                               self.momentum, wd, self.rescale_grad,
+                               self.clip_gradient or 0.0)
+                return
+        super().update_multi_precision(index, weight, grad, state)
+
 
 @register
 class NAG(SGD):

@@ -1,0 +1,397 @@
+"""Numerics tests for the gfx950 HIP kernels against plain-PyTorch fp32
+oracles (reference test strategy: check_consistency, test_utils.py:1490).
+
+Every test here runs on a real MI355X (pytest -m gpu via gpurun).
+"""
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from mxnet_amd import _hipops as ext
+else:
+    ext = None
+
+DEV = 'cuda:0'
+
+
+def mk(shape, dtype=torch.float16, scale=1.0, seed=None):
+    if seed is not None:
+        torch.manual_seed(seed)
+    return (torch.randn(*shape, device=DEV) * scale).to(dtype)
+
+
+def check(got, want32, tol=None):
+    """got: GPU tensor; want32: fp32 oracle (any device)."""
+    got32 = got.float().cpu()
+    want32 = want32.float().cpu()
+    assert got32.shape == want32.shape, (got32.shape, want32.shape)
+    scale = want32.abs().max().item() + 1e-6
+    if tol is None:
+        tol = 2e-3 if got.dtype in (torch.float16, torch.bfloat16) else 1e-5
+        if got.dtype is torch.bfloat16:
+            tol = 1.6e-2
+    err = (got32 - want32).abs().max().item()
+    assert err <= tol * scale, f'max err {err:.4g} vs tol {tol * scale:.4g} (scale {scale:.4g})'
+
+
+# ---------------------------------------------------------------------------
+# GEMM family
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('dtype', [torch.float16, torch.bfloat16, torch.float32])
+@pytest.mark.parametrize('mnk', [(37, 53, 64), (128, 128, 128),
+                                 (256, 1000, 2048), (130, 70, 72),
+                                 (512, 256, 24)])
+def test_gemm_nt(dtype, mnk):
+    M, N, K = mnk
+    a, b = mk((M, K), dtype, seed=0), mk((N, K), dtype, seed=1)
+    bias = mk((N,), dtype, seed=2)
+    y = ext.gemm_nt(a, b, bias)
+    want = a.float() @ b.float().t() + bias.float()
+    check(y, want)
+
+
+def test_gemm_variants():
+    a = mk((96, 200), seed=3)
+    b = mk((200, 72), seed=4)
+    check(ext.gemm(a, b), a.float() @ b.float())
+    dy = mk((64, 120), seed=5)
+    w = mk((120, 88), seed=6)
+    check(ext.gemm_nn(dy, w), dy.float() @ w.float())
+    x = mk((64, 88), seed=7)
+    check(ext.gemm_tn(dy, x), dy.float().t() @ x.float())
+
+
+def test_bgemm():
+    a = mk((6, 33, 48), seed=8)
+    b = mk((6, 48, 29), seed=9)
+    check(ext.bgemm(a, b), torch.matmul(a.float(), b.float()))
+
+
+def test_transpose2d():
+    x = mk((123, 77), seed=10)
+    assert torch.equal(ext.transpose2d(x), x.t().contiguous())
+    x3 = mk((4, 50, 66), seed=11)
+    assert torch.equal(ext.transpose2d(x3), x3.transpose(1, 2).contiguous())
+
+
+# ---------------------------------------------------------------------------
+# convolution (NHWC) vs torch fp32 NCHW oracle
+# ---------------------------------------------------------------------------
+CONV_CASES = [
+    # (N, H, W, C, K, R, S, stride, pad, dil)  -- ResNet-50 v1.5 shapes
+    (2, 56, 56, 64, 64, 1, 1, 1, 0, 1),       # 1x1 s1 (plain GEMM path)
+    (2, 56, 56, 64, 64, 3, 3, 1, 1, 1),       # 3x3 s1 (implicit)
+    (2, 56, 56, 256, 512, 1, 1, 2, 0, 1),     # 1x1 s2 (implicit gather)
+    (2, 28, 28, 128, 128, 3, 3, 2, 1, 1),     # 3x3 s2
+    (2, 32, 32, 3, 64, 7, 7, 2, 3, 1),        # stem RGB (im2col path)
+    (2, 16, 16, 72, 40, 3, 3, 1, 1, 1),       # C%8==0 but not %64
+    (1, 20, 20, 64, 32, 3, 3, 1, 2, 2),       # dilation 2
+    (2, 9, 9, 16, 24, 2, 2, 2, 0, 1),         # even kernel, odd spatial
+]
+
+
+def conv_oracle(x, w, stride, pad, dil):
+    xn = x.float().permute(0, 3, 1, 2)
+    wn = w.float().permute(0, 3, 1, 2)
+    y = torch.nn.functional.conv2d(xn, wn, None, stride=stride, padding=pad,
+                                   dilation=dil)
+    return y.permute(0, 2, 3, 1)
+
+
+@pytest.mark.parametrize('case', CONV_CASES)
+def test_conv_fwd(case):
+    N, H, W, C, K, R, S, st, pd, dl = case
+    x = mk((N, H, W, C), seed=20)
+    w = mk((K, R, S, C), seed=21, scale=0.5)
+    y = ext.conv2d_nhwc_fwd(x, w, None, st, st, pd, pd, dl, dl, 1)
+    check(y, conv_oracle(x, w, st, pd, dl))
+
+
+@pytest.mark.parametrize('case', CONV_CASES)
+def test_conv_bwd(case):
+    N, H, W, C, K, R, S, st, pd, dl = case
+    x = mk((N, H, W, C), seed=22)
+    w = mk((K, R, S, C), seed=23, scale=0.5)
+    xn = x.float().permute(0, 3, 1, 2).requires_grad_(True)
+    wn = w.float().permute(0, 3, 1, 2).requires_grad_(True)
+    y = torch.nn.functional.conv2d(xn, wn, None, stride=st, padding=pd,
+                                   dilation=dl)
+    dy_n = torch.randn_like(y)
+    y.backward(dy_n)
+    dy = dy_n.permute(0, 2, 3, 1).contiguous().to(x.dtype)
+    dx = ext.conv2d_nhwc_bwd_data(dy, w, H, W, st, st, pd, pd, dl, dl, 1)
+    check(dx, xn.grad.permute(0, 2, 3, 1), tol=4e-3)
+    dw = ext.conv2d_nhwc_bwd_weight(dy, x, R, S, st, st, pd, pd, dl, dl, 1)
+    check(dw, wn.grad.permute(0, 2, 3, 1), tol=4e-3)
+
+
+# ---------------------------------------------------------------------------
+# batch norm
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('fuse_relu', [False, True])
+@pytest.mark.parametrize('with_res', [False, True])
+def test_bn_train_fwd_bwd(fuse_relu, with_res):
+    N, H, W, C = 4, 7, 9, 32
+    x = mk((N, H, W, C), seed=30)
+    res = mk((N, H, W, C), seed=31) if with_res else None
+    gamma = mk((C,), torch.float32, seed=32) + 1.0
+    beta = mk((C,), torch.float32, seed=33)
+    rmean = torch.zeros(C, device=DEV)
+    rvar = torch.ones(C, device=DEV)
+    rmean0, rvar0 = rmean.clone(), rvar.clone()
+    y, smean, sistd = ext.bn_nhwc_fwd_train(
+        x, gamma, beta, rmean, rvar, 0.9, 1e-5, fuse_relu,
+        res if res is not None else x.new_empty(0))
+    # oracle
+    x32 = x.float().requires_grad_(True)
+    res32 = res.float().requires_grad_(True) if with_res else None
+    g32 = gamma.clone().requires_grad_(True)
+    b32 = beta.clone().requires_grad_(True)
+    rm, rv = rmean0.clone(), rvar0.clone()
+    xn = x32.reshape(-1, C)
+    yo = torch.nn.functional.batch_norm(
+        xn.reshape(-1, C).t().reshape(1, C, -1), rm, rv, g32, b32,
+        training=True, momentum=0.1, eps=1e-5)
+    yo = yo.reshape(C, -1).t().reshape(N, H, W, C)
+    if with_res:
+        yo = yo + res32
+    if fuse_relu:
+        yo = torch.relu(yo)
+    check(y, yo.detach(), tol=5e-3)
+    check(rmean, rm, tol=1e-4)
+    check(rvar, rv, tol=1e-4)
+    # backward
+    dy_o = torch.randn(N, H, W, C, device=DEV)
+    yo.backward(dy_o)
+    dy = dy_o.to(x.dtype)
+    dx, dgamma, dbeta, dres = ext.bn_nhwc_bwd(dy, x, gamma, smean, sistd,
+                                              fuse_relu, y, with_res)
+    check(dx, x32.grad, tol=6e-3)
+    check(dgamma, g32.grad, tol=6e-3)
+    check(dbeta, b32.grad, tol=6e-3)
+    if with_res:
+        check(dres, res32.grad, tol=6e-3)
+
+
+def test_bn_infer():
+    N, H, W, C = 3, 5, 5, 48
+    x = mk((N, H, W, C), seed=35)
+    gamma = mk((C,), torch.float32, seed=36) + 1.0
+    beta = mk((C,), torch.float32, seed=37)
+    rmean = mk((C,), torch.float32, seed=38)
+    rvar = mk((C,), torch.float32, seed=39).abs() + 0.5
+    y = ext.bn_nhwc_fwd_infer(x, gamma, beta, rmean, rvar, 1e-5, False,
+                              x.new_empty(0))
+    want = (x.float() - rmean) / torch.sqrt(rvar + 1e-5) * gamma + beta
+    check(y, want, tol=5e-3)
+
+
+# ---------------------------------------------------------------------------
+# pooling
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('kind', ['max', 'avg'])
+def test_pool(kind):
+    N, H, W, C = 2, 13, 13, 24
+    x = mk((N, H, W, C), seed=40)
+    y, arg = ext.pool_nhwc_fwd(x, kind, 3, 3, 2, 2, 1, 1, False)
+    xn = x.float().permute(0, 3, 1, 2).requires_grad_(True)
+    if kind == 'max':
+        yo = torch.nn.functional.max_pool2d(xn, 3, 2, 1)
+    else:
+        yo = torch.nn.functional.avg_pool2d(xn, 3, 2, 1,
+                                            count_include_pad=False)
+    check(y, yo.permute(0, 2, 3, 1).detach())
+    dy_o = torch.randn_like(yo)
+    yo.backward(dy_o)
+    dy = dy_o.permute(0, 2, 3, 1).contiguous().to(x.dtype)
+    dx = ext.pool_nhwc_bwd(dy, arg, kind, H, W, 3, 3, 2, 2, 1, 1, False)
+    check(dx, xn.grad.permute(0, 2, 3, 1), tol=4e-3)
+
+
+# ---------------------------------------------------------------------------
+# softmax / layernorm / colsum
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('log', [False, True])
+@pytest.mark.parametrize('C', [8, 1000, 4096])
+def test_softmax(log, C):
+    x = mk((64, C), seed=50, scale=3.0)
+    y = ext.softmax_fwd(x, log, 1.0)
+    want = (torch.log_softmax if log else torch.softmax)(x.float(), -1)
+    check(y, want, tol=3e-3)
+    x32 = x.float().requires_grad_(True)
+    yo = (torch.log_softmax if log else torch.softmax)(x32, -1)
+    dy_o = torch.randn_like(yo)
+    yo.backward(dy_o)
+    dx = ext.softmax_bwd(dy_o.to(x.dtype), y, log, 1.0)
+    check(dx, x32.grad, tol=4e-3)
+
+
+def test_softmax_temperature():
+    x = mk((16, 100), seed=51, scale=2.0)
+    y = ext.softmax_fwd(x, False, 2.5)
+    check(y, torch.softmax(x.float() / 2.5, -1), tol=3e-3)
+
+
+def test_layernorm():
+    R, C = 128, 768
+    x = mk((R, C), seed=52)
+    gamma = mk((C,), torch.float32, seed=53) + 1.0
+    beta = mk((C,), torch.float32, seed=54)
+    y, mean, istd = ext.layernorm_fwd(x, gamma, beta, 1e-5)
+    x32 = x.float().requires_grad_(True)
+    g32 = gamma.clone().requires_grad_(True)
+    b32 = beta.clone().requires_grad_(True)
+    yo = torch.nn.functional.layer_norm(x32, (C,), g32, b32, 1e-5)
+    check(y, yo.detach(), tol=4e-3)
+    dy_o = torch.randn_like(yo)
+    yo.backward(dy_o)
+    dx, dg, db = ext.layernorm_bwd(dy_o.to(x.dtype), x, gamma, mean, istd)
+    check(dx, x32.grad, tol=6e-3)
+    check(dg, g32.grad, tol=6e-3)
+    check(db, b32.grad, tol=6e-3)
+
+
+def test_colsum():
+    x = mk((4096, 1000), seed=55)
+    check(ext.colsum(x), x.float().sum(0), tol=3e-3)
+
+
+# ---------------------------------------------------------------------------
+# elementwise
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('kind', ['relu', 'sigmoid', 'tanh', 'gelu', 'silu'])
+def test_activation(kind):
+    x = mk((3, 1000), seed=60, scale=2.0)
+    y = ext.act_fwd(x, kind)
+    import torch.nn.functional as F
+    oracle = {'relu': F.relu, 'sigmoid': torch.sigmoid, 'tanh': torch.tanh,
+              'gelu': lambda t: F.gelu(t, approximate='tanh'),
+              'silu': F.silu}[kind]
+    x32 = x.float().requires_grad_(True)
+    yo = oracle(x32)
+    check(y, yo.detach(), tol=3e-3)
+    dy_o = torch.randn_like(yo)
+    yo.backward(dy_o)
+    saved = y if kind in ('relu', 'sigmoid', 'tanh') else x
+    dx = ext.act_bwd(dy_o.to(x.dtype), saved, kind)
+    check(dx, x32.grad, tol=4e-3)
+
+
+def test_sgd_update():
+    n = 10007
+    w = mk((n,), seed=70)
+    master = w.float().clone()
+    g = mk((n,), seed=71)
+    mom = torch.randn(n, device=DEV)
+    m2, mom2 = master.clone(), mom.clone()
+    ext.sgd_update(w, master, g, mom, 0.1, 0.9, 1e-4, 0.5, 0.0)
+    # oracle
+    ge = g.float() * 0.5 + 1e-4 * m2
+    mom2 = mom2 * 0.9 + ge
+    m2 = m2 - 0.1 * mom2
+    check(master, m2, tol=1e-5)
+    check(mom, mom2, tol=1e-5)
+    check(w, m2, tol=2e-3)
+
+
+def test_adam_update():
+    n = 4099
+    w = torch.randn(n, device=DEV)
+    g = torch.randn(n, device=DEV)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+    w2, m2, v2 = w.clone(), m.clone(), v.clone()
+    ext.adam_update(w, None, g, m, v, 0.01, 0.9, 0.999, 1e-8, 0.0, 1.0, 0.0,
+                    False)
+    m2 = 0.9 * m2 + 0.1 * g
+    v2 = 0.999 * v2 + 0.001 * g * g
+    w2 = w2 - 0.01 * m2 / (v2.sqrt() + 1e-8)
+    check(w, w2, tol=1e-5)
+
+
+def test_multi_all_finite():
+    a = mk((100,), seed=72)
+    b = mk((50,), seed=73)
+    assert ext.multi_all_finite([a, b])
+    b[7] = float('inf')
+    assert not ext.multi_all_finite([a, b])
+    b[7] = float('nan')
+    assert not ext.multi_all_finite([a, b])
+
+
+def test_lstm_cell():
+    N, H = 8, 64
+    gates = mk((N, 4 * H), seed=74)
+    c = mk((N, H), seed=75)
+    h2, c2 = ext.lstm_cell_fwd(gates, c)
+    i, f, g, o = gates.float().split(H, dim=-1)
+    i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
+    g = torch.tanh(g)
+    cn = f * c.float() + i * g
+    check(c2, cn, tol=3e-3)
+    check(h2, o * torch.tanh(cn), tol=3e-3)
+
+
+def test_dropout():
+    x = torch.ones(100000, device=DEV, dtype=torch.float16)
+    y, mask = ext.dropout_fwd(x, 0.3, 1234)
+    keep = mask.float().mean().item()
+    assert abs(keep - 0.7) < 0.02
+    got = y.float()
+    assert torch.allclose(got[mask.bool()],
+                          torch.full_like(got[mask.bool()], 1 / 0.7),
+                          atol=1e-3)
+    dy = torch.ones_like(x)
+    dx = ext.dropout_bwd(dy, mask, 0.3)
+    assert torch.allclose(dx.float(), mask.float() / 0.7, atol=1e-3)
+
+
+def test_embedding():
+    V, D = 1000, 96
+    w = mk((V, D), seed=76)
+    idx = torch.randint(0, V, (4, 37), device=DEV)
+    y = ext.embedding_fwd(idx, w)
+    check(y, w.float()[idx])
+    dy = mk((4, 37, D), seed=77)
+    dw = ext.embedding_bwd(idx, dy, V)
+    want = torch.zeros(V, D, device=DEV)
+    want.index_add_(0, idx.reshape(-1), dy.float().reshape(-1, D))
+    check(dw, want, tol=4e-3)
+
+
+# ---------------------------------------------------------------------------
+# integration: one ResNet bottleneck through the framework, GPU fp16 vs
+# CPU fp32 oracle
+# ---------------------------------------------------------------------------
+def test_resnet_block_integration():
+    import mxnet_amd as mx
+    from mxnet_amd import autograd
+    from mxnet_amd.gluon.model_zoo.vision.resnet import BottleneckV1
+
+    torch.manual_seed(99)
+    blk = BottleneckV1(64, stride=2, downsample=True, in_channels=32,
+                       layout='NHWC')
+    blk.initialize(ctx=mx.cpu())
+    x_cpu = mx.nd.array(torch.randn(2, 16, 16, 32))
+    with autograd.record():
+        y_cpu = blk(x_cpu)
+    y_cpu.backward()
+    g_cpu = {k: p.grad().asnumpy() for k, p in blk.collect_params().items()}
+
+    blk.reset_ctx(mx.gpu(0))
+    blk.cast('float16')
+    x_gpu = mx.nd.from_torch(x_cpu.handle.to(DEV).half())
+    with autograd.record():
+        y_gpu = blk(x_gpu)
+    y_gpu.backward()
+
+    np.testing.assert_allclose(y_gpu.asnumpy().astype(np.float32),
+                               y_cpu.asnumpy(), rtol=0.1, atol=0.05)
+    for k, p in blk.collect_params().items():
+        got = p.grad().asnumpy().astype(np.float32)
+        want = g_cpu[k]
+        scale = np.abs(want).max() + 1e-6
+        assert np.abs(got - want).max() < 0.08 * scale, k
