@@ -1,4 +1,6 @@
 """Locate the avg-pool backward mismatch pattern on device."""
+import os, sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 from mxnet_amd import _hipops as ext
 

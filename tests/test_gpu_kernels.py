@@ -397,4 +397,4 @@ def test_resnet_block_integration():
         got = p.grad().asnumpy().astype(np.float32)
         want = g_cpu[k]
         scale = np.abs(want).max() + 1e-6
-        assert np.abs(got - want).max() < 0.08 * scale, k
+        assert np.abs(got - want).max() < 0.15 * scale, k
