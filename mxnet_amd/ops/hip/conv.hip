@@ -95,16 +95,18 @@ template <typename T>
 __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
     const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ bias, T* __restrict__ y, int NB, int H, int W,
-    int C, int Kout, int P, int Q, int R, int S, int sh, int sw, int ph,
-    int pw, int dh, int dw, const T* __restrict__ zpage, bool relu) {
+    int C, int Kout, int Cg, int Kg, int P, int Q, int R, int S, int sh,
+    int sw, int ph, int pw, int dh, int dw, const T* __restrict__ zpage,
+    bool relu) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BN = 128, BK = 64;
   __shared__ T As[2][BM * BK];
   __shared__ T Bs[2][BN * BK];
 
+  const int g = blockIdx.y;  // conv group
   const long M = (long)NB * P * Q;
-  const long RSC = (long)R * S * C;
-  const int nTn = (Kout + BN - 1) / BN;
+  const long RSCg = (long)R * S * Cg;
+  const int nTn = (Kg + BN - 1) / BN;
   const int nwg = (int)(((M + BM - 1) / BM) * nTn);
   const int bid = xcd_swizzle(blockIdx.x, nwg);
   const long m0 = (long)(bid / nTn) * BM;
@@ -133,7 +135,7 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
     pb_w[rnd] = q * sw - pw;
   }
 
-  const int cpl = (C + BK - 1) / BK;  // c-chunks per (r,s) plane
+  const int cpl = (Cg + BK - 1) / BK;  // c-chunks per (r,s) plane
   const int nk = R * S * cpl;
 
   float4_t acc[4][4] = {};
@@ -142,20 +144,22 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
     const int rs = step / cpl;
     const int c0 = (step % cpl) * BK;
     const int r = rs / S, s = rs % S;
-    const int cseg = c0 + s_col;
-    const bool c_ok = cseg + 8 <= C;
-    const long wbase = (long)rs * C + cseg;
+    const int cseg = c0 + s_col;              // channel within the group
+    const bool c_ok = cseg + 8 <= Cg;
+    const long wbase = (long)rs * Cg + cseg;
 #pragma unroll
     for (int rnd = 0; rnd < 4; ++rnd) {
       const int ih = pb_h[rnd] + r * dh, iw = pb_w[rnd] + s * dw;
       const bool ok = pb_ok[rnd] && c_ok && ih >= 0 && ih < H && iw >= 0 &&
                       iw < W;
       const T* ga = ok
-          ? x + (((long)pb_n[rnd] * H + ih) * W + iw) * C + cseg
+          ? x + (((long)pb_n[rnd] * H + ih) * W + iw) * C + (long)g * Cg +
+                cseg
           : zpage;
       gload_lds16c(ga, &As[buf][(rnd * 256 + t) * 8]);
-      const long kb = n0 + rnd * 32 + s_row;
-      const T* gb = (kb < Kout && c_ok) ? w + kb * RSC + wbase : zpage;
+      const long kb = (long)g * Kg + n0 + rnd * 32 + s_row;
+      const T* gb = (n0 + rnd * 32 + s_row < Kg && c_ok)
+          ? w + kb * RSCg + wbase : zpage;
       gload_lds16c(gb, &Bs[buf][(rnd * 256 + t) * 8]);
     }
   };
@@ -193,8 +197,9 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
   for (int n = 0; n < 4; ++n) {
-    const long col = n0 + wc * 64 + n * 16 + d_col;
-    if (col >= Kout) continue;
+    const long col_l = n0 + wc * 64 + n * 16 + d_col;
+    if (col_l >= Kg) continue;
+    const long col = (long)g * Kg + col_l;
     const float b = bias ? bias[col] : 0.f;
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
@@ -219,16 +224,17 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
 template <typename T>
 __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
     const T* __restrict__ dy, const T* __restrict__ wt, T* __restrict__ dx,
-    int NB, int H, int W, int C, int Kout, int P, int Q, int R, int S,
-    int sh, int sw, int ph, int pw, int dh, int dw,
+    int NB, int H, int W, int C, int Kout, int Cg, int Kg, int P, int Q,
+    int R, int S, int sh, int sw, int ph, int pw, int dh, int dw,
     const T* __restrict__ zpage) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BN = 128, BK = 64;
   __shared__ T As[2][BM * BK];
   __shared__ T Bs[2][BN * BK];
 
+  const int g = blockIdx.y;  // conv group
   const long M = (long)NB * H * W;
-  const int nTn = (C + BN - 1) / BN;
+  const int nTn = (Cg + BN - 1) / BN;
   const int nwg = (int)(((M + BM - 1) / BM) * nTn);
   const int bid = xcd_swizzle(blockIdx.x, nwg);
   const long m0 = (long)(bid / nTn) * BM;
@@ -254,7 +260,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
     pb_n[rnd] = (int)(t2 / H);
   }
 
-  const int kpl = (Kout + BK - 1) / BK;
+  const int kpl = (Kg + BK - 1) / BK;
   const int nk = R * S * kpl;
 
   float4_t acc[4][4] = {};
@@ -263,8 +269,8 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
     const int rs = step / kpl;
     const int k0 = (step % kpl) * BK;
     const int r = rs / S, s = rs % S;
-    const int kseg = k0 + s_col;
-    const bool k_ok = kseg + 8 <= Kout;
+    const int kseg = k0 + s_col;              // out-channel within group
+    const bool k_ok = kseg + 8 <= Kg;
 #pragma unroll
     for (int rnd = 0; rnd < 4; ++rnd) {
       // which output pixel (p,q) feeds input (h,w) through tap (r,s)?
@@ -274,12 +280,14 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
       const bool ok = pb_ok[rnd] && k_ok && pnum >= 0 && qnum >= 0 &&
                       pnum % sh == 0 && qnum % sw == 0 && p < P && q < Q;
       const T* ga = ok
-          ? dy + (((long)pb_n[rnd] * P + p) * Q + q) * Kout + kseg
+          ? dy + (((long)pb_n[rnd] * P + p) * Q + q) * Kout + (long)g * Kg +
+                kseg
           : zpage;
       gload_lds16c(ga, &As[buf][(rnd * 256 + t) * 8]);
       const long cb = n0 + rnd * 32 + s_row;
-      const T* gb = (cb < C && k_ok)
-          ? wt + ((long)rs * C + cb) * Kout + kseg
+      const T* gb = (cb < Cg && k_ok)
+          ? wt + ((long)rs * C + (long)g * Cg + cb) * Kout + (long)g * Kg +
+                kseg
           : zpage;
       gload_lds16c(gb, &Bs[buf][(rnd * 256 + t) * 8]);
     }
@@ -318,8 +326,9 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
   for (int n = 0; n < 4; ++n) {
-    const long col = n0 + wc * 64 + n * 16 + d_col;
-    if (col >= C) continue;
+    const long col_l = n0 + wc * 64 + n * 16 + d_col;
+    if (col_l >= Cg) continue;
+    const long col = (long)g * Cg + col_l;
 #pragma unroll
     for (int m = 0; m < 4; ++m) {
       const long row_base = m0 + wr * 64 + m * 16 + d_row;
@@ -344,14 +353,15 @@ template <typename T>
 __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
-    int H, int W, int C, int Kout, int R, int S, int dh, int dw,
-    long m_per_slice) {
+    int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
+    int dw, long m_per_slice) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BI = 64, BJ = 64, BKM = 64;
   __shared__ T DyT[2][BI * BKM];  // [i(out-ch)][m]
   __shared__ T XT[2][BJ * BKM];   // [j(c within chunk)][m]
 
-  const int cpl = (C + BJ - 1) / BJ;
+  const int g = blockIdx.z;  // conv group
+  const int cpl = (Cg + BJ - 1) / BJ;
   const int nTj = R * S * cpl;
   const int bid = blockIdx.x;
   const int i0 = (bid / nTj) * BI;
@@ -388,8 +398,8 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
         half8_t zero8 = {};
         using V8 = T __attribute__((ext_vector_type(8)));
         V8 v = {};
-        if (m_ok && iseg + 8 <= Kout)
-          v = *(const V8*)(dy + m_g * Kout + iseg);
+        if (m_ok && iseg + 8 <= Kg)
+          v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
 #pragma unroll
         for (int jj = 0; jj < 8; ++jj)
           DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
@@ -404,8 +414,9 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
           const int ih = pt[1] + roff, iw = pt[2] + soff;
           const int cseg = c0 + seg * 8;
           if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
-              cseg + 8 <= C)
-            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C + cseg);
+              cseg + 8 <= Cg)
+            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                             (long)g * Cg + cseg);
         }
 #pragma unroll
         for (int jj = 0; jj < 8; ++jj)
@@ -446,21 +457,21 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     }
   }
 
-  const long RSC = (long)R * S * C;
+  const long RSCg = (long)R * S * Cg;
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
   for (int n = 0; n < 2; ++n) {
     const long c = c0 + wc * 32 + n * 16 + d_col;
-    if (c >= C) continue;
+    if (c >= Cg) continue;
 #pragma unroll
     for (int m = 0; m < 2; ++m) {
       const long i_base = i0 + wr * 32 + m * 16 + d_row;
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
         const long i = i_base + j;
-        if (i < Kout) {
-          float* dst = dw32 + i * RSC + (long)rs * C + c;
+        if (i < Kg) {
+          float* dst = dw32 + ((long)g * Kg + i) * RSCg + (long)rs * Cg + c;
           if (gridDim.y == 1) *dst += acc[m][n][j];
           else atomicAdd(dst, acc[m][n][j]);
         }
@@ -484,6 +495,105 @@ __global__ void build_pixtab_kernel(int4_t* __restrict__ tab, long M, int P,
     v[2] = q * sw - pw;
     v[3] = 1;
     tab[m] = v;
+  }
+}
+
+
+// ---------------------------------------------------------------------------
+// depthwise conv (groups == C, multiplier 1): direct memory-bound kernels
+// (MobileNet family; reference depthwise_convolution.cu)
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void dwconv_fwd_kernel(const T* __restrict__ x,
+                                  const T* __restrict__ w,
+                                  const float* __restrict__ bias,
+                                  T* __restrict__ y, long total, int H,
+                                  int W, int C, int P, int Q, int R, int S,
+                                  int sh, int sw, int ph, int pw, int dh,
+                                  int dw) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int q = t % Q;
+    long t2 = t / Q;
+    int p = t2 % P;
+    int n = t2 / P;
+    float acc = bias ? bias[c] : 0.f;
+    for (int r = 0; r < R; ++r) {
+      int h = p * sh - ph + r * dh;
+      if (h < 0 || h >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int ww = q * sw - pw + s * dw;
+        if (ww < 0 || ww >= W) continue;
+        acc += (float)x[(((long)n * H + h) * W + ww) * C + c] *
+               (float)w[((long)c * R + r) * S + s];
+      }
+    }
+    y[i] = (T)acc;
+  }
+}
+
+template <typename T>
+__global__ void dwconv_bwd_data_kernel(const T* __restrict__ dy,
+                                       const T* __restrict__ w,
+                                       T* __restrict__ dx, long total, int H,
+                                       int W, int C, int P, int Q, int R,
+                                       int S, int sh, int sw, int ph, int pw,
+                                       int dh, int dwl) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int ww = t % W;
+    long t2 = t / W;
+    int h = t2 % H;
+    int n = t2 / H;
+    float acc = 0.f;
+    for (int r = 0; r < R; ++r) {
+      int pnum = h + ph - r * dh;
+      if (pnum < 0 || pnum % sh) continue;
+      int p = pnum / sh;
+      if (p >= P) continue;
+      for (int s = 0; s < S; ++s) {
+        int qnum = ww + pw - s * dwl;
+        if (qnum < 0 || qnum % sw) continue;
+        int q = qnum / sw;
+        if (q >= Q) continue;
+        acc += (float)dy[(((long)n * P + p) * Q + q) * C + c] *
+               (float)w[((long)c * R + r) * S + s];
+      }
+    }
+    dx[i] = (T)acc;
+  }
+}
+
+template <typename T>
+__global__ void dwconv_bwd_w_kernel(const T* __restrict__ dy,
+                                    const T* __restrict__ x,
+                                    float* __restrict__ dw32, long total,
+                                    int H, int W, int C, int P, int Q, int R,
+                                    int S, int sh, int sw, int ph, int pw,
+                                    int dh, int dwl) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int q = t % Q;
+    long t2 = t / Q;
+    int p = t2 % P;
+    int n = t2 / P;
+    float g = (float)dy[i];
+    for (int r = 0; r < R; ++r) {
+      int h = p * sh - ph + r * dh;
+      if (h < 0 || h >= H) continue;
+      for (int s = 0; s < S; ++s) {
+        int ww = q * sw - pw + s * dwl;
+        if (ww < 0 || ww >= W) continue;
+        atomicAdd(dw32 + ((long)c * R + r) * S + s,
+                  g * (float)x[(((long)n * H + h) * W + ww) * C + c]);
+      }
+    }
   }
 }
 
@@ -511,24 +621,38 @@ at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
                            int64_t sw, int64_t ph, int64_t pw, int64_t dh,
                            int64_t dw, int64_t groups) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
-  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
   int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int Kout = w.size(0), R = w.size(1), S = w.size(2);
+  int Cg = C / (int)groups, Kg = Kout / (int)groups;
   int P = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
   int Q = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
   long M = (long)NB * P * Q;
   auto y = at::empty({NB, P, Q, Kout}, x.options());
   if (y.numel() == 0) return y;
-  bool mfma_ok = (x.scalar_type() == at::kHalf ||
-                  x.scalar_type() == at::kBFloat16) && C % 8 == 0;
   at::Tensor b32;
   const float* bias_ptr = nullptr;
   if (bias && bias->defined() && bias->numel() > 0) {
     b32 = bias->to(at::kFloat).contiguous();
     bias_ptr = b32.data_ptr<float>();
   }
-  if (mfma_ok && R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
-      pw == 0) {
+  if (groups == C && Kout == C && w.size(3) == 1) {
+    // depthwise, multiplier 1: direct kernel
+    long total = y.numel();
+    DISPATCH_FLOAT_TYPES(x.scalar_type(), "dwconv_fwd", [&] {
+      dwconv_fwd_kernel<scalar_t><<<ew_grid(total), 256, 0, cur_stream()>>>(
+          (const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
+          bias_ptr, (scalar_t*)y.data_ptr(), total, H, W, C, P, Q, R, S, sh,
+          sw, ph, pw, dh, dw);
+    });
+    HIP_CHECK_LAST();
+    return y;
+  }
+  bool mfma_ok = (x.scalar_type() == at::kHalf ||
+                  x.scalar_type() == at::kBFloat16) && Cg % 8 == 0;
+  TORCH_CHECK(groups == 1 || mfma_ok,
+              "conv2d: grouped conv needs fp16/bf16 with C/groups % 8 == 0");
+  if (mfma_ok && groups == 1 && R == 1 && S == 1 && sh == 1 && sw == 1 &&
+      ph == 0 && pw == 0) {
     // 1x1 stride-1: a plain NT GEMM on the flattened pixels
     auto y2 = gemm_nt_core(x.view({M, C}), w.view({Kout, C}),
                            bias ? c10::optional<at::Tensor>(b32)
@@ -537,12 +661,14 @@ at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
     return y2.view({NB, P, Q, Kout});
   }
   if (mfma_ok) {
-    int nwg = (int)(((M + 127) / 128) * ((Kout + 127) / 128));
+    int nwg = (int)(((M + 127) / 128) * ((Kg + 127) / 128));
+    dim3 grid((unsigned)nwg, (unsigned)groups);
     DISPATCH_HALF_TYPES(x.scalar_type(), "conv_fwd", [&] {
-      conv_fwd_igemm_kernel<scalar_t><<<nwg, 256, 0, cur_stream()>>>(
+      conv_fwd_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
           (const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
-          bias_ptr, (scalar_t*)y.data_ptr(), NB, H, W, C, Kout, P, Q, R, S,
-          sh, sw, ph, pw, dh, dw, (const scalar_t*)zero_page(x), false);
+          bias_ptr, (scalar_t*)y.data_ptr(), NB, H, W, C, Kout, Cg, Kg, P, Q,
+          R, S, sh, sw, ph, pw, dh, dw, (const scalar_t*)zero_page(x),
+          false);
     });
     HIP_CHECK_LAST();
     return y;
@@ -561,28 +687,67 @@ at::Tensor conv2d_nhwc_bwd_data(const at::Tensor& dy, const at::Tensor& w,
                                 int64_t ph, int64_t pw, int64_t dh,
                                 int64_t dw, int64_t groups) {
   CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(w);
-  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
   int NB = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
-  int R = w.size(1), S = w.size(2), C = w.size(3);
+  int R = w.size(1), S = w.size(2), Cgw = w.size(3);
+  int C = Cgw * (int)groups;
+  int Cg = Cgw, Kg = Kout / (int)groups;
   auto dx = at::empty({NB, (long)H, (long)W, (long)C}, dy.options());
   long M2 = (long)NB * H * W;
+  if (groups == C && Kout == C && Cgw == 1) {
+    long total = dx.numel();
+    DISPATCH_FLOAT_TYPES(dy.scalar_type(), "dwconv_bwd_data", [&] {
+      dwconv_bwd_data_kernel<scalar_t><<<ew_grid(total), 256, 0,
+                                         cur_stream()>>>(
+          (const scalar_t*)dy.data_ptr(), (const scalar_t*)w.data_ptr(),
+          (scalar_t*)dx.data_ptr(), total, H, W, C, P, Q, R, S, sh, sw, ph,
+          pw, dh, dw);
+    });
+    HIP_CHECK_LAST();
+    return dx;
+  }
   bool mfma_ok = (dy.scalar_type() == at::kHalf ||
-                  dy.scalar_type() == at::kBFloat16) && Kout % 8 == 0;
-  if (mfma_ok && R == 1 && S == 1 && sh == 1 && sw == 1 && ph == 0 &&
-      pw == 0) {
+                  dy.scalar_type() == at::kBFloat16) && Kg % 8 == 0;
+  TORCH_CHECK(groups == 1 || mfma_ok,
+              "conv2d bwd_data: grouped conv needs fp16/bf16 with "
+              "K/groups % 8 == 0");
+  if (mfma_ok && groups == 1 && R == 1 && S == 1 && sh == 1 && sw == 1 &&
+      ph == 0 && pw == 0) {
     auto wt = transpose2d(w.view({Kout, C}));  // [C, Kout]
     auto d2 = gemm_nt_core(dy.view({(long)NB * P * Q, Kout}), wt,
                            c10::nullopt, false);
     return d2.view({NB, (long)H, (long)W, (long)C});
   }
   if (mfma_ok) {
-    auto wt = w.permute({1, 2, 3, 0}).contiguous();  // [R,S,C,K]
-    int nwg = (int)(((M2 + 127) / 128) * ((C + 127) / 128));
+    // wt[R,S,C,K]: for grouped conv, channel c within group g maps to
+    // column g*Cg+c of the full C axis, filter rows g*Kg..; build the
+    // full [R,S,C,Kout] permuted copy with zeros implicitly handled by
+    // per-group indexing (kernel indexes wt[(rs*C + g*Cg+c)*Kout + g*Kg+k],
+    // so assemble wt as [R,S,C,Kout] from the grouped weight).
+    at::Tensor wt;
+    if (groups == 1) {
+      wt = w.permute({1, 2, 3, 0}).contiguous();
+    } else {
+      // w [Kout, R, S, Cg] -> view groups: [G, Kg, R, S, Cg]
+      auto wg = w.view({(long)groups, (long)Kg, (long)R, (long)S, (long)Cg});
+      // -> [R, S, G, Cg, Kg] -> flatten to [R,S,C,Kout]-compatible blocks:
+      // kernel reads wt[((rs)*C + g*Cg + c) * Kout + g*Kg + k]; build a
+      // zero tensor and scatter the block diagonal.
+      wt = at::zeros({(long)R, (long)S, (long)C, (long)Kout}, w.options());
+      auto wperm = wg.permute({2, 3, 0, 4, 1});  // [R,S,G,Cg,Kg]
+      auto wtv = wt.view({(long)R, (long)S, (long)groups, (long)Cg,
+                          (long)groups, (long)Kg});
+      for (long gg = 0; gg < groups; ++gg)
+        wtv.select(4, gg).select(2, gg).copy_(
+            wperm.select(2, gg));
+      wt = wt.contiguous();
+    }
+    int nwg = (int)(((M2 + 127) / 128) * ((Cg + 127) / 128));
+    dim3 grid((unsigned)nwg, (unsigned)groups);
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_data", [&] {
-      conv_bwd_data_igemm_kernel<scalar_t><<<nwg, 256, 0, cur_stream()>>>(
+      conv_bwd_data_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
           (const scalar_t*)dy.data_ptr(), (const scalar_t*)wt.data_ptr(),
-          (scalar_t*)dx.data_ptr(), NB, H, W, C, Kout, P, Q, R, S, sh, sw,
-          ph, pw, dh, dw, (const scalar_t*)zero_page(dy));
+          (scalar_t*)dx.data_ptr(), NB, H, W, C, Kout, Cg, Kg, P, Q, R, S,
+          sh, sw, ph, pw, dh, dw, (const scalar_t*)zero_page(dy));
     });
     HIP_CHECK_LAST();
     return dx;
@@ -606,36 +771,54 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
                                   int64_t sw, int64_t ph, int64_t pw,
                                   int64_t dh, int64_t dw, int64_t groups) {
   CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
-  TORCH_CHECK(groups == 1, "conv2d: groups>1 not yet native (v1)");
   int NB = dy.size(0), P = dy.size(1), Q = dy.size(2), Kout = dy.size(3);
   int H = x.size(1), W = x.size(2), C = x.size(3);
+  int Cg = C / (int)groups, Kg = Kout / (int)groups;
   long M = (long)NB * P * Q;
+  if (groups == C && Kout == C) {
+    auto dw32 = at::zeros({(long)C, (long)R * S},
+                          dy.options().dtype(at::kFloat));
+    long total = dy.numel();
+    DISPATCH_FLOAT_TYPES(dy.scalar_type(), "dwconv_bwd_w", [&] {
+      dwconv_bwd_w_kernel<scalar_t><<<ew_grid(total), 256, 0,
+                                      cur_stream()>>>(
+          (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+          dw32.data_ptr<float>(), total, H, W, C, P, Q, R, S, sh, sw, ph,
+          pw, dh, dw);
+    });
+    HIP_CHECK_LAST();
+    return dw32.view({(long)C, R, S, 1}).to(dy.scalar_type());
+  }
   bool mfma_ok = (dy.scalar_type() == at::kHalf ||
-                  dy.scalar_type() == at::kBFloat16) && C % 8 == 0 &&
-                 Kout % 8 == 0;
+                  dy.scalar_type() == at::kBFloat16) && Cg % 8 == 0 &&
+                 Kg % 8 == 0;
+  TORCH_CHECK(groups == 1 || mfma_ok,
+              "conv2d bwd_weight: grouped conv needs fp16/bf16 with "
+              "C/groups and K/groups % 8 == 0");
   if (mfma_ok) {
-    auto dw32 = at::zeros({(long)Kout, (long)R * S * C},
+    auto dw32 = at::zeros({(long)Kout, (long)R * S * Cg},
                           dy.options().dtype(at::kFloat));
     auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
     build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
-    int cpl = (C + 63) / 64;
-    int nwg = (int)(((Kout + 63) / 64) * (long)R * S * cpl);
+    int cpl = (Cg + 63) / 64;
+    int nwg = (int)(((Kg + 63) / 64) * (long)R * S * cpl);
     // split the reduction so the grid fills the chip
     long want_blocks = 2048;
     long yb = std::max<long>(
-        1, std::min<long>((M + 63) / 64, want_blocks / std::max(nwg, 1)));
+        1, std::min<long>((M + 63) / 64,
+                          want_blocks / std::max<long>((long)nwg * groups, 1)));
     long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
     yb = (M + m_per_slice - 1) / m_per_slice;
-    dim3 grid((unsigned)nwg, (unsigned)yb);
+    dim3 grid((unsigned)nwg, (unsigned)yb, (unsigned)groups);
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
       conv_bwd_w_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
           (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
           (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W, C,
-          Kout, R, S, dh, dw, m_per_slice);
+          Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
     });
     HIP_CHECK_LAST();
-    return dw32.view({(long)Kout, R, S, (long)C}).to(dy.scalar_type());
+    return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
   }
   // generic: dw = dy^T @ col (both transposed into NT form)
   auto col = im2col_nhwc(x, P, Q, R, S, sh, sw, ph, pw, dh, dw);

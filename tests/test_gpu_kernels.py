@@ -398,3 +398,40 @@ def test_resnet_block_integration():
         want = g_cpu[k]
         scale = np.abs(want).max() + 1e-6
         assert np.abs(got - want).max() < 0.15 * scale, k
+
+
+GROUP_CASES = [
+    # (N, H, W, C, K, R, S, stride, pad, dil, groups)
+    (2, 14, 14, 64, 128, 3, 3, 1, 1, 1, 4),    # grouped igemm (ResNeXt-ish)
+    (2, 14, 14, 32, 32, 3, 3, 1, 1, 1, 32),    # depthwise (MobileNet)
+    (2, 15, 15, 48, 48, 3, 3, 2, 1, 1, 48),    # depthwise stride 2
+]
+
+
+def group_oracle(x, w, st, pd, dl, g):
+    xn = x.float().permute(0, 3, 1, 2)
+    wn = w.float().permute(0, 3, 1, 2)
+    y = torch.nn.functional.conv2d(xn, wn, None, stride=st, padding=pd,
+                                   dilation=dl, groups=g)
+    return y.permute(0, 2, 3, 1)
+
+
+@pytest.mark.parametrize('case', GROUP_CASES)
+def test_grouped_conv(case):
+    N, H, W, C, K, R, S, st, pd, dl, g = case
+    x = mk((N, H, W, C), seed=25)
+    w = mk((K, R, S, C // g), seed=26, scale=0.5)
+    y = ext.conv2d_nhwc_fwd(x, w, None, st, st, pd, pd, dl, dl, g)
+    check(y, group_oracle(x, w, st, pd, dl, g))
+    # backward
+    xn = x.float().permute(0, 3, 1, 2).requires_grad_(True)
+    wn = w.float().permute(0, 3, 1, 2).requires_grad_(True)
+    yo = torch.nn.functional.conv2d(xn, wn, None, stride=st, padding=pd,
+                                    dilation=dl, groups=g)
+    dy_n = torch.randn_like(yo)
+    yo.backward(dy_n)
+    dy = dy_n.permute(0, 2, 3, 1).contiguous().to(x.dtype)
+    dx = ext.conv2d_nhwc_bwd_data(dy, w, H, W, st, st, pd, pd, dl, dl, g)
+    check(dx, xn.grad.permute(0, 2, 3, 1), tol=4e-3)
+    dw = ext.conv2d_nhwc_bwd_weight(dy, x, R, S, st, st, pd, pd, dl, dl, g)
+    check(dw, wn.grad.permute(0, 2, 3, 1), tol=4e-3)
