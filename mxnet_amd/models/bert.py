@@ -1,0 +1,187 @@
+"""BERT encoder (Gluon-NLP parity: scripts/bert / gluon-nlp BERTModel).
+
+BASELINE config 4: BERT-base fp16 seq=128 on MI355X — the hot path is
+FC GEMM (MFMA gemm_nt), fused LayerNorm, masked softmax and batched
+attention GEMMs (bgemm), all native gfx950 kernels via ops.nn.
+
+Layout: [batch, seq, hidden] (NT GEMM-friendly: hidden contiguous).
+"""
+import math
+
+from ..gluon.block import HybridBlock
+from ..gluon import nn
+from ..gluon.parameter import Parameter
+from .. import initializer as init
+
+__all__ = ['BERTEncoder', 'BERTModel', 'bert_base', 'bert_large',
+           'BERTSelfAttention', 'PositionwiseFFN']
+
+
+class BERTSelfAttention(HybridBlock):
+    """Multi-head self-attention (QKV fused into one Dense)."""
+
+    def __init__(self, units, num_heads, dropout=0.0, **kwargs):
+        super().__init__(**kwargs)
+        assert units % num_heads == 0
+        self._units = units
+        self._num_heads = num_heads
+        self._head_dim = units // num_heads
+        self.qkv = nn.Dense(3 * units, flatten=False, use_bias=True,
+                            weight_initializer=init.Normal(0.02))
+        self.proj = nn.Dense(units, flatten=False, use_bias=True,
+                             weight_initializer=init.Normal(0.02))
+        self.dropout = nn.Dropout(dropout) if dropout else None
+
+    def forward(self, x, mask=None):
+        # x: [B, S, U]
+        from ..ops import nn as F
+        import torch
+        t = x.handle if hasattr(x, 'handle') else x
+        B, S, U = t.shape
+        H, D = self._num_heads, self._head_dim
+        qkv = self.qkv(x)
+        qt = qkv.handle if hasattr(qkv, 'handle') else qkv
+        q, k, v = qt.split(U, dim=-1)
+        # [B,S,H,D] -> [B*H, S, D]
+        def heads(z):
+            return z.reshape(B, S, H, D).permute(0, 2, 1, 3) \
+                    .reshape(B * H, S, D).contiguous()
+        q, k, v = heads(q), heads(k), heads(v)
+        scores = F.batch_dot(q, k, transpose_b=True)  # [B*H, S, S]
+        scores = scores / math.sqrt(D)
+        if mask is not None:
+            m = mask.handle if hasattr(mask, 'handle') else mask
+            # mask: [B, S] valid-token mask -> broadcast [B*H, S, S]
+            m2 = m[:, None, None, :].expand(B, H, S, S).reshape(B * H, S, S)
+            att = F.masked_softmax(scores, m2, axis=-1)
+        else:
+            att = F.softmax(scores, axis=-1)
+        if self.dropout is not None:
+            att = self.dropout(self._wrap(att, x)).handle
+        out = F.batch_dot(att, v)  # [B*H, S, D]
+        out = out.reshape(B, H, S, D).permute(0, 2, 1, 3).reshape(B, S, U)
+        out = self._wrap(out.contiguous(), x)
+        return self.proj(out)
+
+    @staticmethod
+    def _wrap(t, like):
+        if hasattr(like, 'handle'):
+            from ..ndarray.ndarray import NDArray
+            return NDArray(t)
+        return t
+
+
+class PositionwiseFFN(HybridBlock):
+    def __init__(self, units, hidden_size, dropout=0.0, activation='gelu',
+                 **kwargs):
+        super().__init__(**kwargs)
+        self.ffn1 = nn.Dense(hidden_size, flatten=False,
+                             weight_initializer=init.Normal(0.02))
+        self.ffn2 = nn.Dense(units, flatten=False,
+                             weight_initializer=init.Normal(0.02))
+        self.act = nn.Activation(activation)
+        self.dropout = nn.Dropout(dropout) if dropout else None
+
+    def forward(self, x):
+        y = self.ffn2(self.act(self.ffn1(x)))
+        if self.dropout is not None:
+            y = self.dropout(y)
+        return y
+
+
+class BERTEncoderLayer(HybridBlock):
+    """Post-LN transformer layer (BERT convention)."""
+
+    def __init__(self, units, hidden_size, num_heads, dropout=0.0, **kwargs):
+        super().__init__(**kwargs)
+        self.attention = BERTSelfAttention(units, num_heads, dropout)
+        self.ln1 = nn.LayerNorm(epsilon=1e-12)
+        self.ffn = PositionwiseFFN(units, hidden_size, dropout)
+        self.ln2 = nn.LayerNorm(epsilon=1e-12)
+        self.dropout = nn.Dropout(dropout) if dropout else None
+
+    def forward(self, x, mask=None):
+        att = self.attention(x, mask)
+        if self.dropout is not None:
+            att = self.dropout(att)
+        x = self.ln1(x + att)
+        ffn = self.ffn(x)
+        x = self.ln2(x + ffn)
+        return x
+
+
+class BERTEncoder(HybridBlock):
+    def __init__(self, num_layers, units, hidden_size, num_heads,
+                 dropout=0.0, **kwargs):
+        super().__init__(**kwargs)
+        self.layers = []
+        for i in range(num_layers):
+            layer = BERTEncoderLayer(units, hidden_size, num_heads, dropout)
+            setattr(self, f'layer{i}', layer)
+            self.layers.append(layer)
+
+    def forward(self, x, mask=None):
+        for layer in self.layers:
+            x = layer(x, mask)
+        return x
+
+
+class BERTModel(HybridBlock):
+    """Embeddings + encoder + pooler + MLM/NSP heads (pretraining shape)."""
+
+    def __init__(self, vocab_size=30522, units=768, hidden_size=3072,
+                 num_layers=12, num_heads=12, max_length=512,
+                 type_vocab_size=2, dropout=0.1, **kwargs):
+        super().__init__(**kwargs)
+        self._units = units
+        self.word_embed = nn.Embedding(vocab_size, units,
+                                       weight_initializer=init.Normal(0.02))
+        self.token_type_embed = nn.Embedding(type_vocab_size, units,
+                                             weight_initializer=init.Normal(0.02))
+        self.position_embed = Parameter('position_embed',
+                                        shape=(max_length, units),
+                                        init=init.Normal(0.02))
+        self.embed_ln = nn.LayerNorm(epsilon=1e-12)
+        self.embed_dropout = nn.Dropout(dropout) if dropout else None
+        self.encoder = BERTEncoder(num_layers, units, hidden_size, num_heads,
+                                   dropout)
+        self.pooler = nn.Dense(units, activation='tanh', flatten=False,
+                               weight_initializer=init.Normal(0.02))
+        # MLM decoder (ties would share word_embed.weight; kept separate
+        # like gluon-nlp's default decoder for benchmark parity)
+        self.mlm_dense = nn.Dense(units, flatten=False, activation=None,
+                                  weight_initializer=init.Normal(0.02))
+        self.mlm_ln = nn.LayerNorm(epsilon=1e-12)
+        self.mlm_decoder = nn.Dense(vocab_size, flatten=False,
+                                    weight_initializer=init.Normal(0.02))
+        self.nsp_classifier = nn.Dense(2, flatten=False,
+                                       weight_initializer=init.Normal(0.02))
+
+    def forward(self, tokens, token_types=None, valid_mask=None):
+        from ..ndarray.ndarray import NDArray
+        t = tokens.handle if hasattr(tokens, 'handle') else tokens
+        S = t.shape[1]
+        emb = self.word_embed(tokens)
+        if token_types is not None:
+            emb = emb + self.token_type_embed(token_types)
+        pos = self.position_embed.data(
+            emb.context if hasattr(emb, 'context') else None)
+        emb = emb + NDArray(pos.handle[:S].unsqueeze(0).to(emb.handle.dtype))
+        emb = self.embed_ln(emb)
+        if self.embed_dropout is not None:
+            emb = self.embed_dropout(emb)
+        seq = self.encoder(emb, valid_mask)
+        pooled = self.pooler(NDArray(seq.handle[:, 0]))
+        mlm = self.mlm_decoder(self.mlm_ln(self.mlm_dense(seq)))
+        nsp = self.nsp_classifier(pooled)
+        return seq, pooled, mlm, nsp
+
+
+def bert_base(**kwargs):
+    return BERTModel(units=768, hidden_size=3072, num_layers=12,
+                     num_heads=12, **kwargs)
+
+
+def bert_large(**kwargs):
+    return BERTModel(units=1024, hidden_size=4096, num_layers=24,
+                     num_heads=16, **kwargs)
