@@ -85,7 +85,11 @@ class Block:
                                          force_reinit=force_reinit)
 
     def cast(self, dtype):
-        for p in self.collect_params().values():
+        # recurse so blocks can override (e.g. BatchNorm pins fp32 params
+        # under fp16 training, matching the reference AMP fp32 list)
+        for child in self._children.values():
+            child.cast(dtype)
+        for p in self._reg_params.values():
             p.cast(dtype)
         return self
 

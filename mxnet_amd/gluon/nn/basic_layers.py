@@ -112,6 +112,14 @@ class BatchNorm(HybridBlock):
         from ...ndarray import ops as F
         return self.hybrid_forward(F, x, residual, **params)
 
+    def cast(self, dtype):
+        """BN statistics/affine params stay fp32 under fp16 training
+        (reference AMP keeps BatchNorm in the fp32 list, amp/lists);
+        also lets the GPU kernels skip a per-call cast."""
+        if dtype in ('float16', 'bfloat16'):
+            return self
+        return super().cast(dtype)
+
     def __repr__(self):
         return f'BatchNorm(axis={self._axis}, fuse_relu={self._fuse_relu})'
 

@@ -36,12 +36,15 @@ DEV_INLINE void gload_lds16(const void* g, void* lds) {
 // A, B row-major, contiguous along K; K % 8 == 0 (host pads otherwise).
 // Batched via blockIdx.y with element strides (0 = broadcast).
 // ---------------------------------------------------------------------------
+// split-K: when out32 != nullptr, blockIdx.z owns k-tiles
+// [z*tiles_per_slice, ...) and accumulates fp32 partials with atomics
+// (small-M*N huge-K problems, e.g. the conv-stem weight gradient).
 template <typename T>
 __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
     long strideA, long strideB, long strideC, const T* __restrict__ zpage,
-    bool relu) {
+    bool relu, float* __restrict__ out32, int tiles_per_slice) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BN = 128, BK = 64;
   __shared__ T As[2][BM * BK];
@@ -86,15 +89,22 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     }
   };
 
-  stage(0, 0);
+  int kt0 = 0, kt1 = nk;
+  if (out32) {
+    kt0 = blockIdx.z * tiles_per_slice;
+    kt1 = min(nk, kt0 + tiles_per_slice);
+    if (kt0 >= kt1) return;
+    out32 += batch * strideC;
+  }
+  stage(0, kt0);
   __syncthreads();
 
   const int a_row = (lane & 15);
   const int k_off = (lane >> 4) * 8;
 
-  for (int kt = 0; kt < nk; ++kt) {
-    const int buf = kt & 1;
-    if (kt + 1 < nk) stage(buf ^ 1, kt + 1);
+  for (int kt = kt0; kt < kt1; ++kt) {
+    const int buf = (kt - kt0) & 1;
+    if (kt + 1 < kt1) stage(buf ^ 1, kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       Frag af[4], bf[4];
@@ -130,9 +140,13 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
       for (int j = 0; j < 4; ++j) {
         const long row = row_base + j;
         if (row < M) {
-          float v = acc[m][n][j] + b;
-          if (relu) v = fmaxf(v, 0.f);
-          C[row * N + col] = (T)v;
+          if (out32) {
+            atomicAdd(out32 + row * N + col, acc[m][n][j]);
+          } else {
+            float v = acc[m][n][j] + b;
+            if (relu) v = fmaxf(v, 0.f);
+            C[row * N + col] = (T)v;
+          }
         }
       }
     }
@@ -299,14 +313,31 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   sA = batched ? M * K : 0;
   sB = batched ? N * K : 0;
   long nwg = ((M + 127) / 128) * ((N + 127) / 128);
-  dim3 grid((unsigned)nwg, (unsigned)nb);
+  int nk_total = (int)((K + 63) / 64);
+  int ksplit = 1, tps = nk_total;
+  at::Tensor out32;
+  if (nwg * nb < 512 && nk_total > 16) {
+    ksplit = (int)std::min<long>((2048 + nwg * nb - 1) / (nwg * nb),
+                                 (nk_total + 15) / 16);
+    tps = (nk_total + ksplit - 1) / ksplit;
+    ksplit = (nk_total + tps - 1) / tps;
+    out32 = at::zeros(out.sizes(), out.options().dtype(at::kFloat));
+  }
+  dim3 grid((unsigned)nwg, (unsigned)nb, (unsigned)ksplit);
   DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt", [&] {
     gemm_nt_mfma_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
         (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
         bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
-        (const scalar_t*)zero_page(A), relu);
+        (const scalar_t*)zero_page(A), relu,
+        ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps);
   });
   HIP_CHECK_LAST();
+  if (ksplit > 1) {
+    auto o = out32;
+    if (bias_ptr) o = o + b32;
+    if (relu) o = at::relu(o);
+    out.copy_(o.to(out.scalar_type()));
+  }
   return out;
 }
 

@@ -294,9 +294,12 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
     else launch_apply(std::false_type{}, std::false_type{});
   });
   HIP_CHECK_LAST();
-  // write back running stats in their original dtype
-  rmean.copy_(rm32.to(rmean.scalar_type()));
-  rvar.copy_(rv32.to(rvar.scalar_type()));
+  // write back running stats (no-op when they were already fp32 contiguous:
+  // the kernel then updated them in place)
+  if (rm32.data_ptr() != rmean.data_ptr())
+    rmean.copy_(rm32.to(rmean.scalar_type()));
+  if (rv32.data_ptr() != rvar.data_ptr())
+    rvar.copy_(rv32.to(rvar.scalar_type()));
   return {y, save_mean, save_istd};
 }
 

@@ -51,14 +51,15 @@ class _FullyConnected(torch.autograd.Function):
     def backward(ctx, dy):
         x, w = ctx.saved_tensors
         dy = dy.contiguous()
+        need_dx, need_dw = ctx.needs_input_grad[0], ctx.needs_input_grad[1]
         if use_hip(x):
             ext = hip_required('fully_connected')
-            dx = ext.gemm_nn(dy, w)          # [M,N]@[N,K] -> [M,K]
-            dw = ext.gemm_tn(dy, x)          # [N,M]@[M,K] -> [N,K]
+            dx = ext.gemm_nn(dy, w) if need_dx else None
+            dw = ext.gemm_tn(dy, x) if need_dw else None
             db = ext.colsum(dy) if ctx.has_bias else None
         else:
-            dx = dy @ w
-            dw = dy.t() @ x
+            dx = dy @ w if need_dx else None
+            dw = dy.t() @ x if need_dw else None
             db = dy.sum(0) if ctx.has_bias else None
         return dx, dw, db
 
@@ -117,13 +118,17 @@ class _Conv2dNHWC(torch.autograd.Function):
         stride, pad, dilation, groups = ctx.conf
         dy = dy.contiguous()
         ext = hip_required('conv2d')
-        dx = ext.conv2d_nhwc_bwd_data(dy, w, x.shape[1], x.shape[2],
-                                      stride[0], stride[1], pad[0], pad[1],
-                                      dilation[0], dilation[1], groups)
-        dw = ext.conv2d_nhwc_bwd_weight(dy, x, w.shape[1], w.shape[2],
-                                        stride[0], stride[1], pad[0], pad[1],
-                                        dilation[0], dilation[1], groups)
-        db = ext.colsum(dy.reshape(-1, dy.shape[-1])) if ctx.has_bias else None
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv2d_nhwc_bwd_data(dy, w, x.shape[1], x.shape[2],
+                                          stride[0], stride[1], pad[0], pad[1],
+                                          dilation[0], dilation[1], groups)
+        if ctx.needs_input_grad[1]:
+            dw = ext.conv2d_nhwc_bwd_weight(dy, x, w.shape[1], w.shape[2],
+                                            stride[0], stride[1], pad[0], pad[1],
+                                            dilation[0], dilation[1], groups)
+        if ctx.has_bias:
+            db = ext.colsum(dy.reshape(-1, dy.shape[-1]))
         return dx, dw, db, None, None, None, None
 
 
