@@ -214,6 +214,31 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ dy,
     }
     __syncthreads();
   }
+  using VecT = T __attribute__((ext_vector_type(8)));
+  if (use_lds && C % 8 == 0) {
+    long nv = total / 8;
+    for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nv;
+         i += (long)gridDim.x * blockDim.x) {
+      long c0 = (i * 8) % C;
+      VecT vg = reinterpret_cast<const VecT*>(dy)[i];
+      VecT vx = reinterpret_cast<const VecT*>(x)[i];
+      VecT vy;
+      if (RELU) vy = reinterpret_cast<const VecT*>(y)[i];
+      VecT odx, ores;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        long c = c0 + j;
+        float g = (float)vg[j];
+        if (RELU && (float)vy[j] <= 0.f) g = 0.f;
+        float xhat = ((float)vx[j] - c_mu[c]) * c_is[c];
+        odx[j] = (T)(c_gis[c] * (g - c_s1[c] - xhat * c_s2[c]));
+        if (RES) ores[j] = (T)g;
+      }
+      reinterpret_cast<VecT*>(dx)[i] = odx;
+      if (RES) reinterpret_cast<VecT*>(dres)[i] = ores;
+    }
+    return;
+  }
   for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += (long)gridDim.x * blockDim.x) {
     long c = i % C;
@@ -373,7 +398,7 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
     auto launch_apply = [&](auto relu_c, auto res_c) {
       bn_bwd_apply_kernel<scalar_t, decltype(relu_c)::value,
                           decltype(res_c)::value>
-          <<<ew_grid(total), 256, lds, stream>>>(
+          <<<ew_grid(total / 8 + 1), 256, lds, stream>>>(
               (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
               (const scalar_t*)y.data_ptr(), (scalar_t*)dx.data_ptr(),
               has_res ? (scalar_t*)dres.data_ptr() : nullptr, total, C,
