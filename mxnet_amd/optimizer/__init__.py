@@ -407,6 +407,161 @@ class FTRL(Optimizer):
             w.copy_(wnew.to(w.dtype))
 
 
+
+@register
+class FTML(Optimizer):
+    """FTML (reference ftml_update, optimizer_op.cc)."""
+
+    def __init__(self, beta1=0.6, beta2=0.999, epsilon=1e-8, **kwargs):
+        super().__init__(**kwargs)
+        self.beta1, self.beta2, self.epsilon = beta1, beta2, epsilon
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return tuple(torch.zeros_like(w, dtype=torch.float32)
+                     for _ in range(3))
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        d, v, z = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            v.mul_(self.beta2).add_(g * g, alpha=1 - self.beta2)
+            coef = (1 - self.beta1 ** t)
+            d_t = coef * ((v / (1 - self.beta2 ** t)).sqrt() + self.epsilon) / lr
+            sigma = d_t - self.beta1 * d
+            z.mul_(self.beta1).add_((1 - self.beta1) * g - sigma * w.float())
+            d.copy_(d_t)
+            w.copy_((-z / d_t).to(w.dtype))
+
+
+@register
+class Nadam(Adam):
+    """Nesterov Adam (reference nadam in optimizer.py)."""
+
+    def __init__(self, learning_rate=0.001, schedule_decay=0.004, **kwargs):
+        super().__init__(learning_rate=learning_rate, **kwargs)
+        self.schedule_decay = schedule_decay
+        self.m_schedule = 1.0
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        m, v = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            mom_t = self.beta1 * (1 - 0.5 * 0.96 ** (t * self.schedule_decay))
+            mom_t1 = self.beta1 * (1 - 0.5 * 0.96 ** ((t + 1) * self.schedule_decay))
+            self.m_schedule *= mom_t
+            ms1 = self.m_schedule * mom_t1
+            gp = g / (1 - self.m_schedule)
+            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
+            mp = m / (1 - ms1)
+            vp = v / (1 - self.beta2 ** t)
+            mbar = (1 - mom_t) * gp + mom_t1 * mp
+            w.sub_((lr * mbar / (vp.sqrt() + self.epsilon)).to(w.dtype))
+
+
+@register
+class AdaBelief(Adam):
+    """AdaBelief (reference contrib adabelief)."""
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        lr_t = lr * math.sqrt(1 - self.beta2 ** t) / (1 - self.beta1 ** t)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        m, v = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            m.mul_(self.beta1).add_(g, alpha=1 - self.beta1)
+            diff = g - m
+            v.mul_(self.beta2).addcmul_(diff, diff, value=1 - self.beta2)
+            w.sub_((lr_t * m / (v.sqrt() + self.epsilon)).to(w.dtype))
+
+
+@register
+class LARS(SGD):
+    """Layer-wise adaptive rate scaling (reference multi_lars.cc)."""
+
+    def __init__(self, eta=0.001, epsilon=1e-8, **kwargs):
+        super().__init__(**kwargs)
+        self.eta, self.epsilon = eta, epsilon
+
+    def update(self, index, weight, grad, state):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            wnorm = w.float().norm()
+            gnorm = (g.float() * self.rescale_grad).norm()
+            if wnorm > 0 and gnorm > 0:
+                trust = self.eta * wnorm / (gnorm + self._get_wd(index) * wnorm
+                                            + self.epsilon)
+            else:
+                trust = 1.0
+        saved_lr = self.lr
+        try:
+            self.lr = float(trust) * self._get_lr(index)
+            super().update(index, weight, grad, state)
+        finally:
+            self.lr = saved_lr
+
+
+@register
+class SGLD(Optimizer):
+    """Stochastic gradient Langevin dynamics (reference sgld)."""
+
+    def create_state(self, index, weight):
+        return None
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            noise = torch.randn_like(w, dtype=torch.float32) * math.sqrt(lr)
+            w.sub_((0.5 * lr * g + noise).to(w.dtype))
+
+
+@register
+class DCASGD(Optimizer):
+    """Delay-compensated async SGD (reference dcasgd)."""
+
+    def __init__(self, momentum=0.0, lamda=0.04, **kwargs):
+        super().__init__(**kwargs)
+        self.momentum, self.lamda = momentum, lamda
+
+    def create_state(self, index, weight):
+        w = weight._t if isinstance(weight, NDArray) else weight
+        return (torch.zeros_like(w, dtype=torch.float32),
+                w.detach().float().clone())
+
+    def update(self, index, weight, grad, state):
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        mom, prev = state
+        with torch.no_grad():
+            g = self._preprocess_grad(g) + wd * w.float()
+            comp = g + self.lamda * g * g * (w.float() - prev)
+            mom.mul_(self.momentum).sub_(lr * comp)
+            prev.copy_(w.float())
+            w.add_(mom.to(w.dtype))
+
+
 # lowercase aliases matching mx.optimizer.create names
 sgd = SGD
 adam = Adam
@@ -418,6 +573,12 @@ adadelta = AdaDelta
 lamb = LAMB
 ftrl = FTRL
 signum = Signum
+ftml = FTML
+nadam = Nadam
+adabelief = AdaBelief
+lars = LARS
+sgld = SGLD
+dcasgd = DCASGD
 
 
 class Updater:
