@@ -46,11 +46,11 @@ from .util import is_np_array, set_np, use_np
 # mx.metric alias (pre-2.0 location), mirroring gluon.metric
 from .gluon import metric
 
-test_utils = None  # populated lazily to avoid import cycles
-
-
 def __getattr__(name):
-    if name == 'test_utils':
-        from . import test_utils as tu
-        return tu
+    # lazily imported to avoid import cycles
+    if name in ('test_utils', 'runtime', 'visualization'):
+        import importlib
+        mod = importlib.import_module('.' + name, __name__)
+        globals()[name] = mod
+        return mod
     raise AttributeError(name)
