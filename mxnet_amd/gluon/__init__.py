@@ -7,3 +7,4 @@ from . import nn, loss, metric
 from . import rnn
 from . import data
 from . import model_zoo
+from . import probability  # noqa: F401
