@@ -16,6 +16,62 @@
 // threads = 64 cols x 4 row-lanes.
 // Computes sum and sum-of-squares (fwd) or the two backward sums.
 // ---------------------------------------------------------------------------
+// vectorized variant (C % 8 == 0): 256 threads = 8 channel-groups x 32
+// row-lanes, each lane streams half8 (16 B - Guideline 13) and keeps 8
+// per-channel fp32 partials in registers; LDS tree-reduce over row-lanes.
+template <typename T>
+__global__ void bn_reduce_vec_kernel(const T* __restrict__ x, long M, long C,
+                                     long rows_per_block,
+                                     float* __restrict__ sum,
+                                     float* __restrict__ sumsq) {
+  using V8 = T __attribute__((ext_vector_type(8)));
+  __shared__ float b0[256][8];
+  __shared__ float b1[256][8];
+  const int t = threadIdx.x;
+  const int cg = t & 7;            // which 8-channel group
+  const int rl = t >> 3;           // row lane 0..31
+  const long c0 = (long)blockIdx.x * 64 + cg * 8;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min(M, r0 + rows_per_block);
+  float a0[8] = {}, a1[8] = {};
+  if (c0 + 8 <= C) {
+    for (long r = r0 + rl; r < r1; r += 32) {
+      V8 v = *(const V8*)(x + r * C + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float f = (float)v[j];
+        a0[j] += f;
+        a1[j] += f * f;
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    b0[t][j] = a0[j];
+    b1[t][j] = a1[j];
+  }
+  __syncthreads();
+  // threads 0..63 finalize channel c = blockIdx.x*64 + t
+  if (t < 64) {
+    long c = (long)blockIdx.x * 64 + t;
+    if (c < C) {
+      int g = t >> 3, j = t & 7;
+      float s0 = 0.f, s1 = 0.f;
+      for (int r = 0; r < 32; ++r) {
+        s0 += b0[r * 8 + g][j];
+        s1 += b1[r * 8 + g][j];
+      }
+      if (gridDim.y == 1) {
+        sum[c] = s0;
+        sumsq[c] = s1;
+      } else {
+        atomicAdd(sum + c, s0);
+        atomicAdd(sumsq + c, s1);
+      }
+    }
+  }
+}
+
 template <typename T>
 __global__ void bn_reduce_kernel(const T* __restrict__ x, long M, long C,
                                  long rows_per_block,
@@ -142,6 +198,73 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 
 // backward column reduction: s1 = sum(dy_eff), s2 = sum(dy_eff * xhat)
 // dy_eff = relu-masked dy (mask from the saved post-activation y)
+// vectorized backward reduction (same geometry as bn_reduce_vec_kernel)
+template <typename T, bool RELU>
+__global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ dy,
+                                         const T* __restrict__ x,
+                                         const T* __restrict__ y, long M,
+                                         long C, long rows_per_block,
+                                         const float* __restrict__ mean,
+                                         const float* __restrict__ istd,
+                                         float* __restrict__ s1,
+                                         float* __restrict__ s2) {
+  using V8 = T __attribute__((ext_vector_type(8)));
+  __shared__ float b1[256][8];
+  __shared__ float b2[256][8];
+  const int t = threadIdx.x;
+  const int cg = t & 7;
+  const int rl = t >> 3;
+  const long c0 = (long)blockIdx.x * 64 + cg * 8;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min(M, r0 + rows_per_block);
+  float a1[8] = {}, a2[8] = {};
+  if (c0 + 8 <= C) {
+    float mu[8], is[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      mu[j] = mean[c0 + j];
+      is[j] = istd[c0 + j];
+    }
+    for (long r = r0 + rl; r < r1; r += 32) {
+      V8 vg = *(const V8*)(dy + r * C + c0);
+      V8 vx = *(const V8*)(x + r * C + c0);
+      V8 vy;
+      if (RELU) vy = *(const V8*)(y + r * C + c0);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = (float)vg[j];
+        if (RELU && (float)vy[j] <= 0.f) g = 0.f;
+        a1[j] += g;
+        a2[j] += g * ((float)vx[j] - mu[j]) * is[j];
+      }
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    b1[t][j] = a1[j];
+    b2[t][j] = a2[j];
+  }
+  __syncthreads();
+  if (t < 64) {
+    long c = (long)blockIdx.x * 64 + t;
+    if (c < C) {
+      int g = t >> 3, j = t & 7;
+      float r1v = 0.f, r2v = 0.f;
+      for (int r = 0; r < 32; ++r) {
+        r1v += b1[r * 8 + g][j];
+        r2v += b2[r * 8 + g][j];
+      }
+      if (gridDim.y == 1) {
+        s1[c] = r1v;
+        s2[c] = r2v;
+      } else {
+        atomicAdd(s1 + c, r1v);
+        atomicAdd(s2 + c, r2v);
+      }
+    }
+  }
+}
+
 template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
@@ -290,9 +413,14 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
   dim3 grid = bn_reduce_grid(M, C, &rpb);
   auto stream = cur_stream();
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_fwd", [&] {
-    bn_reduce_kernel<scalar_t><<<grid, 256, 0, stream>>>(
-        (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
-        sumsq.data_ptr<float>());
+    if (C % 8 == 0 && sizeof(scalar_t) == 2)
+      bn_reduce_vec_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+          (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
+          sumsq.data_ptr<float>());
+    else
+      bn_reduce_kernel<scalar_t><<<grid, 256, 0, stream>>>(
+          (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
+          sumsq.data_ptr<float>());
     bn_finalize_kernel<<<(int)((C + 255) / 256), 256, 0, stream>>>(
         sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C,
         (float)momentum, (float)eps, save_mean.data_ptr<float>(),
@@ -384,12 +512,20 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto stream = cur_stream();
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_bwd", [&] {
     auto launch_red = [&](auto relu_c) {
-      bn_bwd_reduce_kernel<scalar_t, decltype(relu_c)::value>
-          <<<grid, 256, 0, stream>>>(
-              (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
-              (const scalar_t*)y.data_ptr(), M, C, rpb,
-              save_mean.data_ptr<float>(), save_istd.data_ptr<float>(),
-              s1.data_ptr<float>(), s2.data_ptr<float>());
+      if (C % 8 == 0 && sizeof(scalar_t) == 2)
+        bn_bwd_reduce_vec_kernel<scalar_t, decltype(relu_c)::value>
+            <<<grid, 256, 0, stream>>>(
+                (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+                (const scalar_t*)y.data_ptr(), M, C, rpb,
+                save_mean.data_ptr<float>(), save_istd.data_ptr<float>(),
+                s1.data_ptr<float>(), s2.data_ptr<float>());
+      else
+        bn_bwd_reduce_kernel<scalar_t, decltype(relu_c)::value>
+            <<<grid, 256, 0, stream>>>(
+                (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+                (const scalar_t*)y.data_ptr(), M, C, rpb,
+                save_mean.data_ptr<float>(), save_istd.data_ptr<float>(),
+                s1.data_ptr<float>(), s2.data_ptr<float>());
     };
     if (fuse_relu) launch_red(std::true_type{});
     else launch_red(std::false_type{});
