@@ -35,6 +35,7 @@ class Trainer:
         self._kv_initialized = False
         self._update_on_kvstore = update_on_kvstore
         self._kvstore_kind = kvstore
+        self._compression_params = compression_params
         self._distributed = False
 
     # ------------------------------------------------------------------
@@ -60,6 +61,8 @@ class Trainer:
         else:
             self._kvstore = kvs_mod.create(kind)
         self._distributed = isinstance(self._kvstore, kvs_mod.DistKVStore)
+        if self._compression_params and self._kvstore is not None:
+            self._kvstore.set_gradient_compression(self._compression_params)
         if self._distributed:
             # sync initial parameters across ranks (reference: kv.init
             # broadcasts rank-0 values)

@@ -77,6 +77,13 @@ class KVStore(KVStoreBase):
         self._data = {}           # key -> merged NDArray (on merge ctx)
         self._updater = None
         self._optimizer = None
+        self._compression = None
+
+    def set_gradient_compression(self, compression_params):
+        """2-bit/1-bit gradient compression (reference
+        kvstore.cc SetGradientCompression)."""
+        from .gradient_compression import GradientCompression
+        self._compression = GradientCompression(**compression_params)
 
     # -- init / push / pull ---------------------------------------------
     def init(self, key, value):
@@ -104,6 +111,9 @@ class KVStore(KVStoreBase):
                 self.push(k, v, priority)
             return
         values = value if isinstance(value, (list, tuple)) else [value]
+        if self._compression is not None:
+            for i, v in enumerate(values):
+                self._compression.compress_decompress((key, i), v._t)
         merged = self._reduce(values)
         if self._updater is not None:
             self._updater(key, merged, self._data[key])
@@ -192,6 +202,11 @@ class DistKVStore(KVStoreBase):
             torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
         self._comm_stream = (torch.cuda.Stream()
                              if torch.cuda.is_available() else None)
+        self._compression = None
+
+    def set_gradient_compression(self, compression_params):
+        from .gradient_compression import GradientCompression
+        self._compression = GradientCompression(**compression_params)
 
     @property
     def rank(self):
@@ -221,6 +236,9 @@ class DistKVStore(KVStoreBase):
         v = value[0] if isinstance(value, (list, tuple)) else value
         t = v._t.grad if isinstance(v._t, torch.Tensor) and v._t.grad is not None \
             and out is None else v._t
+        if self._compression is not None:
+            # lossy quantize-with-error-feedback before the collective
+            self._compression.compress_decompress(key, t.data)
         work = dist.all_reduce(t.data, op=dist.ReduceOp.SUM, async_op=async_op)
         if async_op:
             self._handles.append(work)
