@@ -14,6 +14,7 @@ from setuptools import setup
 os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
 
 from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
+from pybind11.setup_helpers import Pybind11Extension  # noqa: E402
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
 SRC = sorted(glob.glob(os.path.join(ROOT, 'mxnet_amd', 'ops', 'hip', '*.hip'))) \
@@ -29,7 +30,13 @@ setup(
                 'cxx': ['-O3', '-std=c++17'],
                 'nvcc': ['-O3', '-std=c++17', '--offload-arch=gfx950'],
             },
-        )
+        ),
+        Pybind11Extension(
+            'mxnet_amd._engine',
+            [os.path.join(ROOT, 'src', 'engine.cc')],
+            cxx_std=17,
+            extra_compile_args=['-O2', '-pthread'],
+        ),
     ],
     cmdclass={'build_ext': BuildExtension.with_options(use_ninja=True)},
 )
