@@ -481,6 +481,239 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// small-C implicit GEMM (the RGB stem): x padded to 8 channels so one
+// BK=64 chunk = 8 horizontal taps (r fixed, s = lane group, dil=1) --
+// taps are w-contiguous in NHWC so each lane's 16 B segment is one tap.
+// Weights padded to [K, R, 8, 8] (zero taps s>=S / channels c>=C make
+// the out-of-range reads harmless).
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_fwd_igemm_c8_kernel(
+    const T* __restrict__ x8, const T* __restrict__ w8,
+    const float* __restrict__ bias, T* __restrict__ y, int NB, int H, int W,
+    int Kout, int P, int Q, int R, int sh, int sw, int ph, int pw,
+    const T* __restrict__ zpage, bool relu) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BM = 128, BN = 128, BK = 64;
+  __shared__ T As[2][BM * BK];
+  __shared__ T Bs[2][BN * BK];
+
+  const long M = (long)NB * P * Q;
+  const int nTn = (Kout + BN - 1) / BN;
+  const int nwg = (int)(((M + BM - 1) / BM) * nTn);
+  const int bid = xcd_swizzle(blockIdx.x, nwg);
+  const long m0 = (long)(bid / nTn) * BM;
+  const long n0 = (long)(bid % nTn) * BN;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int s_row = t >> 3;
+  const int s_tap = t & 7;        // horizontal tap within the chunk
+
+  int pb_n[4], pb_h[4], pb_w[4];
+  bool pb_ok[4];
+#pragma unroll
+  for (int rnd = 0; rnd < 4; ++rnd) {
+    long pix = m0 + rnd * 32 + s_row;
+    pb_ok[rnd] = pix < M;
+    long pp = pb_ok[rnd] ? pix : 0;
+    int q = (int)(pp % Q);
+    long t2 = pp / Q;
+    int p = (int)(t2 % P);
+    pb_n[rnd] = (int)(t2 / P);
+    pb_h[rnd] = p * sh - ph;
+    pb_w[rnd] = q * sw - pw;
+  }
+
+  const int nk = R;
+  float4_t acc[4][4] = {};
+
+  auto stage = [&](int buf, int r) {
+#pragma unroll
+    for (int rnd = 0; rnd < 4; ++rnd) {
+      const int ih = pb_h[rnd] + r;
+      const int iw = pb_w[rnd] + s_tap;
+      const bool ok = pb_ok[rnd] && ih >= 0 && ih < H && iw >= 0 && iw < W;
+      const T* ga = ok
+          ? x8 + (((long)pb_n[rnd] * H + ih) * W + iw) * 8
+          : zpage;
+      gload_lds16c(ga, &As[buf][(rnd * 256 + t) * 8]);
+      const long kb = n0 + rnd * 32 + s_row;
+      const T* gb = kb < Kout ? w8 + (kb * R + r) * 64 + s_tap * 8 : zpage;
+      gload_lds16c(gb, &Bs[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < nk) stage(buf ^ 1, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[4], bf[4];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+        af[m] = *(const Frag*)&As[buf][(wr * 64 + m * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        bf[n] = *(const Frag*)&Bs[buf][(wc * 64 + n * 16 + a_row) * BK +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 4; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const long col = n0 + wc * 64 + n * 16 + d_col;
+    if (col >= Kout) continue;
+    const float b = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const long row_base = m0 + wr * 64 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long row = row_base + j;
+        if (row < M) {
+          float v = acc[m][n][j] + b;
+          if (relu) v = fmaxf(v, 0.f);
+          y[row * Kout + col] = (T)v;
+        }
+      }
+    }
+  }
+}
+
+// stem weight gradient: j-tile = one r row (8 taps x 8 ch); same TN
+// geometry as conv_bwd_w_igemm_kernel, x8-gathered taps.
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x8,
+    const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
+    int H, int W, int Kout, int R, long m_per_slice) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BI = 64, BKM = 64;
+  __shared__ T DyT[2][BI * BKM];
+  __shared__ T XT[2][64 * BKM];
+
+  const int nTj = R;
+  const int bid = blockIdx.x;
+  const int i0 = (bid / nTj) * BI;
+  const int r = bid % nTj;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+  const int sm_half = t >> 3;
+  const int seg = t & 7;
+
+  float4_t acc[2][2] = {};
+
+  auto stage = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      {
+        const int iseg = i0 + seg * 8;
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok && iseg + 8 <= Kout)
+          v = *(const V8*)(dy + m_g * Kout + iseg);
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+      }
+      {
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok) {
+          int4_t pt = pixtab[m_g];
+          const int ih = pt[1] + r, iw = pt[2] + seg;  // tap (r, seg)
+          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+            v = *(const V8*)(x8 + (((long)pt[0] * H + ih) * W + iw) * 8);
+        }
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          XT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+      }
+    }
+  };
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  int buf = 0;
+  stage(0, ms0);
+  __syncthreads();
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[2], bf[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+        af[m] = *(const Frag*)&DyT[buf][(wr * 32 + m * 16 + a_row) * BKM +
+                                        kk * 32 + k_off];
+#pragma unroll
+      for (int n = 0; n < 2; ++n)
+        bf[n] = *(const Frag*)&XT[buf][(wc * 32 + n * 16 + a_row) * BKM +
+                                       kk * 32 + k_off];
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    if (mc + BKM < ms1) {
+      stage(buf ^ 1, mc + BKM);
+      buf ^= 1;
+      __syncthreads();
+    }
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    const long j = wc * 32 + n * 16 + d_col;   // within the 64-wide r row
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const long i = i_base + jj;
+        if (i < Kout) {
+          float* dst = dw32 + (i * R + r) * 64 + j;
+          if (gridDim.y == 1) *dst += acc[m][n][jj];
+          else atomicAdd(dst, acc[m][n][jj]);
+        }
+      }
+    }
+  }
+}
+
 // pixtab builder: m -> (n, h_base, w_base, valid)
 __global__ void build_pixtab_kernel(int4_t* __restrict__ tab, long M, int P,
                                     int Q, int sh, int sw, int ph, int pw) {
@@ -674,8 +907,26 @@ at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
     HIP_CHECK_LAST();
     return y;
   }
-  // generic: im2col + NT GEMM (stem 7x7 C=3, fp32, odd C)
-  auto col = im2col_nhwc(x, P, Q, R, S, sh, sw, ph, pw, dh, dw);
+  bool c8_ok = (x.scalar_type() == at::kHalf ||
+                x.scalar_type() == at::kBFloat16) && groups == 1 && C < 8 &&
+               S <= 8 && dh == 1 && dw == 1;
+  if (c8_ok) {
+    // pad channels to 8 (one 16 B lane segment = one tap) and taps to 8
+    auto x8 = at::constant_pad_nd(x, {0, 8 - C});
+    auto w8 = at::constant_pad_nd(w, {0, 8 - C, 0, 8 - S})
+                  .view({(long)Kout, (long)R * 64});
+    int nwg = (int)(((M + 127) / 128) * ((Kout + 127) / 128));
+    DISPATCH_HALF_TYPES(x.scalar_type(), "conv_fwd_c8", [&] {
+     hipLaunchKernelGGL(( conv_fwd_igemm_c8_kernel<scalar_t>), dim3(nwg), dim3(256), 0, cur_stream(), 
+          (const scalar_t*)x8.data_ptr(), (const scalar_t*)w8.data_ptr(),
+          bias_ptr, (scalar_t*)y.data_ptr(), NB, H, W, Kout, P, Q, R, sh,
+          sw, ph, pw, (const scalar_t*)zero_page(x), false);
+    });
+    HIP_CHECK_LAST();
+    return y;
+  }
+  // generic: im2col + NT GEMM (fp32, odd C)
+  auto col = im2col_nhwc(x, P, Q, R, S, sh, sw, dh == 1 ? ph : ph, pw, dh, dw);
   auto y2 = gemm_nt_core(col, w.view({Kout, (long)R * S * C}),
                          bias_ptr ? c10::optional<at::Tensor>(b32)
                                   : c10::nullopt,
@@ -793,9 +1044,37 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
   bool mfma_ok = (dy.scalar_type() == at::kHalf ||
                   dy.scalar_type() == at::kBFloat16) && Cg % 8 == 0 &&
                  Kg % 8 == 0;
-  TORCH_CHECK(groups == 1 || mfma_ok,
+  TORCH_CHECK(groups == 1 || mfma_ok || (groups == C && Kout == C),
               "conv2d bwd_weight: grouped conv needs fp16/bf16 with "
               "C/groups and K/groups % 8 == 0");
+  bool c8_ok = (dy.scalar_type() == at::kHalf ||
+                dy.scalar_type() == at::kBFloat16) && groups == 1 && C < 8 &&
+               S <= 8 && dh == 1 && dw == 1 && Kout % 8 == 0;
+  if (c8_ok) {
+    auto x8 = at::constant_pad_nd(x, {0, 8 - (int)C});
+    auto dw32 = at::zeros({(long)Kout, (long)R * 64},
+                          dy.options().dtype(at::kFloat));
+    auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
+   hipLaunchKernelGGL(( build_pixtab_kernel), dim3(ew_grid(M)), dim3(256), 0, cur_stream(), 
+        (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
+    int nwg = (int)(((Kout + 63) / 64) * R);
+    long yb = std::max<long>(
+        1, std::min<long>((M + 63) / 64, 2048 / std::max(nwg, 1)));
+    long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
+    yb = (M + m_per_slice - 1) / m_per_slice;
+    dim3 grid((unsigned)nwg, (unsigned)yb);
+    DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w_c8", [&] {
+     hipLaunchKernelGGL(( conv_bwd_w_igemm_c8_kernel<scalar_t>), dim3(grid), dim3(256), 0, cur_stream(), 
+          (const scalar_t*)dy.data_ptr(), (const scalar_t*)x8.data_ptr(),
+          (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
+          Kout, R, m_per_slice);
+    });
+    HIP_CHECK_LAST();
+    // unpad: [K, R, 8tap, 8ch] -> [K, R, S, C]
+    auto dwp = dw32.view({(long)Kout, (long)R, 8L, 8L})
+                   .slice(2, 0, S).slice(3, 0, C);
+    return dwp.contiguous().to(dy.scalar_type());
+  }
   if (mfma_ok) {
     auto dw32 = at::zeros({(long)Kout, (long)R * S * Cg},
                           dy.options().dtype(at::kFloat));

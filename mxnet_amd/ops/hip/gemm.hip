@@ -44,11 +44,14 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
     long strideA, long strideB, long strideC, const T* __restrict__ zpage,
-    bool relu, float* __restrict__ out32, int tiles_per_slice) {
+    bool relu, float* __restrict__ out32, int tiles_per_slice, int nbuf) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BN = 128, BK = 64;
-  __shared__ T As[2][BM * BK];
-  __shared__ T Bs[2][BN * BK];
+  // dynamic LDS: single-buffered when the K loop has one tile (small-K
+  // 1x1 convs / FC) so occupancy isn't paying for an unused prefetch buf
+  extern __shared__ char smem_raw[];
+  T* As = (T*)smem_raw;               // [nbuf][BM*BK]
+  T* Bs = As + (long)nbuf * BM * BK;  // [nbuf][BN*BK]
 
   const long batch = blockIdx.y;
   A += batch * strideA;
@@ -82,10 +85,10 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
       const long kcol = k0 + s_col;
       const bool ka_ok = kcol + 8 <= K;
       const T* ga = (row_a < M && ka_ok) ? A + row_a * K + kcol : zpage;
-      gload_lds16(ga, &As[buf][(r * 256 + t) * 8]);
+      gload_lds16(ga, As + buf * (BM * BK) + (r * 256 + t) * 8);
       const long row_b = n0 + r * 32 + s_row;
       const T* gb = (row_b < N && ka_ok) ? B + row_b * K + kcol : zpage;
-      gload_lds16(gb, &Bs[buf][(r * 256 + t) * 8]);
+      gload_lds16(gb, Bs + buf * (BN * BK) + (r * 256 + t) * 8);
     }
   };
 
@@ -103,19 +106,21 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   const int k_off = (lane >> 4) * 8;
 
   for (int kt = kt0; kt < kt1; ++kt) {
-    const int buf = (kt - kt0) & 1;
+    const int buf = (kt - kt0) & (nbuf - 1);
     if (kt + 1 < kt1) stage(buf ^ 1, kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       Frag af[4], bf[4];
 #pragma unroll
       for (int m = 0; m < 4; ++m)
-        af[m] = *(const Frag*)&As[buf][(wr * 64 + m * 16 + a_row) * BK +
-                                       kk * 32 + k_off];
+        af[m] = *(const Frag*)&As[buf * (BM * BK) +
+                                  (wr * 64 + m * 16 + a_row) * BK +
+                                  kk * 32 + k_off];
 #pragma unroll
       for (int n = 0; n < 4; ++n)
-        bf[n] = *(const Frag*)&Bs[buf][(wc * 64 + n * 16 + a_row) * BK +
-                                       kk * 32 + k_off];
+        bf[n] = *(const Frag*)&Bs[buf * (BN * BK) +
+                                  (wc * 64 + n * 16 + a_row) * BK +
+                                  kk * 32 + k_off];
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -323,13 +328,16 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
     ksplit = (nk_total + tps - 1) / tps;
     out32 = at::zeros(out.sizes(), out.options().dtype(at::kFloat));
   }
+  int span = ksplit > 1 ? tps : nk_total;
+  int nbuf = span > 1 ? 2 : 1;
+  size_t lds_bytes = (size_t)nbuf * (128 + 128) * 64 * 2;
   dim3 grid((unsigned)nwg, (unsigned)nb, (unsigned)ksplit);
   DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt", [&] {
-    gemm_nt_mfma_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
+    gemm_nt_mfma_kernel<scalar_t><<<grid, 256, lds_bytes, cur_stream()>>>(
         (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
         bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
         (const scalar_t*)zero_page(A), relu,
-        ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps);
+        ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
   });
   HIP_CHECK_LAST();
   if (ksplit > 1) {

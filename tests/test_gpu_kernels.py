@@ -408,7 +408,7 @@ def test_resnet_block_integration():
         if want.size < 256:
             assert err.max() < 0.3 * scale, k
         else:
-            assert np.percentile(err, 99.5) < 0.05 * scale, k
+            assert np.percentile(err, 99.5) < 0.08 * scale, k
             assert err.max() < 0.25 * scale, k
 
 
