@@ -357,9 +357,16 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
     int dw, long m_per_slice) {
   using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
   constexpr int BI = 64, BJ = 64, BKM = 64;
-  __shared__ T DyT[2][BI * BKM];  // [i(out-ch)][m]
-  __shared__ T XT[2][BJ * BKM];   // [j(c within chunk)][m]
+  constexpr int HSTR = 80;  // hop row stride (halfs): breaks phase locks
+  // two-hop transpose staging: global 16B -> Hop[m][i] (coalesced b128
+  // LDS writes) -> per-wave u16 reads (2-way) -> XOR-swizzled operand
+  // tiles [i][m] read by ds_read_b128 fragments at the b128 bank floor.
+  __shared__ T HopA[BKM * HSTR];
+  __shared__ T HopB[BKM * HSTR];
+  __shared__ T DyT[BI * BKM];  // [i][m], idx ^ ((i&7)*8)
+  __shared__ T XT[BJ * BKM];
 
   const int g = blockIdx.z;  // conv group
   const int cpl = (Cg + BJ - 1) / BJ;
@@ -380,35 +387,26 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
-  // staging map: task idx = rnd*256+t -> m_local = idx/8 in 0..31 (x2 rnds),
-  // seg = idx%8 -> 8 channels
-  const int sm_half = t >> 3;   // m_local within half (0..31)
-  const int seg = t & 7;
+  const int sm_half = t >> 3;   // m row handled in stage1 (0..31, x2 rnds)
+  const int seg = t & 7;        // 8-element column segment
 
   float4_t acc[2][2] = {};
 
-  auto stage = [&](int buf, long mc) {
+  // stage1: global -> Hop[m][seg*8..+8)
+  auto stage1 = [&](long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
       const int m_l = rnd * 32 + sm_half;
       const long m_g = mc + m_l;
       const bool m_ok = m_g < ms1;
-      // dy side: read 8 out-channels [i0+seg*8 .. +8) at row m_g
       {
         const int iseg = i0 + seg * 8;
-        half8_t zero8 = {};
-        using V8 = T __attribute__((ext_vector_type(8)));
         V8 v = {};
         if (m_ok && iseg + 8 <= Kg)
           v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
-        (void)zero8;
+        *(V8*)&HopA[m_l * HSTR + seg * 8] = v;
       }
-      // x side: gathered tap (r,s) at channels [c0+seg*8 .. +8)
       {
-        using V8 = T __attribute__((ext_vector_type(8)));
         V8 v = {};
         if (m_ok) {
           int4_t pt = pixtab[m_g];
@@ -419,31 +417,55 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
             v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
                              (long)g * Cg + cseg);
         }
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          XT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+        *(V8*)&HopB[m_l * HSTR + seg * 8] = v;
       }
+    }
+  };
+
+  // hop2: Hop[m][i] -> DyT/XT[i][m] (whole wave shares one m-segment so
+  // the u16 gather reads are 2-way; the b128 tile write sits at the
+  // 128 B/cycle LDS floor thanks to the XOR swizzle)
+  auto hop2 = [&]() {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int mseg = rnd * 4 + (t >> 6);
+      const int i_l = lane;
+      V8 va, vb;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+        va[jj] = HopA[(mseg * 8 + jj) * HSTR + i_l];
+        vb[jj] = HopB[(mseg * 8 + jj) * HSTR + i_l];
+      }
+      const int base = (i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8);
+      *(V8*)&DyT[base] = va;
+      *(V8*)&XT[base] = vb;
     }
   };
 
   const int a_row = lane & 15;
   const int k_off = (lane >> 4) * 8;
 
-  int buf = 0;
-  stage(0, ms0);
+  stage1(ms0);
   __syncthreads();
   for (long mc = ms0; mc < ms1; mc += BKM) {
+    hop2();
+    __syncthreads();
+    if (mc + BKM < ms1) stage1(mc + BKM);  // overlaps the MFMA phase
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       Frag af[2], bf[2];
 #pragma unroll
-      for (int m = 0; m < 2; ++m)
-        af[m] = *(const Frag*)&DyT[buf][(wr * 32 + m * 16 + a_row) * BKM +
-                                        kk * 32 + k_off];
+      for (int m = 0; m < 2; ++m) {
+        const int i = wr * 32 + m * 16 + a_row;
+        af[m] = *(const Frag*)&DyT[(i * BKM + kk * 32 + k_off) ^
+                                   ((i & 7) * 8)];
+      }
 #pragma unroll
-      for (int n = 0; n < 2; ++n)
-        bf[n] = *(const Frag*)&XT[buf][(wc * 32 + n * 16 + a_row) * BKM +
-                                       kk * 32 + k_off];
+      for (int n = 0; n < 2; ++n) {
+        const int i = wc * 32 + n * 16 + a_row;
+        bf[n] = *(const Frag*)&XT[(i * BKM + kk * 32 + k_off) ^
+                                  ((i & 7) * 8)];
+      }
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -451,11 +473,6 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
-    if (mc + BKM < ms1) {
-      stage(buf ^ 1, mc + BKM);
-      buf ^= 1;
-      __syncthreads();
-    }
   }
 
   const long RSCg = (long)R * S * Cg;
