@@ -5,3 +5,5 @@ from .ndarray import (NDArray, array, zeros, ones, full, empty, arange,
 from .ops import *  # noqa: F401,F403
 from . import ops
 from .ndarray import concat, stack  # keep creation-module versions authoritative
+
+from . import contrib  # noqa: F401
