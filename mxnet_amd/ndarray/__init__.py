@@ -7,3 +7,4 @@ from . import ops
 from .ndarray import concat, stack  # keep creation-module versions authoritative
 
 from . import contrib  # noqa: F401
+from . import sparse  # noqa: F401

@@ -113,3 +113,20 @@ def test_topk_sort():
 
 def test_waitall():
     nd.waitall()  # no-op on CPU, must not raise
+
+
+def test_sparse_ndarray():
+    """row_sparse / CSR storage (reference test_sparse_ndarray.py)."""
+    import numpy as np
+    import torch
+    from mxnet_amd.ndarray import sparse as S
+    rs = S.row_sparse_array((np.ones((2, 3)), [1, 4]), shape=(6, 3))
+    d = rs.tostype('default').asnumpy()
+    assert d[1].sum() == 3 and d[0].sum() == 0
+    csr = S.csr_matrix(np.eye(4, dtype=np.float32))
+    out = S.sparse_dot(csr, torch.ones(4, 2))
+    assert out.shape == (4, 2)
+    both = S.add(rs, S.row_sparse_array((np.ones((1, 3)), [4]), shape=(6, 3)))
+    assert both.tostype('default').asnumpy()[4].sum() == 6
+    assert S.retain(rs, [4]).indices.tolist() == [4]
+    assert S.zeros('row_sparse', (5, 2)).tostype('default').asnumpy().sum() == 0
