@@ -92,7 +92,14 @@ class Trainer:
         if self._kvstore is None:
             return
         if self._distributed:
-            # late layers first (their grads were computed first)
+            n = self._kvstore.num_workers
+            if n > 1 and self._compression_params is None:
+                # bucketed flat all-reduce: xGMI rings are per-link bound,
+                # so few large collectives beat 100+ tiny ones (reference
+                # KVStoreNCCL groups ops the same way, kvstore_nccl.h:62)
+                self._bucketed_allreduce(n)
+                return
+            # per-key path (compression or single worker)
             handles = []
             for i in reversed(range(len(self._params))):
                 grads = self._params[i].list_grad()
@@ -102,8 +109,6 @@ class Trainer:
             for h in handles:
                 if h is not None:
                     h.wait()
-            # mean over workers
-            n = self._kvstore.num_workers
             if n > 1:
                 with torch.no_grad():
                     for p in self._params:
@@ -115,6 +120,37 @@ class Trainer:
             grads = p.list_grad()
             if len(grads) > 1 or isinstance(self._kvstore, kvs_mod.KVStore):
                 self._kvstore.pushpull(i, grads, out=grads, priority=-i)
+
+    def _bucketed_allreduce(self, world, bucket_bytes=1 << 27):
+        import torch.distributed as dist
+        from torch._utils import _flatten_dense_tensors, _unflatten_dense_tensors
+        # group grads by dtype, then into <=bucket_bytes buckets
+        groups = {}
+        for p in self._params:
+            for g in p.list_grad():
+                groups.setdefault(g._t.dtype, []).append(g._t)
+        with torch.no_grad():
+            for dtype, tensors in groups.items():
+                bucket, size = [], 0
+                buckets = []
+                for t in tensors:
+                    bucket.append(t)
+                    size += t.numel() * t.element_size()
+                    if size >= bucket_bytes:
+                        buckets.append(bucket)
+                        bucket, size = [], 0
+                if bucket:
+                    buckets.append(bucket)
+                handles = []
+                for b in buckets:
+                    flat = _flatten_dense_tensors(b)
+                    handles.append((dist.all_reduce(flat, async_op=True),
+                                    flat, b))
+                for h, flat, b in handles:
+                    h.wait()
+                    flat.div_(world)
+                    for t, u in zip(b, _unflatten_dense_tensors(flat, b)):
+                        t.copy_(u)
 
     def update(self, batch_size, ignore_stale_grad=False):
         self._init_kvstore()
