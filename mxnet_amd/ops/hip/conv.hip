@@ -349,7 +349,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
 // grid.y m-split into the dw workspace.
 // pixtab[m] = (n, h_base, w_base, valid) precomputed once per call.
 // ---------------------------------------------------------------------------
-template <typename T>
+template <typename T, int NJ>
 __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
@@ -357,13 +357,16 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     int dw, long m_per_slice) {
   using Frag = typename DTraits<T>::frag8;
   using V8 = T __attribute__((ext_vector_type(8)));
-  constexpr int BI = 64, BJ = 64, BKM = 64;
-  constexpr int HSTR = 80;  // hop row stride (halfs): breaks phase locks
+  constexpr int BI = 64, BJ = 64 * NJ, BKM = 64;
+  constexpr int HSTR = 80;         // dy hop row stride (halfs)
+  constexpr int HSTRB = BJ + 16;   // x hop row stride
   // two-hop transpose staging: global 16B -> Hop[m][i] (coalesced b128
   // LDS writes) -> per-wave u16 reads (2-way) -> XOR-swizzled operand
   // tiles [i][m] read by ds_read_b128 fragments at the b128 bank floor.
+  // NJ=2 widens the j (filter-input) tile to 128 halving how often the
+  // dy panel is re-staged from HBM (this kernel is staging-BW bound).
   __shared__ T HopA[BKM * HSTR];
-  __shared__ T HopB[BKM * HSTR];
+  __shared__ T HopB[BKM * HSTRB];
   __shared__ T DyT[BI * BKM];  // [i][m], idx ^ ((i&7)*8)
   __shared__ T XT[BJ * BKM];
 
@@ -389,7 +392,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int sm_half = t >> 3;   // m row handled in stage1 (0..31, x2 rnds)
   const int seg = t & 7;        // 8-element column segment
 
-  float4_t acc[2][2] = {};
+  float4_t acc[2][2 * NJ] = {};
 
   // stage1: global -> Hop[m][seg*8..+8)
   auto stage1 = [&](long mc) {
@@ -405,18 +408,20 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
           v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
         *(V8*)&HopA[m_l * HSTR + seg * 8] = v;
       }
-      {
+#pragma unroll
+      for (int part = 0; part < NJ; ++part) {
         V8 v = {};
+        const int jseg = part * 64 + seg * 8;
         if (m_ok) {
           int4_t pt = pixtab[m_g];
           const int ih = pt[1] + roff, iw = pt[2] + soff;
-          const int cseg = c0 + seg * 8;
+          const int cseg = c0 + jseg;
           if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
               cseg + 8 <= Cg)
             v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
                              (long)g * Cg + cseg);
         }
-        *(V8*)&HopB[m_l * HSTR + seg * 8] = v;
+        *(V8*)&HopB[m_l * HSTRB + jseg] = v;
       }
     }
   };
@@ -429,15 +434,20 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     for (int rnd = 0; rnd < 2; ++rnd) {
       const int mseg = rnd * 4 + (t >> 6);
       const int i_l = lane;
-      V8 va, vb;
+      V8 va;
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
+      for (int jj = 0; jj < 8; ++jj)
         va[jj] = HopA[(mseg * 8 + jj) * HSTR + i_l];
-        vb[jj] = HopB[(mseg * 8 + jj) * HSTR + i_l];
+      *(V8*)&DyT[(i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8)] = va;
+#pragma unroll
+      for (int part = 0; part < NJ; ++part) {
+        const int j_l = part * 64 + i_l;
+        V8 vb;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
+        *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
       }
-      const int base = (i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8);
-      *(V8*)&DyT[base] = va;
-      *(V8*)&XT[base] = vb;
     }
   };
 
@@ -452,7 +462,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     if (mc + BKM < ms1) stage1(mc + BKM);  // overlaps the MFMA phase
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      Frag af[2], bf[2];
+      Frag af[2], bf[2 * NJ];
 #pragma unroll
       for (int m = 0; m < 2; ++m) {
         const int i = wr * 32 + m * 16 + a_row;
@@ -460,15 +470,15 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
                                    ((i & 7) * 8)];
       }
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        const int i = wc * 32 + n * 16 + a_row;
+      for (int n = 0; n < 2 * NJ; ++n) {
+        const int i = wc * 32 * NJ + n * 16 + a_row;
         bf[n] = *(const Frag*)&XT[(i * BKM + kk * 32 + k_off) ^
                                   ((i & 7) * 8)];
       }
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int n = 0; n < 2; ++n)
+        for (int n = 0; n < 2 * NJ; ++n)
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
@@ -478,8 +488,8 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
-  for (int n = 0; n < 2; ++n) {
-    const long c = c0 + wc * 32 + n * 16 + d_col;
+  for (int n = 0; n < 2 * NJ; ++n) {
+    const long c = c0 + wc * 32 * NJ + n * 16 + d_col;
     if (c >= Cg) continue;
 #pragma unroll
     for (int m = 0; m < 2; ++m) {
@@ -489,7 +499,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
         const long i = i_base + j;
         if (i < Kg) {
           float* dst = dw32 + ((long)g * Kg + i) * RSCg + (long)rs * Cg + c;
-          if (gridDim.y == 1) *dst += acc[m][n][j];
+          if (gridDim.y == 1) *dst = acc[m][n][j];
           else atomicAdd(dst, acc[m][n][j]);
         }
       }
@@ -722,7 +732,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
         const long i = i_base + jj;
         if (i < Kout) {
           float* dst = dw32 + (i * R + r) * 64 + j;
-          if (gridDim.y == 1) *dst += acc[m][n][jj];
+          if (gridDim.y == 1) *dst = acc[m][n][jj];
           else atomicAdd(dst, acc[m][n][jj]);
         }
       }
@@ -1068,8 +1078,6 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
                S <= 8 && dh == 1 && dw == 1 && Kout % 8 == 0;
   if (c8_ok) {
     auto x8 = at::constant_pad_nd(x, {0, 8 - (int)C});
-    auto dw32 = at::zeros({(long)Kout, (long)R * 64},
-                          dy.options().dtype(at::kFloat));
     auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
     build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
@@ -1078,6 +1086,11 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
         1, std::min<long>((M + 63) / 64, 2048 / std::max(nwg, 1)));
     long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
     yb = (M + m_per_slice - 1) / m_per_slice;
+    auto dw32 = yb == 1
+        ? at::empty({(long)Kout, (long)R * 64},
+                    dy.options().dtype(at::kFloat))
+        : at::zeros({(long)Kout, (long)R * 64},
+                    dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb);
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w_c8", [&] {
       conv_bwd_w_igemm_c8_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
@@ -1092,12 +1105,11 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
     return dwp.contiguous().to(dy.scalar_type());
   }
   if (mfma_ok) {
-    auto dw32 = at::zeros({(long)Kout, (long)R * S * Cg},
-                          dy.options().dtype(at::kFloat));
     auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
     build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
-    int cpl = (Cg + 63) / 64;
+    int nj = (Cg % 128 == 0) ? 2 : 1;
+    int cpl = (Cg + 64 * nj - 1) / (64 * nj);
     int nwg = (int)(((Kg + 63) / 64) * (long)R * S * cpl);
     // split the reduction so the grid fills the chip
     long want_blocks = 2048;
@@ -1106,12 +1118,23 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
                           want_blocks / std::max<long>((long)nwg * groups, 1)));
     long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
     yb = (M + m_per_slice - 1) / m_per_slice;
+    auto dw32 = yb == 1
+        ? at::empty({(long)Kout, (long)R * S * Cg},
+                    dy.options().dtype(at::kFloat))
+        : at::zeros({(long)Kout, (long)R * S * Cg},
+                    dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb, (unsigned)groups);
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
-      conv_bwd_w_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
-          (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
-          (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W, C,
-          Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
+      if (nj == 2)
+        conv_bwd_w_igemm_kernel<scalar_t, 2><<<grid, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+            (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
+            C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
+      else
+        conv_bwd_w_igemm_kernel<scalar_t, 1><<<grid, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
+            (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
+            C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
     });
     HIP_CHECK_LAST();
     return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
