@@ -59,8 +59,9 @@ def param_size(mode, input_size, hidden_size, num_layers, bidirectional=False):
 
 
 def _lstm_cell(x_gates, h, c, wh, bh):
-    """One LSTM step given precomputed input gates. Fused HIP kernel on GPU."""
-    gates = x_gates + torch.nn.functional.linear(h, wh, bh)
+    """One LSTM step given precomputed input gates. Fused HIP kernel on GPU;
+    the recurrent gate GEMM runs the MFMA gemm_nt path."""
+    gates = x_gates + _nn.fully_connected(h.contiguous(), wh, bh)
     if use_hip(gates):
         ext = hipops()
         if ext is not None and hasattr(ext, 'lstm_cell_fwd'):
@@ -76,7 +77,7 @@ def _lstm_cell(x_gates, h, c, wh, bh):
 
 def _gru_cell(x_gates, h, wh, bh):
     H = h.shape[-1]
-    hg = torch.nn.functional.linear(h, wh, bh)
+    hg = _nn.fully_connected(h.contiguous(), wh, bh)
     xr, xz, xn = x_gates.split(H, dim=-1)
     hr, hz, hn = hg.split(H, dim=-1)
     r = torch.sigmoid(xr + hr)
@@ -86,7 +87,7 @@ def _gru_cell(x_gates, h, wh, bh):
 
 
 def _rnn_cell(x_gates, h, wh, bh, act):
-    pre = x_gates + torch.nn.functional.linear(h, wh, bh)
+    pre = x_gates + _nn.fully_connected(h.contiguous(), wh, bh)
     return torch.relu(pre) if act == 'relu' else torch.tanh(pre)
 
 
