@@ -1109,7 +1109,8 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
     auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
    hipLaunchKernelGGL(( build_pixtab_kernel), dim3(ew_grid(M)), dim3(256), 0, cur_stream(), 
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
-    int nj = (Cg % 128 == 0) ? 2 : 1;
+    int nj = 1;  // NJ=2 measured net-negative (L3 absorbs the
+                 // dy re-reads; the kernel is LDS-pipeline bound)
     int cpl = (Cg + 64 * nj - 1) / (64 * nj);
     int nwg = (int)(((Kg + 63) / 64) * (long)R * S * cpl);
     // split the reduction so the grid fills the chip

@@ -39,14 +39,15 @@ DEV_INLINE void gload_lds16(const void* g, void* lds) {
 // split-K: when out32 != nullptr, blockIdx.z owns k-tiles
 // [z*tiles_per_slice, ...) and accumulates fp32 partials with atomics
 // (small-M*N huge-K problems, e.g. the conv-stem weight gradient).
-template <typename T>
+template <typename T, int BN = 128>
 __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
     long strideA, long strideB, long strideC, const T* __restrict__ zpage,
     bool relu, float* __restrict__ out32, int tiles_per_slice, int nbuf) {
   using Frag = typename DTraits<T>::frag8;
-  constexpr int BM = 128, BN = 128, BK = 64;
+  constexpr int BM = 128, BK = 64;
+  constexpr int NW = BN / 32;  // n-fragments per wave
   // dynamic LDS: single-buffered when the K loop has one tile (small-K
   // 1x1 convs / FC) so occupancy isn't paying for an unused prefetch buf
   extern __shared__ char smem_raw[];
@@ -73,19 +74,22 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   const int s_row = t >> 3;
   const int s_col = (t & 7) * 8;
 
-  float4_t acc[4][4] = {};
+  float4_t acc[4][NW] = {};
 
   const int nk = (int)((K + BK - 1) / BK);
 
   auto stage = [&](int buf, int kt) {
     const long k0 = (long)kt * BK;
+    const long kcol = k0 + s_col;
+    const bool ka_ok = kcol + 8 <= K;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const long row_a = m0 + r * 32 + s_row;
-      const long kcol = k0 + s_col;
-      const bool ka_ok = kcol + 8 <= K;
       const T* ga = (row_a < M && ka_ok) ? A + row_a * K + kcol : zpage;
       gload_lds16(ga, As + buf * (BM * BK) + (r * 256 + t) * 8);
+    }
+#pragma unroll
+    for (int r = 0; r < BN / 32; ++r) {
       const long row_b = n0 + r * 32 + s_row;
       const T* gb = (row_b < N && ka_ok) ? B + row_b * K + kcol : zpage;
       gload_lds16(gb, Bs + buf * (BN * BK) + (r * 256 + t) * 8);
@@ -110,21 +114,21 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     if (kt + 1 < kt1) stage(buf ^ 1, kt + 1);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      Frag af[4], bf[4];
+      Frag af[4], bf[NW];
 #pragma unroll
       for (int m = 0; m < 4; ++m)
         af[m] = *(const Frag*)&As[buf * (BM * BK) +
                                   (wr * 64 + m * 16 + a_row) * BK +
                                   kk * 32 + k_off];
 #pragma unroll
-      for (int n = 0; n < 4; ++n)
+      for (int n = 0; n < NW; ++n)
         bf[n] = *(const Frag*)&Bs[buf * (BN * BK) +
-                                  (wc * 64 + n * 16 + a_row) * BK +
+                                  (wc * (BN / 2) + n * 16 + a_row) * BK +
                                   kk * 32 + k_off];
 #pragma unroll
       for (int m = 0; m < 4; ++m)
 #pragma unroll
-        for (int n = 0; n < 4; ++n)
+        for (int n = 0; n < NW; ++n)
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
@@ -134,8 +138,8 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
-  for (int n = 0; n < 4; ++n) {
-    const long col = n0 + wc * 64 + n * 16 + d_col;
+  for (int n = 0; n < NW; ++n) {
+    const long col = n0 + wc * (BN / 2) + n * 16 + d_col;
     if (col >= N) continue;
     const float b = bias ? bias[col] : 0.f;
 #pragma unroll
@@ -330,14 +334,25 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   }
   int span = ksplit > 1 ? tps : nk_total;
   int nbuf = span > 1 ? 2 : 1;
-  size_t lds_bytes = (size_t)nbuf * (128 + 128) * 64 * 2;
+  bool narrow = N <= 64;
+  if (narrow) nwg = (long)((M + 127) / 128) * ((N + 63) / 64);
+  size_t lds_bytes = (size_t)nbuf * (128 + (narrow ? 64 : 128)) * 64 * 2;
   dim3 grid((unsigned)nwg, (unsigned)nb, (unsigned)ksplit);
   DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt", [&] {
-    gemm_nt_mfma_kernel<scalar_t><<<grid, 256, lds_bytes, cur_stream()>>>(
-        (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
-        bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
-        (const scalar_t*)zero_page(A), relu,
-        ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
+    if (narrow)
+      gemm_nt_mfma_kernel<scalar_t, 64>
+          <<<grid, 256, lds_bytes, cur_stream()>>>(
+              (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+              bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
+              (const scalar_t*)zero_page(A), relu,
+              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
+    else
+      gemm_nt_mfma_kernel<scalar_t, 128>
+          <<<grid, 256, lds_bytes, cur_stream()>>>(
+              (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+              bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
+              (const scalar_t*)zero_page(A), relu,
+              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
   });
   HIP_CHECK_LAST();
   if (ksplit > 1) {
