@@ -123,3 +123,56 @@ class Counter:
 class Domain:
     def __init__(self, name='domain'):
         self.name = name
+
+
+# ---------------------------------------------------------------------------
+# GPU memory profiler (reference src/profiler/storage_profiler.{h,cc}:
+# per-scope Alloc/Free accounting dumped as a usage table)
+# ---------------------------------------------------------------------------
+_mem_scopes = {}
+
+
+class profile_scope:
+    """Tag allocations in a region (reference AssignStorageInfo)."""
+
+    def __init__(self, name):
+        self.name = name
+
+    def __enter__(self):
+        if torch.cuda.is_available():
+            self._start = torch.cuda.memory_allocated()
+        return self
+
+    def __exit__(self, *a):
+        if torch.cuda.is_available():
+            delta = torch.cuda.memory_allocated() - self._start
+            cur = _mem_scopes.get(self.name, 0)
+            _mem_scopes[self.name] = cur + delta
+
+
+def memory_stats(device=None):
+    """Allocator counters (reference storage_profiler dump)."""
+    if not torch.cuda.is_available():
+        return {}
+    s = torch.cuda.memory_stats(device)
+    out = {
+        'allocated_bytes': s.get('allocated_bytes.all.current', 0),
+        'allocated_peak': s.get('allocated_bytes.all.peak', 0),
+        'reserved_bytes': s.get('reserved_bytes.all.current', 0),
+        'reserved_peak': s.get('reserved_bytes.all.peak', 0),
+        'num_allocs': s.get('allocation.all.allocated', 0),
+    }
+    out['scopes'] = dict(_mem_scopes)
+    return out
+
+
+def dump_memory_profile(fname='gpu_memory_profile.csv', device=None):
+    stats = memory_stats(device)
+    with open(fname, 'w') as f:
+        f.write('entry,bytes\n')
+        for k, v in stats.items():
+            if k != 'scopes':
+                f.write(f'{k},{v}\n')
+        for k, v in stats.get('scopes', {}).items():
+            f.write(f'scope:{k},{v}\n')
+    return fname
