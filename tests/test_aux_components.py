@@ -1,0 +1,140 @@
+"""Coverage for auxiliary subsystems: estimator, probability, contrib
+transformer ops, gradient compression, runtime, visualization,
+profiler memory API, control flow, im2rec round-trip."""
+import json
+import math
+import os
+import subprocess
+
+import numpy as np
+import pytest
+import torch
+
+import mxnet_amd as mx
+
+
+def test_estimator_fit_and_eval(tmp_path):
+    from mxnet_amd.gluon import nn
+    from mxnet_amd.gluon.contrib.estimator import (Estimator,
+                                                   CheckpointHandler)
+    from mxnet_amd.gluon.loss import SoftmaxCrossEntropyLoss
+    from mxnet_amd.gluon.metric import Accuracy
+    net = nn.Dense(3, in_units=5)
+    net.initialize()
+    data = [(mx.nd.array(torch.randn(8, 5)),
+             mx.nd.array(torch.randint(0, 3, (8,)))) for _ in range(3)]
+    est = Estimator(net, SoftmaxCrossEntropyLoss(), train_metrics=Accuracy())
+    ckpt = CheckpointHandler(str(tmp_path), epoch_period=1)
+    est.fit(data, epochs=2, event_handlers=[ckpt])
+    assert any(f.endswith('.params') for f in os.listdir(tmp_path))
+    res = est.evaluate(data)
+    assert 'accuracy' in res
+
+
+def test_probability_distributions():
+    from mxnet_amd.gluon import probability as mgp
+    n = mgp.Normal(0.0, 1.0)
+    assert abs(float(n.log_prob(0.0).asnumpy()) + 0.5 * math.log(2 * math.pi)) < 1e-5
+    assert n.sample((5,)).shape == (5,)
+    kl = mgp.kl_divergence(n, mgp.Normal(0.0, 2.0))
+    assert float(kl.asnumpy()) > 0
+    d = mgp.Dirichlet(torch.ones(3))
+    s = d.sample()
+    assert abs(float(s.handle.sum()) - 1.0) < 1e-5
+    b = mgp.Bernoulli(probs=0.3)
+    assert abs(float(b.mean.asnumpy()) - 0.3) < 1e-6
+    td = mgp.TransformedDistribution(n, mgp.ExpTransform())
+    ln = mgp.LogNormal(0.0, 1.0)
+    assert abs(float(td.log_prob(2.0).asnumpy())
+               - float(ln.log_prob(2.0).asnumpy())) < 1e-5
+
+
+def test_contrib_transformer_matches_torch():
+    from mxnet_amd.ops import contrib_transformer as CT
+    S, B, H, D = 6, 2, 2, 8
+    qkv = torch.randn(S, B, H * 3 * D)
+    sc = CT.interleaved_matmul_selfatt_qk(qkv, H)
+    att = torch.softmax(sc, -1)
+    out = CT.interleaved_matmul_selfatt_valatt(qkv, att, H)
+    x = qkv.reshape(S, B, H, 3, D)
+    q, k, v = (x[:, :, :, i].permute(1, 2, 0, 3) for i in range(3))
+    sc_o = (q @ k.transpose(-1, -2) / math.sqrt(D)).reshape(B * H, S, S)
+    assert torch.allclose(sc, sc_o, atol=1e-5)
+    out_o = (torch.softmax(sc_o.reshape(B, H, S, S), -1) @ v) \
+        .permute(2, 0, 1, 3).reshape(S, B, H * D)
+    assert torch.allclose(out, out_o, atol=1e-5)
+    # sliding-window band mask
+    m = CT.sldwin_atten_mask_like(sc, 1, H, 1)
+    assert bool(m[0, 0, 0]) and bool(m[0, 0, 1]) and not bool(m[0, 0, 3])
+
+
+def test_gradient_compression_error_feedback():
+    from mxnet_amd.parallel.gradient_compression import GradientCompression
+    gc = GradientCompression('2bit', threshold=0.5)
+    g = torch.tensor([0.3, 0.3, -0.8, 0.0])
+    q1, s1 = gc.compress('k', g.clone())
+    assert s1 == 0.5 and q1.tolist() == [0, 0, -1, 0]
+    # residual carries the 0.3s; after the second step they cross threshold
+    q2, _ = gc.compress('k', g.clone())
+    assert q2[0] == 1 and q2[1] == 1
+
+
+def test_runtime_and_visualization():
+    feats = mx.runtime.Features()
+    assert feats.is_enabled('ROCM') and not feats.is_enabled('CUDA')
+    from mxnet_amd import symbol as S
+    y = S.Activation(S.FullyConnected(S.var('data'), num_hidden=4, name='fc'),
+                     act_type='relu', name='act')
+    dot = mx.visualization.plot_network(y)
+    assert 'digraph' in dot and 'fc' in dot
+
+
+def test_profiler_roundtrip(tmp_path):
+    from mxnet_amd import profiler
+    profiler.set_config(filename=str(tmp_path / 'trace.json'))
+    profiler.set_state('run')
+    _ = mx.nd.array(torch.randn(8, 8)) * 2
+    profiler.set_state('stop')
+    fname = profiler.dump()
+    assert os.path.exists(fname)
+    json.load(open(fname))
+    with profiler.profile_scope('region'):
+        pass
+    assert isinstance(profiler.memory_stats(), dict)
+
+
+def test_im2rec_cli(tmp_path):
+    src = tmp_path / 'imgs'
+    src.mkdir()
+    (src / 'a.raw').write_bytes(b'abcdef')
+    (src / 'b.raw').write_bytes(b'0123456789')
+    lst = tmp_path / 'list.txt'
+    lst.write_text('0\t1.0\ta.raw\n1\t2.0\tb.raw\n')
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    exe = tmp_path / 'im2rec'
+    subprocess.check_call(['g++', '-O2', '-o', str(exe),
+                           os.path.join(root, 'tools', 'im2rec.cc')])
+    subprocess.check_call([str(exe), str(lst), str(src),
+                           str(tmp_path / 'out')])
+    from mxnet_amd.io.recordio import MXRecordIO, unpack
+    r = MXRecordIO(str(tmp_path / 'out.rec'), 'r')
+    h1, p1 = unpack(r.read())
+    h2, p2 = unpack(r.read())
+    assert p1 == b'abcdef' and p2 == b'0123456789'
+    assert h1.label == 1.0 and h2.label == 2.0
+    idx = (tmp_path / 'out.idx').read_text().splitlines()
+    assert len(idx) == 2
+
+
+def test_control_flow_gradients():
+    from mxnet_amd.ndarray import contrib as C
+    from mxnet_amd import autograd
+    from mxnet_amd.ndarray.ndarray import NDArray
+    x = mx.nd.array(torch.randn(4, 3))
+    x.attach_grad()
+    with autograd.record():
+        out, _ = C.foreach(lambda xi, s: (xi * 2, s), x, [])
+        loss = NDArray(out.handle.sum())
+    loss.backward()
+    np.testing.assert_allclose(x.grad.asnumpy(), np.full((4, 3), 2.0),
+                               rtol=1e-6)
