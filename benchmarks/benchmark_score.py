@@ -28,6 +28,7 @@ def score(model_name, batch, steps, warmup, dtype, image_size=224):
     net = getattr(vision, model_name)(**kwargs)
     net.initialize(ctx=ctx)
     net.cast(dtype)
+    net.hybridize(static_alloc=True, static_shape=True)  # hipGraph capture
     dev = torch.device('cuda', 0) if on_gpu else torch.device('cpu')
     tdt = {'float16': torch.float16, 'float32': torch.float32}[dtype]
     shape = (batch, image_size, image_size, 3) if 'layout' in kwargs \

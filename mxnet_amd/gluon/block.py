@@ -231,11 +231,72 @@ class HybridBlock(Block):
         from .. import symbol as _sym_mod
         if args and isinstance(args[0], _sym_mod.Symbol):
             out = self._symbolic_call(*args, **kwargs)
+        elif self._active and not kwargs and self._graph_eligible(args):
+            out = self._graph_call(args)
         else:
             out = self.forward(*args, **kwargs)
         for hook in self._forward_hooks:
             hook(self, args, out)
         return out
+
+    # -- hipGraph-captured inference (CachedOp static execution) --------
+    # Reference parity: CachedOp with static_alloc/static_shape
+    # (src/imperative/cached_op.cc) replays a planned graph; on MI355X the
+    # natural equivalent is HIP graph capture — one graph launch replaces
+    # hundreds of kernel launches, which dominates small-batch scoring.
+    _graph_guard = False  # a parent is already capturing/replaying
+
+    def _graph_eligible(self, args):
+        import os
+        import torch
+        from .. import autograd as _ag
+        from ..ndarray.ndarray import NDArray
+        if HybridBlock._graph_guard:
+            return False
+        if os.environ.get('MXNET_ENABLE_HIPGRAPH', '1') != '1':
+            return False
+        return (torch.cuda.is_available() and not _ag.is_recording()
+                and all(isinstance(a, NDArray) and a.handle.is_cuda
+                        for a in args))
+
+    def _graph_call(self, args):
+        HybridBlock._graph_guard = True
+        try:
+            return self._graph_call_impl(args)
+        finally:
+            HybridBlock._graph_guard = False
+
+    def _graph_call_impl(self, args):
+        import torch
+        from ..ndarray.ndarray import NDArray
+        key = tuple((tuple(a.shape), a.dtype) for a in args)
+        if self._graph is not None and self._graph_key == key:
+            graph, s_in, s_out = self._graph
+            for buf, a in zip(s_in, args):
+                buf.copy_(a.handle)
+            graph.replay()
+            return s_out[0] if len(s_out) == 1 else s_out
+        # (re)capture: warm up twice on a side stream, then record
+        s_in = [a.handle.clone() for a in args]
+        nd_in = [NDArray(t) for t in s_in]
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                warm = self.forward(*nd_in)
+        torch.cuda.current_stream().wait_stream(stream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            out = self.forward(*nd_in)
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        self._graph = (graph, s_in, [o if isinstance(o, NDArray) else o
+                                     for o in outs])
+        self._graph_key = key
+        del warm
+        for buf, a in zip(s_in, args):
+            buf.copy_(a.handle)
+        graph.replay()
+        return outs[0] if len(outs) == 1 else list(outs)
 
     def _symbolic_call(self, *args, **kwargs):
         from .. import symbol as _sym_mod

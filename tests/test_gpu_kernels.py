@@ -447,3 +447,27 @@ def test_grouped_conv(case):
     check(dx, xn.grad.permute(0, 2, 3, 1), tol=4e-3)
     dw = ext.conv2d_nhwc_bwd_weight(dy, x, R, S, st, st, pd, pd, dl, dl, g)
     check(dw, wn.grad.permute(0, 2, 3, 1), tol=4e-3)
+
+
+def test_hybridize_hipgraph_inference():
+    """hybridize(static_alloc) captures the forward in a hipGraph; replay
+    must match eager output (reference CachedOp static execution)."""
+    import mxnet_amd as mx
+    from mxnet_amd.gluon.model_zoo.vision import resnet18_v1
+    torch.manual_seed(3)
+    net = resnet18_v1(classes=10, layout='NHWC')
+    net.initialize(ctx=mx.gpu(0))
+    net.cast('float16')
+    x = mx.nd.from_torch(torch.randn(2, 64, 64, 3, device=DEV).half())
+    y_eager = net(x).asnumpy()
+    net.hybridize(static_alloc=True, static_shape=True)
+    y_graph1 = net(x).asnumpy()
+    np.testing.assert_allclose(y_graph1, y_eager, rtol=2e-2, atol=2e-2)
+    # second input through the captured graph
+    x2 = mx.nd.from_torch(torch.randn(2, 64, 64, 3, device=DEV).half())
+    net.hybridize(False)
+    y2_eager = net(x2).asnumpy()
+    net.hybridize(static_alloc=True, static_shape=True)
+    net(x)  # capture
+    y2_graph = net(x2).asnumpy()
+    np.testing.assert_allclose(y2_graph, y2_eager, rtol=2e-2, atol=2e-2)
