@@ -94,10 +94,15 @@ class Trainer:
         if self._distributed:
             n = self._kvstore.num_workers
             if n > 1 and self._compression_params is None:
-                # bucketed flat all-reduce: xGMI rings are per-link bound,
-                # so few large collectives beat 100+ tiny ones (reference
-                # KVStoreNCCL groups ops the same way, kvstore_nccl.h:62)
-                self._bucketed_allreduce(n)
+                # overlapped path: grad hooks launched each bucket's
+                # all-reduce during backward; here we only drain
+                if getattr(self, '_ov_buckets', None) is None:
+                    self._init_overlap_hooks()
+                    # hooks weren't active during this first backward:
+                    # reduce everything synchronously once
+                    self._bucketed_allreduce(n)
+                    return
+                self._finish_overlap(n)
                 return
             # per-key path (compression or single worker)
             handles = []
@@ -120,6 +125,79 @@ class Trainer:
             grads = p.list_grad()
             if len(grads) > 1 or isinstance(self._kvstore, kvs_mod.KVStore):
                 self._kvstore.pushpull(i, grads, out=grads, priority=-i)
+
+    # -- overlapped all-reduce (reference: per-layer priority pushes that
+    # hide gradient sync behind the remaining backward; here: DDP-style
+    # post-accumulate-grad hooks launch each bucket's RCCL all-reduce as
+    # soon as its last gradient lands) ---------------------------------
+    def _init_overlap_hooks(self, bucket_bytes=1 << 25):
+        import torch.distributed as dist
+        params = [p for p in self._params]
+        # reverse order: late layers' grads arrive first in backward
+        order = list(reversed(params))
+        buckets, cur, size = [], [], 0
+        for p in order:
+            t = p.list_grad()[0]._t
+            cur.append(p)
+            size += t.numel() * t.element_size()
+            if size >= bucket_bytes:
+                buckets.append(cur)
+                cur, size = [], 0
+        if cur:
+            buckets.append(cur)
+        self._ov_buckets = []
+        param_slot = {}
+        for bi, plist in enumerate(buckets):
+            tensors = [p.list_grad()[0]._t for p in plist]
+            state = {'params': plist, 'tensors': tensors,
+                     'pending': len(plist), 'handle': None, 'flat': None}
+            self._ov_buckets.append(state)
+            for p in plist:
+                param_slot[id(p.list_grad()[0]._t)] = state
+
+        from torch._utils import _flatten_dense_tensors
+
+        def make_hook(state):
+            def hook(_tensor):
+                state['pending'] -= 1
+                if state['pending'] == 0:
+                    flat = _flatten_dense_tensors(state['tensors'])
+                    state['flat'] = flat
+                    state['handle'] = dist.all_reduce(flat, async_op=True)
+            return hook
+
+        self._ov_hook_handles = []
+        for state in self._ov_buckets:
+            for p in state['params']:
+                # the hook fires on the leaf (the weight) once its .grad
+                # accumulation for this backward is complete
+                w = p.list_data()[0]._t
+                h = w.register_post_accumulate_grad_hook(make_hook(state))
+                self._ov_hook_handles.append(h)
+
+    def _finish_overlap(self, world):
+        import torch.distributed as dist
+        from torch._utils import (_flatten_dense_tensors,
+                                  _unflatten_dense_tensors)
+        with torch.no_grad():
+            for state in self._ov_buckets:
+                if state['handle'] is None:
+                    # grads of this bucket never all arrived (e.g. a branch
+                    # unused this step): reduce synchronously now
+                    flat = _flatten_dense_tensors(state['tensors'])
+                    state['flat'] = flat
+                    state['handle'] = dist.all_reduce(flat, async_op=True)
+            for state in self._ov_buckets:
+                state['handle'].wait()
+                flat = state['flat']
+                flat.div_(world)
+                for t, u in zip(state['tensors'],
+                                _unflatten_dense_tensors(flat,
+                                                         state['tensors'])):
+                    t.copy_(u)
+                state['handle'] = None
+                state['flat'] = None
+                state['pending'] = len(state['params'])
 
     def _bucketed_allreduce(self, world, bucket_bytes=1 << 27):
         import torch.distributed as dist

@@ -36,11 +36,12 @@ X = torch.randn(8, 6)
 Y = torch.randint(0, 4, (8,))
 xs = mx.nd.from_torch(X[rank * 4:(rank + 1) * 4])
 ys = mx.nd.from_torch(Y[rank * 4:(rank + 1) * 4])
-with autograd.record():
-    out = net(xs)
-    L = loss_fn(out, ys)
-L.backward()
-tr.step(4)  # per-rank batch size; grads averaged over workers
+for _ in range(3):  # step 1 = sync path, steps 2-3 = overlap hooks
+    with autograd.record():
+        out = net(xs)
+        L = loss_fn(out, ys)
+    L.backward()
+    tr.step(4)  # per-rank batch size; grads averaged over workers
 w = net.weight.data().asnumpy()
 np.save(os.environ['OUT_PREFIX'] + f'_r{rank}.npy', w)
 print('STEP_OK', rank)
@@ -83,9 +84,10 @@ def test_dp_trainer_matches_single_process(tmp_path):
     loss_fn = SoftmaxCrossEntropyLoss()
     tr = Trainer(net.collect_params(), 'sgd',
                  {'learning_rate': 0.5, 'momentum': 0.9}, kvstore=None)
-    with autograd.record():
-        L = loss_fn(net(mx.nd.from_torch(X)), mx.nd.from_torch(Y))
-    L.backward()
-    tr.step(8)
+    for _ in range(3):
+        with autograd.record():
+            L = loss_fn(net(mx.nd.from_torch(X)), mx.nd.from_torch(Y))
+        L.backward()
+        tr.step(8)
     np.testing.assert_allclose(w0, net.weight.data().asnumpy(), rtol=1e-5,
                                atol=1e-6)
