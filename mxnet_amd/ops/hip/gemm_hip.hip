@@ -138,27 +138,58 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   // epilogue: D frag layout col=lane&15, row=(lane>>4)*4+j (guide §3)
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
+  // Split-K accumulates fp32 atomics directly; the normal path restages
+  // the tile through LDS so global stores are coalesced 16 B (scalar 2 B
+  // stores were the bottleneck on write-heavy small-K GEMMs).
+  if (out32) {
 #pragma unroll
-  for (int n = 0; n < NW; ++n) {
-    const long col = n0 + wc * (BN / 2) + n * 16 + d_col;
-    if (col >= N) continue;
-    const float b = bias ? bias[col] : 0.f;
+    for (int n = 0; n < NW; ++n) {
+      const long col = n0 + wc * (BN / 2) + n * 16 + d_col;
+      if (col >= N) continue;
 #pragma unroll
-    for (int m = 0; m < 4; ++m) {
-      const long row_base = m0 + wr * 64 + m * 16 + d_row;
+      for (int m = 0; m < 4; ++m) {
+        const long row_base = m0 + wr * 64 + m * 16 + d_row;
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const long row = row_base + j;
-        if (row < M) {
-          if (out32) {
-            atomicAdd(out32 + row * N + col, acc[m][n][j]);
-          } else {
-            float v = acc[m][n][j] + b;
-            if (relu) v = fmaxf(v, 0.f);
-            C[row * N + col] = (T)v;
-          }
+        for (int j = 0; j < 4; ++j) {
+          const long row = row_base + j;
+          if (row < M) atomicAdd(out32 + row * N + col, acc[m][n][j]);
         }
       }
+    }
+    return;
+  }
+  __syncthreads();                       // done with the K-loop buffers
+  T* tile = As;                          // [BM][BN] fp16 staging (fits)
+#pragma unroll
+  for (int n = 0; n < NW; ++n) {
+    const int colL = wc * (BN / 2) + n * 16 + d_col;
+    const long col = n0 + colL;
+    const float b = (bias && col < N) ? bias[col] : 0.f;
+#pragma unroll
+    for (int m = 0; m < 4; ++m) {
+      const int rowL = wr * 64 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        float v = acc[m][n][j] + b;
+        if (relu) v = fmaxf(v, 0.f);
+        tile[(rowL + j) * BN + colL] = (T)v;
+      }
+    }
+  }
+  __syncthreads();
+  using V8e = T __attribute__((ext_vector_type(8)));
+  constexpr int SEGS = BM * BN / 8;      // 8-half segments in the tile
+  for (int sidx = t; sidx < SEGS; sidx += 256) {
+    const int rowL = sidx / (BN / 8);
+    const int colL = (sidx % (BN / 8)) * 8;
+    const long row = m0 + rowL;
+    const long col = n0 + colL;
+    if (row >= M) continue;
+    if (col + 8 <= N && (N % 8) == 0) {   // 16 B-aligned fast path
+      *(V8e*)(C + row * N + col) = *(const V8e*)&tile[rowL * BN + colL];
+    } else {
+      for (int j = 0; j < 8 && col + j < N; ++j)
+        C[row * N + col + j] = tile[rowL * BN + colL + j];
     }
   }
 }

@@ -53,7 +53,12 @@ def test_dp_trainer_matches_single_process(tmp_path):
     script.write_text(_WORKER)
     repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     env = dict(os.environ)
-    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': '29517',
+    import socket
+    sock = socket.socket()
+    sock.bind(('127.0.0.1', 0))
+    port = str(sock.getsockname()[1])
+    sock.close()
+    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': port,
                 'OUT_PREFIX': str(tmp_path / 'w'),
                 'PYTHONPATH': repo_root + os.pathsep + env.get('PYTHONPATH', '')})
     procs = []
