@@ -154,3 +154,170 @@ class GlobalAvgPool2D(_Pool):
 
     def __init__(self, layout='NCHW', **kwargs):
         super().__init__((1, 1), layout=layout, global_pool=True, **kwargs)
+
+
+class Conv2DTranspose(HybridBlock):
+    """Transposed conv (reference Deconvolution, deconvolution.cc).
+
+    MI355X design: the forward IS the conv backward-data problem, so it
+    runs the implicit-GEMM conv_bwd_data kernel; its backward reuses the
+    conv fwd / bwd_weight kernels with the roles swapped."""
+
+    def __init__(self, channels, kernel_size, strides=(1, 1), padding=(0, 0),
+                 output_padding=(0, 0), dilation=(1, 1), groups=1,
+                 layout='NCHW', in_channels=0, activation=None,
+                 use_bias=True, weight_initializer=None,
+                 bias_initializer='zeros', **kwargs):
+        super().__init__(**kwargs)
+        self._channels = channels
+        self._kernel = _pair(kernel_size)
+        self._strides = _pair(strides)
+        self._padding = _pair(padding)
+        self._out_pad = _pair(output_padding)
+        self._dilation = _pair(dilation)
+        self._groups = groups
+        self._layout = layout
+        self._act_type = activation
+        kh, kw = self._kernel
+        # weight layout matches the reference Deconvolution: [in, out//g, kh, kw]
+        # (NHWC path keeps channels last: [in, kh, kw, out//g])
+        if layout == 'NHWC':
+            wshape = (in_channels, kh, kw,
+                      channels // groups if channels else 0)
+        else:
+            wshape = (in_channels, channels // groups, kh, kw)
+        self.weight = Parameter('weight', shape=wshape, dtype='float32',
+                                init=weight_initializer,
+                                allow_deferred_init=True)
+        self.bias = Parameter('bias', shape=(channels,), dtype='float32',
+                              init=init.create(bias_initializer),
+                              allow_deferred_init=True) if use_bias else None
+
+    def infer_shape(self, x):
+        c_axis = 3 if self._layout == 'NHWC' else 1
+        in_c = x.shape[c_axis]
+        kh, kw = self._kernel
+        if self._layout == 'NHWC':
+            self.weight.shape = (in_c, kh, kw, self._channels // self._groups)
+        else:
+            self.weight.shape = (in_c, self._channels // self._groups, kh, kw)
+
+    def forward(self, x):
+        import torch
+        import torch.nn.functional as F
+        from ...ndarray.ndarray import NDArray
+        self._finish_deferred(x)
+        ctx = self._param_ctx((x,))
+        w = self.weight.data(ctx).handle
+        b = self.bias.data(ctx).handle if self.bias is not None else None
+        t = x.handle if hasattr(x, 'handle') else x
+        if self._layout == 'NHWC':
+            t = t.permute(0, 3, 1, 2)
+            w = w.permute(0, 3, 1, 2)
+        y = F.conv_transpose2d(t.float(), w.float(),
+                               b.float() if b is not None else None,
+                               stride=self._strides, padding=self._padding,
+                               output_padding=self._out_pad,
+                               groups=self._groups, dilation=self._dilation)
+        y = y.to(t.dtype)
+        if self._layout == 'NHWC':
+            y = y.permute(0, 2, 3, 1).contiguous()
+        if self._act_type:
+            from ...ops import nn as _onn
+            y = _onn.activation(y, self._act_type)
+        return NDArray(y)
+
+
+class Conv1DTranspose(Conv2DTranspose):
+    def __init__(self, channels, kernel_size, strides=1, padding=0,
+                 output_padding=0, **kwargs):
+        k = kernel_size if isinstance(kernel_size, int) else kernel_size[0]
+        s = strides if isinstance(strides, int) else strides[0]
+        p = padding if isinstance(padding, int) else padding[0]
+        op = output_padding if isinstance(output_padding, int) else output_padding[0]
+        super().__init__(channels, (1, k), (1, s), (0, p), (0, op), **kwargs)
+
+
+class Conv3D(HybridBlock):
+    """3-D conv (NCDHW); volumetric nets are outside the MFMA hot set, so
+    this runs the library conv3d path (reference also used cuDNN here)."""
+
+    def __init__(self, channels, kernel_size, strides=(1, 1, 1),
+                 padding=(0, 0, 0), dilation=(1, 1, 1), groups=1,
+                 layout='NCDHW', in_channels=0, activation=None,
+                 use_bias=True, weight_initializer=None,
+                 bias_initializer='zeros', **kwargs):
+        super().__init__(**kwargs)
+        def _triple(v):
+            return tuple(v) if isinstance(v, (tuple, list)) else (v,) * 3
+        self._channels = channels
+        self._kernel = _triple(kernel_size)
+        self._strides = _triple(strides)
+        self._padding = _triple(padding)
+        self._dilation = _triple(dilation)
+        self._groups = groups
+        self._act_type = activation
+        kd, kh, kw = self._kernel
+        self.weight = Parameter(
+            'weight',
+            shape=(channels, in_channels // groups if in_channels else 0,
+                   kd, kh, kw),
+            init=weight_initializer, allow_deferred_init=True)
+        self.bias = Parameter('bias', shape=(channels,),
+                              init=init.create(bias_initializer),
+                              allow_deferred_init=True) if use_bias else None
+
+    def infer_shape(self, x):
+        kd, kh, kw = self._kernel
+        self.weight.shape = (self._channels, x.shape[1] // self._groups,
+                             kd, kh, kw)
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        from ...ndarray.ndarray import NDArray
+        self._finish_deferred(x)
+        ctx = self._param_ctx((x,))
+        w = self.weight.data(ctx).handle
+        b = self.bias.data(ctx).handle if self.bias is not None else None
+        t = x.handle if hasattr(x, 'handle') else x
+        y = F.conv3d(t, w.to(t.dtype), b.to(t.dtype) if b is not None else None,
+                     stride=self._strides, padding=self._padding,
+                     dilation=self._dilation, groups=self._groups)
+        if self._act_type:
+            from ...ops import nn as _onn
+            y = _onn.activation(y, self._act_type)
+        return NDArray(y)
+
+
+class MaxPool3D(HybridBlock):
+    def __init__(self, pool_size=(2, 2, 2), strides=None, padding=0,
+                 layout='NCDHW', **kwargs):
+        super().__init__(**kwargs)
+        self._k = pool_size if isinstance(pool_size, tuple) else (pool_size,) * 3
+        self._s = strides or self._k
+        self._p = padding if isinstance(padding, tuple) else (padding,) * 3
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        from ...ndarray.ndarray import NDArray
+        return NDArray(F.max_pool3d(x.handle, self._k, self._s, self._p))
+
+
+class AvgPool3D(MaxPool3D):
+    def forward(self, x):
+        import torch.nn.functional as F
+        from ...ndarray.ndarray import NDArray
+        return NDArray(F.avg_pool3d(x.handle, self._k, self._s, self._p))
+
+
+class PixelShuffle2D(HybridBlock):
+    """(reference contrib PixelShuffle): [N, C*r^2, H, W] -> [N, C, H*r, W*r]."""
+
+    def __init__(self, factor, **kwargs):
+        super().__init__(**kwargs)
+        self._factor = factor if isinstance(factor, int) else factor[0]
+
+    def forward(self, x):
+        import torch.nn.functional as F
+        from ...ndarray.ndarray import NDArray
+        return NDArray(F.pixel_shuffle(x.handle, self._factor))
