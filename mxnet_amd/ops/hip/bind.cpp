@@ -36,7 +36,8 @@ at::Tensor bn_nhwc_fwd_infer(const at::Tensor&, const at::Tensor&,
 std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor&, const at::Tensor&,
                                     const at::Tensor&, const at::Tensor&,
                                     const at::Tensor&, bool,
-                                    const at::Tensor&, bool);
+                                    const at::Tensor&, bool,
+                                    const at::Tensor&);
 std::vector<at::Tensor> layernorm_fwd(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, double);
 std::vector<at::Tensor> layernorm_bwd(const at::Tensor&, const at::Tensor&,

@@ -156,7 +156,8 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const T* __restrict__ res,
                                 T* __restrict__ y, long total, long C,
                                 const float* __restrict__ scale,
-                                const float* __restrict__ shift) {
+                                const float* __restrict__ shift,
+                                unsigned char* __restrict__ mask) {
   extern __shared__ float lds[];
   float* s_scale = lds;
   float* s_shift = lds + C;
@@ -178,13 +179,19 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
       long c0 = (i * 8) % C;
       VecT v = reinterpret_cast<const VecT*>(x)[i];
       VecT o;
+      unsigned char mb = 0;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float z = (float)v[j] * sc[c0 + j] + sh[c0 + j];
         if (RES) z += (float)res[i * 8 + j];
-        o[j] = (T)(RELU ? fmaxf(z, 0.f) : z);
+        if (RELU) {
+          if (z > 0.f) mb |= (1u << j);
+          z = fmaxf(z, 0.f);
+        }
+        o[j] = (T)z;
       }
       reinterpret_cast<VecT*>(y)[i] = o;
+      if (RELU && mask) mask[i] = mb;
     }
   } else {
     for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
@@ -208,7 +215,9 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ dy,
                                          const float* __restrict__ mean,
                                          const float* __restrict__ istd,
                                          float* __restrict__ s1,
-                                         float* __restrict__ s2) {
+                                         float* __restrict__ s2,
+                                         const unsigned char* __restrict__
+                                             mask) {
   using V8 = T __attribute__((ext_vector_type(8)));
   __shared__ float b1[256][8];
   __shared__ float b2[256][8];
@@ -230,11 +239,18 @@ __global__ void bn_bwd_reduce_vec_kernel(const T* __restrict__ dy,
       V8 vg = *(const V8*)(dy + r * C + c0);
       V8 vx = *(const V8*)(x + r * C + c0);
       V8 vy;
-      if (RELU) vy = *(const V8*)(y + r * C + c0);
+      unsigned char mb = 0xff;
+      if (RELU) {
+        if (mask) mb = mask[(r * C + c0) / 8];
+        else vy = *(const V8*)(y + r * C + c0);
+      }
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         float g = (float)vg[j];
-        if (RELU && (float)vy[j] <= 0.f) g = 0.f;
+        if (RELU) {
+          bool on = mask ? ((mb >> j) & 1) : ((float)vy[j] > 0.f);
+          if (!on) g = 0.f;
+        }
         a1[j] += g;
         a2[j] += g * ((float)vx[j] - mu[j]) * is[j];
       }
@@ -320,7 +336,8 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ dy,
                                     const float* __restrict__ mean,
                                     const float* __restrict__ istd,
                                     const float* __restrict__ s1,
-                                    const float* __restrict__ s2) {
+                                    const float* __restrict__ s2,
+                                    const unsigned char* __restrict__ mask) {
   extern __shared__ float lds[];  // [C] x5: ga*is, mean, istd, s1/M, s2/M
   float* c_gis = lds;
   float* c_mu = lds + C;
@@ -347,13 +364,20 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ dy,
       VecT vg = reinterpret_cast<const VecT*>(dy)[i];
       VecT vx = reinterpret_cast<const VecT*>(x)[i];
       VecT vy;
-      if (RELU) vy = reinterpret_cast<const VecT*>(y)[i];
+      unsigned char mb = 0xff;
+      if (RELU) {
+        if (mask) mb = mask[i];
+        else vy = reinterpret_cast<const VecT*>(y)[i];
+      }
       VecT odx, ores;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         long c = c0 + j;
         float g = (float)vg[j];
-        if (RELU && (float)vy[j] <= 0.f) g = 0.f;
+        if (RELU) {
+          bool on = mask ? ((mb >> j) & 1) : ((float)vy[j] > 0.f);
+          if (!on) g = 0.f;
+        }
         float xhat = ((float)vx[j] - c_mu[c]) * c_is[c];
         odx[j] = (T)(c_gis[c] * (g - c_s1[c] - xhat * c_s2[c]));
         if (RES) ores[j] = (T)g;
@@ -410,6 +434,13 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
   auto rm32 = rmean.to(at::kFloat).contiguous();
   auto rv32 = rvar.to(at::kFloat).contiguous();
   bool has_res = residual.numel() > 0;
+  // packed relu mask: backward then skips re-reading y (1 bit/elem)
+  bool want_mask = fuse_relu && C % 8 == 0 &&
+                   (x.scalar_type() == at::kHalf ||
+                    x.scalar_type() == at::kBFloat16);
+  auto mask = want_mask
+      ? at::empty({x.numel() / 8}, x.options().dtype(at::kByte))
+      : at::empty({0}, x.options().dtype(at::kByte));
   long rpb;
   dim3 grid = bn_reduce_grid(M, C, &rpb);
   auto stream = cur_stream();
@@ -439,7 +470,8 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
               (const scalar_t*)x.data_ptr(),
               has_res ? (const scalar_t*)residual.data_ptr() : nullptr,
               (scalar_t*)y.data_ptr(), total, C, scale.data_ptr<float>(),
-              shift.data_ptr<float>());
+              shift.data_ptr<float>(),
+              want_mask ? mask.data_ptr<unsigned char>() : nullptr);
     };
     if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
     else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});
@@ -453,7 +485,7 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
     rmean.copy_(rm32.to(rmean.scalar_type()));
   if (rv32.data_ptr() != rvar.data_ptr())
     rvar.copy_(rv32.to(rvar.scalar_type()));
-  return {y, save_mean, save_istd};
+  return {y, save_mean, save_istd, mask};
 }
 
 at::Tensor bn_nhwc_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
@@ -484,7 +516,7 @@ at::Tensor bn_nhwc_fwd_infer(const at::Tensor& x, const at::Tensor& gamma,
               (const scalar_t*)x.data_ptr(),
               has_res ? (const scalar_t*)residual.data_ptr() : nullptr,
               (scalar_t*)y.data_ptr(), total, C, scale.data_ptr<float>(),
-              shift.data_ptr<float>());
+              shift.data_ptr<float>(), nullptr);
     };
     if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
     else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});
@@ -500,7 +532,9 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
                                     const at::Tensor& save_mean,
                                     const at::Tensor& save_istd,
                                     bool fuse_relu, const at::Tensor& y,
-                                    bool has_res) {
+                                    bool has_res, const at::Tensor& mask) {
+  const unsigned char* mask_ptr =
+      mask.numel() > 0 ? mask.data_ptr<unsigned char>() : nullptr;
   CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
   long C = x.size(-1), M = x.numel() / C;
   auto dx = at::empty_like(x);
@@ -519,7 +553,7 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
                 (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
                 (const scalar_t*)y.data_ptr(), M, C, rpb,
                 save_mean.data_ptr<float>(), save_istd.data_ptr<float>(),
-                s1.data_ptr<float>(), s2.data_ptr<float>());
+                s1.data_ptr<float>(), s2.data_ptr<float>(), mask_ptr);
       else
        hipLaunchKernelGGL(( bn_bwd_reduce_kernel<scalar_t, decltype(relu_c)::value>)
             , dim3(grid), dim3(256), 0, stream, 
@@ -541,7 +575,7 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
               has_res ? (scalar_t*)dres.data_ptr() : nullptr, total, C,
               1.f / M, g32.data_ptr<float>(), save_mean.data_ptr<float>(),
               save_istd.data_ptr<float>(), s1.data_ptr<float>(),
-              s2.data_ptr<float>());
+              s2.data_ptr<float>(), mask_ptr);
     };
     if (fuse_relu && has_res) launch_apply(std::true_type{}, std::true_type{});
     else if (fuse_relu) launch_apply(std::true_type{}, std::false_type{});

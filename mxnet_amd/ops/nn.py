@@ -173,10 +173,10 @@ class _BatchNormNHWC(torch.autograd.Function):
                 fuse_relu, residual):
         ext = hip_required('batch_norm')
         if training:
-            y, save_mean, save_istd = ext.bn_nhwc_fwd_train(
+            y, save_mean, save_istd, mask = ext.bn_nhwc_fwd_train(
                 x, gamma, beta, rmean, rvar, momentum, eps, fuse_relu,
                 residual if residual is not None else x.new_empty(0))
-            ctx.save_for_backward(x, gamma, save_mean, save_istd, y)
+            ctx.save_for_backward(x, gamma, save_mean, save_istd, y, mask)
             ctx.fuse_relu = fuse_relu
             ctx.has_res = residual is not None
         else:
@@ -186,11 +186,12 @@ class _BatchNormNHWC(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dy):
-        x, gamma, save_mean, save_istd, y = ctx.saved_tensors
+        x, gamma, save_mean, save_istd, y, mask = ctx.saved_tensors
         ext = hip_required('batch_norm')
         dy = dy.contiguous()
         dx, dgamma, dbeta, dres = ext.bn_nhwc_bwd(
-            dy, x, gamma, save_mean, save_istd, ctx.fuse_relu, y, ctx.has_res)
+            dy, x, gamma, save_mean, save_istd, ctx.fuse_relu, y,
+            ctx.has_res, mask)
         return (dx, dgamma, dbeta, None, None, None, None, None, None,
                 dres if ctx.has_res else None)
 

@@ -142,7 +142,7 @@ def test_bn_train_fwd_bwd(fuse_relu, with_res):
     rmean = torch.zeros(C, device=DEV)
     rvar = torch.ones(C, device=DEV)
     rmean0, rvar0 = rmean.clone(), rvar.clone()
-    y, smean, sistd = ext.bn_nhwc_fwd_train(
+    y, smean, sistd, mask = ext.bn_nhwc_fwd_train(
         x, gamma, beta, rmean, rvar, 0.9, 1e-5, fuse_relu,
         res if res is not None else x.new_empty(0))
     # oracle
@@ -168,7 +168,7 @@ def test_bn_train_fwd_bwd(fuse_relu, with_res):
     yo.backward(dy_o)
     dy = dy_o.to(x.dtype)
     dx, dgamma, dbeta, dres = ext.bn_nhwc_bwd(dy, x, gamma, smean, sistd,
-                                              fuse_relu, y, with_res)
+                                              fuse_relu, y, with_res, mask)
     check(dx, x32.grad, tol=6e-3)
     check(dgamma, g32.grad, tol=6e-3)
     check(dbeta, b32.grad, tol=6e-3)
