@@ -503,3 +503,73 @@ def sample_multinomial(data, shape=1, get_prob=False, **kwargs):
     x = _t(data)
     n = shape if isinstance(shape, int) else shape[0]
     return NDArray(torch.multinomial(x, n, replacement=True).to(torch.int32))
+
+
+def LRN(data, alpha=1e-4, beta=0.75, knorm=2, nsize=5, **kwargs):
+    """Local response norm (reference lrn.cc; AlexNet-era)."""
+    import torch.nn.functional as F
+    t = _t(data)
+    y = F.local_response_norm(t.float(), size=nsize, alpha=alpha, beta=beta,
+                              k=knorm).to(t.dtype)
+    return NDArray(y)
+
+
+def UpSampling(data, scale=2, sample_type='nearest', **kwargs):
+    """(reference upsampling.cc): nearest or bilinear spatial upsampling."""
+    import torch.nn.functional as F
+    t = _t(data)
+    mode = 'nearest' if sample_type == 'nearest' else 'bilinear'
+    y = F.interpolate(t.float(), scale_factor=scale, mode=mode,
+                      align_corners=False if mode == 'bilinear' else None)
+    return NDArray(y.to(t.dtype))
+
+
+def BilinearResize2D(data, height=None, width=None, scale_height=None,
+                     scale_width=None, **kwargs):
+    """(reference contrib bilinear_resize.cc)."""
+    import torch.nn.functional as F
+    t = _t(data)
+    if height is None:
+        height = int(t.shape[2] * scale_height)
+        width = int(t.shape[3] * scale_width)
+    y = F.interpolate(t.float(), size=(height, width), mode='bilinear',
+                      align_corners=True)
+    return NDArray(y.to(t.dtype))
+
+
+def SequenceLast(data, sequence_length=None, use_sequence_length=False,
+                 axis=0):
+    """(reference sequence_last.cc): last valid step of [T, N, ...]."""
+    t = _t(data)
+    if not use_sequence_length or sequence_length is None:
+        return NDArray(t.select(axis, t.shape[axis] - 1))
+    import torch
+    sl = _t(sequence_length).long() - 1
+    tt = t.movedim(axis, 0)
+    idx = sl.view(-1, *([1] * (tt.dim() - 2))).expand(1, *tt.shape[1:])
+    return NDArray(tt.gather(0, idx).squeeze(0))
+
+
+def SequenceReverse(data, sequence_length=None, use_sequence_length=False,
+                    axis=0):
+    """(reference sequence_reverse.cc)."""
+    import torch
+    t = _t(data)
+    if not use_sequence_length or sequence_length is None:
+        return NDArray(torch.flip(t, dims=[axis]))
+    tt = t.movedim(axis, 0).clone()
+    sl = _t(sequence_length).long()
+    for n in range(tt.shape[1]):
+        L = int(sl[n])
+        tt[:L, n] = torch.flip(tt[:L, n], dims=[0])
+    return NDArray(tt.movedim(0, axis))
+
+
+def smooth_l1(data, scalar=1.0, **kwargs):
+    """(reference smooth_l1 op)."""
+    import torch
+    t = _t(data)
+    s2 = scalar * scalar
+    absd = t.abs()
+    y = torch.where(absd < 1.0 / s2, 0.5 * s2 * t * t, absd - 0.5 / s2)
+    return NDArray(y)
