@@ -56,12 +56,12 @@ def test_independent_parallelism(eng):
     v1, v2 = eng.new_variable(), eng.new_variable()
     t0 = time.time()
     for _ in range(4):
-        eng.push(lambda: time.sleep(0.05), (), (v1,))
-        eng.push(lambda: time.sleep(0.05), (), (v2,))
+        eng.push(lambda: time.sleep(0.08), (), (v1,))
+        eng.push(lambda: time.sleep(0.08), (), (v2,))
     eng.wait_for_all()
-    # serial would be 0.4s; parallel chains ~0.2s. The sleeps release the
-    # GIL so workers genuinely overlap.
-    assert time.time() - t0 < 0.35
+    # serial would be 0.64s; two parallel chains ~0.32s. The sleeps
+    # release the GIL so workers genuinely overlap.
+    assert time.time() - t0 < 0.55
 
 
 def test_exception_propagation(eng):
