@@ -45,10 +45,12 @@ class _RNNLayer(HybridBlock):
             return [shape, shape]
         return [shape]
 
-    def begin_state(self, batch_size=0, func=zeros, ctx=None, **kwargs):
+    def begin_state(self, batch_size=0, func=zeros, ctx=None, dtype=None,
+                    **kwargs):
         states = []
         for shape in self.state_info(batch_size):
-            states.append(func(shape, ctx=ctx, dtype=self._dtype, **kwargs))
+            states.append(func(shape, ctx=ctx, dtype=dtype or self._dtype,
+                               **kwargs))
         return states
 
     def forward(self, x, states=None):
@@ -61,7 +63,9 @@ class _RNNLayer(HybridBlock):
         N = t.shape[1]
         return_states = states is not None
         if states is None:
-            states = self.begin_state(N, ctx=ctx)
+            # states follow the runtime compute dtype (fp16 after cast)
+            states = self.begin_state(N, ctx=ctx, dtype=str(x.dtype)
+                                      if hasattr(x, 'dtype') else None)
         if isinstance(states, NDArray):
             states = [states]
         from ... import autograd as _ag
