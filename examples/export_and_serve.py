@@ -22,6 +22,8 @@ def main():
     ctx = mx.gpu(0) if on_gpu else mx.cpu()
     net = vision.resnet18_v1(classes=1000)
     net.initialize(ctx=ctx)
+    dev0 = torch.device('cuda', 0) if on_gpu else torch.device('cpu')
+    net(mx.nd.from_torch(torch.randn(1, 3, 64, 64, device=dev0)))  # shapes
 
     workdir = tempfile.mkdtemp()
     prefix = os.path.join(workdir, 'resnet18')

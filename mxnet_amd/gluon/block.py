@@ -331,6 +331,12 @@ class HybridBlock(Block):
         # give every parameter its flattened structural name for tracing
         for name, p in self._collect_params_with_prefix().items():
             p._structure = name
+        uninit = [n for n, p in self._collect_params_with_prefix().items()
+                  if p._data is None]
+        if uninit:
+            raise RuntimeError(
+                'export() needs fully-initialized parameters; run one '
+                f'forward pass first (deferred: {uninit[:3]}...)')
         data = _sym_mod.var('data')
         out = self(data)
         if isinstance(out, (list, tuple)):
@@ -364,15 +370,23 @@ class SymbolBlock(HybridBlock):
         self._out_sym = outputs
         self._in_syms = inputs if isinstance(inputs, (list, tuple)) else [inputs]
         in_names = {s.name for s in self._in_syms}
+        aux = set(outputs.list_auxiliary_states())
         for name in outputs.list_arguments():
             if name in in_names:
                 continue
-            p = Parameter(name=name, allow_deferred_init=True)
+            # the nnvm JSON does not persist aux-ness; recover it from the
+            # reference naming convention (running/moving statistics)
+            is_aux = name in aux or 'running_' in name or 'moving_' in name
+            p = Parameter(name=name,
+                          grad_req='null' if is_aux else 'write',
+                          allow_deferred_init=True)
             p._structure = name
             self._reg_params[name] = p
-        for name in outputs.list_auxiliary_states():
-            p = Parameter(name=name, grad_req='null', allow_deferred_init=True)
-            self._reg_params[name] = p
+        for name in aux:
+            if name not in self._reg_params:
+                p = Parameter(name=name, grad_req='null',
+                              allow_deferred_init=True)
+                self._reg_params[name] = p
 
     @staticmethod
     def imports(symbol_file, input_names, param_file=None, ctx=None):

@@ -193,7 +193,7 @@ class ResNetV1(_ResNetBase):
     def forward(self, x):
         x = self._maybe_to_layout(x)
         x = self.features(x)
-        x = x.reshape((x.shape[0], -1))
+        x = x.flatten()
         return self.output(x)
 
 
@@ -230,7 +230,7 @@ class ResNetV2(_ResNetBase):
     def forward(self, x):
         x = self._maybe_to_layout(x)
         x = self.features(x)
-        x = x.reshape((x.shape[0], -1))
+        x = x.flatten()
         return self.output(x)
 
 

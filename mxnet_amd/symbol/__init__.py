@@ -191,6 +191,12 @@ class Symbol:
             if fn is None:
                 raise NotImplementedError(f'symbol eval: op {n.op}')
             kwargs = dict(n.attrs)
+            kw_in = kwargs.pop('__kw_inputs__', None)
+            if kw_in:
+                names = [k for k in str(kw_in).split(',') if k]
+                for k, v in zip(names, args[len(args) - len(names):]):
+                    kwargs[k] = v
+                args = args[:len(args) - len(names)]
             out = fn(*args, **kwargs)
             values[id(n)] = out if isinstance(out, tuple) else \
                 (tuple(out) if isinstance(out, list) else (out,))
@@ -334,6 +340,17 @@ def _apply(op, inputs, name=None, num_outputs=1, **attrs):
 def _make_op(opname, arity='var'):
     def op(*inputs, name=None, **attrs):
         syms = [i for i in inputs if isinstance(i, Symbol)]
+        # Symbol-valued keyword args (e.g. BatchNorm residual=...) become
+        # extra graph inputs; __kw_inputs__ records their keywords so the
+        # evaluator can reconstruct the call (JSON-stringified like all
+        # attrs, reference-compatible: unknown attrs are ignored there)
+        kw_syms = [(k, v) for k, v in list(attrs.items())
+                   if isinstance(v, Symbol)]
+        for k, v in kw_syms:
+            del attrs[k]
+            syms.append(v)
+        if kw_syms:
+            attrs['__kw_inputs__'] = ','.join(k for k, _ in kw_syms)
         return _apply(opname, syms, name=name, **attrs)
     op.__name__ = opname
     return op
