@@ -358,15 +358,18 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
     int dw, long m_per_slice) {
   using Frag = typename DTraits<T>::frag8;
   using V8 = T __attribute__((ext_vector_type(8)));
-  constexpr int BI = 64, BJ = 64, BKM = 64;
-  // v3 staging: the TN transpose is free at global-load time — each lane
-  // owns ONE column (an output channel / filter input) and gathers its 8
-  // reduction-rows with strided loads that are coalesced ACROSS lanes
-  // (consecutive lanes read consecutive columns of the same dy/x row).
-  // One XOR-swizzled b128 LDS write per 8 rows; no hop, ~4x less VALU
-  // than the two-hop variant (PMC showed 13.9:1 VALU:MFMA there).
-  __shared__ T DyT[2][BI * BKM];  // [i][m], idx ^ ((i&7)*8)
-  __shared__ T XT[2][BJ * BKM];
+  constexpr int BI = 64, BJ = 64 * NJ, BKM = 64;
+  constexpr int HSTR = 80;         // dy hop row stride (halfs)
+  constexpr int HSTRB = BJ + 16;   // x hop row stride
+  // two-hop transpose staging: global 16B -> Hop[m][i] (coalesced b128
+  // LDS writes) -> per-wave u16 reads (2-way) -> XOR-swizzled operand
+  // tiles [i][m] read by ds_read_b128 fragments at the b128 bank floor.
+  // NJ=2 widens the j (filter-input) tile to 128 halving how often the
+  // dy panel is re-staged from HBM (this kernel is staging-BW bound).
+  __shared__ T HopA[BKM * HSTR];
+  __shared__ T HopB[BKM * HSTRB];
+  __shared__ T DyT[BI * BKM];  // [i][m], idx ^ ((i&7)*8)
+  __shared__ T XT[BJ * BKM];
 
   const int g = blockIdx.z;  // conv group
   const int cpl = (Cg + BJ - 1) / BJ;
@@ -376,8 +379,8 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int jt = bid % nTj;
   const int rs = jt / cpl;
   const int c0 = (jt % cpl) * BJ;
-  const int r = rs / S, sst = rs % S;
-  const int roff = r * dh, soff = sst * dw;
+  const int r = rs / S, s = rs % S;
+  const int roff = r * dh, soff = s * dw;
 
   const long ms0 = (long)blockIdx.y * m_per_slice;
   const long ms1 = min(M, ms0 + m_per_slice);
@@ -386,78 +389,108 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int lane = t & 63;
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
-  const int i_l = lane;            // column owned by this lane
-  const int mhalf = t >> 6;        // 0..3: which 8-row group (x2 rounds)
 
-  const bool dy_ok = i0 + i_l < Kg;
-  const bool x_ok = c0 + i_l < Cg;
-  const T* dy_col = dy + (long)g * Kg + i0 + i_l;
-  const long x_coff = (long)g * Cg + c0 + i_l;
+  const int sm_half = t >> 3;   // m row handled in stage1 (0..31, x2 rnds)
+  const int seg = t & 7;        // 8-element column segment
 
-  float4_t acc[2][2] = {};
+  float4_t acc[2][2 * NJ] = {};
 
-  auto stage = [&](int buf, long mc) {
+  // stage1: global -> Hop[m][seg*8..+8)
+  auto stage1 = [&](long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
-      const int mseg = rnd * 4 + mhalf;
-      const long m0 = mc + mseg * 8;
-      V8 va = {}, vb = {};
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      {
+        const int iseg = i0 + seg * 8;
+        V8 v = {};
+        if (m_ok && iseg + 8 <= Kg)
+          v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
+        *(V8*)&HopA[m_l * HSTR + seg * 8] = v;
+      }
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
-        const long m_g = m0 + jj;
-        if (m_g < ms1) {
-          if (dy_ok) va[jj] = dy_col[m_g * Kout];
+      for (int part = 0; part < NJ; ++part) {
+        V8 v = {};
+        const int jseg = part * 64 + seg * 8;
+        if (m_ok) {
           int4_t pt = pixtab[m_g];
           const int ih = pt[1] + roff, iw = pt[2] + soff;
-          if (x_ok && ih >= 0 && ih < H && iw >= 0 && iw < W)
-            vb[jj] = x[(((long)pt[0] * H + ih) * W + iw) * C + x_coff];
+          const int cseg = c0 + jseg;
+          if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+              cseg + 8 <= Cg)
+            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                             (long)g * Cg + cseg);
         }
+        *(V8*)&HopB[m_l * HSTRB + jseg] = v;
       }
-      const int base = (i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8);
-      *(V8*)&DyT[buf][base] = va;
-      *(V8*)&XT[buf][base] = vb;
+    }
+  };
+
+  // hop2: Hop[m][i] -> DyT/XT[i][m] (whole wave shares one m-segment so
+  // the u16 gather reads are 2-way; the b128 tile write sits at the
+  // 128 B/cycle LDS floor thanks to the XOR swizzle)
+  auto hop2 = [&]() {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int mseg = rnd * 4 + (t >> 6);
+      const int i_l = lane;
+      V8 va;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj)
+        va[jj] = HopA[(mseg * 8 + jj) * HSTR + i_l];
+      *(V8*)&DyT[(i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8)] = va;
+#pragma unroll
+      for (int part = 0; part < NJ; ++part) {
+        const int j_l = part * 64 + i_l;
+        V8 vb;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
+        *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
+      }
     }
   };
 
   const int a_row = lane & 15;
   const int k_off = (lane >> 4) * 8;
 
-  stage(0, ms0);
+  stage1(ms0);
   __syncthreads();
-  int buf = 0;
   for (long mc = ms0; mc < ms1; mc += BKM) {
-    if (mc + BKM < ms1) stage(buf ^ 1, mc + BKM);  // overlaps the MFMAs
+    hop2();
+    __syncthreads();
+    if (mc + BKM < ms1) stage1(mc + BKM);  // overlaps the MFMA phase
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      Frag af[2], bf[2];
+      Frag af[2], bf[2 * NJ];
 #pragma unroll
       for (int m = 0; m < 2; ++m) {
         const int i = wr * 32 + m * 16 + a_row;
-        af[m] = *(const Frag*)&DyT[buf][(i * BKM + kk * 32 + k_off) ^
-                                        ((i & 7) * 8)];
+        af[m] = *(const Frag*)&DyT[(i * BKM + kk * 32 + k_off) ^
+                                   ((i & 7) * 8)];
       }
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        const int i = wc * 32 + n * 16 + a_row;
-        bf[n] = *(const Frag*)&XT[buf][(i * BKM + kk * 32 + k_off) ^
-                                       ((i & 7) * 8)];
+      for (int n = 0; n < 2 * NJ; ++n) {
+        const int i = wc * 32 * NJ + n * 16 + a_row;
+        bf[n] = *(const Frag*)&XT[(i * BKM + kk * 32 + k_off) ^
+                                  ((i & 7) * 8)];
       }
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int n = 0; n < 2; ++n)
+        for (int n = 0; n < 2 * NJ; ++n)
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
-    buf ^= 1;
   }
 
   const long RSCg = (long)R * S * Cg;
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
-  for (int n = 0; n < 2; ++n) {
-    const long c = c0 + wc * 32 + n * 16 + d_col;
+  for (int n = 0; n < 2 * NJ; ++n) {
+    const long c = c0 + wc * 32 * NJ + n * 16 + d_col;
     if (c >= Cg) continue;
 #pragma unroll
     for (int m = 0; m < 2; ++m) {
@@ -602,10 +635,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
     int H, int W, int Kout, int R, long m_per_slice) {
   using Frag = typename DTraits<T>::frag8;
-  using V8 = T __attribute__((ext_vector_type(8)));
   constexpr int BI = 64, BKM = 64;
-  // v3 staging (see conv_bwd_w_igemm_kernel): lane owns one column,
-  // strided-coalesced gathers, XOR-swizzled b128 tile writes.
   __shared__ T DyT[2][BI * BKM];
   __shared__ T XT[2][64 * BKM];
 
@@ -621,62 +651,61 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
   const int lane = t & 63;
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
-  const int i_l = lane;
-  const int mhalf = t >> 6;
-  const int tap = i_l >> 3;       // x column: (tap s, channel)
-  const int ch = i_l & 7;
-
-  const bool dy_ok = i0 + i_l < Kout;
-  const T* dy_col = dy + i0 + i_l;
+  const int sm_half = t >> 3;
+  const int seg = t & 7;
 
   float4_t acc[2][2] = {};
 
   auto stage = [&](int buf, long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
-      const int mseg = rnd * 4 + mhalf;
-      const long m0 = mc + mseg * 8;
-      V8 va = {}, vb = {};
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      {
+        const int iseg = i0 + seg * 8;
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok && iseg + 8 <= Kout)
+          v = *(const V8*)(dy + m_g * Kout + iseg);
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
-        const long m_g = m0 + jj;
-        if (m_g < ms1) {
-          if (dy_ok) va[jj] = dy_col[m_g * Kout];
-          int4_t pt = pixtab[m_g];
-          const int ih = pt[1] + r, iw = pt[2] + tap;
-          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-            vb[jj] = x8[(((long)pt[0] * H + ih) * W + iw) * 8 + ch];
-        }
+        for (int jj = 0; jj < 8; ++jj)
+          DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
       }
-      const int base = (i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8);
-      *(V8*)&DyT[buf][base] = va;
-      *(V8*)&XT[buf][base] = vb;
+      {
+        using V8 = T __attribute__((ext_vector_type(8)));
+        V8 v = {};
+        if (m_ok) {
+          int4_t pt = pixtab[m_g];
+          const int ih = pt[1] + r, iw = pt[2] + seg;  // tap (r, seg)
+          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+            v = *(const V8*)(x8 + (((long)pt[0] * H + ih) * W + iw) * 8);
+        }
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          XT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+      }
     }
   };
 
   const int a_row = lane & 15;
   const int k_off = (lane >> 4) * 8;
 
+  int buf = 0;
   stage(0, ms0);
   __syncthreads();
-  int buf = 0;
   for (long mc = ms0; mc < ms1; mc += BKM) {
-    if (mc + BKM < ms1) stage(buf ^ 1, mc + BKM);
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       Frag af[2], bf[2];
 #pragma unroll
-      for (int m = 0; m < 2; ++m) {
-        const int i = wr * 32 + m * 16 + a_row;
-        af[m] = *(const Frag*)&DyT[buf][(i * BKM + kk * 32 + k_off) ^
-                                        ((i & 7) * 8)];
-      }
+      for (int m = 0; m < 2; ++m)
+        af[m] = *(const Frag*)&DyT[buf][(wr * 32 + m * 16 + a_row) * BKM +
+                                        kk * 32 + k_off];
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        const int i = wc * 32 + n * 16 + a_row;
-        bf[n] = *(const Frag*)&XT[buf][(i * BKM + kk * 32 + k_off) ^
-                                       ((i & 7) * 8)];
-      }
+      for (int n = 0; n < 2; ++n)
+        bf[n] = *(const Frag*)&XT[buf][(wc * 32 + n * 16 + a_row) * BKM +
+                                       kk * 32 + k_off];
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -684,7 +713,11 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
-    buf ^= 1;
+    if (mc + BKM < ms1) {
+      stage(buf ^ 1, mc + BKM);
+      buf ^= 1;
+      __syncthreads();
+    }
   }
 
   const int d_col = lane & 15;
