@@ -741,6 +741,29 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
   }
 }
 
+
+// scatter for the strided-1x1 backward-data fast path: dx is zero except
+// at the stride lattice, which receives the compact GEMM result
+template <typename T>
+__global__ void scatter_stride_rows_kernel(const T* __restrict__ compact,
+                                           T* __restrict__ dx, long totalv,
+                                           int H, int W, int C, int P,
+                                           int Q, int sh, int sw) {
+  using V8 = T __attribute__((ext_vector_type(8)));
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < totalv;
+       i += (long)gridDim.x * blockDim.x) {
+    long e = i * 8;
+    int c = e % C;
+    long t = e / C;
+    int q = t % Q;
+    long t2 = t / Q;
+    int p = t2 % P;
+    int n = t2 / P;
+    V8 v = reinterpret_cast<const V8*>(compact)[i];
+    *(V8*)(dx + ((((long)n * H + (long)p * sh) * W + (long)q * sw) * C + c)) = v;
+  }
+}
+
 // pixtab builder: m -> (n, h_base, w_base, valid)
 __global__ void build_pixtab_kernel(int4_t* __restrict__ tab, long M, int P,
                                     int Q, int sh, int sw, int ph, int pw) {
@@ -995,6 +1018,25 @@ at::Tensor conv2d_nhwc_bwd_data(const at::Tensor& dy, const at::Tensor& w,
     auto d2 = gemm_nt_core(dy.view({(long)NB * P * Q, Kout}), wt,
                            c10::nullopt, false);
     return d2.view({NB, (long)H, (long)W, (long)C});
+  }
+  if (mfma_ok && groups == 1 && R == 1 && S == 1 && (sh > 1 || sw > 1) &&
+      ph == 0 && pw == 0 && C % 8 == 0) {
+    // strided 1x1: only the stride lattice has gradient — GEMM on the
+    // compact P*Q grid, then scatter (the implicit path wastes
+    // (sh*sw-1)/(sh*sw) of its MFMA work on zero rows)
+    auto wt = transpose2d(w.view({Kout, C}));
+    auto d2 = gemm_nt_core(dy.view({(long)NB * P * Q, Kout}), wt,
+                           c10::nullopt, false);
+    dx.zero_();
+    long totalv = d2.numel() / 8;
+    DISPATCH_HALF_TYPES(dy.scalar_type(), "scatter_s11", [&] {
+     hipLaunchKernelGGL(( scatter_stride_rows_kernel<scalar_t>), dim3(ew_grid(totalv)), dim3(256), 0,
+                                             cur_stream(), 
+          (const scalar_t*)d2.data_ptr(), (scalar_t*)dx.data_ptr(), totalv,
+          H, W, C, P, Q, sh, sw);
+    });
+    HIP_CHECK_LAST();
+    return dx;
   }
   if (mfma_ok) {
     // wt[R,S,C,K]: for grouped conv, channel c within group g maps to
