@@ -1,4 +1,5 @@
 """VGG (reference gluon/model_zoo/vision/vgg.py)."""
+from .... import initializer as init
 from ...block import HybridBlock
 from ... import nn
 
@@ -21,18 +22,23 @@ class VGG(HybridBlock):
         self.features = nn.HybridSequential()
         for i, num in enumerate(layers):
             for _ in range(num):
-                self.features.add(nn.Conv2D(filters[i], kernel_size=3,
-                                            padding=1, layout=layout,
-                                            use_bias=not batch_norm))
+                self.features.add(nn.Conv2D(
+                    filters[i], kernel_size=3, padding=1, layout=layout,
+                    use_bias=not batch_norm,
+                    weight_initializer=init.Xavier(
+                        rnd_type='gaussian', factor_type='out', magnitude=2)))
                 if batch_norm:
                     self.features.add(nn.BatchNormReLU(axis=ax))
                 else:
                     self.features.add(nn.Activation('relu'))
             self.features.add(nn.MaxPool2D(strides=2, layout=layout))
-        self.features.add(nn.Flatten(),
-                          nn.Dense(4096, activation='relu'), nn.Dropout(0.5),
-                          nn.Dense(4096, activation='relu'), nn.Dropout(0.5))
-        self.output = nn.Dense(classes)
+        self.features.add(
+            nn.Flatten(),
+            nn.Dense(4096, activation='relu',
+                     weight_initializer=init.Normal(0.01)), nn.Dropout(0.5),
+            nn.Dense(4096, activation='relu',
+                     weight_initializer=init.Normal(0.01)), nn.Dropout(0.5))
+        self.output = nn.Dense(classes, weight_initializer=init.Normal(0.01))
 
     def forward(self, x):
         return self.output(self.features(x))
