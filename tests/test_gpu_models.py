@@ -111,3 +111,24 @@ def test_lstm_gpu():
     L.backward()
     tr.step(4)
     assert np.isfinite(L.asnumpy()).all()
+
+
+def test_export_symbolblock_serve_gpu(tmp_path):
+    """Reference deploy workflow on GPU: export -> SymbolBlock.imports ->
+    hipGraph-captured serving matches the exporting net."""
+    import os
+    import mxnet_amd as mx
+    from mxnet_amd.gluon import SymbolBlock
+    from mxnet_amd.gluon.model_zoo import vision
+    net = vision.resnet18_v1(classes=10)
+    net.initialize(ctx=mx.gpu(0))
+    net.cast('float16')
+    x = mx.nd.from_torch(torch.randn(2, 3, 64, 64, device=DEV).half())
+    ref = net(x).asnumpy()
+    sym_f, par_f = net.export(str(tmp_path / 'm'))
+    assert os.path.exists(sym_f) and os.path.exists(par_f)
+    served = SymbolBlock.imports(sym_f, ['data'], par_f, ctx=mx.gpu(0))
+    served.cast('float16')
+    out = served(x).asnumpy()
+    np.testing.assert_allclose(out.astype(np.float32),
+                               ref.astype(np.float32), rtol=2e-2, atol=2e-2)
