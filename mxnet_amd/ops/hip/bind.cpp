@@ -54,6 +54,12 @@ std::vector<at::Tensor> pool_nhwc_fwd(const at::Tensor&, const std::string&,
 at::Tensor pool_nhwc_bwd(const at::Tensor&, const at::Tensor&,
                          const std::string&, int64_t, int64_t, int64_t,
                          int64_t, int64_t, int64_t, int64_t, int64_t, bool);
+// quant.hip
+at::Tensor quantize_i8(const at::Tensor&, double);
+at::Tensor dequantize_i8(const at::Tensor&, double,
+                         c10::optional<at::ScalarType>);
+at::Tensor gemm_nt_i8(const at::Tensor&, const at::Tensor&, double,
+                      c10::optional<at::ScalarType>);
 // elemwise.hip
 at::Tensor act_fwd(const at::Tensor&, const std::string&);
 at::Tensor act_bwd(const at::Tensor&, const at::Tensor&, const std::string&);
@@ -105,4 +111,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dropout_bwd", &dropout_bwd);
   m.def("embedding_fwd", &embedding_fwd);
   m.def("embedding_bwd", &embedding_bwd);
+  m.def("quantize_i8", &quantize_i8);
+  m.def("dequantize_i8", &dequantize_i8, py::arg("x"), py::arg("scale"),
+        py::arg("dtype") = py::none());
+  m.def("gemm_nt_i8", &gemm_nt_i8, py::arg("a"), py::arg("b"),
+        py::arg("scale"), py::arg("out_dtype") = py::none());
 }

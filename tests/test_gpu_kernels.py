@@ -471,3 +471,18 @@ def test_hybridize_hipgraph_inference():
     net(x)  # capture
     y2_graph = net(x2).asnumpy()
     np.testing.assert_allclose(y2_graph, y2_eager, rtol=2e-2, atol=2e-2)
+
+
+def test_int8_gemm_and_quantize():
+    """int8 MFMA GEMM vs fp32 oracle (reference quantized FC path)."""
+    from mxnet_amd import _hipops as hx
+    M, N, K = 128, 96, 256
+    a = (torch.randn(M, K, device=DEV) * 20).clamp(-127, 127).round().to(torch.int8)
+    b = (torch.randn(N, K, device=DEV) * 20).clamp(-127, 127).round().to(torch.int8)
+    y = hx.gemm_nt_i8(a, b, 0.5, torch.float32)
+    want = (a.float() @ b.float().t()) * 0.5
+    check(y, want, tol=1e-5)
+    x = torch.randn(1000, device=DEV).half()
+    q = hx.quantize_i8(x, 0.01)
+    d = hx.dequantize_i8(q, 0.01, torch.float32)
+    assert (d - x.float()).abs().max().item() <= 0.0051
