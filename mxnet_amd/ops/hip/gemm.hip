@@ -193,6 +193,166 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// 256x256 8-phase NT kernel (guide "The 256^2 8-phase template", m194-m201):
+// 512 threads = 8 waves (2M x 4N), BK=64, per-wave output 128x64
+// (acc[8][4]), 128 KiB dynamic LDS (double-buffered A/B tiles),
+// st_16x32 XOR swizzle (pre-swizzled global source + swizzled ds_read),
+// one counted s_waitcnt vmcnt(8) per K-tile (next tile's 8 staging loads
+// stay in flight across the boundary), s_setprio(1) around each MFMA
+// quadrant.  Routed for large compute-bound shapes; the 128x128 2-phase
+// kernel remains the general path.
+template <typename T>
+__global__ __launch_bounds__(512, 1) void gemm_nt_8ph_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    long M, long N, long K, const T* __restrict__ zpage) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BM = 256, BN = 256, BK = 64;
+  extern __shared__ char smem8[];
+  T* As = (T*)smem8;                    // [2][256*64]
+  T* Bs = As + 2 * BM * BK;
+
+  const int nTn = (N + BN - 1) / BN;
+  const int nwg = ((M + BM - 1) / BM) * nTn;
+  const int bid = xcd_swizzle(blockIdx.x, nwg);
+  const long m0 = (long)(bid / nTn) * BM;
+  const long n0 = (long)(bid % nTn) * BN;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 2;              // 0..1: row half
+  const int wc = wid & 3;               // 0..3: 64-col panel
+
+  // staging map for ONE half-tile (128 rows x 64 halfs = 16 KiB):
+  // 2 rounds of 512 threads x 16 B; row = rnd*64 + t/8, col = (t%8)*8,
+  // with the st_16x32 source pre-swizzle col ^= ((row>>2)&1)*16
+  const int s_row_base = t >> 3;        // 0..63 (+64 second round)
+  const int s_col_base = (t & 7) * 8;
+
+  float4_t acc[8][4] = {};
+  const int nk = (int)((K + BK - 1) / BK);
+
+  // stage half-tile h (0:A-rows0,1:A-rows1,2:B-rows0,3:B-rows1) of tile kt
+  auto stage_half = [&](int buf, int kt, int h) {
+    const long k0 = (long)kt * BK;
+    const bool is_a = h < 2;
+    const int rh = (h & 1) * 128;
+    const T* src = is_a ? A : B;
+    const long lim = is_a ? M : N;
+    const long base0 = is_a ? m0 : n0;
+    T* dst = (is_a ? As : Bs) + buf * (BM * BK) + rh * BK;
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int row = rnd * 64 + s_row_base;
+      const int col = s_col_base ^ (((row >> 2) & 1) << 4);  // pre-swizzle
+      const long grow = base0 + rh + row;
+      const long kcol = k0 + col;
+      const bool ok = grow < lim && kcol + 8 <= K;
+      const T* g = ok ? src + grow * K + kcol : zpage;
+      // LDS dest is LINEAR (gload_lds constraint); the source column was
+      // pre-swizzled above so the swizzled ds_read finds the right data
+      gload_lds16(g, dst + (long)(rnd * 512 + t) * 8);
+    }
+  };
+
+  // prologue: tile 0 fully staged, then drain
+  for (int h = 0; h < 4; ++h) stage_half(0, 0, h);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __syncthreads();
+
+  const int a_row16 = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  for (int kt = 0; kt < nk; ++kt) {
+    const int buf = kt & 1;
+    const T* Asb = As + buf * (BM * BK);
+    const T* Bsb = Bs + buf * (BM * BK);
+    const bool more = kt + 1 < nk;
+    // 4 phases: quadrant q = (qm, qn); stage one next-tile half per phase
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {
+      const int qm = q >> 1, qn = q & 1;
+      if (more) stage_half(buf ^ 1, kt + 1, q);
+      Frag af[4][2], bf[2][2];
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        const int row = wr * 128 + (qm * 4 + m) * 16 + a_row16;
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int col = (kk * 32 + k_off) ^ (((row >> 2) & 1) << 4);
+          af[m][kk] = *(const Frag*)&Asb[row * BK + col];
+        }
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const int row = wc * 64 + (qn * 2 + n) * 16 + a_row16;
+#pragma unroll
+        for (int kk = 0; kk < 2; ++kk) {
+          const int col = (kk * 32 + k_off) ^ (((row >> 2) & 1) << 4);
+          bf[n][kk] = *(const Frag*)&Bsb[row * BK + col];
+        }
+      }
+      __builtin_amdgcn_s_barrier();        // align waves into the MFMA
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+#pragma unroll
+          for (int kk = 0; kk < 2; ++kk)
+            acc[qm * 4 + m][qn * 2 + n] = DTraits<T>::mfma_16x16x32(
+                af[m][kk], bf[n][kk], acc[qm * 4 + m][qn * 2 + n]);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    // tile boundary: wait for THIS tile's successor data only — the next
+    // tile's 8 in-flight staging loads stay outstanding (T4 counted
+    // vmcnt; a compiler __syncthreads would drain to 0)
+    if (more) {
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // epilogue through LDS for coalesced 16 B stores (tile = 128 KiB fits)
+  __syncthreads();
+  T* tile = As;
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 4; ++n) {
+    const int colL = wc * 64 + n * 16 + d_col;
+#pragma unroll
+    for (int m = 0; m < 8; ++m) {
+      const int rowL = wr * 128 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j)
+        tile[(rowL + j) * BN + colL] = (T)acc[m][n][j];
+    }
+  }
+  __syncthreads();
+  using V8e = T __attribute__((ext_vector_type(8)));
+  constexpr int SEGS = BM * BN / 8;
+  const bool vec_ok = (N % 8) == 0;
+  for (int sidx = t; sidx < SEGS; sidx += 512) {
+    const int rowL = sidx / (BN / 8);
+    const int colL = (sidx % (BN / 8)) * 8;
+    const long row = m0 + rowL;
+    const long col = n0 + colL;
+    if (row >= M) continue;
+    if (vec_ok && col + 8 <= N) {
+      *(V8e*)(C + row * N + col) = *(const V8e*)&tile[rowL * BN + colL];
+    } else {
+      for (int j = 0; j < 8 && col + j < N; ++j)
+        C[row * N + col + j] = tile[rowL * BN + colL + j];
+    }
+  }
+}
+
 // ---------------------------------------------------------------------------
 // fp32 NT fallback: classic 64x64 LDS tile, 4x4 per thread, VALU FMA
 // (CDNA4 has no fp32 MFMA)
@@ -365,6 +525,28 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   }
   int span = ksplit > 1 ? tps : nk_total;
   int nbuf = span > 1 ? 2 : 1;
+  // large compute-bound shapes take the 256^2 8-phase kernel
+  bool big = !batched && ksplit == 1 && M >= 512 && N >= 256 && K >= 256 &&
+             !bias_ptr && !relu;
+  if (big) {
+    long nwg8 = ((M + 255) / 256) * ((N + 255) / 256);
+    DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt8", [&] {
+      static bool attr_set = false;
+      if (!attr_set) {
+        hipFuncSetAttribute(
+            (const void*)&gemm_nt_8ph_kernel<scalar_t>,
+            hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+        attr_set = true;
+      }
+      gemm_nt_8ph_kernel<scalar_t>
+          <<<(unsigned)nwg8, 512, 131072, cur_stream()>>>(
+              (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+              (scalar_t*)out.data_ptr(), M, N, K,
+              (const scalar_t*)zero_page(A));
+    });
+    HIP_CHECK_LAST();
+    return out;
+  }
   bool narrow = N <= 64;
   if (narrow) nwg = (long)((M + 127) / 128) * ((N + 63) / 64);
   size_t lds_bytes = (size_t)nbuf * (128 + (narrow ? 64 : 128)) * 64 * 2;

@@ -486,3 +486,16 @@ def test_int8_gemm_and_quantize():
     q = hx.quantize_i8(x, 0.01)
     d = hx.dequantize_i8(q, 0.01, torch.float32)
     assert (d - x.float()).abs().max().item() <= 0.0051
+
+
+@pytest.mark.parametrize('dtype', [torch.float16, torch.bfloat16])
+@pytest.mark.parametrize('mnk', [(640, 512, 512), (2048, 1024, 1024),
+                                 (513, 300, 264), (768, 768, 768)])
+def test_gemm_nt_8phase_path(dtype, mnk):
+    """Large shapes route to the 256^2 8-phase kernel (raw barriers +
+    counted vmcnt) — refcheck against fp32 (guide two-lane discipline)."""
+    M, N, K = mnk
+    a, b = mk((M, K), dtype, seed=90), mk((N, K), dtype, seed=91)
+    y = ext.gemm_nt(a, b, None)
+    want = a.float() @ b.float().t()
+    check(y, want)
