@@ -81,3 +81,22 @@ def test_list_arguments_order():
     args = out.list_arguments()
     assert args[0] == 'data'
     assert any('weight' in a for a in args)
+
+
+def test_subgraph_partition():
+    """Subgraph framework (reference build_subgraph.cc): pointwise chains
+    collapse into a _fused_subgraph node carrying the sub-symbol."""
+    import json
+    from mxnet_amd import symbol as S
+    from mxnet_amd.symbol.subgraph import partition_graph
+    x = S.var('data')
+    y = S.Activation(S.FullyConnected(x, num_hidden=8, name='fc'),
+                     act_type='relu')
+    y = S.tanh(y + 1.0)
+    out = S.FullyConnected(y, num_hidden=4, name='fc2')
+    p = partition_graph(out)
+    conf = json.loads(p.tojson())
+    ops = [n['op'] for n in conf['nodes'] if n['op'] != 'null']
+    assert ops.count('_fused_subgraph') == 1
+    fused = [n for n in conf['nodes'] if n['op'] == '_fused_subgraph'][0]
+    assert fused['attrs']['ops'] == 'Activation,_plus_scalar,tanh'
