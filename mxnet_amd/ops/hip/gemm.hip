@@ -307,14 +307,11 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_8ph_kernel(
       __builtin_amdgcn_s_setprio(0);
       __builtin_amdgcn_s_barrier();
     }
-    // tile boundary: wait for THIS tile's successor data only — the next
-    // tile's 8 in-flight staging loads stay outstanding (T4 counted
-    // vmcnt; a compiler __syncthreads would drain to 0)
-    if (more) {
-      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
-    } else {
-      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
-    }
+    // tile boundary: drain the staging queue before the next tile's
+    // ds_reads touch that buffer (gload_lds->ds_read has no automatic
+    // waitcnt — the compiler cannot see the dependency).  The counted
+    // T4 form needs >1-tile lookahead (a 3-buffer ring); next round.
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
     __builtin_amdgcn_s_barrier();
   }
 

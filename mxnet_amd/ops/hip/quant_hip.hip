@@ -189,7 +189,7 @@ at::Tensor gemm_nt_i8(const at::Tensor& a, const at::Tensor& b, double scale,
   auto dt = out_dtype.value_or(at::kHalf);
   auto out = at::empty({M, N}, a.options().dtype(dt));
   long nwg = ((M + 127) / 128) * ((N + 127) / 128);
-  DISPATCH_HALF_TYPES(dt, "gemm_i8", [&] {
+  DISPATCH_FLOAT_TYPES(dt, "gemm_i8", [&] {
    hipLaunchKernelGGL(( gemm_nt_i8_kernel<scalar_t>), dim3((unsigned)nwg), dim3(256), 0, cur_stream(), 
         (const signed char*)a.data_ptr(), (const signed char*)b.data_ptr(),
         (scalar_t*)out.data_ptr(), M, N, K, (float)scale,
