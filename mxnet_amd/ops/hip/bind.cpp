@@ -12,6 +12,7 @@ at::Tensor gemm_nn(const at::Tensor&, const at::Tensor&);
 at::Tensor gemm_tn(const at::Tensor&, const at::Tensor&);
 at::Tensor bgemm(const at::Tensor&, const at::Tensor&);
 at::Tensor transpose2d(const at::Tensor&);
+at::Tensor gemm_nt_8ph(const at::Tensor&, const at::Tensor&);
 // conv.hip
 at::Tensor conv2d_nhwc_fwd(const at::Tensor&, const at::Tensor&,
                            c10::optional<at::Tensor>, int64_t, int64_t,
@@ -86,6 +87,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn", &gemm_tn);
   m.def("bgemm", &bgemm);
   m.def("transpose2d", &transpose2d);
+  m.def("gemm_nt_8ph", &gemm_nt_8ph);
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   m.def("conv2d_nhwc_bwd_data", &conv2d_nhwc_bwd_data);
   m.def("conv2d_nhwc_bwd_weight", &conv2d_nhwc_bwd_weight);

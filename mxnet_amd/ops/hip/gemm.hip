@@ -522,9 +522,15 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   }
   int span = ksplit > 1 ? tps : nk_total;
   int nbuf = span > 1 ? 2 : 1;
-  // large compute-bound shapes take the 256^2 8-phase kernel
-  bool big = !batched && ksplit == 1 && M >= 512 && N >= 256 && K >= 256 &&
-             !bias_ptr && !relu;
+  // 256^2 8-phase kernel: refchecked, currently at parity with the
+  // 128^2 path (drain-at-boundary; the counted-vmcnt form needs a
+  // 3-buffer ring — round-2 work), so routing is opt-in
+  static const bool use8ph = [] {
+    const char* e = getenv("MXNET_GEMM_8PH");
+    return e && e[0] == '1';
+  }();
+  bool big = use8ph && !batched && ksplit == 1 && M >= 512 && N >= 256 &&
+             K >= 256 && !bias_ptr && !relu;
   if (big) {
     long nwg8 = ((M + 255) / 256) * ((N + 255) / 256);
     DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt8", [&] {
@@ -603,4 +609,29 @@ at::Tensor bgemm(const at::Tensor& a, const at::Tensor& b) {
   TORCH_CHECK(a.dim() == 3 && b.dim() == 3, "bgemm expects 3-D");
   return gemm_nt_core(a.contiguous(), transpose2d(b.contiguous()),
                       c10::nullopt, false);
+}
+
+
+// explicit 8-phase entry (testing / iteration)
+at::Tensor gemm_nt_8ph(const at::Tensor& A_, const at::Tensor& B_) {
+  auto A = pad_k8(A_.contiguous()).contiguous();
+  auto B = pad_k8(B_.contiguous()).contiguous();
+  long M = A.size(-2), N = B.size(-2), K = A.size(-1);
+  auto out = at::empty({M, N}, A.options());
+  long nwg8 = ((M + 255) / 256) * ((N + 255) / 256);
+  DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt8x", [&] {
+    static bool attr_set = false;
+    if (!attr_set) {
+      hipFuncSetAttribute((const void*)&gemm_nt_8ph_kernel<scalar_t>,
+                          hipFuncAttributeMaxDynamicSharedMemorySize, 131072);
+      attr_set = true;
+    }
+    gemm_nt_8ph_kernel<scalar_t>
+        <<<(unsigned)nwg8, 512, 131072, cur_stream()>>>(
+            (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+            (scalar_t*)out.data_ptr(), M, N, K,
+            (const scalar_t*)zero_page(A));
+  });
+  HIP_CHECK_LAST();
+  return out;
 }

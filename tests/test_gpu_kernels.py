@@ -492,10 +492,10 @@ def test_int8_gemm_and_quantize():
 @pytest.mark.parametrize('mnk', [(640, 512, 512), (2048, 1024, 1024),
                                  (513, 300, 264), (768, 768, 768)])
 def test_gemm_nt_8phase_path(dtype, mnk):
-    """Large shapes route to the 256^2 8-phase kernel (raw barriers +
-    counted vmcnt) — refcheck against fp32 (guide two-lane discipline)."""
+    """256^2 8-phase kernel (raw barriers, st_16x32 swizzle) — refcheck
+    against fp32 (guide two-lane discipline)."""
     M, N, K = mnk
     a, b = mk((M, K), dtype, seed=90), mk((N, K), dtype, seed=91)
-    y = ext.gemm_nt(a, b, None)
+    y = ext.gemm_nt_8ph(a, b)
     want = a.float() @ b.float().t()
     check(y, want)
