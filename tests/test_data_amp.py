@@ -1,0 +1,100 @@
+"""Data pipeline + AMP coverage (reference tests/python/unittest/
+test_gluon_data.py, test_amp.py shapes)."""
+import numpy as np
+import pytest
+import torch
+
+import mxnet_amd as mx
+from mxnet_amd.gluon import data as gdata
+from mxnet_amd.gluon.data.vision import transforms
+
+
+def test_array_dataset_and_loader():
+    X = np.random.randn(37, 4).astype(np.float32)
+    Y = np.arange(37).astype(np.float32)
+    ds = gdata.ArrayDataset(X, Y)
+    assert len(ds) == 37
+    loader = gdata.DataLoader(ds, batch_size=8, shuffle=True,
+                              last_batch='keep')
+    seen = 0
+    for xb, yb in loader:
+        assert xb.shape[1] == 4
+        seen += xb.shape[0]
+    assert seen == 37
+
+
+def test_dataloader_multiworker():
+    ds = gdata.ArrayDataset(np.arange(64, dtype=np.float32).reshape(32, 2))
+    loader = gdata.DataLoader(ds, batch_size=4, num_workers=2)
+    total = sum(b.shape[0] for b in loader)
+    assert total == 32
+
+
+def test_transforms_pipeline():
+    tr = transforms.Compose([transforms.ToTensor(),
+                             transforms.Normalize(0.5, 0.5)])
+    img = mx.nd.array(np.random.rand(8, 8, 3).astype(np.float32))
+    out = tr(img)
+    assert out.shape == (3, 8, 8)
+    assert abs(float(out.handle.mean())) < 2.0
+
+
+def test_synthetic_dataset_with_transform():
+    from mxnet_amd.gluon.data.vision.datasets import SyntheticImageDataset
+    ds = SyntheticImageDataset(length=16, shape=(8, 8, 3), num_classes=4)
+    ds2 = ds.transform_first(transforms.ToTensor())
+    x, y = ds2[0]
+    assert x.shape == (3, 8, 8) and 0 <= int(y) < 4
+
+
+def test_sampler_batchify():
+    from mxnet_amd.gluon.data import sampler as smp
+    s = list(smp.SequentialSampler(5))
+    assert s == [0, 1, 2, 3, 4]
+    r = list(smp.RandomSampler(5))
+    assert sorted(r) == [0, 1, 2, 3, 4]
+    from mxnet_amd.gluon.data.batchify import Stack
+    out = Stack()([np.ones((2, 2)), np.zeros((2, 2))])
+    assert out.shape == (2, 2, 2)
+
+
+def test_amp_loss_scaler_overflow_skip():
+    from mxnet_amd import amp
+    from mxnet_amd.gluon import nn, Trainer
+    from mxnet_amd import autograd
+    from mxnet_amd.ndarray.ndarray import NDArray
+    net = nn.Dense(2, in_units=3)
+    net.initialize()
+    tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.1},
+                 kvstore=None)
+    amp.init()
+    amp.init_trainer(tr)
+    x = mx.nd.array(torch.randn(4, 3))
+    w_before = net.weight.data().asnumpy().copy()
+    with autograd.record():
+        out = net(x)
+        loss = NDArray(out.handle.mean())
+    with amp.scale_loss(loss, tr) as scaled:
+        scaled.backward()
+    # poison a gradient: all_finite must veto the step
+    net.weight.grad().handle[0, 0] = float('inf')
+    grads = [p.grad() for p in tr._params]
+    assert not amp.all_finite(grads)
+    scaler = tr._amp_loss_scaler
+    before = scaler.loss_scale
+    scaler.update(False)   # overflow -> halve scale
+    assert scaler.loss_scale < before
+    scaler.update(True)
+    assert scaler.loss_scale >= before / 2
+
+
+def test_recordfile_dataset(tmp_path):
+    from mxnet_amd.io.recordio import MXIndexedRecordIO
+    path = str(tmp_path / 'x')
+    w = MXIndexedRecordIO(path + '.idx', path + '.rec', 'w')
+    for i in range(5):
+        w.write_idx(i, bytes([i] * 4))
+    w.close()
+    ds = gdata.RecordFileDataset(path + '.rec')
+    assert len(ds) == 5
+    assert ds[3] == bytes([3] * 4)
