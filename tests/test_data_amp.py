@@ -82,10 +82,11 @@ def test_amp_loss_scaler_overflow_skip():
     assert not amp.all_finite(grads)
     scaler = tr._amp_loss_scaler
     before = scaler.loss_scale
-    scaler.update(False)   # overflow -> halve scale
+    scaler.update_scale(True)   # overflow -> halve scale
     assert scaler.loss_scale < before
-    scaler.update(True)
-    assert scaler.loss_scale >= before / 2
+    after = scaler.loss_scale
+    scaler.update_scale(False)
+    assert scaler.loss_scale >= after
 
 
 def test_recordfile_dataset(tmp_path):
