@@ -236,6 +236,8 @@ class Trainer:
         self._update(ignore_stale_grad)
 
     def _update(self, ignore_stale_grad=False):
+        if self._try_fused_update():
+            return
         for i, p in enumerate(self._params):
             self._check_states(i, p)
             if len(p.list_data()) == 1:
@@ -251,6 +253,53 @@ class Trainer:
             # restores reference behavior; 'add' keeps accumulating)
             if p.grad_req == 'write':
                 p.zero_grad()
+
+    def _try_fused_update(self):
+        """One multi-tensor kernel updates every parameter (reference
+        multi_sgd_mom_update / preloaded_multi_sgd): GPU, plain SGD."""
+        opt = self._optimizer
+        if type(opt).__name__ != 'SGD':
+            return False
+        from ..ops.dispatch import hipops, use_hip
+        groups = {}
+        for i, p in enumerate(self._params):
+            datas = p.list_data()
+            if len(datas) != 1:
+                return False
+            w = datas[0]._t
+            if not w.is_cuda:
+                return False
+            g = p.list_grad()[0]._t
+            if g.dtype != w.dtype or not g.is_contiguous():
+                return False
+            self._check_states(i, p)
+            groups.setdefault(w.dtype, []).append((i, p, w, g))
+        ext = hipops() if groups and use_hip(next(iter(groups.values()))[0][2])             else None
+        if ext is None or not hasattr(ext, 'multi_sgd_update'):
+            return False
+        empty = torch.empty(0)
+        for dtype, items in groups.items():
+            ws, masters, grads, moms, lrs, wds = [], [], [], [], [], []
+            for i, p, w, g in items:
+                opt._update_count(i)
+                st = self._states[i]
+                if opt.multi_precision and isinstance(st, tuple) and                         isinstance(st[0], torch.Tensor) and                         st[0].dtype is torch.float32 and                         w.dtype in (torch.float16, torch.bfloat16):
+                    master, mom = st
+                else:
+                    master, mom = None, st
+                ws.append(w)
+                masters.append(master if master is not None else empty)
+                grads.append(g)
+                moms.append(mom if mom is not None else empty)
+                lrs.append(opt._get_lr(i))
+                wds.append(opt._get_wd(i))
+            ext.multi_sgd_update(ws, masters, grads, moms, lrs, wds,
+                                 opt.momentum, opt.rescale_grad,
+                                 opt.clip_gradient or 0.0)
+        for p in self._params:
+            if p.grad_req == 'write':
+                p.zero_grad()
+        return True
 
     # -- AMP hook (loss scaler rescales via _scale) ----------------------
     @property

@@ -67,6 +67,10 @@ at::Tensor act_bwd(const at::Tensor&, const at::Tensor&, const std::string&);
 void sgd_update(at::Tensor, c10::optional<at::Tensor>, at::Tensor,
                 c10::optional<at::Tensor>, double, double, double, double,
                 double);
+void multi_sgd_update(std::vector<at::Tensor>, std::vector<at::Tensor>,
+                      std::vector<at::Tensor>, std::vector<at::Tensor>,
+                      std::vector<double>, std::vector<double>, double,
+                      double, double);
 void adam_update(at::Tensor, c10::optional<at::Tensor>, at::Tensor,
                  at::Tensor, at::Tensor, double, double, double, double,
                  double, double, double, bool);
@@ -107,6 +111,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("grad"), py::arg("mom"), py::arg("lr"), py::arg("mu"),
         py::arg("wd"), py::arg("rescale"), py::arg("clip"));
   m.def("adam_update", &adam_update);
+  m.def("multi_sgd_update", &multi_sgd_update);
   m.def("multi_all_finite", &multi_all_finite);
   m.def("lstm_cell_fwd", &lstm_cell_fwd);
   m.def("dropout_fwd", &dropout_fwd);

@@ -434,3 +434,86 @@ at::Tensor embedding_bwd(const at::Tensor& indices, const at::Tensor& dy,
   HIP_CHECK_LAST();
   return dw32.to(dyc.scalar_type());
 }
+
+// multi-tensor fused SGD (reference multi_sgd_mom_update /
+// preloaded_multi_sgd, optimizer_op.cc): ONE launch updates every
+// parameter; chunk table in device memory, binary search per block.
+struct MTChunk {
+  void* w;
+  float* master;
+  const void* grad;
+  float* mom;
+  long start;   // global element offset of this tensor
+  long len;
+  float lr, wd;
+};
+
+template <typename T>
+__global__ void multi_sgd_kernel(const MTChunk* __restrict__ chunks,
+                                 int nchunks, long total, float mu,
+                                 float rescale, float clip) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    // binary search the owning tensor
+    int lo = 0, hi = nchunks - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (chunks[mid].start <= i) lo = mid;
+      else hi = mid - 1;
+    }
+    const MTChunk c = chunks[lo];
+    long j = i - c.start;
+    if (j >= c.len) continue;
+    T* w = (T*)c.w;
+    const T* g = (const T*)c.grad;
+    float wm = c.master ? c.master[j] : (float)w[j];
+    float gv = (float)g[j] * rescale;
+    if (clip > 0.f) gv = fminf(fmaxf(gv, -clip), clip);
+    gv += c.wd * wm;
+    float upd;
+    if (c.mom) {
+      float m = c.mom[j] * mu + gv;
+      c.mom[j] = m;
+      upd = m;
+    } else {
+      upd = gv;
+    }
+    wm -= c.lr * upd;
+    if (c.master) c.master[j] = wm;
+    w[j] = (T)wm;
+  }
+}
+
+void multi_sgd_update(std::vector<at::Tensor> ws,
+                      std::vector<at::Tensor> masters,
+                      std::vector<at::Tensor> grads,
+                      std::vector<at::Tensor> moms,
+                      std::vector<double> lrs, std::vector<double> wds,
+                      double mu, double rescale, double clip) {
+  int n = (int)ws.size();
+  if (n == 0) return;
+  std::vector<MTChunk> host(n);
+  long total = 0;
+  for (int i = 0; i < n; ++i) {
+    host[i].w = ws[i].data_ptr();
+    host[i].master = masters[i].numel() ? masters[i].data_ptr<float>()
+                                        : nullptr;
+    host[i].grad = grads[i].data_ptr();
+    host[i].mom = moms[i].numel() ? moms[i].data_ptr<float>() : nullptr;
+    host[i].start = total;
+    host[i].len = ws[i].numel();
+    host[i].lr = (float)lrs[i];
+    host[i].wd = (float)wds[i];
+    total += host[i].len;
+  }
+  auto table = at::from_blob(host.data(), {(long)(n * sizeof(MTChunk))},
+                             at::TensorOptions().dtype(at::kByte))
+                   .to(ws[0].device(), /*non_blocking=*/false);
+  DISPATCH_FLOAT_TYPES(ws[0].scalar_type(), "multi_sgd", [&] {
+    multi_sgd_kernel<scalar_t><<<ew_grid(total), kEwBlock, 0,
+                                 cur_stream()>>>(
+        (const MTChunk*)table.data_ptr(), n, total, (float)mu,
+        (float)rescale, (float)clip);
+  });
+  HIP_CHECK_LAST();
+}

@@ -500,3 +500,28 @@ def test_gemm_nt_8phase_path(dtype, mnk):
     y = ext.gemm_nt_8ph(a, b)
     want = a.float() @ b.float().t()
     check(y, want)
+
+
+def test_multi_sgd_matches_single():
+    """multi-tensor SGD == per-tensor fused SGD (reference multi_sgd)."""
+    from mxnet_amd import _hipops as hx
+    torch.manual_seed(5)
+    shapes = [(1000,), (64, 32), (7,), (128, 3, 3, 8)]
+    ws = [torch.randn(*s, device=DEV).half() for s in shapes]
+    masters = [w.float().clone() for w in ws]
+    grads = [torch.randn_like(w) for w in ws]
+    moms = [torch.randn(w.numel(), device=DEV).reshape(w.shape) for w in ws]
+    # reference: per-tensor kernel
+    ws_r = [w.clone() for w in ws]
+    ms_r = [m.clone() for m in masters]
+    mo_r = [m.clone() for m in moms]
+    for i in range(len(ws)):
+        hx.sgd_update(ws_r[i], ms_r[i], grads[i], mo_r[i], 0.1 + 0.01 * i,
+                      0.9, 1e-4, 0.5, 0.0)
+    hx.multi_sgd_update(ws, masters, grads, moms,
+                        [0.1 + 0.01 * i for i in range(len(ws))],
+                        [1e-4] * len(ws), 0.9, 0.5, 0.0)
+    for a, b in zip(masters, ms_r):
+        assert torch.allclose(a, b, atol=1e-6), 'master mismatch'
+    for a, b in zip(moms, mo_r):
+        assert torch.allclose(a, b, atol=1e-6), 'momentum mismatch'
