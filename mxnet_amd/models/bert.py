@@ -48,14 +48,16 @@ class BERTSelfAttention(HybridBlock):
                     .reshape(B * H, S, D).contiguous()
         q, k, v = heads(q), heads(k), heads(v)
         scores = F.batch_dot(q, k, transpose_b=True)  # [B*H, S, S]
-        scores = scores / math.sqrt(D)
+        # 1/sqrt(D) folds into the softmax temperature (saves a pass)
         if mask is not None:
             m = mask.handle if hasattr(mask, 'handle') else mask
             # mask: [B, S] valid-token mask -> broadcast [B*H, S, S]
-            m2 = m[:, None, None, :].expand(B, H, S, S).reshape(B * H, S, S)
-            att = F.masked_softmax(scores, m2, axis=-1)
+            m2 = m[:, None, None, :].expand(B, H, S, S) \
+                .reshape(B * H, S, S)
+            att = F.masked_softmax(scores, m2, axis=-1,
+                                   temperature=math.sqrt(D))
         else:
-            att = F.softmax(scores, axis=-1)
+            att = F.softmax(scores, axis=-1, temperature=math.sqrt(D))
         if self.dropout is not None:
             att = self.dropout(self._wrap(att, x)).handle
         out = F.batch_dot(att, v)  # [B*H, S, D]

@@ -45,7 +45,8 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&, const at::Tensor&,
                                       const at::Tensor&);
 // softmax.hip
-at::Tensor softmax_fwd(const at::Tensor&, bool, double);
+at::Tensor softmax_fwd(const at::Tensor&, bool, double,
+                       c10::optional<at::Tensor>);
 at::Tensor softmax_bwd(const at::Tensor&, const at::Tensor&, bool, double);
 at::Tensor colsum(const at::Tensor&);
 // pool.hip
@@ -100,7 +101,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_nhwc_bwd", &bn_nhwc_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
-  m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_fwd", &softmax_fwd, py::arg("x"), py::arg("log"),
+        py::arg("temperature"), py::arg("mask") = py::none());
   m.def("softmax_bwd", &softmax_bwd);
   m.def("colsum", &colsum);
   m.def("pool_nhwc_fwd", &pool_nhwc_fwd);

@@ -12,11 +12,13 @@
 template <typename T, int VEC>
 __global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                    long rows, long C, bool log_mode,
-                                   float invT) {
+                                   float invT,
+                                   const unsigned char* __restrict__ mask) {
   __shared__ float sred[16];
   using VecT = T __attribute__((ext_vector_type(VEC)));
   for (long r = blockIdx.x; r < rows; r += gridDim.x) {
     const T* xr = x + r * C;
+    const unsigned char* mr = mask ? mask + r * C : nullptr;
     T* yr = y + r * C;
     long cv = C / VEC;
     // pass 1: max
@@ -24,37 +26,44 @@ __global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     for (long i = threadIdx.x; i < cv; i += blockDim.x) {
       VecT v = reinterpret_cast<const VecT*>(xr)[i];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) m = fmaxf(m, (float)v[j]);
+      for (int j = 0; j < VEC; ++j)
+        if (!mr || mr[i * VEC + j]) m = fmaxf(m, (float)v[j]);
     }
     for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
-      m = fmaxf(m, (float)xr[i]);
+      if (!mr || mr[i]) m = fmaxf(m, (float)xr[i]);
     m = block_reduce(m, sred, MaxOp(), -INFINITY);
     // pass 2: sum of exp
     float s = 0.f;
     for (long i = threadIdx.x; i < cv; i += blockDim.x) {
       VecT v = reinterpret_cast<const VecT*>(xr)[i];
 #pragma unroll
-      for (int j = 0; j < VEC; ++j) s += __expf(((float)v[j] - m) * invT);
+      for (int j = 0; j < VEC; ++j)
+        if (!mr || mr[i * VEC + j])
+          s += __expf(((float)v[j] - m) * invT);
     }
     for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
-      s += __expf(((float)xr[i] - m) * invT);
+      if (!mr || mr[i]) s += __expf(((float)xr[i] - m) * invT);
     __syncthreads();  // reuse of sred
     s = block_reduce(s, sred, SumOp(), 0.f);
-    float inv_s = 1.f / s, log_s = __logf(s);
-    // pass 3: write
+    float inv_s = s > 0.f ? 1.f / s : 0.f, log_s = __logf(s);
+    // pass 3: write (masked entries get 0 / -inf)
     for (long i = threadIdx.x; i < cv; i += blockDim.x) {
       VecT v = reinterpret_cast<const VecT*>(xr)[i];
       VecT o;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
+        bool on = !mr || mr[i * VEC + j];
         float z = ((float)v[j] - m) * invT;
-        o[j] = (T)(log_mode ? z - log_s : __expf(z) * inv_s);
+        o[j] = (T)(log_mode ? (on ? z - log_s : -INFINITY)
+                            : (on ? __expf(z) * inv_s : 0.f));
       }
       reinterpret_cast<VecT*>(yr)[i] = o;
     }
     for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      bool on = !mr || mr[i];
       float z = ((float)xr[i] - m) * invT;
-      yr[i] = (T)(log_mode ? z - log_s : __expf(z) * inv_s);
+      yr[i] = (T)(log_mode ? (on ? z - log_s : -INFINITY)
+                           : (on ? __expf(z) * inv_s : 0.f));
     }
     __syncthreads();
   }
@@ -107,17 +116,25 @@ __global__ void softmax_bwd_kernel(const T* __restrict__ dy,
 }
 
 at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode,
-                       double temperature) {
+                       double temperature,
+                       c10::optional<at::Tensor> mask) {
   CHECK_GPU(x); CHECK_CONTIG(x);
   auto y = at::empty_like(x);
   long C = x.size(-1), rows = x.numel() / (C > 0 ? C : 1);
   if (x.numel() == 0) return y;
+  const unsigned char* mp = nullptr;
+  at::Tensor mc;
+  if (mask && mask->defined() && mask->numel() > 0) {
+    mc = mask->to(at::kByte).contiguous();
+    TORCH_CHECK(mc.numel() == x.numel(), "softmax mask shape mismatch");
+    mp = mc.data_ptr<unsigned char>();
+  }
   int grid = (int)std::min<long>(rows, 4096);
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "softmax_fwd", [&] {
     constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
    hipLaunchKernelGGL(( softmax_fwd_kernel<scalar_t, VEC>), dim3(grid), dim3(256), 0, cur_stream(), 
         (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), rows, C,
-        log_mode, (float)(1.0 / temperature));
+        log_mode, (float)(1.0 / temperature), mp);
   });
   HIP_CHECK_LAST();
   return y;

@@ -331,13 +331,14 @@ def relu(x):
 # ---------------------------------------------------------------------------
 
 class _Softmax(torch.autograd.Function):
-    """Row softmax, stride-1 vectorized with fp32 accumulation
-    (reference softmax-inl.h:351-820)."""
+    """Row softmax, stride-1 vectorized with fp32 accumulation, optional
+    in-kernel bool mask (reference softmax-inl.h:351-820 masked
+    variants — saves the masked_fill materialization pass)."""
 
     @staticmethod
-    def forward(ctx, x, log, temperature):
+    def forward(ctx, x, log, temperature, mask=None):
         ext = hip_required('softmax')
-        y = ext.softmax_fwd(x, log, temperature)
+        y = ext.softmax_fwd(x, log, temperature, mask)
         ctx.save_for_backward(y)
         ctx.log = log
         ctx.temperature = temperature
@@ -347,7 +348,9 @@ class _Softmax(torch.autograd.Function):
     def backward(ctx, dy):
         (y,) = ctx.saved_tensors
         ext = hip_required('softmax')
-        return ext.softmax_bwd(dy.contiguous(), y, ctx.log, ctx.temperature), None, None
+        # masked entries have y==0 so the standard vjp already zeroes them
+        return (ext.softmax_bwd(dy.contiguous(), y, ctx.log,
+                                ctx.temperature), None, None, None)
 
 
 def _rows_last(x, axis):
@@ -374,7 +377,13 @@ def log_softmax(x, axis=-1, temperature=1.0):
 
 
 def masked_softmax(x, mask, axis=-1, temperature=1.0):
-    x = x.masked_fill(~mask.bool(), float('-inf')) if mask is not None else x
+    if mask is None:
+        return softmax(x, axis, temperature)
+    if use_hip(x) and axis in (-1, x.dim() - 1):
+        m = mask.expand_as(x).contiguous() if mask.shape != x.shape \
+            else mask.contiguous()
+        return _Softmax.apply(x.contiguous(), False, float(temperature), m)
+    x = x.masked_fill(~mask.bool(), float('-inf'))
     return softmax(x, axis, temperature)
 
 
