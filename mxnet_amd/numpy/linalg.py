@@ -78,3 +78,47 @@ def tensorsolve(a, b, axes=None):
 
 def tensorinv(a, ind=2):
     return NDArray(torch.linalg.tensorinv(_t(a), ind=ind))
+
+
+def eig(a):
+    """(reference np_eig op)."""
+    import torch
+    from ..ndarray.ndarray import NDArray
+    w, v = torch.linalg.eig(_t(a))
+    return NDArray(w), NDArray(v)
+
+
+def eigh(a, UPLO='L'):
+    import torch
+    from ..ndarray.ndarray import NDArray
+    w, v = torch.linalg.eigh(_t(a), UPLO=UPLO)
+    return NDArray(w), NDArray(v)
+
+
+def eigvals(a):
+    import torch
+    from ..ndarray.ndarray import NDArray
+    return NDArray(torch.linalg.eigvals(_t(a)))
+
+
+def lstsq(a, b, rcond='warn'):
+    """(reference np_lstsq)."""
+    import torch
+    from ..ndarray.ndarray import NDArray
+    rc = None if rcond in ('warn', None) else rcond
+    res = torch.linalg.lstsq(_t(a), _t(b), rcond=rc)
+    return (NDArray(res.solution), NDArray(res.residuals),
+            int(res.rank) if res.rank.numel() == 1 else NDArray(res.rank),
+            NDArray(res.singular_values))
+
+
+def matrix_rank(a, tol=None):
+    import torch
+    from ..ndarray.ndarray import NDArray
+    return NDArray(torch.linalg.matrix_rank(_t(a), tol=tol))
+
+
+def tensorinv(a, ind=2):
+    import torch
+    from ..ndarray.ndarray import NDArray
+    return NDArray(torch.linalg.tensorinv(_t(a), ind=ind))
