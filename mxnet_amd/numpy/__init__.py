@@ -181,8 +181,8 @@ def _red(fn):
             out = fn(t)
         else:
             out = fn(t, dim=axis, keepdim=keepdims)
-            if isinstance(out, tuple) or hasattr(out, 'values'):
-                out = out.values if hasattr(out, 'values') else out[0]
+            if not isinstance(out, torch.Tensor):   # (values, indices)
+                out = out[0]
         if dtype is not None:
             out = out.to(torch_dtype(dtype))
         return NDArray(out)
