@@ -352,24 +352,21 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_data_igemm_kernel(
 // ---------------------------------------------------------------------------
 template <typename T, int NJ>
 __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
-    const T* __restrict__ dy, const T* __restrict__ x,
+    const T* __restrict__ dyT, const T* __restrict__ x,
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
     int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
-    int dw, long m_per_slice) {
+    int dw, long m_per_slice, const T* __restrict__ zpage) {
   using Frag = typename DTraits<T>::frag8;
   using V8 = T __attribute__((ext_vector_type(8)));
-  constexpr int BI = 64, BJ = 64 * NJ, BKM = 64;
-  constexpr int HSTR = 80;         // dy hop row stride (halfs)
-  constexpr int HSTRB = BJ + 16;   // x hop row stride
-  // two-hop transpose staging: global 16B -> Hop[m][i] (coalesced b128
-  // LDS writes) -> per-wave u16 reads (2-way) -> XOR-swizzled operand
-  // tiles [i][m] read by ds_read_b128 fragments at the b128 bank floor.
-  // NJ=2 widens the j (filter-input) tile to 128 halving how often the
-  // dy panel is re-staged from HBM (this kernel is staging-BW bound).
-  __shared__ T HopA[BKM * HSTR];
-  __shared__ T HopB[BKM * HSTRB];
-  __shared__ T DyT[BI * BKM];  // [i][m], idx ^ ((i&7)*8)
-  __shared__ T XT[BJ * BKM];
+  constexpr int BI = 64, BJ = 64, BKM = 64;
+  constexpr int HSTRB = BJ + 16;   // x hop row stride (halfs)
+  // v2.5 staging: dy arrives PRE-TRANSPOSED ([Kout, M], one cheap global
+  // transpose per call) so its tile stages like a plain NT operand via
+  // global_load_lds — zero transpose work on chip.  Only the gathered x
+  // side still needs the two-hop LDS transpose (its rows are virtual).
+  __shared__ T DyT[2][BI * BKM];   // [i][m], linear (gload_lds dest)
+  __shared__ T HopB[BKM * HSTRB];  // x hop [m][j]
+  __shared__ T XT[BJ * BKM];       // [j][m], idx ^ ((j&7)*8)
 
   const int g = blockIdx.z;  // conv group
   const int cpl = (Cg + BJ - 1) / BJ;
@@ -379,8 +376,8 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int jt = bid % nTj;
   const int rs = jt / cpl;
   const int c0 = (jt % cpl) * BJ;
-  const int r = rs / S, s = rs % S;
-  const int roff = r * dh, soff = s * dw;
+  const int r = rs / S, sst = rs % S;
+  const int roff = r * dh, soff = sst * dw;
 
   const long ms0 = (long)blockIdx.y * m_per_slice;
   const long ms1 = min(M, ms0 + m_per_slice);
@@ -390,107 +387,104 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
-  const int sm_half = t >> 3;   // m row handled in stage1 (0..31, x2 rnds)
-  const int seg = t & 7;        // 8-element column segment
+  const int sm_half = t >> 3;   // staging row (x2 rounds)
+  const int seg = t & 7;        // 8-half column segment
 
-  float4_t acc[2][2 * NJ] = {};
+  float4_t acc[2][2] = {};
 
-  // stage1: global -> Hop[m][seg*8..+8)
-  auto stage1 = [&](long mc) {
+  // dy side: linear NT staging of dyT rows [i0..i0+64) x m-chunk
+  auto stage_dy = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int i_l = rnd * 32 + sm_half;
+      const long mcol = mc + seg * 8;
+      const bool ok = i0 + i_l < Kg && mcol + 8 <= M;
+      const T* ga = ok
+          ? dyT + ((long)g * Kg + i0 + i_l) * M + mcol
+          : zpage;
+      gload_lds16c(ga, &DyT[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  // x side stage1: gathered global 16B -> HopB[m][j]
+  auto stage1_x = [&](long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
       const int m_l = rnd * 32 + sm_half;
       const long m_g = mc + m_l;
       const bool m_ok = m_g < ms1;
-      {
-        const int iseg = i0 + seg * 8;
-        V8 v = {};
-        if (m_ok && iseg + 8 <= Kg)
-          v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
-        *(V8*)&HopA[m_l * HSTR + seg * 8] = v;
+      V8 v = {};
+      if (m_ok) {
+        int4_t pt = pixtab[m_g];
+        const int ih = pt[1] + roff, iw = pt[2] + soff;
+        const int cseg = c0 + seg * 8;
+        if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+            cseg + 8 <= Cg)
+          v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                           (long)g * Cg + cseg);
       }
-#pragma unroll
-      for (int part = 0; part < NJ; ++part) {
-        V8 v = {};
-        const int jseg = part * 64 + seg * 8;
-        if (m_ok) {
-          int4_t pt = pixtab[m_g];
-          const int ih = pt[1] + roff, iw = pt[2] + soff;
-          const int cseg = c0 + jseg;
-          if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
-              cseg + 8 <= Cg)
-            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
-                             (long)g * Cg + cseg);
-        }
-        *(V8*)&HopB[m_l * HSTRB + jseg] = v;
-      }
+      *(V8*)&HopB[m_l * HSTRB + seg * 8] = v;
     }
   };
 
-  // hop2: Hop[m][i] -> DyT/XT[i][m] (whole wave shares one m-segment so
-  // the u16 gather reads are 2-way; the b128 tile write sits at the
-  // 128 B/cycle LDS floor thanks to the XOR swizzle)
-  auto hop2 = [&]() {
+  // x side hop2: HopB[m][j] -> XT[j][m] (swizzled b128 writes)
+  auto hop2_x = [&]() {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
       const int mseg = rnd * 4 + (t >> 6);
-      const int i_l = lane;
-      V8 va;
+      const int j_l = lane;
+      V8 vb;
 #pragma unroll
       for (int jj = 0; jj < 8; ++jj)
-        va[jj] = HopA[(mseg * 8 + jj) * HSTR + i_l];
-      *(V8*)&DyT[(i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8)] = va;
-#pragma unroll
-      for (int part = 0; part < NJ; ++part) {
-        const int j_l = part * 64 + i_l;
-        V8 vb;
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
-        *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
-      }
+        vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
+      *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
     }
   };
 
   const int a_row = lane & 15;
   const int k_off = (lane >> 4) * 8;
 
-  stage1(ms0);
+  stage_dy(0, ms0);
+  stage1_x(ms0);
   __syncthreads();
+  int buf = 0;
   for (long mc = ms0; mc < ms1; mc += BKM) {
-    hop2();
-    __syncthreads();
-    if (mc + BKM < ms1) stage1(mc + BKM);  // overlaps the MFMA phase
+    hop2_x();                              // HopB -> XT for this chunk
+    __syncthreads();                       // XT ready; HopB reusable
+    if (mc + BKM < ms1) {
+      stage1_x(mc + BKM);                  // overlap next x gather
+      stage_dy(buf ^ 1, mc + BKM);         // async next dy tile
+    }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      Frag af[2], bf[2 * NJ];
+      Frag af[2], bf[2];
 #pragma unroll
       for (int m = 0; m < 2; ++m) {
         const int i = wr * 32 + m * 16 + a_row;
-        af[m] = *(const Frag*)&DyT[(i * BKM + kk * 32 + k_off) ^
-                                   ((i & 7) * 8)];
+        af[m] = *(const Frag*)&DyT[buf][i * BKM + kk * 32 + k_off];
       }
 #pragma unroll
-      for (int n = 0; n < 2 * NJ; ++n) {
-        const int i = wc * 32 * NJ + n * 16 + a_row;
-        bf[n] = *(const Frag*)&XT[(i * BKM + kk * 32 + k_off) ^
-                                  ((i & 7) * 8)];
+      for (int n = 0; n < 2; ++n) {
+        const int j = wc * 32 + n * 16 + a_row;
+        bf[n] = *(const Frag*)&XT[(j * BKM + kk * 32 + k_off) ^
+                                  ((j & 7) * 8)];
       }
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
-        for (int n = 0; n < 2 * NJ; ++n)
+        for (int n = 0; n < 2; ++n)
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
+    buf ^= 1;
   }
 
   const long RSCg = (long)R * S * Cg;
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
-  for (int n = 0; n < 2 * NJ; ++n) {
-    const long c = c0 + wc * 32 * NJ + n * 16 + d_col;
+  for (int n = 0; n < 2; ++n) {
+    const long c = c0 + wc * 32 + n * 16 + d_col;
     if (c >= Cg) continue;
 #pragma unroll
     for (int m = 0; m < 2; ++m) {
@@ -1168,17 +1162,13 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
         : at::zeros({(long)Kout, (long)R * S * Cg},
                     dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb, (unsigned)groups);
+    auto dyT = transpose2d(dy.view({M, (long)Kout}));  // [Kout, M]
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
-      if (nj == 2)
-       hipLaunchKernelGGL(( conv_bwd_w_igemm_kernel<scalar_t, 2>), dim3(grid), dim3(256), 0, cur_stream(), 
-            (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
-            (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
-            C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
-      else
-       hipLaunchKernelGGL(( conv_bwd_w_igemm_kernel<scalar_t, 1>), dim3(grid), dim3(256), 0, cur_stream(), 
-            (const scalar_t*)dy.data_ptr(), (const scalar_t*)x.data_ptr(),
-            (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
-            C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
+     hipLaunchKernelGGL(( conv_bwd_w_igemm_kernel<scalar_t, 1>), dim3(grid), dim3(256), 0, cur_stream(), 
+          (const scalar_t*)dyT.data_ptr(), (const scalar_t*)x.data_ptr(),
+          (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
+          C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice,
+          (const scalar_t*)zero_page(dy));
     });
     HIP_CHECK_LAST();
     return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
