@@ -47,8 +47,14 @@ from .util import is_np_array, set_np, use_np
 from .gluon import metric
 
 def __getattr__(name):
+    if name == 'npx':
+        import importlib
+        mod = importlib.import_module('.numpy_extension', __name__)
+        globals()['npx'] = mod
+        return mod
     # lazily imported to avoid import cycles
-    if name in ('test_utils', 'runtime', 'visualization'):
+    if name in ('test_utils', 'runtime', 'visualization',
+                'numpy_extension'):
         import importlib
         mod = importlib.import_module('.' + name, __name__)
         globals()[name] = mod
