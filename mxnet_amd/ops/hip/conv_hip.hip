@@ -758,6 +758,166 @@ __global__ void scatter_stride_rows_kernel(const T* __restrict__ compact,
   }
 }
 
+// big-M variant: both operands two-hop transposed on chip (the global
+// dy transpose would dominate for large M x Kout panels)
+template <typename T, int NJ>
+__global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_hop2_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
+    int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
+    int dw, long m_per_slice) {
+  using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
+  constexpr int BI = 64, BJ = 64 * NJ, BKM = 64;
+  constexpr int HSTR = 80;         // dy hop row stride (halfs)
+  constexpr int HSTRB = BJ + 16;   // x hop row stride
+  // two-hop transpose staging: global 16B -> Hop[m][i] (coalesced b128
+  // LDS writes) -> per-wave u16 reads (2-way) -> XOR-swizzled operand
+  // tiles [i][m] read by ds_read_b128 fragments at the b128 bank floor.
+  // NJ=2 widens the j (filter-input) tile to 128 halving how often the
+  // dy panel is re-staged from HBM (this kernel is staging-BW bound).
+  __shared__ T HopA[BKM * HSTR];
+  __shared__ T HopB[BKM * HSTRB];
+  __shared__ T DyT[BI * BKM];  // [i][m], idx ^ ((i&7)*8)
+  __shared__ T XT[BJ * BKM];
+
+  const int g = blockIdx.z;  // conv group
+  const int cpl = (Cg + BJ - 1) / BJ;
+  const int nTj = R * S * cpl;
+  const int bid = blockIdx.x;
+  const int i0 = (bid / nTj) * BI;
+  const int jt = bid % nTj;
+  const int rs = jt / cpl;
+  const int c0 = (jt % cpl) * BJ;
+  const int r = rs / S, s = rs % S;
+  const int roff = r * dh, soff = s * dw;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int sm_half = t >> 3;   // m row handled in stage1 (0..31, x2 rnds)
+  const int seg = t & 7;        // 8-element column segment
+
+  float4_t acc[2][2 * NJ] = {};
+
+  // stage1: global -> Hop[m][seg*8..+8)
+  auto stage1 = [&](long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      {
+        const int iseg = i0 + seg * 8;
+        V8 v = {};
+        if (m_ok && iseg + 8 <= Kg)
+          v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + iseg);
+        *(V8*)&HopA[m_l * HSTR + seg * 8] = v;
+      }
+#pragma unroll
+      for (int part = 0; part < NJ; ++part) {
+        V8 v = {};
+        const int jseg = part * 64 + seg * 8;
+        if (m_ok) {
+          int4_t pt = pixtab[m_g];
+          const int ih = pt[1] + roff, iw = pt[2] + soff;
+          const int cseg = c0 + jseg;
+          if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+              cseg + 8 <= Cg)
+            v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                             (long)g * Cg + cseg);
+        }
+        *(V8*)&HopB[m_l * HSTRB + jseg] = v;
+      }
+    }
+  };
+
+  // hop2: Hop[m][i] -> DyT/XT[i][m] (whole wave shares one m-segment so
+  // the u16 gather reads are 2-way; the b128 tile write sits at the
+  // 128 B/cycle LDS floor thanks to the XOR swizzle)
+  auto hop2 = [&]() {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int mseg = rnd * 4 + (t >> 6);
+      const int i_l = lane;
+      V8 va;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj)
+        va[jj] = HopA[(mseg * 8 + jj) * HSTR + i_l];
+      *(V8*)&DyT[(i_l * BKM + mseg * 8) ^ ((i_l & 7) * 8)] = va;
+#pragma unroll
+      for (int part = 0; part < NJ; ++part) {
+        const int j_l = part * 64 + i_l;
+        V8 vb;
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj)
+          vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
+        *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
+      }
+    }
+  };
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  stage1(ms0);
+  __syncthreads();
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+    hop2();
+    __syncthreads();
+    if (mc + BKM < ms1) stage1(mc + BKM);  // overlaps the MFMA phase
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[2], bf[2 * NJ];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int i = wr * 32 + m * 16 + a_row;
+        af[m] = *(const Frag*)&DyT[(i * BKM + kk * 32 + k_off) ^
+                                   ((i & 7) * 8)];
+      }
+#pragma unroll
+      for (int n = 0; n < 2 * NJ; ++n) {
+        const int i = wc * 32 * NJ + n * 16 + a_row;
+        bf[n] = *(const Frag*)&XT[(i * BKM + kk * 32 + k_off) ^
+                                  ((i & 7) * 8)];
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 2 * NJ; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+  }
+
+  const long RSCg = (long)R * S * Cg;
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 2 * NJ; ++n) {
+    const long c = c0 + wc * 32 * NJ + n * 16 + d_col;
+    if (c >= Cg) continue;
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long i = i_base + j;
+        if (i < Kg) {
+          float* dst = dw32 + ((long)g * Kg + i) * RSCg + (long)rs * Cg + c;
+          if (gridDim.y == 1) *dst = acc[m][n][j];
+          else atomicAdd(dst, acc[m][n][j]);
+        }
+      }
+    }
+  }
+}
+
 // pixtab builder: m -> (n, h_base, w_base, valid)
 __global__ void build_pixtab_kernel(int4_t* __restrict__ tab, long M, int P,
                                     int Q, int sh, int sw, int ph, int pw) {
@@ -1162,14 +1322,30 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
         : at::zeros({(long)Kout, (long)R * S * Cg},
                     dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb, (unsigned)groups);
-    auto dyT = transpose2d(dy.view({M, (long)Kout}));  // [Kout, M]
-    DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
-     hipLaunchKernelGGL(( conv_bwd_w_igemm_kernel<scalar_t, 1>), dim3(grid), dim3(256), 0, cur_stream(), 
-          (const scalar_t*)dyT.data_ptr(), (const scalar_t*)x.data_ptr(),
-          (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
-          C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice,
-          (const scalar_t*)zero_page(dy));
-    });
+    // small-M panels: one cheap global dy transpose, then NT staging;
+    // big-M panels: the transpose itself would dominate -> two-hop
+    bool pretranspose = M * (long)Kout * 2 < (80L << 20);
+    if (pretranspose) {
+      auto dyT = transpose2d(dy.view({M, (long)Kout}));  // [Kout, M]
+      DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
+       hipLaunchKernelGGL(( conv_bwd_w_igemm_kernel<scalar_t, 1>)
+            , dim3(grid), dim3(256), 0, cur_stream(), 
+                (const scalar_t*)dyT.data_ptr(),
+                (const scalar_t*)x.data_ptr(),
+                (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H,
+                W, C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice,
+                (const scalar_t*)zero_page(dy));
+      });
+    } else {
+      DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w2", [&] {
+       hipLaunchKernelGGL(( conv_bwd_w_igemm_hop2_kernel<scalar_t, 1>)
+            , dim3(grid), dim3(256), 0, cur_stream(), 
+                (const scalar_t*)dy.data_ptr(),
+                (const scalar_t*)x.data_ptr(),
+                (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H,
+                W, C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice);
+      });
+    }
     HIP_CHECK_LAST();
     return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
   }
