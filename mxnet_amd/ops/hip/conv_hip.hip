@@ -625,13 +625,19 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_c8_kernel(
 // geometry as conv_bwd_w_igemm_kernel, x8-gathered taps.
 template <typename T>
 __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
-    const T* __restrict__ dy, const T* __restrict__ x8,
+    const T* __restrict__ dyT, const T* __restrict__ x8,
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
-    int H, int W, int Kout, int R, long m_per_slice) {
+    int H, int W, int Kout, int R, long m_per_slice,
+    const T* __restrict__ zpage) {
   using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
   constexpr int BI = 64, BKM = 64;
+  constexpr int HSTRB = 80;
+  // stem variant of the v2.5 staging: dy pre-transposed globally (small
+  // Kout panel), x8 taps two-hop transposed (j = tap*8 + channel)
   __shared__ T DyT[2][BI * BKM];
-  __shared__ T XT[2][64 * BKM];
+  __shared__ T HopB[BKM * HSTRB];
+  __shared__ T XT[64 * BKM];
 
   const int nTj = R;
   const int bid = blockIdx.x;
@@ -646,60 +652,78 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
   const int sm_half = t >> 3;
-  const int seg = t & 7;
+  const int seg = t & 7;        // tap index (8 halfs = 1 tap x 8 ch)
 
   float4_t acc[2][2] = {};
 
-  auto stage = [&](int buf, long mc) {
+  auto stage_dy = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int i_l = rnd * 32 + sm_half;
+      const long mcol = mc + seg * 8;
+      const bool ok = i0 + i_l < Kout && mcol + 8 <= M;
+      const T* ga = ok ? dyT + (long)(i0 + i_l) * M + mcol : zpage;
+      gload_lds16c(ga, &DyT[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  auto stage1_x = [&](long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < 2; ++rnd) {
       const int m_l = rnd * 32 + sm_half;
       const long m_g = mc + m_l;
-      const bool m_ok = m_g < ms1;
-      {
-        const int iseg = i0 + seg * 8;
-        using V8 = T __attribute__((ext_vector_type(8)));
-        V8 v = {};
-        if (m_ok && iseg + 8 <= Kout)
-          v = *(const V8*)(dy + m_g * Kout + iseg);
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          DyT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
+      V8 v = {};
+      if (m_g < ms1) {
+        int4_t pt = pixtab[m_g];
+        const int ih = pt[1] + r, iw = pt[2] + seg;   // tap (r, seg)
+        if (ih >= 0 && ih < H && iw >= 0 && iw < W)
+          v = *(const V8*)(x8 + (((long)pt[0] * H + ih) * W + iw) * 8);
       }
-      {
-        using V8 = T __attribute__((ext_vector_type(8)));
-        V8 v = {};
-        if (m_ok) {
-          int4_t pt = pixtab[m_g];
-          const int ih = pt[1] + r, iw = pt[2] + seg;  // tap (r, seg)
-          if (ih >= 0 && ih < H && iw >= 0 && iw < W)
-            v = *(const V8*)(x8 + (((long)pt[0] * H + ih) * W + iw) * 8);
-        }
+      *(V8*)&HopB[m_l * HSTRB + seg * 8] = v;
+    }
+  };
+
+  auto hop2_x = [&]() {
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj)
-          XT[buf][(seg * 8 + jj) * BKM + m_l] = v[jj];
-      }
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int mseg = rnd * 4 + (t >> 6);
+      const int j_l = lane;
+      V8 vb;
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj)
+        vb[jj] = HopB[(mseg * 8 + jj) * HSTRB + j_l];
+      *(V8*)&XT[(j_l * BKM + mseg * 8) ^ ((j_l & 7) * 8)] = vb;
     }
   };
 
   const int a_row = lane & 15;
   const int k_off = (lane >> 4) * 8;
 
-  int buf = 0;
-  stage(0, ms0);
+  stage_dy(0, ms0);
+  stage1_x(ms0);
   __syncthreads();
+  int buf = 0;
   for (long mc = ms0; mc < ms1; mc += BKM) {
+    hop2_x();
+    __syncthreads();
+    if (mc + BKM < ms1) {
+      stage1_x(mc + BKM);
+      stage_dy(buf ^ 1, mc + BKM);
+    }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
       Frag af[2], bf[2];
 #pragma unroll
-      for (int m = 0; m < 2; ++m)
-        af[m] = *(const Frag*)&DyT[buf][(wr * 32 + m * 16 + a_row) * BKM +
-                                        kk * 32 + k_off];
+      for (int m = 0; m < 2; ++m) {
+        const int i = wr * 32 + m * 16 + a_row;
+        af[m] = *(const Frag*)&DyT[buf][i * BKM + kk * 32 + k_off];
+      }
 #pragma unroll
-      for (int n = 0; n < 2; ++n)
-        bf[n] = *(const Frag*)&XT[buf][(wc * 32 + n * 16 + a_row) * BKM +
-                                       kk * 32 + k_off];
+      for (int n = 0; n < 2; ++n) {
+        const int j = wc * 32 + n * 16 + a_row;
+        bf[n] = *(const Frag*)&XT[(j * BKM + kk * 32 + k_off) ^
+                                  ((j & 7) * 8)];
+      }
 #pragma unroll
       for (int m = 0; m < 2; ++m)
 #pragma unroll
@@ -707,11 +731,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_c8_kernel(
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
-    if (mc + BKM < ms1) {
-      stage(buf ^ 1, mc + BKM);
-      buf ^= 1;
-      __syncthreads();
-    }
+    buf ^= 1;
   }
 
   const int d_col = lane & 15;
@@ -1289,11 +1309,12 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
         : at::zeros({(long)Kout, (long)R * 64},
                     dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb);
+    auto dyT8 = transpose2d(dy.view({M, (long)Kout}));
     DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w_c8", [&] {
      hipLaunchKernelGGL(( conv_bwd_w_igemm_c8_kernel<scalar_t>), dim3(grid), dim3(256), 0, cur_stream(), 
-          (const scalar_t*)dy.data_ptr(), (const scalar_t*)x8.data_ptr(),
+          (const scalar_t*)dyT8.data_ptr(), (const scalar_t*)x8.data_ptr(),
           (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
-          Kout, R, m_per_slice);
+          Kout, R, m_per_slice, (const scalar_t*)zero_page(dy));
     });
     HIP_CHECK_LAST();
     // unpad: [K, R, 8tap, 8ch] -> [K, R, S, C]
