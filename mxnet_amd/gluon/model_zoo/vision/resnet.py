@@ -14,10 +14,11 @@ from ... import nn
 from .... import initializer as init
 
 __all__ = ['ResNetV1', 'ResNetV2', 'BasicBlockV1', 'BasicBlockV2',
-           'BottleneckV1', 'BottleneckV2',
+           'BottleneckV1', 'BottleneckV2', 'ResNext', 'ResNextBlock',
            'resnet18_v1', 'resnet34_v1', 'resnet50_v1', 'resnet101_v1',
            'resnet152_v1', 'resnet18_v2', 'resnet34_v2', 'resnet50_v2',
-           'resnet101_v2', 'resnet152_v2', 'get_resnet']
+           'resnet101_v2', 'resnet152_v2', 'resnext50_32x4d',
+           'resnext101_32x4d', 'get_resnet']
 
 
 def _conv3x3(channels, stride, in_channels, layout):
@@ -296,3 +297,86 @@ def resnet101_v2(**kwargs):
 
 def resnet152_v2(**kwargs):
     return get_resnet(2, 152, **kwargs)
+
+
+class ResNextBlock(HybridBlock):
+    """ResNeXt bottleneck (reference resnext.py): grouped 3x3 conv —
+    runs the per-group implicit-GEMM kernels on gfx950."""
+
+    def __init__(self, channels, cardinality, bottleneck_width, stride,
+                 downsample=False, in_channels=0, layout='NCHW', **kwargs):
+        super().__init__(**kwargs)
+        ax = -1 if layout == 'NHWC' else 1
+        D = int(channels * bottleneck_width / 64) * cardinality // 2
+        group_width = max(D, cardinality)
+        self.conv1 = nn.Conv2D(group_width, kernel_size=1, use_bias=False,
+                               in_channels=in_channels, layout=layout)
+        self.bn1 = nn.BatchNormReLU(axis=ax)
+        self.conv2 = nn.Conv2D(group_width, kernel_size=3, strides=stride,
+                               padding=1, groups=cardinality, use_bias=False,
+                               in_channels=group_width, layout=layout)
+        self.bn2 = nn.BatchNormReLU(axis=ax)
+        self.conv3 = nn.Conv2D(channels, kernel_size=1, use_bias=False,
+                               in_channels=group_width, layout=layout)
+        self.bn3 = nn.BatchNorm(axis=ax, fuse_relu=True)
+        if downsample:
+            self.ds_conv = nn.Conv2D(channels, kernel_size=1, strides=stride,
+                                     use_bias=False, in_channels=in_channels,
+                                     layout=layout)
+            self.ds_bn = nn.BatchNorm(axis=ax)
+        else:
+            self.ds_conv = None
+
+    def forward(self, x):
+        residual = x
+        out = self.bn1(self.conv1(x))
+        out = self.bn2(self.conv2(out))
+        out = self.conv3(out)
+        if self.ds_conv is not None:
+            residual = self.ds_bn(self.ds_conv(x))
+        return self.bn3(out, residual)
+
+
+class ResNext(_ResNetBase):
+    """ResNeXt (reference gluon/model_zoo/vision/resnext.py)."""
+
+    def __init__(self, layers, cardinality=32, bottleneck_width=4,
+                 classes=1000, layout='NCHW', **kwargs):
+        super().__init__(layout=layout, **kwargs)
+        channels = [256, 512, 1024, 2048]
+        self.features = nn.HybridSequential()
+        ax = -1 if layout == 'NHWC' else 1
+        self.features.add(
+            nn.Conv2D(64, kernel_size=7, strides=2, padding=3,
+                      use_bias=False, in_channels=3, layout=layout),
+            nn.BatchNormReLU(axis=ax),
+            nn.MaxPool2D(pool_size=3, strides=2, padding=1, layout=layout))
+        in_c = 64
+        for i, num in enumerate(layers):
+            stride = 1 if i == 0 else 2
+            blk = nn.HybridSequential()
+            blk.add(ResNextBlock(channels[i], cardinality, bottleneck_width,
+                                 stride, True, in_channels=in_c,
+                                 layout=layout))
+            for _ in range(num - 1):
+                blk.add(ResNextBlock(channels[i], cardinality,
+                                     bottleneck_width, 1, False,
+                                     in_channels=channels[i], layout=layout))
+            self.features.add(blk)
+            in_c = channels[i]
+        self.features.add(nn.GlobalAvgPool2D(layout=layout))
+        self.output = nn.Dense(classes)
+
+    def forward(self, x):
+        x = self._maybe_to_layout(x)
+        x = self.features(x)
+        x = x.flatten()
+        return self.output(x)
+
+
+def resnext50_32x4d(**kwargs):
+    return ResNext([3, 4, 6, 3], cardinality=32, bottleneck_width=4, **kwargs)
+
+
+def resnext101_32x4d(**kwargs):
+    return ResNext([3, 4, 23, 3], cardinality=32, bottleneck_width=4, **kwargs)

@@ -45,6 +45,7 @@ def _train_step(net, x_shape, classes=10, dtype='float16'):
     ('mobilenet0_5', (4, 3, 64, 64)),       # depthwise kernels
     ('mobilenet_v2_1_0', (4, 3, 64, 64)),
     ('inception_v3', (2, 3, 299, 299)),
+    ('resnext50_32x4d', (2, 64, 64, 3)),  # grouped igemm kernels
 ])
 def test_vision_family_gpu(name, shape):
     from mxnet_amd.gluon.model_zoo import vision

@@ -25,7 +25,7 @@ def _train_step(net, x, label, classes):
 
 @pytest.mark.parametrize('name', ['resnet18_v1', 'resnet50_v1', 'resnet18_v2',
                                   'vgg11', 'alexnet', 'squeezenet1_0',
-                                  'densenet121', 'mobilenet0_5',
+                                  'densenet121', 'mobilenet0_5', 'resnext50_32x4d',
                                   'inception_v3'])
 def test_vision_model(name):
     from mxnet_amd.gluon.model_zoo import vision
