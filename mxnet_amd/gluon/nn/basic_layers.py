@@ -130,15 +130,94 @@ class BatchNormReLU(BatchNorm):
         super().__init__(**kwargs)
 
 
+class _SyncBNFn(torch.autograd.Function):
+    """Cross-rank BN: batch statistics (and their gradients) are
+    all-reduced over torch.distributed (RCCL on GPU / gloo on CPU) —
+    the reference used an intra-process barrier registry
+    (sync_batch_norm-inl.h:79-195); one collective per direction here."""
+
+    @staticmethod
+    def forward(ctx, x, gamma, beta, rmean, rvar, momentum, eps, axis):
+        import torch.distributed as dist
+        dims = [d for d in range(x.dim()) if d != axis % x.dim()]
+        n_local = x.numel() // x.shape[axis]
+        xs = x.float()
+        stats = torch.cat([xs.sum(dims), (xs * xs).sum(dims),
+                           torch.tensor([float(n_local)], device=x.device)])
+        world = dist.get_world_size() if dist.is_initialized() else 1
+        if world > 1:
+            dist.all_reduce(stats)
+        C = x.shape[axis]
+        n_total = stats[-1]
+        mean = stats[:C] / n_total
+        var = (stats[C:2 * C] / n_total - mean * mean).clamp_min(0)
+        istd = (var + eps).rsqrt()
+        with torch.no_grad():
+            unbias = var * n_total / (n_total - 1).clamp_min(1)
+            rmean.mul_(momentum).add_((1 - momentum) * mean)
+            rvar.mul_(momentum).add_((1 - momentum) * unbias)
+        shape = [1] * x.dim()
+        shape[axis % x.dim()] = C
+        xhat = (xs - mean.view(shape)) * istd.view(shape)
+        y = (xhat * gamma.view(shape) + beta.view(shape)).to(x.dtype)
+        ctx.save_for_backward(x, gamma, mean, istd)
+        ctx.axis, ctx.n_total, ctx.world = axis, float(n_total), world
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        import torch.distributed as dist
+        x, gamma, mean, istd = ctx.saved_tensors
+        axis = ctx.axis % x.dim()
+        dims = [d for d in range(x.dim()) if d != axis]
+        C = x.shape[axis]
+        shape = [1] * x.dim()
+        shape[axis] = C
+        g = dy.float()
+        xhat = (x.float() - mean.view(shape)) * istd.view(shape)
+        s1 = g.sum(dims)
+        s2 = (g * xhat).sum(dims)
+        if ctx.world > 1:
+            pair = torch.cat([s1, s2])
+            dist.all_reduce(pair)
+            s1, s2 = pair[:C], pair[C:]
+        invN = 1.0 / ctx.n_total
+        dx = (gamma.view(shape) * istd.view(shape) *
+              (g - invN * s1.view(shape) - xhat * invN * s2.view(shape)))
+        # s1/s2 are GLOBAL sums after the all-reduce (identical on every
+        # rank); the trainer AVERAGES grads across ranks, which preserves
+        # them — so return the global sums directly.
+        return dx.to(x.dtype), s2, s1, None, None, None, None, None
+
+
 class SyncBatchNorm(BatchNorm):
-    """Cross-rank synchronized BN. On the 1-proc-per-GPU RCCL layout,
-    stats are all-reduced over torch.distributed (reference
-    sync_batch_norm-inl.h used an intra-process barrier registry)."""
+    """Cross-rank synchronized BN (see _SyncBNFn)."""
 
     def __init__(self, in_channels=0, num_devices=None, **kwargs):
         super().__init__(in_channels=in_channels, **kwargs)
-        # round-1: per-rank stats (correct single-process); RCCL stat
-        # all-reduce lands with the distributed trainer work.
+
+    def forward(self, x, residual=None):
+        import torch.distributed as dist
+        from ... import autograd as _ag
+        self._finish_deferred(x)
+        if not (_ag.is_training() and dist.is_initialized()
+                and dist.get_world_size() > 1):
+            return super().forward(x, residual)
+        ctx = self._param_ctx((x,))
+        from ...ndarray.ndarray import NDArray
+        t = x.handle if hasattr(x, 'handle') else x
+        y = _SyncBNFn.apply(
+            t, self.gamma.data(ctx).handle.float(),
+            self.beta.data(ctx).handle.float(),
+            self.running_mean.data(ctx).handle,
+            self.running_var.data(ctx).handle,
+            self._momentum, self._epsilon, self._axis)
+        if residual is not None:
+            y = y + (residual.handle if hasattr(residual, 'handle')
+                     else residual)
+        if self._fuse_relu:
+            y = torch.relu(y)
+        return NDArray(y)
 
 
 class Embedding(HybridBlock):

@@ -96,3 +96,96 @@ def test_dp_trainer_matches_single_process(tmp_path):
         tr.step(8)
     np.testing.assert_allclose(w0, net.weight.data().asnumpy(), rtol=1e-5,
                                atol=1e-6)
+
+
+_SYNCBN_WORKER = '''
+import os
+import numpy as np
+import torch
+import mxnet_amd as mx
+from mxnet_amd import autograd
+from mxnet_amd.gluon import nn
+
+rank = int(os.environ['RANK'])
+import torch.distributed as dist
+dist.init_process_group('gloo', rank=rank,
+                        world_size=int(os.environ['WORLD_SIZE']))
+torch.manual_seed(3)
+bn = nn.SyncBatchNorm(in_channels=3)
+bn.initialize()
+torch.manual_seed(11)
+X = torch.randn(8, 3, 5, 5)
+W = torch.randn(8, 3, 5, 5)  # fixed weights make dy vary per element
+x = mx.nd.from_torch(X[rank * 4:(rank + 1) * 4].clone())
+with autograd.record():
+    y = bn(x)
+    L = mx.nd.from_torch((y.handle * W[rank * 4:(rank + 1) * 4]).sum())
+L.backward()
+pre = os.environ['OUT_PREFIX']
+np.save(pre + f'_y{rank}.npy', y.asnumpy())
+np.save(pre + f'_gg{rank}.npy', bn.gamma.grad().asnumpy())
+np.save(pre + f'_gb{rank}.npy', bn.beta.grad().asnumpy())
+np.save(pre + f'_rm{rank}.npy', bn.running_mean.data().asnumpy())
+np.save(pre + f'_rv{rank}.npy', bn.running_var.data().asnumpy())
+print('SYNCBN_OK', rank)
+'''
+
+
+def test_sync_batchnorm_matches_full_batch(tmp_path):
+    """SyncBN over 2 ranks (half-batch each) == plain BN on the full batch:
+    outputs, gamma/beta grads, and running statistics all agree."""
+    script = tmp_path / 'sbn.py'
+    script.write_text(_SYNCBN_WORKER)
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    import socket
+    sock = socket.socket()
+    sock.bind(('127.0.0.1', 0))
+    port = str(sock.getsockname()[1])
+    sock.close()
+    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': port,
+                'OUT_PREFIX': str(tmp_path / 's'),
+                'PYTHONPATH': repo_root + os.pathsep + env.get('PYTHONPATH', '')})
+    procs = []
+    for rank in range(2):
+        e = dict(env, RANK=str(rank), WORLD_SIZE='2', LOCAL_RANK=str(rank))
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        assert p.returncode == 0, out.decode()
+
+    # single-process oracle: plain BN over the concatenated batch
+    import mxnet_amd as mx
+    from mxnet_amd import autograd
+    from mxnet_amd.gluon import nn
+    torch.manual_seed(3)
+    bn = nn.BatchNorm(in_channels=3)
+    bn.initialize()
+    torch.manual_seed(11)
+    X = torch.randn(8, 3, 5, 5)
+    W = torch.randn(8, 3, 5, 5)
+    x = mx.nd.from_torch(X.clone())
+    with autograd.record():
+        y = bn(x)
+        L = mx.nd.from_torch((y.handle * W).sum())
+    L.backward()
+
+    y_ref = y.asnumpy()
+    for rank in range(2):
+        yr = np.load(tmp_path / f's_y{rank}.npy')
+        np.testing.assert_allclose(yr, y_ref[rank * 4:(rank + 1) * 4],
+                                   rtol=1e-4, atol=1e-4)
+        np.testing.assert_allclose(np.load(tmp_path / f's_gg{rank}.npy'),
+                                   bn.gamma.grad().asnumpy(),
+                                   rtol=1e-4, atol=1e-3)
+        np.testing.assert_allclose(np.load(tmp_path / f's_gb{rank}.npy'),
+                                   bn.beta.grad().asnumpy(),
+                                   rtol=1e-4, atol=1e-3)
+        np.testing.assert_allclose(np.load(tmp_path / f's_rm{rank}.npy'),
+                                   bn.running_mean.data().asnumpy(),
+                                   rtol=1e-4, atol=1e-5)
+        np.testing.assert_allclose(np.load(tmp_path / f's_rv{rank}.npy'),
+                                   bn.running_var.data().asnumpy(),
+                                   rtol=1e-4, atol=1e-4)
