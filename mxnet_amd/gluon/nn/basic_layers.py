@@ -226,12 +226,16 @@ class Embedding(HybridBlock):
         super().__init__(**kwargs)
         self._input_dim = input_dim
         self._output_dim = output_dim
+        self._sparse_grad = sparse_grad
         self.weight = Parameter('weight', shape=(input_dim, output_dim),
-                                dtype=dtype, init=weight_initializer)
+                                dtype=dtype, init=weight_initializer,
+                                grad_stype='row_sparse' if sparse_grad
+                                else 'default')
 
     def hybrid_forward(self, F, x, weight):
         return F.Embedding(x, weight, input_dim=self._input_dim,
-                           output_dim=self._output_dim)
+                           output_dim=self._output_dim,
+                           sparse_grad=self._sparse_grad)
 
     def __repr__(self):
         return f'Embedding({self._input_dim} -> {self._output_dim})'

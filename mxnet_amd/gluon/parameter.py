@@ -37,6 +37,7 @@ class Parameter:
         self.wd_mult = wd_mult
         self.init = init
         self.allow_deferred_init = allow_deferred_init
+        self.grad_stype = grad_stype  # 'row_sparse' => lazy sparse update
         self._data = None          # OrderedDict[Context, NDArray]
         self._grad = None
         self._deferred_init = None

@@ -92,6 +92,7 @@ class Trainer:
         if self._kvstore is None:
             return
         if self._distributed:
+            self._densify_rowsparse()
             n = self._kvstore.num_workers
             if n > 1 and self._compression_params is None:
                 # overlapped path: grad hooks launched each bucket's
@@ -235,9 +236,50 @@ class Trainer:
         self._optimizer.rescale_grad = self._scale / batch_size
         self._update(ignore_stale_grad)
 
+    @staticmethod
+    def _pop_rowsparse(p):
+        """Collect and coalesce the row-sparse grad parts stashed by
+        _SparseEmbedding.backward (None if the param is dense)."""
+        if getattr(p, 'grad_stype', 'default') != 'row_sparse':
+            return None
+        wt = p.list_data()[0]._t
+        parts = getattr(wt, '_rowsparse_parts', None)
+        if not parts:
+            return None
+        from ..ndarray.sparse import RowSparseNDArray
+        if len(parts) == 1:
+            rows, vals = parts[0]
+        else:
+            allr = torch.cat([r for r, _ in parts])
+            allv = torch.cat([v for _, v in parts])
+            rows, inv = torch.unique(allr, sorted=True, return_inverse=True)
+            vals = torch.zeros(rows.numel(), allv.shape[-1],
+                               dtype=allv.dtype, device=allv.device)
+            vals.index_add_(0, inv, allv)
+        wt._rowsparse_parts = []
+        return RowSparseNDArray(vals, rows, tuple(wt.shape))
+
+    def _densify_rowsparse(self):
+        """Distributed path: fold sparse payloads into the dense grad so
+        the bucketed all-reduce stays uniform across ranks (row-sparse
+        all-reduce lands with multi-node KVStore work)."""
+        with torch.no_grad():
+            for p in self._params:
+                rs = self._pop_rowsparse(p)
+                if rs is not None:
+                    g = p.list_grad()[0]._t
+                    g.index_add_(0, rs.indices, rs.data.to(g.dtype))
+
     def _update(self, ignore_stale_grad=False):
         for i, p in enumerate(self._params):
             self._check_states(i, p)
+            rs = self._pop_rowsparse(p)
+            if rs is not None:
+                self._optimizer.update_multi_precision(
+                    i, p.list_data()[0], rs, self._states[i])
+                if p.grad_req == 'write':
+                    p.zero_grad()
+                continue
             if len(p.list_data()) == 1:
                 self._optimizer.update_multi_precision(
                     i, p.list_data()[0], p.list_grad()[0], self._states[i])

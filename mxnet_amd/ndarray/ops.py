@@ -90,8 +90,9 @@ def LayerNorm(data, gamma, beta, axis=-1, eps=1e-5, **kwargs):
     return NDArray(_nn.layer_norm(_t(data), _t(gamma), _t(beta), axis, eps))
 
 
-def Embedding(data, weight, input_dim=None, output_dim=None, dtype=None, **kwargs):
-    return NDArray(_nn.embedding(_t(data), _t(weight)))
+def Embedding(data, weight, input_dim=None, output_dim=None, dtype=None,
+              sparse_grad=False, **kwargs):
+    return NDArray(_nn.embedding(_t(data), _t(weight), sparse_grad))
 
 
 def Dropout(data, p=0.5, mode='training', **kwargs):

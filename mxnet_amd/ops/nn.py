@@ -459,8 +459,39 @@ class _Embedding(torch.autograd.Function):
         return None, dw
 
 
-def embedding(indices, weight):
+def embedding(indices, weight, sparse_grad=False):
+    if sparse_grad:
+        return _SparseEmbedding.apply(indices, weight)
     return _Embedding.apply(indices, weight)
+
+
+class _SparseEmbedding(torch.autograd.Function):
+    """Embedding whose weight gradient is ROW-SPARSE: backward coalesces
+    dy into (rows, vals) and stashes it on the weight tensor for the
+    Trainer's lazy update instead of materializing a dense [V, D] grad
+    (reference: Embedding sparse_grad + kRowSparseStorage,
+    src/operator/tensor/indexing_op.cc EmbeddingOpBackwardEx)."""
+
+    @staticmethod
+    def forward(ctx, indices, weight):
+        ctx.save_for_backward(indices)
+        ctx.weight_ref = weight
+        return weight[indices.long()]
+
+    @staticmethod
+    def backward(ctx, dy):
+        (indices,) = ctx.saved_tensors
+        flat = indices.long().reshape(-1)
+        dy2 = dy.contiguous().reshape(-1, dy.shape[-1])
+        rows, inv = torch.unique(flat, sorted=True, return_inverse=True)
+        vals = torch.zeros(rows.numel(), dy2.shape[-1], dtype=dy2.dtype,
+                           device=dy2.device)
+        vals.index_add_(0, inv, dy2)
+        w = ctx.weight_ref
+        if not hasattr(w, '_rowsparse_parts'):
+            w._rowsparse_parts = []
+        w._rowsparse_parts.append((rows, vals))
+        return None, None  # weight grad delivered out-of-band (row-sparse)
 
 
 def dropout(x, p=0.5, training=True):

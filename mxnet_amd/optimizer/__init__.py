@@ -107,9 +107,10 @@ class Optimizer:
 class SGD(Optimizer):
     """SGD with momentum (reference sgd_mom_update, optimizer_op-inl.h)."""
 
-    def __init__(self, momentum=0.0, lazy_update=False, **kwargs):
+    def __init__(self, momentum=0.0, lazy_update=True, **kwargs):
         super().__init__(**kwargs)
         self.momentum = momentum
+        self.lazy_update = lazy_update
 
     def create_state(self, index, weight):
         if self.momentum == 0:
@@ -118,6 +119,9 @@ class SGD(Optimizer):
         return torch.zeros_like(w, dtype=torch.float32)
 
     def update(self, index, weight, grad, state):
+        from ..ndarray.sparse import RowSparseNDArray
+        if isinstance(grad, RowSparseNDArray):
+            return self._update_row_sparse(index, weight, grad, state)
         self._update_count(index)
         lr, wd = self._get_lr(index), self._get_wd(index)
         w = weight._t if isinstance(weight, NDArray) else weight
@@ -132,7 +136,40 @@ class SGD(Optimizer):
                 upd = g
             w.sub_((lr * upd).to(w.dtype))
 
+    def _update_row_sparse(self, index, weight, grad, state):
+        """Lazy row-sparse SGD (reference sgd_mom_update on
+        kRowSparseStorage, optimizer_op.cc SGDMomLazyUpdate): only rows
+        present in the gradient touch weights / momentum — O(nnz·D)
+        instead of O(V·D) for embedding tables."""
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        w = weight._t if isinstance(weight, NDArray) else weight
+        rows, vals = grad.indices, grad.data
+        with torch.no_grad():
+            g = self._preprocess_grad(vals)
+            wr = w.index_select(0, rows).float()
+            g = g + wd * wr
+            if state is not None and self.momentum != 0:
+                if self.lazy_update:
+                    mr = state.index_select(0, rows)
+                    mr.mul_(self.momentum).add_(g)
+                    state.index_copy_(0, rows, mr)
+                    upd = mr
+                else:
+                    state.mul_(self.momentum)
+                    state.index_add_(0, rows, g)
+                    upd = state.index_select(0, rows)
+            else:
+                upd = g
+            w.index_copy_(0, rows, (wr - lr * upd).to(w.dtype))
+
     def update_multi_precision(self, index, weight, grad, state):
+        from ..ndarray.sparse import RowSparseNDArray
+        if isinstance(grad, RowSparseNDArray):
+            return self._update_row_sparse(index, weight, grad, state)
+        return self._ump_dense(index, weight, grad, state)
+
+    def _ump_dense(self, index, weight, grad, state):
         """Single fused HIP kernel on GPU (reference mp_sgd_mom_update):
         rescale+clip+wd+momentum+master-weight update+fp16 cast in one pass."""
         w = weight._t if isinstance(weight, NDArray) else weight

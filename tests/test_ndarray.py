@@ -130,3 +130,43 @@ def test_sparse_ndarray():
     assert both.tostype('default').asnumpy()[4].sum() == 6
     assert S.retain(rs, [4]).indices.tolist() == [4]
     assert S.zeros('row_sparse', (5, 2)).tostype('default').asnumpy().sum() == 0
+
+
+def test_embedding_sparse_grad_end_to_end():
+    """Embedding(sparse_grad=True) -> RowSparse grad -> SGD lazy update
+    matches the dense path; untouched rows never move under momentum
+    (reference: sparse_grad Embedding + sgd lazy_update)."""
+    import torch
+    from mxnet_amd import autograd
+    from mxnet_amd.gluon import Trainer, nn as gnn
+
+    results = {}
+    for sparse in (False, True):
+        torch.manual_seed(5)
+        emb = gnn.Embedding(50, 8, sparse_grad=sparse)
+        emb.initialize()
+        tr = Trainer(emb.collect_params(), 'sgd',
+                     {'learning_rate': 0.1, 'momentum': 0.9,
+                      'lazy_update': sparse}, kvstore=None)
+        idx = mx.nd.from_torch(torch.tensor([[1, 3, 3], [7, 1, 4]]))
+        for _ in range(2):
+            with autograd.record():
+                out = emb(idx)
+                L = mx.nd.from_torch((out.handle ** 2).sum())
+            L.backward()
+            tr.step(1)
+        results[sparse] = emb.weight.data().asnumpy().copy()
+
+    touched = [1, 3, 4, 7]
+    onp = np
+    # touched rows agree between dense and sparse paths
+    onp.testing.assert_allclose(results[True][touched],
+                                results[False][touched], rtol=1e-5, atol=1e-6)
+    # untouched rows identical to init in both
+    torch.manual_seed(5)
+    emb0 = gnn.Embedding(50, 8)
+    emb0.initialize()
+    w0 = emb0.weight.data().asnumpy()
+    untouched = [r for r in range(50) if r not in touched]
+    onp.testing.assert_allclose(results[True][untouched], w0[untouched])
+    onp.testing.assert_allclose(results[False][untouched], w0[untouched])
