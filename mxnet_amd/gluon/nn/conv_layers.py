@@ -203,27 +203,19 @@ class Conv2DTranspose(HybridBlock):
             self.weight.shape = (in_c, self._channels // self._groups, kh, kw)
 
     def forward(self, x):
-        import torch
-        import torch.nn.functional as F
         from ...ndarray.ndarray import NDArray
+        from ...ops import nn as _onn
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
         w = self.weight.data(ctx).handle
         b = self.bias.data(ctx).handle if self.bias is not None else None
         t = x.handle if hasattr(x, 'handle') else x
-        if self._layout == 'NHWC':
-            t = t.permute(0, 3, 1, 2)
-            w = w.permute(0, 3, 1, 2)
-        y = F.conv_transpose2d(t.float(), w.float(),
-                               b.float() if b is not None else None,
-                               stride=self._strides, padding=self._padding,
-                               output_padding=self._out_pad,
-                               groups=self._groups, dilation=self._dilation)
-        y = y.to(t.dtype)
-        if self._layout == 'NHWC':
-            y = y.permute(0, 2, 3, 1).contiguous()
+        if b is not None:
+            b = b.to(t.dtype)
+        y = _onn.deconv2d(t, w.to(t.dtype), b, self._strides, self._padding,
+                          self._out_pad, self._dilation, self._groups,
+                          layout=self._layout)
         if self._act_type:
-            from ...ops import nn as _onn
             y = _onn.activation(y, self._act_type)
         return NDArray(y)
 

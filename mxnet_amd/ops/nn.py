@@ -132,6 +132,88 @@ class _Conv2dNHWC(torch.autograd.Function):
         return dx, dw, db, None, None, None, None
 
 
+class _Deconv2dNHWC(torch.autograd.Function):
+    """Transposed conv on the conv implicit-GEMM MFMA kernels with the
+    roles swapped (reference deconvolution.cc/.cu computes it the same
+    way from the conv primitives): forward = conv_bwd_data, input grad =
+    conv_fwd, weight grad = conv_bwd_weight with (input, out-grad) =
+    (dy, x).  x: [N,H,W,Cin], w: [Cin,R,S,Cout/g], y: [N,Ho,Wo,Cout].
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, b, stride, pad, dilation, groups, out_hw):
+        ctx.save_for_backward(x, w)
+        ctx.conf = (stride, pad, dilation, groups)
+        ctx.has_bias = b is not None
+        ext = hip_required('deconv2d')
+        y = ext.conv2d_nhwc_bwd_data(x, w, out_hw[0], out_hw[1],
+                                     stride[0], stride[1], pad[0], pad[1],
+                                     dilation[0], dilation[1], groups)
+        if b is not None:
+            y = y + b.to(y.dtype)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        stride, pad, dilation, groups = ctx.conf
+        dy = dy.contiguous()
+        ext = hip_required('deconv2d')
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext.conv2d_nhwc_fwd(dy, w, None, stride[0], stride[1],
+                                     pad[0], pad[1], dilation[0],
+                                     dilation[1], groups)
+        if ctx.needs_input_grad[1]:
+            # conv picture: input = dy, out-grad = x
+            dw = ext.conv2d_nhwc_bwd_weight(x, dy, w.shape[1], w.shape[2],
+                                            stride[0], stride[1],
+                                            pad[0], pad[1], dilation[0],
+                                            dilation[1], groups)
+        if ctx.has_bias:
+            db = ext.colsum(dy.reshape(-1, dy.shape[-1]))
+        return dx, dw, db, None, None, None, None, None
+
+
+def deconv2d(x, w, b=None, stride=(1, 1), pad=(0, 0), out_pad=(0, 0),
+             dilation=(1, 1), groups=1, layout='NCHW'):
+    """Transposed convolution. GPU: native igemm kernels (NHWC internal);
+    CPU: torch conv_transpose2d oracle.  w is [Cin, R, S, Cout/g] for
+    NHWC, [Cin, Cout/g, R, S] for NCHW (reference Deconvolution layout)."""
+    if use_hip(x):
+        if layout != 'NHWC':
+            xh = x.permute(0, 2, 3, 1).contiguous()
+            wh = w.permute(0, 2, 3, 1).contiguous()
+        else:
+            xh, wh = x.contiguous(), w.contiguous()
+        H, W = xh.shape[1], xh.shape[2]
+        R, S = wh.shape[1], wh.shape[2]
+        ho = (H - 1) * stride[0] - 2 * pad[0] + dilation[0] * (R - 1) + 1 \
+            + out_pad[0]
+        wo = (W - 1) * stride[1] - 2 * pad[1] + dilation[1] * (S - 1) + 1 \
+            + out_pad[1]
+        y = _Deconv2dNHWC.apply(xh, wh, b, stride, pad, dilation, groups,
+                                (ho, wo))
+        if layout != 'NHWC':
+            y = y.permute(0, 3, 1, 2).contiguous()
+        return y
+    if layout == 'NHWC':
+        xn = x.permute(0, 3, 1, 2)
+        wn = w.permute(0, 3, 1, 2)
+        y = F.conv_transpose2d(xn.float(), wn.float(),
+                               b.float() if b is not None else None,
+                               stride=stride, padding=pad,
+                               output_padding=out_pad, groups=groups,
+                               dilation=dilation)
+        return y.to(x.dtype).permute(0, 2, 3, 1).contiguous()
+    y = F.conv_transpose2d(x.float(), w.float(),
+                           b.float() if b is not None else None,
+                           stride=stride, padding=pad,
+                           output_padding=out_pad, groups=groups,
+                           dilation=dilation)
+    return y.to(x.dtype)
+
+
 def conv2d(x, w, b=None, stride=(1, 1), pad=(0, 0), dilation=(1, 1),
            groups=1, layout='NCHW'):
     if layout == 'NHWC':
