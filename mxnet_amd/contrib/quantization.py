@@ -8,7 +8,7 @@ import torch
 from ..ndarray.ndarray import NDArray
 
 __all__ = ['quantize', 'dequantize', 'calib_minmax', 'QuantizedDense',
-           'quantize_net']
+           'QuantizedConv2D', 'quantize_net']
 
 
 def _t(x):
@@ -75,17 +75,81 @@ class QuantizedDense:
         return NDArray(y.reshape(*t.shape[:-1], y.shape[-1]))
 
 
+class QuantizedConv2D:
+    """Int8 inference conv (NHWC, groups=1): weight pre-quantized to
+    [K, R*S*C] int8 rows; per batch the input is im2col'd (native
+    kernel), quantized, and multiplied on the i8 MFMA GEMM with fused
+    rescale (reference quantized_conv.cc — cuDNN int8 there; gfx950
+    ``v_mfma_i32_16x16x64_i8`` here)."""
+
+    def __init__(self, conv_layer):
+        assert conv_layer._groups == 1, 'int8 conv: groups==1 only'
+        assert conv_layer._layout == 'NHWC', 'int8 conv: NHWC only'
+        w = conv_layer.weight.data().handle  # [K, R, S, C]
+        self._K, self._R, self._S, self._C = w.shape
+        self._stride = conv_layer._strides
+        self._pad = conv_layer._padding
+        self._dil = conv_layer._dilation
+        self._w_scale = calib_minmax(w)
+        w2 = w.reshape(self._K, -1).contiguous()
+        if w.is_cuda:
+            from ..ops.dispatch import hip_required
+            self._wq = hip_required('quantized_conv').quantize_i8(
+                w2, self._w_scale)
+        else:
+            self._wq = torch.clamp(torch.round(w2.float() / self._w_scale),
+                                   -127, 127).to(torch.int8)
+        b = conv_layer.bias
+        self._bias = b.data().handle.float() if b is not None else None
+        self._out_dtype = w.dtype
+        self._act = conv_layer._act_type
+
+    def __call__(self, x):
+        t = _t(x)
+        N, H, W, C = t.shape
+        (sh, sw), (ph, pw), (dh, dw) = self._stride, self._pad, self._dil
+        P = (H + 2 * ph - dh * (self._R - 1) - 1) // sh + 1
+        Q = (W + 2 * pw - dw * (self._S - 1) - 1) // sw + 1
+        xs = calib_minmax(t)
+        if t.is_cuda:
+            from ..ops.dispatch import hip_required
+            ext = hip_required('quantized_conv')
+            col = ext.im2col_nhwc(t.contiguous(), self._R, self._S,
+                                  sh, sw, ph, pw, dh, dw)
+            xq = ext.quantize_i8(col, xs)
+            y = ext.gemm_nt_i8(xq, self._wq, xs * self._w_scale,
+                               self._out_dtype)
+        else:
+            xn = t.float().permute(0, 3, 1, 2)
+            col = torch.nn.functional.unfold(
+                xn, (self._R, self._S), dilation=(dh, dw),
+                padding=(ph, pw), stride=(sh, sw))  # [N, C*R*S, P*Q]
+            # unfold is c-major [C,R,S]; our layout is [R,S,C]
+            col = col.reshape(N, C, self._R * self._S, -1)                      .permute(0, 3, 2, 1).reshape(-1, self._R * self._S * C)
+            xq = torch.clamp(torch.round(col / xs), -127, 127)
+            y = (xq @ self._wq.float().t() * (xs * self._w_scale))                 .to(self._out_dtype)
+        if self._bias is not None:
+            y = y + self._bias.to(y.dtype)
+        y = y.reshape(N, P, Q, self._K)
+        if self._act == 'relu':
+            y = torch.relu(y)
+        return NDArray(y)
+
+
 def quantize_net(net, quantized_dtype='int8', exclude_layers=None):
-    """Swap Dense layers for int8 inference versions (reference
-    quantize_model; conv quantization lands with the int8 conv kernel)."""
+    """Swap Dense / NHWC Conv2D layers for int8 inference versions
+    (reference quantize_model)."""
     from ..gluon import nn
     swapped = []
     def visit(block):
         for name, child in list(block._children.items()):
-            if isinstance(child, nn.Dense) and \
-                    (not exclude_layers or name not in exclude_layers):
-                q = QuantizedDense(child)
-                swapped.append((block, name, q))
+            if exclude_layers and name in exclude_layers:
+                continue
+            if isinstance(child, nn.Dense):
+                swapped.append((block, name, QuantizedDense(child)))
+            elif isinstance(child, nn.Conv2D) and child._groups == 1 and \
+                    child._layout == 'NHWC':
+                swapped.append((block, name, QuantizedConv2D(child)))
             else:
                 visit(child)
     visit(net)

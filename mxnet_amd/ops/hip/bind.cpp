@@ -60,6 +60,8 @@ at::Tensor pool_nhwc_bwd(const at::Tensor&, const at::Tensor&,
 at::Tensor quantize_i8(const at::Tensor&, double);
 at::Tensor dequantize_i8(const at::Tensor&, double,
                          c10::optional<at::ScalarType>);
+at::Tensor im2col_nhwc_op(const at::Tensor&, int, int, int, int, int, int,
+                          int, int);
 at::Tensor gemm_nt_i8(const at::Tensor&, const at::Tensor&, double,
                       c10::optional<at::ScalarType>);
 // elemwise.hip
@@ -123,6 +125,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_i8", &quantize_i8);
   m.def("dequantize_i8", &dequantize_i8, py::arg("x"), py::arg("scale"),
         py::arg("dtype") = py::none());
+  m.def("im2col_nhwc", &im2col_nhwc_op);
   m.def("gemm_nt_i8", &gemm_nt_i8, py::arg("a"), py::arg("b"),
         py::arg("scale"), py::arg("out_dtype") = py::none());
 }

@@ -1058,6 +1058,10 @@ __global__ void dwconv_bwd_w_kernel(const T* __restrict__ dy,
 // host wrappers
 // ---------------------------------------------------------------------------
 
+// public im2col (int8 quantized conv builds its GEMM operand with it)
+at::Tensor im2col_nhwc_op(const at::Tensor& x, int R, int S, int sh, int sw,
+                          int ph, int pw, int dh, int dw);
+
 static at::Tensor im2col_nhwc(const at::Tensor& x, int P, int Q, int R, int S,
                               int sh, int sw, int ph, int pw, int dh, int dw) {
   int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
@@ -1375,4 +1379,14 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
   auto colT = transpose2d(col);
   auto dwf = gemm_nt_core(dyT, colT, c10::nullopt, false);  // [K, RSC]
   return dwf.view({(long)Kout, R, S, (long)C});
+}
+
+
+at::Tensor im2col_nhwc_op(const at::Tensor& x, int R, int S, int sh, int sw,
+                          int ph, int pw, int dh, int dw) {
+  CHECK_GPU(x); CHECK_CONTIG(x);
+  int H = x.size(1), W = x.size(2);
+  int P = (H + 2 * ph - dh * (R - 1) - 1) / sh + 1;
+  int Q = (W + 2 * pw - dw * (S - 1) - 1) / sw + 1;
+  return im2col_nhwc(x, P, Q, R, S, sh, sw, ph, pw, dh, dw);
 }

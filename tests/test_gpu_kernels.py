@@ -584,3 +584,24 @@ def test_conv2dtranspose_layer_gpu():
     assert y.shape == (2, 14, 14, 32)
     g = net.weight.grad().handle
     assert g is not None and torch.isfinite(g.float()).all()
+
+
+def test_quantized_conv_gpu():
+    """Int8 conv (im2col + i8 MFMA GEMM) tracks the fp16 conv within
+    quantization error."""
+    import mxnet_amd as mx
+    from mxnet_amd.gluon import nn as gnn
+    from mxnet_amd.contrib.quantization import QuantizedConv2D
+    torch.manual_seed(1)
+    c = gnn.Conv2D(32, 3, strides=1, padding=1, layout='NHWC',
+                   in_channels=64, use_bias=True)
+    c.initialize(ctx=mx.gpu(0))
+    c.cast('float16')
+    x = mx.nd.from_torch(torch.randn(2, 14, 14, 64, device='cuda',
+                                     dtype=torch.float16))
+    y_fp = c(x).handle.float()
+    q = QuantizedConv2D(c)
+    y_q = q(x).handle.float()
+    assert y_q.shape == y_fp.shape
+    rel = (y_q - y_fp).abs().max() / (y_fp.abs().max() + 1e-9)
+    assert float(rel) < 0.06, float(rel)
