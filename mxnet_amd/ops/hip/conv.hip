@@ -503,6 +503,161 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_kernel(
 
 
 // ---------------------------------------------------------------------------
+// bwd-weight v3: x side consumed via ds_read_b64_tr_b16 hardware
+// transpose reads (guide T10) -- the two-hop LDS transpose (16 u16
+// reads + swizzled writes per 16 B) that made v2.5 instruction-bound
+// (13.9:1 VALU:MFMA) is deleted.  x tiles are stored ROW-major in a
+// [kk][t][cblk][16][16] subtile layout whose row placement is permuted
+// at write time (phys row ((k>>3)<<2)|(k&3), tile t=(k>>2)&1) so the
+// fixed tr delivery (lane l elem j = tile[(l>>4)*4+j][l&15]) lands the
+// MFMA k = (l>>4)*8 + j' order exactly.  dy side unchanged (linear
+// gload_lds of the pre-transposed [Kout,M]).
+// ---------------------------------------------------------------------------
+typedef short trs4 __attribute__((ext_vector_type(4)));
+__device__ inline trs4 tr_read16(const void* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) trs4*)(uintptr_t)p);
+}
+
+template <typename T>
+__global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_tr_kernel(
+    const T* __restrict__ dyT, const T* __restrict__ x,
+    const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
+    int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
+    int dw, long m_per_slice, const T* __restrict__ zpage) {
+  using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
+  constexpr int BI = 64, BJ = 64, BKM = 64;
+  __shared__ T DyT[2][BI * BKM];   // [i][m], linear (gload_lds dest)
+  __shared__ T XS[2][BKM * BJ];    // x subtiled [kk][t][cblk][16][16]
+
+  const int g = blockIdx.z;
+  const int cpl = (Cg + BJ - 1) / BJ;
+  const int nTj = R * S * cpl;
+  const int bid = blockIdx.x;
+  const int i0 = (bid / nTj) * BI;
+  const int jt = bid % nTj;
+  const int rs = jt / cpl;
+  const int c0 = (jt % cpl) * BJ;
+  const int r = rs / S, sst = rs % S;
+  const int roff = r * dh, soff = sst * dw;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int sm_half = t >> 3;
+  const int seg = t & 7;
+
+  float4_t acc[2][2] = {};
+
+  auto stage_dy = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int i_l = rnd * 32 + sm_half;
+      const long mcol = mc + seg * 8;
+      const bool ok = i0 + i_l < Kg && mcol + 8 <= M;
+      const T* ga = ok
+          ? dyT + ((long)g * Kg + i0 + i_l) * M + mcol
+          : zpage;
+      gload_lds16c(ga, &DyT[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  // gathered x rows -> permuted-row subtiles (tr-read source)
+  auto stage_x = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      const bool m_ok = m_g < ms1;
+      V8 v = {};
+      if (m_ok) {
+        int4_t pt = pixtab[m_g];
+        const int ih = pt[1] + roff, iw = pt[2] + soff;
+        const int cseg = c0 + seg * 8;
+        if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+            cseg + 8 <= Cg)
+          v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                           (long)g * Cg + cseg);
+      }
+      const int kk = m_l >> 5, rem = m_l & 31;
+      const int tt = (rem >> 2) & 1;
+      const int prow = ((rem >> 3) << 2) | (rem & 3);
+      *(V8*)&XS[buf][(((kk * 2 + tt) * 4 + (seg >> 1)) << 8) +
+                     prow * 16 + (seg & 1) * 8] = v;
+    }
+  };
+
+  const int a_row = lane & 15;
+  const int k_off = (lane >> 4) * 8;
+
+  stage_dy(0, ms0);
+  stage_x(0, ms0);
+  __syncthreads();
+  int buf = 0;
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+    if (mc + BKM < ms1) {
+      stage_x(buf ^ 1, mc + BKM);
+      stage_dy(buf ^ 1, mc + BKM);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[2], bf[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int i = wr * 32 + m * 16 + a_row;
+        af[m] = *(const Frag*)&DyT[buf][i * BKM + kk * 32 + k_off];
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const int cblk = wc * 2 + n;
+        union { trs4 h[2]; Frag f; } u;
+        u.h[0] = tr_read16(&XS[buf][(((kk * 2 + 0) * 4 + cblk) << 8) +
+                                    lane * 4]);
+        u.h[1] = tr_read16(&XS[buf][(((kk * 2 + 1) * 4 + cblk) << 8) +
+                                    lane * 4]);
+        bf[n] = u.f;
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const long RSCg = (long)R * S * Cg;
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    const long c = c0 + wc * 32 + n * 16 + d_col;
+    if (c >= Cg) continue;
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const long i = i_base + j;
+        if (i < Kg) {
+          float* dst = dw32 + ((long)g * Kg + i) * RSCg + (long)rs * Cg + c;
+          if (gridDim.y == 1) *dst = acc[m][n][j];
+          else atomicAdd(dst, acc[m][n][j]);
+        }
+      }
+    }
+  }
+}
+
+
+// ---------------------------------------------------------------------------
 // small-C implicit GEMM (the RGB stem): x padded to 8 channels so one
 // BK=64 chunk = 8 horizontal taps (r fixed, s = lane group, dil=1) --
 // taps are w-contiguous in NHWC so each lane's 16 B segment is one tap.
@@ -1347,8 +1502,31 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
                     dy.options().dtype(at::kFloat));
     dim3 grid((unsigned)nwg, (unsigned)yb, (unsigned)groups);
     // small-M panels: one cheap global dy transpose, then NT staging;
-    // big-M panels: the transpose itself would dominate -> two-hop
-    bool pretranspose = M * (long)Kout * 2 < (80L << 20);
+    // big-M panels: the transpose itself would dominate -> two-hop.
+    // With the tr-read x side the staging win is larger, so the
+    // crossover (MXNET_BWDW_PRET_MB, in MB of dy) is retuned on HW.
+    static const long pret_mb = [] {
+      const char* e = getenv("MXNET_BWDW_PRET_MB");
+      return e ? atol(e) : 80L;
+    }();
+    bool pretranspose = M * (long)Kout * 2 < (pret_mb << 20);
+    static const bool use_tr = [] {
+      const char* e = getenv("MXNET_BWDW_TR");
+      return !e || e[0] != '0';
+    }();
+    if (pretranspose && use_tr) {
+      auto dyT = transpose2d(dy.view({M, (long)Kout}));  // [Kout, M]
+      DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w_tr", [&] {
+        conv_bwd_w_igemm_tr_kernel<scalar_t>
+            <<<grid, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dyT.data_ptr(), (const scalar_t*)x.data_ptr(),
+            (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M, H, W,
+            C, Kout, Cg, Kg, R, S, dh, dw, m_per_slice,
+            (const scalar_t*)zero_page(dy));
+      });
+      HIP_CHECK_LAST();
+      return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
+    }
     if (pretranspose) {
       auto dyT = transpose2d(dy.view({M, (long)Kout}));  // [Kout, M]
       DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w", [&] {
