@@ -33,14 +33,30 @@ class BERTSelfAttention(HybridBlock):
         self.dropout = nn.Dropout(dropout) if dropout else None
 
     def forward(self, x, mask=None):
-        # x: [B, S, U]
+        # x: [B, S, U]; mask: [B, S] valid-token or prebuilt [B*H, S, S]
         from ..ops import nn as F
+        from ..ops.dispatch import use_hip
+        import math as _m
         import torch
         t = x.handle if hasattr(x, 'handle') else x
         B, S, U = t.shape
         H, D = self._num_heads, self._head_dim
         qkv = self.qkv(x)
         qt = qkv.handle if hasattr(qkv, 'handle') else qkv
+        m2 = None
+        if mask is not None:
+            m = mask.handle if hasattr(mask, 'handle') else mask
+            if m.dim() == 3:          # prebuilt [B*H, S, S] (model-level)
+                m2 = m
+            else:
+                m2 = m[:, None, None, :].expand(B, H, S, S) \
+                    .reshape(B * H, S, S)
+        if use_hip(qt) and D % 8 == 0 and S % 8 == 0 \
+                and self.dropout is None:
+            out = F.attention_core(
+                qt, m2.to(torch.uint8).contiguous() if m2 is not None
+                else None, H, _m.sqrt(D))
+            return self.proj(self._wrap(out, x))
         q, k, v = qt.split(U, dim=-1)
         # [B,S,H,D] -> [B*H, S, D]
         def heads(z):
@@ -49,11 +65,7 @@ class BERTSelfAttention(HybridBlock):
         q, k, v = heads(q), heads(k), heads(v)
         scores = F.batch_dot(q, k, transpose_b=True)  # [B*H, S, S]
         # 1/sqrt(D) folds into the softmax temperature (saves a pass)
-        if mask is not None:
-            m = mask.handle if hasattr(mask, 'handle') else mask
-            # mask: [B, S] valid-token mask -> broadcast [B*H, S, S]
-            m2 = m[:, None, None, :].expand(B, H, S, S) \
-                .reshape(B * H, S, S)
+        if m2 is not None:
             att = F.masked_softmax(scores, m2, axis=-1,
                                    temperature=math.sqrt(D))
         else:
@@ -136,6 +148,7 @@ class BERTModel(HybridBlock):
                  type_vocab_size=2, dropout=0.1, **kwargs):
         super().__init__(**kwargs)
         self._units = units
+        self._num_heads = num_heads
         self.word_embed = nn.Embedding(vocab_size, units,
                                        weight_initializer=init.Normal(0.02))
         self.token_type_embed = nn.Embedding(type_vocab_size, units,
@@ -172,6 +185,19 @@ class BERTModel(HybridBlock):
         emb = self.embed_ln(emb)
         if self.embed_dropout is not None:
             emb = self.embed_dropout(emb)
+        if valid_mask is not None:
+            # expand the [B, S] token mask to [B*H, S, S] ONCE here --
+            # every layer consumes the same tensor (the reference
+            # rebuilt it per layer inside the attention cell)
+            import torch as _th
+            m = valid_mask.handle if hasattr(valid_mask, 'handle') \
+                else valid_mask
+            if m.dim() == 2:
+                B, S = m.shape
+                H = self._num_heads
+                m2 = m[:, None, None, :].expand(B, H, S, S) \
+                    .reshape(B * H, S, S).to(_th.uint8).contiguous()
+                valid_mask = NDArray(m2)
         seq = self.encoder(emb, valid_mask)
         pooled = self.pooler(NDArray(seq.handle[:, 0]))
         mlm = self.mlm_decoder(self.mlm_ln(self.mlm_dense(seq)))

@@ -62,6 +62,11 @@ at::Tensor dequantize_i8(const at::Tensor&, double,
                          c10::optional<at::ScalarType>);
 at::Tensor im2col_nhwc_op(const at::Tensor&, int, int, int, int, int, int,
                           int, int);
+std::vector<at::Tensor> attention_fwd(const at::Tensor&,
+                                      c10::optional<at::Tensor>, int64_t,
+                                      double);
+at::Tensor attention_bwd(const at::Tensor&, const at::Tensor&,
+                         const at::Tensor&, int64_t, double);
 at::Tensor gemm_nt_i8(const at::Tensor&, const at::Tensor&, double,
                       c10::optional<at::ScalarType>);
 // elemwise.hip
@@ -126,6 +131,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_i8", &dequantize_i8, py::arg("x"), py::arg("scale"),
         py::arg("dtype") = py::none());
   m.def("im2col_nhwc", &im2col_nhwc_op);
+  m.def("attention_fwd", &attention_fwd, py::arg("qkv"),
+        py::arg("mask") = c10::nullopt, py::arg("heads"),
+        py::arg("temperature") = 1.0);
+  m.def("attention_bwd", &attention_bwd);
   m.def("gemm_nt_i8", &gemm_nt_i8, py::arg("a"), py::arg("b"),
         py::arg("scale"), py::arg("out_dtype") = py::none());
 }

@@ -40,12 +40,22 @@ DEV_INLINE void gload_lds16(const void* g, void* lds) {
 // split-K: when out32 != nullptr, blockIdx.z owns k-tiles
 // [z*tiles_per_slice, ...) and accumulates fp32 partials with atomics
 // (small-M*N huge-K problems, e.g. the conv-stem weight gradient).
+// row strides + 2-level batch (outer x heads) so strided views (e.g.
+// attention Q/K/V slices of a fused [B,S,3U] projection) run with no
+// contiguous() copy.  Plain GEMM passes {K, K, N, 1, 0, 0, 0}.
+struct GemmLd {
+  long lda, ldb, ldc;   // row strides (elements); K-contiguous always
+  int bh;               // inner batch extent (heads); 1 = plain
+  long sAh, sBh, sCh;   // inner-batch element strides
+};
+
 template <typename T, int BN = 128>
 __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
     long strideA, long strideB, long strideC, const T* __restrict__ zpage,
-    bool relu, float* __restrict__ out32, int tiles_per_slice, int nbuf) {
+    bool relu, float* __restrict__ out32, int tiles_per_slice, int nbuf,
+    const GemmLd ld) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BK = 64;
   constexpr int NW = BN / 32;  // n-fragments per wave
@@ -56,9 +66,10 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   T* Bs = As + (long)nbuf * BM * BK;  // [nbuf][BN*BK]
 
   const long batch = blockIdx.y;
-  A += batch * strideA;
-  B += batch * strideB;
-  C += batch * strideC;
+  const long bo = batch / ld.bh, bi = batch % ld.bh;
+  A += bo * strideA + bi * ld.sAh;
+  B += bo * strideB + bi * ld.sBh;
+  C += bo * strideC + bi * ld.sCh;
 
   const int nTn = (N + BN - 1) / BN;
   const int nwg = ((M + BM - 1) / BM) * nTn;
@@ -86,13 +97,13 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
       const long row_a = m0 + r * 32 + s_row;
-      const T* ga = (row_a < M && ka_ok) ? A + row_a * K + kcol : zpage;
+      const T* ga = (row_a < M && ka_ok) ? A + row_a * ld.lda + kcol : zpage;
       gload_lds16(ga, As + buf * (BM * BK) + (r * 256 + t) * 8);
     }
 #pragma unroll
     for (int r = 0; r < BN / 32; ++r) {
       const long row_b = n0 + r * 32 + s_row;
-      const T* gb = (row_b < N && ka_ok) ? B + row_b * K + kcol : zpage;
+      const T* gb = (row_b < N && ka_ok) ? B + row_b * ld.ldb + kcol : zpage;
       gload_lds16(gb, Bs + buf * (BN * BK) + (r * 256 + t) * 8);
     }
   };
@@ -185,11 +196,11 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const long row = m0 + rowL;
     const long col = n0 + colL;
     if (row >= M) continue;
-    if (col + 8 <= N && (N % 8) == 0) {   // 16 B-aligned fast path
-      *(V8e*)(C + row * N + col) = *(const V8e*)&tile[rowL * BN + colL];
+    if (col + 8 <= N && (N % 8) == 0 && (ld.ldc % 8) == 0) {  // 16 B fast
+      *(V8e*)(C + row * ld.ldc + col) = *(const V8e*)&tile[rowL * BN + colL];
     } else {
       for (int j = 0; j < 8 && col + j < N; ++j)
-        C[row * N + col + j] = tile[rowL * BN + colL + j];
+        C[row * ld.ldc + col + j] = tile[rowL * BN + colL + j];
     }
   }
 }
@@ -437,6 +448,33 @@ __global__ void transpose_kernel(const T* __restrict__ in, T* __restrict__ out,
   }
 }
 
+// strided variant: per (outer, inner) batch reads in[r*ldin + c] from
+// base + outer*sOut + inner*sIn, writes contiguous out[batch][C][R]
+// (attention builds V^T / Q^T / K^T panels straight from the fused
+// [B,S,3U] projection with this -- one pass, no permute+contiguous).
+template <typename T>
+__global__ void transpose_strided_kernel(const T* __restrict__ in,
+                                         T* __restrict__ out, long R, long C,
+                                         long ldin, int bh, long sOut,
+                                         long sIn) {
+  __shared__ T tile[32][33];
+  const long batch = blockIdx.z;
+  in += (batch / bh) * sOut + (batch % bh) * sIn;
+  out += batch * R * C;
+  long c0 = (long)blockIdx.x * 32;
+  long r0 = (long)blockIdx.y * 32;
+  int tx = threadIdx.x, ty = threadIdx.y;
+  for (int i = 0; i < 32; i += 8) {
+    long r = r0 + ty + i, c = c0 + tx;
+    if (r < R && c < C) tile[ty + i][tx] = in[r * ldin + c];
+  }
+  __syncthreads();
+  for (int i = 0; i < 32; i += 8) {
+    long c = c0 + ty + i, r = r0 + tx;
+    if (c < C && r < R) out[c * R + r] = tile[tx][ty + i];
+  }
+}
+
 // ---------------------------------------------------------------------------
 // host side
 // ---------------------------------------------------------------------------
@@ -562,14 +600,16 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
               (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
               bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
               (const scalar_t*)zero_page(A), relu,
-              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
+              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
+              GemmLd{K, K, N, 1, 0, 0, 0});
     else
      hipLaunchKernelGGL(( gemm_nt_mfma_kernel<scalar_t, 128>)
           , dim3(grid), dim3(256), lds_bytes, cur_stream(), 
               (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
               bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
               (const scalar_t*)zero_page(A), relu,
-              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf);
+              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
+              GemmLd{K, K, N, 1, 0, 0, 0});
   });
   HIP_CHECK_LAST();
   if (ksplit > 1) {
@@ -635,4 +675,131 @@ at::Tensor gemm_nt_8ph(const at::Tensor& A_, const at::Tensor& B_) {
   });
   HIP_CHECK_LAST();
   return out;
+}
+
+
+// ---------------------------------------------------------------------------
+// fused multi-head attention core on the strided NT GEMM
+// (reference transformer attention ran separate transpose/reshape +
+// batch_dot ops, src/operator/contrib/transformer.cc interleaved path;
+// here the Q/K/V panels are consumed as strided views of the fused
+// [B, S, 3U] projection -- zero permute/contiguous copies).
+// ---------------------------------------------------------------------------
+
+// NT GEMM over a (outerB x H) batch of strided panels.
+static void launch_nt_strided(const at::Tensor& like, const void* A,
+                              const void* B, void* C, long M, long N, long K,
+                              long sAb, long sBb, long sCb, const GemmLd& ld,
+                              long nb) {
+  TORCH_CHECK(K % 8 == 0 && ld.lda % 8 == 0 && ld.ldb % 8 == 0,
+              "attention: K and row strides must be 8-element aligned");
+  bool narrow = N <= 64;
+  long nwg = narrow ? (long)((M + 127) / 128) * ((N + 63) / 64)
+                    : (long)((M + 127) / 128) * ((N + 127) / 128);
+  int nk = (int)((K + 63) / 64);
+  int nbuf = nk > 1 ? 2 : 1;
+  size_t lds_bytes = (size_t)nbuf * (128 + (narrow ? 64 : 128)) * 64 * 2;
+  dim3 grid((unsigned)nwg, (unsigned)nb);
+  DISPATCH_HALF_TYPES(like.scalar_type(), "attn_nt", [&] {
+    if (narrow)
+     hipLaunchKernelGGL(( gemm_nt_mfma_kernel<scalar_t, 64>), dim3(grid), dim3(256), lds_bytes,
+                                          cur_stream(), 
+          (const scalar_t*)A, (const scalar_t*)B, nullptr, (scalar_t*)C, M,
+          N, K, sAb, sBb, sCb, (const scalar_t*)zero_page(like), false,
+          nullptr, 0, nbuf, ld);
+    else
+     hipLaunchKernelGGL(( gemm_nt_mfma_kernel<scalar_t, 128>), dim3(grid), dim3(256), lds_bytes,
+                                           cur_stream(), 
+          (const scalar_t*)A, (const scalar_t*)B, nullptr, (scalar_t*)C, M,
+          N, K, sAb, sBb, sCb, (const scalar_t*)zero_page(like), false,
+          nullptr, 0, nbuf, ld);
+  });
+  HIP_CHECK_LAST();
+}
+
+// strided transpose: [R, C] panels at base + b*sOut + h*sIn (row stride
+// ldin) -> contiguous [nb, C, R]
+static at::Tensor transpose_strided(const at::Tensor& src, const void* base,
+                                    long R, long C, long ldin, int bh,
+                                    long sOut, long sIn, long nb) {
+  auto out = at::empty({nb, C, R}, src.options());
+  dim3 grid((unsigned)((C + 31) / 32), (unsigned)((R + 31) / 32),
+            (unsigned)nb);
+  DISPATCH_HALF_TYPES(src.scalar_type(), "transpose_strided", [&] {
+   hipLaunchKernelGGL(( transpose_strided_kernel<scalar_t>), dim3(grid), dim3(dim3(32, 8)), 0,
+                                         cur_stream(), 
+        (const scalar_t*)base, (scalar_t*)out.data_ptr(), R, C, ldin, bh,
+        sOut, sIn);
+  });
+  HIP_CHECK_LAST();
+  return out;
+}
+
+// qkv: [B, S, 3U] (U = H*D); mask: byte [B*H, S, S] or undefined.
+// returns {out [B, S, U], att [B*H, S, S]} -- att saved for backward.
+std::vector<at::Tensor> attention_fwd(const at::Tensor& qkv,
+                                      c10::optional<at::Tensor> mask,
+                                      int64_t H, double temperature) {
+  CHECK_GPU(qkv); CHECK_CONTIG(qkv);
+  long B = qkv.size(0), S = qkv.size(1), U3 = qkv.size(2);
+  long U = U3 / 3, D = U / H, BH = B * H;
+  TORCH_CHECK(D % 8 == 0 && S % 8 == 0, "attention: D, S must be %8");
+  auto T_ = qkv.scalar_type();
+  auto att_raw = at::empty({BH, S, S}, qkv.options());
+  const char* qp = (const char*)qkv.data_ptr();
+  long es = qkv.element_size();
+  // scores = Q K^T: A = q panel, B = k panel (both strided in qkv)
+  launch_nt_strided(qkv, qp, qp + U * es, att_raw.data_ptr(), S, S, D,
+                    S * U3, S * U3, (long)H * S * S,
+                    GemmLd{U3, U3, S, (int)H, D, D, S * S},
+                    BH);
+  auto att = softmax_fwd(att_raw, false, temperature, mask);
+  // V^T panels: [BH, D, S] contiguous
+  auto vt = transpose_strided(qkv, qp + 2 * U * es, S, D, U3, (int)H,
+                              S * U3, D, BH);
+  auto out = at::empty({B, S, U}, qkv.options());
+  // out = att @ V = NT(att, V^T), C written strided into [B, S, U]
+  launch_nt_strided(qkv, att.data_ptr(), vt.data_ptr(), out.data_ptr(), S,
+                    D, S, (long)H * S * S /*per-outer att*/, (long)H * D * S,
+                    S * U, GemmLd{S, S, U, (int)H, S * S, D * S, D}, BH);
+  return {out, att};
+}
+
+// dout: [B, S, U]; returns dqkv [B, S, 3U]
+at::Tensor attention_bwd(const at::Tensor& dout, const at::Tensor& qkv,
+                         const at::Tensor& att, int64_t H,
+                         double temperature) {
+  CHECK_GPU(dout); CHECK_CONTIG(dout); CHECK_CONTIG(qkv); CHECK_CONTIG(att);
+  long B = qkv.size(0), S = qkv.size(1), U3 = qkv.size(2);
+  long U = U3 / 3, D = U / H, BH = B * H;
+  const char* qp = (const char*)qkv.data_ptr();
+  const char* dp = (const char*)dout.data_ptr();
+  long es = qkv.element_size();
+  // datt = dOut V^T-strided NT: A = dout panel, B = v panel
+  auto datt = at::empty({BH, S, S}, qkv.options());
+  launch_nt_strided(qkv, dp, qp + 2 * U * es, datt.data_ptr(), S, S, D,
+                    S * U, S * U3, (long)H * S * S,
+                    GemmLd{U, U3, S, (int)H, D, D, S * S}, BH);
+  auto ds = softmax_bwd(datt, att, false, temperature);
+  auto dqkv = at::empty_like(qkv);
+  char* dq = (char*)dqkv.data_ptr();
+  // dQ = ds K: B-operand = K^T panels
+  auto kt = transpose_strided(qkv, qp + U * es, S, D, U3, (int)H, S * U3, D,
+                              BH);
+  launch_nt_strided(qkv, ds.data_ptr(), kt.data_ptr(), dq, S, D, S,
+                    (long)H * S * S, (long)H * D * S, S * U3,
+                    GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
+  // dK = ds^T Q: A = ds^T (batched transpose), B = Q^T panels
+  auto dsT = transpose2d(ds);
+  auto qt = transpose_strided(qkv, qp, S, D, U3, (int)H, S * U3, D, BH);
+  launch_nt_strided(qkv, dsT.data_ptr(), qt.data_ptr(), dq + U * es, S, D,
+                    S, (long)H * S * S, (long)H * D * S, S * U3,
+                    GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
+  // dV = att^T dOut: A = att^T, B = dOut^T panels
+  auto attT = transpose2d(att);
+  auto dot_t = transpose_strided(dout, dp, S, D, U, (int)H, S * U, D, BH);
+  launch_nt_strided(qkv, attT.data_ptr(), dot_t.data_ptr(), dq + 2 * U * es,
+                    S, D, S, (long)H * S * S, (long)H * D * S, S * U3,
+                    GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
+  return dqkv;
 }

@@ -60,6 +60,10 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
                         c10::optional<at::Tensor> bias, bool relu);
 at::Tensor transpose2d(const at::Tensor& x);
 const void* zero_page(const at::Tensor& like);
+at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode, double temperature,
+                       c10::optional<at::Tensor> mask);
+at::Tensor softmax_bwd(const at::Tensor& dy, const at::Tensor& y,
+                       bool log_mode, double temperature);
 
 constexpr int kEwBlock = 256;
 // memory-bound launch cap: ~8 blocks/CU on 256 CUs (Guideline 11)

@@ -469,6 +469,38 @@ def masked_softmax(x, mask, axis=-1, temperature=1.0):
     return softmax(x, axis, temperature)
 
 
+class _AttentionCore(torch.autograd.Function):
+    """Fused multi-head attention on the strided NT MFMA GEMM: Q/K/V are
+    consumed as strided views of the fused [B,S,3U] projection -- no
+    permute/contiguous head reshapes (reference transformer.cc
+    interleaved_matmul_selfatt_* ops serve the same purpose)."""
+
+    @staticmethod
+    def forward(ctx, qkv, mask, heads, temperature):
+        ext = hip_required('attention')
+        out, att = ext.attention_fwd(qkv, mask, heads, temperature)
+        ctx.save_for_backward(qkv, att)
+        ctx.ht = (heads, temperature)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, att = ctx.saved_tensors
+        heads, temp = ctx.ht
+        ext = hip_required('attention')
+        dqkv = ext.attention_bwd(dout.contiguous(), qkv, att, heads, temp)
+        return dqkv, None, None, None
+
+
+def attention_core(qkv, mask, heads, temperature=1.0):
+    """qkv [B,S,3U] (U=heads*D, q|k|v along last dim, heads split D-wise)
+    -> [B,S,U].  mask: uint8 [B*heads,S,S] or None.  GPU-only fused path;
+    callers fall back to explicit bgemm composition on CPU."""
+    return _AttentionCore.apply(qkv.contiguous(),
+                                mask.contiguous() if mask is not None
+                                else None, heads, temperature)
+
+
 def softmax_cross_entropy(logits, labels, sparse=True):
     """Fused softmax+CE (loss path); logits [N, C]."""
     lsm = log_softmax(logits, axis=-1)

@@ -605,3 +605,46 @@ def test_quantized_conv_gpu():
     assert y_q.shape == y_fp.shape
     rel = (y_q - y_fp).abs().max() / (y_fp.abs().max() + 1e-9)
     assert float(rel) < 0.06, float(rel)
+
+
+# ---------------------------------------------------------------------------
+# fused attention core (strided NT GEMM, no head reshapes)
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('masked', [False, True])
+def test_attention_core(masked):
+    from mxnet_amd.ops import nn as onn
+    torch.manual_seed(33)
+    B, S, H, D = 2, 64, 4, 16
+    U = H * D
+    temp = float(D) ** 0.5
+    qkv = torch.randn(B, S, 3 * U, device='cuda', dtype=torch.float16) \
+        .requires_grad_(True)
+    if masked:
+        valid = torch.ones(B, S, device='cuda', dtype=torch.uint8)
+        valid[:, S // 2:] = 0
+        m2 = valid[:, None, None, :].expand(B, H, S, S) \
+            .reshape(B * H, S, S).contiguous()
+    else:
+        m2 = None
+    out = onn.attention_core(qkv, m2, H, temp)
+    # fp32 oracle: explicit head-split composition on CPU
+    qkv32 = qkv.detach().float().cpu().requires_grad_(True)
+    q, k, v = qkv32.split(U, dim=-1)
+    def heads(z):
+        return z.reshape(B, S, H, D).permute(0, 2, 1, 3) \
+                .reshape(B * H, S, D)
+    qh, kh, vh = heads(q), heads(k), heads(v)
+    sc = qh @ kh.transpose(1, 2) / temp
+    if masked:
+        sc = sc.masked_fill(m2.cpu() == 0, float('-inf'))
+    att = torch.softmax(sc, dim=-1)
+    if masked:
+        att = torch.nan_to_num(att)
+    ref = (att @ vh).reshape(B, H, S, D).permute(0, 2, 1, 3) \
+        .reshape(B, S, U)
+    check(out, ref.to(out.device), tol=3e-3)
+    # backward parity
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+    out.backward(dy.to(out.device).to(out.dtype))
+    check(qkv.grad, qkv32.grad.to(out.device), tol=6e-3)
