@@ -23,6 +23,10 @@ from .context import Context, cpu, gpu, cpu_pinned, current_context, num_gpus
 from .base import MXNetError
 from . import ndarray
 from . import ndarray as nd
+from . import operator
+from . import rtc
+from . import image
+from . import image as img
 from . import numpy as np  # mx.np numpy-compatible namespace
 from . import symbol
 from . import symbol as sym

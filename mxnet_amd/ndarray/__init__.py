@@ -8,3 +8,9 @@ from .ndarray import concat, stack  # keep creation-module versions authoritativ
 
 from . import contrib  # noqa: F401
 from . import sparse  # noqa: F401
+
+
+def Custom(*inputs, op_type=None, **kwargs):
+    """Run a registered python CustomOp (reference nd.Custom)."""
+    from ..operator import invoke
+    return invoke(op_type, *inputs, **kwargs)

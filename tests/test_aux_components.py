@@ -138,3 +138,66 @@ def test_control_flow_gradients():
     loss.backward()
     np.testing.assert_allclose(x.grad.asnumpy(), np.full((4, 3), 2.0),
                                rtol=1e-6)
+
+
+def test_custom_operator():
+    """mx.operator.CustomOp with autograd (reference operator.py custom op
+    bridge, src/operator/custom/custom.cc)."""
+    import torch
+    import mxnet_amd as mx
+    from mxnet_amd import operator as op, autograd
+
+    @op.register("sq_plus_one")
+    class P(op.CustomOpProp):
+        def create_operator(self, ctx, shapes, dtypes):
+            class O(op.CustomOp):
+                def forward(self, is_train, req, in_data, out_data, aux):
+                    self.assign(out_data[0], req[0],
+                                in_data[0].handle ** 2 + 1)
+
+                def backward(self, req, out_grad, in_data, out_data,
+                             in_grad, aux):
+                    self.assign(in_grad[0], req[0],
+                                2 * in_data[0].handle * out_grad[0].handle)
+            return O()
+
+    x = mx.nd.from_torch(torch.randn(3, 4))
+    x.attach_grad()
+    with autograd.record():
+        y = mx.nd.Custom(x, op_type="sq_plus_one")
+        L = mx.nd.from_torch(y.handle.sum())
+    L.backward()
+    assert torch.allclose(y.handle, x.handle.detach() ** 2 + 1)
+    assert torch.allclose(x.grad.handle, 2 * x.handle.detach())
+
+
+def test_rtc_compiles():
+    """mx.rtc.HipModule hiprtc compilation (launch covered by the GPU
+    test)."""
+    import mxnet_amd as mx
+    mod = mx.rtc.HipModule(
+        'extern "C" __global__ void mul2(float* y, int n)'
+        '{ int i = blockIdx.x * blockDim.x + threadIdx.x;'
+        '  if (i < n) y[i] *= 2.f; }')
+    k = mod.get_kernel('mul2', 'float* y, int n')
+    assert k._types == ['ptr', 'i32']
+
+
+def test_image_transforms():
+    """mx.image tensor-domain transforms (reference image/image.py)."""
+    import torch
+    import mxnet_amd as mx
+    src = mx.nd.from_torch(torch.randint(0, 255, (20, 30, 3),
+                                         dtype=torch.uint8))
+    r = mx.image.imresize(src, 15, 10)
+    assert r.shape == (10, 15, 3) and r.handle.dtype == torch.uint8
+    s = mx.image.resize_short(src, 10)
+    assert min(s.shape[:2]) == 10
+    c, rect = mx.image.center_crop(src, (12, 8))
+    assert c.shape == (8, 12, 3)
+    n = mx.image.color_normalize(mx.nd.from_torch(torch.ones(4, 4, 3)),
+                                 mean=(0.5, 0.5, 0.5), std=(0.5, 0.5, 0.5))
+    assert float(n.handle.mean()) == 1.0
+    import pytest as _pt
+    with _pt.raises(NotImplementedError):
+        mx.image.imread('x.jpg')

@@ -648,3 +648,20 @@ def test_attention_core(masked):
     ref.backward(dy)
     out.backward(dy.to(out.device).to(out.dtype))
     check(qkv.grad, qkv32.grad.to(out.device), tol=6e-3)
+
+
+def test_rtc_launch_gpu():
+    """mx.rtc kernel launch on the MI355X via hipModuleLaunchKernel."""
+    import mxnet_amd as mx
+    mod = mx.rtc.HipModule(
+        'extern "C" __global__ void axpy(const float* x, float* y,'
+        ' float a, int n) { int i = blockIdx.x * blockDim.x + threadIdx.x;'
+        ' if (i < n) y[i] += a * x[i]; }')
+    k = mod.get_kernel('axpy', 'const float *x, float *y, float a, int n')
+    x = torch.randn(1000, device='cuda')
+    y = torch.randn(1000, device='cuda')
+    y0 = y.clone()
+    k.launch((x, y, 3.0, 1000), mx.gpu(0), ((1000 + 255) // 256, 1, 1),
+             (256, 1, 1))
+    torch.cuda.synchronize()
+    assert torch.allclose(y, y0 + 3.0 * x, atol=1e-5)
