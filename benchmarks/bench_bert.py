@@ -78,13 +78,36 @@ def main():
         import torch.distributed as dist
     for _ in range(args.warmup):
         step()
+
+    # whole-step hipGraph capture (same pattern as bench.py)
+    graph = None
+    if on_gpu and world == 1 and \
+            os.environ.get('MXNET_BENCH_HIPGRAPH', '1') != '0':
+        try:
+            torch.cuda.synchronize()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                step()
+            g.replay()
+            torch.cuda.synchronize()
+            graph = g
+        except Exception as e:
+            print('# hipgraph capture unavailable:', e)
+            graph = None
+
     if on_gpu:
         torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        graph.replay() if graph is not None else step()
     if on_gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0

@@ -71,11 +71,34 @@ def main():
 
     for _ in range(args.warmup):
         step()
+
+    # whole-step hipGraph capture (same pattern as bench.py): the LSTM
+    # step is launch-bound, so replay removes the per-kernel gaps.
+    graph = None
+    if on_gpu and os.environ.get('MXNET_BENCH_HIPGRAPH', '1') != '0':
+        try:
+            torch.cuda.synchronize()
+            side = torch.cuda.Stream()
+            side.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(side):
+                step()
+            torch.cuda.current_stream().wait_stream(side)
+            torch.cuda.synchronize()
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                step()
+            g.replay()
+            torch.cuda.synchronize()
+            graph = g
+        except Exception as e:
+            print('# hipgraph capture unavailable:', e)
+            graph = None
+
     if on_gpu:
         torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        graph.replay() if graph is not None else step()
     if on_gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0

@@ -552,9 +552,11 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   int nk_total = (int)((K + 63) / 64);
   int ksplit = 1, tps = nk_total;
   at::Tensor out32;
-  if (nwg * nb < 512 && nk_total > 16) {
+  // small grids: split the reduction until the chip fills (256 CUs want
+  // >=512 workgroups; recurrent-GEMM shapes like M=128,K=1024 give 32)
+  if (nwg * nb < 512 && nk_total >= 8) {
     ksplit = (int)std::min<long>((2048 + nwg * nb - 1) / (nwg * nb),
-                                 (nk_total + 15) / 16);
+                                 (nk_total + 3) / 4);
     tps = (nk_total + ksplit - 1) / ksplit;
     ksplit = (nk_total + tps - 1) / tps;
     out32 = at::zeros(out.sizes(), out.options().dtype(at::kFloat));
