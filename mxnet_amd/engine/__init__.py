@@ -97,3 +97,38 @@ def engine_type():
     if _cpp is not None:
         return os.environ.get('MXNET_ENGINE_TYPE', 'ThreadedEnginePerDevice')
     return 'NaiveEngine'
+
+
+# -- fork safety (reference LibraryInitializer pthread_atfork handlers,
+# src/initialize.cc:71-83: stop engine threads around fork so DataLoader
+# worker processes never inherit held queue mutexes) -------------------
+def _before_fork():
+    try:
+        if _cpp is not None:
+            eng = _cpp.get()
+            eng.wait_for_all()
+            eng.stop()
+    except Exception:
+        pass
+
+
+def _after_fork_parent():
+    try:
+        if _cpp is not None:
+            _cpp.get().start()
+    except Exception:
+        pass
+
+
+def _after_fork_child():
+    # child gets a fresh worker pool; inherited thread state is gone
+    try:
+        if _cpp is not None:
+            _cpp.get().start()
+    except Exception:
+        pass
+
+
+os.register_at_fork(before=_before_fork,
+                    after_in_parent=_after_fork_parent,
+                    after_in_child=_after_fork_child)

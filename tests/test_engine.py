@@ -112,3 +112,35 @@ def test_stress_many_vars(eng):
             eng.push(bump, (vs[(i + 1) % 64],), (v,))
     eng.wait_for_all()
     assert counters == [20] * 64
+
+
+def test_engine_fork_safety():
+    """Engine worker pool survives fork (reference initialize.cc:71-83
+    pthread_atfork stop/restart): child gets a working engine, parent
+    keeps processing."""
+    import multiprocessing as mp
+    from mxnet_amd import engine
+    eng = engine.get()
+    v = eng.new_variable()
+    res = []
+    eng.push(lambda: res.append(1), mutable_vars=(v,))
+    eng.wait_for_all()
+
+    def child(q):
+        e = engine.get()
+        w = e.new_variable()
+        out = []
+        e.push(lambda: out.append(42), mutable_vars=(w,))
+        e.wait_for_all()
+        q.put(out[0])
+
+    ctx = mp.get_context('fork')
+    q = ctx.Queue()
+    p = ctx.Process(target=child, args=(q,))
+    p.start()
+    p.join(timeout=60)
+    assert p.exitcode == 0
+    assert q.get(timeout=10) == 42
+    eng.push(lambda: res.append(2), mutable_vars=(v,))
+    eng.wait_for_all()
+    assert res == [1, 2]
