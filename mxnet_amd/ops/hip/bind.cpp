@@ -29,7 +29,8 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor&,
                                           const at::Tensor&,
                                           const at::Tensor&, at::Tensor,
                                           at::Tensor, double, double, bool,
-                                          const at::Tensor&);
+                                          const at::Tensor&,
+                                          c10::optional<at::Tensor>);
 at::Tensor bn_nhwc_fwd_infer(const at::Tensor&, const at::Tensor&,
                              const at::Tensor&, const at::Tensor&,
                              const at::Tensor&, double, bool,
@@ -62,6 +63,9 @@ at::Tensor dequantize_i8(const at::Tensor&, double,
                          c10::optional<at::ScalarType>);
 at::Tensor im2col_nhwc_op(const at::Tensor&, int, int, int, int, int, int,
                           int, int);
+std::vector<at::Tensor> conv2d_nhwc_fwd_stats(
+    const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>, int64_t,
+    int64_t, int64_t, int64_t, int64_t, int64_t, int64_t);
 std::vector<at::Tensor> attention_fwd(const at::Tensor&,
                                       c10::optional<at::Tensor>, int64_t,
                                       double);
@@ -103,7 +107,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("conv2d_nhwc_fwd", &conv2d_nhwc_fwd);
   m.def("conv2d_nhwc_bwd_data", &conv2d_nhwc_bwd_data);
   m.def("conv2d_nhwc_bwd_weight", &conv2d_nhwc_bwd_weight);
-  m.def("bn_nhwc_fwd_train", &bn_nhwc_fwd_train);
+  m.def("bn_nhwc_fwd_train", &bn_nhwc_fwd_train, py::arg("x"),
+        py::arg("gamma"), py::arg("beta"), py::arg("rmean"),
+        py::arg("rvar"), py::arg("momentum"), py::arg("eps"),
+        py::arg("fuse_relu"), py::arg("residual"),
+        py::arg("presums") = c10::nullopt);
   m.def("bn_nhwc_fwd_infer", &bn_nhwc_fwd_infer);
   m.def("bn_nhwc_bwd", &bn_nhwc_bwd);
   m.def("layernorm_fwd", &layernorm_fwd);
@@ -131,6 +139,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequantize_i8", &dequantize_i8, py::arg("x"), py::arg("scale"),
         py::arg("dtype") = py::none());
   m.def("im2col_nhwc", &im2col_nhwc_op);
+  m.def("conv2d_nhwc_fwd_stats", &conv2d_nhwc_fwd_stats);
   m.def("attention_fwd", &attention_fwd, py::arg("qkv"),
         py::arg("mask") = c10::nullopt, py::arg("heads"),
         py::arg("temperature") = 1.0);

@@ -97,7 +97,7 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
     const float* __restrict__ bias, T* __restrict__ y, int NB, int H, int W,
     int C, int Kout, int Cg, int Kg, int P, int Q, int R, int S, int sh,
     int sw, int ph, int pw, int dh, int dw, const T* __restrict__ zpage,
-    bool relu) {
+    bool relu, float* __restrict__ stats = nullptr) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BN = 128, BK = 64;
   __shared__ T As[2][BM * BK];
@@ -195,6 +195,40 @@ __global__ __launch_bounds__(256, 2) void conv_fwd_igemm_kernel(
 
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
+  // fused BN-forward reduction over the raw accumulators (see gemm.hip):
+  // [64 slices][2][Kout] fp32, slice by block id
+  __shared__ float s_st[2][128];
+  if (stats) {
+    for (int i = t; i < 256; i += 256) s_st[i >> 7][i & 127] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      const int colL = wc * 64 + n * 16 + d_col;
+      if (n0 + colL >= Kg) continue;
+      float ps = 0.f, pq = 0.f;
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        const long row_base = m0 + wr * 64 + m * 16 + d_row;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (row_base + j < M) {
+            float v = acc[m][n][j];
+            ps += v;
+            pq += v * v;
+          }
+      }
+      atomicAdd(&s_st[0][colL], ps);
+      atomicAdd(&s_st[1][colL], pq);
+    }
+    __syncthreads();
+    float* slice = stats + (long)(bid & 63) * 2 * Kout;
+    const long cbase = (long)g * Kg + n0;
+    for (int i = t; i < 128 && n0 + i < Kg; i += 256) {
+      atomicAdd(slice + cbase + i, s_st[0][i]);
+      atomicAdd(slice + Kout + cbase + i, s_st[1][i]);
+    }
+    __syncthreads();
+  }
 #pragma unroll
   for (int n = 0; n < 4; ++n) {
     const long col_l = n0 + wc * 64 + n * 16 + d_col;
@@ -1232,10 +1266,10 @@ static at::Tensor im2col_nhwc(const at::Tensor& x, int P, int Q, int R, int S,
   return col;
 }
 
-at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
-                           c10::optional<at::Tensor> bias, int64_t sh,
-                           int64_t sw, int64_t ph, int64_t pw, int64_t dh,
-                           int64_t dw, int64_t groups) {
+static at::Tensor conv2d_nhwc_fwd_impl(
+    const at::Tensor& x, const at::Tensor& w, c10::optional<at::Tensor> bias,
+    int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+    int64_t groups, at::Tensor* stats_out) {
   CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(w);
   int NB = x.size(0), H = x.size(1), W = x.size(2), C = x.size(3);
   int Kout = w.size(0), R = w.size(1), S = w.size(2);
@@ -1273,18 +1307,24 @@ at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
     auto y2 = gemm_nt_core(x.view({M, C}), w.view({Kout, C}),
                            bias ? c10::optional<at::Tensor>(b32)
                                 : c10::nullopt,
-                           false);
+                           false, stats_out);
     return y2.view({NB, P, Q, Kout});
   }
   if (mfma_ok) {
     int nwg = (int)(((M + 127) / 128) * ((Kg + 127) / 128));
     dim3 grid((unsigned)nwg, (unsigned)groups);
+    float* stats_ptr = nullptr;
+    if (stats_out) {
+      *stats_out = at::zeros({64, 2, (long)Kout},
+                             x.options().dtype(at::kFloat));
+      stats_ptr = stats_out->data_ptr<float>();
+    }
     DISPATCH_HALF_TYPES(x.scalar_type(), "conv_fwd", [&] {
       conv_fwd_igemm_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
           (const scalar_t*)x.data_ptr(), (const scalar_t*)w.data_ptr(),
           bias_ptr, (scalar_t*)y.data_ptr(), NB, H, W, C, Kout, Cg, Kg, P, Q,
           R, S, sh, sw, ph, pw, dh, dw, (const scalar_t*)zero_page(x),
-          false);
+          false, stats_ptr);
     });
     HIP_CHECK_LAST();
     return y;
@@ -1314,6 +1354,29 @@ at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
                                   : c10::nullopt,
                          false);
   return y2.view({NB, P, Q, Kout});
+}
+
+at::Tensor conv2d_nhwc_fwd(const at::Tensor& x, const at::Tensor& w,
+                           c10::optional<at::Tensor> bias, int64_t sh,
+                           int64_t sw, int64_t ph, int64_t pw, int64_t dh,
+                           int64_t dw, int64_t groups) {
+  return conv2d_nhwc_fwd_impl(x, w, bias, sh, sw, ph, pw, dh, dw, groups,
+                              nullptr);
+}
+
+// forward + fused BN-forward per-channel {sum, ssq} ([64,2,Kout] fp32
+// partial slices; empty tensor when the executing path has no fused
+// reduction -- caller falls back to the standalone BN reduce).
+std::vector<at::Tensor> conv2d_nhwc_fwd_stats(
+    const at::Tensor& x, const at::Tensor& w, c10::optional<at::Tensor> bias,
+    int64_t sh, int64_t sw, int64_t ph, int64_t pw, int64_t dh, int64_t dw,
+    int64_t groups) {
+  at::Tensor stats;
+  auto y = conv2d_nhwc_fwd_impl(x, w, bias, sh, sw, ph, pw, dh, dw, groups,
+                                &stats);
+  if (!stats.defined())
+    stats = at::empty({0}, x.options().dtype(at::kFloat));
+  return {y, stats};
 }
 
 at::Tensor conv2d_nhwc_bwd_data(const at::Tensor& dy, const at::Tensor& w,

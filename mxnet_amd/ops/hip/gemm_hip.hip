@@ -55,7 +55,7 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
     long strideA, long strideB, long strideC, const T* __restrict__ zpage,
     bool relu, float* __restrict__ out32, int tiles_per_slice, int nbuf,
-    const GemmLd ld) {
+    const GemmLd ld, float* __restrict__ stats = nullptr) {
   using Frag = typename DTraits<T>::frag8;
   constexpr int BM = 128, BK = 64;
   constexpr int NW = BN / 32;  // n-fragments per wave
@@ -170,6 +170,40 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     return;
   }
   __syncthreads();                       // done with the K-loop buffers
+  // optional fused per-channel sum/ssq of the raw fp32 accumulators
+  // (BatchNorm's forward reduction -- saves re-reading y; layout
+  // [64 slices][2][N], slice = bid&63 to spread the atomic traffic)
+  __shared__ float s_st[2][BN];
+  if (stats) {
+    for (int i = t; i < 2 * BN; i += 256) s_st[i / BN][i % BN] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int n = 0; n < NW; ++n) {
+      const int colL = wc * (BN / 2) + n * 16 + d_col;
+      if (n0 + colL >= N) continue;
+      float ps = 0.f, pq = 0.f;
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        const long row_base = m0 + wr * 64 + m * 16 + d_row;
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          if (row_base + j < M) {
+            float v = acc[m][n][j];
+            ps += v;
+            pq += v * v;
+          }
+      }
+      atomicAdd(&s_st[0][colL], ps);
+      atomicAdd(&s_st[1][colL], pq);
+    }
+    __syncthreads();
+    float* slice = stats + (long)(bid & 63) * 2 * N;
+    for (int i = t; i < BN && n0 + i < N; i += 256) {
+      atomicAdd(slice + n0 + i, s_st[0][i]);
+      atomicAdd(slice + N + n0 + i, s_st[1][i]);
+    }
+    __syncthreads();
+  }
   T* tile = As;                          // [BM][BN] fp16 staging (fits)
 #pragma unroll
   for (int n = 0; n < NW; ++n) {
@@ -516,7 +550,8 @@ static at::Tensor pad_k8(const at::Tensor& x) {
 
 // core: C[.., M, N] = A[.., M, K] x B[.., N, K]^T (+bias) (opt relu)
 at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
-                        c10::optional<at::Tensor> bias, bool relu) {
+                        c10::optional<at::Tensor> bias, bool relu,
+                        at::Tensor* stats_out) {
   CHECK_GPU(A);
   TORCH_CHECK(A.scalar_type() == B.scalar_type(), "gemm dtype mismatch");
   bool batched = A.dim() == 3;
@@ -563,6 +598,14 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   }
   int span = ksplit > 1 ? tps : nk_total;
   int nbuf = span > 1 ? 2 : 1;
+  // fused BN-forward reduction: per-channel sum/ssq accumulated by the
+  // epilogue ([64 slices][2][N] fp32; caller folds).  Split-K and the
+  // 8-phase path skip it -- *stats_out stays undefined then.
+  float* stats_ptr = nullptr;
+  if (stats_out && ksplit == 1 && !batched) {
+    *stats_out = at::zeros({64, 2, N}, A.options().dtype(at::kFloat));
+    stats_ptr = stats_out->data_ptr<float>();
+  }
   // 256^2 8-phase kernel: refchecked, currently at parity with the
   // 128^2 path (drain-at-boundary; the counted-vmcnt form needs a
   // 3-buffer ring — round-2 work), so routing is opt-in
@@ -572,6 +615,7 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   }();
   bool big = use8ph && !batched && ksplit == 1 && M >= 512 && N >= 256 &&
              K >= 256 && !bias_ptr && !relu;
+  if (big && stats_out) *stats_out = at::Tensor();  // 8ph has no stats
   if (big) {
     long nwg8 = ((M + 255) / 256) * ((N + 255) / 256);
     DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt8", [&] {
@@ -603,7 +647,8 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
               bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
               (const scalar_t*)zero_page(A), relu,
               ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
-              GemmLd{K, K, N, 1, 0, 0, 0});
+              GemmLd{K, K, N, 1, 0, 0, 0},
+              ksplit > 1 ? nullptr : stats_ptr);
     else
      hipLaunchKernelGGL(( gemm_nt_mfma_kernel<scalar_t, 128>)
           , dim3(grid), dim3(256), lds_bytes, cur_stream(), 
@@ -611,7 +656,8 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
               bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
               (const scalar_t*)zero_page(A), relu,
               ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
-              GemmLd{K, K, N, 1, 0, 0, 0});
+              GemmLd{K, K, N, 1, 0, 0, 0},
+              ksplit > 1 ? nullptr : stats_ptr);
   });
   HIP_CHECK_LAST();
   if (ksplit > 1) {

@@ -519,12 +519,24 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
                                           at::Tensor rmean, at::Tensor rvar,
                                           double momentum, double eps,
                                           bool fuse_relu,
-                                          const at::Tensor& residual) {
+                                          const at::Tensor& residual,
+                                          c10::optional<at::Tensor> presums) {
   CHECK_GPU(x); CHECK_CONTIG(x);
   long C = x.size(-1), M = x.numel() / C;
   auto y = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
-  auto sum = at::zeros({C}, opts), sumsq = at::zeros({C}, opts);
+  // precomputed per-channel {sum, ssq} from the producing conv/GEMM
+  // epilogue ([64,2,C] partial slices) -- skips the reduce pass
+  bool have_pre = presums && presums->defined() && presums->numel() > 0;
+  at::Tensor sum, sumsq;
+  if (have_pre) {
+    auto folded = presums->sum(0);  // [2, C]
+    sum = folded[0].contiguous();
+    sumsq = folded[1].contiguous();
+  } else {
+    sum = at::zeros({C}, opts);
+    sumsq = at::zeros({C}, opts);
+  }
   auto save_mean = at::empty({C}, opts), save_istd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts), shift = at::empty({C}, opts);
   auto g32 = gamma.to(at::kFloat).contiguous();
@@ -543,14 +555,16 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
   dim3 grid = bn_reduce_grid(M, C, &rpb);
   auto stream = cur_stream();
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "bn_fwd", [&] {
-    if (C % 8 == 0 && sizeof(scalar_t) == 2)
-     hipLaunchKernelGGL(( bn_reduce_vec_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
-          (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
-          sumsq.data_ptr<float>());
-    else
-     hipLaunchKernelGGL(( bn_reduce_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
-          (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
-          sumsq.data_ptr<float>());
+    if (!have_pre) {
+      if (C % 8 == 0 && sizeof(scalar_t) == 2)
+       hipLaunchKernelGGL(( bn_reduce_vec_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
+            (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
+            sumsq.data_ptr<float>());
+      else
+       hipLaunchKernelGGL(( bn_reduce_kernel<scalar_t>), dim3(grid), dim3(256), 0, stream, 
+            (const scalar_t*)x.data_ptr(), M, C, rpb, sum.data_ptr<float>(),
+            sumsq.data_ptr<float>());
+    }
    hipLaunchKernelGGL(( bn_finalize_kernel), dim3((int)((C + 255) / 256)), dim3(256), 0, stream, 
         sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C,
         (float)momentum, (float)eps, save_mean.data_ptr<float>(),

@@ -109,11 +109,22 @@ class _Conv2dNHWC(torch.autograd.Function):
         ctx.conf = (stride, pad, dilation, groups)
         ctx.has_bias = b is not None
         ext = hip_required('conv2d')
-        return ext.conv2d_nhwc_fwd(x, w, b, stride[0], stride[1],
-                                   pad[0], pad[1], dilation[0], dilation[1], groups)
+        if b is None and torch.is_grad_enabled():
+            # also harvest per-channel {sum, ssq} from the epilogue --
+            # a following BatchNorm consumes it and skips its forward
+            # reduction pass (empty when the path has no fused stats)
+            y, stats = ext.conv2d_nhwc_fwd_stats(
+                x, w, None, stride[0], stride[1], pad[0], pad[1],
+                dilation[0], dilation[1], groups)
+            ctx.mark_non_differentiable(stats)
+            return y, stats
+        y = ext.conv2d_nhwc_fwd(x, w, b, stride[0], stride[1],
+                                pad[0], pad[1], dilation[0], dilation[1],
+                                groups)
+        return y, y.new_empty(0)
 
     @staticmethod
-    def backward(ctx, dy):
+    def backward(ctx, dy, _dstats=None):
         x, w = ctx.saved_tensors
         stride, pad, dilation, groups = ctx.conf
         dy = dy.contiguous()
@@ -218,8 +229,11 @@ def conv2d(x, w, b=None, stride=(1, 1), pad=(0, 0), dilation=(1, 1),
            groups=1, layout='NCHW'):
     if layout == 'NHWC':
         if use_hip(x):
-            return _Conv2dNHWC.apply(x.contiguous(), w.contiguous(), b,
-                                     stride, pad, dilation, groups)
+            y, stats = _Conv2dNHWC.apply(x.contiguous(), w.contiguous(), b,
+                                         stride, pad, dilation, groups)
+            if stats.numel():
+                y._bn_presums = stats
+            return y
         # CPU oracle for the NHWC kernels: permute through torch NCHW conv
         xn = x.permute(0, 3, 1, 2)
         wn = w.permute(0, 3, 1, 2)
@@ -255,9 +269,13 @@ class _BatchNormNHWC(torch.autograd.Function):
                 fuse_relu, residual):
         ext = hip_required('batch_norm')
         if training:
+            presums = getattr(x, '_bn_presums', None)
+            if presums is not None:
+                del x._bn_presums  # consume once
             y, save_mean, save_istd, mask = ext.bn_nhwc_fwd_train(
                 x, gamma, beta, rmean, rvar, momentum, eps, fuse_relu,
-                residual if residual is not None else x.new_empty(0))
+                residual if residual is not None else x.new_empty(0),
+                presums)
             ctx.save_for_backward(x, gamma, save_mean, save_istd, y, mask)
             ctx.fuse_relu = fuse_relu
             ctx.has_res = residual is not None

@@ -665,3 +665,51 @@ def test_rtc_launch_gpu():
              (256, 1, 1))
     torch.cuda.synchronize()
     assert torch.allclose(y, y0 + 3.0 * x, atol=1e-5)
+
+
+def test_conv_bn_fused_presums():
+    """Conv epilogue's fused per-channel {sum, ssq} matches a direct
+    reduction of the conv output (BN forward-reduce fusion)."""
+    from mxnet_amd.ops import nn as onn
+    torch.manual_seed(44)
+    for shape in [((4, 16, 16, 64), (64, 3, 3, 64), (1, 1)),   # igemm
+                  ((4, 14, 14, 128), (256, 1, 1, 128), (1, 0))]:  # gemm 1x1
+        xs, ws, (st_, pd_) = shape
+        x = torch.randn(*xs, device='cuda', dtype=torch.float16)
+        w = (torch.randn(*ws, device='cuda', dtype=torch.float16) * 0.1) \
+            .requires_grad_(True)
+        y = onn.conv2d(x, w, None, (st_, st_), (pd_, pd_), (1, 1), 1,
+                       layout='NHWC')
+        st = getattr(y, '_bn_presums', None)
+        assert st is not None and st.numel(), 'fused stats missing'
+        folded = st.sum(0)
+        yf = y.detach().float().reshape(-1, y.shape[-1])
+        ref_sum = yf.sum(0)
+        ref_ssq = (yf * yf).sum(0)
+        assert torch.allclose(folded[0], ref_sum, rtol=2e-2, atol=2.0), \
+            (folded[0] - ref_sum).abs().max()
+        assert torch.allclose(folded[1], ref_ssq, rtol=2e-2, atol=2.0), \
+            (folded[1] - ref_ssq).abs().max()
+
+
+def test_bn_with_presums_matches_plain():
+    """bn_nhwc_fwd_train(presums) == bn_nhwc_fwd_train() (same stats)."""
+    torch.manual_seed(45)
+    M, C = 512, 64
+    x = torch.randn(M, 1, 1, C, device='cuda', dtype=torch.float16)
+    g = torch.ones(C, device='cuda')
+    b = torch.zeros(C, device='cuda')
+    rm = torch.zeros(C, device='cuda')
+    rv = torch.ones(C, device='cuda')
+    y0, m0, i0, _ = ext.bn_nhwc_fwd_train(x, g, b, rm.clone(), rv.clone(),
+                                          0.9, 1e-5, False, x.new_empty(0))
+    xf = x.float().reshape(-1, C)
+    pre = torch.zeros(64, 2, C, device='cuda')
+    pre[0, 0] = xf.sum(0)
+    pre[0, 1] = (xf * xf).sum(0)
+    y1, m1, i1, _ = ext.bn_nhwc_fwd_train(x, g, b, rm.clone(), rv.clone(),
+                                          0.9, 1e-5, False, x.new_empty(0),
+                                          pre)
+    assert torch.allclose(m0, m1, rtol=1e-3, atol=1e-3)
+    assert torch.allclose(i0, i1, rtol=1e-3, atol=1e-3)
+    assert torch.allclose(y0.float(), y1.float(), rtol=2e-2, atol=2e-2)
