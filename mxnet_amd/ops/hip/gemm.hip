@@ -586,11 +586,13 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   int nk_total = (int)((K + 63) / 64);
   int ksplit = 1, tps = nk_total;
   at::Tensor out32;
-  // small grids: split the reduction until the chip fills (256 CUs want
-  // >=512 workgroups; recurrent-GEMM shapes like M=128,K=1024 give 32)
-  if (nwg * nb < 512 && nk_total >= 8) {
+  // split-K gate: nk>16 only -- widening to nk>=8 was MEASURED WORSE on
+  // the LSTM recurrent GEMM (M=128,N=4096,K=1024: the fp32 workspace
+  // zero+cast passes and atomics cost more than the occupancy win;
+  // 1.59M -> 1.19M tokens/s), so small-K chip-filling stays off.
+  if (nwg * nb < 512 && nk_total > 16) {
     ksplit = (int)std::min<long>((2048 + nwg * nb - 1) / (nwg * nb),
-                                 (nk_total + 3) / 4);
+                                 (nk_total + 15) / 16);
     tps = (nk_total + ksplit - 1) / ksplit;
     ksplit = (nk_total + tps - 1) / tps;
     out32 = at::zeros(out.sizes(), out.options().dtype(at::kFloat));
