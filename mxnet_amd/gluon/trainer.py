@@ -61,6 +61,8 @@ class Trainer:
         else:
             self._kvstore = kvs_mod.create(kind)
         self._distributed = isinstance(self._kvstore, kvs_mod.DistKVStore)
+        self._async_ps = isinstance(self._kvstore,
+                                    getattr(kvs_mod, 'AsyncPSKVStore', ()))
         if self._compression_params and self._kvstore is not None:
             self._kvstore.set_gradient_compression(self._compression_params)
         if self._distributed:
@@ -69,6 +71,10 @@ class Trainer:
             for i, p in enumerate(self._params):
                 for d in p.list_data():
                     self._kvstore.broadcast(i, d, d)
+        if self._async_ps:
+            # register every weight on the parameter server
+            for i, p in enumerate(self._params):
+                self._kvstore.init(i, p.list_data()[0])
         self._kv_initialized = True
 
     def _check_states(self, i, p):
@@ -81,6 +87,20 @@ class Trainer:
     def step(self, batch_size, ignore_stale_grad=False):
         self._init_kvstore()
         self._optimizer.rescale_grad = self._scale / batch_size
+        if self._async_ps:
+            # dist_async update-on-kvstore (reference KVStoreDist async
+            # mode: push grad -> server updates immediately -> pull the
+            # current weights; no global barrier, no local optimizer)
+            rescale = self._scale / batch_size
+            with torch.no_grad():
+                for i, p in enumerate(self._params):
+                    g = p.list_grad()[0]
+                    g._t.mul_(rescale)
+                    self._kvstore.push(i, g)
+                    self._kvstore.pull(i, p.list_data())
+                    if p.grad_req == 'write':
+                        p.zero_grad()
+            return
         self._allreduce_grads()
         self._update(ignore_stale_grad)
 

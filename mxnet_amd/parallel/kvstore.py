@@ -60,7 +60,9 @@ class KVStoreBase:
 def create(name='local'):
     """Factory (reference KVStore::Create kvstore.cc:42-80)."""
     name = name.lower()
-    if name in ('dist_sync', 'dist_device_sync', 'dist_async', 'nccl',
+    if name == 'dist_async':
+        return AsyncPSKVStore(name)
+    if name in ('dist_sync', 'dist_device_sync', 'nccl',
                 'dist', 'horovod', 'byteps'):
         return DistKVStore(name)
     if name in ('local', 'device', 'local_allreduce_cpu',
@@ -265,3 +267,132 @@ class DistKVStore(KVStoreBase):
         raise NotImplementedError(
             'server-side optimizer: use update_on_kvstore=False with the '
             'distributed Trainer')
+
+
+# ---------------------------------------------------------------------------
+# dist_async: parameter server with a dedicated server rank
+# ---------------------------------------------------------------------------
+
+class AsyncPSKVStore(KVStoreBase):
+    """Asynchronous parameter server (reference KVStoreDist +
+    KVStoreDistServer, kvstore_dist_server.h:155: workers push grads,
+    the server applies the optimizer immediately — no global barrier —
+    and serves pulls of the current weights).
+
+    Roles: the LAST rank of the torch.distributed world is the server
+    (reference: DMLC_ROLE=server process); ranks 0..world-2 are
+    workers.  Transport is gloo point-to-point (send/recv) — weights
+    live on the server in fp32; GPU workers stage through CPU.  The
+    server process calls ``run_server()`` (blocks until every worker
+    sent the stop command); workers use push/pull, and the Trainer
+    drives update-on-kvstore semantics.
+    """
+
+    _OP_INIT, _OP_PUSH, _OP_PULL, _OP_STOP = 0, 1, 2, 3
+
+    def __init__(self, kind='dist_async'):
+        self._type = kind
+        if not dist.is_initialized():
+            dist.init_process_group(backend='gloo')
+        self._world = dist.get_world_size()
+        assert self._world >= 2, 'dist_async needs >= 2 ranks (+1 server)'
+        self._server = self._world - 1
+        self._rank = dist.get_rank()
+        self._optimizer = None
+        self._store = {}        # server side: key -> fp32 tensor
+        self._states = {}       # server side optimizer states
+        self._workers_group = dist.new_group(
+            ranks=list(range(self._world - 1)))
+
+    @property
+    def rank(self):
+        return self._rank
+
+    @property
+    def num_workers(self):
+        return self._world - 1
+
+    @property
+    def is_server(self):
+        return self._rank == self._server
+
+    def set_optimizer(self, optimizer):
+        """Server-side updater (reference sync/async server optimizer,
+        kvstore_dist_server.h:346-365); call on the server process."""
+        self._optimizer = optimizer
+
+    # -- worker protocol -------------------------------------------------
+    def _send_header(self, op, key, numel):
+        h = torch.tensor([op, key, numel, 0], dtype=torch.long)
+        dist.send(h, dst=self._server)
+
+    def init(self, key, value):
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        t = v._t.detach().float().cpu().contiguous()
+        dist.broadcast(t, src=0, group=self._workers_group)
+        with torch.no_grad():
+            v._t.copy_(t.to(v._t.device, v._t.dtype))
+        if self._rank == 0:
+            self._send_header(self._OP_INIT, key, t.numel())
+            dist.send(t, dst=self._server)
+        dist.barrier(group=self._workers_group)
+
+    def push(self, key, value, priority=0):
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        g = v._t.detach().float().cpu().contiguous()
+        self._send_header(self._OP_PUSH, key, g.numel())
+        dist.send(g, dst=self._server)
+
+    def pull(self, key, out=None, priority=0, ignore_sparse=True):
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        n = outs[0]._t.numel()
+        self._send_header(self._OP_PULL, key, n)
+        buf = torch.empty(n, dtype=torch.float32)
+        dist.recv(buf, src=self._server)
+        with torch.no_grad():
+            for o in outs:
+                o._t.copy_(buf.view(o._t.shape).to(o._t.device, o._t.dtype))
+
+    def pushpull(self, key, value, out=None, priority=0, async_op=False):
+        self.push(key, value, priority)
+        if out is not None:
+            self.pull(key, out, priority)
+
+    def stop(self):
+        self._send_header(self._OP_STOP, 0, 0)
+
+    def barrier_workers(self):
+        dist.barrier(group=self._workers_group)
+
+    # -- server loop -----------------------------------------------------
+    def run_server(self):
+        """Service requests until every worker sent STOP (reference
+        DataHandleEx dispatch, kvstore_dist_server.h:325)."""
+        assert self.is_server
+        live = self.num_workers
+        hdr = torch.empty(4, dtype=torch.long)
+        while live > 0:
+            src = dist.recv(hdr, src=None)
+            op, key, numel = int(hdr[0]), int(hdr[1]), int(hdr[2])
+            if op == self._OP_STOP:
+                live -= 1
+                continue
+            if op == self._OP_INIT:
+                t = torch.empty(numel, dtype=torch.float32)
+                dist.recv(t, src=src)
+                self._store[key] = t
+            elif op == self._OP_PUSH:
+                g = torch.empty(numel, dtype=torch.float32)
+                dist.recv(g, src=src)
+                w = self._store[key]
+                if self._optimizer is not None:
+                    if key not in self._states:
+                        self._states[key] = \
+                            self._optimizer.create_state_multi_precision(
+                                key, NDArray(w))
+                    self._optimizer.update_multi_precision(
+                        key, NDArray(w), NDArray(g), self._states[key])
+                else:
+                    w.sub_(g)  # plain accumulate (reference default)
+            elif op == self._OP_PULL:
+                dist.send(self._store[key], dst=src)

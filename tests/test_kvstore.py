@@ -106,3 +106,87 @@ def test_dist_kvstore_gloo_two_procs(tmp_path):
         assert p.returncode == 0, out.decode()
     assert any('RANK_OK 0' in o for o in outs)
     assert any('RANK_OK 1' in o for o in outs)
+
+
+_PS_WORKER = '''
+import os
+import numpy as np
+import torch
+import torch.distributed as dist
+import mxnet_amd as mx
+from mxnet_amd import autograd
+from mxnet_amd.gluon import Trainer, nn
+from mxnet_amd.parallel import kvstore as kvs
+
+rank = int(os.environ['RANK'])
+world = int(os.environ['WORLD_SIZE'])
+kv = kvs.create('dist_async')
+if kv.is_server:
+    from mxnet_amd import optimizer as opt
+    kv.set_optimizer(opt.create('sgd', learning_rate=0.5))
+    kv.run_server()
+else:
+    torch.manual_seed(4)
+    net = nn.Dense(3, in_units=5, use_bias=False)
+    net.initialize()
+    tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.5},
+                 kvstore=kv)
+    X = torch.full((2, 5), 0.1 * (rank + 1))
+    x = mx.nd.from_torch(X)
+    with autograd.record():
+        L = mx.nd.from_torch(net(x).handle.sum())
+    L.backward()
+    tr.step(1)
+    kv.barrier_workers()       # both pushes applied before final pull
+    kv.pull(0, net.weight.data())
+    np.save(os.environ['OUT_PREFIX'] + f'_w{rank}.npy',
+            net.weight.data().asnumpy())
+    kv.stop()
+print('PS_OK', rank)
+'''
+
+
+def test_dist_async_parameter_server(tmp_path):
+    """dist_async: 2 workers + 1 server rank; server applies SGD per
+    push; after both pushes the pulled weights equal w0 - lr*(g1+g2)
+    (reference KVStoreDistServer async mode)."""
+    import subprocess, sys, socket
+    script = tmp_path / 'ps.py'
+    script.write_text(_PS_WORKER)
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sock = socket.socket(); sock.bind(('127.0.0.1', 0))
+    port = str(sock.getsockname()[1]); sock.close()
+    env = dict(os.environ)
+    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': port,
+                'OUT_PREFIX': str(tmp_path / 'ps'),
+                'PYTHONPATH': repo_root + os.pathsep +
+                env.get('PYTHONPATH', '')})
+    procs = []
+    for rank in range(3):
+        e = dict(env, RANK=str(rank), WORLD_SIZE='3')
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        outs.append(out.decode())
+        assert p.returncode == 0, out.decode()
+
+    import numpy as np
+    import torch
+    w0 = np.load(tmp_path / 'ps_w0.npy')
+    w1 = np.load(tmp_path / 'ps_w1.npy')
+    np.testing.assert_allclose(w0, w1, rtol=1e-6)
+    # oracle: dL/dW for sum(x @ W^T) is ones(3,1) @ x_mean... grad per
+    # worker = column-broadcast of X rows summed: each row of dW =
+    # sum of X rows = 2*0.1*(rank+1) per entry
+    torch.manual_seed(4)
+    import mxnet_amd as mx
+    from mxnet_amd.gluon import nn as gnn
+    net = gnn.Dense(3, in_units=5, use_bias=False)
+    net.initialize()
+    init_w = net.weight.data().asnumpy()
+    g_total = np.full((3, 5), 2 * 0.1 * 1) + np.full((3, 5), 2 * 0.1 * 2)
+    np.testing.assert_allclose(w0, init_w - 0.5 * g_total, rtol=1e-4,
+                               atol=1e-5)
