@@ -109,7 +109,9 @@ class _Conv2dNHWC(torch.autograd.Function):
         ctx.conf = (stride, pad, dilation, groups)
         ctx.has_bias = b is not None
         ext = hip_required('conv2d')
-        if b is None and torch.is_grad_enabled():
+        # NOTE grad mode is force-disabled inside Function.forward --
+        # needs_input_grad carries the caller's training intent
+        if b is None and any(ctx.needs_input_grad):
             # also harvest per-channel {sum, ssq} from the epilogue --
             # a following BatchNorm consumes it and skips its forward
             # reduction pass (empty when the path has no fused stats)
