@@ -229,7 +229,13 @@ def _scalar_op(opname):
     return fn
 
 
+def _eval_fused_subgraph(*args, **kwargs):
+    from ..contrib.fusion import execute_fused
+    return execute_fused(*args, **kwargs)
+
+
 _EVAL_TABLE = {
+    '_fused_subgraph': _eval_fused_subgraph,
     '_plus_scalar': _scalar_op('_plus_scalar'),
     '_minus_scalar': _scalar_op('_minus_scalar'),
     '_mul_scalar': _scalar_op('_mul_scalar'),

@@ -119,7 +119,22 @@ def partition_graph(sym, property='elemwise_fusion'):
                         if id(src) not in member_ids and \
                                 (id(src), oi) not in [(id(a), b) for a, b in ext]:
                             ext.append((src, oi))
-                sub_json = Symbol(members[-1]).tojson()
+                # self-contained sub-symbol: member clones over _in{k}
+                # placeholder vars (externals must NOT leak whole
+                # upstream graphs into the attribute)
+                ext_key = {(id(a), b): k for k, (a, b) in enumerate(ext)}
+                clones = {}
+                for m in members:
+                    mi = []
+                    for src, oi in m.inputs:
+                        if id(src) in member_ids:
+                            mi.append((clones[id(src)], oi))
+                        else:
+                            mi.append((_Node('null',
+                                             f'_in{ext_key[(id(src), oi)]}',
+                                             {}, []), 0))
+                    clones[id(m)] = _Node(m.op, m.name, dict(m.attrs), mi)
+                sub_json = Symbol(clones[id(members[-1])]).tojson()
                 fnode = _Node(
                     prop.subgraph_op(), f'fused_{prop.name}_{gi}',
                     {'ops': ','.join(m.op for m in members),
