@@ -111,17 +111,32 @@ __global__ void bn_reduce_kernel(const T* __restrict__ x, long M, long C,
 // finalize training stats: mean/istd + running-stat update
 // (running = momentum*running + (1-momentum)*batch; running var unbiased,
 // matching the torch CPU oracle)
+// nslices > 0: sum/sumsq point at the conv epilogue's sliced
+// [nslices][2][C] workspace; the 64-way fold happens here (saves the
+// separate torch reduction + contiguous sum buffers).
 __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    const float* __restrict__ sumsq, long M,
                                    long C, float momentum, float eps,
                                    float* __restrict__ save_mean,
                                    float* __restrict__ save_istd,
                                    float* __restrict__ rmean,
-                                   float* __restrict__ rvar) {
+                                   float* __restrict__ rvar,
+                                   int nslices = 0) {
   for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < C;
        c += (long)gridDim.x * blockDim.x) {
-    float mean = sum[c] / M;
-    float var = fmaxf(sumsq[c] / M - mean * mean, 0.f);
+    float s0, s1;
+    if (nslices > 0) {
+      s0 = 0.f; s1 = 0.f;
+      for (int k = 0; k < nslices; ++k) {
+        s0 += sum[(long)k * 2 * C + c];
+        s1 += sum[(long)k * 2 * C + C + c];
+      }
+    } else {
+      s0 = sum[c];
+      s1 = sumsq[c];
+    }
+    float mean = s0 / M;
+    float var = fmaxf(s1 / M - mean * mean, 0.f);
     save_mean[c] = mean;
     save_istd[c] = rsqrtf(var + eps);
     if (rmean) {
@@ -526,13 +541,13 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
   auto y = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
   // precomputed per-channel {sum, ssq} from the producing conv/GEMM
-  // epilogue ([64,2,C] partial slices) -- skips the reduce pass
+  // epilogue ([64,2,C] partial slices; folded inside bn_finalize)
   bool have_pre = presums && presums->defined() && presums->numel() > 0;
+  int nslices = have_pre ? (int)presums->size(0) : 0;
   at::Tensor sum, sumsq;
   if (have_pre) {
-    auto folded = presums->sum(0);  // [2, C]
-    sum = folded[0].contiguous();
-    sumsq = folded[1].contiguous();
+    sum = *presums;           // sliced workspace, finalize folds it
+    sumsq = sum;              // unused in the sliced path
   } else {
     sum = at::zeros({C}, opts);
     sumsq = at::zeros({C}, opts);
@@ -569,7 +584,7 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
         sum.data_ptr<float>(), sumsq.data_ptr<float>(), M, C,
         (float)momentum, (float)eps, save_mean.data_ptr<float>(),
         save_istd.data_ptr<float>(), rm32.data_ptr<float>(),
-        rv32.data_ptr<float>());
+        rv32.data_ptr<float>(), nslices);
    hipLaunchKernelGGL(( bn_scale_shift_kernel), dim3((int)((C + 255) / 256)), dim3(256), 0, stream, 
         g32.data_ptr<float>(), b32.data_ptr<float>(),
         save_mean.data_ptr<float>(), save_istd.data_ptr<float>(), true, 0.f,

@@ -96,6 +96,15 @@ def batch_dot(a, b, transpose_a=False, transpose_b=False):
 # Convolution
 # ---------------------------------------------------------------------------
 
+import functools
+
+
+@functools.lru_cache(None)
+def _bn_fused_stats():
+    import os
+    return os.environ.get('MXNET_BN_FUSED_STATS', '1') != '0'
+
+
 class _Conv2dNHWC(torch.autograd.Function):
     """NHWC conv2d on the native implicit-GEMM/im2col MFMA kernels.
 
@@ -111,7 +120,7 @@ class _Conv2dNHWC(torch.autograd.Function):
         ext = hip_required('conv2d')
         # NOTE grad mode is force-disabled inside Function.forward --
         # needs_input_grad carries the caller's training intent
-        if b is None and any(ctx.needs_input_grad):
+        if b is None and any(ctx.needs_input_grad) and _bn_fused_stats():
             # also harvest per-channel {sum, ssq} from the epilogue --
             # a following BatchNorm consumes it and skips its forward
             # reduction pass (empty when the path has no fused stats)
