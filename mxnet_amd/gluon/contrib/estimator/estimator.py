@@ -40,6 +40,15 @@ class Estimator:
         with autograd.record():
             pred = self.net(x)
             loss = self.loss(pred, y)
+            # StochasticBlock intermediate losses (e.g. KL penalties of
+            # sampled latents) join the objective (reference
+            # gluon/probability StochasticBlock + estimator ELBO usage)
+            extra = getattr(self.net, 'losses', None)
+            if extra:
+                for term in extra:
+                    t = term.handle if hasattr(term, 'handle') else term
+                    loss = loss + type(loss)(t.mean()) if hasattr(
+                        loss, 'handle') else loss + t.mean()
         loss.backward()
         bs = x.shape[batch_axis]
         self.trainer.step(bs)
