@@ -667,10 +667,14 @@ def test_rtc_launch_gpu():
     assert torch.allclose(y, y0 + 3.0 * x, atol=1e-5)
 
 
-def test_conv_bn_fused_presums():
+def test_conv_bn_fused_presums(monkeypatch, request):
     """Conv epilogue's fused per-channel {sum, ssq} matches a direct
-    reduction of the conv output (BN forward-reduce fusion)."""
+    reduction of the conv output (BN forward-reduce fusion; default-off
+    -- measured slower end-to-end -- so enabled explicitly here)."""
     from mxnet_amd.ops import nn as onn
+    monkeypatch.setenv('MXNET_BN_FUSED_STATS', '1')
+    onn._bn_fused_stats.cache_clear()
+    request.addfinalizer(onn._bn_fused_stats.cache_clear)
     torch.manual_seed(44)
     for shape in [((4, 16, 16, 64), (64, 3, 3, 64), (1, 1)),   # igemm
                   ((4, 14, 14, 128), (256, 1, 1, 128), (1, 0))]:  # gemm 1x1
