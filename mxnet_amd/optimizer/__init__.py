@@ -227,6 +227,8 @@ class Adam(Optimizer):
         return (torch.zeros_like(w, dtype=torch.float32),
                 torch.zeros_like(w, dtype=torch.float32))
 
+    _adamw = False
+
     def update(self, index, weight, grad, state):
         self._update_count(index)
         lr, wd = self._get_lr(index), self._get_wd(index)
@@ -241,9 +243,41 @@ class Adam(Optimizer):
             v.mul_(self.beta2).addcmul_(g, g, value=1 - self.beta2)
             w.sub_((lr_t * m / (v.sqrt() + self.epsilon)).to(w.dtype))
 
+    def update_multi_precision(self, index, weight, grad, state):
+        """Fused single-kernel Adam on GPU (reference adam_update /
+        mp_adam_update): rescale+clip+wd+moment updates+bias-corrected
+        step+master-weight cast in ONE launch — the eager torch
+        composition was ~10 elementwise launches per parameter and
+        DOMINATED the BERT step (profiles/r01_summary.md)."""
+        w = weight._t if isinstance(weight, NDArray) else weight
+        g = grad._t if isinstance(grad, NDArray) else grad
+        if w.is_cuda and g.dtype == w.dtype:
+            from ..ops.dispatch import hipops, use_hip
+            ext = hipops() if use_hip(w) else None
+            if ext is not None and g.is_contiguous() and w.is_contiguous():
+                self._update_count(index)
+                lr, wd = self._get_lr(index), self._get_wd(index)
+                t = self._index_update_count[index]
+                lr_t = lr * math.sqrt(1 - self.beta2 ** t) / \
+                    (1 - self.beta1 ** t)
+                if self.multi_precision and isinstance(state, tuple) and \
+                        isinstance(state[0], torch.Tensor) and \
+                        state[0].dtype is torch.float32 and \
+                        w.dtype in (torch.float16, torch.bfloat16):
+                    master, (m, v) = state[0], state[1]
+                else:
+                    master, (m, v) = None, state
+                ext.adam_update(w, master, g, m, v, lr_t, self.beta1,
+                                self.beta2, self.epsilon, wd,
+                                self.rescale_grad,
+                                self.clip_gradient or 0.0, self._adamw)
+                return
+        return super().update_multi_precision(index, weight, grad, state)
+
 
 @register
 class AdamW(Adam):
+    _adamw = True
     """Decoupled weight decay (reference contrib/adamw.cc)."""
 
     def update(self, index, weight, grad, state):

@@ -738,3 +738,32 @@ def test_rtc_fusion_gpu():
         torch.cuda.synchronize()
         ref = torch.relu(xt.float() * 2 + yt.float()) * 0.5
         assert torch.allclose(out.handle.float(), ref, atol=2e-3)
+
+
+def test_adam_optimizer_fused_path():
+    """Adam.update_multi_precision routes to the fused kernel and tracks
+    the eager fp32 composition across steps (incl. master weights)."""
+    import mxnet_amd as mx
+    from mxnet_amd import optimizer as opt
+    from mxnet_amd.ndarray.ndarray import NDArray
+    torch.manual_seed(9)
+    w0 = torch.randn(1000)
+    g_seq = [torch.randn(1000) * 0.1 for _ in range(4)]
+
+    # GPU fused, fp16 weight + fp32 master
+    o = opt.create('adam', learning_rate=0.01, wd=0.01, multi_precision=True)
+    w = NDArray(w0.clone().half().cuda())
+    st = o.create_state_multi_precision(0, w)
+    for g in g_seq:
+        o.update_multi_precision(0, w, NDArray(g.half().cuda()), st)
+    torch.cuda.synchronize()
+
+    # CPU fp32 oracle through the eager update
+    o2 = opt.create('adam', learning_rate=0.01, wd=0.01)
+    w2 = NDArray(w0.clone())
+    st2 = o2.create_state(0, w2)
+    for g in g_seq:
+        o2.update(0, w2, NDArray(g.clone()), st2)
+
+    err = (st[0].cpu() - w2.handle).abs().max()
+    assert float(err) < 2e-3, float(err)
