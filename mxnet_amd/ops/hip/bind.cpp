@@ -66,6 +66,8 @@ at::Tensor im2col_nhwc_op(const at::Tensor&, int, int, int, int, int, int,
 std::vector<at::Tensor> conv2d_nhwc_fwd_stats(
     const at::Tensor&, const at::Tensor&, c10::optional<at::Tensor>, int64_t,
     int64_t, int64_t, int64_t, int64_t, int64_t, int64_t);
+std::vector<at::Tensor> gemm_tn_fused(const at::Tensor&, const at::Tensor&,
+                                      bool);
 std::vector<at::Tensor> attention_fwd(const at::Tensor&,
                                       c10::optional<at::Tensor>, int64_t,
                                       double);
@@ -140,6 +142,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dtype") = py::none());
   m.def("im2col_nhwc", &im2col_nhwc_op);
   m.def("conv2d_nhwc_fwd_stats", &conv2d_nhwc_fwd_stats);
+  m.def("gemm_tn_fused", &gemm_tn_fused);
   m.def("attention_fwd", &attention_fwd, py::arg("qkv"),
         py::arg("mask") = c10::nullopt, py::arg("heads"),
         py::arg("temperature") = 1.0);

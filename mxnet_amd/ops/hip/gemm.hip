@@ -728,6 +728,191 @@ at::Tensor gemm_nt_8ph(const at::Tensor& A_, const at::Tensor& B_) {
 
 
 // ---------------------------------------------------------------------------
+// direct TN GEMM: C[I,J] = A[M,I]^T @ B[M,J]  (+ optional dbias[i] =
+// col-sum of A) — the FullyConnected weight-gradient shape.  The NT
+// library canonicalized TN via TWO global transposes (reference used
+// cublas TN directly); here both operands are consumed straight from
+// their k(=m)-major storage with ds_read_b64_tr_b16 hardware transpose
+// reads over permuted-row [16][16] subtiles (same trick as
+// conv_bwd_w_igemm_tr_kernel), so no transpose passes at all, and the
+// bias gradient rides along on the A tiles (saves the colsum pass).
+// ---------------------------------------------------------------------------
+typedef short trs4g __attribute__((ext_vector_type(4)));
+__device__ inline trs4g tr_read16g(const void* p) {
+  return __builtin_amdgcn_ds_read_tr16_b64_v4i16(
+      (__attribute__((address_space(3))) trs4g*)(uintptr_t)p);
+}
+
+template <typename T>
+__global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    long M, long I, long J, long m_per_slice, float* __restrict__ out32,
+    float* __restrict__ dbias, const T* __restrict__ zpage) {
+  using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
+  constexpr int BI = 64, BJ = 64, BKM = 64;
+  __shared__ T AS[2][BKM * BI];
+  __shared__ T BS[2][BKM * BJ];
+
+  const int nTj = (int)((J + BJ - 1) / BJ);
+  const int bid = blockIdx.x;
+  const long i0 = (long)(bid / nTj) * BI;
+  const long j0 = (long)(bid % nTj) * BJ;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int sm_half = t >> 3;
+  const int seg = t & 7;
+
+  float4_t acc[2][2] = {};
+  float bias_acc[8] = {};
+
+  // stage one [64 m][64 col] panel into the permuted-row subtile layout
+  auto stage = [&](T (*dst)[BKM * BI], int buf, const T* __restrict__ src,
+                   long ld, long W, long col0, long mc, bool want_bias) {
+#pragma unroll
+    for (int rnd = 0; rnd < 2; ++rnd) {
+      const int m_l = rnd * 32 + sm_half;
+      const long m_g = mc + m_l;
+      V8 v = {};
+      const long c = col0 + seg * 8;
+      if (m_g < ms1 && c + 8 <= W)
+        v = *(const V8*)(src + m_g * ld + c);
+      if (want_bias)
+#pragma unroll
+        for (int jj = 0; jj < 8; ++jj) bias_acc[jj] += (float)v[jj];
+      const int kk = m_l >> 5, rem = m_l & 31;
+      const int tt = (rem >> 2) & 1;
+      const int prow = ((rem >> 3) << 2) | (rem & 3);
+      *(V8*)&dst[buf][(((kk * 2 + tt) * 4 + (seg >> 1)) << 8) +
+                      prow * 16 + (seg & 1) * 8] = v;
+    }
+  };
+
+  stage(AS, 0, A, I, I, i0, ms0, dbias != nullptr);
+  stage(BS, 0, B, J, J, j0, ms0, false);
+  __syncthreads();
+  int buf = 0;
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+    if (mc + BKM < ms1) {
+      stage(AS, buf ^ 1, A, I, I, i0, mc + BKM, dbias != nullptr);
+      stage(BS, buf ^ 1, B, J, J, j0, mc + BKM, false);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[2], bf[2];
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int cblk = wr * 2 + m;
+        union { trs4g h[2]; Frag f; } u;
+        u.h[0] = tr_read16g(&AS[buf][(((kk * 2 + 0) * 4 + cblk) << 8) +
+                                     lane * 4]);
+        u.h[1] = tr_read16g(&AS[buf][(((kk * 2 + 1) * 4 + cblk) << 8) +
+                                     lane * 4]);
+        af[m] = u.f;
+      }
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const int cblk = wc * 2 + n;
+        union { trs4g h[2]; Frag f; } u;
+        u.h[0] = tr_read16g(&BS[buf][(((kk * 2 + 0) * 4 + cblk) << 8) +
+                                     lane * 4]);
+        u.h[1] = tr_read16g(&BS[buf][(((kk * 2 + 1) * 4 + cblk) << 8) +
+                                     lane * 4]);
+        bf[n] = u.f;
+      }
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int n = 0; n < 2; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // bias gradient: this block staged columns [i0 + seg*8, +8) of A —
+  // every sm_half row summed its own copies, so LDS-reduce over the 32
+  // stagers, then one global atomic per column per block.  Only the
+  // j-first block column contributes (others stage the same A panel).
+  if (dbias != nullptr && j0 == 0) {
+    __shared__ float bsum[64];
+    for (int i = t; i < 64; i += 256) bsum[i] = 0.f;
+    __syncthreads();
+#pragma unroll
+    for (int jj = 0; jj < 8; ++jj)
+      atomicAdd(&bsum[seg * 8 + jj], bias_acc[jj]);
+    __syncthreads();
+    if (t < 64 && i0 + t < I) atomicAdd(dbias + i0 + t, bsum[t]);
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < 2; ++n) {
+    const long j = j0 + wc * 32 + n * 16 + d_col;
+    if (j >= J) continue;
+#pragma unroll
+    for (int m = 0; m < 2; ++m) {
+      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const long i = i_base + jj;
+        if (i >= I) continue;
+        if (out32) atomicAdd(out32 + i * J + j, acc[m][n][jj]);
+        else C[i * J + j] = (T)acc[m][n][jj];
+      }
+    }
+  }
+}
+
+// host: dw[I,J] = A[M,I]^T B[M,J]; optional dbias = colsum(A) (fp32).
+std::vector<at::Tensor> gemm_tn_fused(const at::Tensor& A,
+                                      const at::Tensor& B, bool want_bias) {
+  CHECK_GPU(A); CHECK_CONTIG(A); CHECK_CONTIG(B);
+  long M = A.size(0), I = A.size(1), J = B.size(1);
+  TORCH_CHECK(B.size(0) == M, "gemm_tn: M mismatch");
+  bool ok = (A.scalar_type() == at::kHalf ||
+             A.scalar_type() == at::kBFloat16) && I % 8 == 0 && J % 8 == 0;
+  if (!ok) {
+    auto C = gemm_nt_core(transpose2d(A), transpose2d(B), c10::nullopt,
+                          false);
+    auto db = want_bias ? colsum(A).to(at::kFloat)
+                        : at::empty({0}, A.options().dtype(at::kFloat));
+    return {C, db};
+  }
+  long nwg = ((I + 63) / 64) * ((J + 63) / 64);
+  long yb = std::max<long>(
+      1, std::min<long>((M + 63) / 64, 1024 / std::max<long>(nwg, 1)));
+  long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
+  yb = (M + m_per_slice - 1) / m_per_slice;
+  auto C = at::empty({I, J}, A.options());
+  at::Tensor o32;
+  if (yb > 1) o32 = at::zeros({I, J}, A.options().dtype(at::kFloat));
+  auto db = want_bias ? at::zeros({I}, A.options().dtype(at::kFloat))
+                      : at::Tensor();
+  dim3 grid((unsigned)nwg, (unsigned)yb);
+  DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_tn_tr", [&] {
+    gemm_tn_tr_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
+        (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+        (scalar_t*)C.data_ptr(), M, I, J, m_per_slice,
+        yb > 1 ? o32.data_ptr<float>() : nullptr,
+        want_bias ? db.data_ptr<float>() : nullptr,
+        (const scalar_t*)zero_page(A));
+  });
+  HIP_CHECK_LAST();
+  if (yb > 1) C.copy_(o32.to(C.scalar_type()));
+  if (!want_bias) db = at::empty({0}, A.options().dtype(at::kFloat));
+  return {C, db};
+}
+
+// ---------------------------------------------------------------------------
 // fused multi-head attention core on the strided NT GEMM
 // (reference transformer attention ran separate transpose/reshape +
 // batch_dot ops, src/operator/contrib/transformer.cc interleaved path;

@@ -60,6 +60,7 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
                         c10::optional<at::Tensor> bias, bool relu,
                         at::Tensor* stats_out = nullptr);
 at::Tensor transpose2d(const at::Tensor& x);
+at::Tensor colsum(const at::Tensor& in);
 const void* zero_page(const at::Tensor& like);
 at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode, double temperature,
                        c10::optional<at::Tensor> mask);

@@ -55,8 +55,14 @@ class _FullyConnected(torch.autograd.Function):
         if use_hip(x):
             ext = hip_required('fully_connected')
             dx = ext.gemm_nn(dy, w) if need_dx else None
-            dw = ext.gemm_tn(dy, x) if need_dw else None
-            db = ext.colsum(dy) if ctx.has_bias else None
+            if need_dw:
+                # direct TN kernel (tr_b16 operand reads, no transpose
+                # passes) with the bias gradient fused into the A tiles
+                dw, db32 = ext.gemm_tn_fused(dy, x, ctx.has_bias)
+                db = db32.to(dy.dtype) if ctx.has_bias else None
+            else:
+                dw = None
+                db = ext.colsum(dy) if ctx.has_bias else None
         else:
             dx = dy @ w if need_dx else None
             dw = dy.t() @ x if need_dw else None
