@@ -114,6 +114,92 @@ __global__ void softmax_bwd_kernel(const T* __restrict__ dy,
   }
 }
 
+// small-C fast path: the whole row lives in registers (VPT values per
+// lane, one wave per row, 4 rows per block) — ONE global read of x (+
+// mask) instead of the generic kernel's three passes.  BERT's masked
+// attention softmax (C = seq = 128) measured 2.05 ms/step on the
+// generic kernel; this is the fix.
+template <typename T, int VPT>
+__global__ void softmax_fwd_rowreg_kernel(
+    const T* __restrict__ x, T* __restrict__ y, long rows, int C,
+    bool log_mode, float invT, const unsigned char* __restrict__ mask) {
+  const int lane = threadIdx.x & 63;
+  const int row_in_blk = threadIdx.x >> 6;
+  for (long r = (long)blockIdx.x * 4 + row_in_blk; r < rows;
+       r += (long)gridDim.x * 4) {
+    const T* xr = x + r * C;
+    const unsigned char* mr = mask ? mask + r * C : nullptr;
+    float v[VPT];
+    bool on[VPT];
+    float m = -INFINITY;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      const bool in = c < C;
+      on[j] = in && (!mr || mr[c]);
+      v[j] = on[j] ? (float)xr[c] : -INFINITY;
+      m = fmaxf(m, v[j]);
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off));
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      v[j] = on[j] ? __expf((v[j] - m) * invT) : 0.f;
+      s += v[j];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      s += __shfl_xor(s, off);
+    const float inv_s = s > 0.f ? 1.f / s : 0.f;
+    const float log_s = __logf(s);
+    T* yr = y + r * C;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      if (c < C)
+        yr[c] = (T)(log_mode
+                        ? (on[j] ? (__logf(v[j]) /*=(x-m)invT*/) - log_s
+                                 : -INFINITY)
+                        : v[j] * inv_s);
+    }
+  }
+}
+
+// backward fast path: dx = (dy - sum(dy*y)) * y * invT, one read of each
+template <typename T, int VPT>
+__global__ void softmax_bwd_rowreg_kernel(const T* __restrict__ dy,
+                                          const T* __restrict__ yv,
+                                          T* __restrict__ dx, long rows,
+                                          int C, float invT) {
+  const int lane = threadIdx.x & 63;
+  const int row_in_blk = threadIdx.x >> 6;
+  for (long r = (long)blockIdx.x * 4 + row_in_blk; r < rows;
+       r += (long)gridDim.x * 4) {
+    const T* gr = dy + r * C;
+    const T* yr = yv + r * C;
+    float g[VPT], yy[VPT];
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      g[j] = c < C ? (float)gr[c] : 0.f;
+      yy[j] = c < C ? (float)yr[c] : 0.f;
+      s += g[j] * yy[j];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      s += __shfl_xor(s, off);
+    T* dr = dx + r * C;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      if (c < C) dr[c] = (T)((g[j] - s) * yy[j] * invT);
+    }
+  }
+}
+
 at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode,
                        double temperature,
                        c10::optional<at::Tensor> mask) {
@@ -129,6 +215,26 @@ at::Tensor softmax_fwd(const at::Tensor& x, bool log_mode,
     mp = mc.data_ptr<unsigned char>();
   }
   int grid = (int)std::min<long>(rows, 4096);
+  if (C <= 256 && !log_mode && rows >= 64) {
+    // register-resident rows: one read, one write
+    int g4 = (int)std::min<long>((rows + 3) / 4, 4096);
+    DISPATCH_FLOAT_TYPES(x.scalar_type(), "softmax_fwd_rr", [&] {
+      if (C <= 64)
+        softmax_fwd_rowreg_kernel<scalar_t, 1><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), rows,
+            (int)C, log_mode, (float)(1.0 / temperature), mp);
+      else if (C <= 128)
+        softmax_fwd_rowreg_kernel<scalar_t, 2><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), rows,
+            (int)C, log_mode, (float)(1.0 / temperature), mp);
+      else
+        softmax_fwd_rowreg_kernel<scalar_t, 4><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)x.data_ptr(), (scalar_t*)y.data_ptr(), rows,
+            (int)C, log_mode, (float)(1.0 / temperature), mp);
+    });
+    HIP_CHECK_LAST();
+    return y;
+  }
   DISPATCH_FLOAT_TYPES(x.scalar_type(), "softmax_fwd", [&] {
     constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
     softmax_fwd_kernel<scalar_t, VEC><<<grid, 256, 0, cur_stream()>>>(
@@ -146,6 +252,28 @@ at::Tensor softmax_bwd(const at::Tensor& dy, const at::Tensor& y,
   long C = dy.size(-1), rows = dy.numel() / (C > 0 ? C : 1);
   if (dy.numel() == 0) return dx;
   int grid = (int)std::min<long>(rows, 4096);
+  if (C <= 256 && !log_mode && rows >= 64) {
+    int g4 = (int)std::min<long>((rows + 3) / 4, 4096);
+    DISPATCH_FLOAT_TYPES(dy.scalar_type(), "softmax_bwd_rr", [&] {
+      if (C <= 64)
+        softmax_bwd_rowreg_kernel<scalar_t, 1><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+            (scalar_t*)dx.data_ptr(), rows, (int)C,
+            (float)(1.0 / temperature));
+      else if (C <= 128)
+        softmax_bwd_rowreg_kernel<scalar_t, 2><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+            (scalar_t*)dx.data_ptr(), rows, (int)C,
+            (float)(1.0 / temperature));
+      else
+        softmax_bwd_rowreg_kernel<scalar_t, 4><<<g4, 256, 0, cur_stream()>>>(
+            (const scalar_t*)dy.data_ptr(), (const scalar_t*)y.data_ptr(),
+            (scalar_t*)dx.data_ptr(), rows, (int)C,
+            (float)(1.0 / temperature));
+    });
+    HIP_CHECK_LAST();
+    return dx;
+  }
   DISPATCH_FLOAT_TYPES(dy.scalar_type(), "softmax_bwd", [&] {
     constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
     softmax_bwd_kernel<scalar_t, VEC><<<grid, 256, 0, cur_stream()>>>(

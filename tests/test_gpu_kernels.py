@@ -802,3 +802,22 @@ def test_fc_backward_tn_path():
     check(w.grad, wo.grad.to(w.device), tol=4e-3)
     check(bb.grad, bo.grad.to(w.device), tol=4e-3)
     check(x.grad, xo.grad.to(w.device), tol=4e-3)
+
+
+@pytest.mark.parametrize('C', [64, 128, 200])
+def test_softmax_rowreg_masked(C):
+    """Register-resident small-C softmax (one-read path) with mask."""
+    rows = 512
+    x = mk((rows, C), seed=75, scale=3.0)
+    mask = (torch.rand(rows, C, device='cuda') > 0.3).to(torch.uint8)
+    mask[:, 0] = 1  # never fully masked
+    y = ext.softmax_fwd(x, False, 1.3, mask)
+    xm = x.float().masked_fill(mask == 0, float('-inf'))
+    want = torch.softmax(xm / 1.3, -1)
+    check(y, want, tol=3e-3)
+    dy = torch.randn_like(want)
+    x32 = xm.clone().requires_grad_(True)
+    yo = torch.softmax(x32 / 1.3, -1)
+    yo.backward(dy)
+    dx = ext.softmax_bwd(dy.to(x.dtype), y, False, 1.3)
+    check(dx, torch.nan_to_num(x32.grad), tol=4e-3)
