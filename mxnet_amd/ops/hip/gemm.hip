@@ -743,21 +743,24 @@ __device__ inline trs4g tr_read16g(const void* p) {
       (__attribute__((address_space(3))) trs4g*)(uintptr_t)p);
 }
 
-template <typename T>
+template <typename T, int BT>
 __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
     const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
     long M, long I, long J, long m_per_slice, float* __restrict__ out32,
     float* __restrict__ dbias, const T* __restrict__ zpage) {
   using Frag = typename DTraits<T>::frag8;
   using V8 = T __attribute__((ext_vector_type(8)));
-  constexpr int BI = 64, BJ = 64, BKM = 64;
-  __shared__ T AS[2][BKM * BI];
-  __shared__ T BS[2][BKM * BJ];
+  constexpr int BKM = 64;
+  constexpr int R = BT / 32;        // 16x16 fragments per wave per dim
+  constexpr int SEGS = BT / 8;      // 16 B column segments per row
+  constexpr int ROWS_PER_RND = 256 / SEGS;
+  __shared__ T AS[2][BKM * BT];
+  __shared__ T BS[2][BKM * BT];
 
-  const int nTj = (int)((J + BJ - 1) / BJ);
+  const int nTj = (int)((J + BT - 1) / BT);
   const int bid = blockIdx.x;
-  const long i0 = (long)(bid / nTj) * BI;
-  const long j0 = (long)(bid % nTj) * BJ;
+  const long i0 = (long)(bid / nTj) * BT;
+  const long j0 = (long)(bid % nTj) * BT;
 
   const long ms0 = (long)blockIdx.y * m_per_slice;
   const long ms1 = min(M, ms0 + m_per_slice);
@@ -767,18 +770,17 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
-  const int sm_half = t >> 3;
-  const int seg = t & 7;
+  const int s_row = t / SEGS;
+  const int seg = t % SEGS;
 
-  float4_t acc[2][2] = {};
+  float4_t acc[R][R] = {};
   float bias_acc[8] = {};
 
-  // stage one [64 m][64 col] panel into the permuted-row subtile layout
-  auto stage = [&](T (*dst)[BKM * BI], int buf, const T* __restrict__ src,
+  auto stage = [&](T (*dst)[BKM * BT], int buf, const T* __restrict__ src,
                    long ld, long W, long col0, long mc, bool want_bias) {
 #pragma unroll
-    for (int rnd = 0; rnd < 2; ++rnd) {
-      const int m_l = rnd * 32 + sm_half;
+    for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
+      const int m_l = rnd * ROWS_PER_RND + s_row;
       const long m_g = mc + m_l;
       V8 v = {};
       const long c = col0 + seg * 8;
@@ -790,7 +792,7 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
       const int kk = m_l >> 5, rem = m_l & 31;
       const int tt = (rem >> 2) & 1;
       const int prow = ((rem >> 3) << 2) | (rem & 3);
-      *(V8*)&dst[buf][(((kk * 2 + tt) * 4 + (seg >> 1)) << 8) +
+      *(V8*)&dst[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
                       prow * 16 + (seg & 1) * 8] = v;
     }
   };
@@ -806,61 +808,58 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
-      Frag af[2], bf[2];
+      Frag af[R], bf[R];
 #pragma unroll
-      for (int m = 0; m < 2; ++m) {
-        const int cblk = wr * 2 + m;
+      for (int m = 0; m < R; ++m) {
+        const int cblk = wr * R + m;
         union { trs4g h[2]; Frag f; } u;
-        u.h[0] = tr_read16g(&AS[buf][(((kk * 2 + 0) * 4 + cblk) << 8) +
-                                     lane * 4]);
-        u.h[1] = tr_read16g(&AS[buf][(((kk * 2 + 1) * 4 + cblk) << 8) +
-                                     lane * 4]);
+        u.h[0] = tr_read16g(&AS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        u.h[1] = tr_read16g(&AS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
         af[m] = u.f;
       }
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
-        const int cblk = wc * 2 + n;
+      for (int n = 0; n < R; ++n) {
+        const int cblk = wc * R + n;
         union { trs4g h[2]; Frag f; } u;
-        u.h[0] = tr_read16g(&BS[buf][(((kk * 2 + 0) * 4 + cblk) << 8) +
-                                     lane * 4]);
-        u.h[1] = tr_read16g(&BS[buf][(((kk * 2 + 1) * 4 + cblk) << 8) +
-                                     lane * 4]);
+        u.h[0] = tr_read16g(&BS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        u.h[1] = tr_read16g(&BS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
         bf[n] = u.f;
       }
 #pragma unroll
-      for (int m = 0; m < 2; ++m)
+      for (int m = 0; m < R; ++m)
 #pragma unroll
-        for (int n = 0; n < 2; ++n)
+        for (int n = 0; n < R; ++n)
           acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
     }
     __syncthreads();
     buf ^= 1;
   }
 
-  // bias gradient: this block staged columns [i0 + seg*8, +8) of A —
-  // every sm_half row summed its own copies, so LDS-reduce over the 32
-  // stagers, then one global atomic per column per block.  Only the
-  // j-first block column contributes (others stage the same A panel).
   if (dbias != nullptr && j0 == 0) {
-    __shared__ float bsum[64];
-    for (int i = t; i < 64; i += 256) bsum[i] = 0.f;
+    __shared__ float bsum[BT];
+    for (int i = t; i < BT; i += 256) bsum[i] = 0.f;
     __syncthreads();
 #pragma unroll
     for (int jj = 0; jj < 8; ++jj)
       atomicAdd(&bsum[seg * 8 + jj], bias_acc[jj]);
     __syncthreads();
-    if (t < 64 && i0 + t < I) atomicAdd(dbias + i0 + t, bsum[t]);
+    for (int i = t; i < BT; i += 256)
+      if (i0 + i < I) atomicAdd(dbias + i0 + i, bsum[i]);
   }
 
   const int d_col = lane & 15;
   const int d_row = (lane >> 4) * 4;
 #pragma unroll
-  for (int n = 0; n < 2; ++n) {
-    const long j = j0 + wc * 32 + n * 16 + d_col;
+  for (int n = 0; n < R; ++n) {
+    const long j = j0 + wc * (BT / 2) + n * 16 + d_col;
     if (j >= J) continue;
 #pragma unroll
-    for (int m = 0; m < 2; ++m) {
-      const long i_base = i0 + wr * 32 + m * 16 + d_row;
+    for (int m = 0; m < R; ++m) {
+      const long i_base = i0 + wr * (BT / 2) + m * 16 + d_row;
 #pragma unroll
       for (int jj = 0; jj < 4; ++jj) {
         const long i = i_base + jj;
@@ -887,7 +886,10 @@ std::vector<at::Tensor> gemm_tn_fused(const at::Tensor& A,
                         : at::empty({0}, A.options().dtype(at::kFloat));
     return {C, db};
   }
-  long nwg = ((I + 63) / 64) * ((J + 63) / 64);
+  // 128^2 tiles when both dims allow (2x the operand reuse of 64^2 —
+  // the 64^2 first cut measured ~1% SLOWER than transpose+NT on BERT)
+  int bt = (I >= 128 && J >= 128) ? 128 : 64;
+  long nwg = ((I + bt - 1) / bt) * ((J + bt - 1) / bt);
   long yb = std::max<long>(
       1, std::min<long>((M + 63) / 64, 1024 / std::max<long>(nwg, 1)));
   long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
@@ -899,12 +901,20 @@ std::vector<at::Tensor> gemm_tn_fused(const at::Tensor& A,
                       : at::Tensor();
   dim3 grid((unsigned)nwg, (unsigned)yb);
   DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_tn_tr", [&] {
-    gemm_tn_tr_kernel<scalar_t><<<grid, 256, 0, cur_stream()>>>(
-        (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
-        (scalar_t*)C.data_ptr(), M, I, J, m_per_slice,
-        yb > 1 ? o32.data_ptr<float>() : nullptr,
-        want_bias ? db.data_ptr<float>() : nullptr,
-        (const scalar_t*)zero_page(A));
+    if (bt == 128)
+      gemm_tn_tr_kernel<scalar_t, 128><<<grid, 256, 0, cur_stream()>>>(
+          (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+          (scalar_t*)C.data_ptr(), M, I, J, m_per_slice,
+          yb > 1 ? o32.data_ptr<float>() : nullptr,
+          want_bias ? db.data_ptr<float>() : nullptr,
+          (const scalar_t*)zero_page(A));
+    else
+      gemm_tn_tr_kernel<scalar_t, 64><<<grid, 256, 0, cur_stream()>>>(
+          (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
+          (scalar_t*)C.data_ptr(), M, I, J, m_per_slice,
+          yb > 1 ? o32.data_ptr<float>() : nullptr,
+          want_bias ? db.data_ptr<float>() : nullptr,
+          (const scalar_t*)zero_page(A));
   });
   HIP_CHECK_LAST();
   if (yb > 1) C.copy_(o32.to(C.scalar_type()));
