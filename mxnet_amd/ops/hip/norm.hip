@@ -841,33 +841,62 @@ __global__ void ln_bwd_dgamma_kernel(const T* __restrict__ dy,
                                      float* __restrict__ dgamma,
                                      float* __restrict__ dbeta, long rows,
                                      long C, long rows_per_block) {
-  __shared__ float l1[4][64], l2[4][64];
-  int cc = threadIdx.x & 63;
-  int rl = threadIdx.x >> 6;
-  long c = (long)blockIdx.x * 64 + cc;
-  long r0 = (long)blockIdx.y * rows_per_block;
-  long r1 = min(rows, r0 + rows_per_block);
-  float a1 = 0.f, a2 = 0.f;
-  if (c < C) {
-    for (long r = r0 + rl; r < r1; r += 4) {
-      float g = (float)dy[r * C + c];
-      float xh = ((float)x[r * C + c] - mean[r]) * istd[r];
-      a1 += g * xh;
-      a2 += g;
+  // vectorized like bn_bwd_reduce: 8 col-groups x 32 row-lanes, half8
+  // loads (the scalar-column first cut measured 10x off HBM roof)
+  using V8 = T __attribute__((ext_vector_type(8)));
+  __shared__ float l1[256][8];
+  __shared__ float l2[256][8];
+  const int t = threadIdx.x;
+  const int cg = t & 7;
+  const int rl = t >> 3;
+  const long c0 = (long)blockIdx.x * 64 + cg * 8;
+  const long r0 = (long)blockIdx.y * rows_per_block;
+  const long r1 = min(rows, r0 + rows_per_block);
+  float a1[8] = {}, a2[8] = {};
+  if (c0 + 8 <= C) {
+    for (long r = r0 + rl; r < r1; r += 32) {
+      V8 g8 = *(const V8*)(dy + r * C + c0);
+      V8 x8 = *(const V8*)(x + r * C + c0);
+      const float mu = mean[r], is = istd[r];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float g = (float)g8[j];
+        a1[j] += g * ((float)x8[j] - mu) * is;
+        a2[j] += g;
+      }
+    }
+  } else if (c0 < C) {
+    for (long r = r0 + rl; r < r1; r += 32) {
+      const float mu = mean[r], is = istd[r];
+      for (int j = 0; j < 8 && c0 + j < C; ++j) {
+        float g = (float)dy[r * C + c0 + j];
+        a1[j] += g * ((float)x[r * C + c0 + j] - mu) * is;
+        a2[j] += g;
+      }
     }
   }
-  l1[rl][cc] = a1;
-  l2[rl][cc] = a2;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    l1[t][j] = a1[j];
+    l2[t][j] = a2[j];
+  }
   __syncthreads();
-  if (rl == 0 && c < C) {
-    a1 = l1[0][cc] + l1[1][cc] + l1[2][cc] + l1[3][cc];
-    a2 = l2[0][cc] + l2[1][cc] + l2[2][cc] + l2[3][cc];
-    if (gridDim.y == 1) {
-      dgamma[c] = a1;
-      dbeta[c] = a2;
-    } else {
-      atomicAdd(dgamma + c, a1);
-      atomicAdd(dbeta + c, a2);
+  if (t < 64) {
+    const long c = (long)blockIdx.x * 64 + t;
+    if (c < C) {
+      const int g = t >> 3, j = t & 7;
+      float s1 = 0.f, s2 = 0.f;
+      for (int r = 0; r < 32; ++r) {
+        s1 += l1[r * 8 + g][j];
+        s2 += l2[r * 8 + g][j];
+      }
+      if (gridDim.y == 1) {
+        dgamma[c] = s1;
+        dbeta[c] = s2;
+      } else {
+        atomicAdd(dgamma + c, s1);
+        atomicAdd(dbeta + c, s2);
+      }
     }
   }
 }
