@@ -692,6 +692,159 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_tr_kernel(
 
 
 // ---------------------------------------------------------------------------
+// bwd-weight v4: the TN formulation done directly.  dw = dy^T @ x_tap is
+// a TN GEMM (reduction m is the stored ROW dim of both operands), so
+// both operands stage ROW-major (dy tiles are plain coalesced loads --
+// no global pre-transpose, no two-hop; x rows gathered via pixtab) into
+// permuted-row [16][16] subtiles and are consumed with
+// ds_read_b64_tr_b16 (same scheme as gemm_tn_tr_kernel, gemm.hip).
+// 128x128 output tiles (64x64 when Kg or Cg < 128).
+// ---------------------------------------------------------------------------
+template <typename T, int BT>
+__global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_tn_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
+    int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
+    int dw, long m_per_slice) {
+  using Frag = typename DTraits<T>::frag8;
+  using V8 = T __attribute__((ext_vector_type(8)));
+  constexpr int BKM = 64;
+  constexpr int RF = BT / 32;
+  constexpr int SEGS = BT / 8;
+  constexpr int ROWS_PER_RND = 256 / SEGS;
+  __shared__ T AS[2][BKM * BT];   // dy subtiles
+  __shared__ T BS[2][BKM * BT];   // gathered-x subtiles
+
+  const int g = blockIdx.z;
+  const int cpl = (Cg + BT - 1) / BT;
+  const int nTj = R * S * cpl;
+  const int bid = blockIdx.x;
+  const long i0 = (long)(bid / nTj) * BT;
+  const int jt = bid % nTj;
+  const int rs = jt / cpl;
+  const long c0 = (long)(jt % cpl) * BT;
+  const int r = rs / S, sst = rs % S;
+  const int roff = r * dh, soff = sst * dw;
+
+  const long ms0 = (long)blockIdx.y * m_per_slice;
+  const long ms1 = min(M, ms0 + m_per_slice);
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  const int s_row = t / SEGS;
+  const int seg = t % SEGS;
+
+  float4_t acc[RF][RF] = {};
+
+  auto stage_dy = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
+      const int m_l = rnd * ROWS_PER_RND + s_row;
+      const long m_g = mc + m_l;
+      V8 v = {};
+      const long i = i0 + seg * 8;
+      if (m_g < ms1 && i + 8 <= Kg)
+        v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + i);
+      const int kk = m_l >> 5, rem = m_l & 31;
+      const int tt = (rem >> 2) & 1;
+      const int prow = ((rem >> 3) << 2) | (rem & 3);
+      *(V8*)&AS[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
+                     prow * 16 + (seg & 1) * 8] = v;
+    }
+  };
+
+  auto stage_x = [&](int buf, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
+      const int m_l = rnd * ROWS_PER_RND + s_row;
+      const long m_g = mc + m_l;
+      V8 v = {};
+      if (m_g < ms1) {
+        int4_t pt = pixtab[m_g];
+        const int ih = pt[1] + roff, iw = pt[2] + soff;
+        const long cseg = c0 + seg * 8;
+        if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
+            cseg + 8 <= Cg)
+          v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
+                           (long)g * Cg + cseg);
+      }
+      const int kk = m_l >> 5, rem = m_l & 31;
+      const int tt = (rem >> 2) & 1;
+      const int prow = ((rem >> 3) << 2) | (rem & 3);
+      *(V8*)&BS[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
+                     prow * 16 + (seg & 1) * 8] = v;
+    }
+  };
+
+  stage_dy(0, ms0);
+  stage_x(0, ms0);
+  __syncthreads();
+  int buf = 0;
+  for (long mc = ms0; mc < ms1; mc += BKM) {
+    if (mc + BKM < ms1) {
+      stage_dy(buf ^ 1, mc + BKM);
+      stage_x(buf ^ 1, mc + BKM);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[RF], bf[RF];
+#pragma unroll
+      for (int m = 0; m < RF; ++m) {
+        const int cblk = wr * RF + m;
+        union { trs4 h[2]; Frag f; } u;
+        u.h[0] = tr_read16(&AS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                     << 8) + lane * 4]);
+        u.h[1] = tr_read16(&AS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                     << 8) + lane * 4]);
+        af[m] = u.f;
+      }
+#pragma unroll
+      for (int n = 0; n < RF; ++n) {
+        const int cblk = wc * RF + n;
+        union { trs4 h[2]; Frag f; } u;
+        u.h[0] = tr_read16(&BS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                     << 8) + lane * 4]);
+        u.h[1] = tr_read16(&BS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                     << 8) + lane * 4]);
+        bf[n] = u.f;
+      }
+#pragma unroll
+      for (int m = 0; m < RF; ++m)
+#pragma unroll
+        for (int n = 0; n < RF; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const long RSCg = (long)R * S * Cg;
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < RF; ++n) {
+    const long c = c0 + wc * (BT / 2) + n * 16 + d_col;
+    if (c >= Cg) continue;
+#pragma unroll
+    for (int m = 0; m < RF; ++m) {
+      const long i_base = i0 + wr * (BT / 2) + m * 16 + d_row;
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const long i = i_base + jj;
+        if (i < Kg) {
+          float* dst = dw32 + ((long)g * Kg + i) * RSCg + (long)rs * Cg + c;
+          if (gridDim.y == 1) *dst = acc[m][n][jj];
+          else atomicAdd(dst, acc[m][n][jj]);
+        }
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // small-C implicit GEMM (the RGB stem): x padded to 8 channels so one
 // BK=64 chunk = 8 horizontal taps (r fixed, s = lane group, dil=1) --
 // taps are w-contiguous in NHWC so each lane's 16 B segment is one tap.
@@ -1547,6 +1700,48 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
     auto tab = at::empty({M, 4}, dy.options().dtype(at::kInt));
     build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
+    // v4: direct TN formulation -- both operands row-major staged +
+    // tr_b16 consumed; no dy pre-transpose, no two-hop (default ON,
+    // MXNET_BWDW_V4=0 falls back to the v2.5/hop2 hybrid)
+    static const bool use_v4 = [] {
+      const char* e = getenv("MXNET_BWDW_V4");
+      return !e || e[0] != '0';
+    }();
+    if (use_v4) {
+      int bt = (Kg >= 128 && Cg >= 128) ? 128 : 64;
+      int cpl4 = (Cg + bt - 1) / bt;
+      int nwg4 = (int)(((Kg + bt - 1) / bt) * (long)R * S * cpl4);
+      long want4 = 2048;
+      long yb4 = std::max<long>(
+          1, std::min<long>((M + 63) / 64,
+                            want4 / std::max<long>((long)nwg4 * groups, 1)));
+      long mps4 = ((M + yb4 - 1) / yb4 + 63) / 64 * 64;
+      yb4 = (M + mps4 - 1) / mps4;
+      auto dw32 = yb4 == 1
+          ? at::empty({(long)Kout, (long)R * S * Cg},
+                      dy.options().dtype(at::kFloat))
+          : at::zeros({(long)Kout, (long)R * S * Cg},
+                      dy.options().dtype(at::kFloat));
+      dim3 grid4((unsigned)nwg4, (unsigned)yb4, (unsigned)groups);
+      DISPATCH_HALF_TYPES(dy.scalar_type(), "conv_bwd_w_tn", [&] {
+        if (bt == 128)
+          conv_bwd_w_igemm_tn_kernel<scalar_t, 128>
+              <<<grid4, 256, 0, cur_stream()>>>(
+                  (const scalar_t*)dy.data_ptr(),
+                  (const scalar_t*)x.data_ptr(),
+                  (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M,
+                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4);
+        else
+          conv_bwd_w_igemm_tn_kernel<scalar_t, 64>
+              <<<grid4, 256, 0, cur_stream()>>>(
+                  (const scalar_t*)dy.data_ptr(),
+                  (const scalar_t*)x.data_ptr(),
+                  (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M,
+                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4);
+      });
+      HIP_CHECK_LAST();
+      return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());
+    }
     int nj = 1;  // NJ=2 measured net-negative (L3 absorbs the
                  // dy re-reads; the kernel is LDS-pipeline bound)
     int cpl = (Cg + 64 * nj - 1) / (64 * nj);
