@@ -1707,7 +1707,12 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
       const char* e = getenv("MXNET_BWDW_V4");
       return !e || e[0] != '0';
     }();
-    if (use_v4) {
+    // A/B at batch 256 (profiles/r01_summary.md): the TN-direct kernel
+    // wins on the 64-wide tiles (l1_3x3 330->297 us) but LOSES to the
+    // tuned v2.5/hop2 hybrid on 128x128-tile deep layers (tr-read bank
+    // conflicts vs their XOR-swizzled b128 reads) -- so it runs only
+    // where a 64-tile would be used.
+    if (use_v4 && (Kg < 128 || Cg < 128)) {
       int bt = (Kg >= 128 && Cg >= 128) ? 128 : 64;
       int cpl4 = (Cg + bt - 1) / bt;
       int nwg4 = (int)(((Kg + bt - 1) / bt) * (long)R * S * cpl4);
