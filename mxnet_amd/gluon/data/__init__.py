@@ -4,3 +4,5 @@ from .sampler import Sampler, SequentialSampler, RandomSampler, BatchSampler
 from .dataloader import DataLoader
 from . import vision
 from . import batchify
+
+from .threaded_loader import ThreadedDataLoader  # noqa: F401

@@ -37,6 +37,12 @@ setup(
             cxx_std=17,
             extra_compile_args=['-O2', '-pthread'],
         ),
+        Pybind11Extension(
+            'mxnet_amd._dataloader',
+            [os.path.join(ROOT, 'src', 'dataloader.cc')],
+            cxx_std=17,
+            extra_compile_args=['-O2', '-pthread'],
+        ),
     ],
     cmdclass={'build_ext': BuildExtension.with_options(use_ninja=True)},
 )
