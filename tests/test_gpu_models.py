@@ -133,3 +133,32 @@ def test_export_symbolblock_serve_gpu(tmp_path):
     out = served(x).asnumpy()
     np.testing.assert_allclose(out.astype(np.float32),
                                ref.astype(np.float32), rtol=2e-2, atol=2e-2)
+
+
+@pytest.mark.gpu
+def test_embedding_sparse_grad_gpu():
+    """Sparse-grad Embedding + lazy SGD on device tensors."""
+    import torch
+    import mxnet_amd as mx
+    from mxnet_amd import autograd
+    from mxnet_amd.gluon import Trainer, nn as gnn
+    torch.manual_seed(2)
+    emb = gnn.Embedding(1000, 64, sparse_grad=True)
+    emb.initialize(ctx=mx.gpu(0))
+    tr = Trainer(emb.collect_params(), 'sgd',
+                 {'learning_rate': 0.1, 'momentum': 0.9}, kvstore=None)
+    w0 = emb.weight.data().handle.clone()
+    idx = mx.nd.from_torch(torch.randint(0, 50, (8, 16), device='cuda'))
+    for _ in range(2):
+        with autograd.record():
+            out = emb(idx)
+            L = mx.nd.from_torch((out.handle.float() ** 2).sum())
+        L.backward()
+        tr.step(1)
+    w1 = emb.weight.data().handle
+    touched = torch.unique(idx.handle)
+    moved = (w1 - w0).abs().sum(dim=1)
+    assert (moved[touched] > 0).all()
+    untouched = torch.ones(1000, dtype=torch.bool, device='cuda')
+    untouched[touched] = False
+    assert moved[untouched].abs().max() == 0
