@@ -548,8 +548,9 @@ std::vector<at::Tensor> bn_nhwc_fwd_train(const at::Tensor& x,
     sum = *presums;           // sliced workspace, finalize folds it
     sumsq = sum;              // unused in the sliced path
   } else {
-    sum = at::zeros({C}, opts);
-    sumsq = at::zeros({C}, opts);
+    auto ws = at::zeros({2, C}, opts);  // one fill for both
+    sum = ws[0];
+    sumsq = ws[1];
   }
   auto save_mean = at::empty({C}, opts), save_istd = at::empty({C}, opts);
   auto scale = at::empty({C}, opts), shift = at::empty({C}, opts);
@@ -679,7 +680,8 @@ std::vector<at::Tensor> bn_nhwc_bwd(const at::Tensor& dy, const at::Tensor& x,
   auto dx = at::empty_like(x);
   auto dres = has_res ? at::empty_like(x) : at::empty({0}, x.options());
   auto opts = x.options().dtype(at::kFloat);
-  auto s1 = at::zeros({C}, opts), s2 = at::zeros({C}, opts);
+  auto ws12 = at::zeros({2, C}, opts);
+  auto s1 = ws12[0], s2 = ws12[1];
   auto g32 = gamma.to(at::kFloat).contiguous();
   long rpb;
   dim3 grid = bn_reduce_grid(M, C, &rpb);
@@ -932,7 +934,8 @@ std::vector<at::Tensor> layernorm_bwd(const at::Tensor& dy,
   long C = x.size(-1), rows = x.numel() / C;
   auto dx = at::empty_like(x);
   auto opts = x.options().dtype(at::kFloat);
-  auto dg = at::zeros({C}, opts), db = at::zeros({C}, opts);
+  auto wsgb = at::zeros({2, C}, opts);
+  auto dg = wsgb[0], db = wsgb[1];
   auto g32 = gamma.to(at::kFloat).contiguous();
   int grid = (int)std::min<long>(rows, 4096);
   long xb = (C + 63) / 64;
