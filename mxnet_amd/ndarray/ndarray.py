@@ -411,10 +411,31 @@ def waitall():
 
 
 def save(fname, data):
+    """Save NDArray(s); ``.npy``/``.npz`` extensions use the numpy
+    formats (reference src/serialization/cnpy.cc), anything else the
+    reference ``.params`` byte format."""
+    import numpy as _np
+    if str(fname).endswith('.npy'):
+        t = data[0] if isinstance(data, (list, tuple)) else data
+        _np.save(fname, t.asnumpy())
+        return
+    if str(fname).endswith('.npz'):
+        if isinstance(data, dict):
+            _np.savez(fname, **{k: v.asnumpy() for k, v in data.items()})
+        else:
+            arrs = data if isinstance(data, (list, tuple)) else [data]
+            _np.savez(fname, *[a.asnumpy() for a in arrs])
+        return
     from ..utils import serialization
     serialization.save_ndarrays(fname, data)
 
 
 def load(fname):
+    import numpy as _np
+    if str(fname).endswith('.npy'):
+        return [NDArray(torch.from_numpy(_np.load(fname)))]
+    if str(fname).endswith('.npz'):
+        z = _np.load(fname)
+        return {k: NDArray(torch.from_numpy(z[k])) for k in z.files}
     from ..utils import serialization
     return serialization.load_ndarrays(fname)
