@@ -291,6 +291,7 @@ class Trainer:
                     g.index_add_(0, rs.indices, rs.data.to(g.dtype))
 
     def _update(self, ignore_stale_grad=False):
+        to_zero = []
         for i, p in enumerate(self._params):
             self._check_states(i, p)
             rs = self._pop_rowsparse(p)
@@ -312,7 +313,13 @@ class Trainer:
             # (torch accumulates, the reference overwrites — clearing here
             # restores reference behavior; 'add' keeps accumulating)
             if p.grad_req == 'write':
-                p.zero_grad()
+                to_zero.extend(g._t for g in (p._grad or {}).values()
+                               if g._t is not None)
+        if to_zero:
+            # one fused launch instead of a fill per parameter
+            # (161 x ~3.7 us/step measured in the final profile)
+            with torch.no_grad():
+                torch._foreach_zero_(to_zero)
 
     def _try_fused_update(self):
         """One multi-tensor kernel updates every parameter (reference
