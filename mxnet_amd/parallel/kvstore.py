@@ -192,7 +192,9 @@ class DistKVStore(KVStoreBase):
         self._type = kind
         self._handles = []
         if not dist.is_initialized():
-            backend = 'nccl' if torch.cuda.is_available() else 'gloo'
+            # composite default (cuda:nccl + cpu:gloo) so CPU tensors on
+            # a GPU box still have a backend (RCCL handles cuda ones)
+            backend = None if torch.cuda.is_available() else 'gloo'
             if 'RANK' in os.environ:
                 dist.init_process_group(backend=backend)
             else:
