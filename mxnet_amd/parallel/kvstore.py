@@ -192,9 +192,10 @@ class DistKVStore(KVStoreBase):
         self._type = kind
         self._handles = []
         if not dist.is_initialized():
-            # composite default (cuda:nccl + cpu:gloo) so CPU tensors on
-            # a GPU box still have a backend (RCCL handles cuda ones)
-            backend = None if torch.cuda.is_available() else 'gloo'
+            # composite (cuda:nccl + cpu:gloo) so CPU tensors on a GPU
+            # box still have a backend (RCCL handles the cuda ones)
+            backend = 'cuda:nccl,cpu:gloo' if torch.cuda.is_available() \
+                else 'gloo'
             if 'RANK' in os.environ:
                 dist.init_process_group(backend=backend)
             else:
