@@ -55,11 +55,17 @@ class _FullyConnected(torch.autograd.Function):
         if use_hip(x):
             ext = hip_required('fully_connected')
             dx = ext.gemm_nn(dy, w) if need_dx else None
-            if need_dw:
+            if need_dw and dy.shape[0] >= 1024:
                 # direct TN kernel (tr_b16 operand reads, no transpose
-                # passes) with the bias gradient fused into the A tiles
+                # passes) with the bias gradient fused into the A tiles.
+                # Small reduction dims (LSTM per-step M=128) measured
+                # SLOWER on it than transpose+NT (prologue-dominated:
+                # 2 k-chunks/block) -- hence the M gate.
                 dw, db32 = ext.gemm_tn_fused(dy, x, ctx.has_bias)
                 db = db32.to(dy.dtype) if ctx.has_bias else None
+            elif need_dw:
+                dw = ext.gemm_tn(dy, x)
+                db = ext.colsum(dy) if ctx.has_bias else None
             else:
                 dw = None
                 db = ext.colsum(dy) if ctx.has_bias else None
