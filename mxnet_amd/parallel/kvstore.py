@@ -204,7 +204,9 @@ class DistKVStore(KVStoreBase):
                 os.environ.setdefault('MASTER_PORT', '29741')
                 dist.init_process_group(backend=backend, rank=0, world_size=1)
         if torch.cuda.is_available():
-            torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)))
+            # clamp for CPU-tensor tests running world>1 on a 1-GPU box
+            torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)) %
+                                  max(torch.cuda.device_count(), 1))
         self._comm_stream = (torch.cuda.Stream()
                              if torch.cuda.is_available() else None)
         self._compression = None
