@@ -48,7 +48,7 @@ struct GemmLd {
   long sAh, sBh, sCh;   // inner-batch element strides
 };
 
-template <typename T, int BN = 128>
+template <typename T, int BN = 128, bool WITH_STATS = false>
 __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
     const T* __restrict__ A, const T* __restrict__ B,
     const float* __restrict__ bias, T* __restrict__ C, long M, long N, long K,
@@ -172,8 +172,9 @@ __global__ __launch_bounds__(256, 2) void gemm_nt_mfma_kernel(
   // optional fused per-channel sum/ssq of the raw fp32 accumulators
   // (BatchNorm's forward reduction -- saves re-reading y; layout
   // [64 slices][2][N], slice = bid&63 to spread the atomic traffic)
-  __shared__ float s_st[2][BN];
-  if (stats) {
+  [[maybe_unused]] __shared__ float s_st[WITH_STATS ? 2 : 1]
+                                        [WITH_STATS ? BN : 1];
+  if (WITH_STATS && stats) {
     for (int i = t; i < 2 * BN; i += 256) s_st[i / BN][i % BN] = 0.f;
     __syncthreads();
 #pragma unroll
@@ -641,8 +642,9 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   size_t lds_bytes = (size_t)nbuf * (128 + (narrow ? 64 : 128)) * 64 * 2;
   dim3 grid((unsigned)nwg, (unsigned)nb, (unsigned)ksplit);
   DISPATCH_HALF_TYPES(A.scalar_type(), "gemm_nt", [&] {
-    if (narrow)
-      gemm_nt_mfma_kernel<scalar_t, 64>
+    auto launch = [&](auto narrow_c, auto stats_c) {
+      constexpr int BNv = decltype(narrow_c)::value ? 64 : 128;
+      gemm_nt_mfma_kernel<scalar_t, BNv, decltype(stats_c)::value>
           <<<grid, 256, lds_bytes, cur_stream()>>>(
               (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
               bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
@@ -650,15 +652,12 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
               ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
               GemmLd{K, K, N, 1, 0, 0, 0},
               ksplit > 1 ? nullptr : stats_ptr);
-    else
-      gemm_nt_mfma_kernel<scalar_t, 128>
-          <<<grid, 256, lds_bytes, cur_stream()>>>(
-              (const scalar_t*)A.data_ptr(), (const scalar_t*)B.data_ptr(),
-              bias_ptr, (scalar_t*)out.data_ptr(), M, N, K, sA, sB, sC,
-              (const scalar_t*)zero_page(A), relu,
-              ksplit > 1 ? out32.data_ptr<float>() : nullptr, tps, nbuf,
-              GemmLd{K, K, N, 1, 0, 0, 0},
-              ksplit > 1 ? nullptr : stats_ptr);
+    };
+    bool want_stats = stats_ptr != nullptr && ksplit == 1;
+    if (narrow && want_stats) launch(std::true_type{}, std::true_type{});
+    else if (narrow) launch(std::true_type{}, std::false_type{});
+    else if (want_stats) launch(std::false_type{}, std::true_type{});
+    else launch(std::false_type{}, std::false_type{});
   });
   HIP_CHECK_LAST();
   if (ksplit > 1) {
