@@ -73,3 +73,37 @@ def test_lstm_model():
         L = NDArray(out.handle.float().mean())
     L.backward()
     assert out.shape == (5, 3, 32)
+
+
+def test_lenet_mnist_cpu_local_kvstore():
+    """BASELINE config 1: Gluon LeNet on MNIST-shaped synthetic tensors,
+    ctx=cpu, KVStore='local' — loss decreases over a few steps."""
+    import torch
+    import mxnet_amd as mx
+    from mxnet_amd import autograd
+    from mxnet_amd.gluon import Trainer, nn
+    from mxnet_amd.gluon.loss import SoftmaxCrossEntropyLoss
+    torch.manual_seed(0)
+    net = nn.HybridSequential()
+    net.add(nn.Conv2D(6, 5, padding=2, activation='relu', in_channels=1),
+            nn.MaxPool2D(2),
+            nn.Conv2D(16, 5, activation='relu', in_channels=6),
+            nn.MaxPool2D(2),
+            nn.Flatten(),
+            nn.Dense(120, activation='relu', in_units=16 * 5 * 5),
+            nn.Dense(84, activation='relu', in_units=120),
+            nn.Dense(10, in_units=84))
+    net.initialize(ctx=mx.cpu())
+    tr = Trainer(net.collect_params(), 'sgd',
+                 {'learning_rate': 0.05, 'momentum': 0.9}, kvstore='local')
+    loss_fn = SoftmaxCrossEntropyLoss()
+    X = mx.nd.from_torch(torch.randn(32, 1, 28, 28))
+    Y = mx.nd.from_torch(torch.randint(0, 10, (32,)))
+    losses = []
+    for _ in range(8):
+        with autograd.record():
+            L = loss_fn(net(X), Y)
+        L.backward()
+        tr.step(32)
+        losses.append(float(L.handle.mean()))
+    assert losses[-1] < losses[0] * 0.8, losses
