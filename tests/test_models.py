@@ -100,10 +100,11 @@ def test_lenet_mnist_cpu_local_kvstore():
     X = mx.nd.from_torch(torch.randn(32, 1, 28, 28))
     Y = mx.nd.from_torch(torch.randint(0, 10, (32,)))
     losses = []
-    for _ in range(8):
+    for _ in range(25):
         with autograd.record():
             L = loss_fn(net(X), Y)
         L.backward()
         tr.step(32)
         losses.append(float(L.handle.mean()))
-    assert losses[-1] < losses[0] * 0.8, losses
+    # memorizing one batch: the loss must fall decisively
+    assert losses[-1] < losses[0] * 0.5, (losses[0], losses[-1])
