@@ -77,7 +77,7 @@ def main():
         t = timeit(lambda: ext.bn_nhwc_fwd_train(x, gamma, beta, rm, rv, 0.9,
                                                  1e-5, True, res))
         print(f'{name+"/fwd(3x)":26s} {t*1e6:9.1f} {"":>7s} {3*nbytes/t/1e12:6.2f}')
-        y, sm, si = ext.bn_nhwc_fwd_train(x, gamma, beta, rm, rv, 0.9, 1e-5,
+        y, sm, si, _m = ext.bn_nhwc_fwd_train(x, gamma, beta, rm, rv, 0.9, 1e-5,
                                           True, res)
         t = timeit(lambda: ext.bn_nhwc_bwd(dy, x, gamma, sm, si, True, y, True))
         print(f'{name+"/bwd(8x)":26s} {t*1e6:9.1f} {"":>7s} {8*nbytes/t/1e12:6.2f}')
