@@ -55,7 +55,7 @@ class _FullyConnected(torch.autograd.Function):
         if use_hip(x):
             ext = hip_required('fully_connected')
             dx = ext.gemm_nn(dy, w) if need_dx else None
-            if need_dw and dy.shape[0] >= 1024:
+            if need_dw and dy.shape[0] >= 1024 and _fc_tn_enabled():
                 # direct TN kernel (tr_b16 operand reads, no transpose
                 # passes) with the bias gradient fused into the A tiles.
                 # Small reduction dims (LSTM per-step M=128) measured
@@ -109,6 +109,12 @@ def batch_dot(a, b, transpose_a=False, transpose_b=False):
 # ---------------------------------------------------------------------------
 
 import functools
+
+
+@functools.lru_cache(None)
+def _fc_tn_enabled():
+    import os
+    return os.environ.get('MXNET_FC_TN', '1') != '0'
 
 
 @functools.lru_cache(None)
