@@ -705,7 +705,7 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_tn_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const int4_t* __restrict__ pixtab, float* __restrict__ dw32, long M,
     int H, int W, int C, int Kout, int Cg, int Kg, int R, int S, int dh,
-    int dw, long m_per_slice) {
+    int dw, long m_per_slice, const T* __restrict__ zpage) {
   using Frag = typename DTraits<T>::frag8;
   using V8 = T __attribute__((ext_vector_type(8)));
   constexpr int BKM = 64;
@@ -734,48 +734,56 @@ __global__ __launch_bounds__(256, 2) void conv_bwd_w_igemm_tn_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
-  const int s_row = t / SEGS;
-  const int seg = t % SEGS;
-
   float4_t acc[RF][RF] = {};
+
+  // source-permuted async staging (see gemm_tn_tr_kernel): lane t's
+  // global row is chosen so the wave-linear global_load_lds destination
+  // IS the tr-read subtile layout
+  const int prow_c = (t >> 1) & 15;
+  const int lsb_c = t & 1;
+  const int shi_c = t >> 5;
+
+  auto src_row = [&](int rnd, int* segv) {
+    const int sgl = shi_c + rnd * 8;
+    const int seg_hi = sgl % (SEGS / 2);
+    const int kk2tt = sgl / (SEGS / 2);
+    const int rem = ((prow_c >> 2) << 3) | ((kk2tt & 1) << 2) |
+                    (prow_c & 3);
+    *segv = seg_hi * 2 + lsb_c;
+    return (kk2tt >> 1) * 32 + rem;
+  };
 
   auto stage_dy = [&](int buf, long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
-      const int m_l = rnd * ROWS_PER_RND + s_row;
+      int segv;
+      const int m_l = src_row(rnd, &segv);
       const long m_g = mc + m_l;
-      V8 v = {};
-      const long i = i0 + seg * 8;
-      if (m_g < ms1 && i + 8 <= Kg)
-        v = *(const V8*)(dy + m_g * Kout + (long)g * Kg + i);
-      const int kk = m_l >> 5, rem = m_l & 31;
-      const int tt = (rem >> 2) & 1;
-      const int prow = ((rem >> 3) << 2) | (rem & 3);
-      *(V8*)&AS[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
-                     prow * 16 + (seg & 1) * 8] = v;
+      const long i = i0 + segv * 8;
+      const T* ga = (m_g < ms1 && i + 8 <= Kg)
+          ? dy + m_g * Kout + (long)g * Kg + i
+          : zpage;
+      gload_lds16c(ga, &AS[buf][(rnd * 256 + t) * 8]);
     }
   };
 
   auto stage_x = [&](int buf, long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
-      const int m_l = rnd * ROWS_PER_RND + s_row;
+      int segv;
+      const int m_l = src_row(rnd, &segv);
       const long m_g = mc + m_l;
-      V8 v = {};
+      const T* ga = zpage;
       if (m_g < ms1) {
         int4_t pt = pixtab[m_g];
         const int ih = pt[1] + roff, iw = pt[2] + soff;
-        const long cseg = c0 + seg * 8;
+        const long cseg = c0 + segv * 8;
         if (pt[3] && ih >= 0 && ih < H && iw >= 0 && iw < W &&
             cseg + 8 <= Cg)
-          v = *(const V8*)(x + (((long)pt[0] * H + ih) * W + iw) * C +
-                           (long)g * Cg + cseg);
+          ga = x + (((long)pt[0] * H + ih) * W + iw) * C +
+               (long)g * Cg + cseg;
       }
-      const int kk = m_l >> 5, rem = m_l & 31;
-      const int tt = (rem >> 2) & 1;
-      const int prow = ((rem >> 3) << 2) | (rem & 3);
-      *(V8*)&BS[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
-                     prow * 16 + (seg & 1) * 8] = v;
+      gload_lds16c(ga, &BS[buf][(rnd * 256 + t) * 8]);
     }
   };
 
@@ -1735,14 +1743,16 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
                   (const scalar_t*)dy.data_ptr(),
                   (const scalar_t*)x.data_ptr(),
                   (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M,
-                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4);
+                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4,
+                  (const scalar_t*)zero_page(dy));
         else
           conv_bwd_w_igemm_tn_kernel<scalar_t, 64>
               <<<grid4, 256, 0, cur_stream()>>>(
                   (const scalar_t*)dy.data_ptr(),
                   (const scalar_t*)x.data_ptr(),
                   (const int4_t*)tab.data_ptr(), dw32.data_ptr<float>(), M,
-                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4);
+                  H, W, C, Kout, Cg, Kg, R, S, dh, dw, mps4,
+                  (const scalar_t*)zero_page(dy));
       });
       HIP_CHECK_LAST();
       return dw32.view({(long)Kout, R, S, (long)Cg}).to(dy.scalar_type());

@@ -769,41 +769,42 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
   const int wid = t >> 6;
   const int wr = wid >> 1, wc = wid & 1;
 
-  const int s_row = t / SEGS;
-  const int seg = t % SEGS;
-
   float4_t acc[R][R] = {};
-  float bias_acc[8] = {};
+  float bias_acc[R] = {};
 
+  // async staging straight into the tr-read subtile layout: the row
+  // permutation is applied at the SOURCE (which global row lane t
+  // loads), so the LDS destination stays wave-linear as
+  // global_load_lds requires -- no VGPR round-trip, no ds_writes.
+  const int prow_c = (t >> 1) & 15;
+  const int lsb_c = t & 1;
+  const int shi_c = t >> 5;  // subtile within the rnd (8 per rnd)
   auto stage = [&](T (*dst)[BKM * BT], int buf, const T* __restrict__ src,
-                   long ld, long W, long col0, long mc, bool want_bias) {
+                   long ld, long W, long col0, long mc) {
 #pragma unroll
     for (int rnd = 0; rnd < BKM / ROWS_PER_RND; ++rnd) {
-      const int m_l = rnd * ROWS_PER_RND + s_row;
+      const int sgl = shi_c + rnd * 8;
+      const int seg_hi = sgl % (SEGS / 2);
+      const int kk2tt = sgl / (SEGS / 2);
+      const int rem = ((prow_c >> 2) << 3) | ((kk2tt & 1) << 2) |
+                      (prow_c & 3);
+      const int m_l = (kk2tt >> 1) * 32 + rem;
+      const int segv = seg_hi * 2 + lsb_c;
       const long m_g = mc + m_l;
-      V8 v = {};
-      const long c = col0 + seg * 8;
-      if (m_g < ms1 && c + 8 <= W)
-        v = *(const V8*)(src + m_g * ld + c);
-      if (want_bias)
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj) bias_acc[jj] += (float)v[jj];
-      const int kk = m_l >> 5, rem = m_l & 31;
-      const int tt = (rem >> 2) & 1;
-      const int prow = ((rem >> 3) << 2) | (rem & 3);
-      *(V8*)&dst[buf][(((kk * 2 + tt) * (SEGS / 2) + (seg >> 1)) << 8) +
-                      prow * 16 + (seg & 1) * 8] = v;
+      const long c = col0 + segv * 8;
+      const T* ga = (m_g < ms1 && c + 8 <= W) ? src + m_g * ld + c : zpage;
+      gload_lds16(ga, &dst[buf][(rnd * 256 + t) * 8]);
     }
   };
 
-  stage(AS, 0, A, I, I, i0, ms0, dbias != nullptr);
-  stage(BS, 0, B, J, J, j0, ms0, false);
+  stage(AS, 0, A, I, I, i0, ms0);
+  stage(BS, 0, B, J, J, j0, ms0);
   __syncthreads();
   int buf = 0;
   for (long mc = ms0; mc < ms1; mc += BKM) {
     if (mc + BKM < ms1) {
-      stage(AS, buf ^ 1, A, I, I, i0, mc + BKM, dbias != nullptr);
-      stage(BS, buf ^ 1, B, J, J, j0, mc + BKM, false);
+      stage(AS, buf ^ 1, A, I, I, i0, mc + BKM);
+      stage(BS, buf ^ 1, B, J, J, j0, mc + BKM);
     }
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {
@@ -817,6 +818,10 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
         u.h[1] = tr_read16g(&AS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
                                       << 8) + lane * 4]);
         af[m] = u.f;
+        if (dbias != nullptr && wc == 0)
+#pragma unroll
+          for (int jj = 0; jj < 8; ++jj)
+            bias_acc[m] += (float)af[m][jj];
       }
 #pragma unroll
       for (int n = 0; n < R; ++n) {
@@ -839,12 +844,16 @@ __global__ __launch_bounds__(256, 2) void gemm_tn_tr_kernel(
   }
 
   if (dbias != nullptr && j0 == 0) {
+    // wc==0 waves each saw columns cblk = wr*R + m, col = lane&15;
+    // zero-padded staging keeps OOB contributions at 0
     __shared__ float bsum[BT];
     for (int i = t; i < BT; i += 256) bsum[i] = 0.f;
     __syncthreads();
+    if (wc == 0) {
 #pragma unroll
-    for (int jj = 0; jj < 8; ++jj)
-      atomicAdd(&bsum[seg * 8 + jj], bias_acc[jj]);
+      for (int m = 0; m < R; ++m)
+        atomicAdd(&bsum[(wr * R + m) * 16 + (lane & 15)], bias_acc[m]);
+    }
     __syncthreads();
     for (int i = t; i < BT; i += 256)
       if (i0 + i < I) atomicAdd(dbias + i0 + i, bsum[i]);
