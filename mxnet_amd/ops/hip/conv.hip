@@ -1720,9 +1720,11 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
     // tuned v2.5/hop2 hybrid on 128x128-tile deep layers (tr-read bank
     // conflicts vs their XOR-swizzled b128 reads) -- so it runs only
     // where a 64-tile would be used.
+    // with async gload_lds staging v4 wins or ties on every measured
+    // shape (kbALL vs kb49 logs), so it now covers all tiles by default
     static const bool v4_all = [] {
       const char* e = getenv("MXNET_BWDW_V4_ALL");
-      return e && e[0] == '1';
+      return !e || e[0] != '0';
     }();
     if (use_v4 && (v4_all || Kg < 128 || Cg < 128)) {
       int bt = (Kg >= 128 && Cg >= 128) ? 128 : 64;
