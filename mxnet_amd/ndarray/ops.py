@@ -574,3 +574,129 @@ def smooth_l1(data, scalar=1.0, **kwargs):
     absd = t.abs()
     y = torch.where(absd < 1.0 / s2, 0.5 * s2 * t * t, absd - 0.5 / s2)
     return NDArray(y)
+
+
+# ---------------------------------------------------------------------------
+# tensor-op tail (reference tensor/matrix_op, spatial ops, special math) —
+# library-backed, same dispatch as the rest of the namespace
+# ---------------------------------------------------------------------------
+
+def moments(data, axes=None, keepdims=False, **kwargs):
+    """(mean, var) pair (reference nn/moments.cc)."""
+    t = _t(data)
+    dims = list(axes) if axes is not None else list(range(t.dim()))
+    mean = t.mean(dim=dims, keepdim=keepdims)
+    var = t.var(dim=dims, unbiased=False, keepdim=keepdims)
+    return NDArray(mean), NDArray(var)
+
+
+def SwapAxis(data, dim1=0, dim2=0, **kwargs):
+    return NDArray(_t(data).transpose(dim1, dim2).contiguous())
+
+
+swapaxes = SwapAxis
+
+
+def depth_to_space(data, block_size, **kwargs):
+    import torch.nn.functional as TF
+    return NDArray(TF.pixel_shuffle(_t(data), block_size))
+
+
+def space_to_depth(data, block_size, **kwargs):
+    import torch.nn.functional as TF
+    return NDArray(TF.pixel_unshuffle(_t(data), block_size))
+
+
+def cumsum(a, axis=None, **kwargs):
+    t = _t(a)
+    if axis is None:
+        return NDArray(t.reshape(-1).cumsum(0))
+    return NDArray(t.cumsum(axis))
+
+
+def cumprod(a, axis=None, **kwargs):
+    t = _t(a)
+    if axis is None:
+        return NDArray(t.reshape(-1).cumprod(0))
+    return NDArray(t.cumprod(axis))
+
+
+def diag(data, k=0, **kwargs):
+    t = _t(data)
+    return NDArray(torch.diagonal(t, offset=k).contiguous() if t.dim() >= 2
+                   else torch.diag(t, k))
+
+
+def trace(data, offset=0, **kwargs):
+    return NDArray(torch.diagonal(_t(data), offset=offset).sum(-1))
+
+
+def meshgrid(*arrays, indexing='xy', **kwargs):
+    outs = torch.meshgrid(*[_t(a) for a in arrays], indexing=indexing)
+    return [NDArray(o.contiguous()) for o in outs]
+
+
+def searchsorted(sorted_sequence, values, right=False, **kwargs):
+    return NDArray(torch.searchsorted(_t(sorted_sequence), _t(values),
+                                      right=right))
+
+
+def bincount(x, weights=None, minlength=0, **kwargs):
+    return NDArray(torch.bincount(
+        _t(x).long(), _t(weights) if weights is not None else None,
+        minlength))
+
+
+def digamma(data, **kwargs):
+    return NDArray(torch.digamma(_t(data)))
+
+
+def ravel_multi_index(data, shape, **kwargs):
+    t = _t(data).long()  # [ndim, n]
+    strides = []
+    acc = 1
+    for s in reversed(shape):
+        strides.append(acc)
+        acc *= s
+    strides = torch.tensor(list(reversed(strides)), device=t.device)
+    return NDArray((t * strides[:, None]).sum(0))
+
+
+def unravel_index(data, shape, **kwargs):
+    t = _t(data).long()
+    out = []
+    for s in reversed(shape):
+        out.append(t % s)
+        t = t // s
+    return NDArray(torch.stack(list(reversed(out)), dim=0))
+
+
+def GridGenerator(data, transform_type='affine', target_shape=None, **kwargs):
+    """Affine sampling grid (reference spatial_transformer GridGenerator)."""
+    import torch.nn.functional as TF
+    t = _t(data)
+    H, W = target_shape
+    theta = t.reshape(-1, 2, 3).float()
+    grid = TF.affine_grid(theta, [theta.shape[0], 1, H, W],
+                          align_corners=False)
+    # reference layout: [N, 2, H, W] (x, y)
+    return NDArray(grid.permute(0, 3, 1, 2).contiguous())
+
+
+def BilinearSampler(data, grid, **kwargs):
+    """Sample data at grid locations (reference bilinear_sampler.cc)."""
+    import torch.nn.functional as TF
+    t = _t(data)
+    g = _t(grid)
+    g4 = g.permute(0, 2, 3, 1) if g.shape[1] == 2 else g
+    return NDArray(TF.grid_sample(t.float(), g4.float(), mode='bilinear',
+                                  align_corners=False).to(t.dtype))
+
+
+def SpatialTransformer(data, loc, target_shape=None,
+                       transform_type='affine',
+                       sampler_type='bilinear', **kwargs):
+    """Affine spatial transformer = GridGenerator + BilinearSampler
+    (reference spatial_transformer.cc)."""
+    grid = GridGenerator(loc, transform_type, target_shape)
+    return BilinearSampler(data, grid)

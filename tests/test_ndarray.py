@@ -195,3 +195,34 @@ def test_fp64_op_coverage():
     net = gnn.Dense(4, in_units=6, dtype='float64')
     net.initialize()
     assert net(x64).handle.dtype == torch.float64
+
+
+def test_tensor_op_tail():
+    """moments/cumsum/diag/trace/meshgrid/spatial-transformer and friends
+    (reference tensor/matrix_op + spatial_transformer.cc)."""
+    import torch
+    x = mx.nd.from_torch(torch.arange(12.).reshape(3, 4))
+    m, v = mx.nd.moments(x, axes=[0])
+    np.testing.assert_allclose(m.asnumpy(),
+                               x.asnumpy().mean(0), rtol=1e-6)
+    assert float(mx.nd.cumsum(x, axis=1).handle[0, -1]) == 6
+    assert float(mx.nd.trace(x).handle) == 15
+    g = mx.nd.meshgrid(mx.nd.from_torch(torch.arange(3.)),
+                       mx.nd.from_torch(torch.arange(2.)))
+    ref = np.meshgrid(np.arange(3.), np.arange(2.))
+    np.testing.assert_allclose(g[0].asnumpy(), ref[0])
+    idx = mx.nd.unravel_index(mx.nd.from_torch(torch.tensor([5, 11])),
+                              (3, 4))
+    assert idx.handle.tolist() == [[1, 2], [1, 3]]
+    assert mx.nd.ravel_multi_index(idx, (3, 4)).handle.tolist() == [5, 11]
+    assert mx.nd.bincount(
+        mx.nd.from_torch(torch.tensor([0, 1, 1, 3]))).handle.tolist() \
+        == [1, 2, 0, 1]
+    # spatial transformer with identity affine returns the input
+    data = mx.nd.from_torch(torch.randn(2, 3, 8, 8))
+    theta = mx.nd.from_torch(torch.tensor([[1., 0, 0, 0, 1, 0]] * 2))
+    out = mx.nd.SpatialTransformer(data, theta, target_shape=(8, 8))
+    np.testing.assert_allclose(out.asnumpy(), data.asnumpy(), atol=1e-4)
+    s2d = mx.nd.space_to_depth(
+        mx.nd.from_torch(torch.arange(16.).reshape(1, 1, 4, 4)), 2)
+    assert s2d.shape == (1, 4, 2, 2)
