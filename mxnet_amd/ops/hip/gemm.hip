@@ -987,6 +987,143 @@ static at::Tensor transpose_strided(const at::Tensor& src, const void* base,
   return out;
 }
 
+// batched TN over (outer x heads) panels: C[z] = A[z]^T B[z] with A
+// contiguous per batch and B a strided view (attention's dK = ds^T Q
+// and dV = att^T dOut run directly on the fused qkv / dout storage --
+// no ds^T / att^T transposes, no Q^T/dOut^T panel builds).
+template <typename T, int BT>
+__global__ __launch_bounds__(256, 2) void gemm_tn_tr_batched_kernel(
+    const T* __restrict__ A, const T* __restrict__ B, T* __restrict__ C,
+    long M, long I, long J, long sAb, long ldb, int bh, long sBo, long sBh,
+    long ldc, long sCo, long sCh, const T* __restrict__ zpage) {
+  using Frag = typename DTraits<T>::frag8;
+  constexpr int BKM = 64;
+  constexpr int R = BT / 32;
+  constexpr int SEGS = BT / 8;
+  __shared__ T AS[2][BKM * BT];
+  __shared__ T BS[2][BKM * BT];
+
+  const long z = blockIdx.z;
+  A += z * sAb;
+  B += (z / bh) * sBo + (z % bh) * sBh;
+  C += (z / bh) * sCo + (z % bh) * sCh;
+
+  const int nTj = (int)((J + BT - 1) / BT);
+  const long i0 = (long)(blockIdx.x / nTj) * BT;
+  const long j0 = (long)(blockIdx.x % nTj) * BT;
+
+  const int t = threadIdx.x;
+  const int lane = t & 63;
+  const int wid = t >> 6;
+  const int wr = wid >> 1, wc = wid & 1;
+
+  float4_t acc[R][R] = {};
+
+  const int prow_c = (t >> 1) & 15;
+  const int lsb_c = t & 1;
+  const int shi_c = t >> 5;
+  auto stage = [&](T (*dst)[BKM * BT], int buf, const T* __restrict__ src,
+                   long ld, long W, long col0, long mc) {
+#pragma unroll
+    for (int rnd = 0; rnd < BKM / (256 / SEGS); ++rnd) {
+      const int sgl = shi_c + rnd * 8;
+      const int seg_hi = sgl % (SEGS / 2);
+      const int kk2tt = sgl / (SEGS / 2);
+      const int rem = ((prow_c >> 2) << 3) | ((kk2tt & 1) << 2) |
+                      (prow_c & 3);
+      const int m_l = (kk2tt >> 1) * 32 + rem;
+      const int segv = seg_hi * 2 + lsb_c;
+      const long m_g = mc + m_l;
+      const long c = col0 + segv * 8;
+      const T* ga = (m_g < M && c + 8 <= W) ? src + m_g * ld + c : zpage;
+      gload_lds16(ga, &dst[buf][(rnd * 256 + t) * 8]);
+    }
+  };
+
+  stage(AS, 0, A, I, I, i0, 0);
+  stage(BS, 0, B, ldb, J, j0, 0);
+  __syncthreads();
+  int buf = 0;
+  for (long mc = 0; mc < M; mc += BKM) {
+    if (mc + BKM < M) {
+      stage(AS, buf ^ 1, A, I, I, i0, mc + BKM);
+      stage(BS, buf ^ 1, B, ldb, J, j0, mc + BKM);
+    }
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      Frag af[R], bf[R];
+#pragma unroll
+      for (int m = 0; m < R; ++m) {
+        const int cblk = wr * R + m;
+        union { trs4g h[2]; Frag f; } u;
+        u.h[0] = tr_read16g(&AS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        u.h[1] = tr_read16g(&AS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        af[m] = u.f;
+      }
+#pragma unroll
+      for (int n = 0; n < R; ++n) {
+        const int cblk = wc * R + n;
+        union { trs4g h[2]; Frag f; } u;
+        u.h[0] = tr_read16g(&BS[buf][(((kk * 2 + 0) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        u.h[1] = tr_read16g(&BS[buf][(((kk * 2 + 1) * (SEGS / 2) + cblk)
+                                      << 8) + lane * 4]);
+        bf[n] = u.f;
+      }
+#pragma unroll
+      for (int m = 0; m < R; ++m)
+#pragma unroll
+        for (int n = 0; n < R; ++n)
+          acc[m][n] = DTraits<T>::mfma_16x16x32(af[m], bf[n], acc[m][n]);
+    }
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  const int d_col = lane & 15;
+  const int d_row = (lane >> 4) * 4;
+#pragma unroll
+  for (int n = 0; n < R; ++n) {
+    const long j = j0 + wc * (BT / 2) + n * 16 + d_col;
+    if (j >= J) continue;
+#pragma unroll
+    for (int m = 0; m < R; ++m) {
+      const long i_base = i0 + wr * (BT / 2) + m * 16 + d_row;
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        const long i = i_base + jj;
+        if (i < I) C[i * ldc + j] = (T)acc[m][n][jj];
+      }
+    }
+  }
+}
+
+static void launch_tn_batched(const at::Tensor& like, const void* A,
+                              const void* B, void* C, long M, long I, long J,
+                              long sAb, long ldb, int bh, long sBo, long sBh,
+                              long ldc, long sCo, long sCh, long nb) {
+  int bt = (I >= 128 && J >= 128) ? 128 : 64;
+  long nwg = ((I + bt - 1) / bt) * ((J + bt - 1) / bt);
+  dim3 grid((unsigned)nwg, 1, (unsigned)nb);
+  DISPATCH_HALF_TYPES(like.scalar_type(), "tn_batched", [&] {
+    if (bt == 128)
+      gemm_tn_tr_batched_kernel<scalar_t, 128><<<grid, 256, 0,
+                                                 cur_stream()>>>(
+          (const scalar_t*)A, (const scalar_t*)B, (scalar_t*)C, M, I, J,
+          sAb, ldb, bh, sBo, sBh, ldc, sCo, sCh,
+          (const scalar_t*)zero_page(like));
+    else
+      gemm_tn_tr_batched_kernel<scalar_t, 64><<<grid, 256, 0,
+                                                cur_stream()>>>(
+          (const scalar_t*)A, (const scalar_t*)B, (scalar_t*)C, M, I, J,
+          sAb, ldb, bh, sBo, sBh, ldc, sCo, sCh,
+          (const scalar_t*)zero_page(like));
+  });
+  HIP_CHECK_LAST();
+}
+
 // qkv: [B, S, 3U] (U = H*D); mask: byte [B*H, S, S] or undefined.
 // returns {out [B, S, U], att [B*H, S, S]} -- att saved for backward.
 std::vector<at::Tensor> attention_fwd(const at::Tensor& qkv,
@@ -1035,23 +1172,17 @@ at::Tensor attention_bwd(const at::Tensor& dout, const at::Tensor& qkv,
   auto ds = softmax_bwd(datt, att, false, temperature);
   auto dqkv = at::empty_like(qkv);
   char* dq = (char*)dqkv.data_ptr();
-  // dQ = ds K: B-operand = K^T panels
+  // dQ = ds K: B-operand = K^T panels (d-contiguity needs the panel)
   auto kt = transpose_strided(qkv, qp + U * es, S, D, U3, (int)H, S * U3, D,
                               BH);
   launch_nt_strided(qkv, ds.data_ptr(), kt.data_ptr(), dq, S, D, S,
                     (long)H * S * S, (long)H * D * S, S * U3,
                     GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
-  // dK = ds^T Q: A = ds^T (batched transpose), B = Q^T panels
-  auto dsT = transpose2d(ds);
-  auto qt = transpose_strided(qkv, qp, S, D, U3, (int)H, S * U3, D, BH);
-  launch_nt_strided(qkv, dsT.data_ptr(), qt.data_ptr(), dq + U * es, S, D,
-                    S, (long)H * S * S, (long)H * D * S, S * U3,
-                    GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
-  // dV = att^T dOut: A = att^T, B = dOut^T panels
-  auto attT = transpose2d(att);
-  auto dot_t = transpose_strided(dout, dp, S, D, U, (int)H, S * U, D, BH);
-  launch_nt_strided(qkv, attT.data_ptr(), dot_t.data_ptr(), dq + 2 * U * es,
-                    S, D, S, (long)H * S * S, (long)H * D * S, S * U3,
-                    GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
+  // dK = ds^T Q and dV = att^T dOut: batched TN straight off the fused
+  // storage -- no ds^T/att^T transposes, no Q^T/dOut^T panel builds
+  launch_tn_batched(qkv, ds.data_ptr(), qp, dq + U * es, S, S, D,
+                    S * S, U3, (int)H, S * U3, D, U3, S * U3, D, BH);
+  launch_tn_batched(qkv, att.data_ptr(), dp, dq + 2 * U * es, S, S, D,
+                    S * S, U, (int)H, S * U, D, U3, S * U3, D, BH);
   return dqkv;
 }
