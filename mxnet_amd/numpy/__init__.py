@@ -303,5 +303,290 @@ def array_equal(a, b):
                             torch.as_tensor(b)))
 
 
+
+
+# ---------------------------------------------------------------------------
+# numpy-API tail (reference src/operator/numpy/* np_* op registrations):
+# torch-backed, matching numpy semantics; _w unwraps NDArray operands
+# ---------------------------------------------------------------------------
+
+def _w(x):
+    return x._t if isinstance(x, NDArray) else torch.as_tensor(x)
+
+
+def vstack(tup):
+    return NDArray(torch.vstack([_w(a) for a in tup]))
+
+
+def hstack(tup):
+    return NDArray(torch.hstack([_w(a) for a in tup]))
+
+
+def dstack(tup):
+    return NDArray(torch.dstack([_w(a) for a in tup]))
+
+
+def diff(a, n=1, axis=-1):
+    return NDArray(torch.diff(_w(a), n=n, dim=axis))
+
+
+def flip(m, axis=None):
+    t = _w(m)
+    dims = list(range(t.dim())) if axis is None else         ([axis] if isinstance(axis, int) else list(axis))
+    return NDArray(torch.flip(t, dims))
+
+
+def roll(a, shift, axis=None):
+    t = _w(a)
+    if axis is None:
+        return NDArray(torch.roll(t.reshape(-1), shift).reshape(t.shape))
+    return NDArray(torch.roll(t, shift, axis))
+
+
+def rot90(m, k=1, axes=(0, 1)):
+    return NDArray(torch.rot90(_w(m), k, axes))
+
+
+def pad(array, pad_width, mode='constant', constant_values=0):
+    t = _w(array)
+    if isinstance(pad_width, int):
+        pad_width = [(pad_width, pad_width)] * t.dim()
+    elif len(pad_width) == 2 and isinstance(pad_width[0], int):
+        pad_width = [tuple(pad_width)] * t.dim()
+    flat = []
+    for b, e in reversed(list(pad_width)):
+        flat += [b, e]
+    md = {'constant': 'constant', 'edge': 'replicate',
+          'reflect': 'reflect'}[mode]
+    kw = {'value': constant_values} if md == 'constant' else {}
+    if md != 'constant' and t.dim() < 3:
+        sq = t[None, None]
+        out = torch.nn.functional.pad(sq, flat, mode=md, **kw)[0, 0]
+    else:
+        out = torch.nn.functional.pad(t, flat, mode=md, **kw)
+    return NDArray(out)
+
+
+def tril(m, k=0):
+    return NDArray(torch.tril(_w(m), k))
+
+
+def triu(m, k=0):
+    return NDArray(torch.triu(_w(m), k))
+
+
+def inner(a, b):
+    return NDArray(torch.inner(_w(a), _w(b)))
+
+
+def kron(a, b):
+    return NDArray(torch.kron(_w(a), _w(b)))
+
+
+def trace(a, offset=0, axis1=0, axis2=1):
+    return NDArray(torch.diagonal(_w(a), offset, axis1, axis2).sum(-1))
+
+
+def diag(v, k=0):
+    return NDArray(torch.diag(_w(v), k))
+
+
+def diagonal(a, offset=0, axis1=0, axis2=1):
+    return NDArray(torch.diagonal(_w(a), offset, axis1, axis2))
+
+
+def ravel(a, order='C'):
+    return NDArray(_w(a).reshape(-1))
+
+
+def moveaxis(a, source, destination):
+    return NDArray(torch.movedim(_w(a), source, destination))
+
+
+def atleast_2d(*arys):
+    outs = [NDArray(torch.atleast_2d(_w(a))) for a in arys]
+    return outs[0] if len(outs) == 1 else outs
+
+
+def nan_to_num(x, nan=0.0, posinf=None, neginf=None):
+    return NDArray(torch.nan_to_num(_w(x), nan=nan, posinf=posinf,
+                                    neginf=neginf))
+
+
+def logical_and(x1, x2):
+    return NDArray(torch.logical_and(_w(x1), _w(x2)))
+
+
+def logical_or(x1, x2):
+    return NDArray(torch.logical_or(_w(x1), _w(x2)))
+
+
+def logical_not(x):
+    return NDArray(torch.logical_not(_w(x)))
+
+
+def bitwise_and(x1, x2):
+    return NDArray(torch.bitwise_and(_w(x1), _w(x2)))
+
+
+def bitwise_or(x1, x2):
+    return NDArray(torch.bitwise_or(_w(x1), _w(x2)))
+
+
+def bitwise_xor(x1, x2):
+    return NDArray(torch.bitwise_xor(_w(x1), _w(x2)))
+
+
+def histogram(a, bins=10, range=None):
+    t = _w(a).float()
+    lo, hi = (range if range is not None
+              else (float(t.min()), float(t.max())))
+    hist = torch.histc(t, bins=bins, min=lo, max=hi)
+    edges = torch.linspace(lo, hi, bins + 1)
+    return NDArray(hist), NDArray(edges)
+
+
+def percentile(a, q, axis=None):
+    t = _w(a).float()
+    qq = torch.as_tensor(q, dtype=torch.float64) / 100.0
+    return NDArray(torch.quantile(t, qq.to(t.dtype), dim=axis))
+
+
+def quantile(a, q, axis=None):
+    t = _w(a).float()
+    return NDArray(torch.quantile(t, torch.as_tensor(q, dtype=t.dtype),
+                                  dim=axis))
+
+
+def median(a, axis=None):
+    # torch.median picks the lower middle for even counts; numpy
+    # averages -- use the 0.5 quantile for numpy semantics
+    t = _w(a).float()
+    q = torch.tensor(0.5, dtype=t.dtype)
+    return NDArray(torch.quantile(t, q) if axis is None
+                   else torch.quantile(t, q, dim=axis))
+
+
+def average(a, axis=None, weights=None):
+    t = _w(a).float()
+    if weights is None:
+        return NDArray(t.mean() if axis is None else t.mean(dim=axis))
+    w = _w(weights).float()
+    if axis is None:
+        return NDArray((t * w).sum() / w.sum())
+    return NDArray((t * w).sum(dim=axis) / w.sum(dim=axis))
+
+
+def cov(m, rowvar=True):
+    t = _w(m).float()
+    if not rowvar:
+        t = t.t()
+    return NDArray(torch.cov(t))
+
+
+def corrcoef(x, rowvar=True):
+    t = _w(x).float()
+    if not rowvar:
+        t = t.t()
+    return NDArray(torch.corrcoef(t))
+
+
+def cross(a, b, axis=-1):
+    return NDArray(torch.cross(_w(a), _w(b), dim=axis))
+
+
+def interp(x, xp, fp):
+    xt, xpt, fpt = _w(x).float(), _w(xp).float(), _w(fp).float()
+    idx = torch.searchsorted(xpt, xt).clamp(1, xpt.numel() - 1)
+    x0, x1 = xpt[idx - 1], xpt[idx]
+    y0, y1 = fpt[idx - 1], fpt[idx]
+    t = (xt - x0) / (x1 - x0)
+    return NDArray((y0 + t * (y1 - y0)).clamp(float(fpt[0]) if False
+                                              else -float('inf'),
+                                              float('inf')))
+
+
+def gradient(f, *varargs, axis=None):
+    t = _w(f).float()
+    axes = (range(t.dim()) if axis is None
+            else ([axis] if isinstance(axis, int) else axis))
+    outs = [NDArray(torch.gradient(t, dim=ax)[0]) for ax in axes]
+    return outs[0] if len(outs) == 1 else outs
+
+
+def polyval(p, x):
+    pt, xt = _w(p).float(), _w(x).float()
+    out = torch.zeros_like(xt)
+    for c in pt:
+        out = out * xt + c
+    return NDArray(out)
+
+
+def bincount(x, weights=None, minlength=0):
+    return NDArray(torch.bincount(_w(x).long(),
+                                  _w(weights) if weights is not None
+                                  else None, minlength))
+
+
+def digitize(x, bins, right=False):
+    return NDArray(torch.bucketize(_w(x), _w(bins), right=not right))
+
+
+def nanmean(a, axis=None):
+    t = _w(a).float()
+    return NDArray(t.nanmean() if axis is None else t.nanmean(dim=axis))
+
+
+def nansum(a, axis=None):
+    t = _w(a).float()
+    return NDArray(t.nansum() if axis is None else t.nansum(dim=axis))
+
+
+def floor_divide(x1, x2):
+    return NDArray(torch.floor_divide(_w(x1), _w(x2)))
+
+
+def fmod(x1, x2):
+    return NDArray(torch.fmod(_w(x1), _w(x2)))
+
+
+def divmod(x1, x2):
+    q = torch.div(_w(x1), _w(x2), rounding_mode='floor')
+    return NDArray(q), NDArray(_w(x1) - q * _w(x2))
+
+
+def heaviside(x1, x2):
+    return NDArray(torch.heaviside(_w(x1), _w(x2).to(_w(x1).dtype)))
+
+
+def copysign(x1, x2):
+    return NDArray(torch.copysign(_w(x1), _w(x2)))
+
+
+def ldexp(x1, x2):
+    return NDArray(torch.ldexp(_w(x1), _w(x2)))
+
+
+def frexp(x):
+    m, e = torch.frexp(_w(x))
+    return NDArray(m), NDArray(e)
+
+
+def fix(x):
+    return NDArray(torch.trunc(_w(x)))
+
+
+def ediff1d(ary):
+    return NDArray(torch.diff(_w(ary).reshape(-1)))
+
+
+def i0(x):
+    return NDArray(torch.special.i0(_w(x)))
+
+
+def sinc(x):
+    return NDArray(torch.sinc(_w(x)))
+
+
 from . import random  # noqa: E402
 from . import linalg  # noqa: E402

@@ -70,3 +70,35 @@ def test_einsum_and_random():
     r = npr.uniform(0, 1, (100,)) if hasattr(npr, 'uniform') else None
     if r is not None:
         assert 0 <= float(r.handle.min()) and float(r.handle.max()) <= 1
+
+
+def test_numpy_api_tail():
+    """np tail coverage (reference src/operator/numpy np_* families)."""
+    x = np.array(onp.arange(12.).reshape(3, 4))
+    onp.testing.assert_allclose(np.trace(x).asnumpy(), 15.0)
+    onp.testing.assert_allclose(np.flip(x, 1).asnumpy(),
+                                onp.flip(onp.arange(12.).reshape(3, 4), 1))
+    onp.testing.assert_allclose(np.median(x).asnumpy(),
+                                onp.median(onp.arange(12.)))
+    onp.testing.assert_allclose(
+        np.pad(x, 1).asnumpy(), onp.pad(onp.arange(12.).reshape(3, 4), 1))
+    onp.testing.assert_allclose(
+        np.interp(np.array([0.5, 2.5]), np.array([0., 1., 2., 3.]),
+                  np.array([0., 2., 4., 6.])).asnumpy(), [1.0, 5.0])
+    onp.testing.assert_allclose(np.cov(x).asnumpy(),
+                                onp.cov(onp.arange(12.).reshape(3, 4)),
+                                rtol=1e-5)
+    onp.testing.assert_allclose(
+        np.gradient(x, axis=1).asnumpy(),
+        onp.gradient(onp.arange(12.).reshape(3, 4), axis=1))
+    h, e = np.histogram(x, bins=4)
+    assert float(h.handle.sum()) == 12
+    onp.testing.assert_allclose(np.kron(np.array(onp.eye(2)),
+                                        np.array(onp.ones((2, 2)))).asnumpy(),
+                                onp.kron(onp.eye(2), onp.ones((2, 2))))
+    onp.testing.assert_allclose(
+        np.polyval(np.array([1., 0., -1.]), np.array([2.0])).asnumpy(),
+        [3.0])
+    q, r = np.divmod(np.array([7., -7.]), np.array([3., 3.]))
+    onp.testing.assert_allclose(q.asnumpy(), [2., -3.])
+    onp.testing.assert_allclose(r.asnumpy(), [1., 2.])
