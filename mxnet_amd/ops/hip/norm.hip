@@ -520,9 +520,13 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ dy,
 // -- host wrappers ----------------------------------------------------------
 
 static dim3 bn_reduce_grid(long M, long C, long* rows_per_block) {
+  static const long want = [] {
+    const char* e = getenv("MXNET_BN_REDUCE_BLOCKS");
+    return e ? atol(e) : 2048L;  // sweepable: 256 CUs want >512 wgs
+  }();
   long xb = (C + 63) / 64;
   long yb = std::max<long>(1, std::min<long>((M + 255) / 256,
-                                             2048 / std::max<long>(xb, 1)));
+                                             want / std::max<long>(xb, 1)));
   *rows_per_block = (M + yb - 1) / yb;
   return dim3((unsigned)xb, (unsigned)yb);
 }
