@@ -69,7 +69,9 @@ at::Tensor softmax_bwd(const at::Tensor& dy, const at::Tensor& y,
 
 constexpr int kEwBlock = 256;
 // memory-bound launch cap: ~8 blocks/CU on 256 CUs (Guideline 11)
-constexpr long kEwMaxGrid = 2048;
+static const long kEwMaxGrid = [] {
+  const char* e = getenv("MXNET_EW_BLOCKS");
+  return e ? atoi(e) : 2048; }();
 
 inline int ew_grid(long work_items) {
   long g = (work_items + kEwBlock - 1) / kEwBlock;
