@@ -1730,7 +1730,10 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
       int bt = (Kg >= 128 && Cg >= 128) ? 128 : 64;
       int cpl4 = (Cg + bt - 1) / bt;
       int nwg4 = (int)(((Kg + bt - 1) / bt) * (long)R * S * cpl4);
-      long want4 = 2048;
+      static const long want4 = [] {
+        const char* e = getenv("MXNET_BWDW_BLOCKS");
+        return e ? atol(e) : 2048L;
+      }();
       long yb4 = std::max<long>(
           1, std::min<long>((M + 63) / 64,
                             want4 / std::max<long>((long)nwg4 * groups, 1)));
