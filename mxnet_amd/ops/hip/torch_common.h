@@ -71,7 +71,8 @@ constexpr int kEwBlock = 256;
 // memory-bound launch cap: ~8 blocks/CU on 256 CUs (Guideline 11)
 static const long kEwMaxGrid = [] {
   const char* e = getenv("MXNET_EW_BLOCKS");
-  return e ? atoi(e) : 1024; }()  // swept: 1024 edges 2048/4096;
+  return e ? atoi(e) : 1024;  // swept: 1024 edges 2048/4096
+}();
 
 inline int ew_grid(long work_items) {
   long g = (work_items + kEwBlock - 1) / kEwBlock;
