@@ -1681,8 +1681,12 @@ at::Tensor conv2d_nhwc_bwd_weight(const at::Tensor& dy, const at::Tensor& x,
     build_pixtab_kernel<<<ew_grid(M), 256, 0, cur_stream()>>>(
         (int4_t*)tab.data_ptr(), M, P, Q, sh, sw, ph, pw);
     int nwg = (int)(((Kout + 63) / 64) * R);
+    static const long c8want = [] {
+      const char* e = getenv("MXNET_C8_BLOCKS");
+      return e ? atol(e) : 2048L;
+    }();
     long yb = std::max<long>(
-        1, std::min<long>((M + 63) / 64, 2048 / std::max(nwg, 1)));
+        1, std::min<long>((M + 63) / 64, c8want / std::max(nwg, 1)));
     long m_per_slice = ((M + yb - 1) / yb + 63) / 64 * 64;
     yb = (M + m_per_slice - 1) / m_per_slice;
     auto dw32 = yb == 1

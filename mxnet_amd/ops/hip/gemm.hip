@@ -591,8 +591,12 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   // the LSTM recurrent GEMM (M=128,N=4096,K=1024: the fp32 workspace
   // zero+cast passes and atomics cost more than the occupancy win;
   // 1.59M -> 1.19M tokens/s), so small-K chip-filling stays off.
+  static const long kWant = [] {
+    const char* e = getenv("MXNET_GEMM_SPLITK_BLOCKS");
+    return e ? atol(e) : 2048L;
+  }();
   if (nwg * nb < 512 && nk_total > 16) {
-    ksplit = (int)std::min<long>((2048 + nwg * nb - 1) / (nwg * nb),
+    ksplit = (int)std::min<long>((kWant + nwg * nb - 1) / (nwg * nb),
                                  (nk_total + 15) / 16);
     tps = (nk_total + ksplit - 1) / ksplit;
     ksplit = (nk_total + tps - 1) / tps;
