@@ -593,7 +593,7 @@ at::Tensor gemm_nt_core(at::Tensor A, at::Tensor B,
   // 1.59M -> 1.19M tokens/s), so small-K chip-filling stays off.
   static const long kWant = [] {
     const char* e = getenv("MXNET_GEMM_SPLITK_BLOCKS");
-    return e ? atol(e) : 2048L;
+    return e ? atol(e) : 1024L;  // swept: 6369 vs 6348(512)/6353(2048)
   }();
   if (nwg * nb < 512 && nk_total > 16) {
     ksplit = (int)std::min<long>((kWant + nwg * nb - 1) / (nwg * nb),
