@@ -133,10 +133,20 @@ def export_model(sym_json, params, in_shapes, in_types=None, onnx_file=None,
                  P.attr_field(P.attr_float('momentum',
                                            float(attrs.get('momentum', 0.9)))))
             fuse = _bool(attrs.get('fuse_relu', 'False'))
-            bn_out = out + '_bn' if fuse else out
+            # fused residual arrives as a keyword input (model-zoo
+            # ResNet hot path): BN -> Add(residual) -> Relu
+            kw_names = [k for k in
+                        str(attrs.get('__kw_inputs__', '')).split(',') if k]
+            res = ins[5] if 'residual' in kw_names and len(ins) > 5 else None
+            bn_out = out + '_bn' if (fuse or res) else out
             emit('BatchNormalization', ins[:5], [bn_out], nname, a)
+            cur = bn_out
+            if res is not None:
+                nxt = out + '_res' if fuse else out
+                emit('Add', [cur, res], [nxt], nname + '_add')
+                cur = nxt
             if fuse:
-                emit('Relu', [bn_out], [out], nname + '_relu')
+                emit('Relu', [cur], [out], nname + '_relu')
         elif op == 'Activation':
             act = {'relu': 'Relu', 'sigmoid': 'Sigmoid', 'tanh': 'Tanh',
                    'softrelu': 'Softplus', 'softsign': 'Softsign'}[

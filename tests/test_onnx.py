@@ -81,3 +81,21 @@ def test_onnx_proto_structure(tmp_path):
         names.add(P.as_str(f, 8))
         assert len(P.as_bytes(f, 9)) > 0
     assert '0.weight' in names and '1.gamma' in names
+
+
+def test_onnx_roundtrip_resnet18(tmp_path):
+    """Model-zoo ResNet-18 (fused BN+residual+relu hot path) exports and
+    re-imports exactly — the residual keyword input becomes an Add node."""
+    from mxnet_amd.gluon.model_zoo import vision
+    torch.manual_seed(0)
+    net = vision.resnet18_v1(classes=10)
+    net.initialize()
+    net.hybridize()
+    x = mx.nd.from_torch(torch.randn(1, 3, 32, 32))
+    y0 = net(x).asnumpy()
+    net.export(str(tmp_path / 'r18'))
+    params = mx.nd.load(str(tmp_path / 'r18-0000.params'))
+    buf = mxonnx.export_model(str(tmp_path / 'r18-symbol.json'), params,
+                              [(1, 3, 32, 32)])
+    net2 = mxonnx.import_to_gluon(buf)
+    np.testing.assert_allclose(net2(x).asnumpy(), y0, rtol=1e-4, atol=1e-4)
