@@ -99,3 +99,27 @@ def test_onnx_roundtrip_resnet18(tmp_path):
                               [(1, 3, 32, 32)])
     net2 = mxonnx.import_to_gluon(buf)
     np.testing.assert_allclose(net2(x).asnumpy(), y0, rtol=1e-4, atol=1e-4)
+
+
+def test_onnx_embedding_gather(tmp_path):
+    """Embedding exports as Gather and re-imports (reference mx2onnx
+    embedding mapping)."""
+    from mxnet_amd.gluon import nn as gnn
+    from mxnet_amd import symbol as S
+    torch.manual_seed(2)
+    emb = gnn.Embedding(20, 8)
+    dense = gnn.Dense(4, in_units=8, flatten=False)
+    emb.initialize()
+    dense.initialize()
+    from mxnet_amd.gluon import nn
+    net = nn.HybridSequential()
+    net.add(emb, dense)
+    net.hybridize()
+    x = mx.nd.from_torch(torch.randint(0, 20, (2, 5)))
+    y0 = net(x).asnumpy()
+    net.export(str(tmp_path / 'e'))
+    params = mx.nd.load(str(tmp_path / 'e-0000.params'))
+    buf = mxonnx.export_model(str(tmp_path / 'e-symbol.json'), params,
+                              [(2, 5)], in_types=[torch.int64])
+    net2 = mxonnx.import_to_gluon(buf)
+    np.testing.assert_allclose(net2(x).asnumpy(), y0, rtol=1e-4, atol=1e-5)
