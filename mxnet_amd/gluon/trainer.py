@@ -185,6 +185,10 @@ class Trainer:
                     flat = _flatten_dense_tensors(state['tensors'])
                     state['flat'] = flat
                     state['handle'] = dist.all_reduce(flat, async_op=True)
+                elif state['pending'] < 0:
+                    # a second backward before step(): the reduce already
+                    # in flight used stale grads — re-reduce at drain time
+                    state['stale'] = True
             return hook
 
         self._ov_hook_handles = []
@@ -210,6 +214,15 @@ class Trainer:
                     state['handle'] = dist.all_reduce(flat, async_op=True)
             for state in self._ov_buckets:
                 state['handle'].wait()
+                if state.pop('stale', False):
+                    # gradients accumulated again after the async launch
+                    # (multi-backward step): the in-flight result is stale
+                    # on every rank symmetrically — discard and re-reduce
+                    # the up-to-date grads.
+                    flat = _flatten_dense_tensors(state['tensors'])
+                    state['flat'] = flat
+                    state['handle'] = dist.all_reduce(flat, async_op=True)
+                    state['handle'].wait()
                 flat = state['flat']
                 flat.div_(world)
                 for t, u in zip(state['tensors'],
@@ -291,6 +304,20 @@ class Trainer:
                     g.index_add_(0, rs.indices, rs.data.to(g.dtype))
 
     def _update(self, ignore_stale_grad=False):
+        # AMP dynamic loss scaling (reference trainer.py:445-447): on an
+        # fp16 gradient overflow skip the whole update, shrink the scale;
+        # grow it after scale_window clean steps.  Without this the first
+        # overflow writes inf/NaN into the weights irrecoverably.
+        scaler = getattr(self, '_amp_loss_scaler', None)
+        if scaler is not None:
+            overflow = scaler.has_overflow(self._params)
+            scaler.update_scale(overflow)
+            self._scale = 1.0 / scaler.loss_scale
+            if overflow:
+                for p in self._params:
+                    if p.grad_req == 'write':
+                        p.zero_grad()
+                return
         to_zero = []
         for i, p in enumerate(self._params):
             self._check_states(i, p)

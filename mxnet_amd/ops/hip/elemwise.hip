@@ -154,15 +154,15 @@ __global__ void sgd_mp_kernel(T* __restrict__ w, float* __restrict__ master,
     float g = (float)grad[i] * rescale;
     if (clip > 0.f) g = fminf(fmaxf(g, -clip), clip);
     g += wd * wm;
-    float upd;
+    // reference rule: lr folded into the momentum buffer
+    // (mom = mu*mom - lr*g; w += mom)
     if (mom) {
-      float m = mom[i] * mu + g;
+      float m = mom[i] * mu - lr * g;
       mom[i] = m;
-      upd = m;
+      wm += m;
     } else {
-      upd = g;
+      wm -= lr * g;
     }
-    wm -= lr * upd;
     if (master) master[i] = wm;
     w[i] = (T)wm;
   }
@@ -470,15 +470,14 @@ __global__ void multi_sgd_kernel(const MTChunk* __restrict__ chunks,
     float gv = (float)g[j] * rescale;
     if (clip > 0.f) gv = fminf(fmaxf(gv, -clip), clip);
     gv += c.wd * wm;
-    float upd;
+    // reference rule: lr folded into the momentum buffer
     if (c.mom) {
-      float m = c.mom[j] * mu + gv;
+      float m = c.mom[j] * mu - c.lr * gv;
       c.mom[j] = m;
-      upd = m;
+      wm += m;
     } else {
-      upd = gv;
+      wm -= c.lr * gv;
     }
-    wm -= c.lr * upd;
     if (c.master) c.master[j] = wm;
     w[j] = (T)wm;
   }

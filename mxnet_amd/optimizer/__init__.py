@@ -130,11 +130,13 @@ class SGD(Optimizer):
             g = self._preprocess_grad(g)
             g = g + wd * w.float()
             if state is not None:
-                state.mul_(self.momentum).add_(g)
-                upd = state
+                # reference sgd_mom_update folds lr INTO the momentum
+                # buffer (mom = mu*mom - lr*g; w += mom) so trajectories
+                # track the reference bit-for-bit under lr schedules
+                state.mul_(self.momentum).sub_(lr * g)
+                w.add_(state.to(w.dtype))
             else:
-                upd = g
-            w.sub_((lr * upd).to(w.dtype))
+                w.sub_((lr * g).to(w.dtype))
 
     def _update_row_sparse(self, index, weight, grad, state):
         """Lazy row-sparse SGD (reference sgd_mom_update on
@@ -150,18 +152,19 @@ class SGD(Optimizer):
             wr = w.index_select(0, rows).float()
             g = g + wd * wr
             if state is not None and self.momentum != 0:
+                # lr folded into the momentum buffer (reference rule)
                 if self.lazy_update:
                     mr = state.index_select(0, rows)
-                    mr.mul_(self.momentum).add_(g)
+                    mr.mul_(self.momentum).sub_(lr * g)
                     state.index_copy_(0, rows, mr)
                     upd = mr
                 else:
                     state.mul_(self.momentum)
-                    state.index_add_(0, rows, g)
+                    state.index_add_(0, rows, -lr * g)
                     upd = state.index_select(0, rows)
+                w.index_copy_(0, rows, (wr + upd).to(w.dtype))
             else:
-                upd = g
-            w.index_copy_(0, rows, (wr - lr * upd).to(w.dtype))
+                w.index_copy_(0, rows, (wr - lr * g).to(w.dtype))
 
     def update_multi_precision(self, index, weight, grad, state):
         from ..ndarray.sparse import RowSparseNDArray
@@ -174,7 +177,7 @@ class SGD(Optimizer):
         rescale+clip+wd+momentum+master-weight update+fp16 cast in one pass."""
         w = weight._t if isinstance(weight, NDArray) else weight
         g = grad._t if isinstance(grad, NDArray) else grad
-        if w.is_cuda and g.dtype == w.dtype:
+        if type(self) is SGD and w.is_cuda and g.dtype == w.dtype:
             from ..ops.dispatch import hipops, use_hip
             ext = hipops() if use_hip(w) else None
             if ext is not None:
@@ -325,7 +328,8 @@ class RMSProp(Optimizer):
                 mom.mul_(self.momentum).add_(lr * g / ((n - mg * mg + self.epsilon).sqrt()))
                 w.sub_(mom.to(w.dtype))
             else:
-                w.sub_((lr * g / (n.sqrt() + self.epsilon)).to(w.dtype))
+                # reference rmsprop_update: epsilon INSIDE the sqrt
+                w.sub_((lr * g / (n + self.epsilon).sqrt()).to(w.dtype))
 
 
 @register

@@ -32,9 +32,12 @@ V1_MAGIC = 0xF993fac8
 def _write_ndarray(f, nd):
     t = nd._t.detach().cpu().contiguous()
     type_flag = TORCH_TO_TYPE_FLAG[t.dtype]
-    f.write(struct.pack('<I', V2_MAGIC))
-    f.write(struct.pack('<i', 0))                      # stype dense
     shape = tuple(t.shape)
+    # 0-dim scalars only exist under np-shape semantics: V3 magic
+    # (reference NDArray::Save ndarray.cc:1730-1738; in V2 ndim==0 means
+    # an empty "none" array and the record stops right after the shape).
+    f.write(struct.pack('<I', V3_MAGIC if len(shape) == 0 else V2_MAGIC))
+    f.write(struct.pack('<i', 0))                      # stype dense
     f.write(struct.pack('<i', len(shape)))
     for s in shape:
         f.write(struct.pack('<q', s))
@@ -61,9 +64,17 @@ def _read_ndarray(f):
         if stype != 0:
             raise NotImplementedError('sparse NDArray load: deferred')
         ndim = struct.unpack('<i', _read_exact(f, 4))[0]
+        # "none" arrays stop right after the shape (reference
+        # NDArray::Save `if (is_none()) return` ndarray.cc:1753):
+        # V2 ndim==0 and V3 ndim==-1 carry no ctx/dtype/data — reading
+        # further would desync every subsequent record in the file.
+        if (magic == V2_MAGIC and ndim == 0) or ndim < 0:
+            return NDArray(torch.empty(0))
         shape = struct.unpack('<%dq' % ndim, _read_exact(f, 8 * ndim)) if ndim else ()
     elif magic == V1_MAGIC:
         ndim = struct.unpack('<i', _read_exact(f, 4))[0]
+        if ndim == 0:
+            return NDArray(torch.empty(0))
         shape = struct.unpack('<%dq' % ndim, _read_exact(f, 8 * ndim)) if ndim else ()
     else:
         # pre-V1 legacy: magic was actually ndim (u32 dims format)
