@@ -1,0 +1,45 @@
+# Native core build — hipcc only, no torch toolchain, no hipify.
+# Produces the in-tree extension mxnet_amd/_core.cpython-310-*.so
+HIPCC    := /opt/rocm/bin/hipcc
+ARCH     := gfx950
+PYINC    := $(shell python3 -c "import sysconfig; print(sysconfig.get_paths()['include'])")
+PYBIND   := $(shell python3 -c "import pybind11; print(pybind11.get_include())")
+EXT      := $(shell python3 -c "import sysconfig; print(sysconfig.get_config_var('EXT_SUFFIX'))")
+NUMPYINC := $(shell python3 -c "import numpy; print(numpy.get_include())")
+
+CXXFLAGS := -fPIC -O3 -std=c++17 --offload-arch=$(ARCH) \
+            -I src -I mxnet_amd/ops/hip -I $(PYINC) -I $(PYBIND) -I $(NUMPYINC) \
+            -Wno-unused-result -parallel-jobs=4
+
+CORE_SRCS := src/core/storage.cc src/core/engine.cc src/core/ndarray.cc \
+             src/core/op.cc
+OPS_SRCS  := $(wildcard src/ops/*.hip)
+PYBIND_SRC := src/core/pybind.cc
+
+CORE_OBJS := $(patsubst src/%.cc,build/core/%.o,$(CORE_SRCS))
+OPS_OBJS  := $(patsubst src/%.hip,build/core/%.o,$(OPS_SRCS))
+PY_OBJ    := build/core/pybind.o
+
+TARGET := mxnet_amd/_core$(EXT)
+
+all: $(TARGET)
+
+build/core/%.o: src/%.cc src/core/*.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+build/core/%.o: src/%.hip src/core/*.h src/ops/*.h mxnet_amd/ops/hip/common.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
+
+$(PY_OBJ): $(PYBIND_SRC) src/core/*.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -c $< -o $@
+
+$(TARGET): $(CORE_OBJS) $(OPS_OBJS) $(PY_OBJ)
+	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64
+
+clean:
+	rm -rf build/core $(TARGET)
+
+.PHONY: all clean

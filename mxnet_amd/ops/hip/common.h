@@ -99,10 +99,14 @@ DEV_INLINE int xcd_swizzle(int bid, int nwg) {
   return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
 }
 
-#define HIP_CHECK_LAST()                                                     \
+#include <stdexcept>
+#include <string>
+
+#define HIP_CHECK_LAST()                                                      \
   do {                                                                        \
     hipError_t e_ = hipGetLastError();                                        \
     if (e_ != hipSuccess) {                                                   \
-      TORCH_CHECK(false, "HIP kernel launch failed: ", hipGetErrorString(e_)); \
+      throw std::runtime_error(std::string("HIP kernel launch failed: ") +    \
+                               hipGetErrorString(e_));                        \
     }                                                                         \
   } while (0)

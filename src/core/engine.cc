@@ -1,0 +1,532 @@
+#include "engine.h"
+
+#include <atomic>
+#include <cassert>
+#include <future>
+
+namespace mxcore {
+
+namespace {
+
+// pooled hipEvents (one pool per device)
+struct EventPool {
+  std::mutex mu;
+  std::vector<hipEvent_t> free_list;
+  hipEvent_t Take(int dev) {
+    {
+      std::lock_guard<std::mutex> g(mu);
+      if (!free_list.empty()) {
+        hipEvent_t e = free_list.back();
+        free_list.pop_back();
+        return e;
+      }
+    }
+    hipEvent_t e;
+    MX_HIP_CALL(hipSetDevice(dev));
+    MX_HIP_CALL(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    return e;
+  }
+  void Put(hipEvent_t e) {
+    std::lock_guard<std::mutex> g(mu);
+    free_list.push_back(e);
+  }
+};
+
+EventPool& GetEventPool(int dev) {
+  static EventPool pools[64];
+  return pools[dev & 63];
+}
+
+// refcounted event: recycled to the pool when the last holder drops it
+struct EventRef {
+  hipEvent_t ev = nullptr;
+  int dev = 0;
+  hipStream_t stream = nullptr;  // stream it was recorded on
+  ~EventRef() {
+    if (ev) GetEventPool(dev).Put(ev);
+  }
+};
+using EventPtr = std::shared_ptr<EventRef>;
+
+}  // namespace
+
+struct Opr;
+
+struct Block {
+  Opr* opr;
+  bool write;
+  bool granted = false;
+};
+
+struct Var {
+  uint64_t version = 0;
+  std::deque<Block> queue;
+  int running_reads = 0;
+  std::exception_ptr exc;
+  EventPtr last_event;  // execution-completion of the last write
+  // execution-completion of reads since that write: a later writer (or the
+  // storage-freeing delete op) must order behind these, not just behind
+  // the producing write (write-after-read across streams)
+  std::vector<EventPtr> read_events;
+  bool to_delete = false;
+};
+
+struct Opr {
+  OpFn fn;
+  std::vector<Var*> const_vars;
+  std::vector<Var*> mutable_vars;
+  int wait = 0;
+  FnProperty prop = FnProperty::kNormal;
+  Context ctx;
+  const char* name = "";
+  VarId delete_var = 0;           // kDeleteVar: var to erase afterwards
+  std::vector<EventPtr> waits;    // foreign events to wait before fn
+};
+
+// one FIFO queue + worker thread(s)
+struct WorkQueue {
+  std::mutex mu;
+  std::condition_variable cv;
+  std::queue<Opr*> q;
+  bool shutdown = false;
+  void Push(Opr* o) {
+    {
+      std::lock_guard<std::mutex> g(mu);
+      q.push(o);
+    }
+    cv.notify_one();
+  }
+  Opr* Pop() {
+    std::unique_lock<std::mutex> lk(mu);
+    cv.wait(lk, [&] { return shutdown || !q.empty(); });
+    if (q.empty()) return nullptr;
+    Opr* o = q.front();
+    q.pop();
+    return o;
+  }
+  void Shutdown() {
+    {
+      std::lock_guard<std::mutex> g(mu);
+      shutdown = true;
+    }
+    cv.notify_all();
+  }
+};
+
+struct DeviceWorkers {
+  hipStream_t compute = nullptr, copy = nullptr, comm = nullptr;
+  WorkQueue compute_q, copy_q, comm_q;
+  std::thread compute_t, copy_t, comm_t;
+  std::atomic<bool> capturing{false};
+  hipGraph_t captured = nullptr;
+};
+
+struct Engine::Impl {
+  std::mutex mu_;                       // protects vars/inflight
+  std::condition_variable done_cv_;
+  std::unordered_map<VarId, std::unique_ptr<Var>> vars_;
+  VarId next_var_ = 1;
+  std::atomic<long> inflight_{0};
+  std::exception_ptr global_exc_;
+
+  WorkQueue cpu_q_, cpu_prio_q_;
+  std::vector<std::thread> cpu_workers_;
+
+  std::mutex dev_mu_;
+  std::unordered_map<int, std::unique_ptr<DeviceWorkers>> devices_;
+
+  Engine* owner_ = nullptr;
+
+  Var* GetVar(VarId id) {
+    auto it = vars_.find(id);
+    MX_CHECK(it != vars_.end(), "unknown engine var " << id);
+    return it->second.get();
+  }
+
+  DeviceWorkers* GetDevice(int dev) {
+    std::lock_guard<std::mutex> g(dev_mu_);
+    auto it = devices_.find(dev);
+    if (it != devices_.end()) return it->second.get();
+    auto dw = std::make_unique<DeviceWorkers>();
+    MX_HIP_CALL(hipSetDevice(dev));
+    MX_HIP_CALL(hipStreamCreateWithFlags(&dw->compute, hipStreamNonBlocking));
+    MX_HIP_CALL(hipStreamCreateWithFlags(&dw->copy, hipStreamNonBlocking));
+    MX_HIP_CALL(hipStreamCreateWithFlags(&dw->comm, hipStreamNonBlocking));
+    DeviceWorkers* p = dw.get();
+    p->compute_t = std::thread([this, p, dev] {
+      WorkerLoop(&p->compute_q, RunContext{Context::GPU(dev), p->compute}, p);
+    });
+    p->copy_t = std::thread([this, p, dev] {
+      WorkerLoop(&p->copy_q, RunContext{Context::GPU(dev), p->copy}, p);
+    });
+    p->comm_t = std::thread([this, p, dev] {
+      WorkerLoop(&p->comm_q, RunContext{Context::GPU(dev), p->comm}, p);
+    });
+    devices_.emplace(dev, std::move(dw));
+    return p;
+  }
+
+  void StartCPUWorkers() {
+    int n = env_int("MXNET_CPU_WORKER_NTHREADS", 4);
+    RunContext cpu_rc{Context::CPU(), nullptr};
+    for (int i = 0; i < n; ++i)
+      cpu_workers_.emplace_back(
+          [this, cpu_rc] { WorkerLoop(&cpu_q_, cpu_rc, nullptr); });
+    for (int i = 0; i < 2; ++i)
+      cpu_workers_.emplace_back(
+          [this, cpu_rc] { WorkerLoop(&cpu_prio_q_, cpu_rc, nullptr); });
+  }
+
+  // ---- dependency tracking (reference threaded_engine.h:120-229) ------
+  void Push(Opr* opr) {
+    bool ready;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      ++inflight_;
+      opr->wait = (int)(opr->const_vars.size() + opr->mutable_vars.size());
+      for (Var* v : opr->const_vars) {
+        bool write_pending = false;
+        for (auto& b : v->queue)
+          if (b.write) {
+            write_pending = true;
+            break;
+          }
+        v->queue.push_back({opr, false, !write_pending});
+        if (!write_pending) {
+          ++v->running_reads;
+          --opr->wait;
+        }
+      }
+      for (Var* v : opr->mutable_vars) {
+        bool idle = v->queue.empty() && v->running_reads == 0;
+        v->queue.push_back({opr, true, idle});
+        if (idle) --opr->wait;
+      }
+      ready = opr->wait == 0;
+    }
+    if (ready) Dispatch(opr);
+  }
+
+  // snapshot foreign completion events, then hand to the right queue
+  void Dispatch(Opr* opr) {
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      for (Var* v : opr->const_vars) {
+        if (v->last_event) opr->waits.push_back(v->last_event);
+      }
+      for (Var* v : opr->mutable_vars) {
+        if (v->last_event) opr->waits.push_back(v->last_event);
+        for (auto& e : v->read_events) opr->waits.push_back(e);
+      }
+    }
+    switch (opr->prop) {
+      case FnProperty::kDeleteVar:
+      case FnProperty::kAsync:
+      case FnProperty::kCPUPrioritized:
+        cpu_prio_q_.Push(opr);
+        return;
+      default:
+        break;
+    }
+    if (!opr->ctx.is_gpu()) {
+      cpu_q_.Push(opr);
+      return;
+    }
+    DeviceWorkers* dw = GetDevice(opr->ctx.dev_id);
+    switch (opr->prop) {
+      case FnProperty::kCopyFromGPU:
+      case FnProperty::kCopyToGPU:
+        dw->copy_q.Push(opr);
+        break;
+      case FnProperty::kGPUPrioritized:
+        dw->comm_q.Push(opr);
+        break;
+      default:
+        dw->compute_q.Push(opr);
+    }
+  }
+
+  void WorkerLoop(WorkQueue* q, RunContext rc, DeviceWorkers* dw) {
+    if (rc.ctx.is_gpu()) (void)hipSetDevice(rc.ctx.dev_id);
+    for (;;) {
+      Opr* opr = q->Pop();
+      if (opr == nullptr) return;
+      std::exception_ptr exc;
+      bool capturing = dw && dw->capturing.load(std::memory_order_relaxed);
+      try {
+        // order execution behind producers on other streams
+        for (auto& w : opr->waits) {
+          if (rc.ctx.is_gpu() && !capturing) {
+            if (w->stream != rc.stream || w->dev != rc.ctx.dev_id)
+              MX_HIP_CALL(hipStreamWaitEvent(rc.stream, w->ev, 0));
+          } else if (!rc.ctx.is_gpu()) {
+            MX_HIP_CALL(hipEventSynchronize(w->ev));
+          }
+        }
+        if (opr->fn) opr->fn(rc);
+      } catch (...) {
+        exc = std::current_exception();
+      }
+      EventPtr done;
+      if (rc.ctx.is_gpu() && !exc && !capturing) {
+        // execution-completion marker for cross-stream/CPU consumers
+        done = std::make_shared<EventRef>();
+        done->dev = rc.ctx.dev_id;
+        done->stream = rc.stream;
+        done->ev = GetEventPool(done->dev).Take(done->dev);
+        hipError_t e = hipEventRecord(done->ev, rc.stream);
+        if (e != hipSuccess) {
+          (void)hipGetLastError();
+          done.reset();
+        }
+      }
+      OnComplete(opr, exc, done);
+    }
+  }
+
+  void GrantHead(Var* v, std::vector<Opr*>* now_ready) {
+    for (auto it = v->queue.begin(); it != v->queue.end() && !it->write; ++it) {
+      if (!it->granted) {
+        it->granted = true;
+        ++v->running_reads;
+        if (--it->opr->wait == 0) now_ready->push_back(it->opr);
+      }
+    }
+    if (!v->queue.empty() && v->queue.front().write && v->running_reads == 0 &&
+        !v->queue.front().granted) {
+      v->queue.front().granted = true;
+      if (--v->queue.front().opr->wait == 0)
+        now_ready->push_back(v->queue.front().opr);
+    }
+  }
+
+  void OnComplete(Opr* opr, std::exception_ptr exc, EventPtr done) {
+    std::vector<Opr*> now_ready;
+    VarId erase_var = 0;
+    {
+      std::lock_guard<std::mutex> g(mu_);
+      if (exc) {
+        global_exc_ = exc;
+        for (Var* v : opr->mutable_vars) v->exc = exc;
+      }
+      for (Var* v : opr->const_vars) {
+        for (auto it = v->queue.begin(); it != v->queue.end(); ++it)
+          if (it->opr == opr && !it->write) {
+            v->queue.erase(it);
+            break;
+          }
+        --v->running_reads;
+        if (done) v->read_events.push_back(done);
+        GrantHead(v, &now_ready);
+      }
+      for (Var* v : opr->mutable_vars) {
+        ++v->version;
+        // this write's execution is ordered behind the reads it waited on,
+        // so its event supersedes them
+        v->read_events.clear();
+        if (done) v->last_event = done;
+        for (auto it = v->queue.begin(); it != v->queue.end(); ++it)
+          if (it->opr == opr && it->write) {
+            v->queue.erase(it);
+            break;
+          }
+        GrantHead(v, &now_ready);
+      }
+      if (opr->delete_var) erase_var = opr->delete_var;
+      --inflight_;
+    }
+    done_cv_.notify_all();
+    for (Opr* o : now_ready) Dispatch(o);
+    if (erase_var) {
+      std::lock_guard<std::mutex> g(mu_);
+      auto it = vars_.find(erase_var);
+      if (it != vars_.end() && it->second->queue.empty() &&
+          it->second->running_reads == 0)
+        vars_.erase(it);
+    }
+    delete opr;
+  }
+};
+
+Engine::Engine() : impl_(new Impl()) {
+  impl_->owner_ = this;
+  impl_->StartCPUWorkers();
+}
+
+Engine* Engine::Get() {
+  static Engine* e = new Engine();
+  return e;
+}
+
+VarId Engine::NewVariable() {
+  std::lock_guard<std::mutex> g(impl_->mu_);
+  VarId id = impl_->next_var_++;
+  impl_->vars_.emplace(id, std::make_unique<Var>());
+  return id;
+}
+
+void Engine::PushDeleteVariable(VarId v, OpFn on_delete) {
+  Var* var;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu_);
+    auto it = impl_->vars_.find(v);
+    if (it == impl_->vars_.end()) return;
+    var = it->second.get();
+    var->to_delete = true;
+  }
+  auto* opr = new Opr();
+  // the generic CPU-op path synchronizes every pending execution event of
+  // a mutable var before fn runs, so the storage is free to release here
+  opr->fn = std::move(on_delete);
+  opr->mutable_vars = {var};
+  opr->prop = FnProperty::kDeleteVar;
+  opr->ctx = Context::CPU();
+  opr->delete_var = v;
+  impl_->Push(opr);
+}
+
+void Engine::PushAsync(OpFn fn, Context ctx, const std::vector<VarId>& cv,
+                       const std::vector<VarId>& mv, FnProperty prop,
+                       const char* name) {
+  auto* opr = new Opr();
+  opr->fn = std::move(fn);
+  opr->prop = prop;
+  opr->ctx = ctx;
+  if (name) opr->name = name;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu_);
+    for (VarId id : cv) opr->const_vars.push_back(impl_->GetVar(id));
+    for (VarId id : mv) opr->mutable_vars.push_back(impl_->GetVar(id));
+  }
+  impl_->Push(opr);
+}
+
+void Engine::WaitForVar(VarId v) {
+  std::exception_ptr exc;
+  EventPtr ev;
+  {
+    std::unique_lock<std::mutex> lk(impl_->mu_);
+    auto it = impl_->vars_.find(v);
+    if (it == impl_->vars_.end()) return;
+    Var* var = it->second.get();
+    impl_->done_cv_.wait(
+        lk, [&] { return var->queue.empty() && var->running_reads == 0; });
+    exc = var->exc;
+    var->exc = nullptr;
+    ev = var->last_event;
+  }
+  if (ev && ev->ev) MX_HIP_CALL(hipEventSynchronize(ev->ev));
+  if (exc) std::rethrow_exception(exc);
+}
+
+void Engine::WaitForAll() {
+  std::exception_ptr exc;
+  {
+    std::unique_lock<std::mutex> lk(impl_->mu_);
+    impl_->done_cv_.wait(lk, [&] { return impl_->inflight_.load() == 0; });
+    exc = impl_->global_exc_;
+    impl_->global_exc_ = nullptr;
+  }
+  // drain execution on every known device stream
+  {
+    std::lock_guard<std::mutex> g(impl_->dev_mu_);
+    for (auto& kv : impl_->devices_) {
+      (void)hipStreamSynchronize(kv.second->compute);
+      (void)hipStreamSynchronize(kv.second->copy);
+      (void)hipStreamSynchronize(kv.second->comm);
+    }
+  }
+  if (exc) std::rethrow_exception(exc);
+}
+
+uint64_t Engine::Version(VarId v) {
+  std::lock_guard<std::mutex> g(impl_->mu_);
+  auto it = impl_->vars_.find(v);
+  return it == impl_->vars_.end() ? 0 : it->second->version;
+}
+
+void Engine::Throw(VarId v) {
+  std::exception_ptr exc;
+  {
+    std::lock_guard<std::mutex> g(impl_->mu_);
+    auto it = impl_->vars_.find(v);
+    if (it == impl_->vars_.end()) return;
+    exc = it->second->exc;
+    it->second->exc = nullptr;
+  }
+  if (exc) std::rethrow_exception(exc);
+}
+
+hipEvent_t Engine::LastEvent(VarId v) {
+  std::lock_guard<std::mutex> g(impl_->mu_);
+  auto it = impl_->vars_.find(v);
+  if (it == impl_->vars_.end() || !it->second->last_event) return nullptr;
+  return it->second->last_event->ev;
+}
+
+hipStream_t Engine::ComputeStream(int dev) {
+  return impl_->GetDevice(dev)->compute;
+}
+hipStream_t Engine::CopyStream(int dev) { return impl_->GetDevice(dev)->copy; }
+hipStream_t Engine::CommStream(int dev) { return impl_->GetDevice(dev)->comm; }
+
+// ---- hipGraph capture -----------------------------------------------------
+void Engine::BeginCapture(int dev) {
+  DeviceWorkers* dw = impl_->GetDevice(dev);
+  std::promise<void> p;
+  auto fut = p.get_future();
+  PushAsync(
+      [dw, &p](const RunContext& rc) {
+        MX_HIP_CALL(
+            hipStreamBeginCapture(rc.stream, hipStreamCaptureModeRelaxed));
+        dw->capturing.store(true);
+        p.set_value();
+      },
+      Context::GPU(dev), {}, {}, FnProperty::kNormal, "BeginCapture");
+  fut.wait();
+}
+
+uintptr_t Engine::EndCapture(int dev) {
+  DeviceWorkers* dw = impl_->GetDevice(dev);
+  std::promise<uintptr_t> p;
+  auto fut = p.get_future();
+  PushAsync(
+      [dw, &p](const RunContext& rc) {
+        dw->capturing.store(false);
+        hipGraph_t g = nullptr;
+        MX_HIP_CALL(hipStreamEndCapture(rc.stream, &g));
+        hipGraphExec_t exec = nullptr;
+        MX_HIP_CALL(hipGraphInstantiate(&exec, g, nullptr, nullptr, 0));
+        (void)hipGraphDestroy(g);
+        p.set_value((uintptr_t)exec);
+      },
+      Context::GPU(dev), {}, {}, FnProperty::kNormal, "EndCapture");
+  return fut.get();
+}
+
+void Engine::LaunchGraph(int dev, uintptr_t exec) {
+  PushAsync(
+      [exec](const RunContext& rc) {
+        MX_HIP_CALL(hipGraphLaunch((hipGraphExec_t)exec, rc.stream));
+      },
+      Context::GPU(dev), {}, {}, FnProperty::kNormal, "GraphLaunch");
+}
+
+void Engine::StopWorkers() {
+  impl_->cpu_q_.Shutdown();
+  impl_->cpu_prio_q_.Shutdown();
+  for (auto& t : impl_->cpu_workers_)
+    if (t.joinable()) t.join();
+  impl_->cpu_workers_.clear();
+  std::lock_guard<std::mutex> g(impl_->dev_mu_);
+  for (auto& kv : impl_->devices_) {
+    kv.second->compute_q.Shutdown();
+    kv.second->copy_q.Shutdown();
+    kv.second->comm_q.Shutdown();
+    if (kv.second->compute_t.joinable()) kv.second->compute_t.join();
+    if (kv.second->copy_t.joinable()) kv.second->copy_t.join();
+    if (kv.second->comm_t.joinable()) kv.second->comm_t.join();
+  }
+}
+
+}  // namespace mxcore

@@ -1,0 +1,247 @@
+#include "op.h"
+
+#include <algorithm>
+#include <mutex>
+
+namespace mxcore {
+
+OpRegistry* OpRegistry::Get() {
+  static OpRegistry inst;
+  return &inst;
+}
+
+OpEntry& OpRegistry::Register(const std::string& name) {
+  auto* e = new OpEntry();
+  e->name = name;
+  ops_[name] = e;
+  return *e;
+}
+
+OpEntry* OpRegistry::Find(const std::string& name) {
+  auto it = ops_.find(name);
+  return it == ops_.end() ? nullptr : it->second;
+}
+
+std::vector<std::string> OpRegistry::List() const {
+  std::vector<std::string> out;
+  for (auto& kv : ops_) out.push_back(kv.first);
+  std::sort(out.begin(), out.end());
+  return out;
+}
+
+Imperative* Imperative::Get() {
+  static Imperative inst;
+  return &inst;
+}
+
+NDArray Make(const TShape& s, Context ctx, int dtype) {
+  return NDArray(s, ctx, dtype);
+}
+NDArray MakeLike(const NDArray& a) {
+  return NDArray(a.shape(), a.ctx(), a.dtype());
+}
+
+// per-device grow-only scratch arena: every user is ordered on the compute
+// stream, so reuse across ops is race-free (reference kTempSpace resource)
+namespace {
+struct Arena {
+  std::mutex mu;
+  void* ptr = nullptr;
+  size_t cap = 0;
+  Context ctx;
+};
+Arena& GetArena(const Context& c) {
+  static Arena arenas[68];
+  int idx = c.is_gpu() ? (c.dev_id & 63) : 64 + (c.dev_type & 3);
+  arenas[idx].ctx = c;
+  return arenas[idx];
+}
+void* ArenaAlloc(const Context& c, size_t n) {
+  Arena& a = GetArena(c);
+  std::lock_guard<std::mutex> g(a.mu);
+  if (a.cap < n) {
+    size_t want = std::max(n, a.cap * 2);
+    if (a.ptr) Storage::Get()->DirectFree({a.ptr, a.cap, c});
+    auto h = Storage::Get()->Alloc(want, c);
+    a.ptr = h.dptr;
+    a.cap = want;
+  }
+  return a.ptr;
+}
+}  // namespace
+
+void Imperative::PushOp(const OpEntry* op, const NodeAttrs& attrs,
+                        const std::vector<NDArray>& inputs,
+                        const std::vector<NDArray>& outputs) {
+  Context ctx = outputs.empty() ? (inputs.empty() ? Context::CPU()
+                                                  : inputs[0].ctx())
+                                : outputs[0].ctx();
+  const FCompute& fc = ctx.is_gpu() ? op->fcompute_gpu : op->fcompute_cpu;
+  MX_CHECK(fc, "op '" << op->name << "' has no " << (ctx.is_gpu() ? "GPU" : "CPU")
+                      << " implementation (native extension required)");
+
+  std::vector<TBlob> in_blobs, out_blobs;
+  in_blobs.reserve(inputs.size());
+  out_blobs.reserve(outputs.size());
+  std::vector<VarId> cvars, mvars;
+  for (auto& o : outputs) {
+    out_blobs.push_back(o.data());
+    mvars.push_back(o.var());
+  }
+  for (auto& a : inputs) {
+    in_blobs.push_back(a.data());
+    // in-place (input aliases an output): keep only the write dep
+    if (std::find(mvars.begin(), mvars.end(), a.var()) == mvars.end() &&
+        std::find(cvars.begin(), cvars.end(), a.var()) == cvars.end())
+      cvars.push_back(a.var());
+  }
+  // keep chunks alive until the op ran
+  std::vector<std::shared_ptr<NDArray::Chunk>> hold;
+  for (auto& a : inputs) hold.push_back(a.chunk_);
+  for (auto& o : outputs) hold.push_back(o.chunk_);
+
+  NodeAttrs at = attrs;  // by value: lambda outlives the caller
+  Engine::Get()->PushAsync(
+      [op, at, fc, in_blobs, out_blobs, ctx, hold](const RunContext& rc) {
+        OpCtx octx;
+        octx.rc = rc;
+        octx.workspace = [&](size_t n) { return ArenaAlloc(ctx, n); };
+        fc(at, octx, in_blobs, out_blobs);
+      },
+      ctx, cvars, mvars, FnProperty::kNormal, op->name.c_str());
+}
+
+std::vector<NDArray> Imperative::Run(const OpEntry* op, const NodeAttrs& attrs,
+                                     const std::vector<NDArray>& inputs) {
+  MX_CHECK(op->infer, "op '" << op->name << "' has no shape inference");
+  std::vector<TShape> in_shapes;
+  std::vector<int> in_dtypes;
+  for (auto& a : inputs) {
+    in_shapes.push_back(a.shape());
+    in_dtypes.push_back(a.dtype());
+  }
+  std::vector<TShape> out_shapes;
+  std::vector<int> out_dtypes;
+  op->infer(attrs, in_shapes, in_dtypes, &out_shapes, &out_dtypes);
+  Context ctx = inputs.empty() ? Context::CPU() : inputs[0].ctx();
+  if (attrs.has("__ctx_gpu__")) {  // source ops (zeros/random) carry ctx
+    int id = (int)attrs.GetInt("__ctx_gpu__", -1);
+    ctx = id >= 0 ? Context::GPU(id) : Context::CPU();
+  }
+  std::vector<NDArray> outputs;
+  for (size_t i = 0; i < out_shapes.size(); ++i)
+    outputs.emplace_back(out_shapes[i], ctx, out_dtypes[i]);
+  PushOp(op, attrs, inputs, outputs);
+  return outputs;
+}
+
+void Imperative::RunInto(const OpEntry* op, const NodeAttrs& attrs,
+                         const std::vector<NDArray>& inputs,
+                         const std::vector<NDArray>& outputs) {
+  PushOp(op, attrs, inputs, outputs);
+}
+
+std::vector<NDArray> Imperative::Invoke(const OpEntry* op,
+                                        const NodeAttrs& attrs,
+                                        const std::vector<NDArray>& inputs) {
+  auto outputs = Run(op, attrs, inputs);
+  if (recording_) {
+    // record every op: if a gradient later flows into one that has no
+    // fbackward, Backward throws instead of silently detaching
+    tape_.push_back(TapeNode{op, attrs, inputs, outputs});
+  }
+  return outputs;
+}
+
+void Imperative::InvokeInto(const OpEntry* op, const NodeAttrs& attrs,
+                            const std::vector<NDArray>& inputs,
+                            const std::vector<NDArray>& outputs) {
+  RunInto(op, attrs, inputs, outputs);
+}
+
+void Imperative::MarkVariable(const NDArray& x, const NDArray& grad, int req) {
+  leaves_[x.chunk_.get()] = LeafInfo{grad, req};
+  leaf_keepalive_[x.chunk_.get()] = x;
+}
+
+void Imperative::DropVariable(const NDArray& x) {
+  leaves_.erase(x.chunk_.get());
+  leaf_keepalive_.erase(x.chunk_.get());
+}
+
+void Imperative::ClearTape() { tape_.clear(); }
+
+void Imperative::Backward(const std::vector<NDArray>& ys,
+                          const std::vector<NDArray>& y_grads,
+                          bool retain_graph) {
+  OpEntry* add_into = OpRegistry::Get()->Find("_grad_add");
+  OpEntry* ones_op = OpRegistry::Get()->Find("ones_like");
+  MX_CHECK(add_into && ones_op, "core grad ops missing");
+
+  std::unordered_map<NDArray::Chunk*, NDArray> grads;
+  for (size_t i = 0; i < ys.size(); ++i) {
+    NDArray g;
+    if (i < y_grads.size() && !y_grads[i].is_none()) {
+      g = y_grads[i];
+    } else {
+      g = Run(ones_op, {}, {ys[i]})[0];
+    }
+    grads[ys[i].chunk_.get()] = g;
+  }
+
+  for (auto it = tape_.rbegin(); it != tape_.rend(); ++it) {
+    TapeNode& node = *it;
+    bool any = false;
+    std::vector<NDArray> ograds(node.outputs.size());
+    for (size_t i = 0; i < node.outputs.size(); ++i) {
+      auto git = grads.find(node.outputs[i].chunk_.get());
+      if (git != grads.end()) {
+        ograds[i] = git->second;
+        any = true;
+      }
+    }
+    if (!any) continue;
+    MX_CHECK(node.op->fbackward,
+             "op '" << node.op->name
+                    << "' is not differentiable but a gradient flows into it");
+    // missing head grads are zeros
+    for (size_t i = 0; i < ograds.size(); ++i)
+      if (ograds[i].is_none()) {
+        OpEntry* zeros = OpRegistry::Get()->Find("zeros_like");
+        ograds[i] = Run(zeros, {}, {node.outputs[i]})[0];
+      }
+    std::vector<NDArray> igrads = node.op->fbackward(node, ograds);
+    MX_CHECK(igrads.size() == node.inputs.size(),
+             "op '" << node.op->name << "' backward returned "
+                    << igrads.size() << " grads for " << node.inputs.size()
+                    << " inputs");
+    for (size_t i = 0; i < igrads.size(); ++i) {
+      if (igrads[i].is_none()) continue;
+      NDArray::Chunk* key = node.inputs[i].chunk_.get();
+      auto git = grads.find(key);
+      if (git == grads.end()) {
+        grads[key] = igrads[i];
+      } else {
+        // out-of-place: a stored grad may be a view of another node's
+        // ograd (reshape-style backwards), so never mutate it
+        OpEntry* add2 = OpRegistry::Get()->Find("elemwise_add");
+        grads[key] = Run(add2, {}, {git->second, igrads[i]})[0];
+      }
+    }
+  }
+
+  // write accumulated grads into attached leaf buffers
+  OpEntry* copy_op = OpRegistry::Get()->Find("_copy_into");
+  for (auto& kv : leaves_) {
+    auto git = grads.find(kv.first);
+    if (git == grads.end()) continue;
+    if (kv.second.req == 2) {  // add
+      RunInto(add_into, {}, {git->second}, {kv.second.grad});
+    } else {
+      RunInto(copy_op, {}, {git->second}, {kv.second.grad});
+    }
+  }
+  if (!retain_graph) tape_.clear();
+}
+
+}  // namespace mxcore
