@@ -1,0 +1,287 @@
+"""Native C++ runtime core: storage pool, threaded engine, NDArray,
+op registry, autograd tape (reference test model: tests/cpp/engine/
+threaded_engine_test.cc + check_numeric_gradient CPU oracles)."""
+import numpy as np
+import pytest
+
+from mxnet_amd import _core as c
+
+CPU = dict(dev_type=1, dev_id=0)
+
+
+def nd(arr):
+    return c.from_numpy(np.asarray(arr, dtype='float32'))
+
+
+# ---------------------------------------------------------------------------
+# storage
+# ---------------------------------------------------------------------------
+def test_storage_pool_reuse():
+    used0 = c.used_size(1, 0)
+    a = c.NDArray([1024], 1, 0, 0)
+    a.wait_to_read()
+    assert c.used_size(1, 0) >= used0 + 4096
+    del a
+    c.wait_all()
+    # freed memory returns to the pool, not the OS
+    assert c.pool_size(1, 0) >= 4096
+    b = c.NDArray([1024], 1, 0, 0)  # same bucket: comes from the pool
+    b.wait_to_read()
+    del b
+    c.wait_all()
+    c.release_all(1, 0)
+    assert c.pool_size(1, 0) == 0
+
+
+# ---------------------------------------------------------------------------
+# ndarray + views
+# ---------------------------------------------------------------------------
+def test_ndarray_numpy_roundtrip():
+    x = np.random.RandomState(0).randn(3, 5).astype('float32')
+    a = c.from_numpy(x)
+    assert a.shape == [3, 5] and a.dtype == 0
+    np.testing.assert_array_equal(a.asnumpy(), x)
+
+
+def test_reshape_slice_views():
+    x = np.arange(24, dtype='float32').reshape(4, 6)
+    a = c.from_numpy(x)
+    r = a.reshape([2, 12])
+    np.testing.assert_array_equal(r.asnumpy(), x.reshape(2, 12))
+    s = a.slice(1, 3)
+    np.testing.assert_array_equal(s.asnumpy(), x[1:3])
+    # views share the chunk (same engine var)
+    assert s.var == a.var
+
+
+def test_copyto():
+    a = nd(np.arange(10))
+    b = c.NDArray([10], 1, 0, 0)
+    a.copyto(b)
+    np.testing.assert_array_equal(b.asnumpy(), np.arange(10, dtype='float32'))
+
+
+# ---------------------------------------------------------------------------
+# elementwise / broadcast / reduce ops vs numpy oracle
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize('op,fn', [
+    ('elemwise_add', np.add), ('elemwise_sub', np.subtract),
+    ('elemwise_mul', np.multiply), ('elemwise_div', np.divide),
+    ('maximum', np.maximum), ('minimum', np.minimum),
+])
+def test_binary_ops(op, fn):
+    rs = np.random.RandomState(1)
+    x = rs.randn(4, 5).astype('float32')
+    y = rs.rand(4, 5).astype('float32') + 0.5
+    out = c.invoke(op, [nd(x), nd(y)], {})[0].asnumpy()
+    np.testing.assert_allclose(out, fn(x, y), rtol=1e-6)
+
+
+def test_broadcast_binary():
+    rs = np.random.RandomState(2)
+    x = rs.randn(4, 3, 5).astype('float32')
+    y = rs.randn(1, 3, 1).astype('float32')
+    out = c.invoke('elemwise_add', [nd(x), nd(y)], {})[0].asnumpy()
+    np.testing.assert_allclose(out, x + y, rtol=1e-6)
+    # bias-style trailing broadcast
+    b = rs.randn(5).astype('float32')
+    out = c.invoke('elemwise_add', [nd(x), nd(b)], {})[0].asnumpy()
+    np.testing.assert_allclose(out, x + b, rtol=1e-6)
+
+
+@pytest.mark.parametrize('op,fn', [
+    ('relu', lambda x: np.maximum(x, 0)),
+    ('sigmoid', lambda x: 1 / (1 + np.exp(-x))),
+    ('tanh', np.tanh), ('exp', np.exp), ('square', np.square),
+    ('negative', np.negative), ('abs', np.abs),
+])
+def test_unary_ops(op, fn):
+    x = np.random.RandomState(3).randn(7, 9).astype('float32')
+    out = c.invoke(op, [nd(x)], {})[0].asnumpy()
+    np.testing.assert_allclose(out, fn(x), rtol=1e-5, atol=1e-6)
+
+
+def test_scalar_ops():
+    x = np.random.RandomState(4).rand(5, 5).astype('float32') + 1
+    got = c.invoke('_mul_scalar', [nd(x)], {'alpha': '2.5'})[0].asnumpy()
+    np.testing.assert_allclose(got, x * 2.5, rtol=1e-6)
+    got = c.invoke('clip', [nd(x)], {'alpha': '1.2', 'beta': '1.6'})[0].asnumpy()
+    np.testing.assert_allclose(got, np.clip(x, 1.2, 1.6), rtol=1e-6)
+
+
+@pytest.mark.parametrize('axis,keepdims', [
+    ('()', False), ('(0,)', False), ('(1,)', True), ('(0,2)', False)])
+def test_sum_axes(axis, keepdims):
+    x = np.random.RandomState(5).randn(3, 4, 5).astype('float32')
+    ax = eval(axis)
+    want = x.sum(axis=ax if ax else None, keepdims=keepdims)
+    if not ax and not keepdims:
+        want = want.reshape(1)
+    got = c.invoke('sum', [nd(x)],
+                   {'axis': axis, 'keepdims': '1' if keepdims else '0'})[0]
+    np.testing.assert_allclose(got.asnumpy().squeeze(), np.asarray(want).squeeze(),
+                               rtol=1e-5)
+
+
+def test_mean_max_min():
+    x = np.random.RandomState(6).randn(6, 7).astype('float32')
+    np.testing.assert_allclose(
+        c.invoke('mean', [nd(x)], {'axis': '(1,)'})[0].asnumpy(),
+        x.mean(axis=1), rtol=1e-5)
+    np.testing.assert_allclose(
+        c.invoke('max', [nd(x)], {'axis': '(0,)'})[0].asnumpy(),
+        x.max(axis=0), rtol=1e-6)
+    np.testing.assert_allclose(
+        c.invoke('min', [nd(x)], {})[0].asnumpy(), [x.min()], rtol=1e-6)
+
+
+def test_transpose():
+    x = np.random.RandomState(7).randn(2, 3, 4).astype('float32')
+    got = c.invoke('transpose', [nd(x)], {'axes': '(2,0,1)'})[0].asnumpy()
+    np.testing.assert_array_equal(got, x.transpose(2, 0, 1))
+    got = c.invoke('transpose', [nd(x)], {})[0].asnumpy()
+    np.testing.assert_array_equal(got, x.T)
+
+
+def test_cast():
+    x = np.random.RandomState(8).randn(4, 4).astype('float32') * 10
+    got = c.invoke('cast', [nd(x)], {'dtype': '4'})[0]  # int32
+    assert got.dtype == 4
+    np.testing.assert_array_equal(got.asnumpy(), x.astype('int32'))
+
+
+def test_fills_and_random():
+    z = c.invoke('_full', [], {'shape': '(3,4)', 'value': '7', 'dtype': '0'})[0]
+    np.testing.assert_array_equal(z.asnumpy(), np.full((3, 4), 7, 'float32'))
+    u = c.invoke('_random_uniform', [],
+                 {'shape': '(10000,)', 'low': '0', 'high': '1',
+                  'seed': '42', 'dtype': '0'})[0].asnumpy()
+    assert 0 <= u.min() and u.max() < 1 and abs(u.mean() - 0.5) < 0.02
+    n = c.invoke('_random_normal', [],
+                 {'shape': '(10000,)', 'loc': '1', 'scale': '2',
+                  'seed': '43', 'dtype': '0'})[0].asnumpy()
+    assert abs(n.mean() - 1) < 0.1 and abs(n.std() - 2) < 0.1
+    # counter-based: same seed, same stream
+    u2 = c.invoke('_random_uniform', [],
+                  {'shape': '(10000,)', 'low': '0', 'high': '1',
+                   'seed': '42', 'dtype': '0'})[0].asnumpy()
+    np.testing.assert_array_equal(u, u2)
+
+
+# ---------------------------------------------------------------------------
+# autograd tape
+# ---------------------------------------------------------------------------
+def _attach(a):
+    g = c.invoke('zeros_like', [a], {})[0]
+    c.mark_variable(a, g, 1)
+    return g
+
+
+def test_backward_chain():
+    x = nd([[1., 2.], [3., 4.]])
+    w = nd([[2., 0.], [1., 3.]])
+    gx, gw = _attach(x), _attach(w)
+    c.set_recording(True)
+    y = c.invoke('elemwise_mul', [x, w], {})[0]
+    z = c.invoke('relu', [y], {})[0]
+    L = c.invoke('sum', [z], {})[0]
+    c.set_recording(False)
+    c.backward([L], [], False)
+    c.wait_all()
+    np.testing.assert_allclose(gx.asnumpy(), [[2, 0], [1, 3]])
+    np.testing.assert_allclose(gw.asnumpy(), [[1, 0], [3, 4]])
+    c.drop_variable(x)
+    c.drop_variable(w)
+
+
+def test_backward_broadcast_and_fanout():
+    # y = x*b + x (residual fan-out; b broadcasts) ; L = sum(y)
+    rs = np.random.RandomState(9)
+    xv = rs.randn(3, 4).astype('float32')
+    bv = rs.randn(4).astype('float32')
+    x, b = nd(xv), nd(bv)
+    gx, gb = _attach(x), _attach(b)
+    c.set_recording(True)
+    t = c.invoke('elemwise_mul', [x, b], {})[0]
+    y = c.invoke('elemwise_add', [t, x], {})[0]
+    L = c.invoke('sum', [y], {})[0]
+    c.set_recording(False)
+    c.backward([L], [], False)
+    c.wait_all()
+    np.testing.assert_allclose(gx.asnumpy(), np.tile(bv + 1, (3, 1)),
+                               rtol=1e-6)
+    np.testing.assert_allclose(gb.asnumpy(), xv.sum(0), rtol=1e-5)
+    c.drop_variable(x)
+    c.drop_variable(b)
+
+
+def test_numeric_gradient():
+    # finite differences vs the tape (reference check_numeric_gradient)
+    rs = np.random.RandomState(10)
+    xv = rs.rand(5, 3).astype('float32') + 0.5
+
+    def f(v):
+        return float(np.sum(np.tanh(v) ** 2))
+
+    x = nd(xv)
+    gx = _attach(x)
+    c.set_recording(True)
+    t = c.invoke('tanh', [x], {})[0]
+    s = c.invoke('square', [t], {})[0]
+    L = c.invoke('sum', [s], {})[0]
+    c.set_recording(False)
+    c.backward([L], [], False)
+    c.wait_all()
+    got = gx.asnumpy()
+    eps = 1e-3
+    num = np.zeros_like(xv)
+    for i in range(xv.shape[0]):
+        for j in range(xv.shape[1]):
+            xp = xv.copy(); xp[i, j] += eps
+            xm = xv.copy(); xm[i, j] -= eps
+            num[i, j] = (f(xp) - f(xm)) / (2 * eps)
+    np.testing.assert_allclose(got, num, rtol=1e-2, atol=1e-3)
+    c.drop_variable(x)
+
+
+def test_nondiff_error():
+    x = nd([1., 2.])
+    g = _attach(x)
+    c.set_recording(True)
+    y = c.invoke('greater', [x, nd([1.5, 1.5])], {})[0]
+    L = c.invoke('sum', [y], {})[0]
+    c.set_recording(False)
+    # greater is registered non-differentiable: returns empty grads, so
+    # backward completes with no grad written into x
+    c.backward([L], [], False)
+    c.wait_all()
+    np.testing.assert_array_equal(g.asnumpy(), [0., 0.])
+    c.drop_variable(x)
+
+
+# ---------------------------------------------------------------------------
+# engine semantics
+# ---------------------------------------------------------------------------
+def test_engine_ordering_stress():
+    # many chained in-place adds must serialize through var deps
+    a = nd(np.zeros(16))
+    one = nd(np.ones(16))
+    for _ in range(200):
+        c.invoke_into('_grad_add', [one], [a], {})
+    a.wait_to_read()
+    np.testing.assert_array_equal(a.asnumpy(), np.full(16, 200, 'float32'))
+
+
+def test_engine_exception_propagation():
+    a = nd(np.ones(4))
+    with pytest.raises(Exception):
+        # unknown op raises synchronously
+        c.invoke('definitely_not_an_op', [a], {})
+
+
+def test_version_bumps():
+    a = nd(np.zeros(4))
+    v0 = c.var_version(a.var)
+    c.invoke_into('_grad_add', [nd(np.ones(4))], [a], {})
+    a.wait_to_read()
+    assert c.var_version(a.var) > v0
