@@ -15,10 +15,12 @@ CORE_SRCS := src/core/storage.cc src/core/engine.cc src/core/ndarray.cc \
              src/core/op.cc
 OPS_SRCS  := $(wildcard src/ops/*.hip)
 PYBIND_SRC := src/core/pybind.cc
+RAW_SRC    := src/core/raw_bind.cc
 
 CORE_OBJS := $(patsubst src/%.cc,build/core/%.o,$(CORE_SRCS))
 OPS_OBJS  := $(patsubst src/%.hip,build/core/%.o,$(OPS_SRCS))
 PY_OBJ    := build/core/pybind.o
+RAW_OBJ   := build/core/raw_bind.o
 
 TARGET := mxnet_amd/_core$(EXT)
 
@@ -36,7 +38,11 @@ $(PY_OBJ): $(PYBIND_SRC) src/core/*.h
 	@mkdir -p $(dir $@)
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 
-$(TARGET): $(CORE_OBJS) $(OPS_OBJS) $(PY_OBJ)
+$(RAW_OBJ): $(RAW_SRC) src/core/*.h src/ops/*.h
+	@mkdir -p $(dir $@)
+	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
+
+$(TARGET): $(CORE_OBJS) $(OPS_OBJS) $(PY_OBJ) $(RAW_OBJ)
 	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64
 
 clean:

@@ -41,34 +41,57 @@ NDArray MakeLike(const NDArray& a) {
   return NDArray(a.shape(), a.ctx(), a.dtype());
 }
 
-// per-device grow-only scratch arena: every user is ordered on the compute
-// stream, so reuse across ops is race-free (reference kTempSpace resource)
+// Per-device grow-only scratch arena for GPU compute ops (reference
+// kTempSpace resource): every user runs on that device's single compute
+// worker in stream order, so cursor-reset-per-op reuse is race-free.
+// Blocks are never reallocated (pointers handed out stay valid); capacity
+// converges to the largest step's scratch footprint.
 namespace {
+struct ArenaBlock {
+  void* ptr;
+  size_t cap;
+  size_t used;
+};
 struct Arena {
-  std::mutex mu;
-  void* ptr = nullptr;
-  size_t cap = 0;
+  std::mutex mu;  // shim path (python thread) vs compute worker
+  std::vector<ArenaBlock> blocks;
+  size_t total = 0;
   Context ctx;
 };
-Arena& GetArena(const Context& c) {
-  static Arena arenas[68];
+Arena& GetArena(const Context& c, int lane) {
+  // lane 0 = engine compute worker, lane 1 = frontend shim (torch stream):
+  // separate arenas so the two stream domains never alias scratch
+  static Arena arenas[136];
   int idx = c.is_gpu() ? (c.dev_id & 63) : 64 + (c.dev_type & 3);
+  idx += lane * 68;
   arenas[idx].ctx = c;
   return arenas[idx];
 }
-void* ArenaAlloc(const Context& c, size_t n) {
-  Arena& a = GetArena(c);
-  std::lock_guard<std::mutex> g(a.mu);
-  if (a.cap < n) {
-    size_t want = std::max(n, a.cap * 2);
-    if (a.ptr) Storage::Get()->DirectFree({a.ptr, a.cap, c});
-    auto h = Storage::Get()->Alloc(want, c);
-    a.ptr = h.dptr;
-    a.cap = want;
-  }
-  return a.ptr;
-}
 }  // namespace
+
+void ArenaReset(const Context& c, int lane) {
+  Arena& a = GetArena(c, lane);
+  std::lock_guard<std::mutex> g(a.mu);
+  for (auto& b : a.blocks) b.used = 0;
+}
+
+void* ArenaAlloc(const Context& c, size_t n, int lane) {
+  n = (n + 255) & ~(size_t)255;  // 256-B aligned carve-outs
+  Arena& a = GetArena(c, lane);
+  std::lock_guard<std::mutex> g(a.mu);
+  for (auto& b : a.blocks) {
+    if (b.cap - b.used >= n) {
+      void* p = (char*)b.ptr + b.used;
+      b.used += n;
+      return p;
+    }
+  }
+  size_t want = std::max(n, a.total);  // at least double the arena
+  auto h = Storage::Get()->Alloc(want, c);
+  a.blocks.push_back({h.dptr, want, n});
+  a.total += want;
+  return h.dptr;
+}
 
 void Imperative::PushOp(const OpEntry* op, const NodeAttrs& attrs,
                         const std::vector<NDArray>& inputs,
@@ -105,7 +128,18 @@ void Imperative::PushOp(const OpEntry* op, const NodeAttrs& attrs,
       [op, at, fc, in_blobs, out_blobs, ctx, hold](const RunContext& rc) {
         OpCtx octx;
         octx.rc = rc;
-        octx.workspace = [&](size_t n) { return ArenaAlloc(ctx, n); };
+        std::vector<std::unique_ptr<char[]>> cpu_scratch;
+        if (ctx.is_gpu()) {
+          // single compute worker per device: cursor reset is race-free
+          ArenaReset(ctx);
+          octx.workspace = [&](size_t n) { return ArenaAlloc(ctx, n); };
+        } else {
+          // CPU pool has several workers: per-op heap scratch
+          octx.workspace = [&](size_t n) -> void* {
+            cpu_scratch.emplace_back(new char[n]);
+            return cpu_scratch.back().get();
+          };
+        }
         fc(at, octx, in_blobs, out_blobs);
       },
       ctx, cvars, mvars, FnProperty::kNormal, op->name.c_str());

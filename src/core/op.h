@@ -173,4 +173,9 @@ class Imperative {
 NDArray MakeLike(const NDArray& a);
 NDArray Make(const TShape& s, Context ctx, int dtype);
 
+// scratch arenas (see op.cc): lane 0 = engine compute worker,
+// lane 1 = frontend shim path on the torch stream
+void ArenaReset(const Context& c, int lane = 0);
+void* ArenaAlloc(const Context& c, size_t n, int lane = 0);
+
 }  // namespace mxcore

@@ -113,7 +113,10 @@ void InvokeIntoPy(const std::string& name, const std::vector<NDArray>& inputs,
 
 }  // namespace
 
+void init_raw(py::module_& m);
+
 PYBIND11_MODULE(_core, m) {
+  init_raw(m);
   m.doc() = "mxnet_amd native runtime (storage + engine + ndarray + ops)";
 
   py::class_<NDArray>(m, "NDArray")

@@ -1,0 +1,340 @@
+// Row softmax / log-softmax fwd+bwd, and column-sum (bias gradient).
+//
+// Reference parity: src/operator/nn/softmax-inl.h:351-820 (softmax_compute,
+// softmax_gradient), broadcast_reduce sum for the bias grad.
+//
+// MI355X design: one 256-thread block per row (4 waves), fp32 accumulation,
+// wave64 __shfl_xor + LDS block reduction, 8-wide vector loads for 16-bit
+// dtypes (Guideline 13).  Rows are the contiguous last axis.
+#include "native_common.h"
+
+using namespace mxcore;
+
+template <typename T, int VEC>
+__global__ void softmax_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   long rows, long C, bool log_mode,
+                                   float invT,
+                                   const unsigned char* __restrict__ mask) {
+  __shared__ float sred[16];
+  using VecT = T __attribute__((ext_vector_type(VEC)));
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* xr = x + r * C;
+    const unsigned char* mr = mask ? mask + r * C : nullptr;
+    T* yr = y + r * C;
+    long cv = C / VEC;
+    // pass 1: max
+    float m = -INFINITY;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        if (!mr || mr[i * VEC + j]) m = fmaxf(m, (float)v[j]);
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      if (!mr || mr[i]) m = fmaxf(m, (float)xr[i]);
+    m = block_reduce(m, sred, MaxOp(), -INFINITY);
+    // pass 2: sum of exp
+    float s = 0.f;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        if (!mr || mr[i * VEC + j])
+          s += __expf(((float)v[j] - m) * invT);
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      if (!mr || mr[i]) s += __expf(((float)xr[i] - m) * invT);
+    __syncthreads();  // reuse of sred
+    s = block_reduce(s, sred, SumOp(), 0.f);
+    float inv_s = s > 0.f ? 1.f / s : 0.f, log_s = __logf(s);
+    // pass 3: write (masked entries get 0 / -inf)
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT v = reinterpret_cast<const VecT*>(xr)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        bool on = !mr || mr[i * VEC + j];
+        float z = ((float)v[j] - m) * invT;
+        o[j] = (T)(log_mode ? (on ? z - log_s : -INFINITY)
+                            : (on ? __expf(z) * inv_s : 0.f));
+      }
+      reinterpret_cast<VecT*>(yr)[i] = o;
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      bool on = !mr || mr[i];
+      float z = ((float)xr[i] - m) * invT;
+      yr[i] = (T)(log_mode ? (on ? z - log_s : -INFINITY)
+                           : (on ? __expf(z) * inv_s : 0.f));
+    }
+    __syncthreads();
+  }
+}
+
+// softmax:     dx = (dy - sum(dy*y)) * y * invT
+// log_softmax: dx = (dy - exp(y) * sum(dy)) * invT
+template <typename T, int VEC>
+__global__ void softmax_bwd_kernel(const T* __restrict__ dy,
+                                   const T* __restrict__ y,
+                                   T* __restrict__ dx, long rows, long C,
+                                   bool log_mode, float invT) {
+  __shared__ float sred[16];
+  using VecT = T __attribute__((ext_vector_type(VEC)));
+  for (long r = blockIdx.x; r < rows; r += gridDim.x) {
+    const T* gr = dy + r * C;
+    const T* yr = y + r * C;
+    T* dr = dx + r * C;
+    long cv = C / VEC;
+    float s = 0.f;
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT g = reinterpret_cast<const VecT*>(gr)[i];
+      VecT v = reinterpret_cast<const VecT*>(yr)[i];
+#pragma unroll
+      for (int j = 0; j < VEC; ++j)
+        s += log_mode ? (float)g[j] : (float)g[j] * (float)v[j];
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x)
+      s += log_mode ? (float)gr[i] : (float)gr[i] * (float)yr[i];
+    s = block_reduce(s, sred, SumOp(), 0.f);
+    for (long i = threadIdx.x; i < cv; i += blockDim.x) {
+      VecT g = reinterpret_cast<const VecT*>(gr)[i];
+      VecT v = reinterpret_cast<const VecT*>(yr)[i];
+      VecT o;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        float d = log_mode ? (float)g[j] - __expf((float)v[j]) * s
+                           : ((float)g[j] - s) * (float)v[j];
+        o[j] = (T)(d * invT);
+      }
+      reinterpret_cast<VecT*>(dr)[i] = o;
+    }
+    for (long i = cv * VEC + threadIdx.x; i < C; i += blockDim.x) {
+      float d = log_mode ? (float)gr[i] - __expf((float)yr[i]) * s
+                         : ((float)gr[i] - s) * (float)yr[i];
+      dr[i] = (T)(d * invT);
+    }
+    __syncthreads();
+  }
+}
+
+// small-C fast path: the whole row lives in registers (VPT values per
+// lane, one wave per row, 4 rows per block) — ONE global read of x (+
+// mask) instead of the generic kernel's three passes.  BERT's masked
+// attention softmax (C = seq = 128) measured 2.05 ms/step on the
+// generic kernel; this is the fix.
+template <typename T, int VPT>
+__global__ void softmax_fwd_rowreg_kernel(
+    const T* __restrict__ x, T* __restrict__ y, long rows, int C,
+    bool log_mode, float invT, const unsigned char* __restrict__ mask) {
+  const int lane = threadIdx.x & 63;
+  const int row_in_blk = threadIdx.x >> 6;
+  for (long r = (long)blockIdx.x * 4 + row_in_blk; r < rows;
+       r += (long)gridDim.x * 4) {
+    const T* xr = x + r * C;
+    const unsigned char* mr = mask ? mask + r * C : nullptr;
+    float v[VPT];
+    bool on[VPT];
+    float m = -INFINITY;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      const bool in = c < C;
+      on[j] = in && (!mr || mr[c]);
+      v[j] = on[j] ? (float)xr[c] : -INFINITY;
+      m = fmaxf(m, v[j]);
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      m = fmaxf(m, __shfl_xor(m, off));
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      v[j] = on[j] ? __expf((v[j] - m) * invT) : 0.f;
+      s += v[j];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      s += __shfl_xor(s, off);
+    const float inv_s = s > 0.f ? 1.f / s : 0.f;
+    const float log_s = __logf(s);
+    T* yr = y + r * C;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      if (c < C)
+        yr[c] = (T)(log_mode
+                        ? (on[j] ? (__logf(v[j]) /*=(x-m)invT*/) - log_s
+                                 : -INFINITY)
+                        : v[j] * inv_s);
+    }
+  }
+}
+
+// backward fast path: dx = (dy - sum(dy*y)) * y * invT, one read of each
+template <typename T, int VPT>
+__global__ void softmax_bwd_rowreg_kernel(const T* __restrict__ dy,
+                                          const T* __restrict__ yv,
+                                          T* __restrict__ dx, long rows,
+                                          int C, float invT) {
+  const int lane = threadIdx.x & 63;
+  const int row_in_blk = threadIdx.x >> 6;
+  for (long r = (long)blockIdx.x * 4 + row_in_blk; r < rows;
+       r += (long)gridDim.x * 4) {
+    const T* gr = dy + r * C;
+    const T* yr = yv + r * C;
+    float g[VPT], yy[VPT];
+    float s = 0.f;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      g[j] = c < C ? (float)gr[c] : 0.f;
+      yy[j] = c < C ? (float)yr[c] : 0.f;
+      s += g[j] * yy[j];
+    }
+#pragma unroll
+    for (int off = 32; off; off >>= 1)
+      s += __shfl_xor(s, off);
+    T* dr = dx + r * C;
+#pragma unroll
+    for (int j = 0; j < VPT; ++j) {
+      const int c = lane + j * 64;
+      if (c < C) dr[c] = (T)((g[j] - s) * yy[j] * invT);
+    }
+  }
+}
+
+
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ in, float* __restrict__ out,
+                              long M, long N, long rows_per_block) {
+  long r0 = (long)blockIdx.y * rows_per_block;
+  long r1 = min(M, r0 + rows_per_block);
+  for (long c = (long)blockIdx.x * blockDim.x + threadIdx.x; c < N;
+       c += (long)gridDim.x * blockDim.x) {
+    float acc = 0.f;
+    for (long r = r0; r < r1; ++r) acc += (float)in[r * N + c];
+    if (gridDim.y == 1) out[c] = acc;
+    else atomicAdd(out + c, acc);
+  }
+}
+
+
+// ===========================================================================
+// native host launchers
+// ===========================================================================
+#include <algorithm>
+
+#include "ops_api.h"
+
+template <typename T>
+__global__ void colsum_cast_kernel(const float* __restrict__ x,
+                                   T* __restrict__ y, long n) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x)
+    y[i] = (T)x[i];
+}
+
+namespace mxcore {
+
+void softmax_fwd_raw(const LaunchCtx& lc, const Arr& x, const Arr& mask,
+                     bool log_mode, double temperature, const Arr& y) {
+  long C = x.size(-1), rows = x.numel() / (C > 0 ? C : 1);
+  if (x.numel() == 0) return;
+  const unsigned char* mp = nullptr;
+  if (mask.defined() && mask.numel() > 0) {
+    MX_CHECK(mask.dtype == kUint8 || mask.dtype == kBool,
+             "softmax mask must be uint8/bool");
+    MX_CHECK(mask.numel() == x.numel(), "softmax mask shape mismatch");
+    mp = mask.data<unsigned char>();
+  }
+  int grid = (int)std::min<long>(rows, 4096);
+  float invT = (float)(1.0 / temperature);
+  if (C <= 256 && !log_mode && rows >= 64) {
+    // register-resident rows: one read, one write
+    int g4 = (int)std::min<long>((rows + 3) / 4, 4096);
+    DISPATCH_FLOAT_NATIVE(x.dtype, "softmax_fwd_rr", [&] {
+      if (C <= 64)
+        softmax_fwd_rowreg_kernel<scalar_t, 1><<<g4, 256, 0, lc.stream>>>(
+            x.data<scalar_t>(), (scalar_t*)y.ptr, rows, (int)C, log_mode,
+            invT, mp);
+      else if (C <= 128)
+        softmax_fwd_rowreg_kernel<scalar_t, 2><<<g4, 256, 0, lc.stream>>>(
+            x.data<scalar_t>(), (scalar_t*)y.ptr, rows, (int)C, log_mode,
+            invT, mp);
+      else
+        softmax_fwd_rowreg_kernel<scalar_t, 4><<<g4, 256, 0, lc.stream>>>(
+            x.data<scalar_t>(), (scalar_t*)y.ptr, rows, (int)C, log_mode,
+            invT, mp);
+    });
+    HIP_CHECK_LAST();
+    return;
+  }
+  DISPATCH_FLOAT_NATIVE(x.dtype, "softmax_fwd", [&] {
+    constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
+    softmax_fwd_kernel<scalar_t, VEC><<<grid, 256, 0, lc.stream>>>(
+        x.data<scalar_t>(), (scalar_t*)y.ptr, rows, C, log_mode, invT, mp);
+  });
+  HIP_CHECK_LAST();
+}
+
+void softmax_bwd_raw(const LaunchCtx& lc, const Arr& dy, const Arr& y,
+                     bool log_mode, double temperature, const Arr& dx) {
+  long C = dy.size(-1), rows = dy.numel() / (C > 0 ? C : 1);
+  if (dy.numel() == 0) return;
+  int grid = (int)std::min<long>(rows, 4096);
+  float invT = (float)(1.0 / temperature);
+  if (C <= 256 && !log_mode && rows >= 64) {
+    int g4 = (int)std::min<long>((rows + 3) / 4, 4096);
+    DISPATCH_FLOAT_NATIVE(dy.dtype, "softmax_bwd_rr", [&] {
+      if (C <= 64)
+        softmax_bwd_rowreg_kernel<scalar_t, 1><<<g4, 256, 0, lc.stream>>>(
+            dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
+            rows, (int)C, invT);
+      else if (C <= 128)
+        softmax_bwd_rowreg_kernel<scalar_t, 2><<<g4, 256, 0, lc.stream>>>(
+            dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
+            rows, (int)C, invT);
+      else
+        softmax_bwd_rowreg_kernel<scalar_t, 4><<<g4, 256, 0, lc.stream>>>(
+            dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
+            rows, (int)C, invT);
+    });
+    HIP_CHECK_LAST();
+    return;
+  }
+  DISPATCH_FLOAT_NATIVE(dy.dtype, "softmax_bwd", [&] {
+    constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
+    softmax_bwd_kernel<scalar_t, VEC><<<grid, 256, 0, lc.stream>>>(
+        dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr, rows, C,
+        log_mode, invT);
+  });
+  HIP_CHECK_LAST();
+}
+
+void colsum_raw(const LaunchCtx& lc, const Arr& in, const Arr& out) {
+  long N = in.size(-1), M = in.numel() / (N > 0 ? N : 1);
+  float* acc;
+  bool direct = out.dtype == kFloat32;
+  if (direct) acc = out.data<float>();
+  else acc = (float*)lc.workspace((size_t)N * 4);
+  MX_HIP_CALL(hipMemsetAsync(acc, 0, (size_t)N * 4, lc.stream));
+  long target_blocks = 2048;
+  long xblocks = (N + 255) / 256;
+  long yblocks = std::max<long>(
+      1, std::min<long>(M, target_blocks / std::max<long>(xblocks, 1)));
+  long rows_per_block = (M + yblocks - 1) / yblocks;
+  dim3 grid((unsigned)std::min<long>(xblocks, 65535), (unsigned)yblocks);
+  DISPATCH_FLOAT_NATIVE(in.dtype, "colsum", [&] {
+    colsum_kernel<scalar_t><<<grid, 256, 0, lc.stream>>>(
+        in.data<scalar_t>(), acc, M, N, rows_per_block);
+  });
+  HIP_CHECK_LAST();
+  if (!direct) {
+    DISPATCH_HALF_NATIVE(out.dtype, "colsum_cast", [&] {
+      colsum_cast_kernel<scalar_t><<<ew_grid_n(N), 256, 0, lc.stream>>>(
+          acc, (scalar_t*)out.ptr, N);
+    });
+    HIP_CHECK_LAST();
+  }
+}
+
+}  // namespace mxcore
