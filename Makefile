@@ -8,7 +8,7 @@ EXT      := $(shell python3 -c "import sysconfig; print(sysconfig.get_config_var
 NUMPYINC := $(shell python3 -c "import numpy; print(numpy.get_include())")
 
 CXXFLAGS := -fPIC -O3 -std=c++17 --offload-arch=$(ARCH) \
-            -I src -I mxnet_amd/ops/hip -I $(PYINC) -I $(PYBIND) -I $(NUMPYINC) \
+            -I src -I $(PYINC) -I $(PYBIND) -I $(NUMPYINC) \
             -Wno-unused-result -parallel-jobs=4
 
 CORE_SRCS := src/core/storage.cc src/core/engine.cc src/core/ndarray.cc \
@@ -30,7 +30,7 @@ build/core/%.o: src/%.cc src/core/*.h
 	@mkdir -p $(dir $@)
 	$(HIPCC) $(CXXFLAGS) -c $< -o $@
 
-build/core/%.o: src/%.hip src/core/*.h src/ops/*.h mxnet_amd/ops/hip/common.h
+build/core/%.o: src/%.hip src/core/*.h src/ops/*.h src/ops/common.h
 	@mkdir -p $(dir $@)
 	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
 

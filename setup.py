@@ -1,36 +1,19 @@
-"""Build the in-tree gfx950 HIP extension ``mxnet_amd._hipops``.
+"""Build the host-side C++ helper extensions (engine/dataloader pybind).
 
-Usage:  PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace
-
-Every kernel is hand-written CDNA4 HIP (MFMA / LDS / wave64) — no hipify,
-no CUDA sources.  torch.utils.cpp_extension drives hipcc and links
-against the PyTorch-ROCm runtime so tensors pass straight through.
+The GPU kernel library ``mxnet_amd._core`` is built by the top-level
+Makefile with hipcc alone (``make`` — no torch toolchain, no hipify);
+this setup.py only covers the two pure-CPU pybind11 modules.
 """
 import os
-import glob
 
 from setuptools import setup
-
-os.environ.setdefault('PYTORCH_ROCM_ARCH', 'gfx950')
-
-from torch.utils.cpp_extension import BuildExtension, CUDAExtension  # noqa: E402
-from pybind11.setup_helpers import Pybind11Extension  # noqa: E402
+from pybind11.setup_helpers import Pybind11Extension
 
 ROOT = os.path.dirname(os.path.abspath(__file__))
-SRC = sorted(glob.glob(os.path.join(ROOT, 'mxnet_amd', 'ops', 'hip', '*.hip'))) \
-    + [os.path.join(ROOT, 'mxnet_amd', 'ops', 'hip', 'bind.cpp')]
 
 setup(
-    name='mxnet_amd_hipops',
+    name='mxnet_amd_host',
     ext_modules=[
-        CUDAExtension(
-            name='mxnet_amd._hipops',
-            sources=SRC,
-            extra_compile_args={
-                'cxx': ['-O3', '-std=c++17'],
-                'nvcc': ['-O3', '-std=c++17', '--offload-arch=gfx950'],
-            },
-        ),
         Pybind11Extension(
             'mxnet_amd._engine',
             [os.path.join(ROOT, 'src', 'engine.cc')],
@@ -44,5 +27,4 @@ setup(
             extra_compile_args=['-O2', '-pthread'],
         ),
     ],
-    cmdclass={'build_ext': BuildExtension.with_options(use_ninja=True)},
 )

@@ -10,7 +10,7 @@ import torch
 pytestmark = pytest.mark.gpu
 
 if torch.cuda.is_available():
-    from mxnet_amd import _hipops as ext
+    from mxnet_amd.ops import hipshim as ext
 else:
     ext = None
 
@@ -475,7 +475,7 @@ def test_hybridize_hipgraph_inference():
 
 def test_int8_gemm_and_quantize():
     """int8 MFMA GEMM vs fp32 oracle (reference quantized FC path)."""
-    from mxnet_amd import _hipops as hx
+    from mxnet_amd.ops import hipshim as hx
     M, N, K = 128, 96, 256
     a = (torch.randn(M, K, device=DEV) * 20).clamp(-127, 127).round().to(torch.int8)
     b = (torch.randn(N, K, device=DEV) * 20).clamp(-127, 127).round().to(torch.int8)
@@ -504,7 +504,7 @@ def test_gemm_nt_8phase_path(dtype, mnk):
 
 def test_multi_sgd_matches_single():
     """multi-tensor SGD == per-tensor fused SGD (reference multi_sgd)."""
-    from mxnet_amd import _hipops as hx
+    from mxnet_amd.ops import hipshim as hx
     torch.manual_seed(5)
     shapes = [(1000,), (64, 32), (7,), (128, 3, 3, 8)]
     ws = [torch.randn(*s, device=DEV).half() for s in shapes]
