@@ -292,10 +292,11 @@ def test_sgd_update():
     mom = torch.randn(n, device=DEV)
     m2, mom2 = master.clone(), mom.clone()
     ext.sgd_update(w, master, g, mom, 0.1, 0.9, 1e-4, 0.5, 0.0)
-    # oracle
+    # oracle — reference sgd_mom_update rule: lr folded into the momentum
+    # buffer (mom = mu*mom - lr*g; w += mom)
     ge = g.float() * 0.5 + 1e-4 * m2
-    mom2 = mom2 * 0.9 + ge
-    m2 = m2 - 0.1 * mom2
+    mom2 = mom2 * 0.9 - 0.1 * ge
+    m2 = m2 + mom2
     check(master, m2, tol=1e-5)
     check(mom, mom2, tol=1e-5)
     check(w, m2, tol=2e-3)
