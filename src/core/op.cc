@@ -111,8 +111,18 @@ void Imperative::PushOp(const OpEntry* op, const NodeAttrs& attrs,
     out_blobs.push_back(o.data());
     mvars.push_back(o.var());
   }
-  for (auto& a : inputs) {
+  for (size_t i = 0; i < inputs.size(); ++i) {
+    const NDArray& a = inputs[i];
     in_blobs.push_back(a.data());
+    bool mut = std::find(op->mutate_inputs.begin(), op->mutate_inputs.end(),
+                         (int)i) != op->mutate_inputs.end();
+    if (mut) {
+      auto cit = std::find(cvars.begin(), cvars.end(), a.var());
+      if (cit != cvars.end()) cvars.erase(cit);  // write dep supersedes
+      if (std::find(mvars.begin(), mvars.end(), a.var()) == mvars.end())
+        mvars.push_back(a.var());
+      continue;
+    }
     // in-place (input aliases an output): keep only the write dep
     if (std::find(mvars.begin(), mvars.end(), a.var()) == mvars.end() &&
         std::find(cvars.begin(), cvars.end(), a.var()) == cvars.end())

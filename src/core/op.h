@@ -85,6 +85,9 @@ struct OpEntry {
   std::string name;
   int n_in = -1;   // -1 = variadic
   int n_out = 1;
+  // input positions the op WRITES (BatchNorm running stats, optimizer
+  // states): tracked as engine mutable vars instead of read deps
+  std::vector<int> mutate_inputs;
   FInferShape infer;
   FCompute fcompute_gpu;
   FCompute fcompute_cpu;

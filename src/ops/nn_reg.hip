@@ -1,0 +1,1470 @@
+// NN operator registrations for the native imperative path: GPU FCompute
+// dispatches to the raw CDNA4 launchers (ops_api.h), CPU FCompute is a
+// plain-loop oracle (reference test strategy: the CPU implementation is
+// the ground truth for the HIP kernels), gradients are eager compositions
+// of registered backward ops so they run on either device.
+//
+// Reference parity: src/operator/nn/ registration surface —
+// FullyConnected (fully_connected.cc:251), Convolution (convolution.cc:405),
+// Activation, Pooling (pool.cuh), BatchNorm (batch_norm.cu), softmax
+// (softmax-inl.h), Dropout (dropout-inl.h), Embedding (indexing_op.cu),
+// pick, dot; optimizer ops (optimizer_op.cc).
+#include <algorithm>
+#include <cmath>
+
+#include "ew_common.h"
+#include "ops_api.h"
+
+namespace mxcore {
+namespace {
+
+using V = const std::vector<TBlob>&;
+
+struct RegN {
+  OpEntry* e;
+  explicit RegN(const char* name) {
+    e = &OpRegistry::Get()->Register(name);
+    e->n_out = 1;
+  }
+  RegN& in(int n) { e->n_in = n; return *this; }
+  RegN& out(int n) { e->n_out = n; return *this; }
+  RegN& mut(std::vector<int> m) { e->mutate_inputs = std::move(m); return *this; }
+  RegN& infer(FInferShape f) { e->infer = std::move(f); return *this; }
+  RegN& gpu(FCompute f) { e->fcompute_gpu = std::move(f); return *this; }
+  RegN& cpu(FCompute f) { e->fcompute_cpu = std::move(f); return *this; }
+  RegN& bwd(FBackward f) { e->fbackward = std::move(f); return *this; }
+};
+
+LaunchCtx LC(const OpCtx& o) {
+  LaunchCtx lc;
+  lc.stream = o.rc.stream;
+  lc.dev = o.rc.ctx.dev_id;
+  lc.workspace = o.workspace;
+  return lc;
+}
+
+FBackward NoGradN() {
+  return [](const TapeNode& n, const std::vector<NDArray>&) {
+    return std::vector<NDArray>(n.inputs.size());
+  };
+}
+
+NDArray RunN(const char* name, const NodeAttrs& attrs,
+             const std::vector<NDArray>& ins) {
+  OpEntry* e = OpRegistry::Get()->Find(name);
+  MX_CHECK(e, "op not registered: " << name);
+  return Imperative::Run(e, attrs, ins)[0];
+}
+
+std::vector<NDArray> RunNMulti(const char* name, const NodeAttrs& attrs,
+                               const std::vector<NDArray>& ins) {
+  OpEntry* e = OpRegistry::Get()->Find(name);
+  MX_CHECK(e, "op not registered: " << name);
+  return Imperative::Run(e, attrs, ins);
+}
+
+std::string ShapeStr(const TShape& s) {
+  std::string r = "(";
+  for (auto d : s) r += std::to_string(d) + ",";
+  return r + ")";
+}
+
+// fp32 accumulate load/store helpers for the CPU loops
+template <typename T>
+float LD(const T* p, long i) { return (float)p[i]; }
+template <typename T>
+void ST(T* p, long i, float v) { p[i] = (T)v; }
+
+// host activation math (mirrors elemwise.hip act_apply/act_grad)
+float h_act(float x, const std::string& k) {
+  if (k == "relu") return x > 0.f ? x : 0.f;
+  if (k == "sigmoid") return 1.f / (1.f + expf(-x));
+  if (k == "tanh") return tanhf(x);
+  if (k == "gelu") {
+    float c = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+    return 0.5f * x * (1.f + tanhf(c));
+  }
+  if (k == "silu" || k == "swish") return x / (1.f + expf(-x));
+  MX_CHECK(false, "unknown activation " << k);
+  return x;
+}
+float h_act_grad(float dy, float s, const std::string& k) {
+  if (k == "relu") return s > 0.f ? dy : 0.f;
+  if (k == "sigmoid") return dy * s * (1.f - s);
+  if (k == "tanh") return dy * (1.f - s * s);
+  if (k == "gelu") {
+    float x = s;
+    float u = 0.7978845608028654f * (x + 0.044715f * x * x * x);
+    float t = tanhf(u);
+    float du = 0.7978845608028654f * (1.f + 3.f * 0.044715f * x * x);
+    return dy * (0.5f * (1.f + t) + 0.5f * x * (1.f - t * t) * du);
+  }
+  if (k == "silu" || k == "swish") {
+    float sig = 1.f / (1.f + expf(-s));
+    return dy * sig * (1.f + s * (1.f - sig));
+  }
+  return dy;
+}
+
+// ---------------------------------------------------------------------------
+// small GPU kernels local to this file: pick / pick-grad (CE loss)
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void pick_kernel(const T* __restrict__ x,
+                            const long* __restrict__ idx, T* __restrict__ y,
+                            long rows, long C) {
+  for (long r = (long)blockIdx.x * blockDim.x + threadIdx.x; r < rows;
+       r += (long)gridDim.x * blockDim.x) {
+    long c = idx[r];
+    y[r] = (c >= 0 && c < C) ? x[r * C + c] : (T)0.f;
+  }
+}
+
+template <typename T>
+__global__ void pick_grad_kernel(const T* __restrict__ dy,
+                                 const long* __restrict__ idx,
+                                 T* __restrict__ dx, long rows, long C) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < rows * C;
+       i += (long)gridDim.x * blockDim.x) {
+    long r = i / C, c = i % C;
+    dx[i] = (idx[r] == c) ? dy[r] : (T)0.f;
+  }
+}
+
+// one-hot (metrics / losses)
+template <typename T>
+__global__ void one_hot_kernel(const long* __restrict__ idx,
+                               T* __restrict__ y, long rows, long depth,
+                               float on, float off) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < rows * depth;
+       i += (long)gridDim.x * blockDim.x) {
+    long r = i / depth, c = i % depth;
+    y[i] = (T)(idx[r] == c ? on : off);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// CPU oracles
+// ---------------------------------------------------------------------------
+#define CPU_FLOAT_ONLY(b, name) \
+  MX_CHECK((b).dtype == kFloat32, name ": CPU path is fp32 (oracle)")
+
+void cpu_gemm_nt(const TBlob& A, const TBlob& B, const float* bias,
+                 const TBlob& out, bool accum_db = false, float* db = nullptr) {
+  // out[M,N] = A[M,K] @ B[N,K]^T
+  long M = A.shape[A.ndim() - 2], K = A.shape[A.ndim() - 1];
+  long N = B.shape[B.ndim() - 2];
+  long nb = A.ndim() == 3 ? A.shape[0] : 1;
+  const float* a = (const float*)A.dptr;
+  const float* b = (const float*)B.dptr;
+  float* y = (float*)out.dptr;
+  for (long z = 0; z < nb; ++z) {
+    const float* az = a + z * M * K;
+    const float* bz = b + (B.ndim() == 3 ? z * N * K : 0);
+    float* yz = y + z * M * N;
+    for (long i = 0; i < M; ++i)
+      for (long j = 0; j < N; ++j) {
+        double acc = 0;
+        for (long k = 0; k < K; ++k) acc += (double)az[i * K + k] * bz[j * K + k];
+        yz[i * N + j] = (float)acc + (bias ? bias[j] : 0.f);
+      }
+  }
+  if (accum_db && db) {
+    for (long j = 0; j < N; ++j) db[j] = 0.f;
+    for (long i = 0; i < M; ++i)
+      for (long j = 0; j < N; ++j) db[j] += a[i * K + j];  // unused path
+  }
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// registrations
+// ---------------------------------------------------------------------------
+namespace {
+
+bool _registered_nn = [] {
+  // ======================= dot family ====================================
+  // dot_nt: [.., M, K] x [.., N, K]^T  (FC fwd, also plain dot via python
+  // pre-transpose); attrs: relu (fused)
+  RegN("dot_nt").in(-1)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        TShape s(is[0].begin(), is[0].end() - 1);
+        s.push_back(is[1][is[1].size() - 2]);
+        os->assign(1, s);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        gemm_nt_raw(LC(o), in[0], in[1],
+                    in.size() > 2 ? Arr(in[2]) : Arr(), out[0],
+                    a.GetBool("relu", false), Arr());
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "dot_nt");
+        const float* bias = in.size() > 2 ? (const float*)in[2].dptr : nullptr;
+        cpu_gemm_nt(in[0], in[1], bias, out[0]);
+        if (a.GetBool("relu", false)) {
+          float* y = (float*)out[0].dptr;
+          for (long i = 0; i < out[0].size(); ++i) y[i] = std::max(y[i], 0.f);
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        MX_CHECK(!n.attrs.GetBool("relu", false),
+                 "dot_nt fused relu has no direct backward");
+        // dy [M,N]; dx = dy @ B  (dot_nn); dB = dy^T @ A (dot_tn)
+        std::vector<NDArray> r(n.inputs.size());
+        r[0] = RunN("dot_nn", {}, {og[0], n.inputs[1]});
+        r[1] = RunN("dot_tn", {}, {og[0], n.inputs[0]});
+        if (n.inputs.size() > 2) r[2] = RunN("colsum", {}, {og[0]});
+        return r;
+      });
+
+  // dot_nn: [M,N] x [N,K]
+  RegN("dot_nn").in(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        TShape s(is[0].begin(), is[0].end() - 1);
+        s.push_back(is[1][is[1].size() - 1]);
+        os->assign(1, s);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        gemm_nn_raw(LC(o), in[0], in[1], out[0]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "dot_nn");
+        long M = in[0].shape[0], N = in[0].shape[1], K = in[1].shape[1];
+        const float* a = (const float*)in[0].dptr;
+        const float* b = (const float*)in[1].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long i = 0; i < M; ++i)
+          for (long k = 0; k < K; ++k) {
+            double acc = 0;
+            for (long j = 0; j < N; ++j) acc += (double)a[i * N + j] * b[j * K + k];
+            y[i * K + k] = (float)acc;
+          }
+      });
+
+  // dot_tn: A[M,I]^T @ B[M,J] -> [I,J]
+  RegN("dot_tn").in(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][1], is[1][1]});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        gemm_tn_fused_raw(LC(o), in[0], in[1], out[0], Arr());
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "dot_tn");
+        long M = in[0].shape[0], I = in[0].shape[1], J = in[1].shape[1];
+        const float* a = (const float*)in[0].dptr;
+        const float* b = (const float*)in[1].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long i = 0; i < I; ++i)
+          for (long j = 0; j < J; ++j) {
+            double acc = 0;
+            for (long m = 0; m < M; ++m) acc += (double)a[m * I + i] * b[m * J + j];
+            y[i * J + j] = (float)acc;
+          }
+      });
+
+  // batched dot: [B,M,K] x [B,K,N]
+  RegN("batch_dot").in(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][0], is[0][1], is[1][2]});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        bgemm_raw(LC(o), in[0], in[1], out[0]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "batch_dot");
+        long B = in[0].shape[0], M = in[0].shape[1], K = in[0].shape[2],
+             N = in[1].shape[2];
+        const float* a = (const float*)in[0].dptr;
+        const float* b = (const float*)in[1].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long z = 0; z < B; ++z)
+          for (long i = 0; i < M; ++i)
+            for (long j = 0; j < N; ++j) {
+              double acc = 0;
+              for (long k = 0; k < K; ++k)
+                acc += (double)a[(z * M + i) * K + k] * b[(z * K + k) * N + j];
+              y[(z * M + i) * N + j] = (float)acc;
+            }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        // da = dy @ b^T ; db = a^T @ dy — via transpose + batch_dot
+        NodeAttrs tr;
+        tr.d["axes"] = "(0,2,1)";
+        NDArray bt = RunN("transpose", tr, {n.inputs[1]});
+        NDArray at = RunN("transpose", tr, {n.inputs[0]});
+        return {RunN("batch_dot", {}, {og[0], bt}),
+                RunN("batch_dot", {}, {at, og[0]})};
+      });
+
+  RegN("colsum").in(1)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][is[0].size() - 1]});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        colsum_raw(LC(o), in[0], out[0]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "colsum");
+        long N = in[0].shape[in[0].ndim() - 1], M = in[0].size() / N;
+        const float* x = (const float*)in[0].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long j = 0; j < N; ++j) y[j] = 0.f;
+        for (long i = 0; i < M; ++i)
+          for (long j = 0; j < N; ++j) y[j] += x[i * N + j];
+      });
+
+  // ======================= FullyConnected ================================
+  RegN("FullyConnected").in(-1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        bool flatten = a.GetBool("flatten", true);
+        int64_t units = is[1][0];
+        TShape s;
+        if (flatten) {
+          s = {is[0][0], units};
+        } else {
+          s = TShape(is[0].begin(), is[0].end() - 1);
+          s.push_back(units);
+        }
+        os->assign(1, s);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        Arr x = in[0];
+        long K = in[1].shape[1];
+        x.shape = {x.numel() / K, K};
+        Arr y = out[0];
+        y.shape = {x.shape[0], in[1].shape[0]};
+        gemm_nt_raw(LC(o), x, in[1], in.size() > 2 ? Arr(in[2]) : Arr(), y,
+                    false, Arr());
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "FullyConnected");
+        TBlob x = in[0];
+        long K = in[1].shape[1];
+        x.shape = {x.size() / K, K};
+        const float* bias = in.size() > 2 ? (const float*)in[2].dptr : nullptr;
+        cpu_gemm_nt(x, in[1], bias, out[0]);
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        long K = n.inputs[1].shape()[1];
+        NDArray x2 = n.inputs[0].Reshape({n.inputs[0].size() / K, K});
+        NDArray dy2 = og[0].Reshape({x2.shape()[0], n.inputs[1].shape()[0]});
+        std::vector<NDArray> r(n.inputs.size());
+        r[0] = RunN("dot_nn", {}, {dy2, n.inputs[1]})
+                   .Reshape(n.inputs[0].shape());
+        r[1] = RunN("dot_tn", {}, {dy2, x2});
+        if (n.inputs.size() > 2) r[2] = RunN("colsum", {}, {dy2});
+        return r;
+      });
+
+  // ======================= Activation ====================================
+  RegN("Activation").in(1).infer(InferSame())
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        act_fwd_raw(LC(o), in[0], a.d.at("act_type"), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "Activation");
+        const std::string& k = a.d.at("act_type");
+        const float* x = (const float*)in[0].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long i = 0; i < in[0].size(); ++i) y[i] = h_act(x[i], k);
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        const std::string& k = n.attrs.d.at("act_type");
+        // saved = y for relu/sigmoid/tanh, x for gelu/silu
+        bool from_out = k == "relu" || k == "sigmoid" || k == "tanh";
+        NDArray saved = from_out ? n.outputs[0] : n.inputs[0];
+        return {RunN("_backward_Activation", n.attrs, {og[0], saved})};
+      });
+  RegN("_backward_Activation").in(2).infer(InferSame())
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        act_bwd_raw(LC(o), in[0], in[1], a.d.at("act_type"), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        const std::string& k = a.d.at("act_type");
+        const float* dy = (const float*)in[0].dptr;
+        const float* s = (const float*)in[1].dptr;
+        float* dx = (float*)out[0].dptr;
+        for (long i = 0; i < in[0].size(); ++i)
+          dx[i] = h_act_grad(dy[i], s[i], k);
+      });
+
+  // ======================= Convolution (NHWC) ============================
+  auto conv_out_shape = [](const NodeAttrs& a, const TShape& xs,
+                           const TShape& ws) {
+    auto stride = a.GetTuple("stride", {1, 1});
+    auto pad = a.GetTuple("pad", {0, 0});
+    auto dil = a.GetTuple("dilate", {1, 1});
+    int64_t R = ws[1], S = ws[2];
+    int64_t P = (xs[1] + 2 * pad[0] - dil[0] * (R - 1) - 1) / stride[0] + 1;
+    int64_t Q = (xs[2] + 2 * pad[1] - dil[1] * (S - 1) - 1) / stride[1] + 1;
+    return TShape{xs[0], P, Q, ws[0]};
+  };
+  RegN("Convolution").in(-1)
+      .infer([conv_out_shape](const NodeAttrs& a,
+                              const std::vector<TShape>& is,
+                              const std::vector<int>& it,
+                              std::vector<TShape>* os, std::vector<int>* ot) {
+        os->assign(1, conv_out_shape(a, is[0], is[1]));
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        conv2d_fwd_raw(LC(o), in[0], in[1],
+                       in.size() > 2 ? Arr(in[2]) : Arr(), stride[0],
+                       stride[1], pad[0], pad[1], dil[0], dil[1],
+                       (int)a.GetInt("num_group", 1), out[0], Arr());
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "Convolution");
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        int G = (int)a.GetInt("num_group", 1);
+        long NB = in[0].shape[0], H = in[0].shape[1], W = in[0].shape[2],
+             C = in[0].shape[3];
+        long Kout = in[1].shape[0], R = in[1].shape[1], S = in[1].shape[2];
+        long Cg = C / G, Kg = Kout / G;
+        long P = out[0].shape[1], Q = out[0].shape[2];
+        const float* x = (const float*)in[0].dptr;
+        const float* w = (const float*)in[1].dptr;
+        const float* b = in.size() > 2 ? (const float*)in[2].dptr : nullptr;
+        float* y = (float*)out[0].dptr;
+        for (long n = 0; n < NB; ++n)
+          for (long p = 0; p < P; ++p)
+            for (long q = 0; q < Q; ++q)
+              for (long k = 0; k < Kout; ++k) {
+                long g = k / Kg;
+                double acc = b ? b[k] : 0.0;
+                for (long r = 0; r < R; ++r) {
+                  long h = p * stride[0] - pad[0] + r * dil[0];
+                  if (h < 0 || h >= H) continue;
+                  for (long s2 = 0; s2 < S; ++s2) {
+                    long ww = q * stride[1] - pad[1] + s2 * dil[1];
+                    if (ww < 0 || ww >= W) continue;
+                    for (long c = 0; c < Cg; ++c)
+                      acc += (double)x[((n * H + h) * W + ww) * C + g * Cg + c] *
+                             w[((k * R + r) * S + s2) * Cg + c];
+                  }
+                }
+                y[((n * P + p) * Q + q) * Kout + k] = (float)acc;
+              }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        std::vector<NDArray> r(n.inputs.size());
+        NodeAttrs a = n.attrs;
+        a.d["__h__"] = std::to_string(n.inputs[0].shape()[1]);
+        a.d["__w__"] = std::to_string(n.inputs[0].shape()[2]);
+        a.d["__r__"] = std::to_string(n.inputs[1].shape()[1]);
+        a.d["__s__"] = std::to_string(n.inputs[1].shape()[2]);
+        a.d["__cg__"] = std::to_string(n.inputs[1].shape()[3]);
+        r[0] = RunN("_conv_bwd_data", a, {og[0], n.inputs[1]});
+        r[1] = RunN("_conv_bwd_weight", a, {og[0], n.inputs[0]});
+        if (n.inputs.size() > 2) {
+          NDArray dy2 = og[0].Reshape(
+              {og[0].size() / og[0].shape().back(), og[0].shape().back()});
+          r[2] = RunN("colsum", {}, {dy2});
+        }
+        return r;
+      });
+
+  RegN("_conv_bwd_data").in(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t C = a.GetInt("__cg__", 1) * a.GetInt("num_group", 1);
+        os->assign(1, TShape{is[0][0], a.GetInt("__h__", 1),
+                             a.GetInt("__w__", 1), C});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        conv2d_bwd_data_raw(LC(o), in[0], in[1], stride[0], stride[1],
+                            pad[0], pad[1], dil[0], dil[1],
+                            (int)a.GetInt("num_group", 1),
+                            (int)a.GetInt("__h__", 1),
+                            (int)a.GetInt("__w__", 1), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "_conv_bwd_data");
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        int G = (int)a.GetInt("num_group", 1);
+        long NB = in[0].shape[0], P = in[0].shape[1], Q = in[0].shape[2],
+             Kout = in[0].shape[3];
+        long R = in[1].shape[1], S = in[1].shape[2], Cg = in[1].shape[3];
+        long C = Cg * G, Kg = Kout / G;
+        long H = out[0].shape[1], W = out[0].shape[2];
+        const float* dy = (const float*)in[0].dptr;
+        const float* w = (const float*)in[1].dptr;
+        float* dx = (float*)out[0].dptr;
+        for (long i = 0; i < out[0].size(); ++i) dx[i] = 0.f;
+        for (long n = 0; n < NB; ++n)
+          for (long p = 0; p < P; ++p)
+            for (long q = 0; q < Q; ++q)
+              for (long k = 0; k < Kout; ++k) {
+                long g = k / Kg;
+                float d = dy[((n * P + p) * Q + q) * Kout + k];
+                for (long r = 0; r < R; ++r) {
+                  long h = p * stride[0] - pad[0] + r * dil[0];
+                  if (h < 0 || h >= H) continue;
+                  for (long s2 = 0; s2 < S; ++s2) {
+                    long ww = q * stride[1] - pad[1] + s2 * dil[1];
+                    if (ww < 0 || ww >= W) continue;
+                    for (long c = 0; c < Cg; ++c)
+                      dx[((n * H + h) * W + ww) * C + g * Cg + c] +=
+                          d * w[((k * R + r) * S + s2) * Cg + c];
+                  }
+                }
+              }
+      });
+
+  RegN("_conv_bwd_weight").in(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][3], a.GetInt("__r__", 1),
+                             a.GetInt("__s__", 1), a.GetInt("__cg__", 1)});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        conv2d_bwd_weight_raw(LC(o), in[0], in[1], stride[0], stride[1],
+                              pad[0], pad[1], dil[0], dil[1],
+                              (int)a.GetInt("num_group", 1),
+                              (int)a.GetInt("__r__", 1),
+                              (int)a.GetInt("__s__", 1), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "_conv_bwd_weight");
+        auto stride = a.GetTuple("stride", {1, 1});
+        auto pad = a.GetTuple("pad", {0, 0});
+        auto dil = a.GetTuple("dilate", {1, 1});
+        int G = (int)a.GetInt("num_group", 1);
+        long NB = in[0].shape[0], P = in[0].shape[1], Q = in[0].shape[2],
+             Kout = in[0].shape[3];
+        long H = in[1].shape[1], W = in[1].shape[2], C = in[1].shape[3];
+        long R = out[0].shape[1], S = out[0].shape[2], Cg = out[0].shape[3];
+        long Kg = Kout / G;
+        const float* dy = (const float*)in[0].dptr;
+        const float* x = (const float*)in[1].dptr;
+        float* dw = (float*)out[0].dptr;
+        for (long i = 0; i < out[0].size(); ++i) dw[i] = 0.f;
+        for (long n = 0; n < NB; ++n)
+          for (long p = 0; p < P; ++p)
+            for (long q = 0; q < Q; ++q)
+              for (long k = 0; k < Kout; ++k) {
+                long g = k / Kg;
+                float d = dy[((n * P + p) * Q + q) * Kout + k];
+                for (long r = 0; r < R; ++r) {
+                  long h = p * stride[0] - pad[0] + r * dil[0];
+                  if (h < 0 || h >= H) continue;
+                  for (long s2 = 0; s2 < S; ++s2) {
+                    long ww = q * stride[1] - pad[1] + s2 * dil[1];
+                    if (ww < 0 || ww >= W) continue;
+                    for (long c = 0; c < Cg; ++c)
+                      dw[((k * R + r) * S + s2) * Cg + c] +=
+                          d * x[((n * H + h) * W + ww) * C + g * Cg + c];
+                  }
+                }
+              }
+      });
+
+  // ======================= Pooling (NHWC) ================================
+  RegN("Pooling").in(1).out(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        auto kernel = a.GetTuple("kernel", {2, 2});
+        auto stride = a.GetTuple("stride", kernel);
+        auto pad = a.GetTuple("pad", {0, 0});
+        bool gp = a.GetBool("global_pool", false);
+        int64_t kh = gp ? is[0][1] : kernel[0], kw = gp ? is[0][2] : kernel[1];
+        int64_t sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
+        int64_t ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
+        int64_t P = (is[0][1] + 2 * ph - kh) / sh + 1;
+        int64_t Q = (is[0][2] + 2 * pw - kw) / sw + 1;
+        os->assign(1, TShape{is[0][0], P, Q, is[0][3]});
+        // argmax plane (max pool; dummy [1] for avg)
+        bool is_max = a.d.count("pool_type") == 0 ||
+                      a.d.at("pool_type") == "max";
+        os->push_back(is_max ? (*os)[0] : TShape{1});
+        ot->assign(1, it[0]);
+        ot->push_back(kInt32);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        auto kernel = a.GetTuple("kernel", {2, 2});
+        auto stride = a.GetTuple("stride", kernel);
+        auto pad = a.GetTuple("pad", {0, 0});
+        bool gp = a.GetBool("global_pool", false);
+        std::string kind = a.d.count("pool_type") ? a.d.at("pool_type")
+                                                  : "max";
+        int64_t kh = gp ? in[0].shape[1] : kernel[0],
+                kw = gp ? in[0].shape[2] : kernel[1];
+        int64_t sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
+        int64_t ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
+        pool_fwd_raw(LC(o), in[0], kind, kh, kw, sh, sw, ph, pw,
+                     a.GetBool("count_include_pad", true), out[0],
+                     kind == "max" ? Arr(out[1]) : Arr());
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "Pooling");
+        auto kernel = a.GetTuple("kernel", {2, 2});
+        auto stride = a.GetTuple("stride", kernel);
+        auto pad = a.GetTuple("pad", {0, 0});
+        bool gp = a.GetBool("global_pool", false);
+        bool cip = a.GetBool("count_include_pad", true);
+        std::string kind = a.d.count("pool_type") ? a.d.at("pool_type")
+                                                  : "max";
+        long NB = in[0].shape[0], H = in[0].shape[1], W = in[0].shape[2],
+             C = in[0].shape[3];
+        long kh = gp ? H : kernel[0], kw = gp ? W : kernel[1];
+        long sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
+        long ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
+        long P = out[0].shape[1], Q = out[0].shape[2];
+        const float* x = (const float*)in[0].dptr;
+        float* y = (float*)out[0].dptr;
+        int* arg = kind == "max" ? (int*)out[1].dptr : nullptr;
+        for (long n = 0; n < NB; ++n)
+          for (long p = 0; p < P; ++p)
+            for (long q = 0; q < Q; ++q)
+              for (long c = 0; c < C; ++c) {
+                long h0 = p * sh - ph, w0 = q * sw - pw;
+                long h1 = std::min(h0 + kh, H), w1 = std::min(w0 + kw, W);
+                long hs = std::max(h0, 0L), ws = std::max(w0, 0L);
+                long oi = ((n * P + p) * Q + q) * C + c;
+                if (kind == "max") {
+                  float best = -3.4e38f;
+                  long bi = hs * W + ws;
+                  for (long h = hs; h < h1; ++h)
+                    for (long w2 = ws; w2 < w1; ++w2) {
+                      float v = x[((n * H + h) * W + w2) * C + c];
+                      if (v > best) { best = v; bi = h * W + w2; }
+                    }
+                  y[oi] = best;
+                  if (arg) arg[oi] = (int)bi;
+                } else {
+                  double acc = 0;
+                  for (long h = hs; h < h1; ++h)
+                    for (long w2 = ws; w2 < w1; ++w2)
+                      acc += x[((n * H + h) * W + w2) * C + c];
+                  long cnt = cip ? kh * kw : (h1 - hs) * (w1 - ws);
+                  y[oi] = (float)(acc / cnt);
+                }
+              }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a = n.attrs;
+        a.d["__h__"] = std::to_string(n.inputs[0].shape()[1]);
+        a.d["__w__"] = std::to_string(n.inputs[0].shape()[2]);
+        return {RunN("_pool_bwd", a, {og[0], n.outputs[1]})};
+      });
+
+  RegN("_pool_bwd").in(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][0], a.GetInt("__h__", 1),
+                             a.GetInt("__w__", 1), is[0][3]});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        auto kernel = a.GetTuple("kernel", {2, 2});
+        auto stride = a.GetTuple("stride", kernel);
+        auto pad = a.GetTuple("pad", {0, 0});
+        bool gp = a.GetBool("global_pool", false);
+        std::string kind = a.d.count("pool_type") ? a.d.at("pool_type")
+                                                  : "max";
+        long H = a.GetInt("__h__", 1), W = a.GetInt("__w__", 1);
+        int64_t kh = gp ? H : kernel[0], kw = gp ? W : kernel[1];
+        int64_t sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
+        int64_t ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
+        pool_bwd_raw(LC(o), in[0], kind == "max" ? Arr(in[1]) : Arr(), kind,
+                     kh, kw, sh, sw, ph, pw, H, W,
+                     a.GetBool("count_include_pad", true), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        auto kernel = a.GetTuple("kernel", {2, 2});
+        auto stride = a.GetTuple("stride", kernel);
+        auto pad = a.GetTuple("pad", {0, 0});
+        bool gp = a.GetBool("global_pool", false);
+        bool cip = a.GetBool("count_include_pad", true);
+        std::string kind = a.d.count("pool_type") ? a.d.at("pool_type")
+                                                  : "max";
+        long NB = in[0].shape[0], P = in[0].shape[1], Q = in[0].shape[2],
+             C = in[0].shape[3];
+        long H = out[0].shape[1], W = out[0].shape[2];
+        long kh = gp ? H : kernel[0], kw = gp ? W : kernel[1];
+        long sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
+        long ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
+        const float* dy = (const float*)in[0].dptr;
+        const int* arg = kind == "max" ? (const int*)in[1].dptr : nullptr;
+        float* dx = (float*)out[0].dptr;
+        for (long i = 0; i < out[0].size(); ++i) dx[i] = 0.f;
+        for (long n = 0; n < NB; ++n)
+          for (long p = 0; p < P; ++p)
+            for (long q = 0; q < Q; ++q)
+              for (long c = 0; c < C; ++c) {
+                long oi = ((n * P + p) * Q + q) * C + c;
+                if (kind == "max") {
+                  long plane = arg[oi];
+                  dx[(n * H * W + plane) * C + c] += dy[oi];
+                } else {
+                  long h0 = p * sh - ph, w0 = q * sw - pw;
+                  long h1 = std::min(h0 + kh, H), w1 = std::min(w0 + kw, W);
+                  long hs = std::max(h0, 0L), ws = std::max(w0, 0L);
+                  long cnt = cip ? kh * kw : (h1 - hs) * (w1 - ws);
+                  float d = dy[oi] / cnt;
+                  for (long h = hs; h < h1; ++h)
+                    for (long w2 = ws; w2 < w1; ++w2)
+                      dx[((n * H + h) * W + w2) * C + c] += d;
+                }
+              }
+      });
+
+
+  // ======================= BatchNorm (NHWC, fused add+relu) ==============
+  // inputs: x, gamma(f32), beta(f32), rmean(f32, mutated), rvar(f32,
+  // mutated) [, residual]; outputs: y, save_mean, save_istd, mask
+  RegN("BatchNorm").in(-1).out(4).mut({3, 4})
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t C = is[0][is[0].size() - 1];
+        os->assign(1, is[0]);
+        os->push_back({C});
+        os->push_back({C});
+        bool training = a.GetBool("training", false);
+        bool mask_ok = training && a.GetBool("fuse_relu", false) &&
+                       C % 8 == 0 &&
+                       (it[0] == kFloat16 || it[0] == kBFloat16);
+        int64_t total = 1;
+        for (auto d : is[0]) total *= d;
+        os->push_back(mask_ok ? TShape{total / 8} : TShape{1});
+        ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
+        ot->push_back(kFloat32);
+        ot->push_back(kUint8);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        bool training = a.GetBool("training", false);
+        bool relu = a.GetBool("fuse_relu", false);
+        Arr res = in.size() > 5 ? Arr(in[5]) : Arr();
+        if (training) {
+          bool mask_ok = out[3].size() > 1;
+          bn_fwd_train_raw(LC(o), in[0], in[1], in[2], in[3], in[4],
+                           a.GetFloat("momentum", 0.9),
+                           a.GetFloat("eps", 1e-5), relu, res, Arr(),
+                           out[0], out[1], out[2],
+                           mask_ok ? Arr(out[3]) : Arr());
+        } else {
+          bn_fwd_infer_raw(LC(o), in[0], in[1], in[2], in[3], in[4],
+                           a.GetFloat("eps", 1e-5), relu, res, out[0]);
+        }
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "BatchNorm");
+        bool training = a.GetBool("training", false);
+        bool relu = a.GetBool("fuse_relu", false);
+        float mom = (float)a.GetFloat("momentum", 0.9);
+        float eps = (float)a.GetFloat("eps", 1e-5);
+        long C = in[0].shape[in[0].ndim() - 1], M = in[0].size() / C;
+        const float* x = (const float*)in[0].dptr;
+        const float* g = (const float*)in[1].dptr;
+        const float* b = (const float*)in[2].dptr;
+        float* rm = (float*)in[3].dptr;
+        float* rv = (float*)in[4].dptr;
+        const float* res =
+            in.size() > 5 ? (const float*)in[5].dptr : nullptr;
+        float* y = (float*)out[0].dptr;
+        float* smean = (float*)out[1].dptr;
+        float* sistd = (float*)out[2].dptr;
+        for (long c = 0; c < C; ++c) {
+          float mean, istd;
+          if (training) {
+            double s = 0, sq = 0;
+            for (long i = 0; i < M; ++i) {
+              double v = x[i * C + c];
+              s += v;
+              sq += v * v;
+            }
+            mean = (float)(s / M);
+            float var = (float)(sq / M - mean * mean);
+            istd = 1.f / sqrtf(var + eps);
+            rm[c] = mom * rm[c] + (1.f - mom) * mean;
+            rv[c] = mom * rv[c] + (1.f - mom) * var;
+          } else {
+            mean = rm[c];
+            istd = 1.f / sqrtf(rv[c] + eps);
+          }
+          smean[c] = mean;
+          sistd[c] = istd;
+          for (long i = 0; i < M; ++i) {
+            float v = (x[i * C + c] - mean) * istd * g[c] + b[c];
+            if (res) v += res[i * C + c];
+            if (relu) v = std::max(v, 0.f);
+            y[i * C + c] = v;
+          }
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a = n.attrs;
+        bool relu = a.GetBool("fuse_relu", false);
+        bool has_res = n.inputs.size() > 5;
+        a.d["has_res"] = has_res ? "1" : "0";
+        bool mask_ok = n.outputs[3].size() > 1;
+        a.d["use_mask"] = mask_ok ? "1" : "0";
+        // inputs to the bwd op: dy, x, gamma, save_mean, save_istd,
+        // y (for un-masked fused relu), mask
+        auto outs = RunNMulti("_bn_bwd", a,
+                              {og[0], n.inputs[0], n.inputs[1],
+                               n.outputs[1], n.outputs[2], n.outputs[0],
+                               n.outputs[3]});
+        std::vector<NDArray> r(n.inputs.size());
+        r[0] = outs[0];
+        r[1] = outs[1];
+        r[2] = outs[2];
+        if (has_res) r[5] = outs[3];
+        return r;
+      });
+
+  RegN("_bn_bwd").in(7).out(4)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t C = is[0][is[0].size() - 1];
+        os->assign(1, is[0]);                          // dx
+        os->push_back({C});                            // dgamma (f32)
+        os->push_back({C});                            // dbeta (f32)
+        os->push_back(a.GetBool("has_res", false) ? is[0] : TShape{1});
+        ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
+        ot->push_back(kFloat32);
+        ot->push_back(it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        bool relu = a.GetBool("fuse_relu", false);
+        bool has_res = a.GetBool("has_res", false);
+        bool use_mask = a.GetBool("use_mask", false);
+        bn_bwd_raw(LC(o), in[0], in[1], in[2], in[3], in[4], relu,
+                   use_mask ? Arr() : Arr(in[5]), has_res,
+                   use_mask ? Arr(in[6]) : Arr(), out[0], out[1], out[2],
+                   has_res ? Arr(out[3]) : Arr());
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        bool relu = a.GetBool("fuse_relu", false);
+        bool has_res = a.GetBool("has_res", false);
+        long C = in[1].shape[in[1].ndim() - 1], M = in[1].size() / C;
+        const float* dyv = (const float*)in[0].dptr;
+        const float* x = (const float*)in[1].dptr;
+        const float* g = (const float*)in[2].dptr;
+        const float* mean = (const float*)in[3].dptr;
+        const float* istd = (const float*)in[4].dptr;
+        const float* y = (const float*)in[5].dptr;
+        float* dx = (float*)out[0].dptr;
+        float* dgamma = (float*)out[1].dptr;
+        float* dbeta = (float*)out[2].dptr;
+        float* dres = has_res ? (float*)out[3].dptr : nullptr;
+        for (long c = 0; c < C; ++c) {
+          double s1 = 0, s2 = 0;
+          for (long i = 0; i < M; ++i) {
+            float d = dyv[i * C + c];
+            if (relu && y[i * C + c] <= 0.f) d = 0.f;
+            float xh = (x[i * C + c] - mean[c]) * istd[c];
+            s1 += d;
+            s2 += d * xh;
+          }
+          dbeta[c] = (float)s1;
+          dgamma[c] = (float)s2;
+          for (long i = 0; i < M; ++i) {
+            float d = dyv[i * C + c];
+            if (relu && y[i * C + c] <= 0.f) d = 0.f;
+            if (dres) dres[i * C + c] = d;
+            float xh = (x[i * C + c] - mean[c]) * istd[c];
+            dx[i * C + c] = g[c] * istd[c] *
+                            (d - (float)s1 / M - xh * (float)s2 / M);
+          }
+        }
+      });
+
+  // ======================= LayerNorm =====================================
+  RegN("LayerNorm").in(3).out(3)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t C = is[0][is[0].size() - 1];
+        int64_t rows = 1;
+        for (size_t i = 0; i + 1 < is[0].size(); ++i) rows *= is[0][i];
+        os->assign(1, is[0]);
+        os->push_back({rows});
+        os->push_back({rows});
+        ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
+        ot->push_back(kFloat32);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        layernorm_fwd_raw(LC(o), in[0], in[1], in[2],
+                          a.GetFloat("eps", 1e-5), out[0], out[1], out[2]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "LayerNorm");
+        float eps = (float)a.GetFloat("eps", 1e-5);
+        long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+        const float* x = (const float*)in[0].dptr;
+        const float* g = (const float*)in[1].dptr;
+        const float* b = (const float*)in[2].dptr;
+        float* y = (float*)out[0].dptr;
+        float* om = (float*)out[1].dptr;
+        float* oi = (float*)out[2].dptr;
+        for (long r = 0; r < rows; ++r) {
+          double s = 0, sq = 0;
+          for (long c = 0; c < C; ++c) {
+            double v = x[r * C + c];
+            s += v;
+            sq += v * v;
+          }
+          float mean = (float)(s / C);
+          float istd = 1.f / sqrtf(std::max((float)(sq / C - mean * mean),
+                                            0.f) + eps);
+          om[r] = mean;
+          oi[r] = istd;
+          for (long c = 0; c < C; ++c)
+            y[r * C + c] = (x[r * C + c] - mean) * istd * g[c] + b[c];
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        auto outs = RunNMulti("_ln_bwd", {},
+                              {og[0], n.inputs[0], n.inputs[1],
+                               n.outputs[1], n.outputs[2]});
+        return {outs[0], outs[1], outs[2]};
+      });
+  RegN("_ln_bwd").in(5).out(3)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t C = is[0][is[0].size() - 1];
+        os->assign(1, is[0]);
+        os->push_back({C});
+        os->push_back({C});
+        ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
+        ot->push_back(kFloat32);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        layernorm_bwd_raw(LC(o), in[0], in[1], in[2], in[3], in[4], out[0],
+                          out[1], out[2]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        long C = in[1].shape[in[1].ndim() - 1], rows = in[1].size() / C;
+        const float* dyv = (const float*)in[0].dptr;
+        const float* x = (const float*)in[1].dptr;
+        const float* g = (const float*)in[2].dptr;
+        const float* mean = (const float*)in[3].dptr;
+        const float* istd = (const float*)in[4].dptr;
+        float* dx = (float*)out[0].dptr;
+        float* dg = (float*)out[1].dptr;
+        float* db = (float*)out[2].dptr;
+        for (long c = 0; c < C; ++c) { dg[c] = 0.f; db[c] = 0.f; }
+        for (long r = 0; r < rows; ++r) {
+          double a1 = 0, b1 = 0;
+          for (long c = 0; c < C; ++c) {
+            float gg = dyv[r * C + c] * g[c];
+            float xh = (x[r * C + c] - mean[r]) * istd[r];
+            a1 += gg * xh;
+            b1 += gg;
+            dg[c] += dyv[r * C + c] * xh;
+            db[c] += dyv[r * C + c];
+          }
+          a1 /= C;
+          b1 /= C;
+          for (long c = 0; c < C; ++c) {
+            float gg = dyv[r * C + c] * g[c];
+            float xh = (x[r * C + c] - mean[r]) * istd[r];
+            dx[r * C + c] = istd[r] * (gg - (float)b1 - xh * (float)a1);
+          }
+        }
+      });
+
+  // ======================= softmax / log_softmax =========================
+  auto softmax_reg = [](const char* name, bool log_mode) {
+    RegN(name).in(1).infer(InferSame())
+        .gpu([log_mode](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+          softmax_fwd_raw(LC(o), in[0], Arr(), log_mode,
+                          a.GetFloat("temperature", 1.0), out[0]);
+        })
+        .cpu([log_mode](const NodeAttrs& a, const OpCtx&, V in, V out) {
+          CPU_FLOAT_ONLY(in[0], "softmax");
+          float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+          long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+          const float* x = (const float*)in[0].dptr;
+          float* y = (float*)out[0].dptr;
+          for (long r = 0; r < rows; ++r) {
+            float m = -3.4e38f;
+            for (long c = 0; c < C; ++c) m = std::max(m, x[r * C + c]);
+            double s = 0;
+            for (long c = 0; c < C; ++c) s += exp((x[r * C + c] - m) * invT);
+            for (long c = 0; c < C; ++c) {
+              float z = (x[r * C + c] - m) * invT;
+              y[r * C + c] = log_mode ? z - (float)log(s)
+                                      : (float)(exp(z) / s);
+            }
+          }
+        })
+        .bwd([log_mode, name](const TapeNode& n,
+                              const std::vector<NDArray>& og)
+                 -> std::vector<NDArray> {
+          NodeAttrs a = n.attrs;
+          a.d["log_mode"] = log_mode ? "1" : "0";
+          return {RunN("_softmax_bwd", a, {og[0], n.outputs[0]})};
+        });
+  };
+  softmax_reg("softmax", false);
+  softmax_reg("log_softmax", true);
+  RegN("_softmax_bwd").in(2).infer(InferSame())
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        softmax_bwd_raw(LC(o), in[0], in[1], a.GetBool("log_mode", false),
+                        a.GetFloat("temperature", 1.0), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        bool log_mode = a.GetBool("log_mode", false);
+        float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+        long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+        const float* dyv = (const float*)in[0].dptr;
+        const float* y = (const float*)in[1].dptr;
+        float* dx = (float*)out[0].dptr;
+        for (long r = 0; r < rows; ++r) {
+          double s = 0;
+          for (long c = 0; c < C; ++c)
+            s += log_mode ? dyv[r * C + c]
+                          : (double)dyv[r * C + c] * y[r * C + c];
+          for (long c = 0; c < C; ++c) {
+            float d = log_mode
+                          ? dyv[r * C + c] - expf(y[r * C + c]) * (float)s
+                          : (dyv[r * C + c] - (float)s) * y[r * C + c];
+            dx[r * C + c] = d * invT;
+          }
+        }
+      });
+
+  // ======================= pick / one_hot ================================
+  RegN("pick").in(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape(is[0].begin(), is[0].end() - 1));
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        MX_CHECK(in[1].dtype == kInt64, "pick index must be int64");
+        long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+        MXC_DISPATCH_FLOAT(in[0].dtype, "pick", {
+          pick_kernel<scalar_t><<<grid_for(rows), kBlock, 0,
+                                  o.rc.stream>>>(
+              (const scalar_t*)in[0].dptr, (const long*)in[1].dptr,
+              (scalar_t*)out[0].dptr, rows, C);
+        });
+        HIP_CHECK_LAST();
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        MX_CHECK(in[1].dtype == kInt64, "pick index must be int64");
+        long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+        MXC_DISPATCH_FLOAT(in[0].dtype, "pick", {
+          auto* x = (const scalar_t*)in[0].dptr;
+          auto* y = (scalar_t*)out[0].dptr;
+          auto* idx = (const long*)in[1].dptr;
+          for (long r = 0; r < rows; ++r) {
+            long c = idx[r];
+            y[r] = (c >= 0 && c < C) ? x[r * C + c] : (scalar_t)0.f;
+          }
+        });
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a;
+        a.d["shape"] = ShapeStr(n.inputs[0].shape());
+        return {RunN("_pick_bwd", a, {og[0], n.inputs[1]}), NDArray()};
+      });
+  RegN("_pick_bwd").in(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>&,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        auto s = a.GetTuple("shape", {});
+        os->assign(1, TShape(s.begin(), s.end()));
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        long C = out[0].shape[out[0].ndim() - 1], rows = out[0].size() / C;
+        MXC_DISPATCH_FLOAT(out[0].dtype, "pick_bwd", {
+          pick_grad_kernel<scalar_t><<<grid_for(rows * C), kBlock, 0,
+                                       o.rc.stream>>>(
+              (const scalar_t*)in[0].dptr, (const long*)in[1].dptr,
+              (scalar_t*)out[0].dptr, rows, C);
+        });
+        HIP_CHECK_LAST();
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        long C = out[0].shape[out[0].ndim() - 1], rows = out[0].size() / C;
+        MXC_DISPATCH_FLOAT(out[0].dtype, "pick_bwd", {
+          auto* dy = (const scalar_t*)in[0].dptr;
+          auto* idx = (const long*)in[1].dptr;
+          auto* dx = (scalar_t*)out[0].dptr;
+          for (long i = 0; i < rows * C; ++i)
+            dx[i] = (idx[i / C] == i % C) ? dy[i / C] : (scalar_t)0.f;
+        });
+      });
+
+  RegN("one_hot").in(1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>&, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        TShape s = is[0];
+        s.push_back(a.GetInt("depth", 1));
+        os->assign(1, s);
+        ot->assign(1, (int)a.GetInt("dtype", kFloat32));
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        long depth = a.GetInt("depth", 1), rows = in[0].size();
+        float on = (float)a.GetFloat("on_value", 1.0);
+        float off = (float)a.GetFloat("off_value", 0.0);
+        MXC_DISPATCH_FLOAT(out[0].dtype, "one_hot", {
+          one_hot_kernel<scalar_t><<<grid_for(rows * depth), kBlock, 0,
+                                     o.rc.stream>>>(
+              (const long*)in[0].dptr, (scalar_t*)out[0].dptr, rows, depth,
+              on, off);
+        });
+        HIP_CHECK_LAST();
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        long depth = a.GetInt("depth", 1), rows = in[0].size();
+        float on = (float)a.GetFloat("on_value", 1.0);
+        float off = (float)a.GetFloat("off_value", 0.0);
+        MXC_DISPATCH_FLOAT(out[0].dtype, "one_hot", {
+          auto* idx = (const long*)in[0].dptr;
+          auto* y = (scalar_t*)out[0].dptr;
+          for (long i = 0; i < rows * depth; ++i)
+            y[i] = (scalar_t)(idx[i / depth] == i % depth ? on : off);
+        });
+      })
+      .bwd(NoGradN());
+
+  // ======================= reshape (recorded copy) =======================
+  RegN("Reshape").in(1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        auto s = a.GetTuple("shape", {});
+        TShape out;
+        int64_t known = 1, infer_at = -1;
+        for (size_t i = 0; i < s.size(); ++i) {
+          if (s[i] == -1) {
+            infer_at = (int64_t)i;
+            out.push_back(1);
+          } else if (s[i] == 0) {
+            out.push_back(is[0][i]);
+            known *= is[0][i];
+          } else {
+            out.push_back(s[i]);
+            known *= s[i];
+          }
+        }
+        int64_t total = 1;
+        for (auto d : is[0]) total *= d;
+        if (infer_at >= 0) out[infer_at] = total / known;
+        os->assign(1, out);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        MX_HIP_CALL(hipMemcpyAsync(
+            out[0].dptr, in[0].dptr,
+            (size_t)in[0].size() * dtype_size(in[0].dtype),
+            hipMemcpyDeviceToDevice, o.rc.stream));
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        memcpy(out[0].dptr, in[0].dptr,
+               (size_t)in[0].size() * dtype_size(in[0].dtype));
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a;
+        a.d["shape"] = ShapeStr(n.inputs[0].shape());
+        return {RunN("Reshape", a, {og[0]})};
+      });
+
+  // ======================= Dropout =======================================
+  RegN("Dropout").in(1).out(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(2, is[0]);
+        ot->assign(1, it[0]);
+        ot->push_back(kUint8);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        dropout_fwd_raw(LC(o), in[0], a.GetFloat("p", 0.5),
+                        a.GetInt("seed", 0), out[0], out[1]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "Dropout");
+        float p = (float)a.GetFloat("p", 0.5);
+        unsigned long long seed = (unsigned long long)a.GetInt("seed", 0);
+        float inv = 1.f / (1.f - p);
+        const float* x = (const float*)in[0].dptr;
+        float* y = (float*)out[0].dptr;
+        unsigned char* m = (unsigned char*)out[1].dptr;
+        for (long i = 0; i < in[0].size(); ++i) {
+          unsigned long long h = seed * 0x9E3779B97F4A7C15ULL + i;
+          h ^= h >> 33; h *= 0xff51afd7ed558ccdULL;
+          h ^= h >> 33; h *= 0xc4ceb9fe1a85ec53ULL;
+          h ^= h >> 33;
+          float u = ((unsigned)h >> 8) * (1.f / 16777216.f);
+          m[i] = u >= p;
+          y[i] = m[i] ? x[i] * inv : 0.f;
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        return {RunN("_dropout_bwd", n.attrs, {og[0], n.outputs[1]})};
+      });
+  RegN("_dropout_bwd").in(2).infer(InferSame())
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        dropout_bwd_raw(LC(o), in[0], in[1], a.GetFloat("p", 0.5), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        float inv = 1.f / (1.f - (float)a.GetFloat("p", 0.5));
+        const float* dy = (const float*)in[0].dptr;
+        const unsigned char* m = (const unsigned char*)in[1].dptr;
+        float* dx = (float*)out[0].dptr;
+        for (long i = 0; i < in[0].size(); ++i) dx[i] = m[i] ? dy[i] * inv : 0.f;
+      });
+
+  // ======================= Embedding =====================================
+  RegN("Embedding").in(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        TShape s = is[0];
+        s.push_back(is[1][1]);
+        os->assign(1, s);
+        ot->assign(1, it[1]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        embedding_fwd_raw(LC(o), in[1], in[0], out[0]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        MX_CHECK(in[0].dtype == kInt64, "Embedding index must be int64");
+        long D = in[1].shape[1], V_ = in[1].shape[0], rows = in[0].size();
+        MXC_DISPATCH_FLOAT(in[1].dtype, "Embedding", {
+          auto* w = (const scalar_t*)in[1].dptr;
+          auto* y = (scalar_t*)out[0].dptr;
+          auto* idx = (const long*)in[0].dptr;
+          for (long r = 0; r < rows; ++r) {
+            long v = idx[r];
+            for (long d = 0; d < D; ++d)
+              y[r * D + d] = (v >= 0 && v < V_) ? w[v * D + d]
+                                                : (scalar_t)0.f;
+          }
+        });
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a;
+        a.d["vocab"] = std::to_string(n.inputs[1].shape()[0]);
+        return {NDArray(), RunN("_embedding_bwd", a, {og[0], n.inputs[0]})};
+      });
+  RegN("_embedding_bwd").in(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{a.GetInt("vocab", 1),
+                             is[0][is[0].size() - 1]});
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        embedding_bwd_raw(LC(o), in[0], in[1], out[0]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        long D = out[0].shape[1], V_ = out[0].shape[0];
+        long rows = in[1].size();
+        MXC_DISPATCH_FLOAT(in[0].dtype, "_embedding_bwd", {
+          auto* dy = (const scalar_t*)in[0].dptr;
+          auto* dw = (scalar_t*)out[0].dptr;
+          auto* idx = (const long*)in[1].dptr;
+          for (long i = 0; i < V_ * D; ++i) dw[i] = (scalar_t)0.f;
+          for (long r = 0; r < rows; ++r) {
+            long v = idx[r];
+            if (v < 0 || v >= V_) continue;
+            for (long d = 0; d < D; ++d)
+              dw[v * D + d] = (scalar_t)((float)dw[v * D + d] +
+                                         (float)dy[r * D + d]);
+          }
+        });
+      });
+
+  // ======================= optimizer updates (InvokeInto) ================
+  // inputs: grad; outputs(mutable): w [, master, mom]
+  RegN("sgd_update").in(1).out(-1)
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        sgd_update_raw(LC(o), out[0], out.size() > 1 ? Arr(out[1]) : Arr(),
+                       in[0], out.size() > 2 ? Arr(out[2]) : Arr(),
+                       a.GetFloat("lr", 0.01), a.GetFloat("momentum", 0.0),
+                       a.GetFloat("wd", 0.0), a.GetFloat("rescale_grad", 1.0),
+                       a.GetFloat("clip_gradient", 0.0));
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        float lr = (float)a.GetFloat("lr", 0.01);
+        float mu = (float)a.GetFloat("momentum", 0.0);
+        float wd = (float)a.GetFloat("wd", 0.0);
+        float rs = (float)a.GetFloat("rescale_grad", 1.0);
+        float clip = (float)a.GetFloat("clip_gradient", 0.0);
+        long n = out[0].size();
+        float* master = out.size() > 1 ? (float*)out[1].dptr : nullptr;
+        float* mom = out.size() > 2 ? (float*)out[2].dptr : nullptr;
+        MXC_DISPATCH_FLOAT(out[0].dtype, "sgd_update", {
+          auto* w = (scalar_t*)out[0].dptr;
+          auto* g = (const scalar_t*)in[0].dptr;
+          for (long i = 0; i < n; ++i) {
+            float wm = master ? master[i] : (float)w[i];
+            float gv = (float)g[i] * rs;
+            if (clip > 0.f) gv = std::min(std::max(gv, -clip), clip);
+            gv += wd * wm;
+            if (mom) {
+              float m = mom[i] * mu - lr * gv;
+              mom[i] = m;
+              wm += m;
+            } else {
+              wm -= lr * gv;
+            }
+            if (master) master[i] = wm;
+            w[i] = (scalar_t)wm;
+          }
+        });
+      });
+
+  RegN("adam_update").in(1).out(-1)
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        adam_update_raw(LC(o), out[0],
+                        out.size() > 3 ? Arr(out[3]) : Arr(), in[0],
+                        out[1], out[2], a.GetFloat("lr_t", 0.001),
+                        a.GetFloat("beta1", 0.9), a.GetFloat("beta2", 0.999),
+                        a.GetFloat("eps", 1e-8), a.GetFloat("wd", 0.0),
+                        a.GetFloat("rescale_grad", 1.0),
+                        a.GetFloat("clip_gradient", 0.0),
+                        a.GetBool("adamw", false));
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        float lr_t = (float)a.GetFloat("lr_t", 0.001);
+        float b1 = (float)a.GetFloat("beta1", 0.9);
+        float b2 = (float)a.GetFloat("beta2", 0.999);
+        float eps = (float)a.GetFloat("eps", 1e-8);
+        float wd = (float)a.GetFloat("wd", 0.0);
+        float rs = (float)a.GetFloat("rescale_grad", 1.0);
+        float clip = (float)a.GetFloat("clip_gradient", 0.0);
+        bool adamw = a.GetBool("adamw", false);
+        long n = out[0].size();
+        float* m = (float*)out[1].dptr;
+        float* v = (float*)out[2].dptr;
+        float* master = out.size() > 3 ? (float*)out[3].dptr : nullptr;
+        MXC_DISPATCH_FLOAT(out[0].dtype, "adam_update", {
+          auto* w = (scalar_t*)out[0].dptr;
+          auto* g = (const scalar_t*)in[0].dptr;
+          for (long i = 0; i < n; ++i) {
+            float wm = master ? master[i] : (float)w[i];
+            float gv = (float)g[i] * rs;
+            if (clip > 0.f) gv = std::min(std::max(gv, -clip), clip);
+            if (!adamw) gv += wd * wm;
+            float mi = m[i] = b1 * m[i] + (1.f - b1) * gv;
+            float vi = v[i] = b2 * v[i] + (1.f - b2) * gv * gv;
+            wm -= lr_t * mi / (sqrtf(vi) + eps);
+            if (adamw) wm -= lr_t * wd * wm;
+            if (master) master[i] = wm;
+            w[i] = (scalar_t)wm;
+          }
+        });
+      });
+
+  // ======================= attention / lstm cell =========================
+  RegN("interleaved_attention").in(-1).out(2)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int64_t B = is[0][0], S = is[0][1], U = is[0][2] / 3;
+        int64_t H = a.GetInt("heads", 1);
+        os->assign(1, TShape{B, S, U});
+        os->push_back({B * H, S, S});
+        ot->assign(2, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        attention_fwd_raw(LC(o), in[0], in.size() > 1 ? Arr(in[1]) : Arr(),
+                          (int)a.GetInt("heads", 1),
+                          a.GetFloat("temperature", 1.0), out[0], out[1]);
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        std::vector<NDArray> r(n.inputs.size());
+        r[0] = RunN("_attention_bwd", n.attrs,
+                    {og[0], n.inputs[0], n.outputs[1]});
+        return r;
+      });
+  RegN("_attention_bwd").in(3)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, is[1]);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        attention_bwd_raw(LC(o), in[0], in[1], in[2],
+                          (int)a.GetInt("heads", 1),
+                          a.GetFloat("temperature", 1.0), out[0]);
+      });
+
+  RegN("lstm_cell").in(2).out(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(2, is[1]);
+        ot->assign(2, it[0]);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        lstm_cell_fwd_raw(LC(o), in[0], in[1], out[0], out[1]);
+      });
+
+  return true;
+}();
+
+}  // namespace
+}  // namespace mxcore
