@@ -11,6 +11,7 @@ import threading
 
 import torch
 
+from . import _core
 from .ndarray.ndarray import NDArray
 
 _STATE = threading.local()
@@ -54,19 +55,25 @@ class _RecordingStateScope:
     def __enter__(self):
         if self._enter_is_record is not None:
             self._prev_is_record = set_recording(self._enter_is_record)
+            self._prev_core_record = _core.set_recording(
+                self._enter_is_record)
             self._grad_ctx = torch.enable_grad() if self._enter_is_record \
                 else torch.no_grad()
             self._grad_ctx.__enter__()
         if self._enter_train_mode is not None:
             self._prev_train_mode = set_training(self._enter_train_mode)
+            self._prev_core_train = _core.set_training(
+                self._enter_train_mode)
         return self
 
     def __exit__(self, *args):
         if self._enter_is_record is not None:
             set_recording(self._prev_is_record)
+            _core.set_recording(self._prev_core_record)
             self._grad_ctx.__exit__(*args)
         if self._enter_train_mode is not None:
             set_training(self._prev_train_mode)
+            _core.set_training(self._prev_core_train)
 
 
 def record(train_mode=True):
@@ -89,7 +96,13 @@ def predict_mode():
 def mark_variables(variables, gradients, grad_reqs='write'):
     if isinstance(variables, NDArray):
         variables, gradients = [variables], [gradients]
-    for v, g in zip(variables, gradients):
+    if isinstance(grad_reqs, str):
+        grad_reqs = [grad_reqs] * len(variables)
+    for v, g, r in zip(variables, gradients, grad_reqs):
+        if v.is_native:
+            v._native_grad = g
+            _core.mark_variable(v._h, g._h, 2 if r == 'add' else 1)
+            continue
         v._t.requires_grad_(True)
         v._t.grad = g._t
 
@@ -103,6 +116,18 @@ def backward(heads, head_grads=None, retain_graph=False, train_mode=True):
     """
     if isinstance(heads, NDArray):
         heads = [heads]
+    if heads and heads[0].is_native:
+        # own C++ tape (Imperative::Backward — torch.autograd is not
+        # involved on the native runtime)
+        hs = [h._h for h in heads]
+        if head_grads is None:
+            gs = []
+        else:
+            if isinstance(head_grads, NDArray):
+                head_grads = [head_grads]
+            gs = [g._h for g in head_grads if g is not None]
+        _core.backward(hs, gs, retain_graph)
+        return
     tensors = [h._t for h in heads]
     if head_grads is None:
         grads = [torch.ones_like(t) for t in tensors]

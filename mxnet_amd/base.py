@@ -85,3 +85,52 @@ class MXNetError(RuntimeError):
 
 def check_sanity():
     return True
+
+
+# ---------------------------------------------------------------------------
+# native runtime (mxnet_amd._core) interop
+# ---------------------------------------------------------------------------
+# _core DTypeFlag values (src/core/base.h) — NOTE: these differ from the
+# .params serialization flags above for bfloat16 (12 vs the file format's 11)
+CORE_FLAG_TO_NP = {
+    0: _np.dtype('float32'), 1: _np.dtype('float64'), 2: _np.dtype('float16'),
+    3: _np.dtype('uint8'), 4: _np.dtype('int32'), 5: _np.dtype('int8'),
+    6: _np.dtype('int64'), 7: _np.dtype('bool'),
+}
+NP_TO_CORE_FLAG = {v: k for k, v in CORE_FLAG_TO_NP.items()}
+CORE_FLAG_BF16 = 12
+
+
+def core_flag(dtype):
+    """dtype spec -> _core DTypeFlag."""
+    if isinstance(dtype, str) and _DTYPE_ALIASES.get(dtype, dtype) == 'bfloat16':
+        return CORE_FLAG_BF16
+    if isinstance(dtype, torch.dtype):
+        if dtype is torch.bfloat16:
+            return CORE_FLAG_BF16
+        return NP_TO_CORE_FLAG[TORCH_TO_NP[dtype]]
+    return NP_TO_CORE_FLAG[np_dtype(dtype)]
+
+
+def core_flag_name(flag):
+    if flag == CORE_FLAG_BF16:
+        return 'bfloat16'
+    return CORE_FLAG_TO_NP[flag].name
+
+
+_NATIVE = [None]  # None = follow MXNET_NATIVE_RUNTIME env; else forced bool
+
+
+def native_mode():
+    """True when new NDArrays should be backed by the native C++ runtime
+    (own storage pool + engine + kernels) instead of torch tensors."""
+    if _NATIVE[0] is not None:
+        return _NATIVE[0]
+    import os
+    return os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
+
+
+def set_native(flag):
+    prev = _NATIVE[0]
+    _NATIVE[0] = bool(flag) if flag is not None else None
+    return prev

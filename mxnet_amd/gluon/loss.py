@@ -32,6 +32,12 @@ class Loss(HybridBlock):
         return loss.mean(dim=dims) if dims else loss
 
     def forward(self, *args, **kwargs):
+        if args and isinstance(args[0], NDArray) and args[0].is_native:
+            fn = getattr(self, '_forward_native', None)
+            if fn is None:
+                raise RuntimeError(
+                    f'{type(self).__name__} has no native-runtime path yet')
+            return fn(*args, **kwargs)
         nds = [a._t if isinstance(a, NDArray) else a for a in args]
         return NDArray(self._forward(*nds, **kwargs))
 
@@ -78,6 +84,19 @@ class SoftmaxCrossEntropyLoss(Loss):
             loss = -(pred * label).sum(self._axis)
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
+
+    def _forward_native(self, pred, label, sample_weight=None):
+        from ..ndarray import ops as F
+        if not self._from_logits:
+            pred = F.log_softmax(pred, self._axis)
+        if self._sparse_label:
+            loss = -F.pick(pred, label)
+        else:
+            loss = -(pred * label).sum(axis=self._axis)
+        if self._weight is not None and self._weight != 1.0:
+            loss = loss * self._weight
+        dims = tuple(d for d in range(loss.ndim) if d != self._batch_axis)
+        return loss.mean(axis=dims) if dims else loss
 
 
 SoftmaxCELoss = SoftmaxCrossEntropyLoss

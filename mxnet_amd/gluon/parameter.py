@@ -90,12 +90,28 @@ class Parameter:
         self._finish_init(init, list(ctx))
 
     def _finish_init(self, init, ctx_list):
+        from ..base import native_mode
         init = _init_mod.create(init)
         td = torch_dtype(self.dtype)
         base = torch.empty(self._shape, dtype=torch.float32)
         init(self._name, NDArray(base))
         self._data = OrderedDict()
         self._grad = OrderedDict()
+        if native_mode():
+            # native runtime: values computed host-side, stored in the own
+            # pooled allocator; grads attach to the own tape
+            from .. import ndarray as _nd
+            host = base.numpy()
+            for c in ctx_list:
+                nd = _nd.array(host, ctx=c)
+                if str(self.dtype) != 'float32':
+                    nd = nd.astype(self.dtype)
+                self._data[c] = nd
+                if self.grad_req != 'null':
+                    nd.attach_grad(self.grad_req)
+                    self._grad[c] = nd.grad
+            self._deferred_init = None
+            return
         for c in ctx_list:
             t = base.to(device=c.torch_device, dtype=td)
             nd = NDArray(t)
@@ -156,15 +172,29 @@ class Parameter:
     def zero_grad(self):
         if self._grad is None:
             return
-        with torch.no_grad():
-            for g in self._grad.values():
-                g._t.zero_()
+        from .. import _core
+        for g in self._grad.values():
+            if g.is_native:
+                _core.invoke_into('_full', [], [g._h], {'value': '0'})
+            else:
+                with torch.no_grad():
+                    g._t.zero_()
 
     def set_data(self, data):
         self._check_init()
-        with torch.no_grad():
-            for nd in self._data.values():
-                nd._t.copy_(data._t.to(nd._t.device, nd._t.dtype))
+        for nd in self._data.values():
+            if nd.is_native:
+                src_nd = data if data.is_native else None
+                if src_nd is None:
+                    from .. import ndarray as _nd
+                    src_nd = _nd.array(data.asnumpy(), ctx=nd.context)
+                if str(src_nd.dtype) != str(nd.dtype):
+                    src_nd = src_nd.astype(nd.dtype)
+                src_nd = src_nd.as_in_context(nd.context)
+                src_nd.copyto(nd)
+            else:
+                with torch.no_grad():
+                    nd._t.copy_(data._t.to(nd._t.device, nd._t.dtype))
 
     def row_sparse_data(self, row_id):
         raise NotImplementedError('row_sparse storage: deferred feature')

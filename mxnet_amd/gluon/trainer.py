@@ -340,8 +340,14 @@ class Trainer:
             # (torch accumulates, the reference overwrites — clearing here
             # restores reference behavior; 'add' keeps accumulating)
             if p.grad_req == 'write':
-                to_zero.extend(g._t for g in (p._grad or {}).values()
-                               if g._t is not None)
+                for g in (p._grad or {}).values():
+                    if g is None:
+                        continue
+                    if g.is_native:
+                        from .. import _core
+                        _core.invoke_into('_full', [], [g._h], {'value': '0'})
+                    else:
+                        to_zero.append(g._t)
         if to_zero:
             # one fused launch instead of a fill per parameter
             # (161 x ~3.7 us/step measured in the final profile)

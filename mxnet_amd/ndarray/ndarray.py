@@ -12,7 +12,9 @@ which our ``mxnet_amd.autograd`` module drives with MXNet semantics.
 import numpy as _np
 import torch
 
-from ..base import torch_dtype, np_dtype, TORCH_TO_NP
+from .. import _core
+from ..base import torch_dtype, np_dtype, TORCH_TO_NP, NP_TO_CORE_FLAG, \
+    core_flag, core_flag_name, native_mode
 from ..context import Context, current_context
 
 __all__ = ['NDArray', 'array', 'zeros', 'ones', 'full', 'empty', 'arange',
@@ -20,41 +22,88 @@ __all__ = ['NDArray', 'array', 'zeros', 'ones', 'full', 'empty', 'arange',
 
 
 class NDArray:
-    """An n-dimensional array on a device, torch.Tensor-backed."""
+    """An n-dimensional array on a device.
 
-    __slots__ = ('_t',)
+    Dual-backend: ``_tt`` is a torch.Tensor (the torch-ROCm compute path)
+    OR ``_h`` is a native ``_core.NDArray`` backed by the own C++ runtime
+    (pooled HIP storage + threaded engine + CDNA4 kernels + own autograd
+    tape).  Touching ``._t`` on a native array raises loudly — every op
+    that supports the native path branches on ``is_native`` explicitly.
+    """
+
+    __slots__ = ('_tt', '_h', '_native_grad')
 
     def __init__(self, data):
-        assert isinstance(data, torch.Tensor), type(data)
-        self._t = data
+        self._native_grad = None
+        if isinstance(data, _core.NDArray):
+            self._h = data
+            self._tt = None
+        else:
+            assert isinstance(data, torch.Tensor), type(data)
+            self._tt = data
+            self._h = None
+
+    @property
+    def is_native(self):
+        return self._h is not None
+
+    @property
+    def _t(self):
+        if self._tt is None:
+            raise RuntimeError(
+                'this NDArray is native-runtime-backed; the torch path is '
+                'not available for it (op lacks a native branch)')
+        return self._tt
+
+    @_t.setter
+    def _t(self, v):
+        self._tt = v
+        self._h = None
+
+    def _invoke(self, name, inputs, attrs=None):
+        """Run a native-registry op on this array (+ extra native inputs)."""
+        hs = [a._h if isinstance(a, NDArray) else a for a in inputs]
+        return NDArray(_core.invoke(name, hs, attrs or {})[0])
 
     # -- properties ------------------------------------------------------
     @property
     def handle(self):
-        return self._t
+        return self._h if self._h is not None else self._tt
 
     @property
     def shape(self):
-        return tuple(self._t.shape)
+        if self._h is not None:
+            return tuple(self._h.shape)
+        return tuple(self._tt.shape)
 
     @property
     def size(self):
-        return self._t.numel()
+        if self._h is not None:
+            return self._h.size
+        return self._tt.numel()
 
     @property
     def ndim(self):
-        return self._t.dim()
+        if self._h is not None:
+            return len(self._h.shape)
+        return self._tt.dim()
 
     @property
     def dtype(self):
-        d = self._t.dtype
+        if self._h is not None:
+            name = core_flag_name(self._h.dtype)
+            return name if name == 'bfloat16' else _np.dtype(name)
+        d = self._tt.dtype
         if d is torch.bfloat16:
             return 'bfloat16'
         return TORCH_TO_NP[d]
 
     @property
     def context(self):
-        return Context.from_torch(self._t.device)
+        if self._h is not None:
+            dev_type, dev_id = self._h.ctx
+            return Context('cpu' if dev_type == 1 else 'gpu', dev_id)
+        return Context.from_torch(self._tt.device)
 
     ctx = context
 
@@ -64,7 +113,9 @@ class NDArray:
 
     @property
     def grad(self):
-        g = self._t.grad
+        if self._h is not None:
+            return self._native_grad
+        g = self._tt.grad
         if g is None:
             return None
         return NDArray(g)
@@ -76,14 +127,19 @@ class NDArray:
 
     # -- sync points -----------------------------------------------------
     def wait_to_read(self):
-        # HIP-stream model: reading on host requires draining the device
-        # stream that produced this tensor (reference: WaitForVar,
-        # threaded_engine.cc:379).
-        if self._t.is_cuda:
-            torch.cuda.synchronize(self._t.device)
+        # native: engine WaitForVar (queue drained + execution event);
+        # torch path: drain the producing device stream
+        if self._h is not None:
+            self._h.wait_to_read()
+            return
+        if self._tt.is_cuda:
+            torch.cuda.synchronize(self._tt.device)
 
     def asnumpy(self):
-        t = self._t.detach()
+        if self._h is not None:
+            a = self._h.asnumpy()
+            return a
+        t = self._tt.detach()
         if t.dtype is torch.bfloat16:
             t = t.float()
         return t.cpu().numpy()
@@ -92,32 +148,42 @@ class NDArray:
         return self.asnumpy().item()
 
     def item(self):
-        return self._t.item()
+        if self._h is not None:
+            return self.asnumpy().item()
+        return self._tt.item()
 
     def __float__(self):
-        return float(self._t.item())
+        return float(self.item())
 
     def __int__(self):
-        return int(self._t.item())
+        return int(self.item())
 
     def __bool__(self):
         return bool(self._t.item()) if self.size == 1 else self.size > 0
 
     def __len__(self):
-        return self._t.shape[0]
+        return self.shape[0]
 
     # -- autograd --------------------------------------------------------
     def attach_grad(self, grad_req='write', stype=None):
         """Allocate gradient buffer and request autograd recording.
 
-        Reference: ndarray.py attach_grad -> MXAutogradMarkVariables.
+        Reference: ndarray.py attach_grad -> MXAutogradMarkVariables;
+        native path: Imperative::MarkVariable over the own tape.
         """
+        if self._h is not None:
+            g = _core.invoke('zeros_like', [self._h], {})[0]
+            self._native_grad = NDArray(g)
+            _core.mark_variable(self._h, g, 2 if grad_req == 'add' else 1)
+            return
         self._t.requires_grad_(True)
         if self._t.grad is None:
             self._t.grad = torch.zeros_like(self._t)
         self._t._mx_grad_req = grad_req
 
     def detach(self):
+        if self._h is not None:
+            return NDArray(self._h)
         return NDArray(self._t.detach())
 
     def backward(self, out_grad=None, retain_graph=False, train_mode=True):
@@ -127,6 +193,9 @@ class NDArray:
 
     # -- conversion / movement -------------------------------------------
     def astype(self, dtype, copy=True):
+        if self._h is not None:
+            return self._invoke('cast', [self],
+                                {'dtype': str(core_flag(dtype))})
         td = torch_dtype(dtype)
         out = self._t.to(td)
         if copy and out is self._t:
@@ -134,6 +203,14 @@ class NDArray:
         return NDArray(out)
 
     def as_in_context(self, ctx):
+        if self._h is not None:
+            want = (1, 0) if ctx.device_type == 'cpu' else (2, ctx.device_id)
+            if tuple(self._h.ctx) == want:
+                return self
+            dst = _core.NDArray(list(self.shape), want[0], want[1],
+                                self._h.dtype)
+            self._h.copyto(dst)
+            return NDArray(dst)
         dev = ctx.torch_device
         if self._t.device == dev:
             return self
@@ -144,6 +221,12 @@ class NDArray:
     def copyto(self, other):
         """Copy to another NDArray or a Context (reference CopyFromTo,
         ndarray.cc — on MI355X this is hipMemcpyAsync on the copy stream)."""
+        if self._h is not None:
+            if isinstance(other, Context):
+                return self.as_in_context(other)
+            assert other._h is not None, 'cannot copy native -> torch NDArray'
+            self._h.copyto(other._h)
+            return other
         if isinstance(other, Context):
             return NDArray(self._t.to(other.torch_device, non_blocking=True).clone()
                            if self._t.device == other.torch_device
@@ -154,6 +237,8 @@ class NDArray:
         return other
 
     def copy(self):
+        if self._h is not None:
+            return self._invoke('_copy', [self])
         return NDArray(self._t.clone())
 
     def to_torch(self):
@@ -171,13 +256,27 @@ class NDArray:
         new = []
         for i, s in enumerate(shape):
             if s == 0:
-                new.append(self._t.shape[i])
+                new.append(self.shape[i])
             else:
                 new.append(s)
+        if self._h is not None:
+            if _core.is_recording():
+                return self._invoke(
+                    'Reshape', [self],
+                    {'shape': '(' + ','.join(str(s) for s in new) + ')'})
+            total, known, at = self.size, 1, -1
+            for i, s in enumerate(new):
+                if s == -1:
+                    at = i
+                else:
+                    known *= s
+            if at >= 0:
+                new[at] = total // known
+            return NDArray(self._h.reshape(new))
         return NDArray(self._t.reshape(new))
 
     def flatten(self):
-        return NDArray(self._t.reshape(self._t.shape[0], -1))
+        return self.reshape(self.shape[0], -1)
 
     def expand_dims(self, axis):
         return NDArray(self._t.unsqueeze(axis))
@@ -188,6 +287,10 @@ class NDArray:
     def transpose(self, axes=None):
         if axes is None:
             axes = tuple(reversed(range(self.ndim)))
+        if self._h is not None:
+            return self._invoke(
+                'transpose', [self],
+                {'axes': '(' + ','.join(str(a) for a in axes) + ')'})
         return NDArray(self._t.permute(*axes).contiguous())
 
     def broadcast_to(self, shape):
@@ -238,20 +341,91 @@ class NDArray:
             return other._t
         return other
 
-    def __add__(self, o): return NDArray(self._t + NDArray._rhs(o, self))
-    def __radd__(self, o): return NDArray(NDArray._rhs(o, self) + self._t)
-    def __sub__(self, o): return NDArray(self._t - NDArray._rhs(o, self))
-    def __rsub__(self, o): return NDArray(NDArray._rhs(o, self) - self._t)
-    def __mul__(self, o): return NDArray(self._t * NDArray._rhs(o, self))
-    def __rmul__(self, o): return NDArray(NDArray._rhs(o, self) * self._t)
-    def __truediv__(self, o): return NDArray(self._t / NDArray._rhs(o, self))
-    def __rtruediv__(self, o): return NDArray(NDArray._rhs(o, self) / self._t)
-    def __pow__(self, o): return NDArray(self._t ** NDArray._rhs(o, self))
+    def __add__(self, o):
+        if self._h is not None:
+            return self._invoke('elemwise_add', [self, o]) \
+                if isinstance(o, NDArray) else \
+                self._invoke('_plus_scalar', [self], {'alpha': str(float(o))})
+        return NDArray(self._t + NDArray._rhs(o, self))
+
+    def __radd__(self, o):
+        if self._h is not None:
+            return self.__add__(o)
+        return NDArray(NDArray._rhs(o, self) + self._t)
+
+    def __sub__(self, o):
+        if self._h is not None:
+            return self._invoke('elemwise_sub', [self, o]) \
+                if isinstance(o, NDArray) else \
+                self._invoke('_plus_scalar', [self],
+                             {'alpha': str(-float(o))})
+        return NDArray(self._t - NDArray._rhs(o, self))
+
+    def __rsub__(self, o):
+        if self._h is not None:
+            if isinstance(o, NDArray):
+                return o.__sub__(self)
+            return self._invoke('_rminus_scalar', [self],
+                                {'alpha': str(float(o))})
+        return NDArray(NDArray._rhs(o, self) - self._t)
+
+    def __mul__(self, o):
+        if self._h is not None:
+            return self._invoke('elemwise_mul', [self, o]) \
+                if isinstance(o, NDArray) else \
+                self._invoke('_mul_scalar', [self], {'alpha': str(float(o))})
+        return NDArray(self._t * NDArray._rhs(o, self))
+
+    def __rmul__(self, o):
+        if self._h is not None:
+            return self.__mul__(o)
+        return NDArray(NDArray._rhs(o, self) * self._t)
+
+    def __truediv__(self, o):
+        if self._h is not None:
+            return self._invoke('elemwise_div', [self, o]) \
+                if isinstance(o, NDArray) else \
+                self._invoke('_mul_scalar', [self],
+                             {'alpha': str(1.0 / float(o))})
+        return NDArray(self._t / NDArray._rhs(o, self))
+
+    def __rtruediv__(self, o):
+        if self._h is not None:
+            if isinstance(o, NDArray):
+                return o.__truediv__(self)
+            return self._invoke('_rdiv_scalar', [self],
+                                {'alpha': str(float(o))})
+        return NDArray(NDArray._rhs(o, self) / self._t)
+
+    def __pow__(self, o):
+        if self._h is not None:
+            return self._invoke('power', [self, o]) \
+                if isinstance(o, NDArray) else \
+                self._invoke('_power_scalar', [self],
+                             {'alpha': str(float(o))})
+        return NDArray(self._t ** NDArray._rhs(o, self))
+
     def __mod__(self, o): return NDArray(self._t % NDArray._rhs(o, self))
-    def __neg__(self): return NDArray(-self._t)
-    def __abs__(self): return NDArray(self._t.abs())
+
+    def __neg__(self):
+        if self._h is not None:
+            return self._invoke('negative', [self])
+        return NDArray(-self._t)
+
+    def __abs__(self):
+        if self._h is not None:
+            return self._invoke('abs', [self])
+        return NDArray(self._t.abs())
 
     def __iadd__(self, o):
+        if self._h is not None:
+            if isinstance(o, NDArray):
+                _core.invoke_into('_grad_add', [o._h], [self._h], {})
+            else:
+                t = self._invoke('_plus_scalar', [self],
+                                 {'alpha': str(float(o))})
+                _core.invoke_into('_copy_into', [t._h], [self._h], {})
+            return self
         with torch.no_grad():
             self._t += NDArray._rhs(o, self)
         return self
@@ -262,6 +436,10 @@ class NDArray:
         return self
 
     def __imul__(self, o):
+        if self._h is not None:
+            t = self.__mul__(o)
+            _core.invoke_into('_copy_into', [t._h], [self._h], {})
+            return self
         with torch.no_grad():
             self._t *= NDArray._rhs(o, self)
         return self
@@ -282,22 +460,43 @@ class NDArray:
         return id(self)
 
     # -- reductions ----------------------------------------------------------
+    @staticmethod
+    def _axis_attr(axis, keepdims):
+        at = {'keepdims': '1' if keepdims else '0'}
+        if axis is not None:
+            if isinstance(axis, int):
+                axis = (axis,)
+            at['axis'] = '(' + ','.join(str(a) for a in axis) + ')'
+        return at
+
     def sum(self, axis=None, keepdims=False):
+        if self._h is not None:
+            return self._invoke('sum', [self],
+                                NDArray._axis_attr(axis, keepdims))
         if axis is None:
             return NDArray(self._t.sum())
         return NDArray(self._t.sum(dim=axis, keepdim=keepdims))
 
     def mean(self, axis=None, keepdims=False):
+        if self._h is not None:
+            return self._invoke('mean', [self],
+                                NDArray._axis_attr(axis, keepdims))
         if axis is None:
             return NDArray(self._t.float().mean().to(self._t.dtype) if not self._t.is_floating_point() else self._t.mean())
         return NDArray(self._t.mean(dim=axis, keepdim=keepdims))
 
     def max(self, axis=None, keepdims=False):
+        if self._h is not None:
+            return self._invoke('max', [self],
+                                NDArray._axis_attr(axis, keepdims))
         if axis is None:
             return NDArray(self._t.max())
         return NDArray(self._t.max(dim=axis, keepdim=keepdims).values)
 
     def min(self, axis=None, keepdims=False):
+        if self._h is not None:
+            return self._invoke('min', [self],
+                                NDArray._axis_attr(axis, keepdims))
         if axis is None:
             return NDArray(self._t.min())
         return NDArray(self._t.min(dim=axis, keepdim=keepdims).values)
@@ -330,7 +529,36 @@ def _device(ctx):
     return (ctx or current_context()).torch_device
 
 
+def _core_ctx(ctx):
+    c = ctx or current_context()
+    return (1, 0) if c.device_type == 'cpu' else (2, c.device_id)
+
+
+def _native_full(shape, val, ctx, dtype):
+    if isinstance(shape, int):
+        shape = (shape,)
+    dt, di = _core_ctx(ctx)
+    at = {'shape': '(' + ','.join(str(s) for s in shape) + ')',
+          'value': str(float(val)), 'dtype': str(core_flag(dtype))}
+    if dt == 2:
+        at['__ctx_gpu__'] = str(di)
+    else:
+        at['__ctx_gpu__'] = '-1'
+    return NDArray(_core.invoke('_full', [], at)[0])
+
+
 def array(source_array, ctx=None, dtype=None):
+    if native_mode():
+        if isinstance(source_array, NDArray):
+            source_array = source_array.asnumpy()
+        from_list = not isinstance(source_array, _np.ndarray)
+        a = _np.asarray(source_array)
+        if dtype is not None:
+            a = a.astype(np_dtype(dtype))
+        elif from_list or a.dtype == _np.float64:
+            a = a.astype(_np.float32)  # mxnet default dtype
+        dt, di = _core_ctx(ctx)
+        return NDArray(_core.from_numpy(_np.ascontiguousarray(a), dt, di))
     if isinstance(source_array, NDArray):
         t = source_array._t
         t = t.to(_device(ctx))
@@ -355,22 +583,33 @@ def from_torch(t):
 
 
 def empty(shape, ctx=None, dtype=None):
+    if native_mode():
+        if isinstance(shape, int):
+            shape = (shape,)
+        dt, di = _core_ctx(ctx)
+        return NDArray(_core.NDArray(list(shape), dt, di, core_flag(dtype)))
     return NDArray(torch.empty(shape, dtype=torch_dtype(dtype), device=_device(ctx)))
 
 
 def zeros(shape, ctx=None, dtype=None, **kwargs):
+    if native_mode():
+        return _native_full(shape, 0, ctx, dtype)
     if isinstance(shape, int):
         shape = (shape,)
     return NDArray(torch.zeros(shape, dtype=torch_dtype(dtype), device=_device(ctx)))
 
 
 def ones(shape, ctx=None, dtype=None, **kwargs):
+    if native_mode():
+        return _native_full(shape, 1, ctx, dtype)
     if isinstance(shape, int):
         shape = (shape,)
     return NDArray(torch.ones(shape, dtype=torch_dtype(dtype), device=_device(ctx)))
 
 
 def full(shape, val, ctx=None, dtype=None):
+    if native_mode():
+        return _native_full(shape, val, ctx, dtype)
     if isinstance(shape, int):
         shape = (shape,)
     return NDArray(torch.full(shape, val, dtype=torch_dtype(dtype), device=_device(ctx)))
@@ -384,10 +623,14 @@ def arange(start, stop=None, step=1.0, ctx=None, dtype=None):
 
 
 def zeros_like(a):
+    if a.is_native:
+        return a._invoke('zeros_like', [a])
     return NDArray(torch.zeros_like(a._t))
 
 
 def ones_like(a):
+    if a.is_native:
+        return a._invoke('ones_like', [a])
     return NDArray(torch.ones_like(a._t))
 
 
@@ -404,7 +647,10 @@ def stack(*arys, axis=0):
 
 
 def waitall():
-    """Block until all async GPU work completes (reference Engine::WaitForAll)."""
+    """Block until all async work completes (reference Engine::WaitForAll:
+    native engine queues drained + device streams synced; plus the torch
+    streams when the torch path is in use)."""
+    _core.wait_all()
     if torch.cuda.is_available():
         for i in range(torch.cuda.device_count()):
             torch.cuda.synchronize(i)

@@ -29,10 +29,34 @@ def _pair(v):
     return (int(v), int(v))
 
 
+# --- native-runtime dispatch helpers ---------------------------------------
+
+def _tup_attr(v):
+    if v is None:
+        return None
+    if isinstance(v, int):
+        v = (v, v)
+    return '(' + ','.join(str(int(x)) for x in v) + ')'
+
+
+def _ninv(name, inputs, attrs=None, nout=1):
+    from .. import _core
+    hs = [a._h for a in inputs if a is not None]
+    outs = _core.invoke(name, hs, {k: v for k, v in (attrs or {}).items()
+                                   if v is not None})
+    if nout == 1:
+        return NDArray(outs[0])
+    return [NDArray(o) for o in outs]
+
+
 # --- NN ops (legacy capitalized names, reference src/operator/nn) -----------
 
 def FullyConnected(data, weight, bias=None, num_hidden=None, no_bias=False,
                    flatten=True, **kwargs):
+    if data.is_native:
+        ins = [data, weight] + ([] if (no_bias or bias is None) else [bias])
+        return _ninv('FullyConnected', ins,
+                     {'flatten': '1' if flatten else '0'})
     return NDArray(_nn.fully_connected(_t(data), _t(weight),
                                        None if no_bias else _t(bias), flatten))
 
@@ -40,6 +64,16 @@ def FullyConnected(data, weight, bias=None, num_hidden=None, no_bias=False,
 def Convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
                 dilate=(1, 1), pad=(0, 0), num_filter=None, num_group=1,
                 no_bias=False, layout='NCHW', **kwargs):
+    if data.is_native:
+        assert layout == 'NHWC', \
+            'native runtime convolution is NHWC (MI355X-first layout)'
+        ins = [data, weight] + ([] if (no_bias or bias is None) else [bias])
+        return _ninv('Convolution', ins,
+                     {'kernel': _tup_attr(kernel),
+                      'stride': _tup_attr(stride), 'pad': _tup_attr(pad),
+                      'dilate': _tup_attr(dilate),
+                      'num_filter': str(num_filter),
+                      'num_group': str(num_group)})
     return NDArray(_nn.conv2d(_t(data), _t(weight),
                               None if no_bias else _t(bias),
                               stride=_pair(stride), pad=_pair(pad),
@@ -48,6 +82,8 @@ def Convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
 
 
 def Activation(data, act_type='relu', **kwargs):
+    if data.is_native:
+        return _ninv('Activation', [data], {'act_type': act_type})
     return NDArray(_nn.activation(_t(data), act_type))
 
 
@@ -66,6 +102,17 @@ def LeakyReLU(data, act_type='leaky', slope=0.25, **kwargs):
 
 def Pooling(data, kernel=(2, 2), pool_type='max', stride=None, pad=(0, 0),
             global_pool=False, layout='NCHW', count_include_pad=True, **kwargs):
+    if data.is_native:
+        assert layout == 'NHWC', \
+            'native runtime pooling is NHWC (MI355X-first layout)'
+        outs = _ninv('Pooling', [data],
+                     {'kernel': _tup_attr(kernel),
+                      'stride': _tup_attr(stride or kernel),
+                      'pad': _tup_attr(pad), 'pool_type': pool_type,
+                      'global_pool': '1' if global_pool else '0',
+                      'count_include_pad':
+                          '1' if count_include_pad else '0'}, nout=2)
+        return outs[0]
     return NDArray(_nn.pooling(_t(data), pool_type, _pair(kernel),
                                _pair(stride), _pair(pad), layout,
                                global_pool, count_include_pad))
@@ -76,6 +123,17 @@ def BatchNorm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
               axis=1, layout=None, fuse_relu=False, residual=None, **kwargs):
     from .. import autograd as _ag
     training = _ag.is_training() and not use_global_stats
+    if data.is_native:
+        assert axis in (-1, data.ndim - 1), \
+            'native runtime BatchNorm is NHWC (channels-last axis)'
+        ins = [data, gamma, beta, moving_mean, moving_var]
+        if residual is not None:
+            ins.append(residual)
+        outs = _ninv('BatchNorm', ins,
+                     {'momentum': str(momentum), 'eps': str(eps),
+                      'training': '1' if training else '0',
+                      'fuse_relu': '1' if fuse_relu else '0'}, nout=4)
+        return outs[0]
     if layout is None:
         layout = 'NHWC' if axis in (-1, _t(data).dim() - 1) else 'NCHW'
     return NDArray(_nn.batch_norm(_t(data), _t(gamma), _t(beta),
@@ -87,24 +145,47 @@ def BatchNorm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
 
 
 def LayerNorm(data, gamma, beta, axis=-1, eps=1e-5, **kwargs):
+    if data.is_native:
+        assert axis in (-1, data.ndim - 1)
+        return _ninv('LayerNorm', [data, gamma, beta],
+                     {'eps': str(eps)}, nout=3)[0]
     return NDArray(_nn.layer_norm(_t(data), _t(gamma), _t(beta), axis, eps))
 
 
 def Embedding(data, weight, input_dim=None, output_dim=None, dtype=None,
               sparse_grad=False, **kwargs):
+    if data.is_native:
+        idx = data if data.dtype == _np.dtype('int64') \
+            else data.astype('int64')
+        return _ninv('Embedding', [idx, weight], {})
     return NDArray(_nn.embedding(_t(data), _t(weight), sparse_grad))
 
 
 def Dropout(data, p=0.5, mode='training', **kwargs):
     from .. import autograd as _ag
+    if data.is_native:
+        if not _ag.is_training() or p == 0:
+            return data
+        import random as _random
+        return _ninv('Dropout', [data],
+                     {'p': str(p),
+                      'seed': str(_random.getrandbits(48))}, nout=2)[0]
     return NDArray(_nn.dropout(_t(data), p, _ag.is_training()))
 
 
 def softmax(data, axis=-1, temperature=None, **kwargs):
+    if data.is_native:
+        assert axis in (-1, data.ndim - 1)
+        return _ninv('softmax', [data],
+                     {'temperature': str(temperature or 1.0)})
     return NDArray(_nn.softmax(_t(data), axis, temperature or 1.0))
 
 
 def log_softmax(data, axis=-1, temperature=None, **kwargs):
+    if data.is_native:
+        assert axis in (-1, data.ndim - 1)
+        return _ninv('log_softmax', [data],
+                     {'temperature': str(temperature or 1.0)})
     return NDArray(_nn.log_softmax(_t(data), axis, temperature or 1.0))
 
 
@@ -117,6 +198,8 @@ def SoftmaxOutput(data, label, **kwargs):
 
 
 def Flatten(data, **kwargs):
+    if data.is_native:
+        return data.reshape(data.shape[0], -1)
     return NDArray(_t(data).reshape(_t(data).shape[0], -1))
 
 
@@ -360,6 +443,11 @@ def take(a, indices, axis=0, **kwargs):
 
 
 def pick(data, index, axis=-1, keepdims=False, **kwargs):
+    if data.is_native:
+        assert axis in (-1, data.ndim - 1) and not keepdims
+        idx = index if index.dtype == _np.dtype('int64') \
+            else index.astype('int64')
+        return _ninv('pick', [data, idx], {})
     x, idx = _t(data), _t(index).long()
     out = torch.gather(x, axis, idx.unsqueeze(axis))
     if not keepdims:
@@ -373,6 +461,14 @@ def gather_nd(data, indices, **kwargs):
 
 
 def one_hot(indices, depth, on_value=1.0, off_value=0.0, dtype='float32', **kwargs):
+    if indices.is_native:
+        from ..base import core_flag
+        idx = indices if indices.dtype == _np.dtype('int64') \
+            else indices.astype('int64')
+        return _ninv('one_hot', [idx],
+                     {'depth': str(depth), 'on_value': str(on_value),
+                      'off_value': str(off_value),
+                      'dtype': str(core_flag(dtype))})
     oh = torch.nn.functional.one_hot(_t(indices).long(), depth)
     oh = oh.to(torch_dtype(dtype)) * (on_value - off_value) + off_value
     return NDArray(oh)

@@ -75,6 +75,12 @@ class Optimizer:
 
     def create_state_multi_precision(self, index, weight):
         """fp32 master weight for low-precision params (mp_* reference ops)."""
+        if isinstance(weight, NDArray) and weight.is_native:
+            if self.multi_precision and str(weight.dtype) in (
+                    'float16', 'bfloat16'):
+                master = weight.astype('float32')
+                return (master, self.create_state(index, master))
+            return self.create_state(index, weight)
         w = weight._t if isinstance(weight, NDArray) else weight
         if self.multi_precision and w.dtype in (torch.float16, torch.bfloat16):
             master = w.detach().float().clone()
@@ -115,6 +121,11 @@ class SGD(Optimizer):
     def create_state(self, index, weight):
         if self.momentum == 0:
             return None
+        if isinstance(weight, NDArray) and weight.is_native:
+            f32 = weight if str(weight.dtype) == 'float32' \
+                else weight.astype('float32')
+            from ..ndarray.ndarray import zeros_like as _zl
+            return _zl(f32)
         w = weight._t if isinstance(weight, NDArray) else weight
         return torch.zeros_like(w, dtype=torch.float32)
 
@@ -175,6 +186,27 @@ class SGD(Optimizer):
     def _ump_dense(self, index, weight, grad, state):
         """Single fused HIP kernel on GPU (reference mp_sgd_mom_update):
         rescale+clip+wd+momentum+master-weight update+fp16 cast in one pass."""
+        if isinstance(weight, NDArray) and weight.is_native:
+            from .. import _core
+            self._update_count(index)
+            lr, wd = self._get_lr(index), self._get_wd(index)
+            master = mom = None
+            if isinstance(state, tuple):
+                master, mom = state
+            else:
+                mom = state
+            outs = [weight._h]
+            if master is not None:
+                outs.append(master._h)
+            if mom is not None:
+                outs.append(mom._h)
+            _core.invoke_into(
+                'sgd_update', [grad._h], outs,
+                {'lr': str(lr), 'momentum': str(self.momentum),
+                 'wd': str(wd), 'rescale_grad': str(self.rescale_grad),
+                 'clip_gradient': str(self.clip_gradient or 0.0),
+                 'has_master': '1' if master is not None else '0'})
+            return
         w = weight._t if isinstance(weight, NDArray) else weight
         g = grad._t if isinstance(grad, NDArray) else grad
         if type(self) is SGD and w.is_cuda and g.dtype == w.dtype:
@@ -390,6 +422,11 @@ class Signum(Optimizer):
     def create_state(self, index, weight):
         if self.momentum == 0:
             return None
+        if isinstance(weight, NDArray) and weight.is_native:
+            f32 = weight if str(weight.dtype) == 'float32' \
+                else weight.astype('float32')
+            from ..ndarray.ndarray import zeros_like as _zl
+            return _zl(f32)
         w = weight._t if isinstance(weight, NDArray) else weight
         return torch.zeros_like(w, dtype=torch.float32)
 

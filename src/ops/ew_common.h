@@ -134,6 +134,9 @@ inline FInferShape InferBroadcast() {
   return [](const NodeAttrs&, const std::vector<TShape>& is,
             const std::vector<int>& it, std::vector<TShape>* os,
             std::vector<int>* ot) {
+    MX_CHECK(it[0] == it[1], "binary op dtype mismatch: "
+                                 << dtype_name(it[0]) << " vs "
+                                 << dtype_name(it[1]));
     os->assign(1, broadcast_shape(is[0], is[1]));
     ot->assign(1, it[0]);
   };

@@ -1338,8 +1338,11 @@ bool _registered_nn = [] {
   // inputs: grad; outputs(mutable): w [, master, mom]
   RegN("sgd_update").in(1).out(-1)
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
-        sgd_update_raw(LC(o), out[0], out.size() > 1 ? Arr(out[1]) : Arr(),
-                       in[0], out.size() > 2 ? Arr(out[2]) : Arr(),
+        bool hm = a.GetBool("has_master", false);
+        size_t i = 1;
+        Arr master = hm ? Arr(out[i++]) : Arr();
+        Arr mom = out.size() > i ? Arr(out[i]) : Arr();
+        sgd_update_raw(LC(o), out[0], master, in[0], mom,
                        a.GetFloat("lr", 0.01), a.GetFloat("momentum", 0.0),
                        a.GetFloat("wd", 0.0), a.GetFloat("rescale_grad", 1.0),
                        a.GetFloat("clip_gradient", 0.0));
@@ -1351,8 +1354,10 @@ bool _registered_nn = [] {
         float rs = (float)a.GetFloat("rescale_grad", 1.0);
         float clip = (float)a.GetFloat("clip_gradient", 0.0);
         long n = out[0].size();
-        float* master = out.size() > 1 ? (float*)out[1].dptr : nullptr;
-        float* mom = out.size() > 2 ? (float*)out[2].dptr : nullptr;
+        bool hm = a.GetBool("has_master", false);
+        size_t oi = 1;
+        float* master = hm ? (float*)out[oi++].dptr : nullptr;
+        float* mom = out.size() > oi ? (float*)out[oi].dptr : nullptr;
         MXC_DISPATCH_FLOAT(out[0].dtype, "sgd_update", {
           auto* w = (scalar_t*)out[0].dptr;
           auto* g = (const scalar_t*)in[0].dptr;

@@ -1,0 +1,143 @@
+"""Native-runtime frontend: Gluon models training on the own C++ runtime
+(pooled HIP storage + threaded engine + registry ops + own autograd tape)
+with torch.autograd not involved (VERDICT item 4 done-criterion:
+LeNet config-1 end-to-end; numerics cross-checked against the
+torch-backed path on identical init/data)."""
+import random
+
+import numpy as np
+import pytest
+import torch
+
+import mxnet_amd as mx
+from mxnet_amd import autograd
+from mxnet_amd.base import set_native
+from mxnet_amd.gluon import nn, Trainer
+from mxnet_amd.gluon.loss import SoftmaxCrossEntropyLoss
+
+
+@pytest.fixture
+def native():
+    prev = set_native(True)
+    yield
+    set_native(prev)
+
+
+def _lenet():
+    net = nn.HybridSequential()
+    net.add(nn.Conv2D(6, kernel_size=5, padding=2, activation='relu',
+                      layout='NHWC'),
+            nn.MaxPool2D(pool_size=2, strides=2, layout='NHWC'),
+            nn.Conv2D(16, kernel_size=5, activation='relu', layout='NHWC'),
+            nn.MaxPool2D(pool_size=2, strides=2, layout='NHWC'),
+            nn.Dense(120, activation='relu'),
+            nn.Dense(84, activation='relu'),
+            nn.Dense(10))
+    return net
+
+
+def test_native_ndarray_basics(native):
+    a = mx.nd.ones((3, 4))
+    assert a.is_native
+    b = mx.nd.array([[1, 2, 3, 4]] * 3)
+    c = (a + b * 2).sum(axis=1)
+    np.testing.assert_allclose(c.asnumpy(), [24, 24, 24])
+    d = a.astype('float16')
+    assert str(d.dtype) == 'float16'
+    np.testing.assert_allclose(d.astype('float32').asnumpy(), a.asnumpy())
+
+
+def test_native_lenet_trains(native):
+    """Gluon LeNet, ctx=cpu, end-to-end on the native runtime
+    (BASELINE config 1; torch.autograd never sees these arrays)."""
+    random.seed(0); np.random.seed(0); torch.manual_seed(0)
+    net = _lenet()
+    net.initialize()
+    tr = Trainer(net.collect_params(), 'sgd',
+                 {'learning_rate': 0.5, 'momentum': 0.9}, kvstore=None)
+    lf = SoftmaxCrossEntropyLoss()
+    rs = np.random.RandomState(0)
+    x = mx.nd.array(rs.randn(16, 28, 28, 1))
+    y = mx.nd.array(rs.randint(0, 10, (16,)).astype('int64'))
+    assert x.is_native
+    losses = []
+    for _ in range(20):
+        with autograd.record():
+            L = lf(net(x), y).mean()
+        L.backward()
+        tr.step(1)
+        losses.append(float(L.asnumpy()))
+    mx.nd.waitall()
+    assert losses[-1] < losses[0] * 0.8, losses
+
+
+def test_native_matches_torch_backend():
+    """Same init/data on both runtimes -> identical loss trajectories."""
+    def run(native_flag):
+        prev = set_native(native_flag)
+        try:
+            random.seed(0); np.random.seed(0); torch.manual_seed(0)
+            net = nn.HybridSequential()
+            net.add(nn.Conv2D(6, kernel_size=5, padding=2,
+                              activation='relu', layout='NHWC'),
+                    nn.MaxPool2D(pool_size=2, strides=2, layout='NHWC'),
+                    nn.Dense(32, activation='relu'),
+                    nn.Dense(10))
+            net.initialize()
+            tr = Trainer(net.collect_params(), 'sgd',
+                         {'learning_rate': 0.5, 'momentum': 0.9},
+                         kvstore=None)
+            lf = SoftmaxCrossEntropyLoss()
+            rs = np.random.RandomState(0)
+            x = mx.nd.array(rs.randn(16, 12, 12, 1))
+            y = mx.nd.array(rs.randint(0, 10, (16,)).astype('int64'))
+            out = []
+            for _ in range(10):
+                with autograd.record():
+                    L = lf(net(x), y).mean()
+                L.backward()
+                tr.step(1)
+                out.append(float(L.asnumpy()))
+            return out
+        finally:
+            set_native(prev)
+
+    torch_traj = run(False)
+    native_traj = run(True)
+    np.testing.assert_allclose(native_traj, torch_traj, atol=2e-3)
+    assert native_traj[-1] < native_traj[0] * 0.7
+
+
+def test_native_bn_dropout_embedding(native):
+    random.seed(0); np.random.seed(0); torch.manual_seed(0)
+    net = nn.HybridSequential()
+    net.add(nn.Conv2D(8, kernel_size=3, padding=1, layout='NHWC'),
+            nn.BatchNorm(axis=-1),
+            nn.Activation('relu'),
+            nn.Dense(16, activation='tanh'),
+            nn.Dropout(0.5),
+            nn.Dense(4))
+    net.initialize()
+    tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.1},
+                 kvstore=None)
+    lf = SoftmaxCrossEntropyLoss()
+    rs = np.random.RandomState(1)
+    x = mx.nd.array(rs.randn(8, 6, 6, 3))
+    y = mx.nd.array(rs.randint(0, 4, (8,)).astype('int64'))
+    for _ in range(3):
+        with autograd.record():
+            L = lf(net(x), y).mean()
+        L.backward()
+        tr.step(1)
+    v = float(L.asnumpy())
+    assert v == v  # finite
+    # eval mode uses running stats
+    out = net(x)
+    assert out.shape == (8, 4)
+
+
+def test_native_gpu_guard(native):
+    """Native ops must fail loudly rather than fall back silently."""
+    a = mx.nd.ones((2, 2))
+    with pytest.raises(Exception):
+        mx.nd.ops._ninv('definitely_not_an_op', [a], {})
