@@ -31,6 +31,9 @@ def parse_args():
     p.add_argument('--image-size', type=int, default=224)
     p.add_argument('--model', default='resnet50_v1')
     p.add_argument('--dtype', default='float16')
+    p.add_argument('--native', action='store_true',
+                   help='run on the native C++ runtime (own engine/'
+                        'storage/tape) instead of the torch frontend')
     return p.parse_args()
 
 
@@ -45,7 +48,20 @@ def main():
     if on_gpu:
         torch.cuda.set_device(local_rank)
 
+    # --native (or MXNET_NATIVE_RUNTIME=1): the whole step runs on the
+    # own C++ runtime — pooled HIP storage, threaded engine streams, own
+    # autograd tape, registry-dispatched CDNA4 kernels; torch only
+    # supplies the host process (no torch tensors, no torch.autograd)
+    native = args.native or os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
+    if native and distributed:
+        print('# native runtime multi-rank path pending RCCL kvstore; '
+              'falling back to torch-frontend', file=sys.stderr)
+        native = False
+
     import mxnet_amd as mx
+    if native:
+        from mxnet_amd.base import set_native
+        set_native(True)
     from mxnet_amd import autograd
     from mxnet_amd.gluon import Trainer
     from mxnet_amd.gluon.loss import SoftmaxCrossEntropyLoss
@@ -76,8 +92,17 @@ def main():
     dev = torch.device('cuda', local_rank) if on_gpu else torch.device('cpu')
     tdt = {'float16': torch.float16, 'bfloat16': torch.bfloat16,
            'float32': torch.float32}[dtype]
-    x = mx.nd.from_torch(torch.randn(B, S, S, 3, device=dev, dtype=tdt))
-    label = mx.nd.from_torch(torch.randint(0, 1000, (B,), device=dev))
+    if native:
+        import numpy as _np
+        rs = _np.random.RandomState(1234 + rank)
+        gctx = mx.gpu(local_rank) if on_gpu else mx.cpu()
+        x = mx.nd.array(rs.randn(B, S, S, 3).astype('float32'),
+                        ctx=gctx).astype(dtype)
+        label = mx.nd.array(rs.randint(0, 1000, (B,)).astype('int64'),
+                            ctx=gctx)
+    else:
+        x = mx.nd.from_torch(torch.randn(B, S, S, 3, device=dev, dtype=tdt))
+        label = mx.nd.from_torch(torch.randint(0, 1000, (B,), device=dev))
 
     def step():
         with autograd.record():
@@ -101,7 +126,22 @@ def main():
     graph = None
     env_g = os.environ.get('MXNET_BENCH_HIPGRAPH', '')
     want_graph = env_g == '1' or (env_g != '0' and not distributed)
-    if on_gpu and want_graph:
+    if on_gpu and want_graph and native:
+        # native runtime: capture the engine's compute stream
+        from mxnet_amd import _core
+        try:
+            mx.nd.waitall()
+            _core.begin_capture(local_rank)
+            step()
+            gexec = _core.end_capture(local_rank)
+            _core.launch_graph(local_rank, gexec)
+            mx.nd.waitall()
+            graph = ('native', gexec)
+        except Exception as e:
+            print(f'# native hipgraph capture unavailable: {e}',
+                  file=sys.stderr)
+            graph = None
+    elif on_gpu and want_graph:
         try:
             torch.cuda.synchronize()
             side = torch.cuda.Stream()
@@ -120,13 +160,23 @@ def main():
             print(f'# hipgraph capture unavailable: {e}', file=sys.stderr)
             graph = None
 
+    if native:
+        mx.nd.waitall()
     if on_gpu:
         torch.cuda.synchronize()
     if distributed:
         dist.barrier()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
-        graph.replay() if graph is not None else step()
+    if native and isinstance(graph, tuple):
+        from mxnet_amd import _core
+        for _ in range(args.steps):
+            _core.launch_graph(local_rank, graph[1])
+        mx.nd.waitall()
+    else:
+        for _ in range(args.steps):
+            graph.replay() if graph is not None else step()
+        if native:
+            mx.nd.waitall()
     if on_gpu:
         torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
@@ -154,6 +204,7 @@ def main():
             'vs_baseline': round(ips / baseline, 3),
             'dtype': dtype,
             'data': 'synthetic',
+            'runtime': 'native' if native else 'torch-frontend',
             'config': {
                 'model': 'resnet50_v1.5',
                 'global_batch': B * world,

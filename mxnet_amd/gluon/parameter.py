@@ -220,7 +220,16 @@ class Parameter:
         if self._data is None:
             return
         td = torch_dtype(dtype)
-        for c, nd in self._data.items():
+        for c, nd in list(self._data.items()):
+            if nd.is_native:
+                from .. import _core
+                _core.drop_variable(nd._h)
+                new = nd.astype(dtype)
+                if self.grad_req != 'null':
+                    new.attach_grad(self.grad_req)
+                    self._grad[c] = new.grad
+                self._data[c] = new
+                continue
             t = nd._t.detach().to(td)
             new = NDArray(t)
             if self.grad_req != 'null':

@@ -1,0 +1,190 @@
+"""Native runtime on the MI355X: own storage/engine/tape with the CDNA4
+registry kernels — numerics against numpy/torch oracles, plus end-to-end
+Gluon training (the torch frontend never sees these arrays)."""
+import random
+
+import numpy as np
+import pytest
+import torch
+
+import mxnet_amd as mx
+from mxnet_amd import _core
+from mxnet_amd import autograd
+from mxnet_amd.base import set_native
+from mxnet_amd.gluon import nn, Trainer
+from mxnet_amd.gluon.loss import SoftmaxCrossEntropyLoss
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture
+def native():
+    prev = set_native(True)
+    yield
+    set_native(prev)
+
+
+def nd_gpu(arr):
+    return _core.from_numpy(np.ascontiguousarray(arr), 2, 0)
+
+
+def test_core_ops_gpu_vs_numpy():
+    rs = np.random.RandomState(0)
+    x = rs.randn(512, 300).astype('float32')
+    y = rs.rand(512, 300).astype('float32') + 0.5
+    a, b = nd_gpu(x), nd_gpu(y)
+    np.testing.assert_allclose(
+        _core.invoke('elemwise_add', [a, b], {})[0].asnumpy(), x + y,
+        rtol=1e-6)
+    np.testing.assert_allclose(
+        _core.invoke('elemwise_div', [a, b], {})[0].asnumpy(), x / y,
+        rtol=1e-5)
+    np.testing.assert_allclose(
+        _core.invoke('sum', [a], {'axis': '(1,)'})[0].asnumpy(),
+        x.sum(axis=1), rtol=1e-4)
+    np.testing.assert_allclose(
+        _core.invoke('mean', [a], {})[0].asnumpy(), [x.mean()], rtol=1e-5)
+    np.testing.assert_allclose(
+        _core.invoke('tanh', [a], {})[0].asnumpy(), np.tanh(x), atol=1e-6)
+    got = _core.invoke('transpose', [a], {})[0].asnumpy()
+    np.testing.assert_array_equal(got, x.T)
+
+
+def test_core_fp16_gemm_gpu():
+    rs = np.random.RandomState(1)
+    x = rs.randn(128, 256).astype('float16')
+    w = rs.randn(64, 256).astype('float16')
+    a, b = nd_gpu(x), nd_gpu(w)
+    got = _core.invoke('dot_nt', [a, b], {})[0].asnumpy().astype('float32')
+    want = x.astype('float32') @ w.astype('float32').T
+    err = np.abs(got - want).max() / (np.abs(want).max() + 1e-6)
+    assert err < 2e-3, err
+
+
+def test_core_backward_gpu():
+    rs = np.random.RandomState(2)
+    xv = rs.randn(64, 32).astype('float32')
+    wv = rs.randn(16, 32).astype('float32')
+    x, w = nd_gpu(xv), nd_gpu(wv)
+    gw = _core.invoke('zeros_like', [w], {})[0]
+    _core.mark_variable(w, gw, 1)
+    _core.set_recording(True)
+    y = _core.invoke('dot_nt', [x, w], {})[0]
+    r = _core.invoke('relu', [y], {})[0]
+    L = _core.invoke('sum', [r], {})[0]
+    _core.set_recording(False)
+    _core.backward([L], [], False)
+    _core.wait_all()
+    y_np = xv @ wv.T
+    want = ((y_np > 0).astype('float32').T @ xv)
+    np.testing.assert_allclose(gw.asnumpy(), want, rtol=1e-4, atol=1e-3)
+    _core.drop_variable(w)
+
+
+def test_native_gluon_lenet_gpu(native):
+    random.seed(0); np.random.seed(0); torch.manual_seed(0)
+    net = nn.HybridSequential()
+    net.add(nn.Conv2D(8, kernel_size=3, padding=1, activation='relu',
+                      layout='NHWC'),
+            nn.MaxPool2D(pool_size=2, strides=2, layout='NHWC'),
+            nn.Dense(32, activation='relu'),
+            nn.Dense(10))
+    net.initialize(ctx=mx.gpu(0))
+    tr = Trainer(net.collect_params(), 'sgd',
+                 {'learning_rate': 0.5, 'momentum': 0.9}, kvstore=None)
+    lf = SoftmaxCrossEntropyLoss()
+    rs = np.random.RandomState(0)
+    x = mx.nd.array(rs.randn(32, 12, 12, 1), ctx=mx.gpu(0))
+    y = mx.nd.array(rs.randint(0, 10, (32,)).astype('int64'), ctx=mx.gpu(0))
+    assert x.is_native
+    losses = []
+    for _ in range(10):
+        with autograd.record():
+            L = lf(net(x), y).mean()
+        L.backward()
+        tr.step(1)
+        losses.append(float(L.asnumpy()))
+    mx.nd.waitall()
+    assert losses[-1] < losses[0] * 0.8, losses
+
+
+def test_native_matches_torch_backend_gpu():
+    """Same init/data: native runtime vs torch frontend on the GPU.
+    Both run OUR kernels; this pins the engine/tape/allocator plumbing."""
+    def run(native_flag):
+        prev = set_native(native_flag)
+        try:
+            random.seed(0); np.random.seed(0); torch.manual_seed(0)
+            net = nn.HybridSequential()
+            net.add(nn.Conv2D(16, kernel_size=3, padding=1,
+                              activation='relu', layout='NHWC'),
+                    nn.Dense(32, activation='relu'),
+                    nn.Dense(10))
+            net.initialize(ctx=mx.gpu(0))
+            tr = Trainer(net.collect_params(), 'sgd',
+                         {'learning_rate': 0.2, 'momentum': 0.9},
+                         kvstore=None)
+            lf = SoftmaxCrossEntropyLoss()
+            rs = np.random.RandomState(3)
+            x = mx.nd.array(rs.randn(16, 8, 8, 8), ctx=mx.gpu(0))
+            yl = mx.nd.array(rs.randint(0, 10, (16,)).astype('int64'),
+                             ctx=mx.gpu(0))
+            out = []
+            for _ in range(8):
+                with autograd.record():
+                    L = lf(net(x), yl).mean()
+                L.backward()
+                tr.step(1)
+                out.append(float(L.asnumpy()))
+            return out
+        finally:
+            set_native(prev)
+
+    torch_traj = run(False)
+    native_traj = run(True)
+    np.testing.assert_allclose(native_traj, torch_traj, rtol=2e-2,
+                               atol=2e-2)
+
+
+def test_native_resnet50_fp16_step_gpu(native):
+    random.seed(0); np.random.seed(0); torch.manual_seed(0)
+    from mxnet_amd.gluon.model_zoo.vision import resnet50_v1
+    net = resnet50_v1(layout='NHWC')
+    net.initialize(ctx=mx.gpu(0))
+    net.cast('float16')
+    tr = Trainer(net.collect_params(), 'sgd',
+                 {'learning_rate': 0.05, 'momentum': 0.9,
+                  'multi_precision': True}, kvstore=None)
+    lf = SoftmaxCrossEntropyLoss()
+    rs = np.random.RandomState(0)
+    x = mx.nd.array(rs.randn(8, 224, 224, 3),
+                    ctx=mx.gpu(0)).astype('float16')
+    y = mx.nd.array(rs.randint(0, 1000, (8,)).astype('int64'),
+                    ctx=mx.gpu(0))
+    for _ in range(2):
+        with autograd.record():
+            L = lf(net(x), y).mean()
+        L.backward()
+        tr.step(8)
+    mx.nd.waitall()
+    v = float(L.asnumpy())
+    assert v == v and abs(v) < 100, v
+
+
+def test_native_hipgraph_capture_gpu(native):
+    """Whole-step capture/replay on the engine's compute stream."""
+    rs = np.random.RandomState(5)
+    a = nd_gpu(rs.randn(1024, 1024).astype('float32'))
+    b = nd_gpu(rs.randn(1024, 1024).astype('float32'))
+    out = _core.invoke('elemwise_add', [a, b], {})[0]
+    _core.wait_all()
+    _core.begin_capture(0)
+    _core.invoke_into('_grad_add', [b], [out], {})
+    g = _core.end_capture(0)
+    for _ in range(3):
+        _core.launch_graph(0, g)
+    _core.wait_all()
+    want = a.asnumpy() + b.asnumpy() * 4  # initial add + 3 replays...
+    # capture executes once at end_capture? No: capture does NOT run.
+    # out = a+b then 3 replays of (+b) => a + 4b? replay count = 3
+    np.testing.assert_allclose(out.asnumpy(), want, rtol=1e-5)
