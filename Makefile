@@ -12,7 +12,7 @@ CXXFLAGS := -fPIC -O3 -std=c++17 --offload-arch=$(ARCH) \
             -Wno-unused-result -parallel-jobs=4
 
 CORE_SRCS := src/core/storage.cc src/core/engine.cc src/core/ndarray.cc \
-             src/core/op.cc
+             src/core/op.cc src/core/rccl_comm.cc
 OPS_SRCS  := $(wildcard src/ops/*.hip)
 PYBIND_SRC := src/core/pybind.cc
 RAW_SRC    := src/core/raw_bind.cc
@@ -43,7 +43,7 @@ $(RAW_OBJ): $(RAW_SRC) src/core/*.h src/ops/*.h
 	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
 
 $(TARGET): $(CORE_OBJS) $(OPS_OBJS) $(PY_OBJ) $(RAW_OBJ)
-	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64
+	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64 -lrccl
 
 clean:
 	rm -rf build/core $(TARGET)

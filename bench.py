@@ -53,10 +53,7 @@ def main():
     # autograd tape, registry-dispatched CDNA4 kernels; torch only
     # supplies the host process (no torch tensors, no torch.autograd)
     native = args.native or os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
-    if native and distributed:
-        print('# native runtime multi-rank path pending RCCL kvstore; '
-              'falling back to torch-frontend', file=sys.stderr)
-        native = False
+    # native+distributed runs the own RCCL binding (NativeDistKVStore)
 
     import mxnet_amd as mx
     if native:

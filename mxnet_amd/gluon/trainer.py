@@ -60,7 +60,9 @@ class Trainer:
             self._kvstore = kind
         else:
             self._kvstore = kvs_mod.create(kind)
-        self._distributed = isinstance(self._kvstore, kvs_mod.DistKVStore)
+        self._distributed = isinstance(
+            self._kvstore,
+            (kvs_mod.DistKVStore, kvs_mod.NativeDistKVStore))
         self._async_ps = isinstance(self._kvstore,
                                     getattr(kvs_mod, 'AsyncPSKVStore', ()))
         if self._compression_params and self._kvstore is not None:
@@ -110,6 +112,15 @@ class Trainer:
 
     def _allreduce_grads(self):
         if self._kvstore is None:
+            return
+        if isinstance(self._kvstore, kvs_mod.NativeDistKVStore):
+            # native runtime: per-key engine-sequenced RCCL all-reduce,
+            # reversed so late layers' grads (ready first) go first;
+            # overlap with any remaining backward is automatic (comm ops
+            # depend only on their own grad var)
+            for i in reversed(range(len(self._params))):
+                g = self._params[i].list_grad()[0]
+                self._kvstore.pushpull(i, g, priority=-i)
             return
         if self._distributed:
             self._densify_rowsparse()

@@ -58,6 +58,10 @@ class KVStoreBase:
 
 
 def create(name='local'):
+    from ..base import native_mode
+    if native_mode() and isinstance(name, str) and (
+            name.startswith('dist') or name in ('nccl', 'device')):
+        return NativeDistKVStore(name)
     """Factory (reference KVStore::Create kvstore.cc:42-80)."""
     name = name.lower()
     if name == 'dist_async':
@@ -175,6 +179,67 @@ class KVStore(KVStoreBase):
             states = pickle.load(f)
         if self._updater:
             self._updater.states.update(states)
+
+
+class NativeDistKVStore(KVStoreBase):
+    """Own RCCL binding, engine-sequenced (native runtime path).
+
+    Reference counterpart: KVStoreNCCL (kvstore_nccl.h:62) — the grouped
+    launch + stream-sync 3-op pattern (:267-443) collapses into ONE engine
+    op per collective on the device's dedicated comm stream
+    (FnProperty::kGPUPrioritized): the comm worker waits the producing
+    compute kernels via hipStreamWaitEvent, launches rcclAllReduce over
+    xGMI, and consumers wait its completion event — overlap with backward
+    falls out of the engine's dependency tracking.
+    """
+
+    def __init__(self, kind='dist_sync'):
+        import os as _os
+        from .. import _core
+        self._type = kind
+        world = int(_os.environ.get('WORLD_SIZE', 1))
+        rank = int(_os.environ.get('RANK', 0))
+        dev = int(_os.environ.get('LOCAL_RANK', 0))
+        _os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        _os.environ.setdefault('MASTER_PORT', '29741')
+        self._core = _core
+        _core.rccl_init(world, rank, dev)
+
+    def set_gradient_compression(self, compression_params):
+        raise NotImplementedError(
+            'gradient compression on the native RCCL path: pending')
+
+    @property
+    def rank(self):
+        return self._core.rccl_rank()
+
+    @property
+    def num_workers(self):
+        return self._core.rccl_world()
+
+    def init(self, key, value):
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        self._core.rccl_broadcast(v._h, 0)
+
+    def broadcast(self, key, value, out, priority=0):
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        self._core.rccl_broadcast(v._h, 0)
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        for o in outs:
+            if o is not v:
+                v.copyto(o)
+
+    def pushpull(self, key, value, out=None, priority=0, async_op=False):
+        """All-reduce-average in place; async by construction (the engine
+        orders it against producers/consumers via vars)."""
+        v = value[0] if isinstance(value, (list, tuple)) else value
+        self._core.rccl_allreduce(v._h, True)
+        if out is not None:
+            outs = out if isinstance(out, (list, tuple)) else [out]
+            for o in outs:
+                if o is not v:
+                    v.copyto(o)
+        return None
 
 
 class DistKVStore(KVStoreBase):

@@ -268,7 +268,11 @@ struct Engine::Impl {
         exc = std::current_exception();
       }
       EventPtr done;
-      if (rc.ctx.is_gpu() && !exc && !capturing) {
+      // re-read AFTER fn: the BeginCapture op flips capturing on inside
+      // its fn — recording an event then would inject a graph node whose
+      // pooled hipEvent outlives the exec (heap corruption on destroy)
+      bool capturing_now = dw && dw->capturing.load(std::memory_order_relaxed);
+      if (rc.ctx.is_gpu() && !exc && !capturing_now) {
         // execution-completion marker for cross-stream/CPU consumers
         done = std::make_shared<EventRef>();
         done->dev = rc.ctx.dev_id;

@@ -115,6 +115,14 @@ void InvokeIntoPy(const std::string& name, const std::vector<NDArray>& inputs,
 
 void init_raw(py::module_& m);
 
+namespace mxcore {
+void RcclInit(int world, int rank, int dev);
+int RcclWorld();
+int RcclRank();
+void RcclAllReduce(const NDArray& a, bool average);
+void RcclBroadcast(const NDArray& a, int root);
+}  // namespace mxcore
+
 PYBIND11_MODULE(_core, m) {
   init_raw(m);
   m.doc() = "mxnet_amd native runtime (storage + engine + ndarray + ops)";
@@ -233,6 +241,22 @@ PYBIND11_MODULE(_core, m) {
   });
   m.def("release_all", [](int dev_type, int dev_id) {
     Storage::Get()->ReleaseAll(MakeCtx(dev_type, dev_id));
+  });
+
+  // RCCL collectives (engine-sequenced on the comm stream)
+  m.def("rccl_init", [](int world, int rank, int dev) {
+    py::gil_scoped_release rel;
+    RcclInit(world, rank, dev);
+  });
+  m.def("rccl_world", &RcclWorld);
+  m.def("rccl_rank", &RcclRank);
+  m.def("rccl_allreduce", [](const NDArray& a, bool average) {
+    py::gil_scoped_release rel;
+    RcclAllReduce(a, average);
+  });
+  m.def("rccl_broadcast", [](const NDArray& a, int root) {
+    py::gil_scoped_release rel;
+    RcclBroadcast(a, root);
   });
 
   m.def("device_count", [] {

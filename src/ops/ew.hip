@@ -782,4 +782,16 @@ bool _registered = [] {
 }();
 
 }  // namespace
+
+// comm-stream helper (rccl_comm.cc): in-place scale without at:: / registry
+void ScaleInPlace(void* p, int64_t n, int dtype, float alpha,
+                  hipStream_t s) {
+  MXC_DISPATCH_FLOAT(dtype, "scale_inplace", {
+    unary_kernel<scalar_t, FMulS><<<grid_for(n, 16 / sizeof(scalar_t)),
+                                    kBlock, 0, s>>>(
+        (const scalar_t*)p, (scalar_t*)p, n, alpha, 0.f);
+  });
+  HIP_CHECK_LAST();
+}
+
 }  // namespace mxcore
