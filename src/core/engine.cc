@@ -499,6 +499,10 @@ uintptr_t Engine::EndCapture(int dev) {
         dw->capturing.store(false);
         hipGraph_t g = nullptr;
         MX_HIP_CALL(hipStreamEndCapture(rc.stream, &g));
+        size_t nnodes = 0;
+        (void)hipGraphGetNodes(g, nullptr, &nnodes);
+        if (env_int("MXNET_ENGINE_DEBUG", 0))
+          fprintf(stderr, "[engine] captured graph: %zu nodes\n", nnodes);
         hipGraphExec_t exec = nullptr;
         MX_HIP_CALL(hipGraphInstantiate(&exec, g, nullptr, nullptr, 0));
         (void)hipGraphDestroy(g);
