@@ -43,7 +43,7 @@ $(RAW_OBJ): $(RAW_SRC) src/core/*.h src/ops/*.h
 	$(HIPCC) $(CXXFLAGS) -x hip -c $< -o $@
 
 $(TARGET): $(CORE_OBJS) $(OPS_OBJS) $(PY_OBJ) $(RAW_OBJ)
-	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64 -lrccl
+	$(HIPCC) -shared -fPIC --offload-arch=$(ARCH) $^ -o $@ -L/opt/rocm/lib -lamdhip64
 
 clean:
 	rm -rf build/core $(TARGET)

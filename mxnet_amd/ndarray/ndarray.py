@@ -31,10 +31,11 @@ class NDArray:
     that supports the native path branches on ``is_native`` explicitly.
     """
 
-    __slots__ = ('_tt', '_h', '_native_grad')
+    __slots__ = ('_tt', '_h', '_native_grad', '_bn_presums')
 
     def __init__(self, data):
         self._native_grad = None
+        self._bn_presums = None
         if isinstance(data, _core.NDArray):
             self._h = data
             self._tt = None

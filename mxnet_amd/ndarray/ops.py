@@ -67,13 +67,23 @@ def Convolution(data, weight, bias=None, kernel=None, stride=(1, 1),
     if data.is_native:
         assert layout == 'NHWC', \
             'native runtime convolution is NHWC (MI355X-first layout)'
+        from .. import _core
         ins = [data, weight] + ([] if (no_bias or bias is None) else [bias])
-        return _ninv('Convolution', ins,
-                     {'kernel': _tup_attr(kernel),
-                      'stride': _tup_attr(stride), 'pad': _tup_attr(pad),
-                      'dilate': _tup_attr(dilate),
-                      'num_filter': str(num_filter),
-                      'num_group': str(num_group)})
+        want_stats = (no_bias or bias is None) and _core.is_recording()
+        y, stats = _ninv('Convolution', ins,
+                         {'kernel': _tup_attr(kernel),
+                          'stride': _tup_attr(stride),
+                          'pad': _tup_attr(pad),
+                          'dilate': _tup_attr(dilate),
+                          'num_filter': str(num_filter),
+                          'num_group': str(num_group),
+                          'want_stats': '1' if want_stats else '0'},
+                         nout=2)
+        if stats.size > 1:
+            # fused per-channel {sum,ssq} from the conv epilogue — the
+            # following BatchNorm consumes it (skips its reduce pass)
+            y._bn_presums = stats
+        return y
     return NDArray(_nn.conv2d(_t(data), _t(weight),
                               None if no_bias else _t(bias),
                               stride=_pair(stride), pad=_pair(pad),
@@ -129,10 +139,18 @@ def BatchNorm(data, gamma, beta, moving_mean, moving_var, eps=1e-5,
         ins = [data, gamma, beta, moving_mean, moving_var]
         if residual is not None:
             ins.append(residual)
+        presums = getattr(data, '_bn_presums', None)
+        if presums is not None and training:
+            ins.append(presums)
+            data._bn_presums = None  # consume once
         outs = _ninv('BatchNorm', ins,
                      {'momentum': str(momentum), 'eps': str(eps),
                       'training': '1' if training else '0',
-                      'fuse_relu': '1' if fuse_relu else '0'}, nout=4)
+                      'fuse_relu': '1' if fuse_relu else '0',
+                      'has_res': '1' if residual is not None else '0',
+                      'has_presums':
+                          '1' if (presums is not None and training)
+                          else '0'}, nout=4)
         return outs[0]
     if layout is None:
         layout = 'NHWC' if axis in (-1, _t(data).dim() - 1) else 'NCHW'

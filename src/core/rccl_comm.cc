@@ -23,6 +23,8 @@
 #include <mutex>
 #include <thread>
 
+#include <dlfcn.h>
+
 #include <rccl/rccl.h>
 
 #include "engine.h"
@@ -30,11 +32,52 @@
 
 namespace mxcore {
 
-#define MX_RCCL_CALL(expr)                                              \
-  do {                                                                  \
-    ncclResult_t r_ = (expr);                                           \
-    MX_CHECK(r_ == ncclSuccess,                                         \
-             "RCCL error: " << ncclGetErrorString(r_) << " at " #expr); \
+// librccl is resolved with dlopen at init time instead of link-time:
+// torch-ROCm already maps its own librccl into the process, and loading a
+// SECOND copy (ld-linked /opt/rocm one) corrupts shared HSA state — the
+// soname lookup below returns the already-loaded copy when there is one.
+namespace {
+
+struct RcclFns {
+  ncclResult_t (*GetUniqueId)(ncclUniqueId*);
+  ncclResult_t (*CommInitRank)(ncclComm_t*, int, ncclUniqueId, int);
+  ncclResult_t (*AllReduce)(const void*, void*, size_t, ncclDataType_t,
+                            ncclRedOp_t, ncclComm_t, hipStream_t);
+  ncclResult_t (*Broadcast)(const void*, void*, size_t, ncclDataType_t,
+                            int, ncclComm_t, hipStream_t);
+  const char* (*GetErrorString)(ncclResult_t);
+};
+
+RcclFns* GetRccl() {
+  static RcclFns* fns = [] {
+    void* h = dlopen("librccl.so.1", RTLD_NOW | RTLD_GLOBAL);
+    if (!h) h = dlopen("librccl.so", RTLD_NOW | RTLD_GLOBAL);
+    MX_CHECK(h, "librccl not found: " << dlerror());
+    auto* f = new RcclFns();
+    f->GetUniqueId =
+        (decltype(f->GetUniqueId))dlsym(h, "ncclGetUniqueId");
+    f->CommInitRank =
+        (decltype(f->CommInitRank))dlsym(h, "ncclCommInitRank");
+    f->AllReduce = (decltype(f->AllReduce))dlsym(h, "ncclAllReduce");
+    f->Broadcast = (decltype(f->Broadcast))dlsym(h, "ncclBroadcast");
+    f->GetErrorString =
+        (decltype(f->GetErrorString))dlsym(h, "ncclGetErrorString");
+    MX_CHECK(f->GetUniqueId && f->CommInitRank && f->AllReduce &&
+                 f->Broadcast && f->GetErrorString,
+             "librccl symbols missing");
+    return f;
+  }();
+  return fns;
+}
+
+}  // namespace
+
+#define MX_RCCL_CALL(expr)                                               \
+  do {                                                                   \
+    ncclResult_t r_ = (expr);                                            \
+    MX_CHECK(r_ == ncclSuccess, "RCCL error: "                           \
+                                    << GetRccl()->GetErrorString(r_)     \
+                                    << " at " #expr);                    \
   } while (0)
 
 namespace {
@@ -124,9 +167,9 @@ void RcclInit(int world, int rank, int dev) {
   if (g_rccl.comm) return;
   MX_HIP_CALL(hipSetDevice(dev));
   ncclUniqueId id;
-  if (rank == 0) MX_RCCL_CALL(ncclGetUniqueId(&id));
+  if (rank == 0) MX_RCCL_CALL(GetRccl()->GetUniqueId(&id));
   if (world > 1) ExchangeId(&id, world, rank);
-  MX_RCCL_CALL(ncclCommInitRank(&g_rccl.comm, world, id, rank));
+  MX_RCCL_CALL(GetRccl()->CommInitRank(&g_rccl.comm, world, id, rank));
   g_rccl.world = world;
   g_rccl.rank = rank;
   g_rccl.dev = dev;
@@ -152,8 +195,8 @@ void RcclAllReduce(const NDArray& a, bool average) {
   Engine::Get()->PushAsync(
       [chunk, p, n, dtype, comm, world, average](const RunContext& rc) {
         if (world > 1)
-          MX_RCCL_CALL(ncclAllReduce(p, p, n, RcclType(dtype), ncclSum,
-                                     comm, rc.stream));
+          MX_RCCL_CALL(GetRccl()->AllReduce(p, p, n, RcclType(dtype),
+                                            ncclSum, comm, rc.stream));
         // average in the same stream order (ring result / world)
         if (average && world > 1) {
           // scale via a small elementwise launch on the comm stream
@@ -175,8 +218,8 @@ void RcclBroadcast(const NDArray& a, int root) {
   ncclComm_t comm = g_rccl.comm;
   Engine::Get()->PushAsync(
       [chunk, p, n, dtype, comm, root](const RunContext& rc) {
-        MX_RCCL_CALL(ncclBroadcast(p, p, n, RcclType(dtype), root, comm,
-                                   rc.stream));
+        MX_RCCL_CALL(GetRccl()->Broadcast(p, p, n, RcclType(dtype), root,
+                                          comm, rc.stream));
       },
       a.ctx(), {}, {a.var()}, FnProperty::kGPUPrioritized, "RcclBroadcast");
 }

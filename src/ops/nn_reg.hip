@@ -423,13 +423,39 @@ bool _registered_nn = [] {
     int64_t Q = (xs[2] + 2 * pad[1] - dil[1] * (S - 1) - 1) / stride[1] + 1;
     return TShape{xs[0], P, Q, ws[0]};
   };
-  RegN("Convolution").in(-1)
+  RegN("Convolution").in(-1).out(2)
       .infer([conv_out_shape](const NodeAttrs& a,
                               const std::vector<TShape>& is,
                               const std::vector<int>& it,
                               std::vector<TShape>* os, std::vector<int>* ot) {
-        os->assign(1, conv_out_shape(a, is[0], is[1]));
+        TShape ys = conv_out_shape(a, is[0], is[1]);
+        os->assign(1, ys);
+        // fused per-channel {sum,ssq} epilogue ([64,2,K] fp32) when the
+        // executing path supports it (mirrors conv2d_fwd_raw routing:
+        // MFMA igemm always; 1x1-s1 GEMM unless split-K engages); the
+        // following BatchNorm consumes it and skips its reduce pass
+        bool want = a.GetBool("want_stats", false);
+        int64_t C = is[0][3], K = is[1][0], R = is[1][1], S = is[1][2];
+        int groups = (int)a.GetInt("num_group", 1);
+        bool mfma = (it[0] == kFloat16 || it[0] == kBFloat16) &&
+                    (C / groups) % 8 == 0 && groups == 1 &&
+                    !(groups == C && K == C && is[1][3] == 1);
+        bool have = want && mfma;
+        if (have && R == 1 && S == 1) {
+          auto stride = a.GetTuple("stride", {1, 1});
+          auto pad = a.GetTuple("pad", {0, 0});
+          if (stride[0] == 1 && stride[1] == 1 && pad[0] == 0 &&
+              pad[1] == 0) {
+            int64_t M = ys[0] * ys[1] * ys[2];
+            int64_t K8 = (C + 7) / 8 * 8;
+            int64_t nwg = ((M + 127) / 128) * ((K + 127) / 128);
+            int64_t nk = (K8 + 63) / 64;
+            if (nwg < 512 && nk > 16) have = false;  // split-K: no stats
+          }
+        }
+        os->push_back(have ? TShape{64, 2, K} : TShape{1});
         ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
       })
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         auto stride = a.GetTuple("stride", {1, 1});
@@ -438,7 +464,8 @@ bool _registered_nn = [] {
         conv2d_fwd_raw(LC(o), in[0], in[1],
                        in.size() > 2 ? Arr(in[2]) : Arr(), stride[0],
                        stride[1], pad[0], pad[1], dil[0], dil[1],
-                       (int)a.GetInt("num_group", 1), out[0], Arr());
+                       (int)a.GetInt("num_group", 1), out[0],
+                       out[1].size() > 1 ? Arr(out[1]) : Arr());
       })
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         CPU_FLOAT_ONLY(in[0], "Convolution");
@@ -455,6 +482,7 @@ bool _registered_nn = [] {
         const float* w = (const float*)in[1].dptr;
         const float* b = in.size() > 2 ? (const float*)in[2].dptr : nullptr;
         float* y = (float*)out[0].dptr;
+        if (out.size() > 1) ((float*)out[1].dptr)[0] = 0.f;  // dummy stats
         for (long n = 0; n < NB; ++n)
           for (long p = 0; p < P; ++p)
             for (long q = 0; q < Q; ++q)
@@ -781,12 +809,15 @@ bool _registered_nn = [] {
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         bool training = a.GetBool("training", false);
         bool relu = a.GetBool("fuse_relu", false);
-        Arr res = in.size() > 5 ? Arr(in[5]) : Arr();
+        bool has_res = a.GetBool("has_res", false);
+        bool has_pre = a.GetBool("has_presums", false);
+        Arr res = has_res ? Arr(in[5]) : Arr();
+        Arr pre = has_pre ? Arr(in[5 + (has_res ? 1 : 0)]) : Arr();
         if (training) {
           bool mask_ok = out[3].size() > 1;
           bn_fwd_train_raw(LC(o), in[0], in[1], in[2], in[3], in[4],
                            a.GetFloat("momentum", 0.9),
-                           a.GetFloat("eps", 1e-5), relu, res, Arr(),
+                           a.GetFloat("eps", 1e-5), relu, res, pre,
                            out[0], out[1], out[2],
                            mask_ok ? Arr(out[3]) : Arr());
         } else {
@@ -806,8 +837,9 @@ bool _registered_nn = [] {
         const float* b = (const float*)in[2].dptr;
         float* rm = (float*)in[3].dptr;
         float* rv = (float*)in[4].dptr;
-        const float* res =
-            in.size() > 5 ? (const float*)in[5].dptr : nullptr;
+        const float* res = a.GetBool("has_res", false)
+                               ? (const float*)in[5].dptr
+                               : nullptr;  // CPU path recomputes stats
         float* y = (float*)out[0].dptr;
         float* smean = (float*)out[1].dptr;
         float* sistd = (float*)out[2].dptr;
@@ -843,8 +875,7 @@ bool _registered_nn = [] {
                -> std::vector<NDArray> {
         NodeAttrs a = n.attrs;
         bool relu = a.GetBool("fuse_relu", false);
-        bool has_res = n.inputs.size() > 5;
-        a.d["has_res"] = has_res ? "1" : "0";
+        bool has_res = a.GetBool("has_res", false);
         bool mask_ok = n.outputs[3].size() > 1;
         a.d["use_mask"] = mask_ok ? "1" : "0";
         // inputs to the bwd op: dy, x, gamma, save_mean, save_istd,
