@@ -2,6 +2,7 @@
 
 #include <algorithm>
 #include <mutex>
+#include <unordered_set>
 
 namespace mxcore {
 
@@ -222,6 +223,22 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
   OpEntry* ones_op = OpRegistry::Get()->Find("ones_like");
   MX_CHECK(add_into && ones_op, "core grad ops missing");
 
+  // forward pass over the tape: which chunks require grad at all
+  // (leaves plus anything computed from them) — reference OpReqType
+  // kNullOp semantics; skips e.g. the input gradient of the first conv
+  std::unordered_set<NDArray::Chunk*> need;
+  for (auto& kv : leaves_) need.insert(kv.first);
+  for (auto& node : tape_) {
+    bool any_in = false;
+    for (auto& in : node.inputs)
+      if (need.count(in.chunk_.get())) {
+        any_in = true;
+        break;
+      }
+    if (any_in)
+      for (auto& out : node.outputs) need.insert(out.chunk_.get());
+  }
+
   std::unordered_map<NDArray::Chunk*, NDArray> grads;
   for (size_t i = 0; i < ys.size(); ++i) {
     NDArray g;
@@ -248,6 +265,13 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
     MX_CHECK(node.op->fbackward,
              "op '" << node.op->name
                     << "' is not differentiable but a gradient flows into it");
+    node.need_igrad.assign(node.inputs.size(), 0);
+    bool any_need = false;
+    for (size_t i = 0; i < node.inputs.size(); ++i) {
+      node.need_igrad[i] = need.count(node.inputs[i].chunk_.get()) ? 1 : 0;
+      any_need |= node.need_igrad[i] != 0;
+    }
+    if (!any_need) continue;
     // missing head grads are zeros
     for (size_t i = 0; i < ograds.size(); ++i)
       if (ograds[i].is_none()) {
@@ -260,7 +284,7 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
                     << igrads.size() << " grads for " << node.inputs.size()
                     << " inputs");
     for (size_t i = 0; i < igrads.size(); ++i) {
-      if (igrads[i].is_none()) continue;
+      if (igrads[i].is_none() || !node.need_igrad[i]) continue;
       NDArray::Chunk* key = node.inputs[i].chunk_.get();
       auto git = grads.find(key);
       if (git == grads.end()) {

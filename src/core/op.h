@@ -115,6 +115,11 @@ struct TapeNode {
   NodeAttrs attrs;
   std::vector<NDArray> inputs;
   std::vector<NDArray> outputs;
+  // filled by Backward before fbackward runs: whether each input needs a
+  // gradient (leaf or derived from one) — impls skip dead grads, e.g.
+  // the data gradient of the stem convolution (reference
+  // needs_input_grad / OpReqType kNullOp)
+  std::vector<char> need_igrad;
 };
 
 // ---------------------------------------------------------------------------

@@ -127,6 +127,24 @@ PYBIND11_MODULE(_core, m) {
   init_raw(m);
   m.doc() = "mxnet_amd native runtime (storage + engine + ndarray + ops)";
 
+  // stop the engine's worker threads BEFORE the interpreter/HIP runtime
+  // tear down — workers still polling streams during libc/HSA teardown
+  // corrupt the heap at exit (reference: engine shutdown in
+  // LibraryInitializer, src/initialize.cc)
+  {
+    auto atexit = py::module_::import("atexit");
+    atexit.attr("register")(py::cpp_function([]() {
+      {
+        py::gil_scoped_release rel;
+        try {
+          Engine::Get()->WaitForAll();
+        } catch (...) {
+        }
+        Engine::Get()->StopWorkers();
+      }
+    }));
+  }
+
   py::class_<NDArray>(m, "NDArray")
       .def(py::init([](const std::vector<int64_t>& shape, int dev_type,
                        int dev_id, int dtype) {

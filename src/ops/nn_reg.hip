@@ -372,8 +372,9 @@ bool _registered_nn = [] {
         NDArray x2 = n.inputs[0].Reshape({n.inputs[0].size() / K, K});
         NDArray dy2 = og[0].Reshape({x2.shape()[0], n.inputs[1].shape()[0]});
         std::vector<NDArray> r(n.inputs.size());
-        r[0] = RunN("dot_nn", {}, {dy2, n.inputs[1]})
-                   .Reshape(n.inputs[0].shape());
+        if (n.need_igrad.empty() || n.need_igrad[0])
+          r[0] = RunN("dot_nn", {}, {dy2, n.inputs[1]})
+                     .Reshape(n.inputs[0].shape());
         r[1] = RunN("dot_tn", {}, {dy2, x2});
         if (n.inputs.size() > 2) r[2] = RunN("colsum", {}, {dy2});
         return r;
@@ -512,7 +513,9 @@ bool _registered_nn = [] {
         a.d["__r__"] = std::to_string(n.inputs[1].shape()[1]);
         a.d["__s__"] = std::to_string(n.inputs[1].shape()[2]);
         a.d["__cg__"] = std::to_string(n.inputs[1].shape()[3]);
-        r[0] = RunN("_conv_bwd_data", a, {og[0], n.inputs[1]});
+        bool need_dx = n.need_igrad.empty() || n.need_igrad[0];
+        if (need_dx)
+          r[0] = RunN("_conv_bwd_data", a, {og[0], n.inputs[1]});
         r[1] = RunN("_conv_bwd_weight", a, {og[0], n.inputs[0]});
         if (n.inputs.size() > 2) {
           NDArray dy2 = og[0].Reshape(
