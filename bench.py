@@ -31,9 +31,14 @@ def parse_args():
     p.add_argument('--image-size', type=int, default=224)
     p.add_argument('--model', default='resnet50_v1')
     p.add_argument('--dtype', default='float16')
+    p.add_argument('--runtime', choices=['native', 'torch'],
+                   default=os.environ.get('MXNET_BENCH_RUNTIME', 'native'),
+                   help='native = the own C++ runtime (pooled HIP storage, '
+                        'threaded engine, own tape, RCCL binding) — the '
+                        'default; torch = torch-tensor frontend over the '
+                        'same CDNA4 kernel library')
     p.add_argument('--native', action='store_true',
-                   help='run on the native C++ runtime (own engine/'
-                        'storage/tape) instead of the torch frontend')
+                   help='(alias for --runtime native)')
     return p.parse_args()
 
 
@@ -52,7 +57,8 @@ def main():
     # own C++ runtime — pooled HIP storage, threaded engine streams, own
     # autograd tape, registry-dispatched CDNA4 kernels; torch only
     # supplies the host process (no torch tensors, no torch.autograd)
-    native = args.native or os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
+    native = args.native or args.runtime == 'native' or \
+        os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
     # native+distributed runs the own RCCL binding (NativeDistKVStore)
 
     import mxnet_amd as mx
@@ -122,23 +128,46 @@ def main():
     # single-process only (MXNET_BENCH_HIPGRAPH=1 forces, =0 disables).
     graph = None
     env_g = os.environ.get('MXNET_BENCH_HIPGRAPH', '')
-    want_graph = env_g == '1' or (env_g != '0' and not distributed)
+    want_graph = env_g == '1' or env_g != '0'
     if on_gpu and want_graph and native:
-        # native runtime: capture the engine's compute stream
+        # Native runtime: capture the engine compute stream.  Two graphs:
+        # G1 = forward+backward, G2 = fused optimizer updates + grad
+        # zeroing; the RCCL all-reduces (engine comm stream) run eagerly
+        # between the replays — for ResNet-50 the whole gradient set is
+        # ~50 MB, well under a millisecond over xGMI, so overlap is not
+        # the constraint the launch gaps are.
         from mxnet_amd import _core
         try:
             mx.nd.waitall()
             _core.begin_capture(local_rank)
-            step()
-            gexec = _core.end_capture(local_rank)
-            _core.launch_graph(local_rank, gexec)
+            with autograd.record():
+                out = net(x)
+                L = loss_fn(out, label)
+            L.backward()
+            g1 = _core.end_capture(local_rank)
             mx.nd.waitall()
-            graph = ('native', gexec)
+            _core.begin_capture(local_rank)
+            trainer._optimizer.rescale_grad = 1.0 / B
+            trainer._update(False)
+            g2 = _core.end_capture(local_rank)
+            mx.nd.waitall()
+            grad_handles = [p.list_grad()[0]._h for p in trainer._params]
+            def replay_step():
+                _core.launch_graph(local_rank, g1)
+                if distributed:
+                    trainer._allreduce_grads()
+                # read-deps on the grads order G2 behind the comm-stream
+                # all-reduces (engine events; no host sync)
+                _core.launch_graph(local_rank, g2,
+                                   grad_handles if distributed else [])
+            replay_step()
+            mx.nd.waitall()
+            graph = ('native', replay_step)
         except Exception as e:
             print(f'# native hipgraph capture unavailable: {e}',
                   file=sys.stderr)
             graph = None
-    elif on_gpu and want_graph:
+    elif on_gpu and want_graph and not distributed:
         try:
             torch.cuda.synchronize()
             side = torch.cuda.Stream()
@@ -165,9 +194,8 @@ def main():
         dist.barrier()
     t0 = time.perf_counter()
     if native and isinstance(graph, tuple):
-        from mxnet_amd import _core
         for _ in range(args.steps):
-            _core.launch_graph(local_rank, graph[1])
+            graph[1]()
         mx.nd.waitall()
     else:
         for _ in range(args.steps):

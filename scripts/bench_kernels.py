@@ -9,7 +9,7 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
-from mxnet_amd import _hipops as ext
+from mxnet_amd.ops import hipshim as ext
 
 DEV = 'cuda:0'
 B = int(os.environ.get('BENCH_BATCH', 256))

@@ -512,12 +512,13 @@ uintptr_t Engine::EndCapture(int dev) {
   return fut.get();
 }
 
-void Engine::LaunchGraph(int dev, uintptr_t exec) {
+void Engine::LaunchGraph(int dev, uintptr_t exec,
+                         const std::vector<VarId>& after) {
   PushAsync(
       [exec](const RunContext& rc) {
         MX_HIP_CALL(hipGraphLaunch((hipGraphExec_t)exec, rc.stream));
       },
-      Context::GPU(dev), {}, {}, FnProperty::kNormal, "GraphLaunch");
+      Context::GPU(dev), after, {}, FnProperty::kNormal, "GraphLaunch");
 }
 
 void Engine::StopWorkers() {

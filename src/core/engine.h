@@ -85,7 +85,9 @@ class Engine {
   // but their launches are captured into a graph instead of executing.
   void BeginCapture(int dev);
   uintptr_t EndCapture(int dev);      // returns hipGraphExec_t
-  void LaunchGraph(int dev, uintptr_t exec);
+  // optional read-deps order the launch behind e.g. comm-stream work
+  void LaunchGraph(int dev, uintptr_t exec,
+                   const std::vector<VarId>& after = {});
 
   void StopWorkers();  // tests / atfork
 

@@ -218,10 +218,15 @@ PYBIND11_MODULE(_core, m) {
     py::gil_scoped_release rel;
     return Engine::Get()->EndCapture(dev);
   });
-  m.def("launch_graph", [](int dev, uintptr_t exec) {
-    py::gil_scoped_release rel;
-    Engine::Get()->LaunchGraph(dev, exec);
-  });
+  m.def("launch_graph",
+        [](int dev, uintptr_t exec, const std::vector<NDArray>& after) {
+          std::vector<VarId> vars;
+          for (auto& a : after) vars.push_back(a.var());
+          py::gil_scoped_release rel;
+          Engine::Get()->LaunchGraph(dev, exec, vars);
+        },
+        py::arg("dev"), py::arg("exec"),
+        py::arg("after") = std::vector<NDArray>());
 
   // autograd tape
   m.def("set_recording", [](bool r) {
