@@ -30,6 +30,24 @@ V1_MAGIC = 0xF993fac8
 
 
 def _write_ndarray(f, nd):
+    if getattr(nd, 'is_native', False):
+        # native runtime: bytes via asnumpy (bf16 comes back as its u16
+        # bit pattern, matching the on-disk representation)
+        arr = nd.asnumpy()
+        if str(nd.dtype) == 'bfloat16':
+            type_flag = 11
+        else:
+            type_flag = NP_TO_TYPE_FLAG[arr.dtype]
+        shape = tuple(arr.shape)
+        f.write(struct.pack('<I', V3_MAGIC if len(shape) == 0 else V2_MAGIC))
+        f.write(struct.pack('<i', 0))
+        f.write(struct.pack('<i', len(shape)))
+        for s in shape:
+            f.write(struct.pack('<q', s))
+        f.write(struct.pack('<ii', 1, 0))
+        f.write(struct.pack('<i', type_flag))
+        f.write(_np.ascontiguousarray(arr).tobytes())
+        return
     t = nd._t.detach().cpu().contiguous()
     type_flag = TORCH_TO_TYPE_FLAG[t.dtype]
     shape = tuple(t.shape)
