@@ -104,3 +104,52 @@ class NDArrayIter(DataIter):
         return DataBatch(data, label, pad=pad,
                          provide_data=self.provide_data,
                          provide_label=self.provide_label)
+
+
+class ImageRecordIter(DataIter):
+    """Threaded RecordIO image iterator (reference ImageRecordIter,
+    src/io/iter_image_recordio_2.cc:887): C++ decode threads run the own
+    baseline JPEG decoder (or raw records) + resize/crop/mirror augment
+    into uint8 NHWC batches; see mxnet_amd/_imageio."""
+
+    def __init__(self, path_imgrec, batch_size, data_shape,
+                 shuffle=False, rand_crop=False, rand_mirror=False,
+                 resize=0, preprocess_threads=0, seed=0, label_width=1,
+                 **kwargs):
+        super().__init__(batch_size)
+        from .. import _imageio
+        # data_shape mxnet-style (C,H,W) or NHWC (H,W,C with C last)
+        if len(data_shape) == 3 and data_shape[0] in (1, 3):
+            c, h, w = data_shape
+        else:
+            h, w, c = data_shape
+        assert c == 3, 'ImageRecordIter decodes RGB'
+        self._h, self._w = h, w
+        self._it = _imageio.ImageRecordIter(
+            path_imgrec, batch_size, h, w, preprocess_threads, shuffle,
+            rand_crop, rand_mirror, resize, seed)
+        self.batch_size = batch_size
+
+    @property
+    def num_records(self):
+        return self._it.size
+
+    def reset(self):
+        self._it.reset()
+
+    def next(self):
+        from ..ndarray import ndarray as nd
+        got, data, labels = self._it.next_batch(self.batch_size, self._h,
+                                                self._w)
+        if got == 0:
+            raise StopIteration
+        batch = DataBatch(
+            data=[nd.array(data[:got])],
+            label=[nd.array(labels[:got])],
+            pad=self.batch_size - got)
+        return batch
+
+    def next_raw(self):
+        """(got, uint8 ndarray [B,H,W,3], float32 labels) — zero-copy for
+        pipelines that upload/normalize on the GPU."""
+        return self._it.next_batch(self.batch_size, self._h, self._w)

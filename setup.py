@@ -26,5 +26,11 @@ setup(
             cxx_std=17,
             extra_compile_args=['-O2', '-pthread'],
         ),
+        Pybind11Extension(
+            'mxnet_amd._imageio',
+            [os.path.join(ROOT, 'src', 'imageio.cc')],
+            cxx_std=17,
+            extra_compile_args=['-O3', '-pthread'],
+        ),
     ],
 )
