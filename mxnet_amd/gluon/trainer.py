@@ -3,9 +3,13 @@
 step(batch_size) = allreduce_grads (KVStore pushpull, per-parameter,
 priority = -index so late layers' gradients — produced first by backward —
 communicate first, reference trainer.py:385-409) + update (fused optimizer
-step per device).  In the one-process-per-GPU RCCL layout the pushpull is
-an async all-reduce launched in backward-completion order, overlapping the
-remaining backward exactly like the reference's kGPUPrioritized engine ops.
+step per device).
+
+Gradient sync backends: on the NATIVE runtime, NativeDistKVStore drives
+the own RCCL binding (engine-sequenced rcclAllReduce on the comm stream;
+overlap with backward comes from the engine's event ordering).  On the
+torch-tensor frontend, DistKVStore uses torch.distributed (RCCL on GPU,
+gloo on CPU) with DDP-style bucketed hooks.
 """
 import torch
 
