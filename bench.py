@@ -123,19 +123,27 @@ def main():
                                       rand_crop=True, rand_mirror=True,
                                       preprocess_threads=0, seed=rank)
         if native:
-            staging = mx.nd.empty((B, S, S, 3), ctx=ctx, dtype='uint8')
+            # decode threads write straight into a PINNED host buffer
+            # (own storage manager, hipHostMalloc pool); upload rides the
+            # engine copy stream; the cast lands in the fixed network
+            # input buffer on the compute queue
+            host_dev = 3 if on_gpu else 1  # pinned on GPU boxes
+            pinned = _core.NDArray([B, S, S, 3], host_dev, 0, 3)  # u8
+            plabels = _core.NDArray([B], host_dev, 0, 0)          # f32
+            staging = mx.nd.empty((B, S, S, 3), ctx=ctx, dtype='uint8') \
+                if on_gpu else None
 
             def feeder():
-                got, data_np, _labels = rec_it.next_raw()
+                got = rec_it._it.next_into(pinned.data_ptr,
+                                           plabels.data_ptr)
                 if got < B:
                     rec_it.reset()
-                    got, data_np, _labels = rec_it.next_raw()
-                host = _core.from_numpy(data_np, 1, 0)
+                    rec_it._it.next_into(pinned.data_ptr, plabels.data_ptr)
                 if on_gpu:
-                    host.copyto(staging._h)
+                    pinned.copyto(staging._h)
                     src_h = staging._h
                 else:
-                    src_h = host
+                    src_h = pinned
                 _core.invoke_into('cast', [src_h], [x._h], {})
         else:
             def feeder():

@@ -591,6 +591,17 @@ PYBIND11_MODULE(_imageio, m) {
            py::arg("seed") = 0)
       .def_property_readonly("size", &ImageRecordIter::size)
       .def("reset", &ImageRecordIter::Reset)
+      .def("next_into",
+           [](ImageRecordIter& it, uintptr_t data_ptr, uintptr_t label_ptr) {
+             // caller-provided buffers (e.g. pinned host memory from the
+             // native runtime) — zero extra host copies
+             int got;
+             {
+               py::gil_scoped_release rel;
+               got = it.NextInto((uint8_t*)data_ptr, (float*)label_ptr);
+             }
+             return got;
+           })
       .def("next_batch", [](ImageRecordIter& it, int batch, int oh, int ow) {
         std::vector<py::ssize_t> shp = {batch, oh, ow, 3};
         py::array_t<uint8_t> data(shp);
