@@ -359,10 +359,11 @@ class Trainer:
                     if g is None:
                         continue
                     if g.is_native:
-                        from .. import _core
-                        _core.invoke_into('_full', [], [g._h], {'value': '0'})
-                    else:
-                        to_zero.append(g._t)
+                        # native backward streams write-req leaf grads
+                        # in-place (first contribution overwrites):
+                        # zeroing here would be a dead fill per param
+                        continue
+                    to_zero.append(g._t)
         if to_zero:
             # one fused launch instead of a fill per parameter
             # (161 x ~3.7 us/step measured in the final profile)

@@ -221,7 +221,10 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
                           bool retain_graph) {
   OpEntry* add_into = OpRegistry::Get()->Find("_grad_add");
   OpEntry* ones_op = OpRegistry::Get()->Find("ones_like");
-  MX_CHECK(add_into && ones_op, "core grad ops missing");
+  OpEntry* copy_into_op = OpRegistry::Get()->Find("_copy_into");
+  MX_CHECK(add_into && ones_op && copy_into_op, "core grad ops missing");
+  // leaves whose attached grad buffer already received this sweep's value
+  std::unordered_set<NDArray::Chunk*> leaf_written;
 
   // forward pass over the tape: which chunks require grad at all
   // (leaves plus anything computed from them) — reference OpReqType
@@ -272,12 +275,10 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
       any_need |= node.need_igrad[i] != 0;
     }
     if (!any_need) continue;
-    // missing head grads are zeros
-    for (size_t i = 0; i < ograds.size(); ++i)
-      if (ograds[i].is_none()) {
-        OpEntry* zeros = OpRegistry::Get()->Find("zeros_like");
-        ograds[i] = Run(zeros, {}, {node.outputs[i]})[0];
-      }
+    // missing head grads stay NONE (= zero): materializing a zeros
+    // array per unused output (BN saved stats/mask, conv stats, pool
+    // argmax) cost ~220 fills per ResNet step; fbackward impls only
+    // read the grads that exist
     std::vector<NDArray> igrads = node.op->fbackward(node, ograds);
     MX_CHECK(igrads.size() == node.inputs.size(),
              "op '" << node.op->name << "' backward returned "
@@ -286,6 +287,18 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
     for (size_t i = 0; i < igrads.size(); ++i) {
       if (igrads[i].is_none() || !node.need_igrad[i]) continue;
       NDArray::Chunk* key = node.inputs[i].chunk_.get();
+      auto lit = leaves_.find(key);
+      if (lit != leaves_.end()) {
+        // leaf: stream contributions straight into the attached buffer
+        // (write req overwrites on the first one; add req accumulates)
+        bool first = leaf_written.insert(key).second;
+        if (first && lit->second.req != 2) {
+          RunInto(copy_into_op, {}, {igrads[i]}, {lit->second.grad});
+        } else {
+          RunInto(add_into, {}, {igrads[i]}, {lit->second.grad});
+        }
+        continue;
+      }
       auto git = grads.find(key);
       if (git == grads.end()) {
         grads[key] = igrads[i];
