@@ -12,6 +12,7 @@ CXXFLAGS := -fPIC -O3 -std=c++17 --offload-arch=$(ARCH) \
             -Wno-unused-result -parallel-jobs=4
 
 CORE_SRCS := src/core/storage.cc src/core/engine.cc src/core/ndarray.cc \
+             src/core/c_api.cc \
              src/core/op.cc src/core/rccl_comm.cc
 OPS_SRCS  := $(wildcard src/ops/*.hip)
 PYBIND_SRC := src/core/pybind.cc
