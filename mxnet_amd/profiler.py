@@ -20,6 +20,13 @@ def set_config(**kwargs):
 
 
 def set_state(state='stop', profile_process='worker'):
+    # native-runtime engine profiler rides along (per-op aggregate stats,
+    # reference src/profiler AggregateStats)
+    try:
+        from . import _core
+        _core.profiler_set_state(state == 'run')
+    except Exception:
+        pass
     if state == 'run' and not _state['running']:
         activities = [torch.profiler.ProfilerActivity.CPU]
         if torch.cuda.is_available():
@@ -58,6 +65,17 @@ def dump(finished=True, profile_process='worker'):
         with open(fname, 'w') as f:
             json.dump({'traceEvents': []}, f)
     return fname
+
+
+def native_summary(top=30):
+    """Aggregate per-op table from the native engine profiler
+    (op name, calls, total ms) — reference profiler aggregate stats."""
+    from . import _core
+    rows = _core.profiler_summary()
+    lines = ['%-32s %8s %12s' % ('op', 'calls', 'total_ms')]
+    for name, calls, ms in rows[:top]:
+        lines.append('%-32s %8d %12.3f' % (name or '<unnamed>', calls, ms))
+    return '\n'.join(lines)
 
 
 def dumps(reset=False):

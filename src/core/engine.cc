@@ -1,6 +1,9 @@
 #include "engine.h"
 
+#include <algorithm>
 #include <atomic>
+#include <chrono>
+#include <unordered_map>
 #include <cassert>
 #include <future>
 
@@ -113,6 +116,14 @@ struct WorkQueue {
   }
 };
 
+// profiler record: timed event pair (GPU) or wall ns (CPU)
+struct ProfRec {
+  std::string name;
+  hipEvent_t ev0 = nullptr, ev1 = nullptr;
+  int dev = 0;
+  double cpu_ms = 0;
+};
+
 struct DeviceWorkers {
   hipStream_t compute = nullptr, copy = nullptr, comm = nullptr;
   WorkQueue compute_q, copy_q, comm_q;
@@ -131,6 +142,10 @@ struct Engine::Impl {
 
   WorkQueue cpu_q_, cpu_prio_q_;
   std::vector<std::thread> cpu_workers_;
+
+  std::atomic<bool> profiling_{false};
+  std::mutex prof_mu_;
+  std::vector<ProfRec> prof_;
 
   std::mutex dev_mu_;
   std::unordered_map<int, std::unique_ptr<DeviceWorkers>> devices_;
@@ -253,6 +268,15 @@ struct Engine::Impl {
       if (opr == nullptr) return;
       std::exception_ptr exc;
       bool capturing = dw && dw->capturing.load(std::memory_order_relaxed);
+      bool prof = profiling_.load(std::memory_order_relaxed) && !capturing;
+      ProfRec rec;
+      auto prof_t0 = std::chrono::steady_clock::now();
+      if (prof && rc.ctx.is_gpu()) {
+        rec.dev = rc.ctx.dev_id;
+        MX_HIP_CALL(hipEventCreate(&rec.ev0));
+        MX_HIP_CALL(hipEventCreate(&rec.ev1));
+        (void)hipEventRecord(rec.ev0, rc.stream);
+      }
       try {
         // order execution behind producers on other streams
         for (auto& w : opr->waits) {
@@ -266,6 +290,18 @@ struct Engine::Impl {
         if (opr->fn) opr->fn(rc);
       } catch (...) {
         exc = std::current_exception();
+      }
+      if (prof) {
+        rec.name = opr->name;
+        if (rc.ctx.is_gpu()) {
+          (void)hipEventRecord(rec.ev1, rc.stream);
+        } else {
+          rec.cpu_ms = std::chrono::duration<double, std::milli>(
+                           std::chrono::steady_clock::now() - prof_t0)
+                           .count();
+        }
+        std::lock_guard<std::mutex> g(prof_mu_);
+        prof_.push_back(std::move(rec));
       }
       EventPtr done;
       // re-read AFTER fn: the BeginCapture op flips capturing on inside
@@ -519,6 +555,41 @@ void Engine::LaunchGraph(int dev, uintptr_t exec,
         MX_HIP_CALL(hipGraphLaunch((hipGraphExec_t)exec, rc.stream));
       },
       Context::GPU(dev), after, {}, FnProperty::kNormal, "GraphLaunch");
+}
+
+void Engine::SetProfiling(bool on) {
+  impl_->profiling_.store(on);
+}
+
+std::vector<std::tuple<std::string, long, double>> Engine::ProfilerSummary() {
+  WaitForAll();
+  std::vector<ProfRec> recs;
+  {
+    std::lock_guard<std::mutex> g(impl_->prof_mu_);
+    recs.swap(impl_->prof_);
+  }
+  std::unordered_map<std::string, std::pair<long, double>> agg;
+  for (auto& r : recs) {
+    double ms = r.cpu_ms;
+    if (r.ev0) {
+      float f = 0;
+      (void)hipEventSynchronize(r.ev1);
+      (void)hipEventElapsedTime(&f, r.ev0, r.ev1);
+      ms = f;
+      (void)hipEventDestroy(r.ev0);
+      (void)hipEventDestroy(r.ev1);
+    }
+    auto& a = agg[r.name];
+    a.first += 1;
+    a.second += ms;
+  }
+  std::vector<std::tuple<std::string, long, double>> out;
+  for (auto& kv : agg)
+    out.emplace_back(kv.first, kv.second.first, kv.second.second);
+  std::sort(out.begin(), out.end(), [](auto& a, auto& b) {
+    return std::get<2>(a) > std::get<2>(b);
+  });
+  return out;
 }
 
 void Engine::StopWorkers() {

@@ -20,6 +20,7 @@
 #pragma once
 
 #include <condition_variable>
+#include <tuple>
 #include <deque>
 #include <functional>
 #include <memory>
@@ -88,6 +89,11 @@ class Engine {
   // optional read-deps order the launch behind e.g. comm-stream work
   void LaunchGraph(int dev, uintptr_t exec,
                    const std::vector<VarId>& after = {});
+
+  // ---- profiler (reference src/profiler: per-op aggregate stats) ----
+  void SetProfiling(bool on);
+  // name -> (calls, gpu_ms or cpu_ms); drains and resets the buffer
+  std::vector<std::tuple<std::string, long, double>> ProfilerSummary();
 
   void StopWorkers();  // tests / atfork
 

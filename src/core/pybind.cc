@@ -255,6 +255,15 @@ PYBIND11_MODULE(_core, m) {
   m.def("clear_tape", [] { Imperative::Get()->ClearTape(); });
   m.def("tape_size", [] { return Imperative::Get()->TapeSize(); });
 
+  // profiler (aggregate per-op stats, reference AggregateStats)
+  m.def("profiler_set_state", [](bool on) {
+    Engine::Get()->SetProfiling(on);
+  });
+  m.def("profiler_summary", [] {
+    py::gil_scoped_release rel;
+    return Engine::Get()->ProfilerSummary();
+  });
+
   // storage telemetry
   m.def("pool_size", [](int dev_type, int dev_id) {
     return Storage::Get()->PoolSize(MakeCtx(dev_type, dev_id));

@@ -246,3 +246,21 @@ def test_stochastic_block_estimator_kl():
     assert len(net.losses) == 1
     # KL term received gradient: encoder params moved
     assert net.enc_ls.weight.grad() is not None
+
+
+def test_native_engine_profiler():
+    """Per-op aggregate stats from the native engine (reference
+    src/profiler AggregateStats table)."""
+    import numpy as np
+    from mxnet_amd import _core, profiler
+    profiler.set_state('run')
+    a = _core.from_numpy(np.ones((64, 64), dtype='float32'))
+    for _ in range(3):
+        a = _core.invoke('elemwise_mul', [a, a], {})[0]
+    _core.invoke('sum', [a], {})[0].asnumpy()
+    rows = dict((n, (c, ms)) for n, c, ms in _core.profiler_summary())
+    profiler.set_state('stop')
+    assert rows['elemwise_mul'][0] == 3
+    assert rows['sum'][0] == 1
+    table = profiler.native_summary()
+    assert 'op' in table and 'calls' in table
