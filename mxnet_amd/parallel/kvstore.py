@@ -59,6 +59,8 @@ class KVStoreBase:
 
 def create(name='local'):
     from ..base import native_mode
+    if isinstance(name, str) and name == 'dist_sync_ps':
+        return SyncPSKVStore(name)
     if native_mode() and isinstance(name, str) and (
             name.startswith('dist') or name in ('nccl', 'device')):
         return NativeDistKVStore(name)
@@ -359,20 +361,29 @@ class AsyncPSKVStore(KVStoreBase):
     """
 
     _OP_INIT, _OP_PUSH, _OP_PULL, _OP_STOP = 0, 1, 2, 3
+    _sync = False
 
     def __init__(self, kind='dist_async'):
+        import os as _os
         self._type = kind
         if not dist.is_initialized():
             dist.init_process_group(backend='gloo')
         self._world = dist.get_world_size()
-        assert self._world >= 2, 'dist_async needs >= 2 ranks (+1 server)'
-        self._server = self._world - 1
+        # key sharding across S servers (reference EncodeDefaultKey
+        # distributes keys over ps-lite servers); the LAST S ranks serve
+        self._nservers = int(_os.environ.get('MXNET_PS_NSERVERS', '1'))
+        assert self._world >= 1 + self._nservers,             'dist PS needs >= 1 worker + MXNET_PS_NSERVERS servers'
         self._rank = dist.get_rank()
         self._optimizer = None
         self._store = {}        # server side: key -> fp32 tensor
         self._states = {}       # server side optimizer states
+        self._pending = {}      # sync mode: key -> (accum, count)
+        self._waiting = {}      # sync mode: key -> queued pull srcs
         self._workers_group = dist.new_group(
-            ranks=list(range(self._world - 1)))
+            ranks=list(range(self._world - self._nservers)))
+
+    def _server_of(self, key):
+        return self._world - 1 - (key % self._nservers)
 
     @property
     def rank(self):
@@ -380,11 +391,11 @@ class AsyncPSKVStore(KVStoreBase):
 
     @property
     def num_workers(self):
-        return self._world - 1
+        return self._world - self._nservers
 
     @property
     def is_server(self):
-        return self._rank == self._server
+        return self._rank >= self._world - self._nservers
 
     def set_optimizer(self, optimizer):
         """Server-side updater (reference sync/async server optimizer,
@@ -392,9 +403,9 @@ class AsyncPSKVStore(KVStoreBase):
         self._optimizer = optimizer
 
     # -- worker protocol -------------------------------------------------
-    def _send_header(self, op, key, numel):
+    def _send_header(self, op, key, numel, dst=None):
         h = torch.tensor([op, key, numel, 0], dtype=torch.long)
-        dist.send(h, dst=self._server)
+        dist.send(h, dst=self._server_of(key) if dst is None else dst)
 
     def init(self, key, value):
         v = value[0] if isinstance(value, (list, tuple)) else value
@@ -404,21 +415,21 @@ class AsyncPSKVStore(KVStoreBase):
             v._t.copy_(t.to(v._t.device, v._t.dtype))
         if self._rank == 0:
             self._send_header(self._OP_INIT, key, t.numel())
-            dist.send(t, dst=self._server)
+            dist.send(t, dst=self._server_of(key))
         dist.barrier(group=self._workers_group)
 
     def push(self, key, value, priority=0):
         v = value[0] if isinstance(value, (list, tuple)) else value
         g = v._t.detach().float().cpu().contiguous()
         self._send_header(self._OP_PUSH, key, g.numel())
-        dist.send(g, dst=self._server)
+        dist.send(g, dst=self._server_of(key))
 
     def pull(self, key, out=None, priority=0, ignore_sparse=True):
         outs = out if isinstance(out, (list, tuple)) else [out]
         n = outs[0]._t.numel()
         self._send_header(self._OP_PULL, key, n)
         buf = torch.empty(n, dtype=torch.float32)
-        dist.recv(buf, src=self._server)
+        dist.recv(buf, src=self._server_of(key))
         with torch.no_grad():
             for o in outs:
                 o._t.copy_(buf.view(o._t.shape).to(o._t.device, o._t.dtype))
@@ -429,10 +440,24 @@ class AsyncPSKVStore(KVStoreBase):
             self.pull(key, out, priority)
 
     def stop(self):
-        self._send_header(self._OP_STOP, 0, 0)
+        for s in range(self._nservers):
+            self._send_header(self._OP_STOP, 0, 0,
+                              dst=self._world - 1 - s)
 
     def barrier_workers(self):
         dist.barrier(group=self._workers_group)
+
+    def _apply(self, key, g):
+        w = self._store[key]
+        if self._optimizer is not None:
+            if key not in self._states:
+                self._states[key] = \
+                    self._optimizer.create_state_multi_precision(
+                        key, NDArray(w))
+            self._optimizer.update_multi_precision(
+                key, NDArray(w), NDArray(g), self._states[key])
+        else:
+            w.sub_(g)  # plain accumulate (reference default)
 
     # -- server loop -----------------------------------------------------
     def run_server(self):
@@ -454,15 +479,38 @@ class AsyncPSKVStore(KVStoreBase):
             elif op == self._OP_PUSH:
                 g = torch.empty(numel, dtype=torch.float32)
                 dist.recv(g, src=src)
-                w = self._store[key]
-                if self._optimizer is not None:
-                    if key not in self._states:
-                        self._states[key] = \
-                            self._optimizer.create_state_multi_precision(
-                                key, NDArray(w))
-                    self._optimizer.update_multi_precision(
-                        key, NDArray(w), NDArray(g), self._states[key])
+                if self._sync:
+                    # sync mode (reference kvstore_dist_server.h:346-365
+                    # ApplyUpdates): aggregate NumWorkers pushes into
+                    # `merged`, run the optimizer ONCE, then answer the
+                    # pulls that queued while waiting
+                    acc, cnt = self._pending.get(key, (None, 0))
+                    acc = g if acc is None else acc.add_(g)
+                    cnt += 1
+                    if cnt == self.num_workers:
+                        self._apply(key, acc)
+                        self._pending[key] = (None, 0)
+                        for wsrc in self._waiting.pop(key, []):
+                            dist.send(self._store[key], dst=wsrc)
+                    else:
+                        self._pending[key] = (acc, cnt)
                 else:
-                    w.sub_(g)  # plain accumulate (reference default)
+                    self._apply(key, g)
             elif op == self._OP_PULL:
-                dist.send(self._store[key], dst=src)
+                if self._sync and self._pending.get(key, (None, 0))[1] > 0:
+                    # update still aggregating: defer the reply
+                    self._waiting.setdefault(key, []).append(src)
+                else:
+                    dist.send(self._store[key], dst=src)
+
+
+class SyncPSKVStore(AsyncPSKVStore):
+    """Synchronous parameter server (reference dist_sync over ps-lite):
+    the server aggregates one push per worker per key, applies the
+    (server-side) optimizer once, and releases the queued pulls — a
+    per-key global barrier.  Key sharding across MXNET_PS_NSERVERS
+    server ranks as in the async store."""
+    _sync = True
+
+    def __init__(self, kind='dist_sync_ps'):
+        super().__init__(kind)

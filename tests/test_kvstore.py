@@ -190,3 +190,75 @@ def test_dist_async_parameter_server(tmp_path):
     g_total = np.full((3, 5), 2 * 0.1 * 1) + np.full((3, 5), 2 * 0.1 * 2)
     np.testing.assert_allclose(w0, init_w - 0.5 * g_total, rtol=1e-4,
                                atol=1e-5)
+
+
+_SYNC_PS_WORKER = '''
+import os
+import numpy as np
+import torch
+from mxnet_amd.ndarray.ndarray import NDArray
+from mxnet_amd.parallel import kvstore as kvs
+
+rank = int(os.environ['RANK'])
+kv = kvs.create('dist_sync_ps')
+if kv.is_server:
+    from mxnet_amd import optimizer as opt
+    kv.set_optimizer(opt.create('sgd', learning_rate=0.5))
+    kv.run_server()
+else:
+    # two keys -> sharded across the two server ranks
+    w0 = NDArray(torch.full((4,), 10.0))
+    w1 = NDArray(torch.full((6,), 20.0))
+    kv.init(0, w0)
+    kv.init(1, w1)
+    g0 = NDArray(torch.full((4,), float(rank + 1)))
+    g1 = NDArray(torch.full((6,), 2.0 * (rank + 1)))
+    kv.push(0, g0)
+    kv.push(1, g1)
+    kv.pull(0, w0)   # blocks until BOTH workers pushed (sync barrier)
+    kv.pull(1, w1)
+    np.save(os.environ['OUT_PREFIX'] + f'_s{rank}.npy',
+            np.concatenate([w0.asnumpy(), w1.asnumpy()]))
+    kv.barrier_workers()
+    kv.stop()   # every worker tells every server (live-count protocol)
+print('SYNC_PS_OK', rank)
+'''
+
+
+def test_dist_sync_parameter_server(tmp_path):
+    """dist_sync_ps: 2 workers + 2 servers (key-sharded).  The server
+    aggregates one push per worker per key, applies SGD ONCE on the
+    merged gradient, and only then releases the queued pulls — per-key
+    global barrier semantics (reference ApplyUpdates,
+    kvstore_dist_server.h:346-365)."""
+    import subprocess, sys, socket
+    script = tmp_path / 'sync_ps.py'
+    script.write_text(_SYNC_PS_WORKER)
+    repo_root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    sock = socket.socket(); sock.bind(('127.0.0.1', 0))
+    port = str(sock.getsockname()[1]); sock.close()
+    env = dict(os.environ)
+    env.update({'MASTER_ADDR': '127.0.0.1', 'MASTER_PORT': port,
+                'MXNET_PS_NSERVERS': '2',
+                'OUT_PREFIX': str(tmp_path / 'ps'),
+                'PYTHONPATH': repo_root + os.pathsep +
+                env.get('PYTHONPATH', '')})
+    procs = []
+    for rank in range(4):
+        e = dict(env, RANK=str(rank), WORLD_SIZE='4')
+        procs.append(subprocess.Popen([sys.executable, str(script)], env=e,
+                                      stdout=subprocess.PIPE,
+                                      stderr=subprocess.STDOUT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        outs.append(out.decode())
+        assert p.returncode == 0, out.decode()
+    import numpy as np
+    s0 = np.load(tmp_path / 'ps_s0.npy')
+    s1 = np.load(tmp_path / 'ps_s1.npy')
+    np.testing.assert_allclose(s0, s1, rtol=1e-6)
+    # merged g0 = (1+2) = 3 -> w0' = 10 - 0.5*3 = 8.5
+    # merged g1 = (2+4) = 6 -> w1' = 20 - 0.5*6 = 17
+    np.testing.assert_allclose(s0[:4], 8.5, rtol=1e-6)
+    np.testing.assert_allclose(s0[4:], 17.0, rtol=1e-6)
