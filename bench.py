@@ -64,7 +64,17 @@ def main():
     # supplies the host process (no torch tensors, no torch.autograd)
     native = args.native or args.runtime == 'native' or \
         os.environ.get('MXNET_NATIVE_RUNTIME', '0') == '1'
-    # native+distributed runs the own RCCL binding (NativeDistKVStore)
+    # native+distributed runs the own RCCL binding (NativeDistKVStore);
+    # pre-flight the communicator before building the model so a
+    # bootstrap failure degrades to the torch frontend instead of dying
+    if native and distributed and on_gpu:
+        try:
+            from mxnet_amd import _core as _c0
+            _c0.rccl_init(world, rank, local_rank)
+        except Exception as e:
+            print(f'# native RCCL preflight failed ({e}); '
+                  f'falling back to the torch frontend', file=sys.stderr)
+            native = False
 
     import mxnet_amd as mx
     if native:
