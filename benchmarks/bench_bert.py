@@ -3,6 +3,8 @@
 
 samples/sec whole job; seq=128 synthetic tokens, random-init weights.
 Launch multi-GPU exactly like bench.py (torch.distributed.run, RCCL).
+Default runtime is the native C++ engine (`--runtime torch` selects the
+torch-tensor frontend over the same gfx950 kernels).
 """
 import argparse
 import json
@@ -12,6 +14,7 @@ import time
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+import numpy as np
 import torch
 
 
@@ -22,6 +25,8 @@ def main():
     p.add_argument('--batch-size', type=int, default=32)
     p.add_argument('--seq-len', type=int, default=128)
     p.add_argument('--dtype', default='float16')
+    p.add_argument('--runtime', choices=['native', 'torch'],
+                   default='native')
     args = p.parse_args()
 
     world = int(os.environ.get('WORLD_SIZE', 1))
@@ -33,12 +38,15 @@ def main():
     else:
         args.batch_size, args.seq_len = 2, 32
 
+    native = args.runtime == 'native'
+    os.environ['MXNET_NATIVE_RUNTIME'] = '1' if native else '0'
+
     import mxnet_amd as mx
     from mxnet_amd import autograd
     from mxnet_amd.gluon import Trainer
     from mxnet_amd.ndarray.ndarray import NDArray
     from mxnet_amd.models.bert import bert_base, BERTModel
-    from mxnet_amd.ops import nn as F
+    from mxnet_amd.ndarray import ops as F
 
     ctx = mx.gpu(local_rank) if on_gpu else mx.cpu()
     dtype = args.dtype if on_gpu else 'float32'
@@ -49,7 +57,7 @@ def main():
                         num_layers=2, num_heads=4)
     net.initialize(ctx=ctx)
     net.cast(dtype)
-    # embeddings + LN params stay functional in fp16 (master weights in opt)
+    # LN/embedding masters come from the optimizer's multi_precision
     trainer = Trainer(net.collect_params(), 'adam',
                       {'learning_rate': 1e-4, 'multi_precision': True},
                       kvstore='dist_device_sync' if world > 1 else None)
@@ -57,20 +65,46 @@ def main():
     B, S = args.batch_size, args.seq_len
     dev = torch.device('cuda', local_rank) if on_gpu else torch.device('cpu')
     vocab = 30522 if on_gpu else 1000
+    rng = np.random.RandomState(1 + rank)
     torch.manual_seed(1 + rank)
-    tokens = mx.nd.from_torch(torch.randint(0, vocab, (B, S), device=dev))
-    types = mx.nd.from_torch(torch.zeros(B, S, dtype=torch.long, device=dev))
-    mask = mx.nd.from_torch(torch.ones(B, S, dtype=torch.bool, device=dev))
-    mlm_label = torch.randint(0, vocab, (B, S), device=dev)
-    nsp_label = torch.randint(0, 2, (B,), device=dev)
+
+    if native:
+        tokens = mx.nd.array(rng.randint(0, vocab, (B, S)), ctx=ctx,
+                             dtype='int64')
+        types = mx.nd.array(np.zeros((B, S)), ctx=ctx, dtype='int64')
+        mask = mx.nd.array(np.ones((B, S)), ctx=ctx, dtype=dtype)
+        mlm_label = mx.nd.array(rng.randint(0, vocab, (B * S,)), ctx=ctx,
+                                dtype='float32')
+        nsp_label = mx.nd.array(rng.randint(0, 2, (B,)), ctx=ctx,
+                                dtype='float32')
+
+        def loss_fn(mlm, nsp):
+            lp1 = F.log_softmax(mlm.reshape(-1, vocab))
+            l1 = F.pick(lp1, mlm_label, axis=-1).mean() * -1.0
+            lp2 = F.log_softmax(nsp)
+            l2 = F.pick(lp2, nsp_label, axis=-1).mean() * -1.0
+            return l1 + l2
+    else:
+        from mxnet_amd.ops import nn as Fnn
+        tokens = mx.nd.from_torch(
+            torch.randint(0, vocab, (B, S), device=dev))
+        types = mx.nd.from_torch(
+            torch.zeros(B, S, dtype=torch.long, device=dev))
+        mask = mx.nd.from_torch(torch.ones(B, S, dtype=torch.bool,
+                                           device=dev))
+        mlm_t = torch.randint(0, vocab, (B * S,), device=dev)
+        nsp_t = torch.randint(0, 2, (B,), device=dev)
+
+        def loss_fn(mlm, nsp):
+            l1 = Fnn.softmax_cross_entropy(
+                mlm.handle.reshape(-1, vocab), mlm_t).mean()
+            l2 = Fnn.softmax_cross_entropy(nsp.handle, nsp_t).mean()
+            return NDArray((l1 + l2).float())
 
     def step():
         with autograd.record():
             _, _, mlm, nsp = net(tokens, types, mask)
-            l1 = F.softmax_cross_entropy(
-                mlm.handle.reshape(-1, vocab), mlm_label.reshape(-1)).mean()
-            l2 = F.softmax_cross_entropy(nsp.handle, nsp_label).mean()
-            L = NDArray((l1 + l2).float())
+            L = loss_fn(mlm, nsp)
         L.backward()
         trainer.step(B)
 
@@ -79,10 +113,43 @@ def main():
     for _ in range(args.warmup):
         step()
 
-    # whole-step hipGraph capture (same pattern as bench.py)
+    # whole-step hipGraph capture (same two-graph scheme as bench.py:
+    # G1 fwd+bwd on the engine compute stream, eager RCCL between, G2
+    # fused Adam updates with read-deps on the grad vars)
     graph = None
-    if on_gpu and world == 1 and \
-            os.environ.get('MXNET_BENCH_HIPGRAPH', '1') != '0':
+    want_graph = os.environ.get('MXNET_BENCH_HIPGRAPH', '1') != '0'
+    if on_gpu and want_graph and native:
+        from mxnet_amd import _core
+        try:
+            mx.nd.waitall()
+            _core.begin_capture(local_rank)
+            with autograd.record():
+                _, _, mlm, nsp = net(tokens, types, mask)
+                L = loss_fn(mlm, nsp)
+            L.backward()
+            g1 = _core.end_capture(local_rank)
+            mx.nd.waitall()
+            _core.begin_capture(local_rank)
+            trainer._optimizer.rescale_grad = 1.0 / B
+            trainer._update(False)
+            g2 = _core.end_capture(local_rank)
+            mx.nd.waitall()
+            grad_handles = [p.list_grad()[0]._h for p in trainer._params]
+
+            def replay_step():
+                _core.launch_graph(local_rank, g1)
+                if world > 1:
+                    trainer._allreduce_grads()
+                _core.launch_graph(local_rank, g2,
+                                   grad_handles if world > 1 else [])
+            replay_step()
+            mx.nd.waitall()
+            graph = ('native', replay_step)
+        except Exception as e:
+            print(f'# native hipgraph capture unavailable: {e}',
+                  file=sys.stderr)
+            graph = None
+    elif on_gpu and want_graph and world == 1:
         try:
             torch.cuda.synchronize()
             side = torch.cuda.Stream()
@@ -98,16 +165,25 @@ def main():
             torch.cuda.synchronize()
             graph = g
         except Exception as e:
-            print('# hipgraph capture unavailable:', e)
+            print('# hipgraph capture unavailable:', e, file=sys.stderr)
             graph = None
 
+    if native:
+        mx.nd.waitall()
     if on_gpu:
         torch.cuda.synchronize()
     if world > 1:
         dist.barrier()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        graph.replay() if graph is not None else step()
+        if graph is None:
+            step()
+        elif isinstance(graph, tuple):
+            graph[1]()
+        else:
+            graph.replay()
+    if native:
+        mx.nd.waitall()
     if on_gpu:
         torch.cuda.synchronize()
     dt = time.perf_counter() - t0
@@ -122,7 +198,7 @@ def main():
             'unit': 'samples/sec', 'n_gpus': world, 'steps': args.steps,
             'warmup': args.warmup, 'ms_per_step': round(dt / args.steps * 1e3, 3),
             'higher_is_better': True, 'scaling': 'weak', 'vs_baseline': None,
-            'dtype': dtype, 'data': 'synthetic',
+            'dtype': dtype, 'data': 'synthetic', 'runtime': args.runtime,
             'config': {'model': 'bert_base', 'global_batch': B * world,
                        'seq_len': S, 'parallelism': f'dp{world}'}}))
 

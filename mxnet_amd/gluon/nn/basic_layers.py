@@ -263,6 +263,14 @@ class LayerNorm(HybridBlock):
     def hybrid_forward(self, F, x, gamma, beta):
         return F.LayerNorm(x, gamma, beta, axis=self._axis, eps=self._epsilon)
 
+    def cast(self, dtype):
+        """LN affine params stay fp32 under half-precision training (same
+        fp32-list treatment as BatchNorm; the fused LN kernel computes
+        stats in fp32 and expects fp32 gamma/beta)."""
+        if dtype in ('float16', 'bfloat16'):
+            return self
+        return super().cast(dtype)
+
 
 class GroupNorm(HybridBlock):
     def __init__(self, num_groups=1, epsilon=1e-5, center=True, scale=True,

@@ -32,8 +32,46 @@ class BERTSelfAttention(HybridBlock):
                              weight_initializer=init.Normal(0.02))
         self.dropout = nn.Dropout(dropout) if dropout else None
 
+    def _forward_native(self, x, addmask):
+        """Native-runtime attention.  ``addmask`` is the model-level
+        additive float mask [B*H, S, S] (0 valid / -1e4 masked) or None.
+        Fused `interleaved_attention` when un-masked, no-dropout and
+        MFMA-aligned; otherwise composed from native registry ops."""
+        import math as _m
+        from ..ndarray import ops as F
+        B, S, U = x.shape
+        H, D = self._num_heads, self._head_dim
+        qkv = self.qkv(x)
+        on_gpu = x.context.device_type == 'gpu'
+        if on_gpu and self.dropout is None and addmask is None \
+                and D % 8 == 0 and S % 8 == 0:
+            out = qkv._invoke('interleaved_attention', [qkv],
+                              {'heads': str(H),
+                               'temperature': str(_m.sqrt(D))})
+            return self.proj(out)
+        q = qkv[:, :, 0:U]
+        k = qkv[:, :, U:2 * U]
+        v = qkv[:, :, 2 * U:3 * U]
+
+        def heads(z):
+            return z.reshape(B, S, H, D).transpose((0, 2, 1, 3)) \
+                    .reshape(B * H, S, D)
+        q, k, v = heads(q), heads(k), heads(v)
+        scores = q._invoke('batch_dot', [q, k.transpose((0, 2, 1))])
+        if addmask is not None:
+            scores = scores + addmask
+        att = F.softmax(scores, axis=-1, temperature=_m.sqrt(D))
+        if self.dropout is not None:
+            att = self.dropout(att)
+        out = att._invoke('batch_dot', [att, v])
+        out = out.reshape(B, H, S, D).transpose((0, 2, 1, 3)) \
+                 .reshape(B, S, U)
+        return self.proj(out)
+
     def forward(self, x, mask=None):
         # x: [B, S, U]; mask: [B, S] valid-token or prebuilt [B*H, S, S]
+        if getattr(x, 'is_native', False):
+            return self._forward_native(x, mask)
         from ..ops import nn as F
         from ..ops.dispatch import use_hip
         import math as _m
@@ -172,8 +210,44 @@ class BERTModel(HybridBlock):
         self.nsp_classifier = nn.Dense(2, flatten=False,
                                        weight_initializer=init.Normal(0.02))
 
+    def _forward_native(self, tokens, token_types, valid_mask):
+        """Native-runtime forward: positional slice / pooler token pick go
+        through the recorded `_strided_copy` op (shape-correct backward
+        into the leaves); the [B,S] valid mask becomes ONE additive float
+        mask [B*H,S,S] shared by every layer."""
+        S = tokens.shape[1]
+        emb = self.word_embed(tokens)
+        if token_types is not None:
+            emb = emb + self.token_type_embed(token_types)
+        pos = self.position_embed.data(emb.context)
+        p = pos[:S].expand_dims(0)
+        if str(p.dtype) != str(emb.dtype):
+            p = p.astype(emb.dtype)
+        emb = emb + p
+        emb = self.embed_ln(emb)
+        if self.embed_dropout is not None:
+            emb = self.embed_dropout(emb)
+        addmask = None
+        if valid_mask is not None:
+            B, Sm = valid_mask.shape
+            H = self._num_heads
+            m = valid_mask
+            if str(m.dtype) != str(emb.dtype):
+                m = m.astype(emb.dtype)
+            add = (m - 1.0) * 10000.0  # 0 valid / -1e4 masked (additive)
+            addmask = add.reshape(B, 1, 1, Sm) \
+                         .broadcast_to((B, H, Sm, Sm)) \
+                         .reshape(B * H, Sm, Sm)
+        seq = self.encoder(emb, addmask)
+        pooled = self.pooler(seq[:, 0])
+        mlm = self.mlm_decoder(self.mlm_ln(self.mlm_dense(seq)))
+        nsp = self.nsp_classifier(pooled)
+        return seq, pooled, mlm, nsp
+
     def forward(self, tokens, token_types=None, valid_mask=None):
         from ..ndarray.ndarray import NDArray
+        if getattr(tokens, 'is_native', False):
+            return self._forward_native(tokens, token_types, valid_mask)
         t = tokens.handle if hasattr(tokens, 'handle') else tokens
         S = t.shape[1]
         emb = self.word_embed(tokens)

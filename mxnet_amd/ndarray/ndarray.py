@@ -280,9 +280,21 @@ class NDArray:
         return self.reshape(self.shape[0], -1)
 
     def expand_dims(self, axis):
+        if self._h is not None:
+            shp = list(self.shape)
+            if axis < 0:
+                axis += len(shp) + 1
+            shp.insert(axis, 1)
+            return self.reshape(shp)
         return NDArray(self._t.unsqueeze(axis))
 
     def squeeze(self, axis=None):
+        if self._h is not None:
+            shp = [d for i, d in enumerate(self.shape)
+                   if not (d == 1 and (axis is None or i == axis
+                                       or (isinstance(axis, (list, tuple))
+                                           and i in axis)))]
+            return self.reshape(shp or [1])
         return NDArray(self._t.squeeze() if axis is None else self._t.squeeze(axis))
 
     def transpose(self, axes=None):
@@ -295,9 +307,17 @@ class NDArray:
         return NDArray(self._t.permute(*axes).contiguous())
 
     def broadcast_to(self, shape):
+        if self._h is not None:
+            return self._invoke(
+                'broadcast_to', [self],
+                {'shape': '(' + ','.join(str(int(d)) for d in shape) + ',)'})
         return NDArray(self._t.broadcast_to(shape).contiguous())
 
     def swapaxes(self, a, b):
+        if self._h is not None:
+            axes = list(range(self.ndim))
+            axes[a], axes[b] = axes[b], axes[a]
+            return self.transpose(tuple(axes))
         return NDArray(self._t.transpose(a, b).contiguous())
 
     def split(self, num_outputs, axis=0):
@@ -313,7 +333,63 @@ class NDArray:
             return tuple(NDArray._unwrap_index(k) for k in key)
         return key
 
+    def _native_basic_index(self, key):
+        """Basic int/slice indexing on a native array as a recorded
+        `_strided_copy` op (gather fwd, scatter-into-zeros bwd) — views
+        would share the chunk without a tape edge (reference
+        ndarray.py:720 slicing semantics)."""
+        if not isinstance(key, tuple):
+            key = (key,)
+        shp = self.shape
+        # row-major element strides of the (contiguous) input
+        istr = [1] * len(shp)
+        for i in range(len(shp) - 2, -1, -1):
+            istr[i] = istr[i + 1] * shp[i + 1]
+        out_shape, out_strides, offset = [], [], 0
+        dim = 0
+        for k in key:
+            if k is Ellipsis:
+                skip = len(shp) - dim - sum(1 for kk in key
+                                            if kk is not Ellipsis)
+                dim += skip
+                for d in range(dim - skip, dim):
+                    out_shape.append(shp[d])
+                    out_strides.append(istr[d])
+                continue
+            if isinstance(k, (int, _np.integer)):
+                i = int(k)
+                if i < 0:
+                    i += shp[dim]
+                offset += i * istr[dim]
+            elif isinstance(k, slice):
+                start, stop, step = k.indices(shp[dim])
+                n = max(0, (stop - start + (step - (1 if step > 0 else -1)))
+                        // step)
+                out_shape.append(n)
+                out_strides.append(istr[dim] * step)
+                offset += start * istr[dim]
+            else:
+                return None  # fancy indexing: unsupported natively
+            dim += 1
+        for d in range(dim, len(shp)):
+            out_shape.append(shp[d])
+            out_strides.append(istr[d])
+        if not out_shape:
+            out_shape, out_strides = [1], [0]
+        return self._invoke(
+            '_strided_copy', [self],
+            {'shape': '(' + ','.join(map(str, out_shape)) + ',)',
+             'strides': '(' + ','.join(map(str, out_strides)) + ',)',
+             'offset': str(offset)})
+
     def __getitem__(self, key):
+        if self._h is not None:
+            out = self._native_basic_index(key)
+            if out is None:
+                raise TypeError(
+                    'native NDArray supports basic (int/slice) indexing '
+                    'only; got %r' % (key,))
+            return out
         out = self._t[NDArray._unwrap_index(key)]
         if not isinstance(out, torch.Tensor):
             out = torch.tensor(out)
@@ -333,7 +409,8 @@ class NDArray:
     def slice_axis(self, axis, begin, end):
         sl = [slice(None)] * self.ndim
         sl[axis] = slice(begin, end)
-        return NDArray(self._t[tuple(sl)])
+        return self[tuple(sl)] if self._h is not None \
+            else NDArray(self._t[tuple(sl)])
 
     # -- arithmetic --------------------------------------------------------
     @staticmethod

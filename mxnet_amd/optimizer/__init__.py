@@ -258,13 +258,44 @@ class Adam(Optimizer):
         self.beta1, self.beta2, self.epsilon = beta1, beta2, epsilon
 
     def create_state(self, index, weight):
+        if isinstance(weight, NDArray) and weight.is_native:
+            from ..ndarray.ndarray import zeros_like as _zl
+            f32 = weight if str(weight.dtype) == 'float32' \
+                else weight.astype('float32')
+            return (_zl(f32), _zl(f32))
         w = weight._t if isinstance(weight, NDArray) else weight
         return (torch.zeros_like(w, dtype=torch.float32),
                 torch.zeros_like(w, dtype=torch.float32))
 
     _adamw = False
 
+    def _native_update(self, index, weight, grad, state):
+        """Fused adam_update through the native registry (out layout:
+        [w, m, v, master?]; kernel src/ops/nn_reg.hip adam_update)."""
+        from .. import _core
+        self._update_count(index)
+        lr, wd = self._get_lr(index), self._get_wd(index)
+        t = self._index_update_count[index]
+        lr_t = lr * math.sqrt(1 - self.beta2 ** t) / (1 - self.beta1 ** t)
+        if isinstance(state, tuple) and len(state) == 2 and \
+                isinstance(state[1], tuple):
+            master, (m, v) = state
+        else:
+            master, (m, v) = None, state
+        outs = [weight._h, m._h, v._h]
+        if master is not None:
+            outs.append(master._h)
+        _core.invoke_into(
+            'adam_update', [grad._h], outs,
+            {'lr_t': str(lr_t), 'beta1': str(self.beta1),
+             'beta2': str(self.beta2), 'eps': str(self.epsilon),
+             'wd': str(wd), 'rescale_grad': str(self.rescale_grad),
+             'clip_gradient': str(self.clip_gradient or 0.0),
+             'adamw': '1' if self._adamw else '0'})
+
     def update(self, index, weight, grad, state):
+        if isinstance(weight, NDArray) and weight.is_native:
+            return self._native_update(index, weight, grad, state)
         self._update_count(index)
         lr, wd = self._get_lr(index), self._get_wd(index)
         t = self._index_update_count[index]
@@ -284,6 +315,8 @@ class Adam(Optimizer):
         step+master-weight cast in ONE launch — the eager torch
         composition was ~10 elementwise launches per parameter and
         DOMINATED the BERT step (profiles/r01_summary.md)."""
+        if isinstance(weight, NDArray) and weight.is_native:
+            return self._native_update(index, weight, grad, state)
         w = weight._t if isinstance(weight, NDArray) else weight
         g = grad._t if isinstance(grad, NDArray) else grad
         if w.is_cuda and g.dtype == w.dtype:
@@ -316,6 +349,8 @@ class AdamW(Adam):
     """Decoupled weight decay (reference contrib/adamw.cc)."""
 
     def update(self, index, weight, grad, state):
+        if isinstance(weight, NDArray) and weight.is_native:
+            return self._native_update(index, weight, grad, state)
         self._update_count(index)
         lr, wd = self._get_lr(index), self._get_wd(index)
         t = self._index_update_count[index]

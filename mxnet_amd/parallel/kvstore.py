@@ -63,7 +63,10 @@ def create(name='local'):
         return SyncPSKVStore(name)
     if native_mode() and isinstance(name, str) and (
             name.startswith('dist') or name in ('nccl', 'device')):
-        return NativeDistKVStore(name)
+        import os as _os
+        if int(_os.environ.get('WORLD_SIZE', '1')) > 1:
+            return NativeDistKVStore(name)
+        return KVStore('local')  # single process: no collective needed
     """Factory (reference KVStore::Create kvstore.cc:42-80)."""
     name = name.lower()
     if name == 'dist_async':
@@ -144,6 +147,13 @@ class KVStore(KVStoreBase):
                            .to(o._t.dtype))
 
     def pushpull(self, key, value, out=None, priority=0):
+        vals = value if isinstance(value, (list, tuple)) else [value]
+        outs = out if isinstance(out, (list, tuple)) else [out]
+        if getattr(vals[0], 'is_native', False) and len(vals) == 1 and \
+                (out is None or (len(outs) == 1 and outs[0] is vals[0])):
+            # single-process native runtime: one device per process, so
+            # reducing one value into itself is the identity
+            return
         self.push(key, value, priority)
         if out is not None:
             self.pull(key, out, priority)

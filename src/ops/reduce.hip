@@ -100,6 +100,25 @@ __global__ void reduce_all_stage2(const float* partial, T* y, int nb, long n,
   if (threadIdx.x == 0) y[0] = (T)(MEAN ? r / (double)n : r);
 }
 
+// scatter for the strided-view backward: dx[off + idx·strides] = dy[i]
+// (slice positions are unique — plain stores into a pre-zeroed buffer)
+template <typename T>
+__global__ void scatter_strided_kernel(const T* __restrict__ dy,
+                                       T* __restrict__ dx, long n,
+                                       Strides8 st, long off0) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (long)gridDim.x * blockDim.x) {
+    long rem = i, off = off0;
+#pragma unroll 4
+    for (int d = st.ndim - 1; d >= 0; --d) {
+      long idx = rem % st.shape[d];
+      rem /= st.shape[d];
+      off += idx * st.s0[d];
+    }
+    dx[off] = dy[i];
+  }
+}
+
 template <typename T>
 __global__ void gather_strided_kernel(const T* __restrict__ x,
                                       T* __restrict__ y, long n, Strides8 st) {
@@ -442,6 +461,106 @@ bool _registered_reduce = [] {
         s += ")";
         b.d["shape"] = s;
         return {RunOp2("_reduce_to", b, {og[0]})};
+      });
+
+  // _strided_copy: out[i] = in[offset + multi_idx(i)·strides] — the
+  // generic strided view materializer (basic __getitem__ slicing,
+  // positional-embedding slices, pooler token picks).  attrs: shape
+  // (output), strides (input element strides per output dim), offset.
+  auto strided_pack = [](const NodeAttrs& a, const TShape& oshape) {
+    auto strv = a.GetTuple("strides", {});
+    Strides8 st;
+    st.ndim = (int)oshape.size();
+    for (int i = 0; i < st.ndim; ++i) {
+      st.shape[i] = oshape[i];
+      st.s0[i] = strv[i];
+    }
+    return st;
+  };
+  Reg2("_strided_copy").in(1).infer(infer_target)
+      .gpu([strided_pack](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        long n = out[0].size();
+        if (n == 0) return;
+        Strides8 st = strided_pack(a, out[0].shape);
+        long off = a.GetInt("offset", 0);
+        MXC_DISPATCH_ALL(out[0].dtype, "_strided_copy", {
+          gather_strided_kernel<scalar_t><<<grid_for(n), kBlock, 0,
+                                            o.rc.stream>>>(
+              (const scalar_t*)in[0].dptr + off, (scalar_t*)out[0].dptr, n,
+              st);
+        });
+        HIP_CHECK_LAST();
+      })
+      .cpu([strided_pack](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        long n = out[0].size();
+        Strides8 st = strided_pack(a, out[0].shape);
+        long off0 = a.GetInt("offset", 0);
+        MXC_DISPATCH_ALL(out[0].dtype, "_strided_copy", {
+          auto* x = (const scalar_t*)in[0].dptr;
+          auto* y = (scalar_t*)out[0].dptr;
+          for (long i = 0; i < n; ++i) {
+            long rem = i, off = off0;
+            for (int d = st.ndim - 1; d >= 0; --d) {
+              long idx = rem % st.shape[d];
+              rem /= st.shape[d];
+              off += idx * st.s0[d];
+            }
+            y[i] = x[off];
+          }
+        });
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        NodeAttrs a = n.attrs;
+        std::string s = "(";
+        for (auto d : n.inputs[0].shape()) s += std::to_string(d) + ",";
+        s += ")";
+        a.d["xshape"] = s;
+        return {RunOp2("_strided_copy_bwd", a, {og[0]})};
+      });
+
+  Reg2("_strided_copy_bwd").in(1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>&,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        auto s = a.GetTuple("xshape", {});
+        os->assign(1, TShape(s.begin(), s.end()));
+        ot->assign(1, it[0]);
+      })
+      .gpu([strided_pack](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        long n = in[0].size();
+        Strides8 st = strided_pack(a, in[0].shape);
+        long off = a.GetInt("offset", 0);
+        MX_HIP_CALL(hipMemsetAsync(
+            out[0].dptr, 0, (size_t)out[0].size() * dtype_size(out[0].dtype),
+            o.rc.stream));
+        MXC_DISPATCH_ALL(in[0].dtype, "_strided_copy_bwd", {
+          scatter_strided_kernel<scalar_t><<<grid_for(n), kBlock, 0,
+                                             o.rc.stream>>>(
+              (const scalar_t*)in[0].dptr, (scalar_t*)out[0].dptr, n, st,
+              off);
+        });
+        HIP_CHECK_LAST();
+      })
+      .cpu([strided_pack](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        long n = in[0].size();
+        Strides8 st = strided_pack(a, in[0].shape);
+        long off0 = a.GetInt("offset", 0);
+        memset(out[0].dptr, 0,
+               (size_t)out[0].size() * dtype_size(out[0].dtype));
+        MXC_DISPATCH_ALL(in[0].dtype, "_strided_copy_bwd", {
+          auto* dy = (const scalar_t*)in[0].dptr;
+          auto* dx = (scalar_t*)out[0].dptr;
+          for (long i = 0; i < n; ++i) {
+            long rem = i, off = off0;
+            for (int d = st.ndim - 1; d >= 0; --d) {
+              long idx = rem % st.shape[d];
+              rem /= st.shape[d];
+              off += idx * st.s0[d];
+            }
+            dx[off] = dy[i];
+          }
+        });
       });
 
   // transpose (general permute)

@@ -143,3 +143,94 @@ def test_native_gpu_guard(native):
     a = mx.nd.ones((2, 2))
     with pytest.raises(Exception):
         mx.nd.ops._ninv('definitely_not_an_op', [a], {})
+
+
+def test_native_basic_indexing(native):
+    """Basic int/slice indexing on native arrays is the recorded
+    `_strided_copy` op with a scatter backward (reference
+    ndarray.py slicing; here reduce.hip _strided_copy)."""
+    ref = np.arange(24, dtype=np.float32).reshape(4, 6)
+    x = mx.nd.array(ref)
+    assert x.is_native
+    for key in [(slice(1, 3), slice(None, None, 2)), (slice(None), 0),
+                2, (3, 5), (slice(None, None, -1),), (Ellipsis, 1)]:
+        np.testing.assert_allclose(x[key].asnumpy(), np.ascontiguousarray(ref[key]))
+    # backward scatters into the leaf
+    w = mx.nd.array(ref)
+    w.attach_grad()
+    with autograd.record():
+        z = (w[1:3, ::3] * 2.0).sum()
+    z.backward()
+    exp = np.zeros_like(ref)
+    exp[1:3, ::3] = 2.0
+    np.testing.assert_allclose(w.grad.asnumpy(), exp)
+
+
+def test_native_bert_matches_torch_backend():
+    """Tiny BERT: native runtime and torch frontend produce the same
+    loss and qkv gradient on identical params/data (pos-embed slice,
+    additive mask, composed attention, pick-NLL loss all native)."""
+    from mxnet_amd.models.bert import BERTModel
+    from mxnet_amd.ndarray import ops as F
+
+    def run(use_native):
+        prev = set_native(use_native)
+        try:
+            torch.manual_seed(0)
+            net = BERTModel(vocab_size=50, units=16, hidden_size=32,
+                            num_layers=2, num_heads=2, max_length=32,
+                            dropout=0.0)
+            net.initialize(ctx=mx.cpu())
+            rng = np.random.RandomState(42)
+            tokens = mx.nd.array(rng.randint(0, 50, (2, 8)), dtype='int64')
+            mask = mx.nd.array((rng.rand(2, 8) > 0.2).astype('float32'))
+            net(tokens, None, mask)  # finish deferred init
+            params = net.collect_params()
+            for k, p in params.items():
+                rs = np.random.RandomState(
+                    sum(ord(c) for c in k) % (2**31))
+                p.set_data(mx.nd.array(
+                    rs.randn(*p.shape).astype('float32') * 0.05))
+            with autograd.record():
+                seq, pooled, mlm, nsp = net(tokens, None, mask)
+                logp = F.log_softmax(mlm.reshape(-1, 50))
+                lab = mx.nd.array(rng.randint(0, 50, (16,)),
+                                  dtype='float32')
+                loss = (F.pick(logp, lab, axis=-1) * -1.0).mean()
+            loss.backward()
+            kq = [k for k in params if 'qkv' in k and 'weight' in k][0]
+            return loss.asscalar(), params[kq].grad(mx.cpu()).asnumpy()
+        finally:
+            set_native(prev)
+
+    l_n, g_n = run(True)
+    l_t, g_t = run(False)
+    np.testing.assert_allclose(l_n, l_t, rtol=2e-4)
+    np.testing.assert_allclose(g_n, g_t, rtol=2e-3, atol=2e-5)
+
+
+def test_native_bert_adam_trains(native):
+    """BERT + Trainer('adam') end-to-end on the native runtime: the
+    fused adam_update registry op reduces the loss (VERDICT item 4
+    breadth: second model family beyond the conv nets)."""
+    from mxnet_amd.models.bert import BERTModel
+    from mxnet_amd.ndarray import ops as F
+    np.random.seed(0)
+    net = BERTModel(vocab_size=100, units=32, hidden_size=64,
+                    num_layers=2, num_heads=4, max_length=64, dropout=0.1)
+    net.initialize(ctx=mx.cpu())
+    tokens = mx.nd.array(np.random.randint(0, 100, (2, 8)), dtype='int64')
+    mask = mx.nd.array(np.ones((2, 8)), dtype='float32')
+    net(tokens, None, mask)
+    tr = Trainer(net.collect_params(), 'adam', {'learning_rate': 1e-3})
+    losses = []
+    for _ in range(4):
+        with autograd.record():
+            _, _, mlm, _ = net(tokens, None, mask)
+            logp = F.log_softmax(mlm.reshape(-1, 100))
+            lab = mx.nd.array(np.zeros((16,)), dtype='float32')
+            loss = (F.pick(logp, lab, axis=-1) * -1.0).mean()
+        loss.backward()
+        tr.step(2)
+        losses.append(loss.asscalar())
+    assert losses[-1] < losses[0]
