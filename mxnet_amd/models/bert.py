@@ -32,40 +32,64 @@ class BERTSelfAttention(HybridBlock):
                              weight_initializer=init.Normal(0.02))
         self.dropout = nn.Dropout(dropout) if dropout else None
 
-    def _forward_native(self, x, addmask):
-        """Native-runtime attention.  ``addmask`` is the model-level
-        additive float mask [B*H, S, S] (0 valid / -1e4 masked) or None.
-        Fused `interleaved_attention` when un-masked, no-dropout and
-        MFMA-aligned; otherwise composed from native registry ops."""
+    def _forward_native(self, x, mask_u8):
+        """Native-runtime attention.  ``mask_u8`` is the model-level
+        uint8 mask [B*H, S, S] (1 valid / 0 masked) or None — it rides
+        the fused softmax kernel's mask slot (no additive pass).
+        GPU half-precision MFMA-aligned shapes take the fused
+        `interleaved_attention` op (strided QKV GEMMs + masked softmax +
+        attention dropout, one tape node); everything else composes from
+        native registry ops (strided gathers + batch_dot + softmax)."""
         import math as _m
         from ..ndarray import ops as F
+        from .. import autograd as _ag
         B, S, U = x.shape
         H, D = self._num_heads, self._head_dim
         qkv = self.qkv(x)
         on_gpu = x.context.device_type == 'gpu'
-        if on_gpu and self.dropout is None and addmask is None \
-                and D % 8 == 0 and S % 8 == 0:
-            out = qkv._invoke('interleaved_attention', [qkv],
-                              {'heads': str(H),
-                               'temperature': str(_m.sqrt(D))})
+        if on_gpu and str(x.dtype) in ('float16', 'bfloat16') \
+                and D % 8 == 0 and S % 8 == 0 \
+                and not getattr(self, '_force_composed', False):
+            # fused path: strided QKV GEMMs + masked softmax + attention
+            # dropout inside ONE registry op (src/ops/gemm.hip
+            # attention_fwd_raw) — no head gather/scatter kernels
+            p = self.dropout._rate if (self.dropout is not None
+                                       and _ag.is_training()) else 0.0
+            import random as _random
+            attrs = {'heads': str(H), 'temperature': str(_m.sqrt(D)),
+                     'p': str(p), 'seed': str(_random.getrandbits(48))}
+            ins = [qkv] if mask_u8 is None else [qkv, mask_u8]
+            out = qkv._invoke('interleaved_attention', ins, attrs)
             return self.proj(out)
-        q = qkv[:, :, 0:U]
-        k = qkv[:, :, U:2 * U]
-        v = qkv[:, :, 2 * U:3 * U]
+        # single-kernel head split via the strided gather (oshape folds
+        # the trailing reshape): q,v as [B*H,S,D]; k gathered directly
+        # TRANSPOSED as [B*H,D,S] — no separate transpose pass
+        U3 = 3 * U
 
-        def heads(z):
-            return z.reshape(B, S, H, D).transpose((0, 2, 1, 3)) \
-                    .reshape(B * H, S, D)
-        q, k, v = heads(q), heads(k), heads(v)
-        scores = q._invoke('batch_dot', [q, k.transpose((0, 2, 1))])
-        if addmask is not None:
-            scores = scores + addmask
-        att = F.softmax(scores, axis=-1, temperature=_m.sqrt(D))
+        def gather(plan_shape, strides, offset, oshape):
+            return qkv._invoke('_strided_copy', [qkv], {
+                'shape': '(' + ','.join(map(str, plan_shape)) + ',)',
+                'strides': '(' + ','.join(map(str, strides)) + ',)',
+                'offset': str(offset),
+                'oshape': '(' + ','.join(map(str, oshape)) + ',)'})
+        q = gather((B, H, S, D), (S * U3, D, U3, 1), 0, (B * H, S, D))
+        kt = gather((B, H, D, S), (S * U3, D, 1, U3), U, (B * H, D, S))
+        v = gather((B, H, S, D), (S * U3, D, U3, 1), 2 * U, (B * H, S, D))
+        scores = q._invoke('batch_dot', [q, kt])
+        if mask_u8 is not None:
+            att = scores._invoke('masked_softmax', [scores, mask_u8],
+                                 {'temperature': str(_m.sqrt(D))})
+        else:
+            att = F.softmax(scores, axis=-1, temperature=_m.sqrt(D))
         if self.dropout is not None:
             att = self.dropout(att)
         out = att._invoke('batch_dot', [att, v])
-        out = out.reshape(B, H, S, D).transpose((0, 2, 1, 3)) \
-                 .reshape(B, S, U)
+        # single-kernel head merge: [B*H,S,D] -> [B,S,U]
+        out = out._invoke('_strided_copy', [out], {
+            'shape': '(%d,%d,%d,%d,)' % (B, S, H, D),
+            'strides': '(%d,%d,%d,1,)' % (H * S * D, D, S * D),
+            'offset': '0',
+            'oshape': '(%d,%d,%d,)' % (B, S, U)})
         return self.proj(out)
 
     def forward(self, x, mask=None):
@@ -227,18 +251,17 @@ class BERTModel(HybridBlock):
         emb = self.embed_ln(emb)
         if self.embed_dropout is not None:
             emb = self.embed_dropout(emb)
-        addmask = None
+        mask_u8 = None
         if valid_mask is not None:
             B, Sm = valid_mask.shape
             H = self._num_heads
             m = valid_mask
-            if str(m.dtype) != str(emb.dtype):
-                m = m.astype(emb.dtype)
-            add = (m - 1.0) * 10000.0  # 0 valid / -1e4 masked (additive)
-            addmask = add.reshape(B, 1, 1, Sm) \
-                         .broadcast_to((B, H, Sm, Sm)) \
-                         .reshape(B * H, Sm, Sm)
-        seq = self.encoder(emb, addmask)
+            if str(m.dtype) != 'uint8':
+                m = m.astype('uint8')
+            mask_u8 = m.reshape(B, 1, 1, Sm) \
+                       .broadcast_to((B, H, Sm, Sm)) \
+                       .reshape(B * H, Sm, Sm)
+        seq = self.encoder(emb, mask_u8)
         pooled = self.pooler(seq[:, 0])
         mlm = self.mlm_decoder(self.mlm_ln(self.mlm_dense(seq)))
         nsp = self.nsp_classifier(pooled)

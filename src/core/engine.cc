@@ -277,6 +277,20 @@ struct Engine::Impl {
         MX_HIP_CALL(hipEventCreate(&rec.ev1));
         (void)hipEventRecord(rec.ev0, rc.stream);
       }
+      // dependency-failure propagation (reference threaded_engine.cc
+      // OnCompleteStatic): if any input/output var carries a pending
+      // exception, skip fn and forward it — downstream consumers see the
+      // ORIGINAL error at their sync point instead of computing on
+      // uninitialized buffers.  Deleters still run (storage must free).
+      std::exception_ptr in_exc;
+      if (!opr->delete_var) {
+        std::lock_guard<std::mutex> g(mu_);
+        for (Var* v : opr->const_vars)
+          if (v->exc) { in_exc = v->exc; break; }
+        if (!in_exc)
+          for (Var* v : opr->mutable_vars)
+            if (v->exc) { in_exc = v->exc; break; }
+      }
       try {
         // order execution behind producers on other streams
         for (auto& w : opr->waits) {
@@ -287,7 +301,10 @@ struct Engine::Impl {
             MX_HIP_CALL(hipEventSynchronize(w->ev));
           }
         }
-        if (opr->fn) opr->fn(rc);
+        if (in_exc)
+          exc = in_exc;
+        else if (opr->fn)
+          opr->fn(rc);
       } catch (...) {
         exc = std::current_exception();
       }

@@ -84,14 +84,14 @@ void init_raw(py::module_& m) {
   r.def("attention_fwd", [](int dev, uintptr_t stream, py::handle qkv,
                             py::handle mask, int H, double temp,
                             py::handle out, py::handle att) {
-    attention_fwd_raw(MakeLC(dev, stream), ToArr(qkv), ToArr(mask), H, temp,
-                      ToArr(out), ToArr(att));
+    attention_fwd_raw(MakeLC(dev, stream), ToArr(qkv), ToArr(mask),
+                      0.0, 0, Arr(), H, temp, ToArr(out), ToArr(att));
   });
   r.def("attention_bwd", [](int dev, uintptr_t stream, py::handle dout,
                             py::handle qkv, py::handle att, int H,
                             double temp, py::handle dqkv) {
     attention_bwd_raw(MakeLC(dev, stream), ToArr(dout), ToArr(qkv),
-                      ToArr(att), H, temp, ToArr(dqkv));
+                      0.0, Arr(), ToArr(att), H, temp, ToArr(dqkv));
   });
 
   r.def("softmax_fwd", [](int dev, uintptr_t stream, py::handle x,

@@ -1192,6 +1192,7 @@ void launch_tn_batched(const LaunchCtx& lc, int dtype, const void* A,
 }  // namespace
 
 void attention_fwd_raw(const LaunchCtx& lc, const Arr& qkv, const Arr& mask,
+                       double p, int64_t seed, const Arr& dropmask,
                        int H, double temperature, const Arr& out,
                        const Arr& att) {
   long B = qkv.size(0), S = qkv.size(1), U3 = qkv.size(2);
@@ -1208,15 +1209,28 @@ void attention_fwd_raw(const LaunchCtx& lc, const Arr& qkv, const Arr& mask,
                     S * U3, S * U3, (long)H * S * S,
                     GemmLd{U3, U3, S, (int)H, D, D, S * S}, BH);
   softmax_fwd_raw(lc, att_raw, mask, false, temperature, att);
+  // attention dropout: att stays PRE-dropout (softmax backward needs
+  // it); the dropped probs go to the arena and feed the second GEMM
+  const void* att2 = att.ptr;
+  if (p > 0) {
+    MX_CHECK(dropmask.defined(), "attention: dropout needs a mask output");
+    Arr dropped;
+    dropped.dtype = att.dtype;
+    dropped.shape = att.shape;
+    dropped.ptr = lc.workspace((size_t)BH * S * S * es);
+    dropout_fwd_raw(lc, att, p, seed, dropped, dropmask);
+    att2 = dropped.ptr;
+  }
   // V^T panels [BH, D, S]
   Arr vt = transpose_strided_ws(lc, qkv.dtype, qp + 2 * U * es, S, D, U3,
                                 (int)H, S * U3, D, BH);
-  launch_nt_strided(lc, qkv.dtype, att.ptr, vt.ptr, out.ptr, S, D, S,
+  launch_nt_strided(lc, qkv.dtype, att2, vt.ptr, out.ptr, S, D, S,
                     (long)H * S * S, (long)H * D * S, S * U,
                     GemmLd{S, S, U, (int)H, S * S, D * S, D}, BH);
 }
 
 void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
+                       double p, const Arr& dropmask,
                        const Arr& att, int H, double temperature,
                        const Arr& dqkv) {
   long B = qkv.size(0), S = qkv.size(1), U3 = qkv.size(2);
@@ -1231,6 +1245,24 @@ void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
   launch_nt_strided(lc, qkv.dtype, dp, qp + 2 * U * es, datt.ptr, S, S, D,
                     S * U, S * U3, (long)H * S * S,
                     GemmLd{U, U3, S, (int)H, D, D, S * S}, BH);
+  // the NT above is d(att_dropped); undo the dropout (same mask/scale)
+  // to get d(att) for the softmax backward
+  Arr att2 = att;  // operand of the dV GEMM (dropped probs if p>0)
+  if (p > 0) {
+    Arr tmp;
+    tmp.dtype = qkv.dtype;
+    tmp.shape = {BH, S, S};
+    tmp.ptr = lc.workspace((size_t)BH * S * S * es);
+    dropout_bwd_raw(lc, datt, dropmask, p, tmp);
+    datt = tmp;
+    // recompute att_dropped = att * mask / (1-p) (same elementwise op)
+    Arr dropped;
+    dropped.dtype = qkv.dtype;
+    dropped.shape = {BH, S, S};
+    dropped.ptr = lc.workspace((size_t)BH * S * S * es);
+    dropout_bwd_raw(lc, att, dropmask, p, dropped);
+    att2 = dropped;
+  }
   Arr ds;
   ds.dtype = qkv.dtype;
   ds.shape = {BH, S, S};
@@ -1244,7 +1276,7 @@ void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
                     GemmLd{S, S, U3, (int)H, S * S, D * S, D}, BH);
   launch_tn_batched(lc, qkv.dtype, ds.ptr, qp, dq + U * es, S, S, D, S * S,
                     U3, (int)H, S * U3, D, U3, S * U3, D, BH);
-  launch_tn_batched(lc, qkv.dtype, att.ptr, dp, dq + 2 * U * es, S, S, D,
+  launch_tn_batched(lc, qkv.dtype, att2.ptr, dp, dq + 2 * U * es, S, S, D,
                     S * S, U, (int)H, S * U, D, U3, S * U3, D, BH);
 }
 

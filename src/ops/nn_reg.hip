@@ -1088,6 +1088,44 @@ bool _registered_nn = [] {
   };
   softmax_reg("softmax", false);
   softmax_reg("log_softmax", true);
+
+  // masked softmax (BERT attention): uint8 mask rides the fused kernel's
+  // mask slot — masked entries get probability 0, and the standard
+  // softmax backward then yields zero grads there (no separate bwd op).
+  RegN("masked_softmax").in(2).infer(InferSame())
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        softmax_fwd_raw(LC(o), in[0], in[1], false,
+                        a.GetFloat("temperature", 1.0), out[0]);
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "masked_softmax");
+        float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+        long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
+        const float* x = (const float*)in[0].dptr;
+        const unsigned char* mk = (const unsigned char*)in[1].dptr;
+        float* y = (float*)out[0].dptr;
+        for (long r = 0; r < rows; ++r) {
+          float m = -3.4e38f;
+          bool any = false;
+          for (long c = 0; c < C; ++c)
+            if (mk[r * C + c]) { m = std::max(m, x[r * C + c]); any = true; }
+          double sum = 0;
+          for (long c = 0; c < C; ++c)
+            if (mk[r * C + c]) sum += exp((x[r * C + c] - m) * invT);
+          for (long c = 0; c < C; ++c)
+            y[r * C + c] = (any && mk[r * C + c])
+                               ? (float)(exp((x[r * C + c] - m) * invT) / sum)
+                               : 0.f;
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        std::vector<NDArray> r(2);
+        NodeAttrs a = n.attrs;
+        a.d["log_mode"] = "0";
+        r[0] = RunN("_softmax_bwd", a, {og[0], n.outputs[0]});
+        return r;
+      });
   RegN("_softmax_bwd").in(2).infer(InferSame())
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         softmax_bwd_raw(LC(o), in[0], in[1], a.GetBool("log_mode", false),
@@ -1456,7 +1494,12 @@ bool _registered_nn = [] {
       });
 
   // ======================= attention / lstm cell =========================
-  RegN("interleaved_attention").in(-1).out(2)
+  // in: qkv [B,S,3U] (+ optional uint8 mask [B*H,S,S] riding the fused
+  // softmax's mask slot); out: y, saved pre-dropout att probs, dropout
+  // mask (uint8; unused when p=0).  Attention dropout runs INSIDE the
+  // fused op (probs dropped before the att@V GEMM) so BERT's training
+  // path needs no composed fallback.
+  RegN("interleaved_attention").in(-1).out(3)
       .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
@@ -1464,29 +1507,33 @@ bool _registered_nn = [] {
         int64_t H = a.GetInt("heads", 1);
         os->assign(1, TShape{B, S, U});
         os->push_back({B * H, S, S});
+        os->push_back({B * H, S, S});
         ot->assign(2, it[0]);
+        ot->push_back(kUint8);
       })
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         attention_fwd_raw(LC(o), in[0], in.size() > 1 ? Arr(in[1]) : Arr(),
-                          (int)a.GetInt("heads", 1),
+                          a.GetFloat("p", 0.0), a.GetInt("seed", 0),
+                          Arr(out[2]), (int)a.GetInt("heads", 1),
                           a.GetFloat("temperature", 1.0), out[0], out[1]);
       })
       .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
                -> std::vector<NDArray> {
         std::vector<NDArray> r(n.inputs.size());
         r[0] = RunN("_attention_bwd", n.attrs,
-                    {og[0], n.inputs[0], n.outputs[1]});
+                    {og[0], n.inputs[0], n.outputs[1], n.outputs[2]});
         return r;
       });
-  RegN("_attention_bwd").in(3)
+  RegN("_attention_bwd").in(4)
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
         os->assign(1, is[1]);
-        ot->assign(1, it[0]);
+        ot->assign(1, it[1]);
       })
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
-        attention_bwd_raw(LC(o), in[0], in[1], in[2],
+        attention_bwd_raw(LC(o), in[0], in[1], a.GetFloat("p", 0.0),
+                          in[3], in[2],
                           (int)a.GetInt("heads", 1),
                           a.GetFloat("temperature", 1.0), out[0]);
       });

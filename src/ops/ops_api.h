@@ -31,9 +31,11 @@ void gemm_tn_fused_raw(const LaunchCtx& lc, const Arr& A, const Arr& B,
                        const Arr& C, const Arr& dbias);
 // qkv [B,S,3U] -> out [B,S,U], att [B*H,S,S] (saved for backward)
 void attention_fwd_raw(const LaunchCtx& lc, const Arr& qkv, const Arr& mask,
+                       double p, int64_t seed, const Arr& dropmask,
                        int H, double temperature, const Arr& out,
                        const Arr& att);
 void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
+                       double p, const Arr& dropmask,
                        const Arr& att, int H, double temperature,
                        const Arr& dqkv);
 

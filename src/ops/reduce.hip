@@ -429,7 +429,7 @@ bool _registered_reduce = [] {
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         long n = out[0].size();
         Strides8 st = make_strides(out[0].shape, in[0].shape, in[0].shape);
-        MXC_DISPATCH_FLOAT(out[0].dtype, "broadcast_to", {
+        MXC_DISPATCH_ALL(out[0].dtype, "broadcast_to", {
           gather_strided_kernel<scalar_t><<<grid_for(n), kBlock, 0,
                                             o.rc.stream>>>(
               (const scalar_t*)in[0].dptr, (scalar_t*)out[0].dptr, n, st);
@@ -439,7 +439,7 @@ bool _registered_reduce = [] {
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         long n = out[0].size();
         Strides8 st = make_strides(out[0].shape, in[0].shape, in[0].shape);
-        MXC_DISPATCH_FLOAT(out[0].dtype, "broadcast_to", {
+        MXC_DISPATCH_ALL(out[0].dtype, "broadcast_to", {
           auto* x = (const scalar_t*)in[0].dptr;
           auto* y = (scalar_t*)out[0].dptr;
           for (long i = 0; i < n; ++i) {
@@ -465,23 +465,38 @@ bool _registered_reduce = [] {
 
   // _strided_copy: out[i] = in[offset + multi_idx(i)·strides] — the
   // generic strided view materializer (basic __getitem__ slicing,
-  // positional-embedding slices, pooler token picks).  attrs: shape
-  // (output), strides (input element strides per output dim), offset.
-  auto strided_pack = [](const NodeAttrs& a, const TShape& oshape) {
+  // positional-embedding slices, pooler token picks, attention head
+  // split/merge).  attrs: shape (gather plan dims), strides (input
+  // element strides per plan dim), offset, and optional oshape — a
+  // same-numel reinterpretation of the contiguous result, folding a
+  // trailing reshape into the same kernel (e.g. gather [B,H,S,D] →
+  // declare [B*H,S,D]).
+  auto strided_pack = [](const NodeAttrs& a) {
+    auto shv = a.GetTuple("shape", {});
     auto strv = a.GetTuple("strides", {});
+    MX_CHECK(shv.size() == strv.size() && shv.size() <= 8,
+             "_strided_copy: shape/strides rank mismatch");
     Strides8 st;
-    st.ndim = (int)oshape.size();
+    st.ndim = (int)shv.size();
     for (int i = 0; i < st.ndim; ++i) {
-      st.shape[i] = oshape[i];
+      st.shape[i] = shv[i];
       st.s0[i] = strv[i];
     }
     return st;
   };
-  Reg2("_strided_copy").in(1).infer(infer_target)
+  Reg2("_strided_copy").in(1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>&,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        auto s = a.has("oshape") ? a.GetTuple("oshape", {})
+                                 : a.GetTuple("shape", {});
+        os->assign(1, TShape(s.begin(), s.end()));
+        ot->assign(1, it[0]);
+      })
       .gpu([strided_pack](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         long n = out[0].size();
         if (n == 0) return;
-        Strides8 st = strided_pack(a, out[0].shape);
+        Strides8 st = strided_pack(a);
         long off = a.GetInt("offset", 0);
         MXC_DISPATCH_ALL(out[0].dtype, "_strided_copy", {
           gather_strided_kernel<scalar_t><<<grid_for(n), kBlock, 0,
@@ -493,7 +508,7 @@ bool _registered_reduce = [] {
       })
       .cpu([strided_pack](const NodeAttrs& a, const OpCtx&, V in, V out) {
         long n = out[0].size();
-        Strides8 st = strided_pack(a, out[0].shape);
+        Strides8 st = strided_pack(a);
         long off0 = a.GetInt("offset", 0);
         MXC_DISPATCH_ALL(out[0].dtype, "_strided_copy", {
           auto* x = (const scalar_t*)in[0].dptr;
@@ -529,7 +544,7 @@ bool _registered_reduce = [] {
       })
       .gpu([strided_pack](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         long n = in[0].size();
-        Strides8 st = strided_pack(a, in[0].shape);
+        Strides8 st = strided_pack(a);
         long off = a.GetInt("offset", 0);
         MX_HIP_CALL(hipMemsetAsync(
             out[0].dptr, 0, (size_t)out[0].size() * dtype_size(out[0].dtype),
@@ -544,7 +559,7 @@ bool _registered_reduce = [] {
       })
       .cpu([strided_pack](const NodeAttrs& a, const OpCtx&, V in, V out) {
         long n = in[0].size();
-        Strides8 st = strided_pack(a, in[0].shape);
+        Strides8 st = strided_pack(a);
         long off0 = a.GetInt("offset", 0);
         memset(out[0].dptr, 0,
                (size_t)out[0].size() * dtype_size(out[0].dtype));

@@ -285,3 +285,20 @@ def test_version_bumps():
     c.invoke_into('_grad_add', [nd(np.ones(4))], [a], {})
     a.wait_to_read()
     assert c.var_version(a.var) > v0
+
+
+def test_engine_exception_propagates_to_consumers():
+    """A failed op poisons its outputs; consumers skip execution and
+    forward the ORIGINAL error to their sync point instead of computing
+    on uninitialized buffers (reference threaded_engine OnComplete)."""
+    import numpy as np
+    from mxnet_amd import _core
+    x = _core.from_numpy(np.ones((2, 2), np.float16), 1, 0)
+    # batch_dot CPU oracle is fp32-only -> fn throws inside the worker
+    bad = _core.invoke('batch_dot', [
+        _core.from_numpy(np.ones((1, 2, 2), np.float16), 1, 0),
+        _core.from_numpy(np.ones((1, 2, 2), np.float16), 1, 0)], {})[0]
+    down = _core.invoke('_mul_scalar', [bad], {'alpha': '2.0'})[0]
+    import pytest as _pt
+    with _pt.raises(RuntimeError, match='batch_dot'):
+        down.asnumpy()

@@ -193,6 +193,57 @@ def test_native_hipgraph_capture_gpu(native):
     np.testing.assert_allclose(out.asnumpy(), want, rtol=1e-4, atol=1e-5)
 
 
+def test_native_bert_fp16_gpu(native):
+    """Small BERT fp16 on the native runtime (GPU): composed attention
+    (dropout path), fused adam_update multi-precision, loss decreases
+    and stays finite."""
+    from mxnet_amd.models.bert import BERTModel
+    from mxnet_amd.ndarray import ops as F
+    np.random.seed(0)
+    net = BERTModel(vocab_size=1000, units=128, hidden_size=256,
+                    num_layers=2, num_heads=4, max_length=64, dropout=0.1)
+    net.initialize(ctx=mx.gpu(0))
+    tokens = mx.nd.array(np.random.randint(0, 1000, (4, 32)),
+                         ctx=mx.gpu(0), dtype='int64')
+    mask = mx.nd.array(np.ones((4, 32)), ctx=mx.gpu(0), dtype='float16')
+    net(tokens, None, mask)
+    net.cast('float16')
+    tr = Trainer(net.collect_params(), 'adam',
+                 {'learning_rate': 1e-3, 'multi_precision': True})
+    losses = []
+    for _ in range(6):
+        with autograd.record():
+            _, _, mlm, _ = net(tokens, None, mask)
+            logp = F.log_softmax(mlm.reshape(-1, 1000).astype('float32'))
+            lab = mx.nd.array(np.zeros((128,)), ctx=mx.gpu(0),
+                              dtype='float32')
+            loss = (F.pick(logp, lab, axis=-1) * -1.0).mean()
+        loss.backward()
+        tr.step(4)
+        losses.append(loss.asscalar())
+    assert all(l == l for l in losses), losses  # finite
+    assert losses[-1] < losses[0], losses
+
+
+def test_native_fused_attention_gpu(native):
+    """Fused interleaved_attention (no dropout, no mask) matches the
+    composed batch_dot/softmax native path in fp16."""
+    from mxnet_amd.models.bert import BERTSelfAttention
+    np.random.seed(3)
+    att = BERTSelfAttention(128, 4, dropout=0.0)
+    att.initialize(ctx=mx.gpu(0))
+    x32 = mx.nd.array(np.random.randn(2, 32, 128) * 0.1, ctx=mx.gpu(0))
+    att(x32)  # finish deferred init
+    att.cast('float16')
+    x = x32.astype('float16')
+    y_fused = att(x).asnumpy().astype(np.float32)
+    att._force_composed = True
+    with autograd.predict_mode():
+        y_comp = att(x).asnumpy().astype(np.float32)
+    att._force_composed = False
+    np.testing.assert_allclose(y_fused, y_comp, rtol=2e-2, atol=2e-2)
+
+
 def test_native_rccl_world1(native):
     """world=1 communicator: allreduce/broadcast are engine-sequenced
     no-ops (average still runs its scale kernel)."""
