@@ -1245,17 +1245,12 @@ void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
   launch_nt_strided(lc, qkv.dtype, dp, qp + 2 * U * es, datt.ptr, S, S, D,
                     S * U, S * U3, (long)H * S * S,
                     GemmLd{U, U3, S, (int)H, D, D, S * S}, BH);
-  // the NT above is d(att_dropped); undo the dropout (same mask/scale)
-  // to get d(att) for the softmax backward
+  // datt above is d(att_dropped); the dropout undo (dy*mask/keep) rides
+  // the softmax-backward kernel inline — no separate full-tensor pass
   Arr att2 = att;  // operand of the dV GEMM (dropped probs if p>0)
   if (p > 0) {
-    Arr tmp;
-    tmp.dtype = qkv.dtype;
-    tmp.shape = {BH, S, S};
-    tmp.ptr = lc.workspace((size_t)BH * S * S * es);
-    dropout_bwd_raw(lc, datt, dropmask, p, tmp);
-    datt = tmp;
-    // recompute att_dropped = att * mask / (1-p) (same elementwise op)
+    // recompute att_dropped = att * mask / (1-p) (same elementwise op
+    // as dropout backward) for the dV GEMM
     Arr dropped;
     dropped.dtype = qkv.dtype;
     dropped.shape = {BH, S, S};
@@ -1263,11 +1258,21 @@ void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
     dropout_bwd_raw(lc, att, dropmask, p, dropped);
     att2 = dropped;
   }
+  bool inline_undo = p > 0 && S <= 256;  // rowreg softmax-bwd constraint
+  if (p > 0 && !inline_undo) {
+    Arr tmp;
+    tmp.dtype = qkv.dtype;
+    tmp.shape = {BH, S, S};
+    tmp.ptr = lc.workspace((size_t)BH * S * S * es);
+    dropout_bwd_raw(lc, datt, dropmask, p, tmp);
+    datt = tmp;
+  }
   Arr ds;
   ds.dtype = qkv.dtype;
   ds.shape = {BH, S, S};
   ds.ptr = lc.workspace((size_t)BH * S * S * es);
-  softmax_bwd_raw(lc, datt, att, false, temperature, ds);
+  softmax_bwd_raw(lc, datt, att, false, temperature, ds,
+                  inline_undo ? dropmask : Arr(), inline_undo ? p : 0.0);
   char* dq = (char*)dqkv.ptr;
   Arr kt = transpose_strided_ws(lc, qkv.dtype, qp + U * es, S, D, U3,
                                 (int)H, S * U3, D, BH);

@@ -222,6 +222,41 @@ bool _registered_nn = [] {
         return r;
       });
 
+  // dot_tn with the bias-gradient fused into the TN GEMM epilogue
+  // (gemm_tn_fused_raw): out(2) = [dw (dtype of dy), dbias fp32] —
+  // replaces the separate colsum pass in FullyConnected backward.
+  RegN("dot_tn_fused").in(2).out(2)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{is[0][1], is[1][1]});
+        os->push_back({is[0][1]});
+        ot->assign(1, it[0]);
+        ot->push_back(kFloat32);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        gemm_tn_fused_raw(LC(o), in[0], in[1], out[0], out[1]);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        CPU_FLOAT_ONLY(in[0], "dot_tn_fused");
+        long M = in[0].shape[0], I = in[0].shape[1], J = in[1].shape[1];
+        const float* a = (const float*)in[0].dptr;
+        const float* b = (const float*)in[1].dptr;
+        float* c = (float*)out[0].dptr;
+        float* db = (float*)out[1].dptr;
+        for (long i = 0; i < I; ++i) {
+          double bs = 0;
+          for (long m = 0; m < M; ++m) bs += a[m * I + i];
+          db[i] = (float)bs;
+          for (long j = 0; j < J; ++j) {
+            double acc = 0;
+            for (long m = 0; m < M; ++m)
+              acc += (double)a[m * I + i] * b[m * J + j];
+            c[i * J + j] = (float)acc;
+          }
+        }
+      });
+
   // dot_nn: [M,N] x [N,K]
   RegN("dot_nn").in(2)
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
@@ -375,8 +410,14 @@ bool _registered_nn = [] {
         if (n.need_igrad.empty() || n.need_igrad[0])
           r[0] = RunN("dot_nn", {}, {dy2, n.inputs[1]})
                      .Reshape(n.inputs[0].shape());
-        r[1] = RunN("dot_tn", {}, {dy2, x2});
-        if (n.inputs.size() > 2) r[2] = RunN("colsum", {}, {dy2});
+        if (n.inputs.size() > 2) {
+          // bias grad rides the TN GEMM epilogue (one launch)
+          auto wb = RunNMulti("dot_tn_fused", {}, {dy2, x2});
+          r[1] = wb[0];
+          r[2] = wb[1];
+        } else {
+          r[1] = RunN("dot_tn", {}, {dy2, x2});
+        }
         return r;
       });
 

@@ -43,7 +43,8 @@ void attention_bwd_raw(const LaunchCtx& lc, const Arr& dout, const Arr& qkv,
 void softmax_fwd_raw(const LaunchCtx& lc, const Arr& x, const Arr& mask,
                      bool log_mode, double temperature, const Arr& out);
 void softmax_bwd_raw(const LaunchCtx& lc, const Arr& dy, const Arr& y,
-                     bool log_mode, double temperature, const Arr& out);
+                     bool log_mode, double temperature, const Arr& out,
+                     const Arr& dropmask = Arr(), double p = 0.0);
 void colsum_raw(const LaunchCtx& lc, const Arr& x, const Arr& out);
 
 // ---- norm.hip -------------------------------------------------------------

@@ -169,24 +169,31 @@ __global__ void softmax_fwd_rowreg_kernel(
   }
 }
 
-// backward fast path: dx = (dy - sum(dy*y)) * y * invT, one read of each
+// backward fast path: dx = (dy - sum(dy*y)) * y * invT, one read of each.
+// dmask/inv_keep: undo an attention dropout inline (dy_eff = dy*mask/keep)
+// instead of a separate full-tensor pass.
 template <typename T, int VPT>
 __global__ void softmax_bwd_rowreg_kernel(const T* __restrict__ dy,
                                           const T* __restrict__ yv,
                                           T* __restrict__ dx, long rows,
-                                          int C, float invT) {
+                                          int C, float invT,
+                                          const unsigned char* __restrict__
+                                              dmask = nullptr,
+                                          float inv_keep = 1.f) {
   const int lane = threadIdx.x & 63;
   const int row_in_blk = threadIdx.x >> 6;
   for (long r = (long)blockIdx.x * 4 + row_in_blk; r < rows;
        r += (long)gridDim.x * 4) {
     const T* gr = dy + r * C;
     const T* yr = yv + r * C;
+    const unsigned char* mr = dmask ? dmask + r * C : nullptr;
     float g[VPT], yy[VPT];
     float s = 0.f;
 #pragma unroll
     for (int j = 0; j < VPT; ++j) {
       const int c = lane + j * 64;
       g[j] = c < C ? (float)gr[c] : 0.f;
+      if (mr && c < C) g[j] = mr[c] ? g[j] * inv_keep : 0.f;
       yy[j] = c < C ? (float)yr[c] : 0.f;
       s += g[j] * yy[j];
     }
@@ -277,30 +284,36 @@ void softmax_fwd_raw(const LaunchCtx& lc, const Arr& x, const Arr& mask,
 }
 
 void softmax_bwd_raw(const LaunchCtx& lc, const Arr& dy, const Arr& y,
-                     bool log_mode, double temperature, const Arr& dx) {
+                     bool log_mode, double temperature, const Arr& dx,
+                     const Arr& dropmask, double p) {
   long C = dy.size(-1), rows = dy.numel() / (C > 0 ? C : 1);
   if (dy.numel() == 0) return;
   int grid = (int)std::min<long>(rows, 4096);
   float invT = (float)(1.0 / temperature);
+  const unsigned char* dm =
+      (p > 0 && dropmask.defined()) ? dropmask.data<unsigned char>()
+                                    : nullptr;
+  float inv_keep = p > 0 ? (float)(1.0 / (1.0 - p)) : 1.f;
   if (C <= 256 && !log_mode && rows >= 64) {
     int g4 = (int)std::min<long>((rows + 3) / 4, 4096);
     DISPATCH_FLOAT_NATIVE(dy.dtype, "softmax_bwd_rr", [&] {
       if (C <= 64)
         softmax_bwd_rowreg_kernel<scalar_t, 1><<<g4, 256, 0, lc.stream>>>(
             dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
-            rows, (int)C, invT);
+            rows, (int)C, invT, dm, inv_keep);
       else if (C <= 128)
         softmax_bwd_rowreg_kernel<scalar_t, 2><<<g4, 256, 0, lc.stream>>>(
             dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
-            rows, (int)C, invT);
+            rows, (int)C, invT, dm, inv_keep);
       else
         softmax_bwd_rowreg_kernel<scalar_t, 4><<<g4, 256, 0, lc.stream>>>(
             dy.data<scalar_t>(), y.data<scalar_t>(), (scalar_t*)dx.ptr,
-            rows, (int)C, invT);
+            rows, (int)C, invT, dm, inv_keep);
     });
     HIP_CHECK_LAST();
     return;
   }
+  MX_CHECK(!dm, "softmax_bwd: inline dropout needs the rowreg path (C<=256)");
   DISPATCH_FLOAT_NATIVE(dy.dtype, "softmax_bwd", [&] {
     constexpr int VEC = sizeof(scalar_t) == 2 ? 8 : 4;
     softmax_bwd_kernel<scalar_t, VEC><<<grid, 256, 0, lc.stream>>>(
