@@ -1,10 +1,15 @@
 #include "op.h"
 
 #include <algorithm>
+#include <map>
 #include <mutex>
 #include <unordered_set>
 
 namespace mxcore {
+
+// defined in src/ops/elemwise.hip (kernel layer); consulted here so the
+// tape only defers leaf-grad writes the multi-copy kernel can service
+int multi_copy_mode(int src_dtype, int dst_dtype);
 
 OpRegistry* OpRegistry::Get() {
   static OpRegistry inst;
@@ -222,9 +227,23 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
   OpEntry* add_into = OpRegistry::Get()->Find("_grad_add");
   OpEntry* ones_op = OpRegistry::Get()->Find("ones_like");
   OpEntry* copy_into_op = OpRegistry::Get()->Find("_copy_into");
-  MX_CHECK(add_into && ones_op && copy_into_op, "core grad ops missing");
+  OpEntry* multi_copy_op = OpRegistry::Get()->Find("_multi_copy");
+  MX_CHECK(add_into && ones_op && copy_into_op && multi_copy_op,
+           "core grad ops missing");
   // leaves whose attached grad buffer already received this sweep's value
   std::unordered_set<NDArray::Chunk*> leaf_written;
+  // single-contribution leaf writes are DEFERRED and issued as ONE
+  // multi-tensor copy at the end of the sweep (a ~200-param model
+  // otherwise pays ~200 tiny copy launches per step); a second
+  // contribution arriving later flushes that leaf's copy immediately so
+  // the following _grad_add lands on a written buffer
+  struct Pending { NDArray src, dst; bool flushed = false; };
+  std::vector<Pending> pending;
+  std::unordered_map<NDArray::Chunk*, size_t> pending_idx;
+  auto deferrable = [](const NDArray& src, const NDArray& dst) {
+    return src.ctx() == dst.ctx() &&
+           multi_copy_mode(src.dtype(), dst.dtype()) >= 0;
+  };
 
   // forward pass over the tape: which chunks require grad at all
   // (leaves plus anything computed from them) — reference OpReqType
@@ -293,8 +312,19 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
         // (write req overwrites on the first one; add req accumulates)
         bool first = leaf_written.insert(key).second;
         if (first && lit->second.req != 2) {
-          RunInto(copy_into_op, {}, {igrads[i]}, {lit->second.grad});
+          if (deferrable(igrads[i], lit->second.grad)) {
+            pending_idx[key] = pending.size();
+            pending.push_back({igrads[i], lit->second.grad});
+          } else {
+            RunInto(copy_into_op, {}, {igrads[i]}, {lit->second.grad});
+          }
         } else {
+          auto pit = pending_idx.find(key);
+          if (pit != pending_idx.end() && !pending[pit->second].flushed) {
+            Pending& pd = pending[pit->second];
+            RunInto(copy_into_op, {}, {pd.src}, {pd.dst});
+            pd.flushed = true;
+          }
           RunInto(add_into, {}, {igrads[i]}, {lit->second.grad});
         }
         continue;
@@ -312,15 +342,31 @@ void Imperative::Backward(const std::vector<NDArray>& ys,
   }
 
   // write accumulated grads into attached leaf buffers
-  OpEntry* copy_op = OpRegistry::Get()->Find("_copy_into");
   for (auto& kv : leaves_) {
     auto git = grads.find(kv.first);
     if (git == grads.end()) continue;
     if (kv.second.req == 2) {  // add
       RunInto(add_into, {}, {git->second}, {kv.second.grad});
+    } else if (deferrable(git->second, kv.second.grad)) {
+      pending.push_back({git->second, kv.second.grad});
     } else {
-      RunInto(copy_op, {}, {git->second}, {kv.second.grad});
+      RunInto(copy_into_op, {}, {git->second}, {kv.second.grad});
     }
+  }
+  // flush the deferred leaf writes: one multi-tensor copy per device
+  std::map<std::pair<int, int>, std::vector<size_t>> by_dev;
+  for (size_t i = 0; i < pending.size(); ++i) {
+    if (pending[i].flushed) continue;
+    Context c = pending[i].dst.ctx();
+    by_dev[{c.dev_type, c.dev_id}].push_back(i);
+  }
+  for (auto& kv : by_dev) {
+    std::vector<NDArray> srcs, dsts;
+    for (size_t i : kv.second) {
+      srcs.push_back(pending[i].src);
+      dsts.push_back(pending[i].dst);
+    }
+    RunInto(multi_copy_op, {}, srcs, dsts);
   }
   if (!retain_graph) tape_.clear();
 }

@@ -1534,6 +1534,38 @@ bool _registered_nn = [] {
         });
       });
 
+  // one-launch multi-tensor copy (tape leaf-grad batching); outputs are
+  // given by RunInto so no inference runs — pure fcompute
+  RegN("_multi_copy").in(-1)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        *os = is;
+        *ot = it;
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        std::vector<Arr> srcs(in.begin(), in.end());
+        std::vector<Arr> dsts(out.begin(), out.end());
+        multi_copy_raw(LC(o), srcs, dsts);
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        for (size_t i = 0; i < in.size(); ++i) {
+          if (in[i].dtype == out[i].dtype) {
+            memcpy(out[i].dptr, in[i].dptr,
+                   (size_t)out[i].size() * dtype_size(out[i].dtype));
+          } else {
+            MX_CHECK(in[i].dtype == kFloat32,
+                     "_multi_copy cpu: f32 source casts only");
+            const float* s = (const float*)in[i].dptr;
+            // CPU leaves are fp32 in practice; cast path kept for parity
+            MXC_DISPATCH_FLOAT(out[i].dtype, "_multi_copy", {
+              auto* d = (scalar_t*)out[i].dptr;
+              for (long e = 0; e < out[i].size(); ++e) d[e] = (scalar_t)s[e];
+            });
+          }
+        }
+      });
+
   // ======================= attention / lstm cell =========================
   // in: qkv [B,S,3U] (+ optional uint8 mask [B*H,S,S] riding the fused
   // softmax's mask slot); out: y, saved pre-dropout att probs, dropout

@@ -113,6 +113,9 @@ void multi_sgd_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
                           const std::vector<double>& wds, double mu,
                           double rescale, double clip);
 // result written into finite_out (int32[1], 1 = all finite)
+void multi_copy_raw(const LaunchCtx& lc, const std::vector<Arr>& srcs,
+                    const std::vector<Arr>& dsts);
+int multi_copy_mode(int src_dtype, int dst_dtype);
 void multi_all_finite_raw(const LaunchCtx& lc, const std::vector<Arr>& ts,
                           const Arr& finite_out);
 void lstm_cell_fwd_raw(const LaunchCtx& lc, const Arr& gates,
