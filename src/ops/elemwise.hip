@@ -503,17 +503,24 @@ struct CopyChunk {
                  // 4 bf16->f32
 };
 
-__global__ void multi_copy_kernel(const CopyChunk* __restrict__ chunks,
-                                  int nchunks, long total_units) {
+// chunks ride the kernarg segment (value struct, max kMCPerLaunch) —
+// no device-side table, no H2D staging, safe under hipGraph capture
+constexpr int kMCPerLaunch = 64;
+struct CopyArgs {
+  CopyChunk c[kMCPerLaunch];
+};
+
+__global__ void multi_copy_kernel(const CopyArgs args, int nchunks,
+                                  long total_units) {
   for (long u = (long)blockIdx.x * blockDim.x + threadIdx.x;
        u < total_units; u += (long)gridDim.x * blockDim.x) {
     int lo = 0, hi = nchunks - 1;
     while (lo < hi) {
       int mid = (lo + hi + 1) >> 1;
-      if (chunks[mid].start <= u) lo = mid;
+      if (args.c[mid].start <= u) lo = mid;
       else hi = mid - 1;
     }
-    const CopyChunk c = chunks[lo];
+    const CopyChunk c = args.c[lo];
     long cu = u - c.start;
     if (cu >= c.units) continue;
     if (c.mode == 0) {
@@ -554,29 +561,32 @@ int multi_copy_mode(int src_dtype, int dst_dtype) {
 void multi_copy_raw(const LaunchCtx& lc, const std::vector<Arr>& srcs,
                     const std::vector<Arr>& dsts) {
   int n = (int)srcs.size();
-  if (n == 0) return;
-  std::vector<CopyChunk> host(n);
-  long total = 0;
-  for (int i = 0; i < n; ++i) {
-    int mode = multi_copy_mode(srcs[i].dtype, dsts[i].dtype);
-    MX_CHECK(mode >= 0, "multi_copy: unsupported cast "
-                            << srcs[i].dtype << "->" << dsts[i].dtype);
-    host[i].src = (const char*)srcs[i].ptr;
-    host[i].dst = (char*)dsts[i].ptr;
-    host[i].mode = mode;
-    long nel = dsts[i].numel();
-    long dbytes = nel * dtype_size(dsts[i].dtype);
-    host[i].n = mode == 0 ? dbytes : nel;
-    host[i].units = (dbytes + 15) / 16;
-    host[i].start = total;
-    total += host[i].units;
+  for (int base = 0; base < n; base += kMCPerLaunch) {
+    int cnt = std::min(n - base, kMCPerLaunch);
+    CopyArgs args{};
+    long total = 0;
+    for (int i = 0; i < cnt; ++i) {
+      const Arr& src = srcs[base + i];
+      const Arr& dst = dsts[base + i];
+      int mode = multi_copy_mode(src.dtype, dst.dtype);
+      MX_CHECK(mode >= 0, "multi_copy: unsupported cast "
+                              << src.dtype << "->" << dst.dtype);
+      CopyChunk& c = args.c[i];
+      c.src = (const char*)src.ptr;
+      c.dst = (char*)dst.ptr;
+      c.mode = mode;
+      long nel = dst.numel();
+      long dbytes = nel * dtype_size(dst.dtype);
+      c.n = mode == 0 ? dbytes : nel;
+      c.units = (dbytes + 15) / 16;
+      c.start = total;
+      total += c.units;
+    }
+    if (total == 0) continue;
+    multi_copy_kernel<<<ew_grid_n(total), 256, 0, lc.stream>>>(args, cnt,
+                                                               total);
+    HIP_CHECK_LAST();
   }
-  void* table = lc.workspace(n * sizeof(CopyChunk));
-  MX_HIP_CALL(hipMemcpyAsync(table, host.data(), n * sizeof(CopyChunk),
-                             hipMemcpyHostToDevice, lc.stream));
-  multi_copy_kernel<<<ew_grid_n(total), 256, 0, lc.stream>>>(
-      (const CopyChunk*)table, n, total);
-  HIP_CHECK_LAST();
 }
 
 void multi_sgd_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
