@@ -589,6 +589,88 @@ void multi_copy_raw(const LaunchCtx& lc, const std::vector<Arr>& srcs,
   }
 }
 
+// multi-tensor fused Adam (kernarg chunk table like multi_copy): ONE
+// launch per ~56 parameters replaces the per-param adam_update storm
+// (~160 launches/step on BERT-base).
+struct AdamChunk {
+  void* w;
+  const void* g;
+  float* m;
+  float* v;
+  float* master;
+  long start, len;
+};
+constexpr int kMAPerLaunch = 56;
+struct AdamArgs {
+  AdamChunk c[kMAPerLaunch];
+};
+
+template <typename T>
+__global__ void multi_adam_kernel(const AdamArgs args, int n, long total,
+                                  float lr_t, float b1, float b2, float eps,
+                                  float wd, float rescale, float clip,
+                                  bool adamw) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += (long)gridDim.x * blockDim.x) {
+    int lo = 0, hi = n - 1;
+    while (lo < hi) {
+      int mid = (lo + hi + 1) >> 1;
+      if (args.c[mid].start <= i) lo = mid;
+      else hi = mid - 1;
+    }
+    const AdamChunk c = args.c[lo];
+    long j = i - c.start;
+    if (j >= c.len) continue;
+    T* w = (T*)c.w;
+    const T* g = (const T*)c.g;
+    float wm = c.master ? c.master[j] : (float)w[j];
+    float gv = (float)g[j] * rescale;
+    if (clip > 0.f) gv = fminf(fmaxf(gv, -clip), clip);
+    if (!adamw) gv += wd * wm;
+    float mi = c.m[j] = b1 * c.m[j] + (1.f - b1) * gv;
+    float vi = c.v[j] = b2 * c.v[j] + (1.f - b2) * gv * gv;
+    wm -= lr_t * mi / (sqrtf(vi) + eps);
+    if (adamw) wm -= lr_t * wd * wm;
+    if (c.master) c.master[j] = wm;
+    w[j] = (T)wm;
+  }
+}
+
+void multi_adam_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
+                           const std::vector<Arr>& gs,
+                           const std::vector<Arr>& ms,
+                           const std::vector<Arr>& vs,
+                           const std::vector<Arr>& masters, double lr_t,
+                           double b1, double b2, double eps, double wd,
+                           double rescale, double clip, bool adamw) {
+  int n = (int)ws.size();
+  for (int base = 0; base < n; base += kMAPerLaunch) {
+    int cnt = std::min(n - base, kMAPerLaunch);
+    AdamArgs args{};
+    long total = 0;
+    for (int i = 0; i < cnt; ++i) {
+      AdamChunk& c = args.c[i];
+      c.w = ws[base + i].ptr;
+      c.g = gs[base + i].ptr;
+      c.m = ms[base + i].data<float>();
+      c.v = vs[base + i].data<float>();
+      c.master = masters[base + i].defined() && masters[base + i].numel()
+                     ? masters[base + i].data<float>()
+                     : nullptr;
+      c.start = total;
+      c.len = ws[base + i].numel();
+      total += c.len;
+    }
+    if (total == 0) continue;
+    DISPATCH_FLOAT_NATIVE(ws[base].dtype, "multi_adam", [&] {
+      multi_adam_kernel<scalar_t><<<ew_grid_n(total), 256, 0, lc.stream>>>(
+          args, cnt, total, (float)lr_t, (float)b1, (float)b2, (float)eps,
+          (float)wd, (float)rescale, (float)clip, adamw);
+    });
+    HIP_CHECK_LAST();
+  }
+}
+
 void multi_sgd_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
                           const std::vector<Arr>& masters,
                           const std::vector<Arr>& grads,

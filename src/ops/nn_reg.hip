@@ -1534,6 +1534,70 @@ bool _registered_nn = [] {
         });
       });
 
+  // multi-tensor fused Adam: in = n grads; out = [w0,m0,v0, w1,m1,v1,
+  // ...] then (has_master) the n fp32 masters appended at the tail.
+  RegN("multi_adam_update").in(-1)
+      .infer([](const NodeAttrs&, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        *os = is;
+        *ot = it;
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        int n = (int)in.size();
+        bool hm = a.GetBool("has_master", false);
+        MX_CHECK((int)out.size() == 3 * n + (hm ? n : 0),
+                 "multi_adam_update: output layout mismatch");
+        std::vector<Arr> ws, gs, ms, vs, masters;
+        for (int i = 0; i < n; ++i) {
+          gs.push_back(in[i]);
+          ws.push_back(out[3 * i]);
+          ms.push_back(out[3 * i + 1]);
+          vs.push_back(out[3 * i + 2]);
+          masters.push_back(hm ? Arr(out[3 * n + i]) : Arr());
+        }
+        multi_adam_update_raw(
+            LC(o), ws, gs, ms, vs, masters, a.GetFloat("lr_t", 0.001),
+            a.GetFloat("beta1", 0.9), a.GetFloat("beta2", 0.999),
+            a.GetFloat("eps", 1e-8), a.GetFloat("wd", 0.0),
+            a.GetFloat("rescale_grad", 1.0),
+            a.GetFloat("clip_gradient", 0.0), a.GetBool("adamw", false));
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        int n = (int)in.size();
+        bool hm = a.GetBool("has_master", false);
+        float lr_t = (float)a.GetFloat("lr_t", 0.001);
+        float b1 = (float)a.GetFloat("beta1", 0.9);
+        float b2 = (float)a.GetFloat("beta2", 0.999);
+        float eps = (float)a.GetFloat("eps", 1e-8);
+        float wd = (float)a.GetFloat("wd", 0.0);
+        float rs = (float)a.GetFloat("rescale_grad", 1.0);
+        float clip = (float)a.GetFloat("clip_gradient", 0.0);
+        bool adamw = a.GetBool("adamw", false);
+        for (int t = 0; t < n; ++t) {
+          long len = out[3 * t].size();
+          float* m = (float*)out[3 * t + 1].dptr;
+          float* v = (float*)out[3 * t + 2].dptr;
+          float* master = hm ? (float*)out[3 * n + t].dptr : nullptr;
+          MXC_DISPATCH_FLOAT(out[3 * t].dtype, "multi_adam_update", {
+            auto* w = (scalar_t*)out[3 * t].dptr;
+            auto* g = (const scalar_t*)in[t].dptr;
+            for (long i = 0; i < len; ++i) {
+              float wm = master ? master[i] : (float)w[i];
+              float gv = (float)g[i] * rs;
+              if (clip > 0.f) gv = std::min(std::max(gv, -clip), clip);
+              if (!adamw) gv += wd * wm;
+              float mi = m[i] = b1 * m[i] + (1.f - b1) * gv;
+              float vi = v[i] = b2 * v[i] + (1.f - b2) * gv * gv;
+              wm -= lr_t * mi / (sqrtf(vi) + eps);
+              if (adamw) wm -= lr_t * wd * wm;
+              if (master) master[i] = wm;
+              w[i] = (scalar_t)wm;
+            }
+          });
+        }
+      });
+
   // one-launch multi-tensor copy (tape leaf-grad batching); outputs are
   // given by RunInto so no inference runs — pure fcompute
   RegN("_multi_copy").in(-1)

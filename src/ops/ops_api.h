@@ -113,6 +113,13 @@ void multi_sgd_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
                           const std::vector<double>& wds, double mu,
                           double rescale, double clip);
 // result written into finite_out (int32[1], 1 = all finite)
+void multi_adam_update_raw(const LaunchCtx& lc, const std::vector<Arr>& ws,
+                           const std::vector<Arr>& gs,
+                           const std::vector<Arr>& ms,
+                           const std::vector<Arr>& vs,
+                           const std::vector<Arr>& masters, double lr_t,
+                           double b1, double b2, double eps, double wd,
+                           double rescale, double clip, bool adamw);
 void multi_copy_raw(const LaunchCtx& lc, const std::vector<Arr>& srcs,
                     const std::vector<Arr>& dsts);
 int multi_copy_mode(int src_dtype, int dst_dtype);

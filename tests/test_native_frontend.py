@@ -234,3 +234,34 @@ def test_native_bert_adam_trains(native):
         tr.step(2)
         losses.append(loss.asscalar())
     assert losses[-1] < losses[0]
+
+
+def test_native_fused_adam_matches_reference(native):
+    """multi_adam_update (one kernarg-batched launch for all params)
+    must produce the same update as the per-parameter reference rule."""
+    from mxnet_amd.gluon.parameter import Parameter
+    from mxnet_amd.gluon import Trainer
+    rs = np.random.RandomState(7)
+    params = []
+    refs = []
+    for i, shape in enumerate([(5, 3), (7,), (2, 2, 2)]):
+        w = rs.randn(*shape).astype('float32')
+        g = rs.randn(*shape).astype('float32')
+        p = Parameter(f'p{i}', shape=shape)
+        p.initialize(ctx=mx.cpu())
+        p.set_data(mx.nd.array(w))
+        mx.nd.array(g).copyto(p.list_grad()[0])
+        params.append(p)
+        refs.append((w, g))
+    tr = Trainer(params, 'adam', {'learning_rate': 0.01}, kvstore=None)
+    tr._optimizer.rescale_grad = 1.0
+    tr._update(False)
+    mx.nd.waitall()
+    b1, b2, eps, lr = 0.9, 0.999, 1e-8, 0.01
+    lr_t = lr * np.sqrt(1 - b2) / (1 - b1)
+    for p, (w, g) in zip(params, refs):
+        m = (1 - b1) * g
+        v = (1 - b2) * g * g
+        exp = w - lr_t * m / (np.sqrt(v) + eps)
+        np.testing.assert_allclose(p.data(mx.cpu()).asnumpy(), exp,
+                                   rtol=1e-5, atol=1e-6)
