@@ -385,14 +385,16 @@ class Trainer:
             return False
         import math as _m
         from .. import _core
-        groups = {}
+        # pass 1: validate with NO side effects (a mid-loop fallback must
+        # not leave some update counts bumped twice)
         lr0 = wd0 = None
+        active = []
         for i, p in enumerate(self._params):
             if p.grad_req == 'null':
                 continue
             self._check_states(i, p)
-            if self._pop_rowsparse(p) is not None:
-                return False
+            if getattr(p, 'grad_stype', 'default') != 'default':
+                return False  # sparse grads: per-param lazy update path
             datas = p.list_data()
             if len(datas) != 1 or not datas[0].is_native:
                 return False
@@ -401,6 +403,11 @@ class Trainer:
                 lr0, wd0 = lr, wd
             elif lr != lr0 or wd != wd0:
                 return False  # per-param schedule: per-param kernels
+            active.append((i, p))
+        # pass 2: count + group
+        groups = {}
+        for i, p in active:
+            datas = p.list_data()
             opt._update_count(i)
             st = self._states[i]
             if isinstance(st, tuple) and len(st) == 2 and                     isinstance(st[1], tuple):
