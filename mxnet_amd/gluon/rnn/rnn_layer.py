@@ -57,6 +57,8 @@ class _RNNLayer(HybridBlock):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
         params = self.parameters.data(ctx)
+        if getattr(x, 'is_native', False):
+            return self._forward_native(x, states, params, ctx)
         t = x._t
         if self._layout == 'NTC':
             t = t.transpose(0, 1).contiguous()
@@ -83,6 +85,35 @@ class _RNNLayer(HybridBlock):
         new_states = [NDArray(hn)]
         if cn is not None:
             new_states.append(NDArray(cn))
+        return out, new_states
+
+    def _forward_native(self, x, states, params, ctx):
+        """Native-runtime RNN: same fused-RNN semantics composed from
+        native registry ops (ops/rnn.py rnn_forward_native) — backward
+        runs on the own tape, no torch involved."""
+        from ... import autograd as _ag
+        t = x
+        if self._layout == 'NTC':
+            t = t.transpose((1, 0, 2))
+        N = t.shape[1]
+        return_states = states is not None
+        if states is None:
+            states = self.begin_state(N, ctx=ctx, dtype=str(x.dtype))
+        if isinstance(states, NDArray):
+            states = [states]
+        out, hn, cn = _rnn_ops.rnn_forward_native(
+            t, params, states[0],
+            states[1] if len(states) > 1 else None,
+            self._mode, self._hidden_size, self._num_layers,
+            self._bidirectional, self._dropout,
+            training=_ag.is_training())
+        if self._layout == 'NTC':
+            out = out.transpose((1, 0, 2))
+        if not return_states:
+            return out
+        new_states = [hn]
+        if cn is not None:
+            new_states.append(cn)
         return out, new_states
 
     def __repr__(self):

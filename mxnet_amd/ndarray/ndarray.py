@@ -715,12 +715,17 @@ def ones_like(a):
 def concat(*arys, dim=1):
     if len(arys) == 1 and isinstance(arys[0], (list, tuple)):
         arys = arys[0]
+    if arys and arys[0].is_native:
+        return arys[0]._invoke('concat', list(arys), {'dim': str(dim)})
     return NDArray(torch.cat([a._t for a in arys], dim=dim))
 
 
 def stack(*arys, axis=0):
     if len(arys) == 1 and isinstance(arys[0], (list, tuple)):
         arys = arys[0]
+    if arys and arys[0].is_native:
+        parts = [a.expand_dims(axis) for a in arys]
+        return parts[0]._invoke('concat', parts, {'dim': str(axis)})
     return NDArray(torch.stack([a._t for a in arys], dim=axis))
 
 

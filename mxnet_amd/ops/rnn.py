@@ -153,3 +153,134 @@ def rnn_ndarray(data, parameters, state, state_cell, mode, state_size,
     if mode == 'lstm':
         return NDArray(out), NDArray(hn), NDArray(cn)
     return NDArray(out), NDArray(hn)
+
+
+# ---------------------------------------------------------------------------
+# native-runtime path: the same fused-RNN semantics composed from native
+# registry ops (FC GEMMs + elementwise + strided slices); every step is a
+# tape node so backward comes from the own autograd, not torch.
+# ---------------------------------------------------------------------------
+def _n_slice_params(params, mode, input_size, hidden_size, num_layers,
+                    bidirectional=False):
+    """slice_params over a native flat vector: 1-D recorded slices (the
+    scatter backward accumulates straight into the flat leaf)."""
+    ng = _gate_count(mode)
+    dirs = 2 if bidirectional else 1
+    shapes = []
+    for layer in range(num_layers):
+        for d in range(dirs):
+            isz = input_size if layer == 0 else hidden_size * dirs
+            shapes.append(('wi', layer, d, (ng * hidden_size, isz)))
+            shapes.append(('wh', layer, d, (ng * hidden_size, hidden_size)))
+    for layer in range(num_layers):
+        for d in range(dirs):
+            shapes.append(('bi', layer, d, (ng * hidden_size,)))
+            shapes.append(('bh', layer, d, (ng * hidden_size,)))
+    out = {}
+    off = 0
+    for kind, layer, d, shp in shapes:
+        n = 1
+        for s in shp:
+            n *= s
+        out[(kind, layer, d)] = params[off:off + n].reshape(shp)
+        off += n
+    assert off == params.size
+    return out
+
+
+def _n_flip0(x):
+    """Reverse along axis 0 as one strided gather (negative stride)."""
+    T = x.shape[0]
+    inner = 1
+    for s in x.shape[1:]:
+        inner *= s
+    strides = [-inner] + [1] * 0
+    # plan: [T, inner] over the flattened tail
+    return x._invoke('_strided_copy', [x], {
+        'shape': '(%d,%d,)' % (T, inner),
+        'strides': '(%d,1,)' % (-inner),
+        'offset': str((T - 1) * inner),
+        'oshape': '(' + ','.join(str(s) for s in x.shape) + ',)'})
+
+
+def _n_concat0(parts):
+    """Stack [N,H] steps into [T,N,H] (one concat + reshape)."""
+    first = parts[0]
+    y = first._invoke('concat', parts, {'dim': '0'})
+    return y.reshape((len(parts),) + tuple(first.shape))
+
+
+def rnn_forward_native(x, params, h0, c0, mode, hidden_size, num_layers,
+                       bidirectional=False, dropout=0.0, training=False):
+    """Native rnn_forward: x [T,N,I] native NDArray; returns
+    (out, hn, cn) native NDArrays (reference rnn-inl.h semantics)."""
+    from ..ndarray import ops as F
+    T, N, I = x.shape
+    H = hidden_size
+    dirs = 2 if bidirectional else 1
+    p = _n_slice_params(params, mode, I, H, num_layers, bidirectional)
+    h0 = h0.reshape(num_layers * dirs, N, H)
+    if mode == 'lstm':
+        c0 = c0.reshape(num_layers * dirs, N, H)
+
+    def sig(z):
+        return F.Activation(z, act_type='sigmoid')
+
+    def tnh(z):
+        return F.Activation(z, act_type='tanh')
+
+    hs, cs = [], []
+    inp = x
+    for layer in range(num_layers):
+        outs_dir = []
+        for d in range(dirs):
+            wi, wh = p[('wi', layer, d)], p[('wh', layer, d)]
+            bi, bh = p[('bi', layer, d)], p[('bh', layer, d)]
+            idx = layer * dirs + d
+            h = h0[idx]
+            c = c0[idx] if mode == 'lstm' else None
+            seq = inp if d == 0 else _n_flip0(inp)
+            xg = F.FullyConnected(seq.reshape(T * N, -1), wi, bi,
+                                  flatten=False).reshape(T, N, -1)
+            outs = []
+            for t in range(T):
+                if mode == 'lstm':
+                    gates = xg[t] + F.FullyConnected(h, wh, bh,
+                                                     flatten=False)
+                    gi = sig(gates[:, 0:H])
+                    gf = sig(gates[:, H:2 * H])
+                    gg = tnh(gates[:, 2 * H:3 * H])
+                    go = sig(gates[:, 3 * H:4 * H])
+                    c = gf * c + gi * gg
+                    h = go * tnh(c)
+                elif mode == 'gru':
+                    # reference gru gate math: r,z from x+h parts; the
+                    # reset gate scales the h-part of the n gate
+                    hg = F.FullyConnected(h, wh, bh, flatten=False)
+                    xt = xg[t]
+                    r = sig(xt[:, 0:H] + hg[:, 0:H])
+                    z = sig(xt[:, H:2 * H] + hg[:, H:2 * H])
+                    n_ = tnh(xt[:, 2 * H:3 * H] + r * hg[:, 2 * H:3 * H])
+                    h = (z * -1.0 + 1.0) * n_ + z * h
+                else:
+                    gates = xg[t] + F.FullyConnected(h, wh, bh,
+                                                     flatten=False)
+                    act = 'relu' if mode == 'rnn_relu' else 'tanh'
+                    h = F.Activation(gates, act_type=act)
+                outs.append(h)
+            out = _n_concat0(outs)
+            if d == 1:
+                out = _n_flip0(out)
+            outs_dir.append(out)
+            hs.append(h)
+            if mode == 'lstm':
+                cs.append(c)
+        if dirs > 1:
+            inp = outs_dir[0]._invoke('concat', outs_dir, {'dim': '2'})
+        else:
+            inp = outs_dir[0]
+        if dropout > 0 and training and layer < num_layers - 1:
+            inp = F.Dropout(inp, p=dropout)
+    hn = _n_concat0(hs)
+    cn = _n_concat0(cs) if mode == 'lstm' else None
+    return inp, hn, cn

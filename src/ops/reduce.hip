@@ -578,6 +578,106 @@ bool _registered_reduce = [] {
         });
       });
 
+  // concat along an axis (reference concat.cc): forward scatters each
+  // input into its strided slab of the output (scatter_strided, one
+  // launch per input, full cover so no memset); backward gathers the
+  // matching slices of the output gradient (_strided_copy).
+  Reg2("concat").in(-1)
+      .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                const std::vector<int>& it, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        int nd = (int)is[0].size();
+        int axis = (int)a.GetInt("dim", 0);
+        if (axis < 0) axis += nd;
+        TShape out = is[0];
+        for (size_t i = 1; i < is.size(); ++i) {
+          MX_CHECK((int)is[i].size() == nd, "concat: rank mismatch");
+          out[axis] += is[i][axis];
+        }
+        os->assign(1, out);
+        ot->assign(1, it[0]);
+      })
+      .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+        int nd = out[0].ndim();
+        int axis = (int)a.GetInt("dim", 0);
+        if (axis < 0) axis += nd;
+        std::vector<int64_t> ostr(nd, 1);
+        for (int d = nd - 2; d >= 0; --d)
+          ostr[d] = ostr[d + 1] * out[0].shape[d + 1];
+        long off_axis = 0;
+        for (auto& x : in) {
+          Strides8 st;
+          st.ndim = nd;
+          for (int d = 0; d < nd; ++d) {
+            st.shape[d] = x.shape[d];
+            st.s0[d] = ostr[d];
+          }
+          long n = x.size();
+          long off = off_axis * ostr[axis];
+          MXC_DISPATCH_ALL(out[0].dtype, "concat", {
+            scatter_strided_kernel<scalar_t><<<grid_for(n), kBlock, 0,
+                                               o.rc.stream>>>(
+                (const scalar_t*)x.dptr, (scalar_t*)out[0].dptr, n, st,
+                off);
+          });
+          off_axis += x.shape[axis];
+        }
+        HIP_CHECK_LAST();
+      })
+      .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
+        int nd = out[0].ndim();
+        int axis = (int)a.GetInt("dim", 0);
+        if (axis < 0) axis += nd;
+        std::vector<int64_t> ostr(nd, 1);
+        for (int d = nd - 2; d >= 0; --d)
+          ostr[d] = ostr[d + 1] * out[0].shape[d + 1];
+        long off_axis = 0;
+        for (auto& x : in) {
+          long n = x.size();
+          long off0 = off_axis * ostr[axis];
+          MXC_DISPATCH_ALL(out[0].dtype, "concat", {
+            auto* src = (const scalar_t*)x.dptr;
+            auto* dst = (scalar_t*)out[0].dptr;
+            for (long i = 0; i < n; ++i) {
+              long rem = i, off = off0;
+              for (int d = nd - 1; d >= 0; --d) {
+                long idx = rem % x.shape[d];
+                rem /= x.shape[d];
+                off += idx * ostr[d];
+              }
+              dst[off] = src[i];
+            }
+          });
+          off_axis += x.shape[axis];
+        }
+      })
+      .bwd([](const TapeNode& n, const std::vector<NDArray>& og)
+               -> std::vector<NDArray> {
+        int nd = (int)og[0].shape().size();
+        int axis = (int)n.attrs.GetInt("dim", 0);
+        if (axis < 0) axis += nd;
+        std::vector<int64_t> ostr(nd, 1);
+        for (int d = nd - 2; d >= 0; --d)
+          ostr[d] = ostr[d + 1] * og[0].shape()[d + 1];
+        std::vector<NDArray> r(n.inputs.size());
+        long off_axis = 0;
+        for (size_t i = 0; i < n.inputs.size(); ++i) {
+          const TShape& xs = n.inputs[i].shape();
+          NodeAttrs a;
+          std::string sh = "(", strd = "(";
+          for (int d = 0; d < nd; ++d) {
+            sh += std::to_string(xs[d]) + ",";
+            strd += std::to_string(ostr[d]) + ",";
+          }
+          a.d["shape"] = sh + ")";
+          a.d["strides"] = strd + ")";
+          a.d["offset"] = std::to_string(off_axis * ostr[axis]);
+          r[i] = RunOp2("_strided_copy", a, {og[0]});
+          off_axis += xs[axis];
+        }
+        return r;
+      });
+
   // transpose (general permute)
   auto infer_perm = [](const NodeAttrs& a, const std::vector<TShape>& is,
                        const std::vector<int>& it, std::vector<TShape>* os,
