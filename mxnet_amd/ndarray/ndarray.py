@@ -123,6 +123,8 @@ class NDArray:
 
     @property
     def T(self):
+        if self._h is not None:
+            return self.transpose()
         return NDArray(self._t.t().contiguous()) if self._t.dim() == 2 \
             else NDArray(self._t.permute(*reversed(range(self._t.dim()))).contiguous())
 
@@ -321,6 +323,15 @@ class NDArray:
         return NDArray(self._t.transpose(a, b).contiguous())
 
     def split(self, num_outputs, axis=0):
+        if self._h is not None:
+            n = self.shape[axis]
+            step = (n + num_outputs - 1) // num_outputs
+            outs = []
+            for b in range(0, n, step):
+                sl = [slice(None)] * self.ndim
+                sl[axis] = slice(b, min(b + step, n))
+                outs.append(self[tuple(sl)])
+            return outs
         outs = torch.chunk(self._t, num_outputs, dim=axis)
         return [NDArray(o) for o in outs]
 
@@ -396,6 +407,27 @@ class NDArray:
         return NDArray(out)
 
     def __setitem__(self, key, value):
+        if self._h is not None:
+            # basic full-slice / scalar fill on native arrays (used by
+            # init code paths); partial writes need the torch frontend
+            if (key is None or key == slice(None) or
+                    (isinstance(key, tuple)
+                     and all(k == slice(None) for k in key))):
+                if isinstance(value, (int, float)):
+                    from .. import _core
+                    _core.invoke_into(
+                        '_full', [], [self._h],
+                        {'value': str(float(value)),
+                         'shape': '(' + ','.join(
+                             str(d) for d in self.shape) + ',)',
+                         'dtype': str(self._h.dtype)})
+                    return
+                if isinstance(value, NDArray) and value.is_native:
+                    from .. import _core
+                    _core.invoke_into('_copy_into', [value._h], [self._h], {})
+                    return
+            raise TypeError('native NDArray supports only full-slice '
+                            'assignment (x[:] = v)')
         key = NDArray._unwrap_index(key)
         with torch.no_grad():
             if isinstance(value, NDArray):
