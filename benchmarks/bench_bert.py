@@ -191,6 +191,13 @@ def main():
         t = torch.tensor([dt], dtype=torch.float64, device=dev)
         dist.all_reduce(t, op=dist.ReduceOp.MAX)
         dt = float(t.item())
+    # post-bench numerics check: one eager step, loss must be finite
+    with autograd.record():
+        _, _, mlm_f, nsp_f = net(tokens, types, mask)
+        Lf = loss_fn(mlm_f, nsp_f)
+    import math as _math
+    final_loss = float(Lf.asnumpy())
+    assert _math.isfinite(final_loss), f'non-finite loss: {final_loss}'
     if rank == 0:
         print(json.dumps({
             'metric': 'samples/sec BERT-base fp16 seq128 (whole node)',
@@ -199,6 +206,7 @@ def main():
             'warmup': args.warmup, 'ms_per_step': round(dt / args.steps * 1e3, 3),
             'higher_is_better': True, 'scaling': 'weak', 'vs_baseline': None,
             'dtype': dtype, 'data': 'synthetic', 'runtime': args.runtime,
+            'final_loss': round(final_loss, 4),
             'config': {'model': 'bert_base', 'global_batch': B * world,
                        'seq_len': S, 'parallelism': f'dp{world}'}}))
 
