@@ -153,3 +153,44 @@ class ImageRecordIter(DataIter):
         """(got, uint8 ndarray [B,H,W,3], float32 labels) — zero-copy for
         pipelines that upload/normalize on the GPU."""
         return self._it.next_batch(self.batch_size, self._h, self._w)
+
+
+class CSVIter(DataIter):
+    """CSV file iterator (reference CSVIter, src/io/iter_csv.cc):
+    C++ parse of data (and optional label) CSV files into float batches
+    via mxnet_amd._imageio.CsvIter."""
+
+    def __init__(self, data_csv, data_shape, label_csv=None,
+                 label_shape=(1,), batch_size=1, round_batch=True,
+                 **kwargs):
+        super().__init__(batch_size)
+        from .. import _imageio
+        self._dshape = tuple(int(d) for d in data_shape)
+        self._lshape = tuple(int(d) for d in label_shape) or (1,)
+        dw = 1
+        for d in self._dshape:
+            dw *= d
+        lw = 1
+        for d in self._lshape:
+            lw *= d
+        self._dw, self._lw = dw, lw
+        self._it = _imageio.CsvIter(data_csv, label_csv or '',
+                                    batch_size, dw, lw)
+
+    @property
+    def num_records(self):
+        return self._it.size
+
+    def reset(self):
+        self._it.reset()
+
+    def next(self):
+        from ..ndarray import ndarray as nd
+        got, data, labels = self._it.next_batch(self.batch_size,
+                                                self._dw, self._lw)
+        if got == 0:
+            raise StopIteration
+        data = data[:got].reshape((got,) + self._dshape)
+        labels = labels[:got].reshape((got,) + self._lshape)
+        return DataBatch(data=[nd.array(data)], label=[nd.array(labels)],
+                         pad=self.batch_size - got)

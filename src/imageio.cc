@@ -21,6 +21,7 @@
 #include <cmath>
 #include <cstring>
 #include <fstream>
+#include <cstdlib>
 #include <random>
 #include <stdexcept>
 #include <string>
@@ -566,6 +567,78 @@ class ImageRecordIter {
   size_t cursor_ = 0;
 };
 
+// ---------------------------------------------------------------------------
+// CSVIter (reference src/io/iter_csv.cc): C++ CSV parse into float
+// row-batches; optional separate label file, round-robin batching.
+// ---------------------------------------------------------------------------
+class CsvIter {
+ public:
+  CsvIter(const std::string& data_path, const std::string& label_path,
+          int batch, int row_width, int label_width)
+      : batch_(batch), dwidth_(row_width), lwidth_(label_width) {
+    ParseFile(data_path, dwidth_, &data_);
+    if (!label_path.empty()) {
+      ParseFile(label_path, lwidth_, &labels_);
+      if (data_.size() / dwidth_ != labels_.size() / lwidth_)
+        throw std::runtime_error("CSVIter: data/label row count mismatch");
+    }
+    nrows_ = data_.size() / dwidth_;
+  }
+
+  size_t size() const { return nrows_; }
+  void Reset() { cursor_ = 0; }
+
+  // fills caller buffers; returns rows delivered (< batch at EOF)
+  int NextInto(float* data_out, float* label_out) {
+    int got = 0;
+    while (got < batch_ && cursor_ < nrows_) {
+      std::memcpy(data_out + (size_t)got * dwidth_,
+                  data_.data() + cursor_ * dwidth_, dwidth_ * 4);
+      if (label_out) {
+        if (!labels_.empty())
+          std::memcpy(label_out + (size_t)got * lwidth_,
+                      labels_.data() + cursor_ * lwidth_, lwidth_ * 4);
+        else
+          label_out[(size_t)got * lwidth_] = 0.f;
+      }
+      ++cursor_;
+      ++got;
+    }
+    return got;
+  }
+
+ private:
+  static void ParseFile(const std::string& path, int width,
+                        std::vector<float>* out) {
+    std::ifstream f(path);
+    if (!f.good())
+      throw std::runtime_error("CSVIter: cannot open " + path);
+    std::string line;
+    while (std::getline(f, line)) {
+      if (line.empty()) continue;
+      const char* p = line.c_str();
+      int n = 0;
+      while (*p && n < width) {
+        char* end = nullptr;
+        float v = std::strtof(p, &end);
+        if (end == p) break;
+        out->push_back(v);
+        ++n;
+        p = end;
+        while (*p == ',' || *p == ' ' || *p == '\t') ++p;
+      }
+      if (n != width)
+        throw std::runtime_error("CSVIter: row has " + std::to_string(n) +
+                                 " fields, expected " +
+                                 std::to_string(width));
+    }
+  }
+
+  int batch_, dwidth_, lwidth_;
+  size_t nrows_ = 0, cursor_ = 0;
+  std::vector<float> data_, labels_;
+};
+
 }  // namespace
 
 PYBIND11_MODULE(_imageio, m) {
@@ -611,6 +684,21 @@ PYBIND11_MODULE(_imageio, m) {
           py::gil_scoped_release rel;
           got = it.NextInto(data.mutable_data(), labels.mutable_data());
         }
+        return py::make_tuple(got, data, labels);
+      });
+
+  py::class_<CsvIter>(m, "CsvIter")
+      .def(py::init<const std::string&, const std::string&, int, int, int>(),
+           py::arg("data_path"), py::arg("label_path") = "",
+           py::arg("batch_size") = 1, py::arg("row_width") = 1,
+           py::arg("label_width") = 1)
+      .def_property_readonly("size", &CsvIter::size)
+      .def("reset", &CsvIter::Reset)
+      .def("next_batch", [](CsvIter& it, int batch, int dwidth, int lwidth) {
+        py::array_t<float> data({(py::ssize_t)batch, (py::ssize_t)dwidth});
+        py::array_t<float> labels({(py::ssize_t)batch,
+                                   (py::ssize_t)lwidth});
+        int got = it.NextInto(data.mutable_data(), labels.mutable_data());
         return py::make_tuple(got, data, labels);
       });
 }

@@ -277,3 +277,32 @@ def test_record_iter_augment(tmp_path):
     assert got == 1
     # crops come from the source image (every pixel must exist in it)
     assert d1.min() >= items[0][1].min() and d1.max() <= items[0][1].max()
+
+
+def test_csv_iter():
+    import os
+    """CSVIter (reference src/io/iter_csv.cc): C++ CSV parse, separate
+    label file, padding on the last batch, reset."""
+    import tempfile
+    import mxnet_amd as mx
+    from mxnet_amd.io import CSVIter
+    d = tempfile.mkdtemp()
+    data = np.arange(20.).reshape(5, 4)
+    lab = np.arange(5.).reshape(5, 1) * 10
+    np.savetxt(os.path.join(d, 'x.csv'), data, delimiter=',')
+    np.savetxt(os.path.join(d, 'y.csv'), lab, delimiter=',')
+    it = CSVIter(os.path.join(d, 'x.csv'), (2, 2),
+                 label_csv=os.path.join(d, 'y.csv'), batch_size=2)
+    assert it.num_records == 5
+    b = next(it)
+    np.testing.assert_array_equal(b.data[0].asnumpy(),
+                                  data[:2].reshape(2, 2, 2))
+    np.testing.assert_array_equal(b.label[0].asnumpy(), lab[:2])
+    next(it)
+    assert next(it).pad == 1
+    it.reset()
+    assert next(it).pad == 0
+    # no-label variant defaults labels to zero
+    it2 = CSVIter(os.path.join(d, 'x.csv'), (4,), batch_size=5)
+    b = next(it2)
+    np.testing.assert_array_equal(b.label[0].asnumpy(), np.zeros((5, 1)))
