@@ -6,6 +6,7 @@
 #include <unordered_map>
 #include <cassert>
 #include <future>
+#include <new>
 
 namespace mxcore {
 
@@ -114,6 +115,24 @@ struct WorkQueue {
     }
     cv.notify_all();
   }
+  // post-fork: worker threads are gone in the child; drop queued work
+  // and reopen
+  void Reset() {
+    std::lock_guard<std::mutex> g(mu);
+    while (!q.empty()) q.pop();
+    shutdown = false;
+  }
+  // child-side rebuild: the glibc condvar keeps internal waiter counts
+  // for the PARENT's blocked workers — a notify in the child can be
+  // consumed by those ghost slots and the new worker never wakes
+  // (observed as a flaky post-fork hang).  No thread of ours exists in
+  // the child yet, so placement-reinit is safe.
+  void ResetAfterFork() {
+    new (&mu) std::mutex();
+    new (&cv) std::condition_variable();
+    q = {};
+    shutdown = false;
+  }
 };
 
 // profiler record: timed event pair (GPU) or wall ns (CPU)
@@ -149,6 +168,9 @@ struct Engine::Impl {
 
   std::mutex dev_mu_;
   std::unordered_map<int, std::unique_ptr<DeviceWorkers>> devices_;
+
+  std::atomic<bool> forked_{false};
+  void ReinitAfterFork();
 
   Engine* owner_ = nullptr;
 
@@ -194,6 +216,9 @@ struct Engine::Impl {
 
   // ---- dependency tracking (reference threaded_engine.h:120-229) ------
   void Push(Opr* opr) {
+    if (forked_.load(std::memory_order_acquire) &&
+        forked_.exchange(false))
+      ReinitAfterFork();
     bool ready;
     {
       std::lock_guard<std::mutex> g(mu_);
@@ -607,6 +632,71 @@ std::vector<std::tuple<std::string, long, double>> Engine::ProfilerSummary() {
     return std::get<2>(a) > std::get<2>(b);
   });
   return out;
+}
+
+// fork support (reference LibraryInitializer::install_pthread_atfork
+// _handlers): prepare drains every queue so no op is mid-flight and no
+// engine lock is held across fork; the child inherits no threads, so it
+// detaches the dead std::thread handles, drops GPU state (HIP contexts
+// do not survive fork) and restarts the CPU workers lazily.
+// prepare: drain, then take EVERY engine mutex so none is mid-acquire
+// in a worker when fork() snapshots the address space (a worker grabs
+// the queue mutex briefly between Pops — fork landing in that window
+// would leave the child's mutex locked forever).  parent/child
+// handlers release them again; the forking thread is the same thread
+// in the child, so the unlocks are well-defined.
+void Engine::AtForkPrepare() {
+  WaitForAll();
+  impl_->mu_.lock();
+  impl_->prof_mu_.lock();
+  impl_->dev_mu_.lock();
+  impl_->cpu_q_.mu.lock();
+  impl_->cpu_prio_q_.mu.lock();
+  for (auto& kv : impl_->devices_) {
+    kv.second->compute_q.mu.lock();
+    kv.second->copy_q.mu.lock();
+    kv.second->comm_q.mu.lock();
+  }
+}
+
+void Engine::AtForkParent() {
+  for (auto& kv : impl_->devices_) {
+    kv.second->comm_q.mu.unlock();
+    kv.second->copy_q.mu.unlock();
+    kv.second->compute_q.mu.unlock();
+  }
+  impl_->cpu_prio_q_.mu.unlock();
+  impl_->cpu_q_.mu.unlock();
+  impl_->dev_mu_.unlock();
+  impl_->prof_mu_.unlock();
+  impl_->mu_.unlock();
+}
+
+// ONLY sets a flag: the handler also fires on every fork()+exec (python
+// subprocess), and allocating/freeing or starting threads in a freshly
+// forked child can deadlock on malloc locks held by OTHER libraries'
+// threads at fork time.  The real rebuild happens lazily on the child's
+// first engine use.
+void Engine::AtForkChild() {
+  AtForkParent();  // release the mutexes taken by prepare
+  impl_->forked_.store(true, std::memory_order_release);
+}
+
+void Engine::Impl::ReinitAfterFork() {
+  for (auto& t : cpu_workers_)
+    if (t.joinable()) t.detach();  // handles of threads that died in fork
+  cpu_workers_.clear();
+  for (auto& kv : devices_) {
+    if (kv.second->compute_t.joinable()) kv.second->compute_t.detach();
+    if (kv.second->copy_t.joinable()) kv.second->copy_t.detach();
+    if (kv.second->comm_t.joinable()) kv.second->comm_t.detach();
+  }
+  devices_.clear();  // HIP contexts do not survive fork
+  cpu_q_.ResetAfterFork();
+  cpu_prio_q_.ResetAfterFork();
+  new (&done_cv_) std::condition_variable();
+  inflight_ = 0;
+  StartCPUWorkers();
 }
 
 void Engine::StopWorkers() {

@@ -95,7 +95,11 @@ class Engine {
   // name -> (calls, gpu_ms or cpu_ms); drains and resets the buffer
   std::vector<std::tuple<std::string, long, double>> ProfilerSummary();
 
-  void StopWorkers();  // tests / atfork
+  void StopWorkers();
+  // pthread_atfork hooks: drain before fork, rebuild workers in child
+  void AtForkPrepare();
+  void AtForkParent();
+  void AtForkChild();  // tests / atfork
 
  private:
   Engine();

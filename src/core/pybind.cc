@@ -1,6 +1,7 @@
 // mxnet_amd._core — Python bindings for the native runtime:
 // Storage (pooled HIP allocator), ThreadedEngine (HIP streams/events),
 // NDArray (chunk/view), op registry + imperative invoke + autograd tape.
+#include <pthread.h>
 #include <pybind11/numpy.h>
 #include <pybind11/pybind11.h>
 #include <pybind11/stl.h>
@@ -144,6 +145,15 @@ PYBIND11_MODULE(_core, m) {
       }
     }));
   }
+
+  // fork safety (reference LibraryInitializer pthread_atfork): drain
+  // the engine before a fork; the child detaches the dead worker
+  // handles, drops GPU state and restarts CPU workers — so python
+  // multiprocessing (fork start method) keeps working after native ops
+  // ran in the parent
+  pthread_atfork([] { Engine::Get()->AtForkPrepare(); },
+                 [] { Engine::Get()->AtForkParent(); },
+                 [] { Engine::Get()->AtForkChild(); });
 
   py::class_<NDArray>(m, "NDArray")
       .def(py::init([](const std::vector<int64_t>& shape, int dev_type,

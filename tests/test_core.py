@@ -308,3 +308,28 @@ def test_engine_exception_propagates_to_consumers():
         _core.wait_all()
     except RuntimeError:
         pass
+
+
+def test_engine_fork_safety():
+    """pthread_atfork handlers (reference LibraryInitializer): the
+    engine drains before fork, the child rebuilds its CPU workers, and
+    both processes keep computing correct results afterwards."""
+    import os
+    import numpy as np
+    from mxnet_amd import _core
+    a = _core.from_numpy(np.ones((64, 64), np.float32), 1, 0)
+    out = _core.invoke('_mul_scalar', [a], {'alpha': '3.0'})[0]
+    assert abs(out.asnumpy().sum() - 64 * 64 * 3) < 1e-3
+    pid = os.fork()
+    if pid == 0:
+        try:
+            c = _core.from_numpy(np.full((32, 32), 2.0, np.float32), 1, 0)
+            d = _core.invoke('elemwise_add', [c, c], {})[0]
+            ok = abs(d.asnumpy().mean() - 4.0) < 1e-5
+            os._exit(0 if ok else 1)
+        except BaseException:
+            os._exit(2)
+    _, status = os.waitpid(pid, 0)
+    assert os.WEXITSTATUS(status) == 0
+    e = _core.invoke('_plus_scalar', [a], {'alpha': '1.0'})[0]
+    assert abs(e.asnumpy().max() - 2.0) < 1e-5
