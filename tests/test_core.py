@@ -313,8 +313,18 @@ def test_engine_exception_propagates_to_consumers():
 def test_engine_fork_safety():
     """pthread_atfork handlers (reference LibraryInitializer): the
     engine drains before fork, the child rebuilds its CPU workers, and
-    both processes keep computing correct results afterwards."""
+    both processes keep computing correct results afterwards.
+
+    CPU-only processes: once the HIP/HSA runtime is initialized its
+    service threads hold locks across fork and the child deadlocks
+    before any of our code runs (same limitation as CUDA in the
+    reference) — fork+exec (spawn) is the supported path on GPU."""
     import os
+    import torch
+    if torch.cuda.is_available():
+        import pytest
+        pytest.skip('fork-reuse unsupported once HIP is initialized '
+                    '(HSA runtime threads are not fork-safe)')
     import numpy as np
     from mxnet_amd import _core
     a = _core.from_numpy(np.ones((64, 64), np.float32), 1, 0)
