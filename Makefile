@@ -50,3 +50,28 @@ clean:
 	rm -rf build/core $(TARGET)
 
 .PHONY: all clean
+
+# AddressSanitizer build of the pure-C++ host modules (engine, imageio,
+# dataloader) + the test subset that exercises them.  The two deselected
+# tests throw C++ exceptions from dlopen'd modules, which ASAN's
+# __cxa_throw interceptor cannot service under LD_PRELOAD (tool
+# limitation, not a code defect).  See profiles/r02_asan.md.
+PYINC := $(shell python3-config --includes) $(shell python3 -c "import pybind11; print('-I'+pybind11.get_include())")
+ASAN_SO := $(shell g++ -print-file-name=libasan.so)
+asan:
+	mkdir -p build/asan
+	for m in engine imageio dataloader; do \
+	  g++ -O1 -g -fsanitize=address -fno-omit-frame-pointer -shared -fPIC \
+	    -std=c++17 $(PYINC) src/$$m.cc -o build/asan/_$$m.so || exit 1; \
+	  cp mxnet_amd/_$$m.cpython-310-x86_64-linux-gnu.so build/asan/orig_$$m.so; \
+	  cp build/asan/_$$m.so mxnet_amd/_$$m.cpython-310-x86_64-linux-gnu.so; \
+	done
+	LD_PRELOAD=$(ASAN_SO) ASAN_OPTIONS=detect_leaks=0 \
+	  python -m pytest tests/test_engine.py tests/test_imageio.py \
+	  tests/test_aux_components.py -q \
+	  --deselect tests/test_engine.py::test_exception_propagation \
+	  --deselect tests/test_engine.py::test_engine_fork_safety; \
+	rc=$$?; \
+	for m in engine imageio dataloader; do \
+	  cp build/asan/orig_$$m.so mxnet_amd/_$$m.cpython-310-x86_64-linux-gnu.so; \
+	done; exit $$rc
