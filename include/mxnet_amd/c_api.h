@@ -53,6 +53,39 @@ int MXAutogradMarkVariables(int num, NDArrayHandle* vars,
 int MXAutogradBackward(int num_heads, NDArrayHandle* heads,
                        NDArrayHandle* head_grads, int retain_graph);
 
+/* External operator libraries (reference include/mxnet/lib_api.h,
+ * MXLoadLib): a user .so exports
+ *     int mxnet_amd_lib_init(MXRegisterOpFn reg, void* reg_ctx);
+ * and calls `reg(reg_ctx, &op)` once per MXCustomOpDef.  Compute
+ * callbacks receive raw buffers + shapes and, on GPU, the hipStream_t
+ * of the engine's compute stream (NULL on CPU). */
+typedef struct {
+  int ndim;
+  const int64_t* shape;
+  int dtype;          /* DTypeFlag: 0=f32 2=f16 6=i64 ... */
+  void* data;
+} MXTensorView;
+
+typedef int (*MXCustomInferFn)(int n_in, const MXTensorView* ins,
+                               int64_t* out_shape, int* out_ndim,
+                               int* out_dtype);
+typedef int (*MXCustomComputeFn)(int n_in, const MXTensorView* ins,
+                                 MXTensorView* out, void* stream);
+
+typedef struct {
+  const char* name;
+  int n_in;
+  MXCustomInferFn infer;          /* single-output shape/dtype */
+  MXCustomComputeFn fcompute_cpu; /* either may be NULL */
+  MXCustomComputeFn fcompute_gpu;
+} MXCustomOpDef;
+
+typedef void (*MXRegisterOpFn)(void* reg_ctx, const MXCustomOpDef* def);
+
+/* dlopen `path` and register every op it defines; ops become invokable
+ * through MXImperativeInvoke / the python frontend like built-ins */
+int MXLoadLib(const char* path);
+
 /* .params list serialization (reference MXNDArraySave/Load,
  * byte format of SURVEY.md Appendix A) */
 int MXNDArraySave(const char* fname, int num, NDArrayHandle* arrays,

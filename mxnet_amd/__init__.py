@@ -25,6 +25,7 @@ from . import ndarray
 from . import ndarray as nd
 from . import operator
 from . import rtc
+from . import library
 from . import image
 from . import image as img
 from . import numpy as np  # mx.np numpy-compatible namespace

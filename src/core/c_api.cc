@@ -2,6 +2,8 @@
 // include/mxnet_amd/c_api.h; reference src/c_api/c_api.cc).
 #include "../../include/mxnet_amd/c_api.h"
 
+#include <dlfcn.h>
+
 #include <cstring>
 #include <fstream>
 #include <string>
@@ -177,6 +179,95 @@ int MXAutogradBackward(int num_heads, NDArrayHandle* heads,
     if (head_grads && head_grads[i]) gs.push_back(*ND(head_grads[i]));
   }
   Imperative::Get()->Backward(ys, gs, retain_graph != 0);
+  API_END()
+}
+
+// ---------------------------------------------------------------------------
+// external operator libraries (reference lib_api.h / MXLoadLib)
+// ---------------------------------------------------------------------------
+namespace {
+
+std::vector<MXTensorView> ViewsOf(const std::vector<TBlob>& blobs,
+                                  std::vector<std::vector<int64_t>>* keep) {
+  std::vector<MXTensorView> v(blobs.size());
+  keep->resize(blobs.size());
+  for (size_t i = 0; i < blobs.size(); ++i) {
+    (*keep)[i] = blobs[i].shape;
+    v[i].ndim = (int)(*keep)[i].size();
+    v[i].shape = (*keep)[i].data();
+    v[i].dtype = blobs[i].dtype;
+    v[i].data = blobs[i].dptr;
+  }
+  return v;
+}
+
+void RegisterExternalOp(void*, const MXCustomOpDef* def) {
+  // copy the def — the pointer is only valid during lib init
+  MXCustomOpDef d = *def;
+  std::string name = def->name;
+  OpEntry& e = OpRegistry::Get()->Register(name);
+  e.n_in = d.n_in;
+  e.n_out = 1;
+  MXCustomInferFn infer = d.infer;
+  e.infer = [infer, name](const NodeAttrs&, const std::vector<TShape>& is,
+                          const std::vector<int>& it,
+                          std::vector<TShape>* os, std::vector<int>* ot) {
+    std::vector<std::vector<int64_t>> keep(is.size());
+    std::vector<MXTensorView> ins(is.size());
+    for (size_t i = 0; i < is.size(); ++i) {
+      keep[i] = is[i];
+      ins[i].ndim = (int)keep[i].size();
+      ins[i].shape = keep[i].data();
+      ins[i].dtype = it[i];
+      ins[i].data = nullptr;  // shapes only at infer time
+    }
+    int64_t oshape[8];
+    int ondim = 0, odtype = it.empty() ? kFloat32 : it[0];
+    MX_CHECK(infer((int)ins.size(), ins.data(), oshape, &ondim,
+                   &odtype) == 0,
+             "external op '" << name << "': infer failed");
+    MX_CHECK(ondim >= 0 && ondim <= 8, "external op: bad out ndim");
+    os->assign(1, TShape(oshape, oshape + ondim));
+    ot->assign(1, odtype);
+  };
+  if (d.fcompute_cpu) {
+    MXCustomComputeFn f = d.fcompute_cpu;
+    e.fcompute_cpu = [f, name](const NodeAttrs&, const OpCtx&,
+                               const std::vector<TBlob>& in,
+                               const std::vector<TBlob>& out) {
+      std::vector<std::vector<int64_t>> ki, ko;
+      auto iv = ViewsOf(in, &ki);
+      auto ov = ViewsOf(out, &ko);
+      MX_CHECK(f((int)iv.size(), iv.data(), ov.data(), nullptr) == 0,
+               "external op '" << name << "' failed (cpu)");
+    };
+  }
+  if (d.fcompute_gpu) {
+    MXCustomComputeFn f = d.fcompute_gpu;
+    e.fcompute_gpu = [f, name](const NodeAttrs&, const OpCtx& o,
+                               const std::vector<TBlob>& in,
+                               const std::vector<TBlob>& out) {
+      std::vector<std::vector<int64_t>> ki, ko;
+      auto iv = ViewsOf(in, &ki);
+      auto ov = ViewsOf(out, &ko);
+      MX_CHECK(f((int)iv.size(), iv.data(), ov.data(),
+                 (void*)o.rc.stream) == 0,
+               "external op '" << name << "' failed (gpu)");
+    };
+  }
+}
+
+}  // namespace
+
+int MXLoadLib(const char* path) {
+  API_BEGIN()
+  void* h = dlopen(path, RTLD_NOW | RTLD_LOCAL);
+  MX_CHECK(h, "MXLoadLib: dlopen failed: " << dlerror());
+  using InitFn = int (*)(MXRegisterOpFn, void*);
+  auto init = (InitFn)dlsym(h, "mxnet_amd_lib_init");
+  MX_CHECK(init, "MXLoadLib: library has no mxnet_amd_lib_init");
+  MX_CHECK(init(&RegisterExternalOp, nullptr) == 0,
+           "MXLoadLib: library init returned nonzero");
   API_END()
 }
 

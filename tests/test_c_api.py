@@ -150,3 +150,25 @@ def test_c_api_save_load(lib, tmp_path):
     from mxnet_amd.utils import serialization as ser
     loaded = ser.load_ndarrays(fname.decode())
     np.testing.assert_array_equal(loaded['w'].asnumpy(), av)
+
+
+def test_load_external_op_library(tmp_path):
+    """MXLoadLib (reference lib_api.h): dlopen a user .so, register its
+    MXCustomOpDef ops into the native registry, invoke like built-ins."""
+    import subprocess
+    import numpy as np
+    so = str(tmp_path / 'libmy_relu.so')
+    subprocess.check_call(['g++', '-shared', '-fPIC', '-I', 'include',
+                           'examples/lib_custom_op/my_relu.cc', '-o', so])
+    import mxnet_amd as mx
+    from mxnet_amd import _core
+    from mxnet_amd.ndarray.ndarray import NDArray
+    mx.library.load(so)
+    x = _core.from_numpy(
+        np.array([[-1.0, 2.0], [3.0, -4.0]], np.float32), 1, 0)
+    y = NDArray(_core.invoke('my_relu', [x], {})[0])
+    np.testing.assert_array_equal(y.asnumpy(), [[0, 2], [3, 0]])
+    # loading a nonexistent library raises through MXGetLastError
+    import pytest
+    with pytest.raises(RuntimeError):
+        mx.library.load(str(tmp_path / 'nope.so'))
