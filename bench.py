@@ -214,7 +214,11 @@ def main():
             mx.nd.waitall()
             grad_handles = [p.list_grad()[0]._h for p in trainer._params]
             def replay_step():
-                _core.launch_graph(local_rank, g1)
+                # G1 declares the grads it writes so the comm-stream
+                # all-reduces order behind the REPLAYED backward (not a
+                # stale pre-capture event)
+                _core.launch_graph(local_rank, g1, [],
+                                   grad_handles if distributed else [])
                 if distributed:
                     trainer._allreduce_grads()
                 # read-deps on the grads order G2 behind the comm-stream

@@ -591,12 +591,17 @@ uintptr_t Engine::EndCapture(int dev) {
 }
 
 void Engine::LaunchGraph(int dev, uintptr_t exec,
-                         const std::vector<VarId>& after) {
+                         const std::vector<VarId>& after,
+                         const std::vector<VarId>& mutate) {
+  // `mutate` declares the buffers the captured graph WRITES (e.g. the
+  // gradient set of a fwd+bwd graph): consumers on other streams (the
+  // RCCL all-reduces on the comm stream) then order behind the graph's
+  // completion event instead of a stale pre-capture event.
   PushAsync(
       [exec](const RunContext& rc) {
         MX_HIP_CALL(hipGraphLaunch((hipGraphExec_t)exec, rc.stream));
       },
-      Context::GPU(dev), after, {}, FnProperty::kNormal, "GraphLaunch");
+      Context::GPU(dev), after, mutate, FnProperty::kNormal, "GraphLaunch");
 }
 
 void Engine::SetProfiling(bool on) {

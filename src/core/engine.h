@@ -88,7 +88,8 @@ class Engine {
   uintptr_t EndCapture(int dev);      // returns hipGraphExec_t
   // optional read-deps order the launch behind e.g. comm-stream work
   void LaunchGraph(int dev, uintptr_t exec,
-                   const std::vector<VarId>& after = {});
+                   const std::vector<VarId>& after = {},
+                   const std::vector<VarId>& mutate = {});
 
   // ---- profiler (reference src/profiler: per-op aggregate stats) ----
   void SetProfiling(bool on);

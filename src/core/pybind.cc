@@ -229,14 +229,17 @@ PYBIND11_MODULE(_core, m) {
     return Engine::Get()->EndCapture(dev);
   });
   m.def("launch_graph",
-        [](int dev, uintptr_t exec, const std::vector<NDArray>& after) {
-          std::vector<VarId> vars;
-          for (auto& a : after) vars.push_back(a.var());
+        [](int dev, uintptr_t exec, const std::vector<NDArray>& after,
+           const std::vector<NDArray>& mutate) {
+          std::vector<VarId> rvars, wvars;
+          for (auto& a : after) rvars.push_back(a.var());
+          for (auto& a : mutate) wvars.push_back(a.var());
           py::gil_scoped_release rel;
-          Engine::Get()->LaunchGraph(dev, exec, vars);
+          Engine::Get()->LaunchGraph(dev, exec, rvars, wvars);
         },
         py::arg("dev"), py::arg("exec"),
-        py::arg("after") = std::vector<NDArray>());
+        py::arg("after") = std::vector<NDArray>(),
+        py::arg("mutate") = std::vector<NDArray>());
 
   // autograd tape
   m.def("set_recording", [](bool r) {
