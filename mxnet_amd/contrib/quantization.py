@@ -7,8 +7,8 @@ import torch
 
 from ..ndarray.ndarray import NDArray
 
-__all__ = ['quantize', 'dequantize', 'calib_minmax', 'QuantizedDense',
-           'QuantizedConv2D', 'quantize_net']
+__all__ = ['quantize', 'dequantize', 'calib_minmax', 'calib_entropy',
+           'QuantizedDense', 'QuantizedConv2D', 'quantize_net']
 
 
 def _t(x):
@@ -19,6 +19,68 @@ def calib_minmax(x):
     """Symmetric per-tensor scale from abs-max (reference quantize_v2
     min/max calibration)."""
     return float(_t(x).abs().max().item()) / 127.0 or 1.0
+
+
+def _calib_sample(t, cap=65536):
+    """Subsampled |activation| values for offline calibration (keeps
+    memory bounded while the KL sweep sees the value distribution)."""
+    import numpy as np
+    a = np.abs(t.detach().float().reshape(-1).cpu().numpy())
+    if a.size > cap:
+        a = a[:: (a.size + cap - 1) // cap]
+    return a
+
+
+def calib_entropy(x, num_bins=2048, num_quantized_bins=255):
+    """KL-divergence-optimal symmetric threshold (reference
+    quantization.py:_get_optimal_threshold / calibrate.cc entropy
+    mode): sweep candidate clip thresholds over an abs-value histogram
+    and pick the one whose 255-bin re-quantization has minimal KL
+    against the clipped reference distribution."""
+    import numpy as np
+    t = _t(x)
+    arr = np.abs(t.detach().float().cpu().numpy()).ravel()
+    amax = float(arr.max()) if arr.size else 0.0
+    if amax == 0.0:
+        return 1.0 / 127.0
+    # histogram over a robust range: a lone extreme outlier would push
+    # all the real mass into a handful of bins and blind the KL sweep
+    # (outliers beyond the range clip into the last candidate bin)
+    hi = min(amax, 4.0 * float(np.percentile(arr, 99.9)) + 1e-12)
+    hist, edges = np.histogram(np.minimum(arr, hi), bins=num_bins,
+                               range=(0, hi))
+    hist = hist.astype(np.float64)
+    best_kl, best_t = None, amax
+    # candidate thresholds: from ~1/8 of the range up to amax
+    for i in range(num_quantized_bins // 2, num_bins + 1,
+                   max(1, num_bins // 128)):
+        threshold = edges[i]
+        p = hist[:i].copy()
+        p[-1] += hist[i:].sum()  # clip outliers into the last bin
+        if p.sum() == 0:
+            continue
+        # quantize the i reference bins down to num_quantized_bins
+        factor = i / num_quantized_bins
+        q = np.zeros(i)
+        for j in range(num_quantized_bins):
+            lo = int(np.floor(j * factor))
+            hi = max(lo + 1, int(np.ceil((j + 1) * factor)))
+            hi = min(hi, i)
+            seg = p[lo:hi]
+            nz = (seg > 0).sum()
+            if nz:
+                q[lo:hi][seg > 0] = seg[seg > 0].sum() / nz
+        pn = p / p.sum()
+        qs = q.sum()
+        if qs == 0:
+            continue
+        qn = q / qs
+        mask = pn > 0
+        kl = float(np.sum(pn[mask] * np.log(
+            pn[mask] / np.maximum(qn[mask], 1e-12))))
+        if best_kl is None or kl < best_kl:
+            best_kl, best_t = kl, threshold
+    return best_t / 127.0
 
 
 def quantize(x, scale=None):
@@ -56,10 +118,20 @@ class QuantizedDense:
         b = dense_layer.bias
         self._bias = b.data().handle.float() if b is not None else None
         self._out_dtype = w.dtype
+        self._act = dense_layer._act_type
+
+    # offline-calibration hooks: when `collect` is armed, record the
+    # activation for scale fitting; when `_x_scale` is fixed, use the
+    # static scale instead of per-batch minmax (reference calibrated
+    # quantize_v2 with out_type-range attrs)
+    _x_scale = None
+    _collect = None
 
     def __call__(self, x):
         t = _t(x)
-        xs = calib_minmax(t)
+        if self._collect is not None:
+            self._collect.append(_calib_sample(t))
+        xs = self._x_scale if self._x_scale is not None else calib_minmax(t)
         if t.is_cuda:
             from ..ops.dispatch import hip_required
             ext = hip_required('quantized_dense')
@@ -72,6 +144,11 @@ class QuantizedDense:
                  self._wq.float().t() * (xs * self._w_scale)).to(self._out_dtype)
         if self._bias is not None:
             y = y + self._bias.to(y.dtype)
+        if self._act == 'relu':
+            y = torch.relu(y)
+        elif self._act:
+            from ..ops import nn as _opsnn
+            y = _opsnn.activation(y, self._act)
         return NDArray(y.reshape(*t.shape[:-1], y.shape[-1]))
 
 
@@ -104,13 +181,18 @@ class QuantizedConv2D:
         self._out_dtype = w.dtype
         self._act = conv_layer._act_type
 
+    _x_scale = None
+    _collect = None
+
     def __call__(self, x):
         t = _t(x)
+        if self._collect is not None:
+            self._collect.append(_calib_sample(t))
         N, H, W, C = t.shape
         (sh, sw), (ph, pw), (dh, dw) = self._stride, self._pad, self._dil
         P = (H + 2 * ph - dh * (self._R - 1) - 1) // sh + 1
         Q = (W + 2 * pw - dw * (self._S - 1) - 1) // sw + 1
-        xs = calib_minmax(t)
+        xs = self._x_scale if self._x_scale is not None else calib_minmax(t)
         if t.is_cuda:
             from ..ops.dispatch import hip_required
             ext = hip_required('quantized_conv')
@@ -136,11 +218,21 @@ class QuantizedConv2D:
         return NDArray(y)
 
 
-def quantize_net(net, quantized_dtype='int8', exclude_layers=None):
-    """Swap Dense / NHWC Conv2D layers for int8 inference versions
-    (reference quantize_model)."""
+def quantize_net(net, quantized_dtype='int8', exclude_layers=None,
+                 calib_data=None, calib_mode='naive', num_calib_batches=5):
+    """Quantize a Gluon net for int8 inference (reference
+    quantize_model / quantize_graph_pass.cc): Dense and NHWC Conv2D
+    children are REPLACED in the block tree with int8 versions.
+
+    calib_data (iterable of input batches) runs an offline calibration
+    pass: activation ranges are collected per quantized layer and fixed
+    as static scales — 'naive' = abs-max over the batches, 'entropy' =
+    KL-optimal threshold over the collected maxima distribution.
+    Without calib_data the layers fall back to per-batch dynamic
+    abs-max.  Returns the (block, name, wrapper) list."""
     from ..gluon import nn
     swapped = []
+
     def visit(block):
         for name, child in list(block._children.items()):
             if exclude_layers and name in exclude_layers:
@@ -153,4 +245,32 @@ def quantize_net(net, quantized_dtype='int8', exclude_layers=None):
             else:
                 visit(child)
     visit(net)
+    # graph surgery: the wrappers take the original layers' places
+    for block, name, q in swapped:
+        object.__setattr__(block, name, q)
+        block._children[name] = q
+    if calib_data is not None and swapped:
+        import numpy as np
+        for _, _, q in swapped:
+            q._collect = []
+        n = 0
+        for batch in calib_data:
+            net(batch)
+            n += 1
+            if n >= num_calib_batches:
+                break
+        for _, _, q in swapped:
+            samples = q._collect
+            q._collect = None
+            if not samples:
+                q._x_scale = 1.0 / 127.0
+                continue
+            pooled = np.concatenate(samples)
+            if calib_mode == 'entropy':
+                # KL threshold over the pooled activation-value samples
+                from ..ndarray.ndarray import array as _arr
+                q._x_scale = calib_entropy(
+                    _arr(pooled.astype('float32')))
+            else:
+                q._x_scale = float(pooled.max()) / 127.0
     return swapped
