@@ -193,10 +193,16 @@ class _OwnDistribution(Distribution):
     """Base for distributions with hand-written math; keeps the public
     Distribution API (log_prob/sample/mean/variance/entropy/cdf...)."""
 
+    _torch_cls = None  # set per class: interop shim for transforms/kl tail
+
     @property
     def _dist(self):
-        raise NotImplementedError(
-            f'{type(self).__name__}: method not implemented')
+        if self._torch_cls is None:
+            raise NotImplementedError(
+                f'{type(self).__name__}: method not implemented')
+        # lazily build the torch counterpart (TransformedDistribution /
+        # unregistered-KL interop only; our own math handles the rest)
+        return self._torch_cls(**self._args)
 
     def __init__(self, *args, **kwargs):  # no torch.distributions object
         names = list(self._arg_names)
@@ -227,6 +233,7 @@ class _OwnDistribution(Distribution):
 
 
 class Normal(_OwnDistribution):
+    _torch_cls = _td.Normal
     """Gaussian (reference distributions/normal.py): every method is the
     closed-form expression over elementwise ops."""
     _arg_names = ('loc', 'scale')
@@ -274,6 +281,7 @@ class Normal(_OwnDistribution):
 
 class LogNormal(Normal):
     """exp of a Normal (reference lognormal.py)."""
+    _torch_cls = _td.LogNormal
 
     @property
     def mean(self):
@@ -310,6 +318,7 @@ class LogNormal(Normal):
 
 
 class Laplace(_OwnDistribution):
+    _torch_cls = _td.Laplace
     _arg_names = ('loc', 'scale')
     has_grad = True
 
@@ -348,6 +357,7 @@ class Laplace(_OwnDistribution):
 
 
 class Uniform(_OwnDistribution):
+    _torch_cls = _td.Uniform
     _arg_names = ('low', 'high')
     has_grad = True
 
@@ -381,6 +391,7 @@ class Uniform(_OwnDistribution):
 
 
 class Exponential(_OwnDistribution):
+    _torch_cls = _td.Exponential
     _arg_names = ('rate',)
     has_grad = True
 
@@ -413,6 +424,7 @@ class Exponential(_OwnDistribution):
 
 
 class Gumbel(_OwnDistribution):
+    _torch_cls = _td.Gumbel
     _arg_names = ('loc', 'scale')
     has_grad = True
     _EULER = 0.57721566490153286555
@@ -455,6 +467,7 @@ def _probs_logits(self):
 
 
 class Bernoulli(_OwnDistribution):
+    _torch_cls = _td.Bernoulli
     _arg_names = ('probs', 'logits')
 
     @property
@@ -483,6 +496,7 @@ class Bernoulli(_OwnDistribution):
 
 
 class Geometric(_OwnDistribution):
+    _torch_cls = _td.Geometric
     """P(X=k) = (1-p)^k p, k = 0,1,2,... (reference geometric.py)."""
     _arg_names = ('probs', 'logits')
 
@@ -512,6 +526,7 @@ class Geometric(_OwnDistribution):
 
 
 class Poisson(_OwnDistribution):
+    _torch_cls = _td.Poisson
     _arg_names = ('rate',)
 
     @property
@@ -554,6 +569,7 @@ def _cat_logits(self):
 
 
 class Categorical(_OwnDistribution):
+    _torch_cls = _td.Categorical
     _arg_names = ('probs', 'logits')
 
     @property
@@ -595,6 +611,8 @@ class Categorical(_OwnDistribution):
 
 
 class OneHotCategorical(Categorical):
+    _torch_cls = _td.OneHotCategorical
+
     def log_prob(self, value):
         lg = _cat_logits(self)
         return _wrap((lg * _t(value)).sum(-1))
