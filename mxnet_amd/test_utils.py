@@ -132,3 +132,49 @@ def check_consistency(fn, inputs, ctx_list=None, dtypes=None, rtol=None,
                                rtol=r * 4, atol=a * 4 * gs), \
                 f'gradient mismatch ({dtype})'
     return results
+
+
+def check_numeric_gradient(fn, inputs, eps=1e-3, rtol=2e-2, atol=2e-3,
+                           ctx=None, dtype='float64'):
+    """Finite-difference gradient check against autograd (reference
+    test_utils.py:1043 check_numeric_gradient).
+
+    ``fn`` takes NDArrays and returns one NDArray; its sum is the scalar
+    objective.  Central differences are computed per input element and
+    compared with the tape's gradients.  Works on either runtime (the
+    native tape or torch.autograd, depending on the array backend).
+    """
+    from . import autograd
+    from .ndarray import ndarray as _nd
+    ctx = ctx or cpu()
+    arrs = [array(np.asarray(x, dtype='float32'), ctx=ctx, dtype='float32')
+            for x in inputs]
+    for a in arrs:
+        a.attach_grad()
+    with autograd.record():
+        out = fn(*arrs)
+        loss = out.sum() if out.size > 1 else out
+    loss.backward()
+    analytic = [a.grad.asnumpy().astype('float64') for a in arrs]
+
+    for i, x in enumerate(inputs):
+        base = np.asarray(x, dtype='float64')
+        num = np.zeros_like(base)
+        flat = base.reshape(-1)
+        nflat = num.reshape(-1)
+        for j in range(flat.size):
+            orig = flat[j]
+            for sign in (+1, -1):
+                flat[j] = orig + sign * eps
+                probe = [array(np.asarray(b if k != i else base,
+                                          dtype='float32'), ctx=ctx)
+                         for k, b in enumerate(inputs)]
+                # re-materialize the perturbed input
+                probe[i] = array(base.astype('float32'), ctx=ctx)
+                val = float(np.asarray(fn(*probe).asnumpy(),
+                                       dtype='float64').sum())
+                nflat[j] += sign * val / (2 * eps)
+            flat[j] = orig
+        assert_almost_equal(analytic[i], num, rtol=rtol, atol=atol,
+                            names=(f'autograd_grad[{i}]',
+                                   f'numeric_grad[{i}]'))
