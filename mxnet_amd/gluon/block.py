@@ -255,9 +255,13 @@ class HybridBlock(Block):
             return False
         if os.environ.get('MXNET_ENABLE_HIPGRAPH', '1') != '1':
             return False
+        # torch-frontend arrays only: the native runtime's pooled
+        # allocator has no graph-private pool, so captured inference
+        # intermediates could be recycled under a replay — the native
+        # path uses the explicit engine capture in the benches instead
         return (torch.cuda.is_available() and not _ag.is_recording()
-                and all(isinstance(a, NDArray) and a.handle.is_cuda
-                        for a in args))
+                and all(isinstance(a, NDArray) and not a.is_native
+                        and a.handle.is_cuda for a in args))
 
     def _graph_call(self, args):
         HybridBlock._graph_guard = True
