@@ -8,6 +8,8 @@
 #include <future>
 #include <new>
 
+#include "storage.h"
+
 namespace mxcore {
 
 namespace {
@@ -171,6 +173,10 @@ struct Engine::Impl {
 
   std::atomic<bool> forked_{false};
   void ReinitAfterFork();
+
+  // blocks freed during a capture, alive for the captured graph's life
+  std::unordered_map<uintptr_t, std::vector<Storage::Handle>>
+      graph_keepalive_;
 
   Engine* owner_ = nullptr;
 
@@ -554,6 +560,9 @@ hipStream_t Engine::CommStream(int dev) { return impl_->GetDevice(dev)->comm; }
 
 // ---- hipGraph capture -----------------------------------------------------
 void Engine::BeginCapture(int dev) {
+  // park pool frees on this device for the capture's duration — the
+  // graph bakes buffer addresses into its nodes (see Storage keepalive)
+  Storage::Get()->BeginCaptureKeepalive(dev);
   DeviceWorkers* dw = impl_->GetDevice(dev);
   std::promise<void> p;
   auto fut = p.get_future();
@@ -587,7 +596,32 @@ uintptr_t Engine::EndCapture(int dev) {
         p.set_value((uintptr_t)exec);
       },
       Context::GPU(dev), {}, {}, FnProperty::kNormal, "EndCapture");
-  return fut.get();
+  uintptr_t exec = fut.get();
+  // drain async deleters pushed during the capture window, then move
+  // every parked block into the graph's keepalive set — they are
+  // pooled again only when the graph is released
+  WaitForAll();
+  auto parked = Storage::Get()->EndCaptureKeepalive(dev);
+  {
+    std::lock_guard<std::mutex> g(impl_->dev_mu_);
+    impl_->graph_keepalive_[exec] = std::move(parked);
+  }
+  return exec;
+}
+
+void Engine::ReleaseGraph(uintptr_t exec) {
+  std::vector<Storage::Handle> parked;
+  {
+    std::lock_guard<std::mutex> g(impl_->dev_mu_);
+    auto it = impl_->graph_keepalive_.find(exec);
+    if (it != impl_->graph_keepalive_.end()) {
+      parked = std::move(it->second);
+      impl_->graph_keepalive_.erase(it);
+    }
+  }
+  WaitForAll();
+  (void)hipGraphExecDestroy((hipGraphExec_t)exec);
+  Storage::Get()->ReleaseHandles(parked);
 }
 
 void Engine::LaunchGraph(int dev, uintptr_t exec,

@@ -87,6 +87,7 @@ class Engine {
   void BeginCapture(int dev);
   uintptr_t EndCapture(int dev);      // returns hipGraphExec_t
   // optional read-deps order the launch behind e.g. comm-stream work
+  void ReleaseGraph(uintptr_t exec);  // destroy exec + pool its keepalive
   void LaunchGraph(int dev, uintptr_t exec,
                    const std::vector<VarId>& after = {},
                    const std::vector<VarId>& mutate = {});

@@ -240,6 +240,10 @@ PYBIND11_MODULE(_core, m) {
         py::arg("dev"), py::arg("exec"),
         py::arg("after") = std::vector<NDArray>(),
         py::arg("mutate") = std::vector<NDArray>());
+  m.def("release_graph", [](uintptr_t exec) {
+    py::gil_scoped_release rel;
+    Engine::Get()->ReleaseGraph(exec);
+  });
 
   // autograd tape
   m.def("set_recording", [](bool r) {

@@ -147,7 +147,37 @@ Storage::Handle Storage::Alloc(size_t size, Context ctx) {
   return GetManager(ctx)->Alloc(size);
 }
 
-void Storage::Free(const Handle& h) { GetManager(h.ctx)->Free(h); }
+void Storage::Free(const Handle& h) {
+  // during hipGraph capture on this device, park GPU frees: the graph
+  // bakes these addresses into its nodes, so recycling them under a
+  // later replay would alias live graph buffers
+  if (h.ctx.dev_type == Context::kGPU) {
+    std::lock_guard<std::mutex> g(cap_mu_);
+    auto it = capturing_.find(h.ctx.dev_id);
+    if (it != capturing_.end() && it->second) {
+      capture_parked_[h.ctx.dev_id].push_back(h);
+      return;
+    }
+  }
+  GetManager(h.ctx)->Free(h);
+}
+
+void Storage::BeginCaptureKeepalive(int dev_id) {
+  std::lock_guard<std::mutex> g(cap_mu_);
+  capturing_[dev_id] = true;
+}
+
+std::vector<Storage::Handle> Storage::EndCaptureKeepalive(int dev_id) {
+  std::lock_guard<std::mutex> g(cap_mu_);
+  capturing_[dev_id] = false;
+  std::vector<Handle> out;
+  out.swap(capture_parked_[dev_id]);
+  return out;
+}
+
+void Storage::ReleaseHandles(const std::vector<Handle>& hs) {
+  for (auto& h : hs) GetManager(h.ctx)->Free(h);
+}
 
 void Storage::DirectFree(const Handle& h) { GetManager(h.ctx)->DirectFree(h); }
 

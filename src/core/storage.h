@@ -34,6 +34,16 @@ class Storage {
   void Free(const Handle& h);        // returns to pool
   void DirectFree(const Handle& h);  // bypasses pool
   void ReleaseAll(Context ctx);      // drop all pooled blocks on ctx
+
+  // hipGraph-capture keepalive (the role torch's graph-private memory
+  // pools play): between BeginCapture/EndCapture the engine arms this;
+  // frees of GPU blocks are PARKED instead of pooled, because the
+  // captured graph references their addresses on every replay.  The
+  // parked set is returned at capture end and pooled only when the
+  // graph is released.
+  void BeginCaptureKeepalive(int dev_id);
+  std::vector<Handle> EndCaptureKeepalive(int dev_id);
+  void ReleaseHandles(const std::vector<Handle>& hs);  // pool them now
   // bytes currently cached in the pool for ctx (testing/telemetry)
   size_t PoolSize(Context ctx);
   // bytes handed out and not yet freed (testing/telemetry)
@@ -45,6 +55,11 @@ class Storage {
 
   std::mutex mu_;
   std::unordered_map<int64_t, Manager*> managers_;
+
+  // capture keepalive state (see BeginCaptureKeepalive)
+  std::mutex cap_mu_;
+  std::unordered_map<int, std::vector<Handle>> capture_parked_;
+  std::unordered_map<int, bool> capturing_;
 };
 
 }  // namespace mxcore
