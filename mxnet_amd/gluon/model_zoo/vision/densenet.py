@@ -26,7 +26,8 @@ class _DenseLayer(HybridBlock):
 
     def forward(self, x):
         out = self.body(x)
-        return NDArray(torch.cat([x._t, out._t], dim=self._dim))
+        from ....ndarray.ndarray import concat
+        return concat([x, out], dim=self._dim)
 
 
 def _make_transition(num_out, layout):

@@ -27,8 +27,9 @@ class _Branches(HybridBlock):
             self.register_child(b, f'b{i}')
 
     def forward(self, x):
-        return NDArray(torch.cat([b(x)._t for b in self._children.values()],
-                                 dim=self._dim))
+        from ....ndarray.ndarray import concat
+        return concat([b(x) for b in self._children.values()],
+                      dim=self._dim)
 
 
 def _seq(*blocks):

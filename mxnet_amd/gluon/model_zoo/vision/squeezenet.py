@@ -21,8 +21,8 @@ class _Fire(HybridBlock):
 
     def forward(self, x):
         x = self.squeeze(x)
-        return NDArray(torch.cat([self.expand1(x)._t, self.expand3(x)._t],
-                                 dim=self._dim))
+        from ....ndarray.ndarray import concat
+        return concat([self.expand1(x), self.expand3(x)], dim=self._dim)
 
 
 class SqueezeNet(HybridBlock):

@@ -316,3 +316,25 @@ def test_native_concat_stack(native):
     np.testing.assert_array_equal(
         stack([a, b], axis=0).asnumpy(),
         np.arange(12.).reshape(2, 2, 3))
+
+
+@pytest.mark.parametrize('name,size', [
+    ('resnet18_v1', 32), ('resnet18_v2', 32), ('vgg11', 32),
+    ('alexnet', 64), ('squeezenet1_0', 64), ('densenet121', 32),
+    ('mobilenet0_5', 32), ('mobilenet_v2_1_0', 32), ('inception_v3', 96),
+    ('resnext50_32x4d', 32)])
+def test_native_model_zoo_family(native, name, size):
+    """Every model-zoo family runs forward+backward on the native
+    runtime (the concat-based families ride the native concat op)."""
+    from mxnet_amd.gluon.model_zoo import vision
+    net = getattr(vision, name)(layout='NHWC', classes=10)
+    net.initialize(ctx=mx.cpu())
+    x = mx.nd.array(np.random.RandomState(0).randn(1, size, size, 3)
+                    .astype('float32'))
+    with autograd.record():
+        y = net(x)
+        L = (y * y).sum()
+    L.backward()
+    assert y.shape == (1, 10)
+    v = L.asscalar()
+    assert v == v
