@@ -21,6 +21,30 @@ def _reshape_like(pred, label):
     return label.reshape(pred.shape)
 
 
+# -- native-runtime composition helpers --------------------------------------
+def _n_mean_nonbatch(loss, batch_axis):
+    dims = tuple(d for d in range(loss.ndim) if d != batch_axis)
+    return loss.mean(axis=dims) if dims else loss
+
+
+def _n_relu(x):
+    from ..ndarray import ops as _F
+    return _F.Activation(x, act_type='relu')
+
+
+def _n_softplus(x):
+    # log(1+exp(x)) = relu(x) + log1p(exp(-|x|)), overflow-safe
+    return _n_relu(x) + ((x.abs() * -1.0).exp() + 1.0).log()
+
+
+def _n_weight(loss, weight, sample_weight):
+    if sample_weight is not None:
+        loss = loss * sample_weight
+    if weight is not None:
+        loss = loss * float(weight)
+    return loss
+
+
 class Loss(HybridBlock):
     def __init__(self, weight=None, batch_axis=0, **kwargs):
         super().__init__(**kwargs)
@@ -51,10 +75,21 @@ class L2Loss(Loss):
         loss = _apply_weighting(loss, self._weight / 2, sample_weight)
         return self._mean_nonbatch(loss)
 
+    def _forward_native(self, pred, label, sample_weight=None):
+        loss = (pred - label.reshape(pred.shape)).square()
+        loss = _n_weight(loss, self._weight / 2 if self._weight else None,
+                         sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
+
 
 class L1Loss(Loss):
     def __init__(self, weight=None, batch_axis=0, **kwargs):
         super().__init__(weight, batch_axis, **kwargs)
+
+    def _forward_native(self, pred, label, sample_weight=None):
+        loss = (pred - label.reshape(pred.shape)).abs()
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
 
     def _forward(self, pred, label, sample_weight=None):
         loss = (pred - _reshape_like(pred, label)).abs()
@@ -119,6 +154,21 @@ class SigmoidBinaryCrossEntropyLoss(Loss):
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
 
+    def _forward_native(self, pred, label, sample_weight=None,
+                        pos_weight=None):
+        assert pos_weight is None, \
+            'pos_weight: use the torch frontend for this option'
+        label = label.reshape(pred.shape)
+        if not self._from_sigmoid:
+            # softplus(pred) - label*pred (standard logits BCE)
+            loss = _n_softplus(pred) - label * pred
+        else:
+            eps = 1e-12
+            loss = ((pred + eps).log() * label
+                    + ((pred * -1.0) + (1 + eps)).log() * (label * -1.0 + 1.0)) * -1.0
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
+
 
 SigmoidBCELoss = SigmoidBinaryCrossEntropyLoss
 
@@ -138,6 +188,14 @@ class KLDivLoss(Loss):
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
 
+    def _forward_native(self, pred, label, sample_weight=None):
+        from ..ndarray import ops as _F
+        if not self._from_logits:
+            pred = _F.log_softmax(pred, axis=self._axis)
+        loss = label * ((label + 1e-12).log() - pred)
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
+
 
 class HuberLoss(Loss):
     def __init__(self, rho=1.0, weight=None, batch_axis=0, **kwargs):
@@ -153,6 +211,15 @@ class HuberLoss(Loss):
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
 
+    def _forward_native(self, pred, label, sample_weight=None):
+        # piecewise huber as clip+relu composition:
+        # (0.5/rho)*min(|d|,rho)^2 + relu(|d|-rho)
+        d = (pred - label.reshape(pred.shape)).abs()
+        rho = float(self._rho)
+        loss = d.clip(0.0, rho).square() * (0.5 / rho) + _n_relu(d - rho)
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
+
 
 class HingeLoss(Loss):
     def __init__(self, margin=1, weight=None, batch_axis=0, **kwargs):
@@ -164,6 +231,14 @@ class HingeLoss(Loss):
         loss = torch.relu(self._margin - pred * label)
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
+
+    def _forward_native(self, pred, label, sample_weight=None):
+        z = pred * label.reshape(pred.shape) * -1.0 + float(self._margin)
+        loss = _n_relu(z)
+        if type(self).__name__ == 'SquaredHingeLoss':
+            loss = loss.square()
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
 
 
 class SquaredHingeLoss(HingeLoss):
@@ -186,6 +261,14 @@ class LogisticLoss(Loss):
         loss = torch.nn.functional.softplus(-pred * label)
         loss = _apply_weighting(loss, self._weight, sample_weight)
         return self._mean_nonbatch(loss)
+
+    def _forward_native(self, pred, label, sample_weight=None):
+        label = label.reshape(pred.shape)
+        if self._label_format == 'binary':
+            label = label * 2.0 - 1.0
+        loss = _n_softplus(pred * label * -1.0)
+        loss = _n_weight(loss, self._weight, sample_weight)
+        return _n_mean_nonbatch(loss, self._batch_axis)
 
 
 class TripletLoss(Loss):

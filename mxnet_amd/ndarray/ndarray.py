@@ -621,9 +621,35 @@ class NDArray:
         return NDArray(self._t.float().norm().to(self._t.dtype))
 
     def abs(self):
+        if self._h is not None:
+            return self._invoke('abs', [self])
         return NDArray(self._t.abs())
 
+    def exp(self):
+        if self._h is not None:
+            return self._invoke('exp', [self])
+        return NDArray(self._t.exp())
+
+    def log(self):
+        if self._h is not None:
+            return self._invoke('log', [self])
+        return NDArray(self._t.log())
+
+    def sqrt(self):
+        if self._h is not None:
+            return self._invoke('sqrt', [self])
+        return NDArray(self._t.sqrt())
+
+    def square(self):
+        if self._h is not None:
+            return self._invoke('square', [self])
+        return NDArray(self._t.square())
+
     def clip(self, a_min, a_max):
+        if self._h is not None:
+            return self._invoke('clip', [self],
+                                {'alpha': str(float(a_min)),
+                                 'beta': str(float(a_max))})
         return NDArray(self._t.clamp(a_min, a_max))
 
     def __repr__(self):

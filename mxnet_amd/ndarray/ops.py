@@ -338,6 +338,8 @@ def where(condition, x, y, **kwargs):
 
 
 def clip(data, a_min, a_max, **kwargs):
+    if data.is_native:
+        return data.clip(a_min, a_max)
     return NDArray(torch.clamp(_t(data), a_min, a_max))
 
 

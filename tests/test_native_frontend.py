@@ -338,3 +338,39 @@ def test_native_model_zoo_family(native, name, size):
     assert y.shape == (1, 10)
     v = L.asscalar()
     assert v == v
+
+
+def test_native_losses_match_torch_backend():
+    """Every composable loss produces identical values on the native
+    runtime and the torch frontend (closed-form compositions over
+    dual-backend nd ops; softplus/huber via relu+clip identities)."""
+    from mxnet_amd.gluon import loss as gl
+    rs = np.random.RandomState(0)
+    p = rs.randn(4, 5).astype('float32')
+    l = rs.randn(4, 5).astype('float32')
+    lb = (rs.rand(4, 5) > 0.5).astype('float32')
+    sign = np.sign(rs.randn(4, 5)).astype('float32')
+    probs = np.abs(rs.rand(4, 5)).astype('float32')
+    probs /= probs.sum(1, keepdims=True)
+    cases = [
+        (gl.L1Loss(), (p, l)), (gl.L2Loss(), (p, l)),
+        (gl.HuberLoss(rho=0.7), (p, l)),
+        (gl.HingeLoss(margin=1.0), (p, sign)),
+        (gl.SquaredHingeLoss(margin=1.0), (p, sign)),
+        (gl.LogisticLoss(), (p, sign)),
+        (gl.SigmoidBinaryCrossEntropyLoss(), (p, lb)),
+        (gl.KLDivLoss(from_logits=False), (p, probs)),
+    ]
+    for loss_fn, args in cases:
+        prev = set_native(True)
+        try:
+            a_n = loss_fn(*[mx.nd.array(a) for a in args]).asnumpy()
+        finally:
+            set_native(prev)
+        prev = set_native(False)
+        try:
+            a_t = loss_fn(*[mx.nd.array(a) for a in args]).asnumpy()
+        finally:
+            set_native(prev)
+        np.testing.assert_allclose(a_n, a_t, rtol=1e-5, atol=1e-6,
+                                   err_msg=type(loss_fn).__name__)
