@@ -21,6 +21,16 @@ class PReLU(HybridBlock):
                                init=alpha_initializer)
 
     def hybrid_forward(self, F, x, alpha):
+        if getattr(x, 'is_native', False):
+            # prelu(x) = relu(x) - alpha * relu(-x); alpha broadcasts
+            # over the channel axis (axis 1 for >1 channels)
+            pos = F.Activation(x, act_type='relu')
+            neg = F.Activation(x * -1.0, act_type='relu')
+            if alpha.size > 1 and len(x.shape) > 1:
+                ash = [1] * len(x.shape)
+                ash[1] = alpha.size
+                alpha = alpha.reshape(ash)
+            return pos - neg * alpha
         import torch
         from ...ndarray.ndarray import NDArray
         return NDArray(torch.nn.functional.prelu(x._t, alpha._t))

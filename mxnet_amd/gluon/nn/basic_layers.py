@@ -290,6 +290,23 @@ class GroupNorm(HybridBlock):
     def forward(self, x):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
+        if getattr(x, 'is_native', False):
+            # composed group-norm over dual-backend nd ops (NCHW-style
+            # axis=1 channels, reference group_norm.cc semantics)
+            N, C = x.shape[0], x.shape[1]
+            G = self._num_groups
+            inner = 1
+            for d in x.shape[2:]:
+                inner *= d
+            xg = x.reshape(N, G, (C // G) * inner)
+            m = xg.mean(axis=2, keepdims=True)
+            d = xg - m
+            v = d.square().mean(axis=2, keepdims=True)
+            y = (d / (v + self._epsilon).sqrt()).reshape(*x.shape)
+            gshape = (1, C) + (1,) * (len(x.shape) - 2)
+            g = self.gamma.data(ctx).reshape(gshape)
+            b = self.beta.data(ctx).reshape(gshape)
+            return y * g + b
         g = self.gamma.data(ctx)._t
         b = self.beta.data(ctx)._t
         from ...ndarray.ndarray import NDArray
@@ -316,6 +333,19 @@ class InstanceNorm(HybridBlock):
     def forward(self, x):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
+        if getattr(x, 'is_native', False):
+            # per-sample per-channel normalization over spatial axes
+            axes = tuple(d for d in range(len(x.shape))
+                         if d not in (0, self._axis))
+            m = x.mean(axis=axes, keepdims=True)
+            d = x - m
+            v = d.square().mean(axis=axes, keepdims=True)
+            y = d / (v + self._epsilon).sqrt()
+            gshape = [1] * len(x.shape)
+            gshape[self._axis] = x.shape[self._axis]
+            g = self.gamma.data(ctx).reshape(gshape)
+            b = self.beta.data(ctx).reshape(gshape)
+            return y * g + b
         from ...ndarray.ndarray import NDArray
         y = torch.nn.functional.instance_norm(
             x._t, weight=self.gamma.data(ctx)._t, bias=self.beta.data(ctx)._t,

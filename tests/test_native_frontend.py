@@ -374,3 +374,33 @@ def test_native_losses_match_torch_backend():
             set_native(prev)
         np.testing.assert_allclose(a_n, a_t, rtol=1e-5, atol=1e-6,
                                    err_msg=type(loss_fn).__name__)
+
+
+def test_native_norm_layers_and_prelu(native):
+    """GroupNorm/InstanceNorm/PReLU compose from dual-backend nd ops on
+    the native runtime and match the torch closed forms."""
+    import torch
+    from mxnet_amd.gluon import nn as gnn
+    rs = np.random.RandomState(0)
+    x = rs.randn(2, 8, 5, 5).astype('float32')
+    gn = gnn.GroupNorm(num_groups=4)
+    gn.initialize()
+    np.testing.assert_allclose(
+        gn(mx.nd.array(x)).asnumpy(),
+        torch.nn.functional.group_norm(torch.tensor(x), 4, torch.ones(8),
+                                       torch.zeros(8), 1e-5).numpy(),
+        rtol=1e-4, atol=1e-5)
+    inorm = gnn.InstanceNorm()
+    inorm.initialize()
+    np.testing.assert_allclose(
+        inorm(mx.nd.array(x)).asnumpy(),
+        torch.nn.functional.instance_norm(
+            torch.tensor(x), weight=torch.ones(8), bias=torch.zeros(8),
+            eps=1e-5).numpy(), rtol=1e-4, atol=1e-5)
+    pr = gnn.PReLU(in_channels=8)
+    pr.initialize()
+    np.testing.assert_allclose(
+        pr(mx.nd.array(x)).asnumpy(),
+        torch.nn.functional.prelu(torch.tensor(x),
+                                  torch.full((8,), 0.25)).numpy(),
+        rtol=1e-5, atol=1e-6)
