@@ -1,10 +1,21 @@
-"""RNN cells (reference gluon/rnn/rnn_cell.py)."""
-import torch
+"""RNN cells (reference gluon/rnn/rnn_cell.py).
 
+Every cell composes dual-backend nd ops (FullyConnected on the MFMA
+GEMM, Activation, slicing, concat), so the same code runs on the torch
+frontend and on the native C++ runtime (own tape)."""
 from ..block import HybridBlock
 from ..parameter import Parameter
-from ...ndarray.ndarray import NDArray, zeros
+from ...ndarray.ndarray import NDArray, zeros, concat, stack
+from ...ndarray import ops as F
 from ... import initializer as init
+
+
+def _sig(x):
+    return F.Activation(x, act_type='sigmoid')
+
+
+def _tanh(x):
+    return F.Activation(x, act_type='tanh')
 
 
 class RecurrentCell(HybridBlock):
@@ -19,12 +30,18 @@ class RecurrentCell(HybridBlock):
         return [func(info, ctx=ctx, dtype=dtype, **kwargs)
                 for info in self.state_info(batch_size)]
 
+    @staticmethod
+    def _split_time(inputs, axis):
+        """[.., T, ..] -> list of T arrays with the axis squeezed."""
+        T = inputs.shape[axis]
+        return [inputs.slice_axis(axis, t, t + 1).squeeze(axis=axis)
+                for t in range(T)]
+
     def unroll(self, length, inputs, begin_state=None, layout='NTC',
                merge_outputs=None):
         axis = layout.find('T')
         if isinstance(inputs, NDArray):
-            inputs = [NDArray(t.squeeze(axis))
-                      for t in inputs._t.split(1, dim=axis)]
+            inputs = self._split_time(inputs, axis)
         batch = inputs[0].shape[0]
         states = begin_state or self.begin_state(
             batch, ctx=inputs[0].context, dtype=str(inputs[0].dtype))
@@ -33,7 +50,7 @@ class RecurrentCell(HybridBlock):
             out, states = self(inputs[t], states)
             outputs.append(out)
         if merge_outputs:
-            outputs = NDArray(torch.stack([o._t for o in outputs], dim=axis))
+            outputs = stack(outputs, axis=axis)
         return outputs, states
 
 
@@ -57,14 +74,13 @@ class RNNCell(RecurrentCell):
     def forward(self, x, states):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
-        h = states[0]._t
-        pre = torch.nn.functional.linear(x._t, self.i2h_weight.data(ctx)._t,
-                                         self.i2h_bias.data(ctx)._t) + \
-            torch.nn.functional.linear(h, self.h2h_weight.data(ctx)._t,
-                                       self.h2h_bias.data(ctx)._t)
-        out = torch.tanh(pre) if self._activation == 'tanh' else torch.relu(pre)
-        nd = NDArray(out)
-        return nd, [nd]
+        pre = F.FullyConnected(x, self.i2h_weight.data(ctx),
+                               self.i2h_bias.data(ctx), flatten=False) + \
+            F.FullyConnected(states[0], self.h2h_weight.data(ctx),
+                             self.h2h_bias.data(ctx), flatten=False)
+        out = _tanh(pre) if self._activation == 'tanh' \
+            else F.Activation(pre, act_type=self._activation)
+        return out, [out]
 
 
 class LSTMCell(RecurrentCell):
@@ -91,18 +107,19 @@ class LSTMCell(RecurrentCell):
     def forward(self, x, states):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
-        h, c = states[0]._t, states[1]._t
-        gates = torch.nn.functional.linear(
-            x._t, self.i2h_weight.data(ctx)._t, self.i2h_bias.data(ctx)._t) + \
-            torch.nn.functional.linear(
-                h, self.h2h_weight.data(ctx)._t, self.h2h_bias.data(ctx)._t)
+        h, c = states
+        gates = F.FullyConnected(x, self.i2h_weight.data(ctx),
+                                 self.i2h_bias.data(ctx), flatten=False) + \
+            F.FullyConnected(h, self.h2h_weight.data(ctx),
+                             self.h2h_bias.data(ctx), flatten=False)
         H = self._hidden_size
-        i, f, g, o = gates.split(H, dim=-1)
-        i, f, o = torch.sigmoid(i), torch.sigmoid(f), torch.sigmoid(o)
-        g = torch.tanh(g)
+        i = _sig(gates[:, 0:H])
+        f = _sig(gates[:, H:2 * H])
+        g = _tanh(gates[:, 2 * H:3 * H])
+        o = _sig(gates[:, 3 * H:4 * H])
         c_new = f * c + i * g
-        h_new = o * torch.tanh(c_new)
-        return NDArray(h_new), [NDArray(h_new), NDArray(c_new)]
+        h_new = o * _tanh(c_new)
+        return h_new, [h_new, c_new]
 
 
 class GRUCell(RecurrentCell):
@@ -128,20 +145,17 @@ class GRUCell(RecurrentCell):
     def forward(self, x, states):
         self._finish_deferred(x)
         ctx = self._param_ctx((x,))
-        h = states[0]._t
-        xg = torch.nn.functional.linear(
-            x._t, self.i2h_weight.data(ctx)._t, self.i2h_bias.data(ctx)._t)
-        hg = torch.nn.functional.linear(
-            h, self.h2h_weight.data(ctx)._t, self.h2h_bias.data(ctx)._t)
+        h = states[0]
+        xg = F.FullyConnected(x, self.i2h_weight.data(ctx),
+                              self.i2h_bias.data(ctx), flatten=False)
+        hg = F.FullyConnected(h, self.h2h_weight.data(ctx),
+                              self.h2h_bias.data(ctx), flatten=False)
         H = self._hidden_size
-        xr, xz, xn = xg.split(H, dim=-1)
-        hr, hz, hn = hg.split(H, dim=-1)
-        r = torch.sigmoid(xr + hr)
-        z = torch.sigmoid(xz + hz)
-        n = torch.tanh(xn + r * hn)
-        out = (1 - z) * n + z * h
-        nd = NDArray(out)
-        return nd, [nd]
+        r = _sig(xg[:, 0:H] + hg[:, 0:H])
+        z = _sig(xg[:, H:2 * H] + hg[:, H:2 * H])
+        n = _tanh(xg[:, 2 * H:3 * H] + r * hg[:, 2 * H:3 * H])
+        out = (z * -1.0 + 1.0) * n + z * h
+        return out, [out]
 
 
 class SequentialRNNCell(RecurrentCell):
@@ -179,7 +193,7 @@ class DropoutCell(RecurrentCell):
     def forward(self, x, states):
         from ... import autograd as _ag
         if self._rate > 0 and _ag.is_training():
-            x = NDArray(torch.nn.functional.dropout(x._t, self._rate, True))
+            x = F.Dropout(x, p=self._rate)
         return x, states
 
 
@@ -198,11 +212,19 @@ class ZoneoutCell(RecurrentCell):
         out, next_states = self.base_cell(x, states)
         from ... import autograd as _ag
         if _ag.is_training():
+            import numpy as _np
+
             def mix(new, old, p):
                 if p == 0:
                     return new
-                mask = torch.bernoulli(torch.full_like(new._t, p))
-                return NDArray(mask * old._t + (1 - mask) * new._t)
+                # bernoulli keep-old mask sampled host-side (parity with
+                # reference zoneout; mask is not differentiated through)
+                from ...ndarray.ndarray import array as _arr
+                m = (_np.random.rand(*new.shape) < p).astype('float32')
+                mask = _arr(m, ctx=new.context)
+                if str(mask.dtype) != str(new.dtype):
+                    mask = mask.astype(new.dtype)
+                return mask * old + (mask * -1.0 + 1.0) * new
             out = mix(out, states[0], self._zo)
             next_states = [mix(n, o, self._zs)
                            for n, o in zip(next_states, states)]
@@ -219,7 +241,7 @@ class ResidualCell(RecurrentCell):
 
     def forward(self, x, states):
         out, next_states = self.base_cell(x, states)
-        return NDArray(out._t + x._t), next_states
+        return out + x, next_states
 
 
 class BidirectionalCell(RecurrentCell):
@@ -236,8 +258,7 @@ class BidirectionalCell(RecurrentCell):
                merge_outputs=None):
         axis = layout.find('T')
         if isinstance(inputs, NDArray):
-            inputs = [NDArray(t.squeeze(axis))
-                      for t in inputs._t.split(1, dim=axis)]
+            inputs = self._split_time(inputs, axis)
         batch = inputs[0].shape[0]
         states = begin_state or self.begin_state(
             batch, ctx=inputs[0].context, dtype=str(inputs[0].dtype))
@@ -251,8 +272,7 @@ class BidirectionalCell(RecurrentCell):
             o, r_states = self.r_cell(inputs[t], r_states)
             r_out.append(o)
         r_out.reverse()
-        outputs = [NDArray(torch.cat([l._t, r._t], dim=-1))
-                   for l, r in zip(l_out, r_out)]
+        outputs = [concat([l, r], dim=-1) for l, r in zip(l_out, r_out)]
         if merge_outputs:
-            outputs = NDArray(torch.stack([o._t for o in outputs], dim=axis))
+            outputs = stack(outputs, axis=axis)
         return outputs, l_states + r_states

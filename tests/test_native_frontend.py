@@ -404,3 +404,39 @@ def test_native_norm_layers_and_prelu(native):
         torch.nn.functional.prelu(torch.tensor(x),
                                   torch.full((8,), 0.25)).numpy(),
         rtol=1e-5, atol=1e-6)
+
+
+@pytest.mark.parametrize('cellf', [
+    lambda: __import__('mxnet_amd.gluon.rnn', fromlist=['x']).LSTMCell(6),
+    lambda: __import__('mxnet_amd.gluon.rnn', fromlist=['x']).GRUCell(6),
+    lambda: __import__('mxnet_amd.gluon.rnn', fromlist=['x']).RNNCell(6),
+], ids=['lstm', 'gru', 'rnn'])
+def test_rnn_cells_native_match_torch(cellf):
+    """RNN cells compose dual-backend nd ops: unrolled outputs and
+    gradients match between the native runtime and the torch frontend."""
+    def run(use_native):
+        prev = set_native(use_native)
+        try:
+            cell = cellf()
+            cell.initialize()
+            x = mx.nd.array(np.random.RandomState(1)
+                            .randn(5, 3, 4).astype('float32'))
+            cell.unroll(3, x, layout='NTC', merge_outputs=True)
+            params = cell.collect_params()
+            for k, p in params.items():
+                rs = np.random.RandomState(sum(ord(c) for c in k) % 1000)
+                p.set_data(mx.nd.array(
+                    rs.randn(*p.shape).astype('float32') * 0.3))
+            with autograd.record():
+                outs, _ = cell.unroll(3, x, layout='NTC',
+                                      merge_outputs=True)
+                L = (outs * outs).sum()
+            L.backward()
+            k0 = sorted(params)[0]
+            return outs.asnumpy(), params[k0].grad(mx.cpu()).asnumpy()
+        finally:
+            set_native(prev)
+    yn, gn = run(True)
+    yt, gt = run(False)
+    np.testing.assert_allclose(yn, yt, rtol=1e-4, atol=1e-5)
+    np.testing.assert_allclose(gn, gt, rtol=1e-3, atol=1e-4)
