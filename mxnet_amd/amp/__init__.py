@@ -62,6 +62,12 @@ def unscale(trainer):
 
 def all_finite(arrays):
     """Multi-tensor finiteness check (native kernel on GPU)."""
+    arrays = [a for a in arrays if a is not None]
+    if arrays and isinstance(arrays[0], NDArray) and arrays[0].is_native:
+        from .. import _core
+        out = _core.invoke('multi_all_finite',
+                           [a._h for a in arrays], {})[0]
+        return bool(int(out.asnumpy()[0]))
     ts = [a._t if isinstance(a, NDArray) else a for a in arrays]
     if not ts:
         return True

@@ -1534,6 +1534,36 @@ bool _registered_nn = [] {
         });
       });
 
+  // multi-tensor finiteness check (AMP loss scaler): out = int32[1],
+  // 1 when every element of every input is finite
+  RegN("multi_all_finite").in(-1)
+      .infer([](const NodeAttrs&, const std::vector<TShape>&,
+                const std::vector<int>&, std::vector<TShape>* os,
+                std::vector<int>* ot) {
+        os->assign(1, TShape{1});
+        ot->assign(1, kInt32);
+      })
+      .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        std::vector<Arr> ts(in.begin(), in.end());
+        multi_all_finite_raw(LC(o), ts, Arr(out[0]));
+      })
+      .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        int ok = 1;
+        for (auto& t : in) {
+          MXC_DISPATCH_FLOAT(t.dtype, "multi_all_finite", {
+            auto* p = (const scalar_t*)t.dptr;
+            for (long i = 0; i < t.size() && ok; ++i) {
+              float v = (float)p[i];
+              if (!(v == v) || v == __builtin_inff() ||
+                  v == -__builtin_inff())
+                ok = 0;
+            }
+          });
+        }
+        *(int*)out[0].dptr = ok;
+      })
+      .bwd(NoGradN());
+
   // multi-tensor fused Adam: in = n grads; out = [w0,m0,v0, w1,m1,v1,
   // ...] then (has_master) the n fp32 masters appended at the tail.
   RegN("multi_adam_update").in(-1)

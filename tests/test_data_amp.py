@@ -122,3 +122,20 @@ def test_threaded_dataloader_native():
     # two epochs over the same loader work (fresh batcher per epoch)
     n2 = sum(1 for _ in dl2)
     assert n2 == 3
+
+
+def test_native_all_finite():
+    """AMP finiteness check on the native runtime (multi_all_finite
+    registry op; loss-scaler overflow detection path)."""
+    import numpy as np
+    import mxnet_amd as mx
+    from mxnet_amd.base import set_native
+    from mxnet_amd.amp import all_finite
+    prev = set_native(True)
+    try:
+        a = mx.nd.array(np.ones((4, 4)))
+        assert all_finite([a]) is True
+        assert all_finite([a, mx.nd.array(np.array([1.0, np.inf]))]) is False
+        assert all_finite([mx.nd.array(np.array([np.nan]))]) is False
+    finally:
+        set_native(prev)
