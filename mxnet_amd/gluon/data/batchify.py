@@ -7,6 +7,10 @@ from ...ndarray.ndarray import NDArray
 
 def _to_tensor(x):
     if isinstance(x, NDArray):
+        if x.is_native:
+            # native samples collate host-side (same as numpy inputs);
+            # the batch goes back through nd.array on the active runtime
+            return torch.as_tensor(x.asnumpy())
         return x._t
     if isinstance(x, torch.Tensor):
         return x
@@ -14,6 +18,15 @@ def _to_tensor(x):
     if a.dtype == _np.float64:
         a = a.astype(_np.float32)
     return torch.as_tensor(a)
+
+
+def _wrap_batch(t):
+    """Return the batch on the ACTIVE runtime (native_mode -> native)."""
+    from ...base import native_mode
+    if native_mode():
+        from ...ndarray.ndarray import array as _arr
+        return _arr(t.numpy())
+    return NDArray(t)
 
 
 class Stack:
@@ -24,9 +37,9 @@ class Stack:
             return tuple(Stack()([d[i] for d in data])
                          for i in range(len(data[0])))
         if isinstance(data[0], (int, float)):
-            return NDArray(torch.tensor(data))
+            return _wrap_batch(torch.tensor(data))
         ts = [_to_tensor(d) for d in data]
-        return NDArray(torch.stack(ts, dim=0))
+        return _wrap_batch(torch.stack(ts, dim=0))
 
 
 class Pad:
@@ -47,7 +60,7 @@ class Pad:
                 filler = torch.full(pad_shape, self._pad_val, dtype=t.dtype)
                 t = torch.cat([t, filler], dim=self._axis)
             padded.append(t)
-        return NDArray(torch.stack(padded, dim=0))
+        return _wrap_batch(torch.stack(padded, dim=0))
 
 
 class Group:
