@@ -462,8 +462,7 @@ def test_native_estimator_dataloader_fit(native):
     tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.2},
                  kvstore=None)
     est = Estimator(net=net, loss=SoftmaxCrossEntropyLoss(), trainer=tr)
-    est.fit(train_data=loader, epochs=4)
-    from mxnet_amd import autograd as _ag
+    est.fit(train_data=loader, epochs=12)
     out = net(mx.nd.array(X))
     acc = (out.asnumpy().argmax(1) == y).mean()
     assert acc > 0.6, acc
