@@ -559,12 +559,43 @@ class NDArray:
             self._t /= NDArray._rhs(o, self)
         return self
 
-    def __eq__(self, o): return NDArray((self._t == NDArray._rhs(o, self)).to(self._t.dtype)) if isinstance(o, (NDArray, int, float, torch.Tensor)) else NotImplemented
-    def __ne__(self, o): return NDArray((self._t != NDArray._rhs(o, self)).to(self._t.dtype)) if isinstance(o, (NDArray, int, float, torch.Tensor)) else NotImplemented
-    def __gt__(self, o): return NDArray((self._t > NDArray._rhs(o, self)).to(self._t.dtype))
-    def __ge__(self, o): return NDArray((self._t >= NDArray._rhs(o, self)).to(self._t.dtype))
-    def __lt__(self, o): return NDArray((self._t < NDArray._rhs(o, self)).to(self._t.dtype))
-    def __le__(self, o): return NDArray((self._t <= NDArray._rhs(o, self)).to(self._t.dtype))
+    def _ncmp(self, o, op):
+        rhs = o if isinstance(o, NDArray) else self * 0.0 + float(o)
+        return self._invoke(op, [self, rhs])
+
+    def __eq__(self, o):
+        if not isinstance(o, (NDArray, int, float, torch.Tensor)):
+            return NotImplemented
+        if self._h is not None:
+            return self._ncmp(o, 'equal')
+        return NDArray((self._t == NDArray._rhs(o, self)).to(self._t.dtype))
+
+    def __ne__(self, o):
+        if not isinstance(o, (NDArray, int, float, torch.Tensor)):
+            return NotImplemented
+        if self._h is not None:
+            return self._ncmp(o, 'not_equal')
+        return NDArray((self._t != NDArray._rhs(o, self)).to(self._t.dtype))
+
+    def __gt__(self, o):
+        if self._h is not None:
+            return self._ncmp(o, 'greater')
+        return NDArray((self._t > NDArray._rhs(o, self)).to(self._t.dtype))
+
+    def __ge__(self, o):
+        if self._h is not None:
+            return self._ncmp(o, 'greater_equal')
+        return NDArray((self._t >= NDArray._rhs(o, self)).to(self._t.dtype))
+
+    def __lt__(self, o):
+        if self._h is not None:
+            return self._ncmp(o, 'less')
+        return NDArray((self._t < NDArray._rhs(o, self)).to(self._t.dtype))
+
+    def __le__(self, o):
+        if self._h is not None:
+            return self._ncmp(o, 'less_equal')
+        return NDArray((self._t <= NDArray._rhs(o, self)).to(self._t.dtype))
 
     def __hash__(self):
         return id(self)
@@ -612,12 +643,20 @@ class NDArray:
         return NDArray(self._t.min(dim=axis, keepdim=keepdims).values)
 
     def argmax(self, axis=None):
+        if self._h is not None:
+            from . import ops as _ops
+            return _ops.argmax(self, axis=axis)
         return NDArray(self._t.argmax() if axis is None else self._t.argmax(dim=axis))
 
     def argmin(self, axis=None):
+        if self._h is not None:
+            from . import ops as _ops
+            return _ops.argmin(self, axis=axis)
         return NDArray(self._t.argmin() if axis is None else self._t.argmin(dim=axis))
 
     def norm(self):
+        if self._h is not None:
+            return self.square().sum().sqrt()
         return NDArray(self._t.float().norm().to(self._t.dtype))
 
     def abs(self):

@@ -334,7 +334,31 @@ ElementWiseSum = add_n
 
 
 def where(condition, x, y, **kwargs):
+    if condition.is_native:
+        # mask-select composition: m*x + (1-m)*y with a float mask
+        m = _ninv('not_equal', [condition, condition * 0.0], {})
+        if str(m.dtype) != str(x.dtype):
+            m = m.astype(x.dtype)
+        return m * x + (m * -1.0 + 1.0) * y
     return NDArray(torch.where(_t(condition).bool(), _t(x), _t(y)))
+
+
+def maximum(lhs, rhs, **kwargs):
+    if isinstance(lhs, NDArray) and lhs.is_native:
+        if isinstance(rhs, (int, float)):
+            return lhs.clip(float(rhs), 3.4e38)  # max(x,c) = clip-below
+        return _ninv('maximum', [lhs, rhs], {})
+    return NDArray(torch.maximum(_t(lhs), torch.as_tensor(
+        _t(rhs) if isinstance(rhs, NDArray) else rhs)))
+
+
+def minimum(lhs, rhs, **kwargs):
+    if isinstance(lhs, NDArray) and lhs.is_native:
+        if isinstance(rhs, (int, float)):
+            return lhs.clip(-3.4e38, float(rhs))  # min(x,c) = clip-above
+        return _ninv('minimum', [lhs, rhs], {})
+    return NDArray(torch.minimum(_t(lhs), torch.as_tensor(
+        _t(rhs) if isinstance(rhs, NDArray) else rhs)))
 
 
 def clip(data, a_min, a_max, **kwargs):
@@ -365,6 +389,11 @@ nansum = _reduce(lambda x, a, k: x.nansum() if a is None else x.nansum(dim=a, ke
 
 
 def norm(data, ord=2, axis=None, keepdims=False, **kwargs):
+    if data.is_native:
+        assert ord == 2
+        sq = data.square().sum(axis=axis, keepdims=keepdims) \
+            if axis is not None else data.square().sum()
+        return sq.sqrt()
     x = _t(data)
     if axis is None:
         return NDArray(torch.linalg.vector_norm(x.float(), ord).to(x.dtype))
@@ -373,18 +402,49 @@ def norm(data, ord=2, axis=None, keepdims=False, **kwargs):
 
 
 def argmax(data, axis=None, keepdims=False, **kwargs):
+    if data.is_native:
+        assert not keepdims
+        if axis is None:
+            flat = data.reshape(int(data.size))
+            return _ninv('argmax', [flat], {'axis': '0'})
+        return _ninv('argmax', [data], {'axis': str(axis)})
     x = _t(data)
     out = x.argmax() if axis is None else x.argmax(dim=axis, keepdim=keepdims)
     return NDArray(out.to(torch.float32))
 
 
 def argmin(data, axis=None, keepdims=False, **kwargs):
+    if data.is_native:
+        assert not keepdims
+        if axis is None:
+            flat = data.reshape(int(data.size))
+            return _ninv('argmin', [flat], {'axis': '0'})
+        return _ninv('argmin', [data], {'axis': str(axis)})
     x = _t(data)
     out = x.argmin() if axis is None else x.argmin(dim=axis, keepdim=keepdims)
     return NDArray(out.to(torch.float32))
 
 
 def topk(data, axis=-1, k=1, ret_typ='indices', is_ascend=False, **kwargs):
+    if data.is_native:
+        # ordering ops on the native runtime run host-side (reference
+        # used cub device radix sort; a CDNA4 sort kernel is future
+        # work) — correct, with a sync cost; rare in training loops
+        import numpy as np
+        arr = data.asnumpy()
+        order = np.argsort(arr, axis=axis)
+        if not is_ascend:
+            order = np.flip(order, axis=axis)
+        sl = [builtins.slice(None)] * arr.ndim
+        sl[axis] = builtins.slice(0, k)
+        idx = np.ascontiguousarray(order[tuple(sl)]).astype('float32')
+        vals = np.take_along_axis(arr, idx.astype('int64'), axis=axis)
+        from .ndarray import array as _mk
+        if ret_typ == 'value':
+            return _mk(vals, ctx=data.context)
+        if ret_typ == 'both':
+            return _mk(vals, ctx=data.context), _mk(idx, ctx=data.context)
+        return _mk(idx, ctx=data.context)
     vals, idx = torch.topk(_t(data), k, dim=axis, largest=not is_ascend)
     if ret_typ == 'value':
         return NDArray(vals)
@@ -394,6 +454,13 @@ def topk(data, axis=-1, k=1, ret_typ='indices', is_ascend=False, **kwargs):
 
 
 def sort(data, axis=-1, is_ascend=True, **kwargs):
+    if data.is_native:
+        import numpy as np
+        arr = np.sort(data.asnumpy(), axis=axis)
+        if not is_ascend:
+            arr = np.flip(arr, axis=axis).copy()
+        from .ndarray import array as _mk
+        return _mk(arr, ctx=data.context)
     return NDArray(torch.sort(_t(data), dim=axis, descending=not is_ascend).values)
 
 
@@ -459,6 +526,14 @@ def slice_like(data, shape_like, axes=None, **kwargs):
 
 
 def take(a, indices, axis=0, **kwargs):
+    if a.is_native:
+        assert axis == 0, 'native take: axis 0 (embedding gather)'
+        idx = indices if str(indices.dtype) == 'int64' \
+            else indices.astype('int64')
+        flat = idx.reshape(int(idx.size))
+        out = _ninv('Embedding', [flat, a], {})
+        oshape = tuple(indices.shape) + tuple(a.shape[1:])
+        return out.reshape(oshape)
     return NDArray(torch.index_select(_t(a), axis, _t(indices).long().reshape(-1)))
 
 
@@ -495,10 +570,28 @@ def one_hot(indices, depth, on_value=1.0, off_value=0.0, dtype='float32', **kwar
 
 
 def tile(data, reps, **kwargs):
+    if data.is_native:
+        out = data
+        from .ndarray import concat as _cat
+        for ax, r in enumerate(reps):
+            if r > 1:
+                out = _cat([out] * int(r), dim=ax)
+        return out
     return NDArray(_t(data).repeat(*reps))
 
 
 def repeat(data, repeats, axis=None, **kwargs):
+    if data.is_native:
+        if axis is None:
+            data = data.reshape(int(data.size))
+            axis = 0
+        # interleaved repeat: expand a size-1 axis then fold it in
+        shp = list(data.shape)
+        x = data.reshape(shp[:axis + 1] + [1] + shp[axis + 1:])
+        bshape = shp[:axis + 1] + [int(repeats)] + shp[axis + 1:]
+        x = x.broadcast_to(tuple(bshape))
+        shp[axis] *= int(repeats)
+        return x.reshape(shp)
     return NDArray(torch.repeat_interleave(_t(data), repeats, dim=axis))
 
 
@@ -530,10 +623,14 @@ def broadcast_axis(data, axis, size, **kwargs):
 
 
 def zeros_like(data, **kwargs):
+    if data.is_native:
+        return _ninv('zeros_like', [data], {})
     return NDArray(torch.zeros_like(_t(data)))
 
 
 def ones_like(data, **kwargs):
+    if data.is_native:
+        return _ninv('ones_like', [data], {})
     return NDArray(torch.ones_like(_t(data)))
 
 
@@ -545,6 +642,11 @@ Cast = cast
 
 
 def dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
+    if lhs.is_native:
+        a = lhs.transpose() if transpose_a else lhs
+        if transpose_b:
+            return _ninv('dot_nt', [a, rhs], {})
+        return _ninv('dot_nn', [a, rhs], {})
     a, b = _t(lhs), _t(rhs)
     if transpose_a:
         a = a.t()
@@ -554,6 +656,10 @@ def dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
 
 
 def batch_dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
+    if lhs.is_native:
+        a = lhs.transpose((0, 2, 1)) if transpose_a else lhs
+        b = rhs.transpose((0, 2, 1)) if transpose_b else rhs
+        return _ninv('batch_dot', [a, b], {})
     return NDArray(_nn.batch_dot(_t(lhs), _t(rhs), transpose_a, transpose_b))
 
 

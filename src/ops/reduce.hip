@@ -119,6 +119,27 @@ __global__ void scatter_strided_kernel(const T* __restrict__ dy,
   }
 }
 
+template <typename T, bool MAX>
+__global__ void arg_reduce_kernel(const T* __restrict__ x,
+                                  float* __restrict__ y, long nout,
+                                  long len, long inner) {
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < nout;
+       i += (long)gridDim.x * blockDim.x) {
+    long outer = i / inner, off = i % inner;
+    const T* base = x + outer * len * inner + off;
+    float best = (float)base[0];
+    long bi = 0;
+    for (long k = 1; k < len; ++k) {
+      float v = (float)base[k * inner];
+      if (MAX ? (v > best) : (v < best)) {
+        best = v;
+        bi = k;
+      }
+    }
+    y[i] = (float)bi;
+  }
+}
+
 template <typename T>
 __global__ void gather_strided_kernel(const T* __restrict__ x,
                                       T* __restrict__ y, long n, Strides8 st) {
@@ -577,6 +598,74 @@ bool _registered_reduce = [] {
           }
         });
       });
+
+  // argmax/argmin over ONE axis (reference ordering ops use topk; the
+  // index-only reductions here run one thread per output element with a
+  // serial strided scan)
+  auto arg_reduce = [](const char* name, bool is_max) {
+    Reg2(name).in(1)
+        .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
+                  const std::vector<int>&, std::vector<TShape>* os,
+                  std::vector<int>* ot) {
+          int nd = (int)is[0].size();
+          int ax = (int)a.GetInt("axis", -1);
+          if (ax < 0) ax += nd;
+          TShape out;
+          for (int d = 0; d < nd; ++d)
+            if (d != ax) out.push_back(is[0][d]);
+          if (out.empty()) out.push_back(1);
+          os->assign(1, out);
+          ot->assign(1, kFloat32);  // mxnet argmax returns float indices
+        })
+        .gpu([is_max](const NodeAttrs& a, const OpCtx& o, V in, V out) {
+          int nd = in[0].ndim();
+          int ax = (int)a.GetInt("axis", -1);
+          if (ax < 0) ax += nd;
+          long n = in[0].size(), len = in[0].shape[ax];
+          long inner = 1;
+          for (int d = ax + 1; d < nd; ++d) inner *= in[0].shape[d];
+          long nout = n / len;
+          MXC_DISPATCH_FLOAT(in[0].dtype, "arg_reduce", {
+            if (is_max)
+              arg_reduce_kernel<scalar_t, true>
+                  <<<grid_for(nout), kBlock, 0, o.rc.stream>>>(
+                      (const scalar_t*)in[0].dptr, (float*)out[0].dptr,
+                      nout, len, inner);
+            else
+              arg_reduce_kernel<scalar_t, false>
+                  <<<grid_for(nout), kBlock, 0, o.rc.stream>>>(
+                      (const scalar_t*)in[0].dptr, (float*)out[0].dptr,
+                      nout, len, inner);
+          });
+          HIP_CHECK_LAST();
+        })
+        .cpu([is_max](const NodeAttrs& a, const OpCtx&, V in, V out) {
+          MX_CHECK(in[0].dtype == kFloat32,
+                   "arg_reduce: CPU path is fp32");
+          int nd = in[0].ndim();
+          int ax = (int)a.GetInt("axis", -1);
+          if (ax < 0) ax += nd;
+          long n = in[0].size(), len = in[0].shape[ax];
+          long inner = 1;
+          for (int d = ax + 1; d < nd; ++d) inner *= in[0].shape[d];
+          long nout = n / len;
+          auto* x = (const float*)in[0].dptr;
+          auto* y = (float*)out[0].dptr;
+          for (long i = 0; i < nout; ++i) {
+            long outer = i / inner, off = i % inner;
+            const float* base = x + outer * len * inner + off;
+            float best = base[0];
+            long bi = 0;
+            for (long k = 1; k < len; ++k) {
+              float v = base[k * inner];
+              if (is_max ? (v > best) : (v < best)) { best = v; bi = k; }
+            }
+            y[i] = (float)bi;
+          }
+        });
+  };
+  arg_reduce("argmax", true);
+  arg_reduce("argmin", false);
 
   // concat along an axis (reference concat.cc): forward scatters each
   // input into its strided slab of the output (scatter_strided, one

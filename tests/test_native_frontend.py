@@ -466,3 +466,45 @@ def test_native_estimator_dataloader_fit(native):
     out = net(mx.nd.array(X))
     acc = (out.asnumpy().argmax(1) == y).mean()
     assert acc > 0.6, acc
+
+
+def test_native_op_batch(native):
+    """Frontend batch that previously had no native branches:
+    maximum/minimum/where/comparisons/dot/batch_dot/norm/argmax/argmin/
+    tile/repeat/take/sort/topk all run on the native runtime and match
+    numpy."""
+    F = mx.nd.ops
+    rs = np.random.RandomState(0)
+    xa = rs.randn(3, 4).astype('float32')
+    ya = rs.randn(3, 4).astype('float32')
+    x, y = mx.nd.array(xa), mx.nd.array(ya)
+    np.testing.assert_allclose(F.maximum(x, y).asnumpy(), np.maximum(xa, ya))
+    np.testing.assert_allclose(F.minimum(x, 0.1).asnumpy(),
+                               np.minimum(xa, 0.1), rtol=1e-6)
+    np.testing.assert_allclose((x > y).asnumpy(),
+                               (xa > ya).astype('float32'))
+    cond = mx.nd.array((xa > 0).astype('float32'))
+    np.testing.assert_allclose(F.where(cond, x, y).asnumpy(),
+                               np.where(xa > 0, xa, ya))
+    np.testing.assert_allclose(F.dot(x, y, transpose_b=True).asnumpy(),
+                               xa @ ya.T, rtol=1e-5)
+    a3 = rs.randn(2, 3, 4).astype('float32')
+    b3 = rs.randn(2, 4, 5).astype('float32')
+    np.testing.assert_allclose(
+        F.batch_dot(mx.nd.array(a3), mx.nd.array(b3)).asnumpy(),
+        a3 @ b3, rtol=1e-5)
+    np.testing.assert_allclose(x.norm().asnumpy(), [np.linalg.norm(xa)],
+                               rtol=1e-5)
+    np.testing.assert_allclose(x.argmax(axis=1).asnumpy(), xa.argmax(1))
+    np.testing.assert_allclose(F.tile(x, (2, 3)).asnumpy(),
+                               np.tile(xa, (2, 3)))
+    np.testing.assert_allclose(F.repeat(x, 2, axis=0).asnumpy(),
+                               np.repeat(xa, 2, 0))
+    w = rs.randn(6, 5).astype('float32')
+    idx = np.array([[0, 2], [5, 1]], 'float32')
+    np.testing.assert_allclose(
+        F.take(mx.nd.array(w), mx.nd.array(idx)).asnumpy(),
+        w[idx.astype(int)])
+    np.testing.assert_allclose(F.sort(x).asnumpy(), np.sort(xa, -1))
+    v, i = F.topk(x, k=2, ret_typ='both')
+    np.testing.assert_allclose(v.asnumpy(), -np.sort(-xa, -1)[:, :2])
