@@ -284,6 +284,28 @@ def test_native_lstm_gpu(native):
     assert losses[-1] < losses[0], losses
 
 
+def test_native_op_batch_gpu(native):
+    """New frontend-native ops on GPU: argmax kernel, comparisons,
+    maximum, where composition, dot."""
+    F = mx.nd.ops
+    rs = np.random.RandomState(0)
+    xa = rs.randn(64, 33).astype('float32')
+    ya = rs.randn(64, 33).astype('float32')
+    x = mx.nd.array(xa, ctx=mx.gpu(0))
+    y = mx.nd.array(ya, ctx=mx.gpu(0))
+    np.testing.assert_allclose(x.argmax(axis=1).asnumpy(), xa.argmax(1))
+    np.testing.assert_allclose(F.argmin(x, axis=0).asnumpy(), xa.argmin(0))
+    np.testing.assert_allclose(F.maximum(x, y).asnumpy(),
+                               np.maximum(xa, ya))
+    np.testing.assert_allclose((x > y).asnumpy(),
+                               (xa > ya).astype('float32'))
+    cond = mx.nd.array((xa > 0).astype('float32'), ctx=mx.gpu(0))
+    np.testing.assert_allclose(F.where(cond, x, y).asnumpy(),
+                               np.where(xa > 0, xa, ya))
+    np.testing.assert_allclose(F.dot(x, y, transpose_b=True).asnumpy(),
+                               xa @ ya.T, rtol=2e-3, atol=2e-3)
+
+
 def test_native_rccl_world1(native):
     """world=1 communicator: allreduce/broadcast are engine-sequenced
     no-ops (average still runs its scale kernel)."""
