@@ -508,3 +508,25 @@ def test_native_op_batch(native):
     np.testing.assert_allclose(F.sort(x).asnumpy(), np.sort(xa, -1))
     v, i = F.topk(x, k=2, ret_typ='both')
     np.testing.assert_allclose(v.asnumpy(), -np.sort(-xa, -1)[:, :2])
+
+
+def test_native_export_symbolblock_roundtrip(native):
+    """export() -> SymbolBlock.imports -> native inference: the symbol
+    interpreter executes through the dual-backend nd ops, so the
+    checkpoint pair serves the native runtime too."""
+    import os
+    import tempfile
+    from mxnet_amd.gluon import SymbolBlock
+    net = nn.HybridSequential()
+    net.add(nn.Dense(8, activation='relu'), nn.Dense(4))
+    net.initialize()
+    x = mx.nd.array(np.random.RandomState(0).randn(2, 6).astype('float32'))
+    ref = net(x).asnumpy()
+    d = tempfile.mkdtemp()
+    path = os.path.join(d, 'm')
+    net.export(path)
+    sb = SymbolBlock.imports(path + '-symbol.json', ['data'],
+                             path + '-0000.params')
+    out = sb(x)
+    assert out.is_native
+    np.testing.assert_allclose(out.asnumpy(), ref, rtol=1e-5, atol=1e-6)
