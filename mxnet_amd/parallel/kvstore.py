@@ -215,6 +215,15 @@ class NativeDistKVStore(KVStoreBase):
         _os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
         _os.environ.setdefault('MASTER_PORT', '29741')
         self._core = _core
+        self._gloo = not torch.cuda.is_available()
+        if self._gloo:
+            # CPU hosts (CI, multi-process tests): RCCL needs a GPU, so
+            # the native arrays bridge through numpy onto gloo
+            import torch.distributed as dist
+            if not dist.is_initialized():
+                dist.init_process_group('gloo', rank=rank, world_size=world)
+            self._world, self._rank = world, rank
+            return
         _core.rccl_init(world, rank, dev)
 
     def set_gradient_compression(self, compression_params):
@@ -223,19 +232,41 @@ class NativeDistKVStore(KVStoreBase):
 
     @property
     def rank(self):
-        return self._core.rccl_rank()
+        return self._rank if self._gloo else self._core.rccl_rank()
 
     @property
     def num_workers(self):
-        return self._core.rccl_world()
+        return self._world if self._gloo else self._core.rccl_world()
+
+    def _gloo_allreduce(self, v, average):
+        import torch.distributed as dist
+        t = torch.from_numpy(v.asnumpy().copy())
+        dist.all_reduce(t)
+        if average:
+            t /= self._world
+        from ..ndarray.ndarray import array as _arr
+        _arr(t.numpy(), ctx=v.context, dtype=str(v.dtype)).copyto(v)
+
+    def _gloo_broadcast(self, v, root):
+        import torch.distributed as dist
+        t = torch.from_numpy(v.asnumpy().copy())
+        dist.broadcast(t, src=root)
+        from ..ndarray.ndarray import array as _arr
+        _arr(t.numpy(), ctx=v.context, dtype=str(v.dtype)).copyto(v)
 
     def init(self, key, value):
         v = value[0] if isinstance(value, (list, tuple)) else value
-        self._core.rccl_broadcast(v._h, 0)
+        if self._gloo:
+            self._gloo_broadcast(v, 0)
+        else:
+            self._core.rccl_broadcast(v._h, 0)
 
     def broadcast(self, key, value, out, priority=0):
         v = value[0] if isinstance(value, (list, tuple)) else value
-        self._core.rccl_broadcast(v._h, 0)
+        if self._gloo:
+            self._gloo_broadcast(v, 0)
+        else:
+            self._core.rccl_broadcast(v._h, 0)
         outs = out if isinstance(out, (list, tuple)) else [out]
         for o in outs:
             if o is not v:
@@ -245,7 +276,10 @@ class NativeDistKVStore(KVStoreBase):
         """All-reduce-average in place; async by construction (the engine
         orders it against producers/consumers via vars)."""
         v = value[0] if isinstance(value, (list, tuple)) else value
-        self._core.rccl_allreduce(v._h, True)
+        if self._gloo:
+            self._gloo_allreduce(v, True)
+        else:
+            self._core.rccl_allreduce(v._h, True)
         if out is not None:
             outs = out if isinstance(out, (list, tuple)) else [out]
             for o in outs:

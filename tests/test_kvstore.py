@@ -262,3 +262,43 @@ def test_dist_sync_parameter_server(tmp_path):
     # merged g1 = (2+4) = 6 -> w1' = 20 - 0.5*6 = 17
     np.testing.assert_allclose(s0[:4], 8.5, rtol=1e-6)
     np.testing.assert_allclose(s0[4:], 17.0, rtol=1e-6)
+
+
+def test_native_dist_kvstore_gloo_bridge():
+    """NativeDistKVStore on CPU hosts bridges native arrays onto gloo
+    (RCCL needs a GPU): 2 processes all-reduce-average and broadcast."""
+    import subprocess
+    import sys
+    import os
+    script = r'''
+import os
+os.environ['MXNET_NATIVE_RUNTIME'] = '1'
+import numpy as np
+import mxnet_amd as mx
+from mxnet_amd.parallel.kvstore import NativeDistKVStore
+rank = int(os.environ['RANK'])
+kv = NativeDistKVStore('dist_sync')
+v = mx.nd.array(np.full((8,), float(rank + 1), dtype='float32'))
+kv.pushpull(0, v, out=v)
+np.testing.assert_allclose(v.asnumpy(), np.full((8,), 1.5))
+b = mx.nd.array(np.full((4,), float(10 * (rank + 1)), dtype='float32'))
+kv.broadcast(1, b, b)
+np.testing.assert_allclose(b.asnumpy(), np.full((4,), 10.0))
+print('GLOO_BRIDGE_OK', rank)
+'''
+    procs = []
+    env_base = dict(os.environ,
+                    MASTER_ADDR='127.0.0.1', MASTER_PORT='29753',
+                    WORLD_SIZE='2')
+    for r in range(2):
+        env = dict(env_base, RANK=str(r), LOCAL_RANK=str(r))
+        procs.append(subprocess.Popen(
+            [sys.executable, '-c', script], env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=180)
+        outs.append(out.decode())
+    for r, (p, out) in enumerate(zip(procs, outs)):
+        assert p.returncode == 0, f'rank {r} failed:\n{out}'
+        assert f'GLOO_BRIDGE_OK {r}' in out
