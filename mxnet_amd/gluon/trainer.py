@@ -520,7 +520,10 @@ class Trainer:
             self._check_states(i, p)
         dev_states = []
         for s, p in zip(blob['states'], self._params):
-            dev = p.list_data()[0]._t.device
+            w = p.list_data()[0]
+            dev = None if w.is_native else w._t.device
+            if w.is_native:
+                dev = w.context.torch_device
             dev_states.append(_state_to_device(s, dev))
         self._states = dev_states
         self._states_init = [True] * len(self._params)
@@ -528,8 +531,12 @@ class Trainer:
 
 
 def _state_to_cpu(s):
+    from ..ndarray.ndarray import NDArray as _ND
     if isinstance(s, torch.Tensor):
         return s.cpu()
+    if isinstance(s, _ND):
+        # native states pickle as tagged numpy (see _state_to_device)
+        return ('__nd__', s.asnumpy(), str(s.dtype))
     if isinstance(s, tuple):
         return tuple(_state_to_cpu(x) for x in s)
     return s
@@ -539,5 +546,10 @@ def _state_to_device(s, dev):
     if isinstance(s, torch.Tensor):
         return s.to(dev)
     if isinstance(s, tuple):
+        if len(s) == 3 and s[0] == '__nd__':
+            from ..ndarray.ndarray import array as _arr
+            from ..context import Context
+            ctx = Context.from_torch(dev) if dev is not None else None
+            return _arr(s[1], ctx=ctx, dtype=s[2])
         return tuple(_state_to_device(x, dev) for x in s)
     return s

@@ -530,3 +530,31 @@ def test_native_export_symbolblock_roundtrip(native):
     out = sb(x)
     assert out.is_native
     np.testing.assert_allclose(out.asnumpy(), ref, rtol=1e-5, atol=1e-6)
+
+
+def test_native_trainer_state_roundtrip(native):
+    """Trainer.save_states/load_states on the native runtime: momentum
+    buffers pickle as tagged numpy, reload onto the right context, and
+    training continues (checkpoint/resume parity)."""
+    import tempfile
+    net = nn.Dense(4)
+    net.initialize()
+    x = mx.nd.array(np.random.RandomState(0).randn(3, 5).astype('float32'))
+    tr = Trainer(net.collect_params(), 'sgd',
+                 {'learning_rate': 0.1, 'momentum': 0.9}, kvstore=None)
+    with autograd.record():
+        L = (net(x) ** 2).sum()
+    L.backward()
+    tr.step(1)
+    s0 = tr._states[0]
+    mom = (s0[1] if isinstance(s0, tuple) else s0).asnumpy()
+    f = tempfile.mktemp()
+    tr.save_states(f)
+    tr.load_states(f)
+    s0 = tr._states[0]
+    np.testing.assert_allclose(
+        (s0[1] if isinstance(s0, tuple) else s0).asnumpy(), mom)
+    with autograd.record():
+        L = (net(x) ** 2).sum()
+    L.backward()
+    tr.step(1)  # training continues on restored state
