@@ -110,6 +110,10 @@ def _read_ndarray(f):
     n = int(_np.prod(shape)) if shape else 1
     raw = _read_exact(f, dtype.itemsize * n)
     arr = _np.frombuffer(raw, dtype=dtype).reshape(shape).copy()
+    from ..base import native_mode
+    if native_mode():
+        from ..ndarray.ndarray import array as _mk
+        return _mk(arr, dtype=str(dtype))
     return NDArray(torch.from_numpy(arr))
 
 
