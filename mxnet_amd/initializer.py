@@ -98,7 +98,12 @@ class Constant(Initializer):
 
     def _init_weight(self, name, arr):
         with torch.no_grad():
-            if hasattr(self.value, '_t'):
+            if hasattr(self.value, 'is_native') and self.value.is_native:
+                # native source (e.g. loaded checkpoint arrays in native
+                # mode): bridge through numpy into the init buffer
+                src = torch.from_numpy(self.value.asnumpy())
+                arr._t.copy_(src.to(arr._t.device, arr._t.dtype))
+            elif hasattr(self.value, '_t'):
                 arr._t.copy_(self.value._t.to(arr._t.device, arr._t.dtype))
             else:
                 arr._t.fill_(float(self.value))
