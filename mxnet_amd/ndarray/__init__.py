@@ -7,6 +7,7 @@ from . import ops
 from .ndarray import concat, stack  # keep creation-module versions authoritative
 
 from . import contrib  # noqa: F401
+from . import random  # noqa: F401
 from . import sparse  # noqa: F401
 
 

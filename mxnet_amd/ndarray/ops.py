@@ -695,13 +695,37 @@ def sequence_mask(*args, **kwargs):
 
 # --- random ------------------------------------------------------------------
 
+def _native_rand(op, shape, dtype, ctx, attrs):
+    import random as _random
+    from .. import _core
+    from ..base import core_flag
+    from .ndarray import _core_ctx
+    if isinstance(shape, int):
+        shape = (shape,)
+    dt, di = _core_ctx(ctx)
+    attrs = dict(attrs)
+    attrs.update({'shape': '(' + ','.join(str(s) for s in shape) + ',)',
+                  'dtype': str(core_flag(dtype)),
+                  'seed': str(_random.getrandbits(48)),
+                  '__ctx_gpu__': str(di) if dt == 2 else '-1'})
+    return NDArray(_core.invoke(op, [], attrs)[0])
+
+
 def random_uniform(low=0.0, high=1.0, shape=(1,), dtype='float32', ctx=None, **kwargs):
+    from ..base import native_mode
+    if native_mode():
+        return _native_rand('_random_uniform', shape, dtype, ctx,
+                            {'low': str(low), 'high': str(high)})
     from ..context import current_context
     dev = (ctx or current_context()).torch_device
     return NDArray(torch.empty(shape, dtype=torch_dtype(dtype), device=dev).uniform_(low, high))
 
 
 def random_normal(loc=0.0, scale=1.0, shape=(1,), dtype='float32', ctx=None, **kwargs):
+    from ..base import native_mode
+    if native_mode():
+        return _native_rand('_random_normal', shape, dtype, ctx,
+                            {'loc': str(loc), 'scale': str(scale)})
     from ..context import current_context
     dev = (ctx or current_context()).torch_device
     return NDArray(torch.empty(shape, dtype=torch_dtype(dtype), device=dev).normal_(loc, scale))
