@@ -34,7 +34,9 @@ class _WorkerDataset(torch.utils.data.Dataset):
 
         def unwrap(x):
             if isinstance(x, NDArray):
-                return x._t
+                # native samples cross the process boundary as numpy
+                # (torch CPU tensors ride shm reductions either way)
+                return torch.as_tensor(x.asnumpy()) if x.is_native else x._t
             if isinstance(x, tuple):
                 return tuple(unwrap(i) for i in x)
             return x
