@@ -691,9 +691,27 @@ class NDArray:
                                  'beta': str(float(a_max))})
         return NDArray(self._t.clamp(a_min, a_max))
 
+    # pickling (DataLoader workers / checkpoint helpers): numpy
+    # round-trip onto the same backend and context
+    def __reduce__(self):
+        return (_rebuild_ndarray,
+                (self.asnumpy(), str(self.dtype),
+                 self.context.device_type, self.context.device_id,
+                 self.is_native))
+
     def __repr__(self):
         return '%s\n<NDArray %s @%s>' % (
             str(self.asnumpy()), 'x'.join(map(str, self.shape)), self.context)
+
+
+def _rebuild_ndarray(arr, dtype, dev_type, dev_id, was_native):
+    from ..base import set_native
+    prev = set_native(was_native)
+    try:
+        ctx = Context(dev_type, dev_id)
+        return array(arr, ctx=ctx, dtype=dtype)
+    finally:
+        set_native(prev)
 
 
 # ---------------------------------------------------------------------------

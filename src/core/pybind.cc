@@ -202,7 +202,26 @@ PYBIND11_MODULE(_core, m) {
            [](const NDArray& src, const NDArray& dst) {
              py::gil_scoped_release rel;
              CopyFromTo(src, dst);
-           });
+           })
+      // pickle via numpy round-trip (multiprocessing DataLoader workers
+      // ship samples/batches through ForkingPickler; GPU arrays
+      // re-materialize on the SAME device id in the consumer)
+      .def(py::pickle(
+          [](const NDArray& a) {
+            return py::make_tuple(ToNumpy(a), a.ctx().dev_type,
+                                  a.ctx().dev_id, a.dtype());
+          },
+          [](py::tuple t) {
+            NDArray host = FromNumpy(t[0].cast<py::array>(), 1, 0);
+            int dev_type = t[1].cast<int>();
+            int dev_id = t[2].cast<int>();
+            int dtype = t[3].cast<int>();
+            if (host.dtype() != dtype) host = host.AsType(dtype);
+            if (dev_type == 1) return host;
+            NDArray dst(host.shape(), MakeCtx(dev_type, dev_id), dtype);
+            CopyFromTo(host, dst);
+            return dst;
+          }));
 
   m.def("from_numpy", &FromNumpy, py::arg("array"), py::arg("dev_type") = 1,
         py::arg("dev_id") = 0);

@@ -558,3 +558,24 @@ def test_native_trainer_state_roundtrip(native):
         L = (net(x) ** 2).sum()
     L.backward()
     tr.step(1)  # training continues on restored state
+
+
+def test_native_pickle_and_mp_dataloader(native):
+    """Native NDArrays pickle (numpy round-trip, same backend/context),
+    so the multiprocessing DataLoader ships native batches across the
+    worker boundary."""
+    import pickle
+    x = mx.nd.array(np.arange(6.).reshape(2, 3))
+    y = pickle.loads(pickle.dumps(x))
+    assert y.is_native
+    np.testing.assert_array_equal(y.asnumpy(), x.asnumpy())
+    from mxnet_amd.gluon.data import ArrayDataset, DataLoader
+    rs = np.random.RandomState(0)
+    X = rs.randn(32, 6).astype('float32')
+    yl = rs.randn(32).astype('float32')
+    loader = DataLoader(ArrayDataset(X, yl), batch_size=8, num_workers=2)
+    tot = 0
+    for bx, by in loader:
+        assert bx.is_native
+        tot += bx.shape[0]
+    assert tot == 32
