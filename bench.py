@@ -178,6 +178,11 @@ def main():
 
     if distributed:
         import torch.distributed as dist
+        if not dist.is_initialized():
+            # host-side gloo group for the timing barrier/max-reduce —
+            # the native runtime's RCCL binding owns the GPU collectives
+            # and must not share a second device communicator
+            dist.init_process_group('gloo')
 
     for _ in range(args.warmup):
         step()
