@@ -579,3 +579,16 @@ def test_native_pickle_and_mp_dataloader(native):
         assert bx.is_native
         tot += bx.shape[0]
     assert tot == 32
+
+
+def test_native_random_creation(native):
+    """nd.random uniform/normal run the native philox kernels with
+    correct moments and bounds."""
+    u = mx.nd.random.uniform(low=2.0, high=5.0, shape=(2000,))
+    assert u.is_native
+    a = u.asnumpy()
+    assert a.min() >= 2.0 and a.max() <= 5.0
+    assert abs(a.mean() - 3.5) < 0.2
+    n = mx.nd.random.normal(loc=1.0, scale=2.0, shape=(20000,))
+    b = n.asnumpy()
+    assert abs(b.mean() - 1.0) < 0.1 and abs(b.std() - 2.0) < 0.1
