@@ -60,6 +60,11 @@ class KVStoreBase:
 def create(name='local'):
     from ..base import native_mode
     if isinstance(name, str) and name == 'dist_sync_ps':
+        if native_mode():
+            raise ValueError(
+                "kvstore 'dist_sync_ps' runs on the torch frontend "
+                "(gloo send/recv internals); with MXNET_NATIVE_RUNTIME "
+                "use 'dist_sync' (own RCCL / gloo-bridged all-reduce)")
         return SyncPSKVStore(name)
     if native_mode() and isinstance(name, str) and (
             name.startswith('dist') or name in ('nccl', 'device')):
