@@ -480,20 +480,32 @@ def transpose(data, axes=None, **kwargs):
 
 
 def expand_dims(data, axis, **kwargs):
+    if data.is_native:
+        return data.expand_dims(axis)
     return NDArray(_t(data).unsqueeze(axis))
 
 
 def squeeze(data, axis=None, **kwargs):
+    if data.is_native:
+        return data.squeeze(axis)
     return NDArray(_t(data).squeeze() if axis is None else _t(data).squeeze(axis))
 
 
 def stack(*data, axis=0, **kwargs):
     if len(data) == 1 and isinstance(data[0], (list, tuple)):
         data = data[0]
+    if data and getattr(data[0], 'is_native', False):
+        from .ndarray import stack as _stk
+        return _stk(list(data), axis=axis)
     return NDArray(torch.stack([_t(d) for d in data], dim=axis))
 
 
 def split(data, num_outputs, axis=1, squeeze_axis=False, **kwargs):
+    if data.is_native:
+        outs = data.split(num_outputs, axis=axis)
+        if squeeze_axis:
+            outs = [o.squeeze(axis=axis) for o in outs]
+        return outs if len(outs) > 1 else outs[0]
     outs = torch.chunk(_t(data), num_outputs, dim=axis)
     if squeeze_axis:
         outs = [o.squeeze(axis) for o in outs]
@@ -502,6 +514,13 @@ def split(data, num_outputs, axis=1, squeeze_axis=False, **kwargs):
 
 
 def slice(data, begin, end, step=None, **kwargs):
+    if data.is_native:
+        sl = []
+        for i in range(len(begin)):
+            sl.append(builtins.slice(
+                begin[i], end[i],
+                step[i] if step and step[i] is not None else None))
+        return data[tuple(sl)]
     x = _t(data)
     sl = []
     for i in range(len(begin)):
@@ -517,6 +536,12 @@ def slice_axis(data, axis, begin, end, **kwargs):
 
 
 def slice_like(data, shape_like, axes=None, **kwargs):
+    if data.is_native:
+        axes2 = axes or range(len(shape_like.shape))
+        sl = [builtins.slice(None)] * len(data.shape)
+        for ax in axes2:
+            sl[ax] = builtins.slice(0, shape_like.shape[ax])
+        return data[tuple(sl)]
     x, ref = _t(data), _t(shape_like)
     axes = axes or range(ref.dim())
     sl = [builtins.slice(None)] * x.dim()
