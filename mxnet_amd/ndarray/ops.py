@@ -240,74 +240,108 @@ def RNN(data, parameters, state, state_cell=None, mode='lstm',
 
 # --- elementwise / math ------------------------------------------------------
 
-def _unary(fn):
+def _unary(fn, native_op=None, native_fn=None, host_fn=None):
+    """Unary factory.  Native routing in priority order: a registry
+    kernel (`native_op`), an nd-composed expression (`native_fn`), or a
+    documented host numpy fallback (`host_fn`, for the exotic tail the
+    reference also ran on generic kernels)."""
     def op(data, **kwargs):
+        if getattr(data, 'is_native', False):
+            if native_op is not None:
+                return _ninv(native_op, [data], {})
+            if native_fn is not None:
+                return native_fn(data)
+            if host_fn is not None:
+                from .ndarray import array as _mk
+                return _mk(host_fn(data.asnumpy()), ctx=data.context)
+            raise RuntimeError(
+                'op has no native-runtime path (torch frontend only)')
         return NDArray(fn(_t(data)))
     return op
 
 
-exp = _unary(torch.exp)
-log = _unary(torch.log)
-log2 = _unary(torch.log2)
-log10 = _unary(torch.log10)
-log1p = _unary(torch.log1p)
-expm1 = _unary(torch.expm1)
-sqrt = _unary(torch.sqrt)
-rsqrt = _unary(torch.rsqrt)
-cbrt = _unary(lambda x: torch.sign(x) * torch.abs(x) ** (1.0 / 3))
-square = _unary(torch.square)
-abs = _unary(torch.abs)
-sign = _unary(torch.sign)
-floor = _unary(torch.floor)
-ceil = _unary(torch.ceil)
-round = _unary(torch.round)
-trunc = _unary(torch.trunc)
-rint = _unary(torch.round)
-fix = _unary(torch.trunc)
-sin = _unary(torch.sin)
-cos = _unary(torch.cos)
-tan = _unary(torch.tan)
-arcsin = _unary(torch.asin)
-arccos = _unary(torch.acos)
-arctan = _unary(torch.atan)
-sinh = _unary(torch.sinh)
-cosh = _unary(torch.cosh)
-tanh = _unary(torch.tanh)
-arcsinh = _unary(torch.asinh)
-arccosh = _unary(torch.acosh)
-arctanh = _unary(torch.atanh)
-sigmoid = _unary(torch.sigmoid)
-erf = _unary(torch.erf)
-erfinv = _unary(torch.erfinv)
-gamma = _unary(lambda x: torch.exp(torch.lgamma(x)))
-gammaln = _unary(torch.lgamma)
-relu = _unary(torch.relu)
-negative = _unary(torch.neg)
-reciprocal = _unary(torch.reciprocal)
-logical_not = _unary(lambda x: (~x.bool()).to(x.dtype))
+import numpy as _onp
+
+exp = _unary(torch.exp, native_op='exp')
+log = _unary(torch.log, native_op='log')
+log2 = _unary(torch.log2,
+              native_fn=lambda x: x.log() * (1.0 / _onp.log(2.0)))
+log10 = _unary(torch.log10,
+               native_fn=lambda x: x.log() * (1.0 / _onp.log(10.0)))
+log1p = _unary(torch.log1p, native_fn=lambda x: (x + 1.0).log())
+expm1 = _unary(torch.expm1, native_fn=lambda x: x.exp() - 1.0)
+sqrt = _unary(torch.sqrt, native_op='sqrt')
+rsqrt = _unary(torch.rsqrt, native_fn=lambda x: 1.0 / x.sqrt())
+cbrt = _unary(lambda x: torch.sign(x) * torch.abs(x) ** (1.0 / 3),
+              host_fn=_onp.cbrt)
+square = _unary(torch.square, native_op='square')
+abs = _unary(torch.abs, native_op='abs')
+sign = _unary(torch.sign,
+              native_fn=lambda x: (x > x * 0.0) - (x < x * 0.0))
+floor = _unary(torch.floor, host_fn=_onp.floor)
+ceil = _unary(torch.ceil, host_fn=_onp.ceil)
+round = _unary(torch.round, host_fn=_onp.round)
+trunc = _unary(torch.trunc, host_fn=_onp.trunc)
+rint = _unary(torch.round, host_fn=_onp.rint)
+fix = _unary(torch.trunc, host_fn=_onp.fix)
+sin = _unary(torch.sin, host_fn=_onp.sin)
+cos = _unary(torch.cos, host_fn=_onp.cos)
+tan = _unary(torch.tan, host_fn=_onp.tan)
+arcsin = _unary(torch.asin, host_fn=_onp.arcsin)
+arccos = _unary(torch.acos, host_fn=_onp.arccos)
+arctan = _unary(torch.atan, host_fn=_onp.arctan)
+sinh = _unary(torch.sinh, host_fn=_onp.sinh)
+cosh = _unary(torch.cosh, host_fn=_onp.cosh)
+tanh = _unary(torch.tanh, native_op='tanh')
+arcsinh = _unary(torch.asinh, host_fn=_onp.arcsinh)
+arccosh = _unary(torch.acosh, host_fn=_onp.arccosh)
+arctanh = _unary(torch.atanh, host_fn=_onp.arctanh)
+sigmoid = _unary(torch.sigmoid, native_op='sigmoid')
+erf = _unary(torch.erf,
+             host_fn=lambda a: __import__('scipy.special',
+                                          fromlist=['erf']).erf(a))
+erfinv = _unary(torch.erfinv,
+                host_fn=lambda a: __import__('scipy.special',
+                                             fromlist=['erfinv']).erfinv(a))
+gamma = _unary(lambda x: torch.exp(torch.lgamma(x)),
+               host_fn=lambda a: __import__('scipy.special',
+                                            fromlist=['gamma']).gamma(a))
+gammaln = _unary(torch.lgamma,
+                 host_fn=lambda a: __import__('scipy.special',
+                                              fromlist=['gammaln'])
+                 .gammaln(a))
+relu = _unary(torch.relu, native_op='relu')
+negative = _unary(torch.neg, native_op='negative')
+reciprocal = _unary(torch.reciprocal, native_fn=lambda x: 1.0 / x)
+logical_not = _unary(lambda x: (~x.bool()).to(x.dtype),
+                     native_fn=lambda x: x == (x * 0.0))
 
 
-def _binary(fn):
+def _binary(fn, native_op=None):
     def op(lhs, rhs, **kwargs):
+        if getattr(lhs, 'is_native', False):
+            if native_op is None:
+                raise RuntimeError(
+                    'op has no native-runtime path (torch frontend only)')
+            r = rhs if isinstance(rhs, NDArray) else lhs * 0.0 + float(rhs)
+            return _ninv(native_op, [lhs, r], {})
         return NDArray(fn(_t(lhs), _t(rhs)))
     return op
 
 
-elemwise_add = _binary(torch.add)
-elemwise_sub = _binary(torch.sub)
-elemwise_mul = _binary(torch.mul)
-elemwise_div = _binary(torch.div)
-broadcast_add = _binary(torch.add)
-broadcast_sub = _binary(torch.sub)
-broadcast_mul = _binary(torch.mul)
-broadcast_div = _binary(torch.div)
-broadcast_power = _binary(torch.pow)
-broadcast_maximum = _binary(torch.maximum)
-broadcast_minimum = _binary(torch.minimum)
+elemwise_add = _binary(torch.add, 'elemwise_add')
+elemwise_sub = _binary(torch.sub, 'elemwise_sub')
+elemwise_mul = _binary(torch.mul, 'elemwise_mul')
+elemwise_div = _binary(torch.div, 'elemwise_div')
+broadcast_add = _binary(torch.add, 'elemwise_add')
+broadcast_sub = _binary(torch.sub, 'elemwise_sub')
+broadcast_mul = _binary(torch.mul, 'elemwise_mul')
+broadcast_div = _binary(torch.div, 'elemwise_div')
+broadcast_power = _binary(torch.pow, 'power')
+broadcast_maximum = _binary(torch.maximum, 'maximum')
+broadcast_minimum = _binary(torch.minimum, 'minimum')
 broadcast_mod = _binary(torch.remainder)
-maximum = _binary(torch.maximum)
-minimum = _binary(torch.minimum)
-power = _binary(torch.pow)
+power = _binary(torch.pow, 'power')
 hypot = _binary(torch.hypot)
 broadcast_equal = _binary(lambda a, b: (a == b).to(a.dtype))
 broadcast_not_equal = _binary(lambda a, b: (a != b).to(a.dtype))
