@@ -592,3 +592,29 @@ def test_native_random_creation(native):
     n = mx.nd.random.normal(loc=1.0, scale=2.0, shape=(20000,))
     b = n.asnumpy()
     assert abs(b.mean() - 1.0) < 0.1 and abs(b.std() - 2.0) < 0.1
+
+
+def test_native_unary_binary_families(native):
+    """The generated unary/binary op families route natively: registry
+    kernels, composed expressions, or documented host fallbacks."""
+    F = mx.nd.ops
+    x = mx.nd.array(np.array([0.2, 0.5, 0.9], np.float32))
+    np.testing.assert_allclose(F.log1p(x).asnumpy(),
+                               np.log1p([0.2, 0.5, 0.9]), rtol=1e-6)
+    np.testing.assert_allclose(F.rsqrt(x).asnumpy(),
+                               1 / np.sqrt([0.2, 0.5, 0.9]), rtol=1e-5)
+    np.testing.assert_allclose(F.sin(x).asnumpy(),
+                               np.sin([0.2, 0.5, 0.9]), rtol=1e-6)
+    np.testing.assert_allclose(
+        F.sign(mx.nd.array(np.array([-2., 0., 3.]))).asnumpy(), [-1, 0, 1])
+    np.testing.assert_allclose(
+        F.broadcast_mul(x, mx.nd.array(np.array([2.0], np.float32)))
+        .asnumpy(), [0.4, 1.0, 1.8], rtol=1e-6)
+    np.testing.assert_allclose(
+        F.logical_not(mx.nd.array(np.array([0., 1., 2.]))).asnumpy(),
+        [1, 0, 0])
+    np.testing.assert_allclose(F.reciprocal(x).asnumpy(),
+                               1 / np.array([0.2, 0.5, 0.9]), rtol=1e-6)
+    import scipy.special as sp
+    np.testing.assert_allclose(F.erf(x).asnumpy(),
+                               sp.erf([0.2, 0.5, 0.9]), rtol=1e-5)
