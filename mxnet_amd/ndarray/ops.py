@@ -98,6 +98,23 @@ def Activation(data, act_type='relu', **kwargs):
 
 
 def LeakyReLU(data, act_type='leaky', slope=0.25, **kwargs):
+    if data.is_native:
+        if act_type == 'leaky':
+            return _ninv('leaky_relu', [data], {'alpha': str(slope)})
+        if act_type == 'gelu':
+            return _ninv('gelu', [data], {})
+        if act_type == 'elu':
+            # elu(x) = x>0 ? x : slope*(exp(x)-1), composed
+            pos = _ninv('relu', [data], {})
+            neg = (((data * -1.0)._invoke('relu', [(data * -1.0)])
+                    * -1.0).exp() - 1.0) * slope
+            mask = (data < data * 0.0).astype(str(data.dtype))                 if str(data.dtype) != 'float32' else (data < data * 0.0)
+            return pos + neg * mask
+        if act_type == 'selu':
+            a, l = 1.6732632423543772, 1.0507009873554805
+            e = LeakyReLU(data, act_type='elu', slope=a)
+            return e * l
+        raise ValueError(act_type)
     x = _t(data)
     if act_type == 'leaky':
         return NDArray(torch.nn.functional.leaky_relu(x, slope))
