@@ -730,6 +730,7 @@ def _core_ctx(ctx):
 def _native_full(shape, val, ctx, dtype):
     if isinstance(shape, int):
         shape = (shape,)
+    shape = tuple(int(s) for s in shape)  # TypeError on non-int dims
     dt, di = _core_ctx(ctx)
     at = {'shape': '(' + ','.join(str(s) for s in shape) + ')',
           'value': str(float(val)), 'dtype': str(core_flag(dtype))}

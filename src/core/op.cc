@@ -173,6 +173,20 @@ std::vector<NDArray> Imperative::Run(const OpEntry* op, const NodeAttrs& attrs,
   std::vector<TShape> out_shapes;
   std::vector<int> out_dtypes;
   op->infer(attrs, in_shapes, in_dtypes, &out_shapes, &out_dtypes);
+  // guard against corrupted shape attrs (e.g. a stringified array
+  // leaking into a shape tuple): negative dims or numel overflow would
+  // otherwise travel into the allocator/kernels
+  for (auto& sh : out_shapes) {
+    __int128 numel = 1;
+    for (auto d : sh) {
+      MX_CHECK(d >= 0 && d <= (int64_t(1) << 40),
+               "op '" << op->name << "': inferred dim " << d
+                      << " out of range (bad shape attr?)");
+      numel *= d;
+    }
+    MX_CHECK(numel <= (__int128(1) << 44),
+             "op '" << op->name << "': inferred tensor too large");
+  }
   Context ctx = inputs.empty() ? Context::CPU() : inputs[0].ctx();
   if (attrs.has("__ctx_gpu__")) {  // source ops (zeros/random) carry ctx
     int id = (int)attrs.GetInt("__ctx_gpu__", -1);
