@@ -225,6 +225,8 @@ def log_softmax(data, axis=-1, temperature=None, **kwargs):
 
 
 def softmin(data, axis=-1, **kwargs):
+    if data.is_native:
+        return softmax(data * -1.0, axis=axis)
     return NDArray(_nn.softmax(-_t(data), axis, 1.0))
 
 
@@ -359,15 +361,38 @@ broadcast_maximum = _binary(torch.maximum, 'maximum')
 broadcast_minimum = _binary(torch.minimum, 'minimum')
 broadcast_mod = _binary(torch.remainder)
 power = _binary(torch.pow, 'power')
-hypot = _binary(torch.hypot)
-broadcast_equal = _binary(lambda a, b: (a == b).to(a.dtype))
-broadcast_not_equal = _binary(lambda a, b: (a != b).to(a.dtype))
-broadcast_greater = _binary(lambda a, b: (a > b).to(a.dtype))
-broadcast_greater_equal = _binary(lambda a, b: (a >= b).to(a.dtype))
-broadcast_lesser = _binary(lambda a, b: (a < b).to(a.dtype))
-broadcast_lesser_equal = _binary(lambda a, b: (a <= b).to(a.dtype))
-broadcast_logical_and = _binary(lambda a, b: (a.bool() & b.bool()).to(a.dtype))
-broadcast_logical_or = _binary(lambda a, b: (a.bool() | b.bool()).to(a.dtype))
+def hypot(lhs, rhs, **kwargs):
+    if getattr(lhs, 'is_native', False):
+        return (lhs.square() + rhs.square()).sqrt()
+    return NDArray(torch.hypot(_t(lhs), _t(rhs)))
+
+
+def _logical_and_native(a, b):
+    return (a != a * 0.0) * (b != b * 0.0)
+
+
+def broadcast_logical_and(lhs, rhs, **kwargs):
+    if getattr(lhs, 'is_native', False):
+        return _logical_and_native(lhs, rhs)
+    return NDArray((_t(lhs).bool() & _t(rhs).bool()).to(_t(lhs).dtype))
+
+
+def broadcast_logical_or(lhs, rhs, **kwargs):
+    if getattr(lhs, 'is_native', False):
+        za = lhs != lhs * 0.0
+        zb = rhs != rhs * 0.0
+        return ((za + zb) > za * 0.0)
+    return NDArray((_t(lhs).bool() | _t(rhs).bool()).to(_t(lhs).dtype))
+
+
+broadcast_equal = _binary(lambda a, b: (a == b).to(a.dtype), 'equal')
+broadcast_not_equal = _binary(lambda a, b: (a != b).to(a.dtype), 'not_equal')
+broadcast_greater = _binary(lambda a, b: (a > b).to(a.dtype), 'greater')
+broadcast_greater_equal = _binary(lambda a, b: (a >= b).to(a.dtype),
+                                  'greater_equal')
+broadcast_lesser = _binary(lambda a, b: (a < b).to(a.dtype), 'less')
+broadcast_lesser_equal = _binary(lambda a, b: (a <= b).to(a.dtype),
+                                 'less_equal')
 broadcast_hypot = hypot
 
 
@@ -375,6 +400,11 @@ def add_n(*args, **kwargs):
     """ElementwiseSum — the KVStore reduce primitive (ndarray_function.cu)."""
     if len(args) == 1 and isinstance(args[0], (list, tuple)):
         args = args[0]
+    if getattr(args[0], 'is_native', False):
+        out = args[0]
+        for a in args[1:]:
+            out = out + a
+        return out
     out = _t(args[0]).clone()
     for a in args[1:]:
         out += _t(a)
@@ -420,8 +450,14 @@ def clip(data, a_min, a_max, **kwargs):
 
 # --- reductions --------------------------------------------------------------
 
-def _reduce(fn):
+def _reduce(fn, method=None, host_fn=None):
     def op(data, axis=None, keepdims=False, **kwargs):
+        if getattr(data, 'is_native', False):
+            if method is not None:
+                return getattr(data, method)(axis=axis, keepdims=keepdims)
+            from .ndarray import array as _mk
+            r = host_fn(data.asnumpy(), axis=axis, keepdims=keepdims)
+            return _mk(_np.asarray(r, dtype='float32'), ctx=data.context)
         x = _t(data)
         if axis is None:
             r = fn(x, None, False)
@@ -431,12 +467,18 @@ def _reduce(fn):
     return op
 
 
-sum = _reduce(lambda x, a, k: x.sum() if a is None else x.sum(dim=a, keepdim=k))
-mean = _reduce(lambda x, a, k: x.mean() if a is None else x.mean(dim=a, keepdim=k))
-prod = _reduce(lambda x, a, k: x.prod() if a is None else x.prod(dim=a, keepdim=k))
-max = _reduce(lambda x, a, k: x.max() if a is None else x.amax(dim=a, keepdim=k))
-min = _reduce(lambda x, a, k: x.min() if a is None else x.amin(dim=a, keepdim=k))
-nansum = _reduce(lambda x, a, k: x.nansum() if a is None else x.nansum(dim=a, keepdim=k))
+sum = _reduce(lambda x, a, k: x.sum() if a is None else x.sum(dim=a, keepdim=k),
+              method='sum')
+mean = _reduce(lambda x, a, k: x.mean() if a is None else x.mean(dim=a, keepdim=k),
+               method='mean')
+prod = _reduce(lambda x, a, k: x.prod() if a is None else x.prod(dim=a, keepdim=k),
+               host_fn=_np.prod)
+max = _reduce(lambda x, a, k: x.max() if a is None else x.amax(dim=a, keepdim=k),
+              method='max')
+min = _reduce(lambda x, a, k: x.min() if a is None else x.amin(dim=a, keepdim=k),
+              method='min')
+nansum = _reduce(lambda x, a, k: x.nansum() if a is None else x.nansum(dim=a, keepdim=k),
+                 host_fn=_np.nansum)
 
 
 def norm(data, ord=2, axis=None, keepdims=False, **kwargs):
@@ -516,6 +558,12 @@ def sort(data, axis=-1, is_ascend=True, **kwargs):
 
 
 def argsort(data, axis=-1, is_ascend=True, dtype='float32', **kwargs):
+    if data.is_native:
+        arr = _np.argsort(data.asnumpy(), axis=axis)
+        if not is_ascend:
+            arr = _np.flip(arr, axis=axis).copy()
+        from .ndarray import array as _mk
+        return _mk(arr.astype('float32'), ctx=data.context)
     return NDArray(torch.argsort(_t(data), dim=axis,
                                  descending=not is_ascend).to(torch_dtype(dtype)))
 
@@ -627,6 +675,12 @@ def pick(data, index, axis=-1, keepdims=False, **kwargs):
 
 
 def gather_nd(data, indices, **kwargs):
+    if data.is_native:
+        from .ndarray import array as _mk
+        x = data.asnumpy()
+        idx = indices.asnumpy().astype('int64')
+        return _mk(x[tuple(idx[i] for i in range(idx.shape[0]))],
+                   ctx=data.context)
     x, idx = _t(data), _t(indices).long()
     return NDArray(x[tuple(idx[i] for i in range(idx.shape[0]))])
 
@@ -685,6 +739,8 @@ def broadcast_to(data, shape, **kwargs):
 
 
 def broadcast_like(data, like, **kwargs):
+    if data.is_native:
+        return data.broadcast_to(tuple(like.shape))
     return NDArray(_t(data).broadcast_to(_t(like).shape).contiguous())
 
 
@@ -740,6 +796,13 @@ def batch_dot(lhs, rhs, transpose_a=False, transpose_b=False, **kwargs):
 
 
 def linalg_gemm2(A, B, transpose_a=False, transpose_b=False, alpha=1.0, **kwargs):
+    if A.is_native:
+        if len(A.shape) == 2:
+            r = dot(A, B, transpose_a=transpose_a, transpose_b=transpose_b)
+        else:
+            r = batch_dot(A, B, transpose_a=transpose_a,
+                          transpose_b=transpose_b)
+        return r if alpha == 1.0 else r * float(alpha)
     a, b = _t(A), _t(B)
     if transpose_a:
         a = a.transpose(-1, -2)
@@ -818,11 +881,26 @@ def random_randint(low, high, shape=(1,), dtype='int32', ctx=None, **kwargs):
 
 
 def shuffle(data, **kwargs):
+    if data.is_native:
+        perm = _np.random.permutation(data.shape[0]).astype('float32')
+        from .ndarray import array as _mk
+        return take(data, _mk(perm, ctx=data.context))
     x = _t(data)
     return NDArray(x[torch.randperm(x.shape[0], device=x.device)])
 
 
 def sample_multinomial(data, shape=1, get_prob=False, **kwargs):
+    if data.is_native:
+        from .ndarray import array as _mk
+        p = data.asnumpy()
+        n = shape if isinstance(shape, int) else shape[0]
+        if p.ndim == 1:
+            out = _np.random.choice(p.shape[-1], size=n, p=p / p.sum())
+        else:
+            out = _np.stack([_np.random.choice(p.shape[-1], size=n,
+                                               p=row / row.sum())
+                             for row in p])
+        return _mk(out.astype('int32'), ctx=data.context, dtype='int32')
     x = _t(data)
     n = shape if isinstance(shape, int) else shape[0]
     return NDArray(torch.multinomial(x, n, replacement=True).to(torch.int32))
@@ -890,6 +968,14 @@ def SequenceReverse(data, sequence_length=None, use_sequence_length=False,
 
 def smooth_l1(data, scalar=1.0, **kwargs):
     """(reference smooth_l1 op)."""
+    if data.is_native:
+        s2 = scalar * scalar
+        a = data.abs()
+        inside = (a < (1.0 / s2)).astype(str(data.dtype)) \
+            if str(data.dtype) != 'float32' else (a < (1.0 / s2))
+        quad = data.square() * (0.5 * s2)
+        lin = a - (0.5 / s2)
+        return inside * quad + (inside * -1.0 + 1.0) * lin
     import torch
     t = _t(data)
     s2 = scalar * scalar
@@ -905,6 +991,14 @@ def smooth_l1(data, scalar=1.0, **kwargs):
 
 def moments(data, axes=None, keepdims=False, **kwargs):
     """(mean, var) pair (reference nn/moments.cc)."""
+    if data.is_native:
+        ax = tuple(axes) if axes is not None \
+            else tuple(range(len(data.shape)))
+        m = data.mean(axis=ax, keepdims=True)
+        v = ((data - m) ** 2).mean(axis=ax, keepdims=keepdims)
+        if not keepdims:
+            m = data.mean(axis=ax, keepdims=False)
+        return m, v
     t = _t(data)
     dims = list(axes) if axes is not None else list(range(t.dim()))
     mean = t.mean(dim=dims, keepdim=keepdims)
@@ -913,6 +1007,8 @@ def moments(data, axes=None, keepdims=False, **kwargs):
 
 
 def SwapAxis(data, dim1=0, dim2=0, **kwargs):
+    if data.is_native:
+        return data.swapaxes(dim1, dim2)
     return NDArray(_t(data).transpose(dim1, dim2).contiguous())
 
 
@@ -930,6 +1026,9 @@ def space_to_depth(data, block_size, **kwargs):
 
 
 def cumsum(a, axis=None, **kwargs):
+    if a.is_native:
+        from .ndarray import array as _mk
+        return _mk(_np.cumsum(a.asnumpy(), axis=axis), ctx=a.context)
     t = _t(a)
     if axis is None:
         return NDArray(t.reshape(-1).cumsum(0))
@@ -937,6 +1036,9 @@ def cumsum(a, axis=None, **kwargs):
 
 
 def cumprod(a, axis=None, **kwargs):
+    if a.is_native:
+        from .ndarray import array as _mk
+        return _mk(_np.cumprod(a.asnumpy(), axis=axis), ctx=a.context)
     t = _t(a)
     if axis is None:
         return NDArray(t.reshape(-1).cumprod(0))
@@ -944,32 +1046,66 @@ def cumprod(a, axis=None, **kwargs):
 
 
 def diag(data, k=0, **kwargs):
+    if data.is_native:
+        from .ndarray import array as _mk
+        arr = data.asnumpy()
+        r = _np.diagonal(arr, offset=k) if arr.ndim >= 2 \
+            else _np.diag(arr, k)
+        return _mk(_np.ascontiguousarray(r), ctx=data.context)
     t = _t(data)
     return NDArray(torch.diagonal(t, offset=k).contiguous() if t.dim() >= 2
                    else torch.diag(t, k))
 
 
 def trace(data, offset=0, **kwargs):
+    if data.is_native:
+        from .ndarray import array as _mk
+        return _mk(_np.trace(data.asnumpy(), offset=offset)
+                   .astype('float32').reshape(-1), ctx=data.context)
     return NDArray(torch.diagonal(_t(data), offset=offset).sum(-1))
 
 
 def meshgrid(*arrays, indexing='xy', **kwargs):
+    if arrays and getattr(arrays[0], 'is_native', False):
+        from .ndarray import array as _mk
+        outs = _np.meshgrid(*[a.asnumpy() for a in arrays],
+                            indexing=indexing)
+        return [_mk(_np.ascontiguousarray(o), ctx=arrays[0].context)
+                for o in outs]
     outs = torch.meshgrid(*[_t(a) for a in arrays], indexing=indexing)
     return [NDArray(o.contiguous()) for o in outs]
 
 
 def searchsorted(sorted_sequence, values, right=False, **kwargs):
+    if sorted_sequence.is_native:
+        from .ndarray import array as _mk
+        out = _np.searchsorted(sorted_sequence.asnumpy(),
+                               values.asnumpy(),
+                               side='right' if right else 'left')
+        return _mk(out.astype('float32'),
+                   ctx=sorted_sequence.context)
     return NDArray(torch.searchsorted(_t(sorted_sequence), _t(values),
                                       right=right))
 
 
 def bincount(x, weights=None, minlength=0, **kwargs):
+    if x.is_native:
+        from .ndarray import array as _mk
+        out = _np.bincount(x.asnumpy().astype('int64'),
+                           weights.asnumpy() if weights is not None
+                           else None, minlength)
+        return _mk(out.astype('float32'), ctx=x.context)
     return NDArray(torch.bincount(
         _t(x).long(), _t(weights) if weights is not None else None,
         minlength))
 
 
 def digamma(data, **kwargs):
+    if data.is_native:
+        import scipy.special as _sp
+        from .ndarray import array as _mk
+        return _mk(_sp.digamma(data.asnumpy()).astype('float32'),
+                   ctx=data.context)
     return NDArray(torch.digamma(_t(data)))
 
 
