@@ -618,3 +618,36 @@ def test_native_unary_binary_families(native):
     import scipy.special as sp
     np.testing.assert_allclose(F.erf(x).asnumpy(),
                                sp.erf([0.2, 0.5, 0.9]), rtol=1e-5)
+
+
+def test_native_reduce_and_tail_ops(native):
+    """The fn-form reductions and tail ops route natively (sum/mean/
+    max/min via the reduce kernels; host fallbacks numerically exact)."""
+    F = mx.nd.ops
+    rs = np.random.RandomState(0)
+    xa = rs.rand(4, 6).astype('float32') + 0.1
+    ya = rs.rand(4, 6).astype('float32') + 0.1
+    x, y = mx.nd.array(xa), mx.nd.array(ya)
+    np.testing.assert_allclose(F.sum(x, axis=1).asnumpy(), xa.sum(1),
+                               rtol=1e-5)
+    np.testing.assert_allclose(F.mean(x, axis=0).asnumpy(), xa.mean(0),
+                               rtol=1e-5)
+    np.testing.assert_allclose(F.max(x, axis=1).asnumpy(), xa.max(1))
+    np.testing.assert_allclose(F.cumsum(x, axis=1).asnumpy(),
+                               np.cumsum(xa, 1), rtol=1e-5)
+    m, v = F.moments(x, axes=(0,))
+    np.testing.assert_allclose(v.asnumpy(), xa.var(0), rtol=1e-4)
+    np.testing.assert_allclose(F.hypot(x, y).asnumpy(), np.hypot(xa, ya),
+                               rtol=1e-5)
+    np.testing.assert_allclose(
+        F.smooth_l1(x, 1.0).asnumpy(),
+        np.where(np.abs(xa) < 1, 0.5 * xa * xa, np.abs(xa) - 0.5),
+        rtol=1e-5)
+    np.testing.assert_allclose(F.add_n(x, y).asnumpy(), xa + ya,
+                               rtol=1e-6)
+    np.testing.assert_allclose(F.broadcast_greater(x, y).asnumpy(),
+                               (xa > ya).astype('f'))
+    np.testing.assert_allclose(F.swapaxes(x, 0, 1).asnumpy(), xa.T)
+    np.testing.assert_allclose(
+        F.linalg_gemm2(x, y, transpose_b=True).asnumpy(), xa @ ya.T,
+        rtol=1e-4)
