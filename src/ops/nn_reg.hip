@@ -229,6 +229,9 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[0].size() == 2 && is[1].size() == 2 &&
+                     is[0][0] == is[1][0],
+                 "dot_tn_fused: needs [M,I]x[M,J]");
         os->assign(1, TShape{is[0][1], is[1][1]});
         os->push_back({is[0][1]});
         ot->assign(1, it[0]);
@@ -289,6 +292,9 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[0].size() == 2 && is[1].size() == 2 &&
+                     is[0][0] == is[1][0],
+                 "dot_tn: needs [M,I]x[M,J]");
         os->assign(1, TShape{is[0][1], is[1][1]});
         ot->assign(1, it[0]);
       })
@@ -314,6 +320,11 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[0].size() == 3 && is[1].size() == 3,
+                 "batch_dot: inputs must be 3-D [B,M,K]x[B,K,N], got "
+                     << is[0].size() << "-D and " << is[1].size() << "-D");
+        MX_CHECK(is[0][0] == is[1][0] && is[0][2] == is[1][1],
+                 "batch_dot: shape mismatch");
         os->assign(1, TShape{is[0][0], is[0][1], is[1][2]});
         ot->assign(1, it[0]);
       })
@@ -476,6 +487,8 @@ bool _registered_nn = [] {
         // executing path supports it (mirrors conv2d_fwd_raw routing:
         // MFMA igemm always; 1x1-s1 GEMM unless split-K engages); the
         // following BatchNorm consumes it and skips its reduce pass
+        MX_CHECK(is[0].size() == 4 && is[1].size() == 4,
+                 "Convolution: NHWC x [K,R,S,C] 4-D tensors required");
         bool want = a.GetBool("want_stats", false);
         int64_t C = is[0][3], K = is[1][0], R = is[1][1], S = is[1][2];
         int groups = (int)a.GetInt("num_group", 1);
@@ -682,6 +695,7 @@ bool _registered_nn = [] {
         auto stride = a.GetTuple("stride", kernel);
         auto pad = a.GetTuple("pad", {0, 0});
         bool gp = a.GetBool("global_pool", false);
+        MX_CHECK(is[0].size() == 4, "Pooling: NHWC 4-D input required");
         int64_t kh = gp ? is[0][1] : kernel[0], kw = gp ? is[0][2] : kernel[1];
         int64_t sh = gp ? 1 : stride[0], sw = gp ? 1 : stride[1];
         int64_t ph = gp ? 0 : pad[0], pw = gp ? 0 : pad[1];
@@ -1389,6 +1403,7 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs&, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[1].size() == 2, "Embedding: weight must be 2-D");
         TShape s = is[0];
         s.push_back(is[1][1]);
         os->assign(1, s);
@@ -1670,8 +1685,11 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[0].size() == 3 && is[0][2] % 3 == 0,
+                 "interleaved_attention: qkv must be [B,S,3U]");
         int64_t B = is[0][0], S = is[0][1], U = is[0][2] / 3;
         int64_t H = a.GetInt("heads", 1);
+        MX_CHECK(H > 0 && U % H == 0, "interleaved_attention: bad heads");
         os->assign(1, TShape{B, S, U});
         os->push_back({B * H, S, S});
         os->push_back({B * H, S, S});
