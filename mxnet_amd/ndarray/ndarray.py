@@ -371,6 +371,12 @@ class NDArray:
                 i = int(k)
                 if i < 0:
                     i += shp[dim]
+                if not 0 <= i < shp[dim]:
+                    # IndexError also terminates python's iteration
+                    # protocol (for d in arr) correctly
+                    raise IndexError(
+                        f'index {k} out of bounds for axis {dim} '
+                        f'with size {shp[dim]}')
                 offset += i * istr[dim]
             elif isinstance(k, slice):
                 start, stop, step = k.indices(shp[dim])

@@ -88,6 +88,12 @@ inline Strides8 make_strides(const TShape& out, const TShape& a,
   for (int i = no - 1; i >= 0; --i) {
     st.shape[i] = out[i];
     int ia = i - (no - na), ib = i - (no - nb);
+    MX_CHECK(ia < 0 || a[ia] == 1 || a[ia] == out[i],
+             "broadcast: dim " << (ia >= 0 ? a[ia] : 1)
+                               << " incompatible with " << out[i]);
+    MX_CHECK(ib < 0 || b[ib] == 1 || b[ib] == out[i],
+             "broadcast: dim " << (ib >= 0 ? b[ib] : 1)
+                               << " incompatible with " << out[i]);
     if (ia >= 0 && a[ia] != 1) {
       st.s0[i] = sa;
       sa *= a[ia];

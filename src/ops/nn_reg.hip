@@ -383,8 +383,18 @@ bool _registered_nn = [] {
       .infer([](const NodeAttrs& a, const std::vector<TShape>& is,
                 const std::vector<int>& it, std::vector<TShape>* os,
                 std::vector<int>* ot) {
+        MX_CHECK(is[1].size() == 2 && is[0].size() >= 1,
+                 "FullyConnected: weight must be [units,K] 2-D");
         bool flatten = a.GetBool("flatten", true);
         int64_t units = is[1][0];
+        int64_t K = is[1][1];
+        int64_t inK = 1;
+        if (flatten)
+          for (size_t i = 1; i < is[0].size(); ++i) inK *= is[0][i];
+        else
+          inK = is[0].empty() ? 0 : is[0][is[0].size() - 1];
+        MX_CHECK(inK == K, "FullyConnected: input feature dim " << inK
+                               << " != weight K " << K);
         TShape s;
         if (flatten) {
           s = {is[0][0], units};
@@ -398,6 +408,7 @@ bool _registered_nn = [] {
       .gpu([](const NodeAttrs& a, const OpCtx& o, V in, V out) {
         Arr x = in[0];
         long K = in[1].shape[1];
+        if (K == 0 || out[0].size() == 0) return;
         x.shape = {x.numel() / K, K};
         Arr y = out[0];
         y.shape = {x.shape[0], in[1].shape[0]};
@@ -408,6 +419,7 @@ bool _registered_nn = [] {
         CPU_FLOAT_ONLY(in[0], "FullyConnected");
         TBlob x = in[0];
         long K = in[1].shape[1];
+        if (K == 0 || out[0].size() == 0) return;
         x.shape = {x.size() / K, K};
         const float* bias = in.size() > 2 ? (const float*)in[2].dptr : nullptr;
         cpu_gemm_nt(x, in[1], bias, out[0]);
@@ -889,6 +901,7 @@ bool _registered_nn = [] {
         bool relu = a.GetBool("fuse_relu", false);
         float mom = (float)a.GetFloat("momentum", 0.9);
         float eps = (float)a.GetFloat("eps", 1e-5);
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], M = in[0].size() / C;
         const float* x = (const float*)in[0].dptr;
         const float* g = (const float*)in[1].dptr;
@@ -976,6 +989,7 @@ bool _registered_nn = [] {
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         bool relu = a.GetBool("fuse_relu", false);
         bool has_res = a.GetBool("has_res", false);
+        if (in[1].size() == 0) return;
         long C = in[1].shape[in[1].ndim() - 1], M = in[1].size() / C;
         const float* dyv = (const float*)in[0].dptr;
         const float* x = (const float*)in[1].dptr;
@@ -1031,6 +1045,7 @@ bool _registered_nn = [] {
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         CPU_FLOAT_ONLY(in[0], "LayerNorm");
         float eps = (float)a.GetFloat("eps", 1e-5);
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
         const float* x = (const float*)in[0].dptr;
         const float* g = (const float*)in[1].dptr;
@@ -1078,6 +1093,7 @@ bool _registered_nn = [] {
                           out[1], out[2]);
       })
       .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        if (in[1].size() == 0) return;
         long C = in[1].shape[in[1].ndim() - 1], rows = in[1].size() / C;
         const float* dyv = (const float*)in[0].dptr;
         const float* x = (const float*)in[1].dptr;
@@ -1118,6 +1134,7 @@ bool _registered_nn = [] {
         .cpu([log_mode](const NodeAttrs& a, const OpCtx&, V in, V out) {
           CPU_FLOAT_ONLY(in[0], "softmax");
           float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+          if (in[0].size() == 0) return;
           long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
           const float* x = (const float*)in[0].dptr;
           float* y = (float*)out[0].dptr;
@@ -1155,6 +1172,7 @@ bool _registered_nn = [] {
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         CPU_FLOAT_ONLY(in[0], "masked_softmax");
         float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
         const float* x = (const float*)in[0].dptr;
         const unsigned char* mk = (const unsigned char*)in[1].dptr;
@@ -1189,6 +1207,7 @@ bool _registered_nn = [] {
       .cpu([](const NodeAttrs& a, const OpCtx&, V in, V out) {
         bool log_mode = a.GetBool("log_mode", false);
         float invT = (float)(1.0 / a.GetFloat("temperature", 1.0));
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
         const float* dyv = (const float*)in[0].dptr;
         const float* y = (const float*)in[1].dptr;
@@ -1217,6 +1236,7 @@ bool _registered_nn = [] {
       })
       .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
         MX_CHECK(in[1].dtype == kInt64, "pick index must be int64");
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
         MXC_DISPATCH_FLOAT(in[0].dtype, "pick", {
           pick_kernel<scalar_t><<<grid_for(rows), kBlock, 0,
@@ -1228,6 +1248,7 @@ bool _registered_nn = [] {
       })
       .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
         MX_CHECK(in[1].dtype == kInt64, "pick index must be int64");
+        if (in[0].size() == 0) return;
         long C = in[0].shape[in[0].ndim() - 1], rows = in[0].size() / C;
         MXC_DISPATCH_FLOAT(in[0].dtype, "pick", {
           auto* x = (const scalar_t*)in[0].dptr;
@@ -1254,6 +1275,7 @@ bool _registered_nn = [] {
         ot->assign(1, it[0]);
       })
       .gpu([](const NodeAttrs&, const OpCtx& o, V in, V out) {
+        if (out[0].size() == 0) return;
         long C = out[0].shape[out[0].ndim() - 1], rows = out[0].size() / C;
         MXC_DISPATCH_FLOAT(out[0].dtype, "pick_bwd", {
           pick_grad_kernel<scalar_t><<<grid_for(rows * C), kBlock, 0,
@@ -1264,6 +1286,7 @@ bool _registered_nn = [] {
         HIP_CHECK_LAST();
       })
       .cpu([](const NodeAttrs&, const OpCtx&, V in, V out) {
+        if (out[0].size() == 0) return;
         long C = out[0].shape[out[0].ndim() - 1], rows = out[0].size() / C;
         MXC_DISPATCH_FLOAT(out[0].dtype, "pick_bwd", {
           auto* dy = (const scalar_t*)in[0].dptr;

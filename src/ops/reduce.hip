@@ -622,6 +622,12 @@ bool _registered_reduce = [] {
           int ax = (int)a.GetInt("axis", -1);
           if (ax < 0) ax += nd;
           long n = in[0].size(), len = in[0].shape[ax];
+          if (n == 0 || len == 0) {
+            MX_HIP_CALL(hipMemsetAsync(out[0].dptr, 0,
+                                       (size_t)out[0].size() * 4,
+                                       o.rc.stream));
+            return;
+          }
           long inner = 1;
           for (int d = ax + 1; d < nd; ++d) inner *= in[0].shape[d];
           long nout = n / len;
@@ -646,6 +652,10 @@ bool _registered_reduce = [] {
           int ax = (int)a.GetInt("axis", -1);
           if (ax < 0) ax += nd;
           long n = in[0].size(), len = in[0].shape[ax];
+          if (n == 0 || len == 0) {
+            memset(out[0].dptr, 0, (size_t)out[0].size() * 4);
+            return;
+          }
           long inner = 1;
           for (int d = ax + 1; d < nd; ++d) inner *= in[0].shape[d];
           long nout = n / len;
