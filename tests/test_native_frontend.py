@@ -655,13 +655,38 @@ def test_native_reduce_and_tail_ops(native):
 
 def test_native_op_surface_fuzz(native):
     """Every public single/two-array op either runs natively or raises
-    a clean error — never a crash; the torch-only tail must not grow."""
+    a clean error — never a crash (incl. empty/int64/zero-dim inputs);
+    the torch-only tail must not grow."""
     import inspect
     F = mx.nd.ops
     x = mx.nd.array(np.random.RandomState(0).rand(4, 6)
                     .astype('float32') + 0.1)
     y = mx.nd.array(np.random.RandomState(1).rand(4, 6)
                     .astype('float32') + 0.1)
+    # degenerate inputs must never crash the process
+    for adv in [mx.nd.array(np.zeros((0,), 'float32')),
+                mx.nd.array(np.arange(6).astype('int64')),
+                mx.nd.array(np.zeros((3, 0), 'float32'))]:
+        for name in sorted(dir(F)):
+            if name.startswith('_'):
+                continue
+            fn = getattr(F, name)
+            if not callable(fn):
+                continue
+            try:
+                sig = inspect.signature(fn)
+                nr = len([p for p in sig.parameters.values()
+                          if p.default is p.empty and p.kind not in
+                          (p.VAR_POSITIONAL, p.VAR_KEYWORD)])
+            except (ValueError, TypeError):
+                continue
+            try:
+                r = fn(adv) if nr <= 1 else \
+                    (fn(adv, adv) if nr == 2 else None)
+                if hasattr(r, 'asnumpy'):
+                    r.asnumpy()
+            except Exception:
+                pass
     allowed_torch_only = {
         'BilinearResize2D', 'BilinearSampler', 'GridGenerator', 'LRN',
         'SequenceLast', 'SequenceMask', 'SequenceReverse',
