@@ -718,3 +718,35 @@ def test_native_op_surface_fuzz(native):
             pass  # arg-shape errors etc. are fine
     unexpected = set(missing) - allowed_torch_only
     assert not unexpected, f'ops lost their native path: {unexpected}'
+
+
+def test_native_layer_sweep(native):
+    """Every basic Gluon layer runs forward+backward on the native
+    runtime (fuzz-style sweep with plausible inputs)."""
+    rs = np.random.RandomState(0)
+    x4 = mx.nd.array(rs.randn(2, 8, 8, 3).astype('float32'))
+    x2 = mx.nd.array(rs.randn(2, 12).astype('float32'))
+    xc = mx.nd.array(rs.randn(2, 4, 5, 5).astype('float32'))
+    cases = [
+        (nn.Dense(5), x2), (nn.Dense(5, flatten=True), x4),
+        (nn.Dropout(0.3), x2), (nn.BatchNorm(axis=-1), x4),
+        (nn.LayerNorm(), x2), (nn.GroupNorm(num_groups=1), xc),
+        (nn.InstanceNorm(), xc), (nn.Flatten(), x4),
+        (nn.Activation('relu'), x2), (nn.LeakyReLU(0.1), x2),
+        (nn.PReLU(), x2), (nn.ELU(), x2), (nn.SELU(), x2),
+        (nn.Swish(), x2), (nn.GELU(), x2),
+        (nn.Conv2D(4, 3, padding=1, layout='NHWC'), x4),
+        (nn.MaxPool2D(2, layout='NHWC'), x4),
+        (nn.AvgPool2D(2, layout='NHWC'), x4),
+        (nn.GlobalAvgPool2D(layout='NHWC'), x4),
+        (nn.Embedding(10, 6),
+         mx.nd.array(rs.randint(0, 10, (2, 3)), dtype='int64')),
+    ]
+    for layer, inp in cases:
+        layer.initialize()
+        with autograd.record():
+            y = layer(inp)
+            L = (y * y).sum()
+        L.backward()
+        v = L.asscalar()
+        assert v == v, type(layer).__name__
