@@ -501,6 +501,9 @@ void Engine::WaitForVar(VarId v) {
         lk, [&] { return var->queue.empty() && var->running_reads == 0; });
     exc = var->exc;
     var->exc = nullptr;
+    // an exception delivered here is CONSUMED: WaitForAll must not
+    // re-throw the same failure into unrelated later sync points
+    if (exc && impl_->global_exc_ == exc) impl_->global_exc_ = nullptr;
     ev = var->last_event;
   }
   if (ev && ev->ev) MX_HIP_CALL(hipEventSynchronize(ev->ev));
@@ -541,6 +544,7 @@ void Engine::Throw(VarId v) {
     if (it == impl_->vars_.end()) return;
     exc = it->second->exc;
     it->second->exc = nullptr;
+    if (exc && impl_->global_exc_ == exc) impl_->global_exc_ = nullptr;
   }
   if (exc) std::rethrow_exception(exc);
 }

@@ -372,3 +372,18 @@ def test_check_numeric_gradient_utility():
         check_numeric_gradient(g, [rs.randn(2, 5)])
     finally:
         set_native(prev)
+
+
+def test_consumed_exception_not_refired():
+    """An async op failure delivered at its own sync point (asnumpy /
+    wait_to_read) is CONSUMED — WaitForAll must not re-throw it into
+    unrelated later sync points (reference exception semantics)."""
+    import numpy as np
+    from mxnet_amd import _core
+    bad = _core.invoke('batch_dot', [
+        _core.from_numpy(np.ones((1, 2, 2), np.float16), 1, 0),
+        _core.from_numpy(np.ones((1, 2, 2), np.float16), 1, 0)], {})[0]
+    import pytest as _pt
+    with _pt.raises(RuntimeError):
+        bad.asnumpy()
+    _core.wait_all()  # clean — the failure was already delivered
