@@ -68,12 +68,25 @@ def main():
     # pre-flight the communicator before building the model so a
     # bootstrap failure degrades to the torch frontend instead of dying
     if native and distributed and on_gpu:
+        ok = 1
         try:
             from mxnet_amd import _core as _c0
             _c0.rccl_init(world, rank, local_rank)
         except Exception as e:
-            print(f'# native RCCL preflight failed ({e}); '
-                  f'falling back to the torch frontend', file=sys.stderr)
+            print(f'# native RCCL preflight failed ({e})', file=sys.stderr)
+            ok = 0
+        # consensus over a host-side gloo group: every rank must take
+        # the SAME runtime or the collectives deadlock
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            dist.init_process_group('gloo')
+        flag = torch.tensor([ok], dtype=torch.int64)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        if int(flag.item()) == 0:
+            if ok:
+                print('# a peer rank failed RCCL preflight; '
+                      'falling back to the torch frontend together',
+                      file=sys.stderr)
             native = False
 
     import mxnet_amd as mx

@@ -39,6 +39,24 @@ def main():
         args.batch_size, args.seq_len = 2, 32
 
     native = args.runtime == 'native'
+    # native+distributed: preflight the RCCL bootstrap with a consensus
+    # fallback (all ranks must take the same runtime — bench.py pattern)
+    if native and world > 1 and on_gpu:
+        ok = 1
+        try:
+            os.environ['MXNET_NATIVE_RUNTIME'] = '1'
+            from mxnet_amd import _core as _c0
+            _c0.rccl_init(world, rank, local_rank)
+        except Exception as e:
+            print(f'# native RCCL preflight failed ({e})', file=sys.stderr)
+            ok = 0
+        import torch.distributed as dist
+        if not dist.is_initialized():
+            dist.init_process_group('gloo')
+        flag = torch.tensor([ok], dtype=torch.int64)
+        dist.all_reduce(flag, op=dist.ReduceOp.MIN)
+        if int(flag.item()) == 0:
+            native = False
     os.environ['MXNET_NATIVE_RUNTIME'] = '1' if native else '0'
 
     import mxnet_amd as mx
