@@ -199,7 +199,9 @@ class Trainer:
                 if state['pending'] == 0:
                     flat = _flatten_dense_tensors(state['tensors'])
                     state['flat'] = flat
-                    state['handle'] = dist.all_reduce(flat, async_op=True)
+                    state['handle'] = dist.all_reduce(
+                        flat, group=self._kvstore._grp(flat),
+                        async_op=True)
                 elif state['pending'] < 0:
                     # a second backward before step(): the reduce already
                     # in flight used stale grads — re-reduce at drain time
@@ -226,7 +228,9 @@ class Trainer:
                     # unused this step): reduce synchronously now
                     flat = _flatten_dense_tensors(state['tensors'])
                     state['flat'] = flat
-                    state['handle'] = dist.all_reduce(flat, async_op=True)
+                    state['handle'] = dist.all_reduce(
+                        flat, group=self._kvstore._grp(flat),
+                        async_op=True)
             for state in self._ov_buckets:
                 state['handle'].wait()
                 if state.pop('stale', False):
@@ -236,7 +240,9 @@ class Trainer:
                     # the up-to-date grads.
                     flat = _flatten_dense_tensors(state['tensors'])
                     state['flat'] = flat
-                    state['handle'] = dist.all_reduce(flat, async_op=True)
+                    state['handle'] = dist.all_reduce(
+                        flat, group=self._kvstore._grp(flat),
+                        async_op=True)
                     state['handle'].wait()
                 flat = state['flat']
                 flat.div_(world)
@@ -271,7 +277,9 @@ class Trainer:
                 handles = []
                 for b in buckets:
                     flat = _flatten_dense_tensors(b)
-                    handles.append((dist.all_reduce(flat, async_op=True),
+                    handles.append((dist.all_reduce(
+                        flat, group=self._kvstore._grp(flat),
+                        async_op=True),
                                     flat, b))
                 for h, flat, b in handles:
                     h.wait()

@@ -307,6 +307,7 @@ class DistKVStore(KVStoreBase):
     def __init__(self, kind='dist_sync'):
         self._type = kind
         self._handles = []
+        self._pg = None  # non-default group when the default is gloo
         if not dist.is_initialized():
             # composite (cuda:nccl + cpu:gloo) so CPU tensors on a GPU
             # box still have a backend (RCCL handles the cuda ones)
@@ -319,6 +320,11 @@ class DistKVStore(KVStoreBase):
                 os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
                 os.environ.setdefault('MASTER_PORT', '29741')
                 dist.init_process_group(backend=backend, rank=0, world_size=1)
+        elif torch.cuda.is_available() and \
+                dist.get_backend() == 'gloo':
+            # a host-side gloo group was created earlier (bench timing /
+            # preflight consensus); GPU gradients need an RCCL group
+            self._pg = dist.new_group(backend='nccl')
         if torch.cuda.is_available():
             # clamp for CPU-tensor tests running world>1 on a 1-GPU box
             torch.cuda.set_device(int(os.environ.get('LOCAL_RANK', 0)) %
@@ -339,14 +345,18 @@ class DistKVStore(KVStoreBase):
     def num_workers(self):
         return dist.get_world_size()
 
+    def _grp(self, t):
+        # GPU tensors ride the RCCL subgroup when the default is gloo
+        return self._pg if (self._pg is not None and t.is_cuda) else None
+
     def init(self, key, value):
         # rank-0 value wins (reference: init broadcasts from root)
         v = value[0] if isinstance(value, (list, tuple)) else value
-        dist.broadcast(v._t.data, src=0)
+        dist.broadcast(v._t.data, src=0, group=self._grp(v._t))
 
     def broadcast(self, key, value, out, priority=0):
         v = value[0] if isinstance(value, (list, tuple)) else value
-        dist.broadcast(v._t.data, src=0)
+        dist.broadcast(v._t.data, src=0, group=self._grp(v._t))
         outs = out if isinstance(out, (list, tuple)) else [out]
         for o in outs:
             if o is not v:
@@ -362,7 +372,8 @@ class DistKVStore(KVStoreBase):
         if self._compression is not None:
             # lossy quantize-with-error-feedback before the collective
             self._compression.compress_decompress(key, t.data)
-        work = dist.all_reduce(t.data, op=dist.ReduceOp.SUM, async_op=async_op)
+        work = dist.all_reduce(t.data, op=dist.ReduceOp.SUM,
+                               group=self._grp(t), async_op=async_op)
         if async_op:
             self._handles.append(work)
         if out is not None:
