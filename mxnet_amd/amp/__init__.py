@@ -38,9 +38,13 @@ def scale_loss(loss, trainer):
                 trainer._amp_loss_scaler = scaler = LossScaler()
             self.scaler = scaler
             trainer._set_scale(1.0 / scaler.loss_scale)
+            def _scale(l):
+                if getattr(l, 'is_native', False):
+                    return l * float(scaler.loss_scale)
+                return NDArray(l._t * scaler.loss_scale)
             if isinstance(loss, (list, tuple)):
-                return [NDArray(l._t * scaler.loss_scale) for l in loss]
-            return NDArray(loss._t * scaler.loss_scale)
+                return [_scale(l) for l in loss]
+            return _scale(loss)
 
         def __exit__(self, *a):
             pass
@@ -56,7 +60,10 @@ def unscale(trainer):
     with torch.no_grad():
         for p in trainer._params:
             for g in p.list_grad():
-                g._t.mul_(inv)
+                if getattr(g, 'is_native', False):
+                    (g * inv).copyto(g)
+                else:
+                    g._t.mul_(inv)
     trainer._set_scale(1.0)
 
 

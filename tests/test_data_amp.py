@@ -139,3 +139,41 @@ def test_native_all_finite():
         assert all_finite([mx.nd.array(np.array([np.nan]))]) is False
     finally:
         set_native(prev)
+
+
+def test_native_amp_loop_and_overflow_skip():
+    """AMP dynamic loss scaling on the native runtime: scale_loss,
+    fused finiteness check, overflow skips the update and shrinks the
+    scale (round-1 ADVICE high-severity behavior, native path)."""
+    import numpy as np
+    import mxnet_amd as mx
+    from mxnet_amd import autograd, amp
+    from mxnet_amd.base import set_native
+    from mxnet_amd.gluon import nn, Trainer
+    prev = set_native(True)
+    try:
+        net = nn.Dense(4)
+        net.initialize()
+        tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.1},
+                     kvstore=None)
+        amp.init_trainer(tr)
+        x = mx.nd.array(np.random.RandomState(0).randn(3, 5)
+                        .astype('float32'))
+        for _ in range(2):
+            with autograd.record():
+                out = net(x)
+                L = (out * out).sum()
+                with amp.scale_loss(L, tr) as scaled:
+                    pass
+            scaled.backward()
+            tr.step(1)
+        w = net.weight.data(mx.cpu()).asnumpy().copy()
+        s0 = tr._amp_loss_scaler.loss_scale
+        g = net.weight.list_grad()[0]
+        mx.nd.array(np.full(g.shape, np.inf, 'float32')).copyto(g)
+        tr._update(False)
+        np.testing.assert_array_equal(
+            net.weight.data(mx.cpu()).asnumpy(), w)
+        assert tr._amp_loss_scaler.loss_scale < s0
+    finally:
+        set_native(prev)
