@@ -298,6 +298,10 @@ class Trainer:
         _SparseEmbedding.backward (None if the param is dense)."""
         if getattr(p, 'grad_stype', 'default') != 'row_sparse':
             return None
+        if p.list_data()[0].is_native:
+            # native runtime computes DENSE embedding grads (row-sparse
+            # lazy update is a torch-frontend feature) — nothing stashed
+            return None
         wt = p.list_data()[0]._t
         parts = getattr(wt, '_rowsparse_parts', None)
         if not parts:
