@@ -759,3 +759,22 @@ def test_native_layer_sweep(native):
         L.backward()
         v = L.asscalar()
         assert v == v, type(layer).__name__
+
+
+def test_native_sparse_grad_embedding_dense_fallback(native):
+    """sparse_grad=True embeddings train natively via correct dense
+    grads (row-sparse laziness is a torch-frontend optimization)."""
+    net = nn.Embedding(20, 4, sparse_grad=True)
+    net.initialize()
+    tr = Trainer(net.collect_params(), 'sgd', {'learning_rate': 0.5},
+                 kvstore=None)
+    idx = mx.nd.array(np.array([1, 3, 3]), dtype='int64')
+    w0 = net.weight.data(mx.cpu()).asnumpy().copy()
+    with autograd.record():
+        e = net(idx)
+        L = (e * e).sum()
+    L.backward()
+    tr.step(1)
+    w1 = net.weight.data(mx.cpu()).asnumpy()
+    assert not np.allclose(w1[1], w0[1])
+    np.testing.assert_array_equal(w1[0], w0[0])
