@@ -57,8 +57,14 @@ def export_model(sym_json, params, in_shapes, in_types=None, onnx_file=None,
             else json.loads(sym_json)
     if hasattr(sym_json, 'tojson'):
         sym_json = json.loads(sym_json.tojson())
-    params = {k.split(':', 1)[-1]: getattr(v, 'handle', v)
-              for k, v in params.items()}
+    def _to_torch(v):
+        if hasattr(v, 'is_native') and v.is_native:
+            # native-runtime arrays bridge through numpy
+            import numpy as _np
+            a = v.asnumpy()
+            return torch.from_numpy(_np.ascontiguousarray(a))
+        return getattr(v, 'handle', v)
+    params = {k.split(':', 1)[-1]: _to_torch(v) for k, v in params.items()}
 
     nodes = sym_json['nodes']
     out_name = {}                       # (node_idx, out_idx) -> onnx name

@@ -123,3 +123,34 @@ def test_onnx_embedding_gather(tmp_path):
                               [(2, 5)], in_types=[torch.int64])
     net2 = mxonnx.import_to_gluon(buf)
     np.testing.assert_allclose(net2(x).asnumpy(), y0, rtol=1e-4, atol=1e-5)
+
+
+def test_onnx_export_native_params():
+    """ONNX export accepts native-runtime parameter arrays (numpy
+    bridge) — a native-trained model exports and re-imports."""
+    import os
+    import tempfile
+    import numpy as np
+    import mxnet_amd as mx
+    from mxnet_amd.base import set_native
+    from mxnet_amd.gluon import nn
+    from mxnet_amd.contrib import onnx as onnx_mod
+    prev = set_native(True)
+    try:
+        net = nn.HybridSequential()
+        net.add(nn.Dense(8, activation='relu'), nn.Dense(4))
+        net.initialize()
+        x = mx.nd.array(np.random.RandomState(0).randn(2, 6)
+                        .astype('float32'))
+        net(x)
+        d = tempfile.mkdtemp()
+        pre = os.path.join(d, 'm')
+        net.export(pre)
+        params = mx.nd.load(pre + '-0000.params')
+        assert next(iter(params.values())).is_native
+        onnx_mod.export_model(pre + '-symbol.json', params, [(2, 6)],
+                              onnx_file=os.path.join(d, 'm.onnx'))
+        sym, arg, aux = onnx_mod.import_model(os.path.join(d, 'm.onnx'))
+        assert len(arg) == 4
+    finally:
+        set_native(prev)
