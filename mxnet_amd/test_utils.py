@@ -110,7 +110,11 @@ def check_consistency(fn, inputs, ctx_list=None, dtypes=None, rtol=None,
             from . import autograd
             with autograd.record():
                 out = fn(*nds)
-            out.backward(NDArray(torch.ones_like(out.handle)))
+            from .ndarray.ndarray import ones as _ones
+            out.backward(_ones(tuple(out.shape), ctx=ctx,
+                               dtype=str(out.dtype))
+                         if out.is_native
+                         else NDArray(torch.ones_like(out.handle)))
             grads = [nd.grad.asnumpy() for nd in nds]
         else:
             out = fn(*nds)
