@@ -6,7 +6,15 @@ from ....ndarray.ndarray import NDArray
 
 
 def _t(x):
-    return x._t if isinstance(x, NDArray) else x
+    if isinstance(x, NDArray):
+        if x.is_native:
+            # transforms are host-side preprocessing: bridge native
+            # samples through numpy (the batch returns to the active
+            # runtime via the loader's batchify)
+            import numpy as _np
+            return torch.from_numpy(_np.ascontiguousarray(x.asnumpy()))
+        return x._t
+    return x
 
 
 class Compose(Block):
