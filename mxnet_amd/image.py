@@ -16,7 +16,14 @@ __all__ = ['imresize', 'resize_short', 'fixed_crop', 'center_crop',
 
 
 def _t(x):
-    return x.handle if isinstance(x, NDArray) else x
+    if isinstance(x, NDArray):
+        if x.is_native:
+            # image preprocessing is host-side: bridge native arrays
+            # through numpy (results wrap back via NDArray(tensor))
+            import numpy as _np
+            return torch.from_numpy(_np.ascontiguousarray(x.asnumpy()))
+        return x.handle
+    return x
 
 
 def imread(*a, **k):
