@@ -204,6 +204,18 @@ class Parameter:
         if isinstance(ctx, Context):
             ctx = [ctx]
         cur = next(iter(self._data.values()))
+        if cur.is_native:
+            self._data = OrderedDict()
+            self._grad = OrderedDict()
+            for c in ctx:
+                nd = cur.as_in_context(c)
+                if nd is cur:
+                    nd = cur.copy()
+                self._data[c] = nd
+                if self.grad_req != 'null':
+                    nd.attach_grad(self.grad_req)
+                    self._grad[c] = nd.grad
+            return
         base = cur._t.detach()
         self._data = OrderedDict()
         self._grad = OrderedDict()
