@@ -79,7 +79,12 @@ def native_summary(top=30):
 
 
 def dumps(reset=False):
-    """Aggregate stats table as a string (reference aggregate_stats.cc)."""
+    """Aggregate stats table as a string (reference aggregate_stats.cc):
+    the native engine's per-op table when the native runtime is active,
+    else the kineto key-averages table."""
+    from .base import native_mode
+    if native_mode():
+        return native_summary()
     prof = _state['prof']
     if prof is None:
         return ''
