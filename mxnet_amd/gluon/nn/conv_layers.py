@@ -203,6 +203,11 @@ class Conv2DTranspose(HybridBlock):
             self.weight.shape = (in_c, self._channels // self._groups, kh, kw)
 
     def forward(self, x):
+        if getattr(x, 'is_native', False):
+            raise NotImplementedError(
+                '%s runs on the torch frontend (set_native(False)); the '
+                'native runtime covers NHWC Conv1D/2D via the MFMA igemm '
+                'kernels' % type(self).__name__)
         from ...ndarray.ndarray import NDArray
         from ...ops import nn as _onn
         self._finish_deferred(x)
@@ -265,6 +270,11 @@ class Conv3D(HybridBlock):
                              kd, kh, kw)
 
     def forward(self, x):
+        if getattr(x, 'is_native', False):
+            raise NotImplementedError(
+                '%s runs on the torch frontend (set_native(False)); the '
+                'native runtime covers NHWC Conv1D/2D via the MFMA igemm '
+                'kernels' % type(self).__name__)
         import torch.nn.functional as F
         from ...ndarray.ndarray import NDArray
         self._finish_deferred(x)
