@@ -778,3 +778,23 @@ def test_native_sparse_grad_embedding_dense_fallback(native):
     w1 = net.weight.data(mx.cpu()).asnumpy()
     assert not np.allclose(w1[1], w0[1])
     np.testing.assert_array_equal(w1[0], w0[0])
+
+
+def test_native_error_paths(native):
+    """Misuse raises clean errors (rank checks, layout asserts, clean
+    torch-frontend pointers) — never UB."""
+    x2 = mx.nd.array(np.ones((4, 6), 'float32'))
+    with pytest.raises(RuntimeError, match='3-D'):
+        mx.nd.ops.batch_dot(x2, x2).asnumpy()
+    with pytest.raises(AssertionError, match='NHWC'):
+        net = nn.Conv2D(4, 3, layout='NCHW')
+        net.initialize()
+        net(mx.nd.array(np.ones((1, 3, 6, 6), 'float32')))
+    with pytest.raises(NotImplementedError, match='torch frontend'):
+        net = nn.Conv3D(4, 3)
+        net.initialize()
+        net(mx.nd.array(np.ones((1, 2, 4, 4, 4), 'float32')))
+    with pytest.raises(IndexError):
+        x2[99]
+    with pytest.raises(RuntimeError, match='incompatible|broadcast'):
+        (x2 + mx.nd.array(np.ones((5, 7), 'float32'))).asnumpy()
