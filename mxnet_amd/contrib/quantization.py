@@ -231,6 +231,13 @@ def quantize_net(net, quantized_dtype='int8', exclude_layers=None,
     Without calib_data the layers fall back to per-batch dynamic
     abs-max.  Returns the (block, name, wrapper) list."""
     from ..gluon import nn
+    from ..base import native_mode
+    if native_mode():
+        raise NotImplementedError(
+            'quantize_net runs on the torch frontend '
+            '(set_native(False) / MXNET_NATIVE_RUNTIME=0): the int8 '
+            'inference wrappers drive the i8 MFMA kernels through '
+            'torch-tensor storage')
     swapped = []
 
     def visit(block):
